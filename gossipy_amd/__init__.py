@@ -263,8 +263,9 @@ class Cache:
         return value
 
     def clear(self) -> None:
-        """Drop every cached item."""
-        self._cache.clear()
+        """Drop every cached item (a previously exported ``get_cache`` dict
+        stays valid — the store is replaced, not emptied in place)."""
+        self._cache = {}
 
     def __getitem__(self, key: CacheKey) -> Any:
         slot = self._cache.get(key)

@@ -564,8 +564,12 @@ class KMeansHandler(ModelHandler):
         if self.matching == "naive":
             self.model = (self.model + other_model_handler.model) / 2
         else:
+            # bug-fix vs reference (gossipy/model/handler.py:629-630): the
+            # reference indexes with hungarian(cost)[0] (= row indices,
+            # an identity permutation), making the matching a no-op; the
+            # matched permutation is the COLUMN index vector.
             cost = torch.cdist(self.model, other_model_handler.model).cpu().numpy()
-            matching_idx = hungarian(cost)[0]
+            matching_idx = hungarian(cost)[1]
             self.model = (self.model + other_model_handler.model[matching_idx]) / 2
 
     def evaluate(self, data: Tuple[torch.Tensor, torch.Tensor]) -> Dict[str, float]:
