@@ -1,0 +1,472 @@
+"""Compute backends for the batched engine.
+
+Two implementations of one op surface:
+
+* :class:`TorchBackend` — plain PyTorch fp32 ops. This is the *semantic
+  oracle* (it mirrors the object-layer handlers batch-for-batch) and the
+  CPU execution path for tests. It is NOT allowed on a GPU device unless
+  explicitly forced — on MI355X the HIP kernels must be the path that runs.
+* :class:`HIPBackend` — the hand-written CDNA4 kernels from
+  ``gossipy_amd/ops`` (built for gfx950). Raises immediately if the
+  extension is missing.
+
+Semantics replicated from the reference (per-family):
+
+* merge = elementwise mean, age = max (gossipy/model/handler.py:260-280,
+  370-373);
+* ``CreateModelMode`` dispatch per received message
+  (gossipy/model/handler.py:117-136);
+* logreg update: ``local_epochs`` x minibatch SGD on
+  ``CrossEntropyLoss(sigmoid(Wx+b), y)`` exactly as the reference composes
+  it (gossipy/model/handler.py:250-258 + gossipy/model/nn.py:162-166);
+* pegasos per-sample: ``lr=1/(t*lam)``; shrink + hinge add
+  (gossipy/model/handler.py:416-423);
+* adaline per-sample delta rule (gossipy/model/handler.py:364-368).
+
+Minor documented divergence: the reference draws a fresh ``randperm`` per
+local epoch (gossipy/model/handler.py:243); the batched engine consumes
+minibatches in shard order. For shards that fit in one batch (every
+BASELINE config) the update math is identical because the batch gradient
+is order-invariant.
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Optional
+
+import numpy as np
+import torch
+
+from ..core import CreateModelMode
+from .arena import DataArena, NodeStateArena, SlotPool
+from .models import AdaLineSpec, LogRegSpec, MLPSpec, PegasosSpec
+from .rng import Purpose, RandomTape
+
+__all__ = ["TorchBackend", "HIPBackend", "make_backend"]
+
+_MODE_ID = {
+    CreateModelMode.UPDATE: 0,
+    CreateModelMode.MERGE_UPDATE: 1,
+    CreateModelMode.UPDATE_MERGE: 2,
+    CreateModelMode.PASS: 3,
+}
+
+
+class TorchBackend:
+    """Eager PyTorch implementation (CPU oracle)."""
+
+    name = "torch"
+
+    # -- init ----------------------------------------------------------------
+
+    def init_params(self, state: NodeStateArena, spec, tape: RandomTape) -> None:
+        """Per-node init. AdaLine/Pegasos start at zero
+        (gossipy/model/nn.py:131); logreg/mlp use torch's Linear default
+        (kaiming-uniform, bounds 1/sqrt(fan_in)) since the reference's
+        ``init_weights`` for LogisticRegression is a no-op
+        (gossipy/model/nn.py:169-170)."""
+        if spec.family in ("pegasos", "adaline"):
+            state.params.zero_()
+            return
+        g = tape.stream(Purpose.INIT)
+        n, D = state.params.shape
+        if spec.family == "logreg":
+            bound = 1.0 / np.sqrt(spec.d_in)
+            w = g.uniform(-bound, bound, size=(n, D))
+            state.params.copy_(torch.from_numpy(w).float().to(state.params.device))
+        elif spec.family == "mlp":
+            rows = np.empty((n, D), dtype=np.float32)
+            for (w_off, b_off, fin, fout) in spec.layer_offsets():
+                # xavier-uniform weights, zero bias (gossipy/model/nn.py:106-110)
+                bound = np.sqrt(6.0 / (fin + fout))
+                rows[:, w_off:b_off] = g.uniform(-bound, bound, size=(n, fout * fin))
+                rows[:, b_off : b_off + fout] = 0.0
+            state.params.copy_(torch.from_numpy(rows).to(state.params.device))
+        else:
+            raise ValueError(spec.family)
+        state.ages.zero_()
+
+    # -- snapshots -----------------------------------------------------------
+
+    def snapshot(
+        self,
+        state: NodeStateArena,
+        pool: SlotPool,
+        nodes: torch.Tensor,
+        slot_ids: torch.Tensor,
+    ) -> None:
+        """Arena row copy — the batched ``ModelHandler.caching``."""
+        pool.slots[slot_ids.long()] = state.params[nodes.long()]
+        pool.slot_ages[slot_ids.long()] = state.ages[nodes.long()]
+
+    # -- local updates -------------------------------------------------------
+
+    def update(
+        self,
+        state: NodeStateArena,
+        data: DataArena,
+        spec,
+        nodes: torch.Tensor,
+    ) -> None:
+        """Local training pass for ``nodes`` (used by init and by the
+        delivery loop)."""
+        if len(nodes) == 0:
+            return
+        if spec.family == "logreg":
+            self._update_logreg(state.params, state.ages, data, spec, nodes.long())
+        elif spec.family == "mlp":
+            self._update_mlp(state.params, state.ages, data, spec, nodes.long())
+        elif spec.family == "pegasos":
+            self._update_pegasos(state.params, state.ages, data, spec, nodes.long())
+        elif spec.family == "adaline":
+            self._update_adaline(state.params, state.ages, data, spec, nodes.long())
+        else:
+            raise ValueError(spec.family)
+
+    def _update_logreg(self, params, ages, data, spec: LogRegSpec, nodes) -> None:
+        d, k = spec.d_in, spec.n_classes
+        for idx in nodes.tolist():
+            c = int(data.counts[idx])
+            if c == 0:
+                continue
+            x = data.x[idx, :c]  # [c, d]
+            y = data.y[idx, :c].long()
+            W = params[idx, : k * d].view(k, d)
+            b = params[idx, k * d :]
+            bs = c if spec.batch_size == 0 else spec.batch_size
+            for _ in range(max(1, spec.local_epochs)):
+                for s in range(0, c, bs):
+                    xb, yb = x[s : s + bs], y[s : s + bs]
+                    m = xb.shape[0]
+                    z = xb @ W.t() + b
+                    a = torch.sigmoid(z)
+                    p = torch.softmax(a, dim=1)
+                    p[torch.arange(m), yb] -= 1.0
+                    dz = (p / m) * a * (1.0 - a)  # CE(softmax(a)) ∘ sigmoid'
+                    gW = dz.t() @ xb
+                    gb = dz.sum(0)
+                    if spec.weight_decay:
+                        gW += spec.weight_decay * W
+                    W -= spec.lr * gW
+                    b -= spec.lr * gb
+                    ages[idx] += 1
+
+    def _update_mlp(self, params, ages, data, spec: MLPSpec, nodes) -> None:
+        offs = spec.layer_offsets()
+        for idx in nodes.tolist():
+            c = int(data.counts[idx])
+            if c == 0:
+                continue
+            x = data.x[idx, :c]
+            y = data.y[idx, :c].long()
+            layers = [
+                (
+                    params[idx, w_off:b_off].view(fout, fin),
+                    params[idx, b_off : b_off + fout],
+                )
+                for (w_off, b_off, fin, fout) in offs
+            ]
+            bs = c if spec.batch_size == 0 else spec.batch_size
+            for _ in range(max(1, spec.local_epochs)):
+                for s in range(0, c, bs):
+                    xb, yb = x[s : s + bs], y[s : s + bs]
+                    m = xb.shape[0]
+                    acts = [xb]
+                    h = xb
+                    for li, (W, b) in enumerate(layers):
+                        h = h @ W.t() + b
+                        if li < len(layers) - 1:
+                            h = torch.relu(h)
+                        acts.append(h)
+                    p = torch.softmax(acts[-1], dim=1)
+                    p[torch.arange(m), yb] -= 1.0
+                    dh = p / m
+                    for li in reversed(range(len(layers))):
+                        W, b = layers[li]
+                        gW = dh.t() @ acts[li]
+                        gb = dh.sum(0)
+                        if li > 0:
+                            dh = (dh @ W) * (acts[li] > 0).float()
+                        if spec.weight_decay:
+                            gW += spec.weight_decay * W
+                        W -= spec.lr * gW
+                        b -= spec.lr * gb
+                    ages[idx] += 1
+
+    def _update_pegasos(self, params, ages, data, spec: PegasosSpec, nodes) -> None:
+        for idx in nodes.tolist():
+            c = int(data.counts[idx])
+            if c == 0:
+                continue
+            w = params[idx]
+            t = int(ages[idx])
+            for s in range(c):
+                t += 1
+                lr = 1.0 / (t * spec.lam)
+                xs = data.x[idx, s]
+                ys = float(data.y[idx, s])
+                pred = torch.dot(w, xs)
+                w *= 1.0 - lr * spec.lam
+                if float(pred) * ys < 1.0:
+                    w += lr * ys * xs
+            ages[idx] = t
+
+    def _update_adaline(self, params, ages, data, spec: AdaLineSpec, nodes) -> None:
+        for idx in nodes.tolist():
+            c = int(data.counts[idx])
+            if c == 0:
+                continue
+            w = params[idx]
+            for s in range(c):
+                xs = data.x[idx, s]
+                ys = float(data.y[idx, s])
+                err = ys - torch.dot(w, xs)
+                w += spec.lr * err * xs
+            ages[idx] += c
+
+    # -- deliveries ----------------------------------------------------------
+
+    def deliver(
+        self,
+        state: NodeStateArena,
+        pool: SlotPool,
+        data: DataArena,
+        spec,
+        recv_nodes: torch.Tensor,
+        recv_ptr: torch.Tensor,
+        del_slots: torch.Tensor,
+        reply_slots: torch.Tensor,
+    ) -> None:
+        """Per receiver, apply its deliveries in order: mode-dispatched
+        merge/update per message, then write the reply snapshot if the
+        message asked for one (PUSH_PULL)."""
+        mode = spec.mode
+        ptr = recv_ptr.tolist()
+        for i, node_t in enumerate(recv_nodes.tolist()):
+            node = torch.tensor([node_t])
+            for j in range(ptr[i], ptr[i + 1]):
+                slot = int(del_slots[j])
+                if mode == CreateModelMode.MERGE_UPDATE:
+                    self._merge_mean(state, pool, node_t, slot)
+                    self.update(state, data, spec, node)
+                elif mode == CreateModelMode.UPDATE:
+                    # train the RECEIVED model on local data, adopt it
+                    # (gossipy/model/handler.py:122-125)
+                    self._adopt_slot(state, pool, node_t, slot)
+                    self.update(state, data, spec, node)
+                elif mode == CreateModelMode.UPDATE_MERGE:
+                    self.update(state, data, spec, node)
+                    self._train_slot(state, pool, data, spec, node_t, slot)
+                    self._merge_mean(state, pool, node_t, slot)
+                elif mode == CreateModelMode.PASS:
+                    self._adopt_slot(state, pool, node_t, slot)
+                else:
+                    raise ValueError(mode)
+                r = int(reply_slots[j])
+                if r >= 0:
+                    self.snapshot(
+                        state,
+                        pool,
+                        torch.tensor([node_t], dtype=torch.long),
+                        torch.tensor([r], dtype=torch.long),
+                    )
+
+    def _merge_mean(self, state, pool, node: int, slot: int) -> None:
+        state.params[node] = (state.params[node] + pool.slots[slot]) * 0.5
+        state.ages[node] = max(int(state.ages[node]), int(pool.slot_ages[slot]))
+
+    def _adopt_slot(self, state, pool, node: int, slot: int) -> None:
+        state.params[node] = pool.slots[slot]
+        state.ages[node] = pool.slot_ages[slot]
+
+    def _train_slot(self, state, pool, data, spec, node: int, slot: int) -> None:
+        """UPDATE_MERGE's 'train the received model too' leg: run the update
+        on the slot copy in place (the merge then averages it in)."""
+        saved_p = state.params[node].clone()
+        saved_a = state.ages[node].clone()
+        state.params[node] = pool.slots[slot]
+        state.ages[node] = pool.slot_ages[slot]
+        self.update(state, data, spec, torch.tensor([node]))
+        pool.slots[slot] = state.params[node]
+        pool.slot_ages[slot] = state.ages[node]
+        state.params[node] = saved_p
+        state.ages[node] = saved_a
+
+    # -- evaluation ----------------------------------------------------------
+
+    def scores(
+        self, state: NodeStateArena, spec, nodes: torch.Tensor, X: torch.Tensor
+    ) -> torch.Tensor:
+        """Class scores ``[len(nodes), n_samples, k]`` of each node's model
+        on a shared input matrix (global eval set)."""
+        nodes = nodes.long()
+        if spec.family in ("pegasos", "adaline"):
+            w = state.params[nodes]  # [R, d]
+            s = X @ w.t()  # [n, R]
+            return s.t().unsqueeze(-1)  # [R, n, 1] margin scores
+        if spec.family == "logreg":
+            d, k = spec.d_in, spec.n_classes
+            W = state.params[nodes, : k * d].view(-1, k, d)
+            b = state.params[nodes, k * d :]
+            z = torch.einsum("nd,rkd->rnk", X, W) + b.unsqueeze(1)
+            return torch.sigmoid(z)
+        if spec.family == "mlp":
+            outs = []
+            for r in nodes.tolist():
+                h = X
+                for (w_off, b_off, fin, fout) in spec.layer_offsets():
+                    W = state.params[r, w_off:b_off].view(fout, fin)
+                    bb = state.params[r, b_off : b_off + fout]
+                    h = h @ W.t() + bb
+                    if b_off + fout < spec.D:
+                        h = torch.relu(h)
+                outs.append(h)
+            return torch.stack(outs)
+        raise ValueError(spec.family)
+
+
+class HIPBackend(TorchBackend):
+    """CDNA4 kernel implementation. Inherits the eval/score helpers (run as
+    regular torch-ROCm GPU ops) and overrides every per-tick hot op with the
+    hand-written gfx950 kernels from :mod:`gossipy_amd.ops`."""
+
+    name = "hip"
+
+    def __init__(self):
+        from .. import ops
+
+        self.ext = ops.load_extension()  # raises if the .so is missing
+
+    def snapshot(self, state, pool, nodes, slot_ids) -> None:
+        if len(nodes) == 0:
+            return
+        self.ext.snapshot(
+            state.params,
+            state.ages,
+            pool.slots,
+            pool.slot_ages,
+            nodes.to(state.params.device, torch.int32),
+            slot_ids.to(state.params.device, torch.int32),
+        )
+
+    def update(self, state, data, spec, nodes) -> None:
+        if len(nodes) == 0:
+            return
+        nodes_dev = nodes.to(state.params.device, torch.int32)
+        empty = torch.zeros(0, dtype=torch.int32, device=state.params.device)
+        self._dispatch(state, None, data, spec, nodes_dev, None, empty, empty, update_only=True)
+
+    def deliver(
+        self, state, pool, data, spec, recv_nodes, recv_ptr, del_slots, reply_slots
+    ) -> None:
+        if len(recv_nodes) == 0:
+            return
+        dev = state.params.device
+        self._dispatch(
+            state,
+            pool,
+            data,
+            spec,
+            recv_nodes.to(dev, torch.int32),
+            recv_ptr.to(dev, torch.int32),
+            del_slots.to(dev, torch.int32),
+            reply_slots.to(dev, torch.int32),
+            update_only=False,
+        )
+
+    def _dispatch(
+        self, state, pool, data, spec, nodes, recv_ptr, del_slots, reply_slots, update_only
+    ):
+        dev = state.params.device
+        if pool is None:
+            # update-only call: fabricate an empty pool of the right width
+            slots = torch.zeros(1, state.D, device=dev)
+            slot_ages = torch.zeros(1, device=dev, dtype=torch.int32)
+            recv_ptr = torch.zeros(len(nodes) + 1, dtype=torch.int32, device=dev)
+        else:
+            slots, slot_ages = pool.slots, pool.slot_ages
+        mode = _MODE_ID[spec.mode]
+        if spec.family == "logreg":
+            self.ext.tick_logreg(
+                state.params,
+                state.ages,
+                slots,
+                slot_ages,
+                nodes,
+                recv_ptr,
+                del_slots,
+                reply_slots,
+                data.x,
+                data.y,
+                data.counts,
+                spec.d_in,
+                spec.n_classes,
+                spec.lr,
+                spec.weight_decay,
+                max(1, spec.local_epochs),
+                spec.batch_size,
+                mode,
+                bool(update_only),
+            )
+        elif spec.family in ("pegasos", "adaline"):
+            self.ext.tick_linear(
+                state.params,
+                state.ages,
+                slots,
+                slot_ages,
+                nodes,
+                recv_ptr,
+                del_slots,
+                reply_slots,
+                data.x,
+                data.y,
+                data.counts,
+                spec.d_in,
+                spec.lam if spec.family == "pegasos" else spec.lr,
+                1 if spec.family == "pegasos" else 0,
+                mode,
+                bool(update_only),
+            )
+        elif spec.family == "mlp":
+            self.ext.tick_mlp(
+                state.params,
+                state.ages,
+                slots,
+                slot_ages,
+                nodes,
+                recv_ptr,
+                del_slots,
+                reply_slots,
+                data.x,
+                data.y,
+                data.counts,
+                torch.tensor(
+                    [x for t in spec.layer_offsets() for x in t],
+                    dtype=torch.int32,
+                    device=dev,
+                ),
+                len(spec.layer_offsets()),
+                spec.lr,
+                spec.weight_decay,
+                max(1, spec.local_epochs),
+                spec.batch_size,
+                mode,
+                bool(update_only),
+            )
+        else:
+            raise ValueError(spec.family)
+
+
+def make_backend(device: torch.device):
+    """HIP on GPU, torch on CPU. A GPU device without the extension is an
+    error — no silent eager fallback (set ``GOSSIPY_AMD_ALLOW_EAGER=1`` to
+    override for debugging only)."""
+    if device.type == "cuda":
+        try:
+            return HIPBackend()
+        except Exception:
+            if os.environ.get("GOSSIPY_AMD_ALLOW_EAGER") == "1":
+                return TorchBackend()
+            raise
+    return TorchBackend()

@@ -1,0 +1,346 @@
+"""Host-side event scheduler for the batched engine.
+
+The object layer discovers events one node at a time inside the timestep
+loop (gossipy/simul.py:389-430). The batched engine instead *precomputes*
+each round's full event schedule from the :class:`~gossipy_amd.engine.rng.
+RandomTape`: which nodes fire at which tick, their peer choices, drop coin
+flips, per-message delays, per-tick online masks, and the resulting delivery
+lists. The schedule is a pure function of ``(seed, config, round index)`` —
+every rank derives the same one, which is what makes node-residency (and GPU
+count) invisible to the simulation.
+
+Each tick becomes a :class:`TickPhase` with flat numpy arrays ready to be
+shipped to the GPU as kernel arguments:
+
+* ``snap_nodes / snap_slots`` — nodes that snapshot their model this tick
+  (the batched replacement of ``ModelHandler.caching``,
+  gossipy/model/handler.py:160-176 — an arena row copy instead of a
+  ``copy.deepcopy``);
+* delivery CSR (``recv_nodes / recv_ptr / del_slots``) — messages delivered
+  this tick, grouped by receiver so one workgroup applies one receiver's
+  merges+update sequentially while receivers run in parallel;
+* ``reply_*`` — the PULL/PUSH_PULL reply sub-phase (same structure).
+
+Snapshot slots are numbered sequentially per round in send order, so the
+slot pool is a dense ``[n_messages, D]`` arena and cross-GPU slot exchange
+is a gather/`ncclSend` per (src GPU, dst GPU) pair.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Tuple
+
+import numpy as np
+
+from ..core import AntiEntropyProtocol, ConstantDelay, Delay, LinearDelay, UniformDelay
+from .rng import Purpose, RandomTape
+
+__all__ = ["EngineConfig", "TickPhase", "RoundSchedule", "Scheduler"]
+
+
+@dataclass
+class EngineConfig:
+    """Static simulation parameters shared by every rank."""
+
+    n_nodes: int
+    delta: int  #: timesteps per round
+    protocol: AntiEntropyProtocol
+    model_size: int  #: scalars per model (message size accounting)
+    drop_prob: float = 0.0
+    online_prob: float = 1.0
+    delay: Delay = field(default_factory=lambda: ConstantDelay(0))
+    sync: bool = True
+    sampling_eval: float = 0.0
+    seed: int = 98765
+    #: adjacency CSR (None = fully connected); built once by the runner
+    peers_indptr: Optional[np.ndarray] = None
+    peers_indices: Optional[np.ndarray] = None
+
+
+@dataclass
+class TickPhase:
+    """All events of one timestep, as flat arrays (see module docstring)."""
+
+    t: int
+    # sub-phase A: snapshots of firing nodes (senders)
+    snap_nodes: np.ndarray  # int32 [S]
+    snap_slots: np.ndarray  # int32 [S]
+    # sub-phase B: model deliveries grouped by receiver (CSR)
+    recv_nodes: np.ndarray  # int32 [R]   unique receivers this tick
+    recv_ptr: np.ndarray  # int32 [R+1]
+    del_slots: np.ndarray  # int32 [recv_ptr[-1]]  slot per delivery, in order
+    # PUSH_PULL / PULL: deliveries that must also emit a reply snapshot
+    # (aligned with del_slots; -1 = no reply)
+    reply_slots: np.ndarray  # int32 [recv_ptr[-1]]
+    # PULL requests delivered this tick: the receiver snapshots (no merge)
+    pull_snap_nodes: np.ndarray  # int32
+    pull_snap_slots: np.ndarray  # int32
+    # sub-phase C: same-tick REPLY deliveries (delay-0 PUSH_PULL/PULL replies
+    # land in the reference's rep_queues[t] and are processed after the main
+    # deliveries, gossipy/simul.py:423-430); their slots are written during
+    # sub-phase B, so they need their own kernel launch.
+    rep_recv_nodes: np.ndarray = None  # int32
+    rep_recv_ptr: np.ndarray = None  # int32
+    rep_del_slots: np.ndarray = None  # int32
+
+    @property
+    def n_events(self) -> int:
+        n_rep = 0 if self.rep_del_slots is None else len(self.rep_del_slots)
+        return (
+            len(self.snap_nodes)
+            + len(self.del_slots)
+            + len(self.pull_snap_nodes)
+            + n_rep
+        )
+
+
+@dataclass
+class RoundSchedule:
+    """One round's ticks plus bookkeeping for the report and the comm plan."""
+
+    round_idx: int
+    ticks: List[TickPhase]
+    n_slots: int  #: size of the snapshot slot pool this round
+    slot_owner: np.ndarray  #: int32 [n_slots] — node that writes each slot
+    sent_messages: int
+    failed_messages: int
+    total_size: int
+    eval_nodes: Optional[np.ndarray]  #: node sample for the round-end sweep
+
+
+class Scheduler:
+    """Derives :class:`RoundSchedule` objects from the random tape.
+
+    Carry-over state between rounds (in-flight messages whose delay crosses
+    the round boundary) is kept internally, so rounds must be generated in
+    order — which every rank does identically.
+    """
+
+    def __init__(self, cfg: EngineConfig):
+        self.cfg = cfg
+        self.tape = RandomTape(cfg.seed)
+        n = cfg.n_nodes
+        # per-node timeout offsets (gossipy/node.py:79): sync U(0, delta),
+        # async N(delta, delta/10) clipped to >=1 (the reference can draw <=0
+        # and crash on t % 0 — SURVEY.md §2.3 quirk 16)
+        g = self.tape.stream(Purpose.TIMEOUT)
+        if cfg.sync:
+            self.deltas = g.integers(0, cfg.delta, size=n).astype(np.int64)
+        else:
+            self.deltas = np.maximum(
+                1, g.normal(cfg.delta, cfg.delta / 10, size=n).astype(np.int64)
+            )
+        # in-flight messages carried across round boundaries:
+        # tick -> list of (receiver, slot, reply_flag, is_pull_request, sender)
+        self._pending: Dict[int, List[Tuple[int, int, int, bool, int]]] = {}
+        # tick -> (sent, failed, size) accounting of replies enqueued there
+        self._reply_accounting: Dict[int, Tuple[int, int, int]] = {}
+
+    # -- internals -----------------------------------------------------------
+
+    def _firing(self, t: int) -> np.ndarray:
+        if self.cfg.sync:
+            return np.where((t % self.cfg.delta) == self.deltas)[0]
+        return np.where((t % self.deltas) == 0)[0]
+
+    def _peers_of(self, nodes: np.ndarray, t: int) -> np.ndarray:
+        """Peer choice per firing node (uniform over its adjacency row;
+        fully-connected fast path excludes self)."""
+        cfg = self.cfg
+        n = len(nodes)
+        if n == 0:
+            return np.empty(0, dtype=np.int64)
+        g = self.tape.stream(Purpose.PEER, t)
+        if cfg.peers_indptr is None:
+            draw = g.integers(0, cfg.n_nodes - 1, size=n)
+            return draw + (draw >= nodes)  # skip self
+        starts = cfg.peers_indptr[nodes]
+        degs = cfg.peers_indptr[nodes + 1] - starts
+        # nodes with no peers never appear here (runner validates topology)
+        offs = np.floor(g.random(n) * degs).astype(np.int64)
+        return cfg.peers_indices[starts + offs]
+
+    def _delays(self, t: int, n: int, sizes: np.ndarray) -> np.ndarray:
+        d = self.cfg.delay
+        if isinstance(d, ConstantDelay):
+            return np.full(n, d._delay, dtype=np.int64)
+        if isinstance(d, UniformDelay):
+            g = self.tape.stream(Purpose.DELAY, t)
+            return g.integers(d._min_delay, d._max_delay + 1, size=n)
+        if isinstance(d, LinearDelay):
+            return (d._timexunit * sizes).astype(np.int64) + d._overhead
+        # custom Delay subclass: fall back to per-message scalar calls
+        return np.array([d.get(None) for _ in range(n)], dtype=np.int64)
+
+    # -- public --------------------------------------------------------------
+
+    def next_round(self, r: int) -> RoundSchedule:
+        """Build the schedule of round ``r`` (rounds must be consumed in
+        order; the in-flight message queue carries over)."""
+        cfg = self.cfg
+        proto = cfg.protocol
+        t0, t1 = r * cfg.delta, (r + 1) * cfg.delta
+        slot_counter = 0
+        slot_owner: List[int] = []
+        sent = failed = total_size = 0
+        ticks: List[TickPhase] = []
+
+        def new_slot(owner: int) -> int:
+            nonlocal slot_counter
+            s = slot_counter
+            slot_counter += 1
+            slot_owner.append(owner)
+            return s
+
+        for t in range(t0, t1):
+            firing = self._firing(t)
+            snap_nodes: List[int] = []
+            snap_slots: List[int] = []
+
+            # --- sends (sub-phase A of the reference loop,
+            #     gossipy/simul.py:393-407)
+            n_f = len(firing)
+            if n_f:
+                peers = self._peers_of(firing, t)
+                drop_u = self.tape.uniform(Purpose.DROP, t, n_f)
+                sizes = np.full(n_f, cfg.model_size if proto != AntiEntropyProtocol.PULL else 1)
+                delays = self._delays(t, n_f, sizes)
+                for j in range(n_f):
+                    sender, receiver = int(firing[j]), int(peers[j])
+                    is_pull = proto == AntiEntropyProtocol.PULL
+                    slot = -1
+                    if not is_pull:
+                        slot = new_slot(sender)
+                        snap_nodes.append(sender)
+                        snap_slots.append(slot)
+                    sent += 1
+                    total_size += int(sizes[j])
+                    if drop_u[j] >= cfg.drop_prob:
+                        due = t + int(delays[j])
+                        wants_reply = proto == AntiEntropyProtocol.PUSH_PULL
+                        self._pending.setdefault(due, []).append(
+                            (receiver, slot, -2 if wants_reply else -1, is_pull, sender)
+                        )
+                    else:
+                        failed += 1
+
+            # --- deliveries due this tick (sub-phase B,
+            #     gossipy/simul.py:409-430). Reply messages generated here are
+            #     enqueued for their own delivery tick; replies to replies are
+            #     discarded (reference parity).
+            online_u = self.tape.uniform(Purpose.ONLINE, t, cfg.n_nodes)
+            online = online_u <= cfg.online_prob
+            due = self._pending.pop(t, [])
+            recv_map: Dict[int, List[Tuple[int, int]]] = {}
+            pull_nodes: List[int] = []
+            pull_slots: List[int] = []
+            for receiver, slot, reply_flag, is_pull, sender in due:
+                if not online[receiver]:
+                    failed += 1
+                    continue
+                if is_pull:
+                    # PULL request: receiver snapshots and replies
+                    rslot = new_slot(receiver)
+                    pull_nodes.append(receiver)
+                    pull_slots.append(rslot)
+                    self._enqueue_reply(t, receiver, sender, rslot)
+                    continue
+                rslot = -1
+                if reply_flag == -2:  # PUSH_PULL: reply with post-merge model
+                    rslot = new_slot(receiver)
+                    self._enqueue_reply(t, receiver, sender, rslot)
+                recv_map.setdefault(receiver, []).append((slot, rslot))
+
+            recv_nodes = np.fromiter(recv_map.keys(), dtype=np.int32, count=len(recv_map))
+            recv_ptr = np.zeros(len(recv_map) + 1, dtype=np.int32)
+            del_slots: List[int] = []
+            reply_slots: List[int] = []
+            for i, rn in enumerate(recv_nodes):
+                pairs = recv_map[int(rn)]
+                del_slots.extend(p[0] for p in pairs)
+                reply_slots.extend(p[1] for p in pairs)
+                recv_ptr[i + 1] = recv_ptr[i] + len(pairs)
+
+            # --- sub-phase C: replies that came due in THIS tick (delay 0).
+            # They were enqueued by the loop above into _pending[t]; pop
+            # again. Replies to replies do not exist (parity).
+            rep_due = self._pending.pop(t, [])
+            rep_map: Dict[int, List[int]] = {}
+            for receiver, slot, _rf, _ip, _sender in rep_due:
+                if not online[receiver]:
+                    failed += 1
+                    continue
+                rep_map.setdefault(receiver, []).append(slot)
+            rep_recv = np.fromiter(rep_map.keys(), dtype=np.int32, count=len(rep_map))
+            rep_ptr = np.zeros(len(rep_map) + 1, dtype=np.int32)
+            rep_slots: List[int] = []
+            for i, rn in enumerate(rep_recv):
+                rep_slots.extend(rep_map[int(rn)])
+                rep_ptr[i + 1] = len(rep_slots)
+
+            phase = TickPhase(
+                t=t,
+                snap_nodes=np.asarray(snap_nodes, dtype=np.int32),
+                snap_slots=np.asarray(snap_slots, dtype=np.int32),
+                recv_nodes=recv_nodes,
+                recv_ptr=recv_ptr,
+                del_slots=np.asarray(del_slots, dtype=np.int32),
+                reply_slots=np.asarray(reply_slots, dtype=np.int32),
+                pull_snap_nodes=np.asarray(pull_nodes, dtype=np.int32),
+                pull_snap_slots=np.asarray(pull_slots, dtype=np.int32),
+                rep_recv_nodes=rep_recv,
+                rep_recv_ptr=rep_ptr,
+                rep_del_slots=np.asarray(rep_slots, dtype=np.int32),
+            )
+            if phase.n_events:
+                ticks.append(phase)
+
+            # count messages the reply bookkeeping generated
+            sent_r, failed_r, size_r = self._reply_accounting.pop(t, (0, 0, 0))
+            sent += sent_r
+            failed += failed_r
+            total_size += size_r
+
+        eval_nodes = None
+        if cfg.sampling_eval > 0:
+            g = self.tape.stream(Purpose.EVAL, t1 - 1)
+            k = max(int(cfg.n_nodes * cfg.sampling_eval), 1)
+            eval_nodes = g.choice(cfg.n_nodes, size=k, replace=True)
+
+        return RoundSchedule(
+            round_idx=r,
+            ticks=ticks,
+            n_slots=slot_counter,
+            slot_owner=np.asarray(slot_owner, dtype=np.int32),
+            sent_messages=sent,
+            failed_messages=failed,
+            total_size=total_size,
+            eval_nodes=eval_nodes,
+        )
+
+    def _enqueue_reply(self, t: int, replier: int, requester: int, slot: int) -> None:
+        """Queue a REPLY message (drop-tested with the reference's
+        ``random() > drop_prob`` variant, gossipy/simul.py:414)."""
+        g = self.tape.stream(Purpose.DROP, t, extra=1 + replier)
+        u = float(g.random())
+        sent, failed, size = self._reply_accounting.get(t, (0, 0, 0))
+        sent += 1
+        size += self.cfg.model_size
+        if u > self.cfg.drop_prob:
+            gd = self.tape.stream(Purpose.DELAY, t, extra=1 + replier)
+            d = self.cfg.delay
+            if isinstance(d, ConstantDelay):
+                dly = d._delay
+            elif isinstance(d, UniformDelay):
+                dly = int(gd.integers(d._min_delay, d._max_delay + 1))
+            elif isinstance(d, LinearDelay):
+                dly = int(d._timexunit * self.cfg.model_size) + d._overhead
+            else:
+                dly = int(d.get(None))
+            self._pending.setdefault(t + dly, []).append(
+                (requester, slot, -1, False, replier)
+            )
+        else:
+            failed += 1
+        self._reply_accounting[t] = (sent, failed, size)

@@ -1,0 +1,39 @@
+"""HIP/CDNA4 kernel extension loader.
+
+The kernels live in ``hip/gossip_kernels.hip`` and are built **in-tree**
+for gfx950 (``python -m gossipy_amd.ops.build`` or the repo's
+``__graft_entry__.build()``), producing ``_gossip_hip.so`` next to this
+file so the snapshot that travels to the GPU box carries it.
+
+On a GPU box the extension is mandatory: :func:`load_extension` raises if
+the ``.so`` is absent rather than silently falling back to eager PyTorch.
+"""
+
+from __future__ import annotations
+
+import os
+
+_EXT = None
+
+
+def extension_path() -> str:
+    return os.path.join(os.path.dirname(__file__), "_gossip_hip.so")
+
+
+def load_extension():
+    """Load (once) and return the compiled kernel module."""
+    global _EXT
+    if _EXT is not None:
+        return _EXT
+    import torch
+
+    path = extension_path()
+    if not os.path.exists(path):
+        raise ImportError(
+            "gossipy_amd HIP extension not built: %s missing. "
+            "Run `python -m gossipy_amd.ops.build` (needs hipcc; "
+            "cross-compiles for gfx950 without a GPU)." % path
+        )
+    torch.ops.load_library(path)
+    _EXT = torch.ops.gossipy_amd
+    return _EXT
