@@ -60,29 +60,41 @@ class TorchBackend:
 
     # -- init ----------------------------------------------------------------
 
-    def init_params(self, state: NodeStateArena, spec, tape: RandomTape) -> None:
+    def init_params(
+        self, state: NodeStateArena, spec, tape: RandomTape, n_total: int = None
+    ) -> None:
         """Per-node init. AdaLine/Pegasos start at zero
         (gossipy/model/nn.py:131); logreg/mlp use torch's Linear default
         (kaiming-uniform, bounds 1/sqrt(fan_in)) since the reference's
         ``init_weights`` for LogisticRegression is a no-op
-        (gossipy/model/nn.py:169-170)."""
+        (gossipy/model/nn.py:169-170).
+
+        The tape is drawn for the FULL node population and the local block
+        sliced out, so every residency map sees identical initial models.
+        """
         if spec.family in ("pegasos", "adaline"):
             state.params.zero_()
             return
         g = tape.stream(Purpose.INIT)
         n, D = state.params.shape
+        n_total = n_total or n
+        lo = state.node_lo
         if spec.family == "logreg":
             bound = 1.0 / np.sqrt(spec.d_in)
-            w = g.uniform(-bound, bound, size=(n, D))
+            w = g.uniform(-bound, bound, size=(n_total, D))[lo : lo + n]
             state.params.copy_(torch.from_numpy(w).float().to(state.params.device))
         elif spec.family == "mlp":
-            rows = np.empty((n, D), dtype=np.float32)
+            rows = np.empty((n_total, D), dtype=np.float32)
             for (w_off, b_off, fin, fout) in spec.layer_offsets():
                 # xavier-uniform weights, zero bias (gossipy/model/nn.py:106-110)
                 bound = np.sqrt(6.0 / (fin + fout))
-                rows[:, w_off:b_off] = g.uniform(-bound, bound, size=(n, fout * fin))
+                rows[:, w_off:b_off] = g.uniform(
+                    -bound, bound, size=(n_total, fout * fin)
+                )
                 rows[:, b_off : b_off + fout] = 0.0
-            state.params.copy_(torch.from_numpy(rows).to(state.params.device))
+            state.params.copy_(
+                torch.from_numpy(rows[lo : lo + n]).to(state.params.device)
+            )
         else:
             raise ValueError(spec.family)
         state.ages.zero_()

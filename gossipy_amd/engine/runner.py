@@ -113,7 +113,7 @@ class BatchedGossipSimulator(SimulationEventSender):
         """Initialize all resident models and run the reference's one
         initial local training pass (gossipy/node.py:82-94)."""
         tape = RandomTape(seed if seed is not None else self.cfg.seed)
-        self.backend.init_params(self.state, self.spec, tape)
+        self.backend.init_params(self.state, self.spec, tape, self.cfg.n_nodes)
         all_local = torch.arange(self.n_local)
         self.backend.update(self.state, self.data, self.spec, all_local)
         self.initialized = True
@@ -159,12 +159,17 @@ class BatchedGossipSimulator(SimulationEventSender):
             self.pool.slot_ages[ids] = buf[:, D].int()
 
     def _plan_exchange(
-        self, recv_nodes: np.ndarray, recv_ptr: np.ndarray, slots: np.ndarray
+        self,
+        recv_nodes: np.ndarray,
+        recv_ptr: np.ndarray,
+        slots: np.ndarray,
+        owners: np.ndarray,
     ) -> List[Tuple[int, int, np.ndarray]]:
-        """(src, dst, slots) transfer plan for one delivery CSR."""
+        """(src, dst, slots) transfer plan for one delivery CSR; ``owners``
+        is the per-delivery writer node (carried in the phase because slot
+        ids are recycled)."""
         if self.world == 1 or len(slots) == 0:
             return []
-        owners = self._slot_owner[slots]
         src = self._rank_of(owners)
         dst = self._rank_of(np.repeat(recv_nodes, np.diff(recv_ptr)))
         plan: Dict[Tuple[int, int], List[int]] = {}
@@ -190,7 +195,9 @@ class BatchedGossipSimulator(SimulationEventSender):
             )
         # move cross-GPU slots needed by this tick's deliveries
         self._exchange(
-            self._plan_exchange(phase.recv_nodes, phase.recv_ptr, phase.del_slots)
+            self._plan_exchange(
+                phase.recv_nodes, phase.recv_ptr, phase.del_slots, phase.del_owners
+            )
         )
 
         # B: deliveries (merge + update [+ reply snapshot]) and PULL snapshots
@@ -231,7 +238,10 @@ class BatchedGossipSimulator(SimulationEventSender):
         if phase.rep_del_slots is not None and len(phase.rep_del_slots):
             self._exchange(
                 self._plan_exchange(
-                    phase.rep_recv_nodes, phase.rep_recv_ptr, phase.rep_del_slots
+                    phase.rep_recv_nodes,
+                    phase.rep_recv_ptr,
+                    phase.rep_del_slots,
+                    phase.rep_del_owners,
                 )
             )
             cmine = self._is_mine(phase.rep_recv_nodes)

@@ -70,6 +70,9 @@ class TickPhase:
     recv_nodes: np.ndarray  # int32 [R]   unique receivers this tick
     recv_ptr: np.ndarray  # int32 [R+1]
     del_slots: np.ndarray  # int32 [recv_ptr[-1]]  slot per delivery, in order
+    #: writer of each delivered slot (slot ids are recycled, so the owner
+    #: must be carried per-delivery — it drives the cross-GPU transfer plan)
+    del_owners: np.ndarray  # int32 [recv_ptr[-1]]
     # PUSH_PULL / PULL: deliveries that must also emit a reply snapshot
     # (aligned with del_slots; -1 = no reply)
     reply_slots: np.ndarray  # int32 [recv_ptr[-1]]
@@ -83,6 +86,7 @@ class TickPhase:
     rep_recv_nodes: np.ndarray = None  # int32
     rep_recv_ptr: np.ndarray = None  # int32
     rep_del_slots: np.ndarray = None  # int32
+    rep_del_owners: np.ndarray = None  # int32
 
     @property
     def n_events(self) -> int:
@@ -136,6 +140,27 @@ class Scheduler:
         self._pending: Dict[int, List[Tuple[int, int, int, bool, int]]] = {}
         # tick -> (sent, failed, size) accounting of replies enqueued there
         self._reply_accounting: Dict[int, Tuple[int, int, int]] = {}
+        # deterministic slot allocator: ids persist across rounds (a delayed
+        # message keeps its slot until delivered), recycled through a free
+        # list once consumed/dropped. This is the arena equivalent of the
+        # reference cache's refcounting (gossipy/__init__.py:283-387) —
+        # except slots of messages lost to drops/offline receivers are
+        # reclaimed instead of leaked (the reference leaks them,
+        # SURVEY.md §5 'failure detection').
+        self._free_slots: List[int] = []
+        self._next_slot = 0
+        self.slot_owner = np.zeros(64, dtype=np.int32)
+
+    def _alloc_slot(self, owner: int) -> int:
+        if self._free_slots:
+            s = self._free_slots.pop()
+        else:
+            s = self._next_slot
+            self._next_slot += 1
+            if s >= len(self.slot_owner):
+                self.slot_owner = np.resize(self.slot_owner, 2 * len(self.slot_owner))
+        self.slot_owner[s] = owner
+        return s
 
     # -- internals -----------------------------------------------------------
 
@@ -181,22 +206,14 @@ class Scheduler:
         cfg = self.cfg
         proto = cfg.protocol
         t0, t1 = r * cfg.delta, (r + 1) * cfg.delta
-        slot_counter = 0
-        slot_owner: List[int] = []
         sent = failed = total_size = 0
         ticks: List[TickPhase] = []
-
-        def new_slot(owner: int) -> int:
-            nonlocal slot_counter
-            s = slot_counter
-            slot_counter += 1
-            slot_owner.append(owner)
-            return s
 
         for t in range(t0, t1):
             firing = self._firing(t)
             snap_nodes: List[int] = []
             snap_slots: List[int] = []
+            freed: List[int] = []  # slots recycled at end of this tick
 
             # --- sends (sub-phase A of the reference loop,
             #     gossipy/simul.py:393-407)
@@ -211,7 +228,7 @@ class Scheduler:
                     is_pull = proto == AntiEntropyProtocol.PULL
                     slot = -1
                     if not is_pull:
-                        slot = new_slot(sender)
+                        slot = self._alloc_slot(sender)
                         snap_nodes.append(sender)
                         snap_slots.append(slot)
                     sent += 1
@@ -224,6 +241,8 @@ class Scheduler:
                         )
                     else:
                         failed += 1
+                        if slot >= 0:
+                            freed.append(slot)
 
             # --- deliveries due this tick (sub-phase B,
             #     gossipy/simul.py:409-430). Reply messages generated here are
@@ -238,45 +257,56 @@ class Scheduler:
             for receiver, slot, reply_flag, is_pull, sender in due:
                 if not online[receiver]:
                     failed += 1
+                    if slot >= 0:
+                        freed.append(slot)
                     continue
                 if is_pull:
                     # PULL request: receiver snapshots and replies
-                    rslot = new_slot(receiver)
+                    rslot = self._alloc_slot(receiver)
                     pull_nodes.append(receiver)
                     pull_slots.append(rslot)
-                    self._enqueue_reply(t, receiver, sender, rslot)
+                    if not self._enqueue_reply(t, receiver, sender, rslot):
+                        freed.append(rslot)
                     continue
                 rslot = -1
                 if reply_flag == -2:  # PUSH_PULL: reply with post-merge model
-                    rslot = new_slot(receiver)
-                    self._enqueue_reply(t, receiver, sender, rslot)
-                recv_map.setdefault(receiver, []).append((slot, rslot))
+                    rslot = self._alloc_slot(receiver)
+                    if not self._enqueue_reply(t, receiver, sender, rslot):
+                        freed.append(rslot)
+                recv_map.setdefault(receiver, []).append((slot, rslot, sender))
+                freed.append(slot)  # consumed by this delivery
 
             recv_nodes = np.fromiter(recv_map.keys(), dtype=np.int32, count=len(recv_map))
             recv_ptr = np.zeros(len(recv_map) + 1, dtype=np.int32)
             del_slots: List[int] = []
+            del_owners: List[int] = []
             reply_slots: List[int] = []
             for i, rn in enumerate(recv_nodes):
                 pairs = recv_map[int(rn)]
                 del_slots.extend(p[0] for p in pairs)
                 reply_slots.extend(p[1] for p in pairs)
+                del_owners.extend(p[2] for p in pairs)
                 recv_ptr[i + 1] = recv_ptr[i] + len(pairs)
 
             # --- sub-phase C: replies that came due in THIS tick (delay 0).
             # They were enqueued by the loop above into _pending[t]; pop
             # again. Replies to replies do not exist (parity).
             rep_due = self._pending.pop(t, [])
-            rep_map: Dict[int, List[int]] = {}
-            for receiver, slot, _rf, _ip, _sender in rep_due:
+            rep_map: Dict[int, List[Tuple[int, int]]] = {}
+            for receiver, slot, _rf, _ip, sender in rep_due:
                 if not online[receiver]:
                     failed += 1
+                    freed.append(slot)
                     continue
-                rep_map.setdefault(receiver, []).append(slot)
+                rep_map.setdefault(receiver, []).append((slot, sender))
+                freed.append(slot)  # consumed by this reply delivery
             rep_recv = np.fromiter(rep_map.keys(), dtype=np.int32, count=len(rep_map))
             rep_ptr = np.zeros(len(rep_map) + 1, dtype=np.int32)
             rep_slots: List[int] = []
+            rep_owners: List[int] = []
             for i, rn in enumerate(rep_recv):
-                rep_slots.extend(rep_map[int(rn)])
+                rep_slots.extend(p[0] for p in rep_map[int(rn)])
+                rep_owners.extend(p[1] for p in rep_map[int(rn)])
                 rep_ptr[i + 1] = len(rep_slots)
 
             phase = TickPhase(
@@ -286,12 +316,14 @@ class Scheduler:
                 recv_nodes=recv_nodes,
                 recv_ptr=recv_ptr,
                 del_slots=np.asarray(del_slots, dtype=np.int32),
+                del_owners=np.asarray(del_owners, dtype=np.int32),
                 reply_slots=np.asarray(reply_slots, dtype=np.int32),
                 pull_snap_nodes=np.asarray(pull_nodes, dtype=np.int32),
                 pull_snap_slots=np.asarray(pull_slots, dtype=np.int32),
                 rep_recv_nodes=rep_recv,
                 rep_recv_ptr=rep_ptr,
                 rep_del_slots=np.asarray(rep_slots, dtype=np.int32),
+                rep_del_owners=np.asarray(rep_owners, dtype=np.int32),
             )
             if phase.n_events:
                 ticks.append(phase)
@@ -302,6 +334,11 @@ class Scheduler:
             failed += failed_r
             total_size += size_r
 
+            # recycle this tick's consumed/dropped slots (safe: any reuse
+            # happens in a later tick's launch, stream-ordered after the
+            # consuming kernel)
+            self._free_slots.extend(freed)
+
         eval_nodes = None
         if cfg.sampling_eval > 0:
             g = self.tape.stream(Purpose.EVAL, t1 - 1)
@@ -311,17 +348,18 @@ class Scheduler:
         return RoundSchedule(
             round_idx=r,
             ticks=ticks,
-            n_slots=slot_counter,
-            slot_owner=np.asarray(slot_owner, dtype=np.int32),
+            n_slots=self._next_slot,
+            slot_owner=self.slot_owner[: self._next_slot].copy(),
             sent_messages=sent,
             failed_messages=failed,
             total_size=total_size,
             eval_nodes=eval_nodes,
         )
 
-    def _enqueue_reply(self, t: int, replier: int, requester: int, slot: int) -> None:
+    def _enqueue_reply(self, t: int, replier: int, requester: int, slot: int) -> bool:
         """Queue a REPLY message (drop-tested with the reference's
-        ``random() > drop_prob`` variant, gossipy/simul.py:414)."""
+        ``random() > drop_prob`` variant, gossipy/simul.py:414). Returns
+        whether the reply was actually enqueued (False = dropped)."""
         g = self.tape.stream(Purpose.DROP, t, extra=1 + replier)
         u = float(g.random())
         sent, failed, size = self._reply_accounting.get(t, (0, 0, 0))
@@ -341,6 +379,8 @@ class Scheduler:
             self._pending.setdefault(t + dly, []).append(
                 (requester, slot, -1, False, replier)
             )
-        else:
-            failed += 1
+            self._reply_accounting[t] = (sent, failed, size)
+            return True
+        failed += 1
         self._reply_accounting[t] = (sent, failed, size)
+        return False
