@@ -25,7 +25,9 @@ def load_extension():
     global _EXT
     if _EXT is not None:
         return _EXT
-    import torch
+    import importlib.util
+
+    import torch  # noqa: F401  (the .so links against torch libs)
 
     path = extension_path()
     if not os.path.exists(path):
@@ -34,6 +36,8 @@ def load_extension():
             "Run `python -m gossipy_amd.ops.build` (needs hipcc; "
             "cross-compiles for gfx950 without a GPU)." % path
         )
-    torch.ops.load_library(path)
-    _EXT = torch.ops.gossipy_amd
+    spec = importlib.util.spec_from_file_location("gossipy_amd.ops._gossip_hip", path)
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    _EXT = mod
     return _EXT

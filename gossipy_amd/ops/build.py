@@ -1,0 +1,42 @@
+"""In-tree build of the gfx950 HIP extension.
+
+Usage: ``python -m gossipy_amd.ops.build``. Cross-compiles on CPU-only
+machines (hipcc needs no GPU); the resulting ``_gossip_hip.so`` sits next
+to ``gossipy_amd/ops/__init__.py`` so it travels with the repo snapshot to
+GPU boxes.
+"""
+
+from __future__ import annotations
+
+import os
+import shutil
+import sys
+
+
+def build(verbose: bool = False) -> str:
+    os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+    os.environ.setdefault("MAX_JOBS", "8")
+    from torch.utils import cpp_extension
+
+    here = os.path.dirname(os.path.abspath(__file__))
+    src = os.path.join(here, "hip", "gossip_kernels.hip")
+    build_dir = os.path.join(here, "hip", "build")
+    os.makedirs(build_dir, exist_ok=True)
+    cpp_extension.load(
+        name="_gossip_hip",
+        sources=[src],
+        build_directory=build_dir,
+        extra_cuda_cflags=["-O3", "--offload-arch=gfx950"],
+        verbose=verbose,
+        is_python_module=False,
+        with_cuda=True,
+    )
+    built = os.path.join(build_dir, "_gossip_hip.so")
+    target = os.path.join(here, "_gossip_hip.so")
+    shutil.copy2(built, target)
+    return target
+
+
+if __name__ == "__main__":
+    path = build(verbose="-v" in sys.argv)
+    print("built:", path)

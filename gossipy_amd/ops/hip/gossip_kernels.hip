@@ -1,0 +1,672 @@
+// gossipy_amd CDNA4 (gfx950) kernels — the batched compute path of the
+// MI355X gossip engine.
+//
+// Design: one kernel launch services EVERY simulated node
+// active in a tick sub-phase. Workgroups map 1:1 to nodes; a node's whole
+// event sequence for the sub-phase (merge -> local SGD -> reply snapshot)
+// runs inside its workgroup so no cross-workgroup ordering is ever needed —
+// ordering between sub-phases comes from stream order of the launches.
+//
+// Kernel inventory (K-numbers from SURVEY.md §2.4):
+//   snapshot_kernel       — K-equivalent of ModelHandler.caching
+//                           (gossipy/model/handler.py:160-176): arena row
+//                           copy instead of copy.deepcopy.
+//   tick_logreg_kernel    — K3+K4+K5 fused: per-receiver merge (mean of
+//                           state, gossipy/model/handler.py:260-280) +
+//                           minibatch SGD on CE(sigmoid(Wx+b))
+//                           (gossipy/model/handler.py:235-258,
+//                            gossipy/model/nn.py:162-166).
+//   tick_linear_kernel    — K1/K2: Pegasos hinge SGD
+//                           (gossipy/model/handler.py:416-423) and AdaLine
+//                           delta rule (gossipy/model/handler.py:364-368),
+//                           one wave per node, weights in registers.
+//   tick_mlp_kernel       — K3/K4 generalized: fused MLP fwd/bwd/SGD
+//                           (gossipy/model/nn.py:91-99).
+//
+// Models are tiny (57x2 logreg ... few-hundred-wide MLPs); per-launch work
+// is launch-latency bound, so kernels are built to let ONE launch cover all
+// active nodes and to keep every intermediate in LDS/registers — the D
+// parameters are read from HBM once per tick and written once.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+
+#define DEV_INLINE __device__ __forceinline__
+
+constexpr int KMAX = 16;     // max classes for the logreg path
+constexpr int WAVE = 64;
+
+// CreateModelMode numbering shared with engine/backend.py (_MODE_ID)
+constexpr int MODE_UPDATE = 0;
+constexpr int MODE_MERGE_UPDATE = 1;
+constexpr int MODE_UPDATE_MERGE = 2;
+constexpr int MODE_PASS = 3;
+
+// ---------------------------------------------------------------------------
+// snapshot: slots[slot_ids[i]] = params[nodes[i]]  (+ age)
+// ---------------------------------------------------------------------------
+
+__global__ void snapshot_kernel(
+    const float* __restrict__ params,
+    const int* __restrict__ ages,
+    float* __restrict__ slots,
+    int* __restrict__ slot_ages,
+    const int* __restrict__ nodes,
+    const int* __restrict__ slot_ids,
+    int n, int D)
+{
+    long total = (long)n * D;
+    for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x; idx < total;
+         idx += (long)gridDim.x * blockDim.x) {
+        int row = idx / D;
+        int col = idx - (long)row * D;
+        int node = nodes[row];
+        int slot = slot_ids[row];
+        slots[(long)slot * D + col] = params[(long)node * D + col];
+        if (col == 0) slot_ages[slot] = ages[node];
+    }
+}
+
+// ---------------------------------------------------------------------------
+// logreg tick: one workgroup per active node
+// ---------------------------------------------------------------------------
+
+struct LogregArgs {
+    float* params; int* ages;
+    float* slots; int* slot_ages;
+    const int* nodes; const int* ptr;
+    const int* dslots; const int* rslots;
+    const float* X; const float* Y; const int* counts;
+    int d, k, Smax, D;
+    float lr, wd;
+    int epochs, bs, mode, update_only;
+};
+
+// Minibatch SGD epochs over the node's shard, on the LDS-resident model W.
+// xb/dz are LDS scratch; age is a wave-uniform register.
+DEV_INLINE void logreg_update(const LogregArgs& a, int node, float* W,
+                              float* xb, float* dz, int& age)
+{
+    int tid = threadIdx.x;
+    int c = a.counts[node];
+    if (c == 0) return;
+    int bsz = (a.bs == 0) ? c : min(a.bs, c);
+    const float* Xn = a.X + (long)node * a.Smax * a.d;
+    const float* Yn = a.Y + (long)node * a.Smax;
+    for (int ep = 0; ep < a.epochs; ++ep) {
+        for (int s0 = 0; s0 < c; s0 += bsz) {
+            int m = min(bsz, c - s0);
+            // stage the batch in LDS (coalesced: consecutive threads read
+            // consecutive floats of the shard)
+            for (int e = tid; e < m * a.d; e += blockDim.x)
+                xb[e] = Xn[(long)s0 * a.d + e];
+            __syncthreads();
+            // per-sample forward + dLoss/dz (thread = sample)
+            if (tid < m) {
+                float z[KMAX];
+                for (int kk = 0; kk < a.k; ++kk) {
+                    float acc = W[a.k * a.d + kk];  // bias
+                    const float* wrow = W + kk * a.d;
+                    const float* xrow = xb + tid * a.d;
+                    for (int dd = 0; dd < a.d; ++dd) acc += wrow[dd] * xrow[dd];
+                    z[kk] = acc;
+                }
+                // a = sigmoid(z); p = softmax(a); dz = (p - 1_y)/m * a*(1-a)
+                float amax = -1e30f;
+                for (int kk = 0; kk < a.k; ++kk) {
+                    z[kk] = 1.0f / (1.0f + __expf(-z[kk]));
+                    amax = fmaxf(amax, z[kk]);
+                }
+                float sum = 0.f;
+                float p[KMAX];
+                for (int kk = 0; kk < a.k; ++kk) {
+                    p[kk] = __expf(z[kk] - amax);
+                    sum += p[kk];
+                }
+                int yi = (int)Yn[s0 + tid];
+                float inv = 1.0f / sum;
+                for (int kk = 0; kk < a.k; ++kk) {
+                    float g = p[kk] * inv - (kk == yi ? 1.0f : 0.0f);
+                    dz[tid * a.k + kk] = (g / m) * z[kk] * (1.0f - z[kk]);
+                }
+            }
+            __syncthreads();
+            // SGD step (thread = parameter)
+            for (int e = tid; e < a.k * a.d; e += blockDim.x) {
+                int kk = e / a.d, dd = e - kk * a.d;
+                float g = 0.f;
+                for (int s = 0; s < m; ++s) g += dz[s * a.k + kk] * xb[s * a.d + dd];
+                if (a.wd != 0.f) g += a.wd * W[e];
+                W[e] -= a.lr * g;
+            }
+            for (int e = tid; e < a.k; e += blockDim.x) {
+                float g = 0.f;
+                for (int s = 0; s < m; ++s) g += dz[s * a.k + e];
+                W[a.k * a.d + e] -= a.lr * g;
+            }
+            __syncthreads();
+            age += 1;  // wave-uniform (gossipy/model/handler.py:258)
+        }
+    }
+}
+
+__global__ void __launch_bounds__(128)
+tick_logreg_kernel(LogregArgs a)
+{
+    int i = blockIdx.x;
+    int node = a.nodes[i];
+    int tid = threadIdx.x;
+    extern __shared__ float sm[];
+    float* W = sm;                         // D
+    float* W2 = W + a.D;                   // D (UPDATE_MERGE scratch)
+    float* xb = W2 + a.D;                  // bsmax*d
+    int bsmax = (a.bs == 0) ? a.Smax : min(a.bs, a.Smax);
+    float* dz = xb + bsmax * a.d;          // bsmax*k
+
+    for (int e = tid; e < a.D; e += blockDim.x)
+        W[e] = a.params[(long)node * a.D + e];
+    __syncthreads();
+    int age = a.ages[node];
+
+    if (a.update_only) {
+        logreg_update(a, node, W, xb, dz, age);
+    } else {
+        for (int j = a.ptr[i]; j < a.ptr[i + 1]; ++j) {
+            int slot = a.dslots[j];
+            const float* srow = a.slots + (long)slot * a.D;
+            int sage = a.slot_ages[slot];
+            if (a.mode == MODE_MERGE_UPDATE) {
+                for (int e = tid; e < a.D; e += blockDim.x)
+                    W[e] = 0.5f * (W[e] + srow[e]);
+                age = max(age, sage);
+                __syncthreads();
+                logreg_update(a, node, W, xb, dz, age);
+            } else if (a.mode == MODE_UPDATE) {
+                // adopt the received model, then train it
+                for (int e = tid; e < a.D; e += blockDim.x) W[e] = srow[e];
+                age = sage;
+                __syncthreads();
+                logreg_update(a, node, W, xb, dz, age);
+            } else if (a.mode == MODE_UPDATE_MERGE) {
+                logreg_update(a, node, W, xb, dz, age);
+                for (int e = tid; e < a.D; e += blockDim.x) W2[e] = srow[e];
+                __syncthreads();
+                int age2 = sage;
+                logreg_update(a, node, W2, xb, dz, age2);
+                for (int e = tid; e < a.D; e += blockDim.x)
+                    W[e] = 0.5f * (W[e] + W2[e]);
+                age = max(age, age2);
+                __syncthreads();
+            } else {  // PASS
+                for (int e = tid; e < a.D; e += blockDim.x) W[e] = srow[e];
+                age = sage;
+                __syncthreads();
+            }
+            int rs = a.rslots[j];
+            if (rs >= 0) {  // PUSH_PULL reply snapshot (post merge+update)
+                for (int e = tid; e < a.D; e += blockDim.x)
+                    a.slots[(long)rs * a.D + e] = W[e];
+                if (tid == 0) a.slot_ages[rs] = age;
+                __syncthreads();
+            }
+        }
+    }
+    __syncthreads();
+    for (int e = tid; e < a.D; e += blockDim.x)
+        a.params[(long)node * a.D + e] = W[e];
+    if (tid == 0) a.ages[node] = age;
+}
+
+// ---------------------------------------------------------------------------
+// pegasos / adaline tick: one WAVE per node, weights in registers
+// ---------------------------------------------------------------------------
+
+DEV_INLINE float wave_sum(float v)
+{
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+        v += __shfl_down(v, off, WAVE);
+    return __shfl(v, 0, WAVE);
+}
+
+struct LinearArgs {
+    float* params; int* ages;
+    float* slots; int* slot_ages;
+    const int* nodes; const int* ptr;
+    const int* dslots; const int* rslots;
+    const float* X; const float* Y; const int* counts;
+    int d, Smax;
+    float lrlam;       // pegasos lambda or adaline lr
+    int is_pegasos, mode, update_only;
+};
+
+constexpr int LIN_MAX_REGS = 16;  // supports d up to 16*64 = 1024
+
+// Per-sample sequential pass (the math is order-dependent: Pegasos' step
+// size is 1/(t*lam) with t = running age). w lives in per-lane registers
+// (lane L owns dims L, L+64, ...).
+DEV_INLINE void linear_update(const LinearArgs& a, int node, float* w,
+                              int nreg, int& age)
+{
+    int lane = threadIdx.x;
+    int c = a.counts[node];
+    const float* Xn = a.X + (long)node * a.Smax * a.d;
+    const float* Yn = a.Y + (long)node * a.Smax;
+    for (int s = 0; s < c; ++s) {
+        const float* x = Xn + (long)s * a.d;
+        float part = 0.f;
+        float xr[LIN_MAX_REGS];
+        for (int r = 0; r < nreg; ++r) {
+            int e = lane + r * WAVE;
+            xr[r] = (e < a.d) ? x[e] : 0.f;
+            part += w[r] * xr[r];
+        }
+        float pred = wave_sum(part);
+        float y = Yn[s];
+        if (a.is_pegasos) {
+            age += 1;
+            float lr = 1.0f / (age * a.lrlam);
+            float hinge = (pred * y - 1.0f < 0.f) ? 1.0f : 0.0f;
+            float scale = 1.0f - lr * a.lrlam;
+            float add = hinge * lr * y;
+            for (int r = 0; r < nreg; ++r) w[r] = w[r] * scale + add * xr[r];
+        } else {
+            float err = y - pred;
+            for (int r = 0; r < nreg; ++r) w[r] += a.lrlam * err * xr[r];
+        }
+    }
+    if (!a.is_pegasos) age += c;  // AdaLine ages by sample count
+}
+
+__global__ void __launch_bounds__(WAVE)
+tick_linear_kernel(LinearArgs a)
+{
+    int i = blockIdx.x;
+    int node = a.nodes[i];
+    int lane = threadIdx.x;
+    int nreg = (a.d + WAVE - 1) / WAVE;
+    float w[LIN_MAX_REGS];
+    for (int r = 0; r < nreg; ++r) {
+        int e = lane + r * WAVE;
+        w[r] = (e < a.d) ? a.params[(long)node * a.d + e] : 0.f;
+    }
+    int age = a.ages[node];
+
+    if (a.update_only) {
+        linear_update(a, node, w, nreg, age);
+    } else {
+        for (int j = a.ptr[i]; j < a.ptr[i + 1]; ++j) {
+            int slot = a.dslots[j];
+            const float* srow = a.slots + (long)slot * a.d;
+            int sage = a.slot_ages[slot];
+            if (a.mode == MODE_MERGE_UPDATE) {
+                for (int r = 0; r < nreg; ++r) {
+                    int e = lane + r * WAVE;
+                    if (e < a.d) w[r] = 0.5f * (w[r] + srow[e]);
+                }
+                age = max(age, sage);
+                linear_update(a, node, w, nreg, age);
+            } else if (a.mode == MODE_UPDATE) {
+                for (int r = 0; r < nreg; ++r) {
+                    int e = lane + r * WAVE;
+                    if (e < a.d) w[r] = srow[e];
+                }
+                age = sage;
+                linear_update(a, node, w, nreg, age);
+            } else if (a.mode == MODE_UPDATE_MERGE) {
+                linear_update(a, node, w, nreg, age);
+                float w2[LIN_MAX_REGS];
+                for (int r = 0; r < nreg; ++r) {
+                    int e = lane + r * WAVE;
+                    w2[r] = (e < a.d) ? srow[e] : 0.f;
+                }
+                int age2 = sage;
+                linear_update(a, node, w2, nreg, age2);
+                for (int r = 0; r < nreg; ++r) w[r] = 0.5f * (w[r] + w2[r]);
+                age = max(age, age2);
+            } else {  // PASS
+                for (int r = 0; r < nreg; ++r) {
+                    int e = lane + r * WAVE;
+                    if (e < a.d) w[r] = srow[e];
+                }
+                age = sage;
+            }
+            int rs = a.rslots[j];
+            if (rs >= 0) {
+                for (int r = 0; r < nreg; ++r) {
+                    int e = lane + r * WAVE;
+                    if (e < a.d) a.slots[(long)rs * a.d + e] = w[r];
+                }
+                if (lane == 0) a.slot_ages[rs] = age;
+            }
+        }
+    }
+    for (int r = 0; r < nreg; ++r) {
+        int e = lane + r * WAVE;
+        if (e < a.d) a.params[(long)node * a.d + e] = w[r];
+    }
+    if (lane == 0) a.ages[node] = age;
+}
+
+// ---------------------------------------------------------------------------
+// MLP tick: one workgroup per node, all activations in LDS
+// ---------------------------------------------------------------------------
+
+struct MlpArgs {
+    float* params; int* ages;
+    float* slots; int* slot_ages;
+    const int* nodes; const int* ptr;
+    const int* dslots; const int* rslots;
+    const float* X; const float* Y; const int* counts;
+    const int* layout;  // [n_layers][4] = (w_off, b_off, in, out)
+    int n_layers, Smax, D, act_max, d_in;
+    float lr, wd;
+    int epochs, bs, mode, update_only;
+};
+
+// Fused fwd+bwd+SGD over the node's shard. LDS holds the model row plus
+// per-batch activations of every layer (act) and their gradients (grad).
+DEV_INLINE void mlp_update(const MlpArgs& a, int node, float* W,
+                           float* act, float* grad, int& age)
+{
+    int tid = threadIdx.x;
+    int c = a.counts[node];
+    if (c == 0) return;
+    int bsz = (a.bs == 0) ? c : min(a.bs, c);
+    const float* Xn = a.X + (long)node * a.Smax * a.d_in;
+    const float* Yn = a.Y + (long)node * a.Smax;
+    for (int ep = 0; ep < a.epochs; ++ep) {
+        for (int s0 = 0; s0 < c; s0 += bsz) {
+            int m = min(bsz, c - s0);
+            // activations laid out layer after layer: act[l] is [m, out_l];
+            // act[-1] (the input) is staged first at act.
+            float* in = act;
+            for (int e = tid; e < m * a.d_in; e += blockDim.x)
+                in[e] = Xn[(long)s0 * a.d_in + e];
+            __syncthreads();
+            // ---- forward
+            float* cur = in;
+            float* nxt = act + m * a.d_in;
+            for (int l = 0; l < a.n_layers; ++l) {
+                int w_off = a.layout[4 * l], b_off = a.layout[4 * l + 1];
+                int fin = a.layout[4 * l + 2], fout = a.layout[4 * l + 3];
+                for (int e = tid; e < m * fout; e += blockDim.x) {
+                    int s = e / fout, o = e - s * fout;
+                    const float* wrow = W + w_off + o * fin;
+                    const float* xrow = cur + s * fin;
+                    float acc = W[b_off + o];
+                    for (int q = 0; q < fin; ++q) acc += wrow[q] * xrow[q];
+                    if (l < a.n_layers - 1) acc = fmaxf(acc, 0.f);  // ReLU
+                    nxt[e] = acc;
+                }
+                __syncthreads();
+                cur = nxt;
+                nxt = nxt + m * fout;
+            }
+            // ---- output grad: softmax-CE on the raw head
+            int k = a.layout[4 * (a.n_layers - 1) + 3];
+            float* gcur = grad;  // [m, k] for the head, reused per layer
+            if (tid < m) {
+                const float* z = cur + tid * k;
+                float zmax = -1e30f;
+                for (int kk = 0; kk < k; ++kk) zmax = fmaxf(zmax, z[kk]);
+                float sum = 0.f;
+                for (int kk = 0; kk < k; ++kk) sum += __expf(z[kk] - zmax);
+                int yi = (int)Yn[s0 + tid];
+                float inv = 1.0f / sum;
+                for (int kk = 0; kk < k; ++kk) {
+                    float p = __expf(z[kk] - zmax) * inv;
+                    gcur[tid * k + kk] = (p - (kk == yi ? 1.0f : 0.0f)) / m;
+                }
+            }
+            __syncthreads();
+            // ---- backward + SGD, layer by layer (grad buffers ping-pong)
+            float* gnext = grad + m * a.act_max;
+            for (int l = a.n_layers - 1; l >= 0; --l) {
+                int w_off = a.layout[4 * l], b_off = a.layout[4 * l + 1];
+                int fin = a.layout[4 * l + 2], fout = a.layout[4 * l + 3];
+                // activation input of this layer
+                float* ain = act;
+                for (int q = 0; q < l; ++q) ain += m * a.layout[4 * q + 2];
+                // (ain now points at act of layer l's input: layers are
+                //  packed input-first, so offset = m*(d_in + hidden_0 + ...))
+                // grad wrt input (needed before W is updated)
+                if (l > 0) {
+                    for (int e = tid; e < m * fin; e += blockDim.x) {
+                        int s = e / fin, q = e - s * fin;
+                        float acc = 0.f;
+                        for (int o = 0; o < fout; ++o)
+                            acc += gcur[s * fout + o] * W[w_off + o * fin + q];
+                        // ReLU mask of the layer-(l-1) activation
+                        acc *= (ain[s * fin + q] > 0.f) ? 1.0f : 0.0f;
+                        gnext[e] = acc;
+                    }
+                    __syncthreads();
+                }
+                // weight/bias SGD
+                for (int e = tid; e < fout * fin; e += blockDim.x) {
+                    int o = e / fin, q = e - o * fin;
+                    float g = 0.f;
+                    for (int s = 0; s < m; ++s)
+                        g += gcur[s * fout + o] * ain[s * fin + q];
+                    if (a.wd != 0.f) g += a.wd * W[w_off + e];
+                    W[w_off + e] -= a.lr * g;
+                }
+                for (int e = tid; e < fout; e += blockDim.x) {
+                    float g = 0.f;
+                    for (int s = 0; s < m; ++s) g += gcur[s * fout + e];
+                    W[b_off + e] -= a.lr * g;
+                }
+                __syncthreads();
+                float* tmp = gcur; gcur = gnext; gnext = tmp;
+            }
+            age += 1;
+        }
+    }
+}
+
+__global__ void __launch_bounds__(256)
+tick_mlp_kernel(MlpArgs a)
+{
+    int i = blockIdx.x;
+    int node = a.nodes[i];
+    int tid = threadIdx.x;
+    extern __shared__ float sm[];
+    float* W = sm;          // D
+    float* W2 = W + a.D;    // D
+    int bsmax = (a.bs == 0) ? a.Smax : min(a.bs, a.Smax);
+    // act: batch input + every layer's activation; grad: 2 ping-pong buffers
+    float* act = W2 + a.D;
+    int act_total = 0;
+    // act layout computed on host side == bs*(d_in + sum(out_l)); the host
+    // passes act_max = max layer width for the grad buffers
+    for (int l = 0; l < a.n_layers; ++l) act_total += a.layout[4 * l + 3];
+    float* grad = act + bsmax * (a.d_in + act_total);
+
+    for (int e = tid; e < a.D; e += blockDim.x)
+        W[e] = a.params[(long)node * a.D + e];
+    __syncthreads();
+    int age = a.ages[node];
+
+    if (a.update_only) {
+        mlp_update(a, node, W, act, grad, age);
+    } else {
+        for (int j = a.ptr[i]; j < a.ptr[i + 1]; ++j) {
+            int slot = a.dslots[j];
+            const float* srow = a.slots + (long)slot * a.D;
+            int sage = a.slot_ages[slot];
+            if (a.mode == MODE_MERGE_UPDATE) {
+                for (int e = tid; e < a.D; e += blockDim.x)
+                    W[e] = 0.5f * (W[e] + srow[e]);
+                age = max(age, sage);
+                __syncthreads();
+                mlp_update(a, node, W, act, grad, age);
+            } else if (a.mode == MODE_UPDATE) {
+                for (int e = tid; e < a.D; e += blockDim.x) W[e] = srow[e];
+                age = sage;
+                __syncthreads();
+                mlp_update(a, node, W, act, grad, age);
+            } else if (a.mode == MODE_UPDATE_MERGE) {
+                mlp_update(a, node, W, act, grad, age);
+                for (int e = tid; e < a.D; e += blockDim.x) W2[e] = srow[e];
+                __syncthreads();
+                int age2 = sage;
+                mlp_update(a, node, W2, act, grad, age2);
+                for (int e = tid; e < a.D; e += blockDim.x)
+                    W[e] = 0.5f * (W[e] + W2[e]);
+                age = max(age, age2);
+                __syncthreads();
+            } else {
+                for (int e = tid; e < a.D; e += blockDim.x) W[e] = srow[e];
+                age = sage;
+                __syncthreads();
+            }
+            int rs = a.rslots[j];
+            if (rs >= 0) {
+                for (int e = tid; e < a.D; e += blockDim.x)
+                    a.slots[(long)rs * a.D + e] = W[e];
+                if (tid == 0) a.slot_ages[rs] = age;
+                __syncthreads();
+            }
+        }
+    }
+    __syncthreads();
+    for (int e = tid; e < a.D; e += blockDim.x)
+        a.params[(long)node * a.D + e] = W[e];
+    if (tid == 0) a.ages[node] = age;
+}
+
+// ---------------------------------------------------------------------------
+// host wrappers
+// ---------------------------------------------------------------------------
+
+#define CHECK_DEV(t) TORCH_CHECK(t.is_cuda() && t.is_contiguous(), #t " must be contiguous on device")
+
+static hipStream_t current_stream()
+{
+    return at::hip::getCurrentHIPStream().stream();
+}
+
+void snapshot(torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
+              torch::Tensor slot_ages, torch::Tensor nodes, torch::Tensor slot_ids)
+{
+    CHECK_DEV(params); CHECK_DEV(slots); CHECK_DEV(nodes); CHECK_DEV(slot_ids);
+    int n = nodes.size(0);
+    if (n == 0) return;
+    int D = params.size(1);
+    long total = (long)n * D;
+    int block = 256;
+    int grid = (int)std::min<long>((total + block - 1) / block, 2048);
+    hipLaunchKernelGGL(snapshot_kernel, dim3(grid), dim3(block), 0, current_stream(),
+        params.data_ptr<float>(), ages.data_ptr<int>(),
+        slots.data_ptr<float>(), slot_ages.data_ptr<int>(),
+        nodes.data_ptr<int>(), slot_ids.data_ptr<int>(), n, D);
+}
+
+void tick_logreg(torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
+                 torch::Tensor slot_ages, torch::Tensor nodes, torch::Tensor recv_ptr,
+                 torch::Tensor del_slots, torch::Tensor reply_slots,
+                 torch::Tensor X, torch::Tensor Y, torch::Tensor counts,
+                 int64_t d, int64_t k, double lr, double wd, int64_t epochs,
+                 int64_t bs, int64_t mode, bool update_only)
+{
+    CHECK_DEV(params); CHECK_DEV(slots); CHECK_DEV(X); CHECK_DEV(Y);
+    TORCH_CHECK(k <= KMAX, "n_classes > ", KMAX, " unsupported");
+    int n = nodes.size(0);
+    if (n == 0) return;
+    LogregArgs a;
+    a.params = params.data_ptr<float>(); a.ages = ages.data_ptr<int>();
+    a.slots = slots.data_ptr<float>(); a.slot_ages = slot_ages.data_ptr<int>();
+    a.nodes = nodes.data_ptr<int>(); a.ptr = recv_ptr.data_ptr<int>();
+    a.dslots = del_slots.numel() ? del_slots.data_ptr<int>() : nullptr;
+    a.rslots = reply_slots.numel() ? reply_slots.data_ptr<int>() : nullptr;
+    a.X = X.data_ptr<float>(); a.Y = Y.data_ptr<float>();
+    a.counts = counts.data_ptr<int>();
+    a.d = d; a.k = k; a.Smax = X.size(1); a.D = params.size(1);
+    a.lr = lr; a.wd = wd; a.epochs = epochs; a.bs = bs; a.mode = mode;
+    a.update_only = update_only;
+    int bsmax = (bs == 0) ? a.Smax : std::min<int>(bs, a.Smax);
+    size_t smem = sizeof(float) * (2 * a.D + (size_t)bsmax * a.d + (size_t)bsmax * a.k);
+    TORCH_CHECK(smem <= 160 * 1024, "logreg LDS budget exceeded: ", smem);
+    hipLaunchKernelGGL(tick_logreg_kernel, dim3(n), dim3(128), smem,
+                       current_stream(), a);
+}
+
+void tick_linear(torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
+                 torch::Tensor slot_ages, torch::Tensor nodes, torch::Tensor recv_ptr,
+                 torch::Tensor del_slots, torch::Tensor reply_slots,
+                 torch::Tensor X, torch::Tensor Y, torch::Tensor counts,
+                 int64_t d, double lrlam, int64_t is_pegasos, int64_t mode,
+                 bool update_only)
+{
+    CHECK_DEV(params); CHECK_DEV(slots); CHECK_DEV(X); CHECK_DEV(Y);
+    TORCH_CHECK(d <= LIN_MAX_REGS * WAVE, "d > ", LIN_MAX_REGS * WAVE, " unsupported");
+    int n = nodes.size(0);
+    if (n == 0) return;
+    LinearArgs a;
+    a.params = params.data_ptr<float>(); a.ages = ages.data_ptr<int>();
+    a.slots = slots.data_ptr<float>(); a.slot_ages = slot_ages.data_ptr<int>();
+    a.nodes = nodes.data_ptr<int>(); a.ptr = recv_ptr.data_ptr<int>();
+    a.dslots = del_slots.numel() ? del_slots.data_ptr<int>() : nullptr;
+    a.rslots = reply_slots.numel() ? reply_slots.data_ptr<int>() : nullptr;
+    a.X = X.data_ptr<float>(); a.Y = Y.data_ptr<float>();
+    a.counts = counts.data_ptr<int>();
+    a.d = d; a.Smax = X.size(1);
+    a.lrlam = lrlam; a.is_pegasos = is_pegasos; a.mode = mode;
+    a.update_only = update_only;
+    hipLaunchKernelGGL(tick_linear_kernel, dim3(n), dim3(WAVE), 0,
+                       current_stream(), a);
+}
+
+void tick_mlp(torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
+              torch::Tensor slot_ages, torch::Tensor nodes, torch::Tensor recv_ptr,
+              torch::Tensor del_slots, torch::Tensor reply_slots,
+              torch::Tensor X, torch::Tensor Y, torch::Tensor counts,
+              torch::Tensor layout, int64_t n_layers, double lr, double wd,
+              int64_t epochs, int64_t bs, int64_t mode, bool update_only)
+{
+    CHECK_DEV(params); CHECK_DEV(slots); CHECK_DEV(X); CHECK_DEV(Y);
+    CHECK_DEV(layout);
+    int n = nodes.size(0);
+    if (n == 0) return;
+    MlpArgs a;
+    a.params = params.data_ptr<float>(); a.ages = ages.data_ptr<int>();
+    a.slots = slots.data_ptr<float>(); a.slot_ages = slot_ages.data_ptr<int>();
+    a.nodes = nodes.data_ptr<int>(); a.ptr = recv_ptr.data_ptr<int>();
+    a.dslots = del_slots.numel() ? del_slots.data_ptr<int>() : nullptr;
+    a.rslots = reply_slots.numel() ? reply_slots.data_ptr<int>() : nullptr;
+    a.X = X.data_ptr<float>(); a.Y = Y.data_ptr<float>();
+    a.counts = counts.data_ptr<int>();
+    a.layout = layout.data_ptr<int>();
+    a.n_layers = n_layers; a.Smax = X.size(1); a.D = params.size(1);
+    a.d_in = layout[2].item<int>();
+    a.lr = lr; a.wd = wd; a.epochs = epochs; a.bs = bs; a.mode = mode;
+    a.update_only = update_only;
+    // LDS: 2*D (model + scratch) + activations + 2 grad buffers
+    auto lay = layout.cpu();
+    const int* L = lay.data_ptr<int>();
+    int act_sum = 0, act_max = L[2];
+    for (int l = 0; l < n_layers; ++l) {
+        act_sum += L[4 * l + 3];
+        act_max = std::max(act_max, L[4 * l + 3]);
+        act_max = std::max(act_max, L[4 * l + 2]);
+    }
+    a.act_max = act_max;
+    a.d_in = L[2];
+    int bsmax = (bs == 0) ? a.Smax : std::min<int>(bs, a.Smax);
+    size_t smem = sizeof(float) *
+        (2 * (size_t)a.D + (size_t)bsmax * (a.d_in + act_sum) +
+         2 * (size_t)bsmax * act_max);
+    TORCH_CHECK(smem <= 160 * 1024,
+        "mlp LDS budget exceeded (", smem, " B); shrink batch_size/hidden");
+    hipLaunchKernelGGL(tick_mlp_kernel, dim3(n), dim3(256), smem,
+                       current_stream(), a);
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m)
+{
+    m.def("snapshot", &snapshot, "batched model snapshot (arena row copy)");
+    m.def("tick_logreg", &tick_logreg, "fused merge + logreg SGD tick");
+    m.def("tick_linear", &tick_linear, "fused merge + pegasos/adaline tick");
+    m.def("tick_mlp", &tick_mlp, "fused merge + MLP SGD tick");
+}

@@ -1,0 +1,219 @@
+"""GPU tests (MI355X): HIP kernel numerics vs the fp32 torch oracle, and
+end-to-end engine equivalence CPU <-> GPU. All marked ``gpu``."""
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from gossipy_amd.core import AntiEntropyProtocol, CreateModelMode
+from gossipy_amd.data import make_synthetic_classification
+from gossipy_amd.engine import (
+    BatchedGossipSimulator,
+    DataArena,
+    EngineConfig,
+    LogRegSpec,
+    MLPSpec,
+    NodeStateArena,
+    PegasosSpec,
+    AdaLineSpec,
+    RandomTape,
+    SlotPool,
+)
+from gossipy_amd.engine.backend import HIPBackend, TorchBackend
+from gossipy_amd.simul import SimulationReport
+
+CUDA = torch.device("cuda:0")
+CPU = torch.device("cpu")
+
+
+def _mk_state(n, spec, device, seed=0):
+    state = NodeStateArena(n, spec.D, device)
+    tape = RandomTape(seed)
+    TorchBackend().init_params(state, spec, tape, n)
+    state.ages += torch.arange(n, device=device, dtype=torch.int32) % 5
+    return state
+
+
+def _mk_data(n, d, device, seed=0, pm1=False, samples=160):
+    X, y = make_synthetic_classification((samples, d, 2), seed=seed)
+    if pm1:
+        y = 2 * y.float() - 1
+    shards = [(X[s], y[s]) for s in np.array_split(np.arange(samples), n)]
+    return DataArena.from_shards(shards, device, global_eval=(X, y))
+
+
+def _pair(n, spec, d, pm1=False, seed=3):
+    """(cpu, gpu) copies of identical state/data."""
+    cs = _mk_state(n, spec, CPU, seed)
+    gs = NodeStateArena(n, spec.D, CUDA)
+    gs.params.copy_(cs.params)
+    gs.ages.copy_(cs.ages)
+    cd = _mk_data(n, d, CPU, seed, pm1)
+    gd = DataArena(
+        cd.x.to(CUDA), cd.y.to(CUDA), cd.counts.to(CUDA),
+        gx=cd.gx.to(CUDA), gy=cd.gy.to(CUDA),
+    )
+    return cs, gs, cd, gd
+
+
+def _close(a, b, tol=1e-4):
+    return torch.allclose(a.cpu(), b.cpu(), atol=tol, rtol=tol)
+
+
+class TestKernelNumerics:
+    def test_extension_loads(self):
+        from gossipy_amd import ops
+
+        ext = ops.load_extension()
+        assert hasattr(ext, "tick_logreg")
+
+    def test_snapshot(self):
+        spec = LogRegSpec(d_in=57)
+        cs, gs, cd, gd = _pair(16, spec, 57)
+        cpool = SlotPool(spec.D, CPU, 32)
+        gpool = SlotPool(spec.D, CUDA, 32)
+        nodes = torch.tensor([3, 7, 11], dtype=torch.int64)
+        slots = torch.tensor([0, 5, 9], dtype=torch.int64)
+        TorchBackend().snapshot(cs, cpool, nodes, slots)
+        HIPBackend().snapshot(gs, gpool, nodes.to(CUDA), slots.to(CUDA))
+        torch.cuda.synchronize()
+        assert _close(cpool.slots[:10], gpool.slots[:10], 0)
+        assert torch.equal(cpool.slot_ages[:10].cpu(), gpool.slot_ages[:10].cpu())
+
+    @pytest.mark.parametrize("mode", [
+        CreateModelMode.MERGE_UPDATE,
+        CreateModelMode.UPDATE,
+        CreateModelMode.UPDATE_MERGE,
+    ])
+    def test_logreg_deliver_matches_oracle(self, mode):
+        spec = LogRegSpec(d_in=57, n_classes=2, lr=0.1, mode=mode)
+        cs, gs, cd, gd = _pair(12, spec, 57)
+        cpool, gpool = SlotPool(spec.D, CPU, 16), SlotPool(spec.D, CUDA, 16)
+        cpool.slots.normal_(generator=torch.Generator().manual_seed(5))
+        cpool.slot_ages.copy_(torch.arange(16, dtype=torch.int32) * 2)
+        gpool.slots.copy_(cpool.slots)
+        gpool.slot_ages.copy_(cpool.slot_ages)
+        recv = torch.tensor([2, 5, 9], dtype=torch.int64)
+        ptr = torch.tensor([0, 2, 3, 4], dtype=torch.int64)  # node2 gets 2 msgs
+        slots = torch.tensor([1, 4, 7, 10], dtype=torch.int64)
+        reply = torch.tensor([12, -1, 13, -1], dtype=torch.int64)
+        TorchBackend().deliver(cs, cpool, cd, spec, recv, ptr, slots, reply)
+        HIPBackend().deliver(gs, gpool, gd, spec, recv, ptr, slots, reply)
+        torch.cuda.synchronize()
+        assert _close(cs.params, gs.params)
+        assert torch.equal(cs.ages.cpu(), gs.ages.cpu())
+        assert _close(cpool.slots[12:14], gpool.slots[12:14])
+
+    def test_logreg_update_only(self):
+        spec = LogRegSpec(d_in=57, n_classes=2, lr=0.1, local_epochs=2)
+        cs, gs, cd, gd = _pair(10, spec, 57)
+        nodes = torch.arange(10)
+        TorchBackend().update(cs, cd, spec, nodes)
+        HIPBackend().update(gs, gd, spec, nodes)
+        torch.cuda.synchronize()
+        assert _close(cs.params, gs.params)
+        assert torch.equal(cs.ages.cpu(), gs.ages.cpu())
+
+    def test_pegasos_matches_oracle(self):
+        spec = PegasosSpec(d_in=57, lam=0.01, mode=CreateModelMode.MERGE_UPDATE)
+        cs, gs, cd, gd = _pair(8, spec, 57, pm1=True)
+        cs.params.normal_(generator=torch.Generator().manual_seed(1))
+        gs.params.copy_(cs.params)
+        cpool, gpool = SlotPool(spec.D, CPU, 8), SlotPool(spec.D, CUDA, 8)
+        cpool.slots.normal_(generator=torch.Generator().manual_seed(2))
+        gpool.slots.copy_(cpool.slots)
+        recv = torch.tensor([1, 4], dtype=torch.int64)
+        ptr = torch.tensor([0, 1, 2], dtype=torch.int64)
+        slots = torch.tensor([0, 3], dtype=torch.int64)
+        reply = torch.tensor([5, -1], dtype=torch.int64)
+        TorchBackend().deliver(cs, cpool, cd, spec, recv, ptr, slots, reply)
+        HIPBackend().deliver(gs, gpool, gd, spec, recv, ptr, slots, reply)
+        torch.cuda.synchronize()
+        assert _close(cs.params, gs.params)
+        assert torch.equal(cs.ages.cpu(), gs.ages.cpu())
+
+    def test_adaline_matches_oracle(self):
+        spec = AdaLineSpec(d_in=57, lr=0.05, mode=CreateModelMode.UPDATE)
+        cs, gs, cd, gd = _pair(8, spec, 57, pm1=True)
+        cpool, gpool = SlotPool(spec.D, CPU, 8), SlotPool(spec.D, CUDA, 8)
+        cpool.slots.normal_(generator=torch.Generator().manual_seed(2))
+        gpool.slots.copy_(cpool.slots)
+        recv = torch.tensor([0, 6], dtype=torch.int64)
+        ptr = torch.tensor([0, 1, 2], dtype=torch.int64)
+        slots = torch.tensor([2, 4], dtype=torch.int64)
+        reply = torch.tensor([-1, -1], dtype=torch.int64)
+        TorchBackend().deliver(cs, cpool, cd, spec, recv, ptr, slots, reply)
+        HIPBackend().deliver(gs, gpool, gd, spec, recv, ptr, slots, reply)
+        torch.cuda.synchronize()
+        assert _close(cs.params, gs.params)
+
+    def test_mlp_matches_oracle(self):
+        spec = MLPSpec(d_in=20, n_classes=3, hidden=(32,), lr=0.05)
+        cs, gs, cd, gd = _pair(6, spec, 20)
+        # 3-class shards
+        X, y = make_synthetic_classification((120, 20, 3), seed=9)
+        shards = [(X[s], y[s]) for s in np.array_split(np.arange(120), 6)]
+        cd = DataArena.from_shards(shards, CPU, global_eval=(X, y))
+        gd = DataArena(cd.x.to(CUDA), cd.y.to(CUDA), cd.counts.to(CUDA),
+                       gx=cd.gx.to(CUDA), gy=cd.gy.to(CUDA))
+        cpool, gpool = SlotPool(spec.D, CPU, 8), SlotPool(spec.D, CUDA, 8)
+        cpool.slots.normal_(generator=torch.Generator().manual_seed(2))
+        cpool.slots.mul_(0.1)
+        gpool.slots.copy_(cpool.slots)
+        recv = torch.tensor([1, 3], dtype=torch.int64)
+        ptr = torch.tensor([0, 1, 2], dtype=torch.int64)
+        slots = torch.tensor([0, 1], dtype=torch.int64)
+        reply = torch.tensor([-1, 6], dtype=torch.int64)
+        TorchBackend().deliver(cs, cpool, cd, spec, recv, ptr, slots, reply)
+        HIPBackend().deliver(gs, gpool, gd, spec, recv, ptr, slots, reply)
+        torch.cuda.synchronize()
+        assert _close(cs.params, gs.params, 1e-3)
+        assert torch.equal(cs.ages.cpu(), gs.ages.cpu())
+
+
+class TestEngineGPU:
+    def _run(self, device, n_nodes=64, rounds=5):
+        X, y = make_synthetic_classification((640, 57, 2), seed=0, margin=2.0)
+        idx = np.random.default_rng(0).permutation(640)
+        shards = [(X[s], y[s]) for s in np.array_split(idx[:576], n_nodes)]
+        data = DataArena.from_shards(
+            shards, device, global_eval=(X[idx[576:]], y[idx[576:]])
+        )
+        cfg = EngineConfig(
+            n_nodes=n_nodes,
+            delta=10,
+            protocol=AntiEntropyProtocol.PUSH_PULL,
+            model_size=116,
+            sampling_eval=0.25,
+            seed=1,
+        )
+        sim = BatchedGossipSimulator(
+            cfg, LogRegSpec(d_in=57, n_classes=2, lr=0.1), data, device=device
+        )
+        rep = SimulationReport()
+        sim.add_receiver(rep)
+        sim.init_nodes()
+        sim.start(n_rounds=rounds)
+        if device.type == "cuda":
+            torch.cuda.synchronize()
+        return sim, rep
+
+    def test_gpu_uses_hip_backend(self):
+        sim, _ = self._run(CUDA, rounds=1)
+        assert sim.backend.name == "hip"
+
+    def test_gpu_matches_cpu_run(self):
+        gsim, grep = self._run(CUDA)
+        csim, crep = self._run(CPU)
+        assert torch.allclose(
+            gsim.local_params().cpu(), csim.local_params(), atol=1e-3, rtol=1e-3
+        )
+        ga = grep.get_evaluation(False)[-1][1]["accuracy"]
+        ca = crep.get_evaluation(False)[-1][1]["accuracy"]
+        assert abs(ga - ca) < 0.05
+
+    def test_gpu_learns(self):
+        sim, rep = self._run(CUDA, rounds=15)
+        assert rep.get_evaluation(False)[-1][1]["accuracy"] > 0.9
