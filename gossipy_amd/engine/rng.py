@@ -3,12 +3,17 @@
 The object layer consumes numpy's *global* RNG in node-iteration order
 (gossipy/simul.py:390-409), which makes results depend on how nodes are
 scheduled. The batched engine instead derives every random decision from a
-Philox counter keyed on ``(seed, purpose, timestep)``, so the full event
-schedule — peer choices, drop/online coin flips, delays, timeout offsets,
-evaluation samples — is a pure function of the seed. Every rank replays the
-same tape, which makes the schedule *identical for any GPU count and any
-node->GPU residency map*: a 1-GPU run and an 8-GPU run execute the same
-simulation event-for-event (SURVEY.md §7 hard-part 3).
+counter-based splitmix64 stream keyed on ``(seed, purpose, t, extra)``, so
+the full event schedule — peer choices, drop/online coin flips, delays,
+timeout offsets, evaluation samples — is a pure function of the seed. Every
+rank replays the same tape, making the schedule *identical for any GPU
+count and any node->GPU residency map* (SURVEY.md §7 hard-part 3).
+
+The generator is splitmix64 (public-domain finalizer constants), chosen
+over numpy's Philox because the exact draw sequence must be reproducible
+bit-for-bit by the native C++ scheduler (``csrc/scheduler.cpp``) — both
+implementations share these few lines of integer arithmetic, which numpy
+replicates exactly with uint64 wraparound ops.
 """
 
 from __future__ import annotations
@@ -17,7 +22,20 @@ from enum import IntEnum
 
 import numpy as np
 
-__all__ = ["Purpose", "RandomTape"]
+__all__ = ["Purpose", "RandomTape", "TapeStream"]
+
+_GOLDEN = np.uint64(0x9E3779B97F4A7C15)
+_MIX1 = np.uint64(0xBF58476D1CE4E5B9)
+_MIX2 = np.uint64(0x94D049BB133111EB)
+_U53_INV = 1.0 / 9007199254740992.0  # 2^-53
+
+
+def _splitmix64(x: np.ndarray) -> np.ndarray:
+    """Vectorized splitmix64 finalizer (uint64 in, uint64 out)."""
+    x = (x + _GOLDEN).astype(np.uint64)
+    x = ((x ^ (x >> np.uint64(30))) * _MIX1).astype(np.uint64)
+    x = ((x ^ (x >> np.uint64(27))) * _MIX2).astype(np.uint64)
+    return (x ^ (x >> np.uint64(31))).astype(np.uint64)
 
 
 class Purpose(IntEnum):
@@ -35,23 +53,63 @@ class Purpose(IntEnum):
     MISC = 9
 
 
-class RandomTape:
-    """Deterministic per-(purpose, timestep) random streams.
+class TapeStream:
+    """Sequential draws from one keyed stream: draw ``i`` is
+    ``splitmix64(key + i)``; the instance keeps a running offset."""
 
-    Each call materializes a fresh ``np.random.Generator`` seeded by the
-    Philox key ``(seed, purpose, t)`` — cheap (µs) and stateless, so any rank
-    can draw any slice of the tape in any order and get identical values.
-    """
+    __slots__ = ("key", "_off")
+
+    def __init__(self, key: int):
+        self.key = np.uint64(key)
+        self._off = 0
+
+    def _raw(self, n: int) -> np.ndarray:
+        idx = np.arange(self._off, self._off + n, dtype=np.uint64)
+        self._off += n
+        return _splitmix64(self.key + idx)
+
+    def random(self, n: int = 1):
+        """``n`` float64 uniforms in [0, 1) (53-bit)."""
+        u = (self._raw(n) >> np.uint64(11)).astype(np.float64) * _U53_INV
+        return u if n != 1 else float(u[0])
+
+    def integers(self, low: int, high: int, size: int = 1):
+        """``size`` ints in [low, high) (scaled-double method — tiny bias,
+        chosen for trivial cross-language reproducibility)."""
+        u = (self._raw(size) >> np.uint64(11)).astype(np.float64) * _U53_INV
+        v = low + np.floor(u * (high - low)).astype(np.int64)
+        return v if size != 1 else int(v[0])
+
+    def uniform(self, low: float, high: float, size) -> np.ndarray:
+        """Uniform floats in [low, high); ``size`` may be a tuple."""
+        n = int(np.prod(size))
+        u = (self._raw(n) >> np.uint64(11)).astype(np.float64) * _U53_INV
+        return (low + u * (high - low)).reshape(size)
+
+    def normal(self, mu: float, sigma: float, size: int) -> np.ndarray:
+        """Box–Muller normals (pairs of uniforms; cos branch only)."""
+        u = (self._raw(2 * size) >> np.uint64(11)).astype(np.float64) * _U53_INV
+        u1 = np.maximum(u[0::2], 1e-300)
+        u2 = u[1::2]
+        z = np.sqrt(-2.0 * np.log(u1)) * np.cos(2.0 * np.pi * u2)
+        return mu + sigma * z
+
+
+class RandomTape:
+    """Deterministic per-(purpose, timestep, extra) streams."""
 
     def __init__(self, seed: int):
-        self.seed = int(seed)
+        self.seed = int(seed) & 0xFFFFFFFFFFFFFFFF
 
-    def stream(self, purpose: Purpose, t: int = 0, extra: int = 0) -> np.random.Generator:
-        """The generator for ``(purpose, t, extra)`` (Philox 2x64 key)."""
-        k0 = (self.seed << 8) ^ int(purpose)
-        k1 = (int(t) << 20) ^ int(extra)
-        key = np.array([k0 & 0xFFFFFFFFFFFFFFFF, k1 & 0xFFFFFFFFFFFFFFFF], dtype=np.uint64)
-        return np.random.Generator(np.random.Philox(key=key))
+    def stream_key(self, purpose: Purpose, t: int = 0, extra: int = 0) -> int:
+        k = np.uint64(self.seed)
+        k = _splitmix64(np.uint64([k ^ np.uint64(int(purpose))]))[0]
+        k = _splitmix64(np.uint64([k ^ np.uint64(int(t))]))[0]
+        k = _splitmix64(np.uint64([k ^ np.uint64(int(extra))]))[0]
+        return int(k)
+
+    def stream(self, purpose: Purpose, t: int = 0, extra: int = 0) -> TapeStream:
+        return TapeStream(self.stream_key(purpose, t, extra))
 
     # -- convenience draws used by the scheduler -----------------------------
 
@@ -59,8 +117,8 @@ class RandomTape:
         self, purpose: Purpose, t: int, n: int, low: int, high: int
     ) -> np.ndarray:
         """``n`` ints in ``[low, high)``."""
-        return self.stream(purpose, t).integers(low, high, size=n)
+        return np.atleast_1d(self.stream(purpose, t).integers(low, high, size=n))
 
     def uniform(self, purpose: Purpose, t: int, n: int) -> np.ndarray:
         """``n`` floats in ``[0, 1)``."""
-        return self.stream(purpose, t).random(n)
+        return np.atleast_1d(self.stream(purpose, t).random(n))

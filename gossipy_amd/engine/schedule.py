@@ -130,7 +130,9 @@ class Scheduler:
         # and crash on t % 0 — SURVEY.md §2.3 quirk 16)
         g = self.tape.stream(Purpose.TIMEOUT)
         if cfg.sync:
-            self.deltas = g.integers(0, cfg.delta, size=n).astype(np.int64)
+            self.deltas = np.atleast_1d(g.integers(0, cfg.delta, size=n)).astype(
+                np.int64
+            )
         else:
             self.deltas = np.maximum(
                 1, g.normal(cfg.delta, cfg.delta / 10, size=n).astype(np.int64)
@@ -178,12 +180,12 @@ class Scheduler:
             return np.empty(0, dtype=np.int64)
         g = self.tape.stream(Purpose.PEER, t)
         if cfg.peers_indptr is None:
-            draw = g.integers(0, cfg.n_nodes - 1, size=n)
+            draw = np.atleast_1d(g.integers(0, cfg.n_nodes - 1, size=n))
             return draw + (draw >= nodes)  # skip self
         starts = cfg.peers_indptr[nodes]
         degs = cfg.peers_indptr[nodes + 1] - starts
         # nodes with no peers never appear here (runner validates topology)
-        offs = np.floor(g.random(n) * degs).astype(np.int64)
+        offs = np.floor(np.atleast_1d(g.random(n)) * degs).astype(np.int64)
         return cfg.peers_indices[starts + offs]
 
     def _delays(self, t: int, n: int, sizes: np.ndarray) -> np.ndarray:
@@ -192,7 +194,7 @@ class Scheduler:
             return np.full(n, d._delay, dtype=np.int64)
         if isinstance(d, UniformDelay):
             g = self.tape.stream(Purpose.DELAY, t)
-            return g.integers(d._min_delay, d._max_delay + 1, size=n)
+            return np.atleast_1d(g.integers(d._min_delay, d._max_delay + 1, size=n))
         if isinstance(d, LinearDelay):
             return (d._timexunit * sizes).astype(np.int64) + d._overhead
         # custom Delay subclass: fall back to per-message scalar calls
@@ -343,7 +345,7 @@ class Scheduler:
         if cfg.sampling_eval > 0:
             g = self.tape.stream(Purpose.EVAL, t1 - 1)
             k = max(int(cfg.n_nodes * cfg.sampling_eval), 1)
-            eval_nodes = g.choice(cfg.n_nodes, size=k, replace=True)
+            eval_nodes = np.atleast_1d(g.integers(0, cfg.n_nodes, size=k))
 
         return RoundSchedule(
             round_idx=r,
