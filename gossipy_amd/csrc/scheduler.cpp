@@ -1,0 +1,423 @@
+// Native event scheduler for the batched gossip engine (CPU, no GPU deps).
+//
+// Bit-exact C++ replica of gossipy_amd/engine/schedule.py — same splitmix64
+// tape (engine/rng.py), same draw order, same slot free-list discipline —
+// but ~100x faster, and it emits the round as FLAT packed arrays (one
+// upload per round) for the GPU round executor instead of per-tick Python
+// objects. Python<->C++ equivalence is enforced by
+// tests/test_native_sched.py on every CPU test run.
+//
+// Output layout per round (all int32 numpy arrays; `delta` ticks):
+//   snap_nodes/snap_slots + snap_tptr[delta+1]      sub-phase A per tick
+//   recv_nodes + recv_nptr (global CSR over msgs)
+//     + recv_tptr[delta+1] (tick -> receiver range)  sub-phase B
+//   del_slots/del_owners/reply_slots  (aligned with recv_nptr[-1] msgs)
+//   pull_nodes/pull_slots + pull_tptr[delta+1]       PULL snapshots
+//   rep_* (same structure as recv_*)                 sub-phase C
+//   eval_nodes; scalars sent/failed/total_size/n_slots(high water)
+
+#include <pybind11/numpy.h>
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <algorithm>
+#include <cmath>
+#include <cstdint>
+#include <map>
+#include <unordered_map>
+#include <vector>
+
+namespace py = pybind11;
+
+// ---------------------------------------------------------------------------
+// tape (must match engine/rng.py exactly)
+// ---------------------------------------------------------------------------
+
+static inline uint64_t splitmix64(uint64_t x)
+{
+    x += 0x9E3779B97F4A7C15ULL;
+    x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ULL;
+    x = (x ^ (x >> 27)) * 0x94D049BB133111EBULL;
+    return x ^ (x >> 31);
+}
+
+struct Stream {
+    uint64_t key;
+    uint64_t off = 0;
+    explicit Stream(uint64_t k) : key(k) {}
+    inline uint64_t raw() { return splitmix64(key + off++); }
+    inline double rnd()
+    {
+        return (double)(raw() >> 11) * (1.0 / 9007199254740992.0);
+    }
+    inline int64_t integers(int64_t lo, int64_t hi)
+    {
+        return lo + (int64_t)std::floor(rnd() * (double)(hi - lo));
+    }
+    inline double normal(double mu, double sigma)
+    {
+        double u1 = std::max(rnd(), 1e-300);
+        double u2 = rnd();
+        return mu + sigma * std::sqrt(-2.0 * std::log(u1)) *
+                        std::cos(2.0 * M_PI * u2);
+    }
+};
+
+enum Purpose {
+    TIMEOUT = 0, PEER = 1, DROP = 2, ONLINE = 3, DELAY = 4,
+    EVAL = 5, INIT = 6, DATA = 7, TOKEN = 8, MISC = 9
+};
+
+static inline uint64_t stream_key(uint64_t seed, uint64_t purpose, uint64_t t,
+                                  uint64_t extra)
+{
+    uint64_t k = seed;
+    k = splitmix64(k ^ purpose);
+    k = splitmix64(k ^ t);
+    k = splitmix64(k ^ extra);
+    return k;
+}
+
+// ---------------------------------------------------------------------------
+// config / protocol enums (numbering matches gossipy_amd.core)
+// ---------------------------------------------------------------------------
+
+enum Proto { PUSH = 1, PULL = 2, PUSH_PULL = 3 };
+enum DelayKind { D_CONST = 0, D_UNIFORM = 1, D_LINEAR = 2 };
+
+struct Msg {  // matches the python pending tuple
+    int32_t receiver;
+    int32_t slot;       // -1 for PULL requests
+    int32_t reply_flag; // -2 = wants reply, -1 = plain
+    bool is_pull;
+    int32_t sender;
+};
+
+class NativeScheduler {
+public:
+    NativeScheduler(int64_t n_nodes, int64_t delta, int proto,
+                    int64_t model_size, double drop_prob, double online_prob,
+                    int delay_kind, int64_t dmin, int64_t dmax,
+                    double timexunit, int64_t overhead, bool sync,
+                    double sampling_eval, uint64_t seed,
+                    py::object peers_indptr, py::object peers_indices)
+        : n_(n_nodes), delta_(delta), proto_(proto), model_size_(model_size),
+          drop_(drop_prob), online_(online_prob), dkind_(delay_kind),
+          dmin_(dmin), dmax_(dmax), timexunit_(timexunit), overhead_(overhead),
+          sync_(sync), sampling_eval_(sampling_eval), seed_(seed)
+    {
+        Stream g(stream_key(seed_, TIMEOUT, 0, 0));
+        deltas_.resize(n_);
+        if (sync_) {
+            for (int64_t i = 0; i < n_; ++i)
+                deltas_[i] = g.integers(0, delta_);
+        } else {
+            for (int64_t i = 0; i < n_; ++i)
+                deltas_[i] = std::max<int64_t>(
+                    1, (int64_t)g.normal((double)delta_, (double)delta_ / 10.0));
+        }
+        if (!peers_indptr.is_none()) {
+            auto ip = peers_indptr.cast<py::array_t<int64_t>>();
+            auto ix = peers_indices.cast<py::array_t<int64_t>>();
+            indptr_.assign(ip.data(), ip.data() + ip.size());
+            indices_.assign(ix.data(), ix.data() + ix.size());
+        }
+        slot_owner_.resize(64);
+    }
+
+    py::dict next_round(int64_t r);
+
+private:
+    int64_t n_, delta_;
+    int proto_;
+    int64_t model_size_;
+    double drop_, online_;
+    int dkind_;
+    int64_t dmin_, dmax_;
+    double timexunit_;
+    int64_t overhead_;
+    bool sync_;
+    double sampling_eval_;
+    uint64_t seed_;
+    std::vector<int64_t> deltas_;
+    std::vector<int64_t> indptr_, indices_;
+    std::unordered_map<int64_t, std::vector<Msg>> pending_;
+    std::vector<int32_t> free_slots_;
+    int64_t next_slot_ = 0;
+    std::vector<int32_t> slot_owner_;
+
+    int32_t alloc_slot(int32_t owner)
+    {
+        int32_t s;
+        if (!free_slots_.empty()) {
+            s = free_slots_.back();
+            free_slots_.pop_back();
+        } else {
+            s = (int32_t)next_slot_++;
+            if (next_slot_ > (int64_t)slot_owner_.size())
+                slot_owner_.resize(slot_owner_.size() * 2);
+        }
+        slot_owner_[s] = owner;
+        return s;
+    }
+
+    int64_t delay_for(Stream& g, int64_t size)
+    {
+        switch (dkind_) {
+        case D_CONST: return dmin_;
+        case D_UNIFORM: return g.integers(dmin_, dmax_ + 1);
+        default: return (int64_t)(timexunit_ * (double)size) + overhead_;
+        }
+    }
+
+    // returns true if the reply was enqueued (false = dropped)
+    bool enqueue_reply(int64_t t, int32_t replier, int32_t requester,
+                       int32_t slot, int64_t& sent, int64_t& failed,
+                       int64_t& total_size)
+    {
+        Stream g(stream_key(seed_, DROP, (uint64_t)t, (uint64_t)(1 + replier)));
+        double u = g.rnd();
+        sent += 1;
+        total_size += model_size_;
+        if (u > drop_) {
+            Stream gd(stream_key(seed_, DELAY, (uint64_t)t,
+                                 (uint64_t)(1 + replier)));
+            int64_t dly = delay_for(gd, model_size_);
+            pending_[t + dly].push_back({requester, slot, -1, false, replier});
+            return true;
+        }
+        failed += 1;
+        return false;
+    }
+};
+
+py::dict NativeScheduler::next_round(int64_t r)
+{
+    const int64_t t0 = r * delta_, t1 = (r + 1) * delta_;
+    int64_t sent = 0, failed = 0, total_size = 0;
+
+    std::vector<int32_t> snap_nodes, snap_slots, snap_tptr{0};
+    std::vector<int32_t> recv_nodes, recv_tptr{0};
+    std::vector<int32_t> recv_nptr{0};
+    std::vector<int32_t> del_slots, del_owners, reply_slots;
+    std::vector<int32_t> pull_nodes, pull_slots, pull_tptr{0};
+    std::vector<int32_t> rep_nodes, rep_tptr{0}, rep_nptr{0};
+    std::vector<int32_t> rep_slots, rep_owners;
+
+    std::vector<double> online(n_);
+    std::vector<int32_t> firing;
+    firing.reserve(64);
+
+    for (int64_t t = t0; t < t1; ++t) {
+        std::vector<int32_t> freed;
+
+        // --- firing set (ascending node id, like np.where)
+        firing.clear();
+        if (sync_) {
+            int64_t phase = t % delta_;
+            for (int64_t i = 0; i < n_; ++i)
+                if (deltas_[i] == phase) firing.push_back((int32_t)i);
+        } else {
+            for (int64_t i = 0; i < n_; ++i)
+                if (t % deltas_[i] == 0) firing.push_back((int32_t)i);
+        }
+
+        // --- sends (gossipy/simul.py:393-407 equivalent)
+        size_t n_f = firing.size();
+        if (n_f) {
+            Stream gp(stream_key(seed_, PEER, (uint64_t)t, 0));
+            std::vector<int32_t> peers(n_f);
+            if (indptr_.empty()) {
+                for (size_t j = 0; j < n_f; ++j) {
+                    int64_t draw = gp.integers(0, n_ - 1);
+                    peers[j] = (int32_t)(draw + (draw >= firing[j] ? 1 : 0));
+                }
+            } else {
+                // python draws ALL uniforms in one batch before indexing
+                std::vector<double> u(n_f);
+                for (size_t j = 0; j < n_f; ++j) u[j] = gp.rnd();
+                for (size_t j = 0; j < n_f; ++j) {
+                    int64_t s = indptr_[firing[j]];
+                    int64_t deg = indptr_[firing[j] + 1] - s;
+                    peers[j] = (int32_t)indices_[s + (int64_t)std::floor(
+                                                          u[j] * (double)deg)];
+                }
+            }
+            Stream gdrop(stream_key(seed_, DROP, (uint64_t)t, 0));
+            std::vector<double> drop_u(n_f);
+            for (size_t j = 0; j < n_f; ++j) drop_u[j] = gdrop.rnd();
+            int64_t msize = (proto_ == PULL) ? 1 : model_size_;
+            std::vector<int64_t> delays(n_f);
+            {
+                Stream gdl(stream_key(seed_, DELAY, (uint64_t)t, 0));
+                for (size_t j = 0; j < n_f; ++j) delays[j] = delay_for(gdl, msize);
+            }
+            for (size_t j = 0; j < n_f; ++j) {
+                int32_t sender = firing[j], receiver = peers[j];
+                bool is_pull = proto_ == PULL;
+                int32_t slot = -1;
+                if (!is_pull) {
+                    slot = alloc_slot(sender);
+                    snap_nodes.push_back(sender);
+                    snap_slots.push_back(slot);
+                }
+                sent += 1;
+                total_size += msize;
+                if (drop_u[j] >= drop_) {
+                    int64_t due = t + delays[j];
+                    int32_t rf = (proto_ == PUSH_PULL) ? -2 : -1;
+                    pending_[due].push_back({receiver, slot, rf, is_pull, sender});
+                } else {
+                    failed += 1;
+                    if (slot >= 0) freed.push_back(slot);
+                }
+            }
+        }
+        snap_tptr.push_back((int32_t)snap_nodes.size());
+
+        // --- deliveries due this tick (sub-phase B)
+        {
+            Stream go(stream_key(seed_, ONLINE, (uint64_t)t, 0));
+            for (int64_t i = 0; i < n_; ++i) online[i] = go.rnd();
+        }
+        std::vector<Msg> due;
+        {
+            auto it = pending_.find(t);
+            if (it != pending_.end()) {
+                due = std::move(it->second);
+                pending_.erase(it);
+            }
+        }
+        // receiver -> (slot, rslot, sender) in first-appearance order
+        std::vector<int32_t> order;
+        std::unordered_map<int32_t, std::vector<std::array<int32_t, 3>>> rmap;
+        for (const Msg& m : due) {
+            if (online[m.receiver] > online_) {
+                failed += 1;
+                if (m.slot >= 0) freed.push_back(m.slot);
+                continue;
+            }
+            if (m.is_pull) {
+                int32_t rslot = alloc_slot(m.receiver);
+                pull_nodes.push_back(m.receiver);
+                pull_slots.push_back(rslot);
+                if (!enqueue_reply(t, m.receiver, m.sender, rslot, sent,
+                                   failed, total_size))
+                    freed.push_back(rslot);
+                continue;
+            }
+            int32_t rslot = -1;
+            if (m.reply_flag == -2) {
+                rslot = alloc_slot(m.receiver);
+                if (!enqueue_reply(t, m.receiver, m.sender, rslot, sent,
+                                   failed, total_size))
+                    freed.push_back(rslot);
+            }
+            auto f = rmap.find(m.receiver);
+            if (f == rmap.end()) {
+                order.push_back(m.receiver);
+                rmap[m.receiver] = {{m.slot, rslot, m.sender}};
+            } else {
+                f->second.push_back({m.slot, rslot, m.sender});
+            }
+            freed.push_back(m.slot);
+        }
+        for (int32_t rn : order) {
+            recv_nodes.push_back(rn);
+            for (auto& p : rmap[rn]) {
+                del_slots.push_back(p[0]);
+                reply_slots.push_back(p[1]);
+                del_owners.push_back(p[2]);
+            }
+            recv_nptr.push_back((int32_t)del_slots.size());
+        }
+        recv_tptr.push_back((int32_t)recv_nodes.size());
+        pull_tptr.push_back((int32_t)pull_nodes.size());
+
+        // --- sub-phase C: same-tick replies
+        std::vector<Msg> rep_due;
+        {
+            auto it = pending_.find(t);
+            if (it != pending_.end()) {
+                rep_due = std::move(it->second);
+                pending_.erase(it);
+            }
+        }
+        std::vector<int32_t> rorder;
+        std::unordered_map<int32_t, std::vector<std::array<int32_t, 2>>> rrmap;
+        for (const Msg& m : rep_due) {
+            if (online[m.receiver] > online_) {
+                failed += 1;
+                freed.push_back(m.slot);
+                continue;
+            }
+            auto f = rrmap.find(m.receiver);
+            if (f == rrmap.end()) {
+                rorder.push_back(m.receiver);
+                rrmap[m.receiver] = {{m.slot, m.sender}};
+            } else {
+                f->second.push_back({m.slot, m.sender});
+            }
+            freed.push_back(m.slot);
+        }
+        for (int32_t rn : rorder) {
+            rep_nodes.push_back(rn);
+            for (auto& p : rrmap[rn]) {
+                rep_slots.push_back(p[0]);
+                rep_owners.push_back(p[1]);
+            }
+            rep_nptr.push_back((int32_t)rep_slots.size());
+        }
+        rep_tptr.push_back((int32_t)rep_nodes.size());
+
+        for (int32_t s : freed) free_slots_.push_back(s);
+    }
+
+    py::dict out;
+    auto arr = [](std::vector<int32_t>& v) {
+        auto a = py::array_t<int32_t>((py::ssize_t)v.size());
+        std::copy(v.begin(), v.end(), a.mutable_data());
+        return a;
+    };
+    out["snap_nodes"] = arr(snap_nodes);
+    out["snap_slots"] = arr(snap_slots);
+    out["snap_tptr"] = arr(snap_tptr);
+    out["recv_nodes"] = arr(recv_nodes);
+    out["recv_nptr"] = arr(recv_nptr);
+    out["recv_tptr"] = arr(recv_tptr);
+    out["del_slots"] = arr(del_slots);
+    out["del_owners"] = arr(del_owners);
+    out["reply_slots"] = arr(reply_slots);
+    out["pull_nodes"] = arr(pull_nodes);
+    out["pull_slots"] = arr(pull_slots);
+    out["pull_tptr"] = arr(pull_tptr);
+    out["rep_nodes"] = arr(rep_nodes);
+    out["rep_nptr"] = arr(rep_nptr);
+    out["rep_tptr"] = arr(rep_tptr);
+    out["rep_slots"] = arr(rep_slots);
+    out["rep_owners"] = arr(rep_owners);
+    out["sent"] = sent;
+    out["failed"] = failed;
+    out["total_size"] = total_size;
+    out["n_slots"] = next_slot_;
+    if (sampling_eval_ > 0) {
+        Stream g(stream_key(seed_, EVAL, (uint64_t)(t1 - 1), 0));
+        int64_t k = std::max<int64_t>((int64_t)(n_ * sampling_eval_), 1);
+        auto ev = py::array_t<int64_t>(k);
+        for (int64_t i = 0; i < k; ++i)
+            ev.mutable_data()[i] = g.integers(0, n_);
+        out["eval_nodes"] = ev;
+    } else {
+        out["eval_nodes"] = py::none();
+    }
+    return out;
+}
+
+PYBIND11_MODULE(_gossip_sched, m)
+{
+    py::class_<NativeScheduler>(m, "NativeScheduler")
+        .def(py::init<int64_t, int64_t, int, int64_t, double, double, int,
+                      int64_t, int64_t, double, int64_t, bool, double,
+                      uint64_t, py::object, py::object>())
+        .def("next_round", &NativeScheduler::next_round);
+}

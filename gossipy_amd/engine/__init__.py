@@ -14,7 +14,7 @@ from .backend import HIPBackend, TorchBackend, make_backend
 from .models import AdaLineSpec, LogRegSpec, MLPSpec, PegasosSpec
 from .rng import Purpose, RandomTape
 from .runner import BatchedGossipSimulator
-from .schedule import EngineConfig, RoundSchedule, Scheduler, TickPhase
+from .schedule import EngineConfig, RoundSchedule, Scheduler, TickPhase, make_scheduler
 
 __all__ = [
     "BatchedGossipSimulator",

@@ -37,7 +37,7 @@ from .arena import DataArena, NodeStateArena, SlotPool
 from .backend import make_backend
 from .metrics import binary_margin_metrics, classification_metrics_shared
 from .rng import Purpose, RandomTape
-from .schedule import EngineConfig, RoundSchedule, Scheduler, TickPhase
+from .schedule import EngineConfig, RoundSchedule, Scheduler, TickPhase, make_scheduler
 
 __all__ = ["BatchedGossipSimulator"]
 
@@ -90,7 +90,7 @@ class BatchedGossipSimulator(SimulationEventSender):
         self.state = NodeStateArena(self.n_local, spec.D, device, self.node_lo)
         self.pool = SlotPool(spec.D, device)
         self.data = data
-        self.scheduler = Scheduler(cfg)
+        self.scheduler = make_scheduler(cfg)
         self.initialized = False
         self.rounds_done = 0
         #: host mirror of per-round slot owners (set per round)
@@ -317,16 +317,111 @@ class BatchedGossipSimulator(SimulationEventSender):
             if results_global:
                 self.notify_evaluation(t, False, results_global)
 
+    def _fast_path_ok(self) -> bool:
+        """Single-GPU fast path: native scheduler + HIP round executor —
+        one python call per round instead of per-tick dispatch."""
+        from .schedule import NativeSchedulerAdapter
+
+        return (
+            self.world == 1
+            and isinstance(self.scheduler, NativeSchedulerAdapter)
+            and getattr(self.backend, "ext", None) is not None
+            and self.spec.family in ("logreg", "pegasos", "adaline")
+        )
+
+    def _run_round_fast(self, f: dict) -> None:
+        """Upload the round's flat event arrays in one H2D copy and hand
+        the whole round to the C++ executor (ops/hip round executors)."""
+        dev = self.device
+        dev_names = (
+            "snap_nodes",
+            "snap_slots",
+            "recv_nodes",
+            "recv_nptr",
+            "del_slots",
+            "reply_slots",
+            "pull_nodes",
+            "pull_slots",
+            "rep_nodes",
+            "rep_nptr",
+            "rep_slots",
+        )
+        parts = [np.ascontiguousarray(f[n], dtype=np.int32) for n in dev_names]
+        lens = [len(p) for p in parts]
+        host = np.concatenate(parts) if sum(lens) else np.zeros(1, np.int32)
+        dbuf = torch.from_numpy(host).to(dev, non_blocking=True)
+        views = {}
+        off = 0
+        for n, l in zip(dev_names, lens):
+            views[n] = dbuf[off : off + l]
+            off += l
+        tp = {
+            n: torch.from_numpy(np.ascontiguousarray(f[n], dtype=np.int32))
+            for n in ("snap_tptr", "recv_tptr", "pull_tptr", "rep_tptr")
+        }
+        ext = self.backend.ext
+        common = (
+            self.state.params,
+            self.state.ages,
+            self.pool.slots,
+            self.pool.slot_ages,
+            views["snap_nodes"],
+            views["snap_slots"],
+            tp["snap_tptr"],
+            views["recv_nodes"],
+            views["recv_nptr"],
+            tp["recv_tptr"],
+            views["del_slots"],
+            views["reply_slots"],
+            views["pull_nodes"],
+            views["pull_slots"],
+            tp["pull_tptr"],
+            views["rep_nodes"],
+            views["rep_nptr"],
+            tp["rep_tptr"],
+            views["rep_slots"],
+            self.data.x,
+            self.data.y,
+            self.data.counts,
+        )
+        from .backend import _MODE_ID
+
+        spec = self.spec
+        if spec.family == "logreg":
+            ext.run_round_logreg(
+                *common,
+                spec.d_in,
+                spec.n_classes,
+                spec.lr,
+                spec.weight_decay,
+                max(1, spec.local_epochs),
+                spec.batch_size,
+                _MODE_ID[spec.mode],
+            )
+        else:
+            ext.run_round_linear(
+                *common,
+                spec.d_in,
+                spec.lam if spec.family == "pegasos" else spec.lr,
+                1 if spec.family == "pegasos" else 0,
+                _MODE_ID[spec.mode],
+            )
+
     def start(self, n_rounds: int = 100) -> None:
         """Run ``n_rounds`` rounds."""
         assert self.initialized, "call init_nodes() first"
+        fast = self._fast_path_ok()
         for _ in range(n_rounds):
             r = self.rounds_done
-            sched = self.scheduler.next_round(r)
-            self.pool.ensure(sched.n_slots)
-            self._slot_owner = sched.slot_owner
-            for phase in sched.ticks:
-                self._run_tick(phase)
+            if fast:
+                sched = self.scheduler.next_round_flat(r)
+                self.pool.ensure(sched.n_slots)
+                self._run_round_fast(self.scheduler.last_flat)
+            else:
+                sched = self.scheduler.next_round(r)
+                self.pool.ensure(sched.n_slots)
+                for phase in sched.ticks:
+                    self._run_tick(phase)
             if self.rank == 0:
                 # report accounting comes from the schedule (host-side)
                 self.notify_message_counts(sched)

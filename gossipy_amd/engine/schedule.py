@@ -36,7 +36,14 @@ import numpy as np
 from ..core import AntiEntropyProtocol, ConstantDelay, Delay, LinearDelay, UniformDelay
 from .rng import Purpose, RandomTape
 
-__all__ = ["EngineConfig", "TickPhase", "RoundSchedule", "Scheduler"]
+__all__ = [
+    "EngineConfig",
+    "TickPhase",
+    "RoundSchedule",
+    "Scheduler",
+    "NativeSchedulerAdapter",
+    "make_scheduler",
+]
 
 
 @dataclass
@@ -386,3 +393,128 @@ class Scheduler:
         failed += 1
         self._reply_accounting[t] = (sent, failed, size)
         return False
+
+
+class NativeSchedulerAdapter:
+    """Adapter over the C++ scheduler (``csrc/scheduler.cpp``).
+
+    Produces the same :class:`RoundSchedule`/:class:`TickPhase` objects as
+    the Python :class:`Scheduler` (bit-exact — enforced by
+    tests/test_native_sched.py) from the native flat per-round arrays, and
+    exposes the flat arrays themselves (``last_flat``) for the GPU round
+    executor's one-upload-per-round fast path.
+    """
+
+    def __init__(self, cfg: EngineConfig):
+        from .. import ops
+
+        mod = ops.load_sched()
+        if mod is None:
+            raise ImportError("_gossip_sched.so not built")
+        self.cfg = cfg
+        d = cfg.delay
+        if isinstance(d, ConstantDelay):
+            kind, dmin, dmax, tx, ov = 0, d._delay, d._delay, 0.0, 0
+        elif isinstance(d, UniformDelay):
+            kind, dmin, dmax, tx, ov = 1, d._min_delay, d._max_delay, 0.0, 0
+        elif isinstance(d, LinearDelay):
+            kind, dmin, dmax, tx, ov = 2, 0, 0, d._timexunit, d._overhead
+        else:
+            raise TypeError("custom Delay subclasses need the python Scheduler")
+        ip = cfg.peers_indptr
+        ix = cfg.peers_indices
+        self._native = mod.NativeScheduler(
+            cfg.n_nodes,
+            cfg.delta,
+            int(cfg.protocol.value),
+            cfg.model_size,
+            cfg.drop_prob,
+            cfg.online_prob,
+            kind,
+            dmin,
+            dmax,
+            tx,
+            ov,
+            cfg.sync,
+            cfg.sampling_eval,
+            cfg.seed & 0xFFFFFFFFFFFFFFFF,
+            None if ip is None else np.ascontiguousarray(ip, dtype=np.int64),
+            None if ix is None else np.ascontiguousarray(ix, dtype=np.int64),
+        )
+        self.last_flat: Optional[dict] = None
+
+    def next_round(self, r: int) -> RoundSchedule:
+        f = self._native.next_round(r)
+        self.last_flat = f
+        ticks: List[TickPhase] = []
+        delta = self.cfg.delta
+        t0 = r * delta
+        for i in range(delta):
+            s0, s1 = int(f["snap_tptr"][i]), int(f["snap_tptr"][i + 1])
+            r0, r1 = int(f["recv_tptr"][i]), int(f["recv_tptr"][i + 1])
+            p0, p1 = int(f["pull_tptr"][i]), int(f["pull_tptr"][i + 1])
+            q0, q1 = int(f["rep_tptr"][i]), int(f["rep_tptr"][i + 1])
+            if s0 == s1 and r0 == r1 and p0 == p1 and q0 == q1:
+                continue
+            d0, d1 = int(f["recv_nptr"][r0]), int(f["recv_nptr"][r1])
+            e0, e1 = int(f["rep_nptr"][q0]), int(f["rep_nptr"][q1])
+            ticks.append(
+                TickPhase(
+                    t=t0 + i,
+                    snap_nodes=f["snap_nodes"][s0:s1],
+                    snap_slots=f["snap_slots"][s0:s1],
+                    recv_nodes=f["recv_nodes"][r0:r1],
+                    recv_ptr=f["recv_nptr"][r0 : r1 + 1] - d0,
+                    del_slots=f["del_slots"][d0:d1],
+                    del_owners=f["del_owners"][d0:d1],
+                    reply_slots=f["reply_slots"][d0:d1],
+                    pull_snap_nodes=f["pull_nodes"][p0:p1],
+                    pull_snap_slots=f["pull_slots"][p0:p1],
+                    rep_recv_nodes=f["rep_nodes"][q0:q1],
+                    rep_recv_ptr=f["rep_nptr"][q0 : q1 + 1] - e0,
+                    rep_del_slots=f["rep_slots"][e0:e1],
+                    rep_del_owners=f["rep_owners"][e0:e1],
+                )
+            )
+        return RoundSchedule(
+            round_idx=r,
+            ticks=ticks,
+            n_slots=int(f["n_slots"]),
+            slot_owner=None,
+            sent_messages=int(f["sent"]),
+            failed_messages=int(f["failed"]),
+            total_size=int(f["total_size"]),
+            eval_nodes=f["eval_nodes"],
+        )
+
+
+def make_scheduler(cfg: EngineConfig):
+    """Native scheduler when available & applicable, else the Python one."""
+    try:
+        return NativeSchedulerAdapter(cfg)
+    except (ImportError, TypeError):
+        return Scheduler(cfg)
+
+
+def _flat_round_summary(f: dict, r: int) -> RoundSchedule:
+    return RoundSchedule(
+        round_idx=r,
+        ticks=[],
+        n_slots=int(f["n_slots"]),
+        slot_owner=None,
+        sent_messages=int(f["sent"]),
+        failed_messages=int(f["failed"]),
+        total_size=int(f["total_size"]),
+        eval_nodes=f["eval_nodes"],
+    )
+
+
+def _adapter_next_round_flat(self, r: int) -> RoundSchedule:
+    """Fast-path variant of next_round: skips the TickPhase conversion and
+    leaves the flat arrays in ``last_flat`` for the GPU round executor."""
+    f = self._native.next_round(r)
+    self.last_flat = f
+    return _flat_round_summary(f, r)
+
+
+NativeSchedulerAdapter.next_round_flat = _adapter_next_round_flat

@@ -14,10 +14,32 @@ from __future__ import annotations
 import os
 
 _EXT = None
+_SCHED = None
 
 
 def extension_path() -> str:
     return os.path.join(os.path.dirname(__file__), "_gossip_hip.so")
+
+
+def sched_path() -> str:
+    return os.path.join(os.path.dirname(__file__), "_gossip_sched.so")
+
+
+def load_sched():
+    """Load (once) the native CPU scheduler, or None if not built."""
+    global _SCHED
+    if _SCHED is not None:
+        return _SCHED
+    import importlib.util
+
+    path = sched_path()
+    if not os.path.exists(path):
+        return None
+    spec = importlib.util.spec_from_file_location("gossipy_amd.ops._gossip_sched", path)
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    _SCHED = mod
+    return _SCHED
 
 
 def load_extension():

@@ -13,6 +13,29 @@ import shutil
 import sys
 
 
+def build_sched(verbose: bool = False) -> str:
+    """Build the native CPU scheduler (plain C++, no HIP)."""
+    from torch.utils import cpp_extension
+
+    here = os.path.dirname(os.path.abspath(__file__))
+    src = os.path.join(here, "..", "csrc", "scheduler.cpp")
+    build_dir = os.path.join(here, "hip", "build_sched")
+    os.makedirs(build_dir, exist_ok=True)
+    cpp_extension.load(
+        name="_gossip_sched",
+        sources=[src],
+        build_directory=build_dir,
+        extra_cflags=["-O3"],
+        verbose=verbose,
+        is_python_module=False,
+        with_cuda=False,
+    )
+    built = os.path.join(build_dir, "_gossip_sched.so")
+    target = os.path.join(here, "_gossip_sched.so")
+    shutil.copy2(built, target)
+    return target
+
+
 def build(verbose: bool = False) -> str:
     os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
     os.environ.setdefault("MAX_JOBS", "8")
@@ -34,6 +57,7 @@ def build(verbose: bool = False) -> str:
     built = os.path.join(build_dir, "_gossip_hip.so")
     target = os.path.join(here, "_gossip_hip.so")
     shutil.copy2(built, target)
+    build_sched(verbose)
     return target
 
 

@@ -663,10 +663,190 @@ void tick_mlp(torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
                        current_stream(), a);
 }
 
+// ---------------------------------------------------------------------------
+// round executors: the C++ loop over one round's ticks. All event arrays
+// arrive on-device ONCE per round; tick boundary pointers stay host-side.
+// Per tick this issues up to 4 stream-ordered launches (snapshot, deliver,
+// pull-snapshot, reply-deliver) with zero per-tick Python involvement —
+// the host cost of a 100-tick round drops from ~10 ms of Python to
+// ~200 launch enqueues of a few µs each.
+// ---------------------------------------------------------------------------
+
+struct RoundArrays {
+    // device pointers
+    const int *snap_nodes, *snap_slots;
+    const int *recv_nodes, *recv_nptr;
+    const int *del_slots, *reply_slots;
+    const int *pull_nodes, *pull_slots;
+    const int *rep_nodes, *rep_nptr, *rep_slots;
+    // host tick pointers [delta+1]
+    const int *snap_tptr, *recv_tptr, *pull_tptr, *rep_tptr;
+    int delta;
+};
+
+static RoundArrays unpack_round(
+    torch::Tensor snap_nodes, torch::Tensor snap_slots, torch::Tensor snap_tptr,
+    torch::Tensor recv_nodes, torch::Tensor recv_nptr, torch::Tensor recv_tptr,
+    torch::Tensor del_slots, torch::Tensor reply_slots,
+    torch::Tensor pull_nodes, torch::Tensor pull_slots, torch::Tensor pull_tptr,
+    torch::Tensor rep_nodes, torch::Tensor rep_nptr, torch::Tensor rep_tptr,
+    torch::Tensor rep_slots)
+{
+    RoundArrays r;
+    r.snap_nodes = snap_nodes.numel() ? snap_nodes.data_ptr<int>() : nullptr;
+    r.snap_slots = snap_slots.numel() ? snap_slots.data_ptr<int>() : nullptr;
+    r.recv_nodes = recv_nodes.numel() ? recv_nodes.data_ptr<int>() : nullptr;
+    r.recv_nptr = recv_nptr.data_ptr<int>();
+    r.del_slots = del_slots.numel() ? del_slots.data_ptr<int>() : nullptr;
+    r.reply_slots = reply_slots.numel() ? reply_slots.data_ptr<int>() : nullptr;
+    r.pull_nodes = pull_nodes.numel() ? pull_nodes.data_ptr<int>() : nullptr;
+    r.pull_slots = pull_slots.numel() ? pull_slots.data_ptr<int>() : nullptr;
+    r.rep_nodes = rep_nodes.numel() ? rep_nodes.data_ptr<int>() : nullptr;
+    r.rep_nptr = rep_nptr.data_ptr<int>();
+    r.rep_slots = rep_slots.numel() ? rep_slots.data_ptr<int>() : nullptr;
+    r.snap_tptr = snap_tptr.data_ptr<int>();   // host (CPU) tensors
+    r.recv_tptr = recv_tptr.data_ptr<int>();
+    r.pull_tptr = pull_tptr.data_ptr<int>();
+    r.rep_tptr = rep_tptr.data_ptr<int>();
+    r.delta = (int)snap_tptr.numel() - 1;
+    return r;
+}
+
+static void launch_snap(const float* params, const int* ages, float* slots,
+                        int* slot_ages, const int* nodes, const int* slot_ids,
+                        int n, int D, hipStream_t s)
+{
+    long total = (long)n * D;
+    int block = 256;
+    int grid = (int)std::min<long>((total + block - 1) / block, 2048);
+    hipLaunchKernelGGL(snapshot_kernel, dim3(grid), dim3(block), 0, s,
+                       params, ages, slots, slot_ages, nodes, slot_ids, n, D);
+}
+
+void run_round_logreg(
+    torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
+    torch::Tensor slot_ages,
+    torch::Tensor snap_nodes, torch::Tensor snap_slots, torch::Tensor snap_tptr,
+    torch::Tensor recv_nodes, torch::Tensor recv_nptr, torch::Tensor recv_tptr,
+    torch::Tensor del_slots, torch::Tensor reply_slots,
+    torch::Tensor pull_nodes, torch::Tensor pull_slots, torch::Tensor pull_tptr,
+    torch::Tensor rep_nodes, torch::Tensor rep_nptr, torch::Tensor rep_tptr,
+    torch::Tensor rep_slots,
+    torch::Tensor X, torch::Tensor Y, torch::Tensor counts,
+    int64_t d, int64_t k, double lr, double wd, int64_t epochs, int64_t bs,
+    int64_t mode)
+{
+    CHECK_DEV(params); CHECK_DEV(slots); CHECK_DEV(X);
+    RoundArrays r = unpack_round(snap_nodes, snap_slots, snap_tptr, recv_nodes,
+                                 recv_nptr, recv_tptr, del_slots, reply_slots,
+                                 pull_nodes, pull_slots, pull_tptr, rep_nodes,
+                                 rep_nptr, rep_tptr, rep_slots);
+    hipStream_t s = current_stream();
+    LogregArgs a;
+    a.params = params.data_ptr<float>(); a.ages = ages.data_ptr<int>();
+    a.slots = slots.data_ptr<float>(); a.slot_ages = slot_ages.data_ptr<int>();
+    a.X = X.data_ptr<float>(); a.Y = Y.data_ptr<float>();
+    a.counts = counts.data_ptr<int>();
+    a.d = d; a.k = k; a.Smax = X.size(1); a.D = params.size(1);
+    a.lr = lr; a.wd = wd; a.epochs = epochs; a.bs = bs; a.mode = mode;
+    a.update_only = 0;
+    int bsmax = (bs == 0) ? a.Smax : std::min<int>(bs, a.Smax);
+    size_t smem = sizeof(float) * (2 * a.D + (size_t)bsmax * a.d + (size_t)bsmax * a.k);
+    TORCH_CHECK(smem <= 160 * 1024, "logreg LDS budget exceeded");
+    for (int t = 0; t < r.delta; ++t) {
+        int s0 = r.snap_tptr[t], s1 = r.snap_tptr[t + 1];
+        if (s1 > s0)
+            launch_snap(a.params, a.ages, a.slots, a.slot_ages,
+                        r.snap_nodes + s0, r.snap_slots + s0, s1 - s0, a.D, s);
+        int r0 = r.recv_tptr[t], r1 = r.recv_tptr[t + 1];
+        if (r1 > r0) {
+            a.nodes = r.recv_nodes + r0;
+            a.ptr = r.recv_nptr + r0;
+            a.dslots = r.del_slots;
+            a.rslots = r.reply_slots;
+            hipLaunchKernelGGL(tick_logreg_kernel, dim3(r1 - r0), dim3(128),
+                               smem, s, a);
+        }
+        int p0 = r.pull_tptr[t], p1 = r.pull_tptr[t + 1];
+        if (p1 > p0)
+            launch_snap(a.params, a.ages, a.slots, a.slot_ages,
+                        r.pull_nodes + p0, r.pull_slots + p0, p1 - p0, a.D, s);
+        int q0 = r.rep_tptr[t], q1 = r.rep_tptr[t + 1];
+        if (q1 > q0) {
+            a.nodes = r.rep_nodes + q0;
+            a.ptr = r.rep_nptr + q0;
+            a.dslots = r.rep_slots;
+            a.rslots = nullptr;
+            hipLaunchKernelGGL(tick_logreg_kernel, dim3(q1 - q0), dim3(128),
+                               smem, s, a);
+        }
+    }
+}
+
+void run_round_linear(
+    torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
+    torch::Tensor slot_ages,
+    torch::Tensor snap_nodes, torch::Tensor snap_slots, torch::Tensor snap_tptr,
+    torch::Tensor recv_nodes, torch::Tensor recv_nptr, torch::Tensor recv_tptr,
+    torch::Tensor del_slots, torch::Tensor reply_slots,
+    torch::Tensor pull_nodes, torch::Tensor pull_slots, torch::Tensor pull_tptr,
+    torch::Tensor rep_nodes, torch::Tensor rep_nptr, torch::Tensor rep_tptr,
+    torch::Tensor rep_slots,
+    torch::Tensor X, torch::Tensor Y, torch::Tensor counts,
+    int64_t d, double lrlam, int64_t is_pegasos, int64_t mode)
+{
+    CHECK_DEV(params); CHECK_DEV(slots); CHECK_DEV(X);
+    RoundArrays r = unpack_round(snap_nodes, snap_slots, snap_tptr, recv_nodes,
+                                 recv_nptr, recv_tptr, del_slots, reply_slots,
+                                 pull_nodes, pull_slots, pull_tptr, rep_nodes,
+                                 rep_nptr, rep_tptr, rep_slots);
+    hipStream_t s = current_stream();
+    LinearArgs a;
+    a.params = params.data_ptr<float>(); a.ages = ages.data_ptr<int>();
+    a.slots = slots.data_ptr<float>(); a.slot_ages = slot_ages.data_ptr<int>();
+    a.X = X.data_ptr<float>(); a.Y = Y.data_ptr<float>();
+    a.counts = counts.data_ptr<int>();
+    a.d = d; a.Smax = X.size(1);
+    a.lrlam = lrlam; a.is_pegasos = is_pegasos; a.mode = mode;
+    a.update_only = 0;
+    for (int t = 0; t < r.delta; ++t) {
+        int s0 = r.snap_tptr[t], s1 = r.snap_tptr[t + 1];
+        if (s1 > s0)
+            launch_snap(a.params, a.ages, a.slots, a.slot_ages,
+                        r.snap_nodes + s0, r.snap_slots + s0, s1 - s0, a.d, s);
+        int r0 = r.recv_tptr[t], r1 = r.recv_tptr[t + 1];
+        if (r1 > r0) {
+            a.nodes = r.recv_nodes + r0;
+            a.ptr = r.recv_nptr + r0;
+            a.dslots = r.del_slots;
+            a.rslots = r.reply_slots;
+            hipLaunchKernelGGL(tick_linear_kernel, dim3(r1 - r0), dim3(WAVE),
+                               0, s, a);
+        }
+        int p0 = r.pull_tptr[t], p1 = r.pull_tptr[t + 1];
+        if (p1 > p0)
+            launch_snap(a.params, a.ages, a.slots, a.slot_ages,
+                        r.pull_nodes + p0, r.pull_slots + p0, p1 - p0, a.d, s);
+        int q0 = r.rep_tptr[t], q1 = r.rep_tptr[t + 1];
+        if (q1 > q0) {
+            a.nodes = r.rep_nodes + q0;
+            a.ptr = r.rep_nptr + q0;
+            a.dslots = r.rep_slots;
+            a.rslots = nullptr;
+            hipLaunchKernelGGL(tick_linear_kernel, dim3(q1 - q0), dim3(WAVE),
+                               0, s, a);
+        }
+    }
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m)
 {
     m.def("snapshot", &snapshot, "batched model snapshot (arena row copy)");
     m.def("tick_logreg", &tick_logreg, "fused merge + logreg SGD tick");
     m.def("tick_linear", &tick_linear, "fused merge + pegasos/adaline tick");
     m.def("tick_mlp", &tick_mlp, "fused merge + MLP SGD tick");
+    m.def("run_round_logreg", &run_round_logreg,
+          "whole-round executor, logreg family");
+    m.def("run_round_linear", &run_round_linear,
+          "whole-round executor, pegasos/adaline family");
 }

@@ -217,3 +217,57 @@ class TestEngineGPU:
     def test_gpu_learns(self):
         sim, rep = self._run(CUDA, rounds=15)
         assert rep.get_evaluation(False)[-1][1]["accuracy"] > 0.9
+
+
+class TestFastPath:
+    """The whole-round C++ executor must match the per-tick python path."""
+
+    def _build(self, protocol=AntiEntropyProtocol.PUSH_PULL):
+        X, y = make_synthetic_classification((640, 57, 2), seed=0, margin=2.0)
+        idx = np.random.default_rng(0).permutation(640)
+        shards = [(X[s], y[s]) for s in np.array_split(idx[:576], 64)]
+        data = DataArena.from_shards(
+            shards, CUDA, global_eval=(X[idx[576:]], y[idx[576:]])
+        )
+        cfg = EngineConfig(
+            n_nodes=64, delta=10, protocol=protocol,
+            model_size=116, sampling_eval=0.0, seed=3,
+        )
+        return cfg, data
+
+    def test_fast_matches_tick_path(self):
+        cfg, data = self._build()
+        spec = LogRegSpec(d_in=57, n_classes=2, lr=0.1)
+
+        fast = BatchedGossipSimulator(cfg, spec, data, device=CUDA)
+        assert fast._fast_path_ok()
+        fast.init_nodes()
+        fast.start(n_rounds=4)
+
+        slow = BatchedGossipSimulator(cfg, spec, data, device=CUDA)
+        slow._fast_path_ok = lambda: False
+        slow.init_nodes()
+        slow.start(n_rounds=4)
+        torch.cuda.synchronize()
+        assert torch.allclose(
+            fast.local_params(), slow.local_params(), atol=1e-5, rtol=1e-5
+        )
+        assert torch.equal(fast.state.ages, slow.state.ages)
+
+    def test_fast_pegasos(self):
+        cfg, data = self._build(AntiEntropyProtocol.PUSH)
+        data.y = torch.where(data.y > 0, 1.0, -1.0)
+        data.gy = torch.where(data.gy > 0, 1.0, -1.0)
+        spec = PegasosSpec(d_in=57, lam=0.01)
+        fast = BatchedGossipSimulator(cfg, spec, data, device=CUDA)
+        assert fast._fast_path_ok()
+        fast.init_nodes()
+        fast.start(n_rounds=4)
+        slow = BatchedGossipSimulator(cfg, spec, data, device=CUDA)
+        slow._fast_path_ok = lambda: False
+        slow.init_nodes()
+        slow.start(n_rounds=4)
+        torch.cuda.synchronize()
+        assert torch.allclose(
+            fast.local_params(), slow.local_params(), atol=1e-5, rtol=1e-5
+        )

@@ -1,0 +1,98 @@
+"""Native (C++) scheduler must reproduce the Python scheduler bit-exactly:
+same ticks, same event arrays, same accounting, for every protocol, delay
+model and fault setting."""
+
+import numpy as np
+import pytest
+
+from gossipy_amd import ops
+from gossipy_amd.core import (
+    AntiEntropyProtocol,
+    ConstantDelay,
+    LinearDelay,
+    UniformDelay,
+)
+from gossipy_amd.engine import EngineConfig, Scheduler
+from gossipy_amd.engine.schedule import NativeSchedulerAdapter
+
+pytestmark = pytest.mark.skipif(
+    ops.load_sched() is None, reason="_gossip_sched.so not built"
+)
+
+
+def _cfgs():
+    base = dict(n_nodes=60, delta=12, model_size=116, seed=9)
+    yield EngineConfig(protocol=AntiEntropyProtocol.PUSH, **base)
+    yield EngineConfig(
+        protocol=AntiEntropyProtocol.PUSH, drop_prob=0.3, online_prob=0.7, **base
+    )
+    yield EngineConfig(
+        protocol=AntiEntropyProtocol.PUSH, delay=UniformDelay(0, 20), **base
+    )
+    yield EngineConfig(
+        protocol=AntiEntropyProtocol.PUSH_PULL, delay=UniformDelay(1, 5), **base
+    )
+    yield EngineConfig(
+        protocol=AntiEntropyProtocol.PUSH_PULL, drop_prob=0.2, **base
+    )
+    yield EngineConfig(protocol=AntiEntropyProtocol.PULL, **base)
+    yield EngineConfig(
+        protocol=AntiEntropyProtocol.PULL,
+        delay=LinearDelay(0.01, 2),
+        online_prob=0.8,
+        **base,
+    )
+    yield EngineConfig(
+        protocol=AntiEntropyProtocol.PUSH, sync=False, sampling_eval=0.2, **base
+    )
+    # explicit topology (ring)
+    n = 60
+    indptr = np.arange(0, 2 * n + 1, 2, dtype=np.int64)
+    indices = np.empty(2 * n, dtype=np.int64)
+    for i in range(n):
+        indices[2 * i] = (i - 1) % n
+        indices[2 * i + 1] = (i + 1) % n
+    yield EngineConfig(
+        protocol=AntiEntropyProtocol.PUSH,
+        peers_indptr=indptr,
+        peers_indices=indices,
+        **base,
+    )
+
+
+@pytest.mark.parametrize("cfg_i", range(9))
+def test_native_matches_python(cfg_i):
+    cfg = list(_cfgs())[cfg_i]
+    py_s = Scheduler(cfg)
+    nat_s = NativeSchedulerAdapter(cfg)
+    for r in range(4):
+        a = py_s.next_round(r)
+        b = nat_s.next_round(r)
+        assert a.sent_messages == b.sent_messages, f"round {r} sent"
+        assert a.failed_messages == b.failed_messages, f"round {r} failed"
+        assert a.total_size == b.total_size, f"round {r} size"
+        assert a.n_slots == b.n_slots, f"round {r} slots"
+        assert len(a.ticks) == len(b.ticks), f"round {r} #ticks"
+        for pa, pb in zip(a.ticks, b.ticks):
+            assert pa.t == pb.t
+            for f in (
+                "snap_nodes",
+                "snap_slots",
+                "recv_nodes",
+                "recv_ptr",
+                "del_slots",
+                "del_owners",
+                "reply_slots",
+                "pull_snap_nodes",
+                "pull_snap_slots",
+                "rep_recv_nodes",
+                "rep_recv_ptr",
+                "rep_del_slots",
+                "rep_del_owners",
+            ):
+                va, vb = getattr(pa, f), getattr(pb, f)
+                assert np.array_equal(
+                    np.asarray(va, dtype=np.int64), np.asarray(vb, dtype=np.int64)
+                ), f"round {r} tick {pa.t} field {f}: {va} vs {vb}"
+        if a.eval_nodes is not None:
+            assert np.array_equal(a.eval_nodes, b.eval_nodes)
