@@ -203,7 +203,9 @@ tick_logreg_kernel(LogregArgs a)
                 age = sage;
                 __syncthreads();
             }
-            int rs = a.rslots[j];
+            // reply-delivery launches pass rslots == nullptr (replies to
+            // replies are discarded, gossipy/simul.py:426)
+            int rs = a.rslots ? a.rslots[j] : -1;
             if (rs >= 0) {  // PUSH_PULL reply snapshot (post merge+update)
                 for (int e = tid; e < a.D; e += blockDim.x)
                     a.slots[(long)rs * a.D + e] = W[e];
@@ -331,7 +333,7 @@ tick_linear_kernel(LinearArgs a)
                 }
                 age = sage;
             }
-            int rs = a.rslots[j];
+            int rs = a.rslots ? a.rslots[j] : -1;
             if (rs >= 0) {
                 for (int r = 0; r < nreg; ++r) {
                     int e = lane + r * WAVE;
@@ -521,7 +523,7 @@ tick_mlp_kernel(MlpArgs a)
                 age = sage;
                 __syncthreads();
             }
-            int rs = a.rslots[j];
+            int rs = a.rslots ? a.rslots[j] : -1;
             if (rs >= 0) {
                 for (int e = tid; e < a.D; e += blockDim.x)
                     a.slots[(long)rs * a.D + e] = W[e];
