@@ -65,7 +65,8 @@ struct Stream {
 
 enum Purpose {
     TIMEOUT = 0, PEER = 1, DROP = 2, ONLINE = 3, DELAY = 4,
-    EVAL = 5, INIT = 6, DATA = 7, TOKEN = 8, MISC = 9
+    EVAL = 5, INIT = 6, DATA = 7, TOKEN = 8, MISC = 9, PART = 10,
+    SAMPLE = 11
 };
 
 static inline uint64_t stream_key(uint64_t seed, uint64_t purpose, uint64_t t,
@@ -91,6 +92,7 @@ struct Msg {  // matches the python pending tuple
     int32_t reply_flag; // -2 = wants reply, -1 = plain
     bool is_pull;
     int32_t sender;
+    int32_t pid;        // partition id (-1 when n_parts == 0)
 };
 
 class NativeScheduler {
@@ -100,11 +102,13 @@ public:
                     int delay_kind, int64_t dmin, int64_t dmax,
                     double timexunit, int64_t overhead, bool sync,
                     double sampling_eval, uint64_t seed,
-                    py::object peers_indptr, py::object peers_indices)
+                    py::object peers_indptr, py::object peers_indices,
+                    int64_t n_parts = 0)
         : n_(n_nodes), delta_(delta), proto_(proto), model_size_(model_size),
           drop_(drop_prob), online_(online_prob), dkind_(delay_kind),
           dmin_(dmin), dmax_(dmax), timexunit_(timexunit), overhead_(overhead),
-          sync_(sync), sampling_eval_(sampling_eval), seed_(seed)
+          sync_(sync), sampling_eval_(sampling_eval), seed_(seed),
+          n_parts_(n_parts)
     {
         Stream g(stream_key(seed_, TIMEOUT, 0, 0));
         deltas_.resize(n_);
@@ -139,6 +143,7 @@ private:
     bool sync_;
     double sampling_eval_;
     uint64_t seed_;
+    int64_t n_parts_ = 0;
     std::vector<int64_t> deltas_;
     std::vector<int64_t> indptr_, indices_;
     std::unordered_map<int64_t, std::vector<Msg>> pending_;
@@ -170,10 +175,19 @@ private:
         }
     }
 
+    // fresh partition id for a reply (keyed on (t, replier) like the
+    // python scheduler's _reply_pid)
+    int32_t reply_pid(int64_t t, int32_t replier)
+    {
+        if (n_parts_ <= 0) return -1;
+        Stream g(stream_key(seed_, PART, (uint64_t)t, (uint64_t)(1 + replier)));
+        return (int32_t)g.integers(0, n_parts_);
+    }
+
     // returns true if the reply was enqueued (false = dropped)
     bool enqueue_reply(int64_t t, int32_t replier, int32_t requester,
-                       int32_t slot, int64_t& sent, int64_t& failed,
-                       int64_t& total_size)
+                       int32_t slot, int32_t pid, int64_t& sent,
+                       int64_t& failed, int64_t& total_size)
     {
         Stream g(stream_key(seed_, DROP, (uint64_t)t, (uint64_t)(1 + replier)));
         double u = g.rnd();
@@ -183,7 +197,7 @@ private:
             Stream gd(stream_key(seed_, DELAY, (uint64_t)t,
                                  (uint64_t)(1 + replier)));
             int64_t dly = delay_for(gd, model_size_);
-            pending_[t + dly].push_back({requester, slot, -1, false, replier});
+            pending_[t + dly].push_back({requester, slot, -1, false, replier, pid});
             return true;
         }
         failed += 1;
@@ -199,10 +213,10 @@ py::dict NativeScheduler::next_round(int64_t r)
     std::vector<int32_t> snap_nodes, snap_slots, snap_tptr{0};
     std::vector<int32_t> recv_nodes, recv_tptr{0};
     std::vector<int32_t> recv_nptr{0};
-    std::vector<int32_t> del_slots, del_owners, reply_slots;
+    std::vector<int32_t> del_slots, del_owners, reply_slots, del_pids;
     std::vector<int32_t> pull_nodes, pull_slots, pull_tptr{0};
     std::vector<int32_t> rep_nodes, rep_tptr{0}, rep_nptr{0};
-    std::vector<int32_t> rep_slots, rep_owners;
+    std::vector<int32_t> rep_slots, rep_owners, rep_pids;
 
     std::vector<double> online(n_);
     std::vector<int32_t> firing;
@@ -252,6 +266,12 @@ py::dict NativeScheduler::next_round(int64_t r)
                 Stream gdl(stream_key(seed_, DELAY, (uint64_t)t, 0));
                 for (size_t j = 0; j < n_f; ++j) delays[j] = delay_for(gdl, msize);
             }
+            std::vector<int32_t> pids(n_f, -1);
+            if (n_parts_ > 0) {
+                Stream gpt(stream_key(seed_, PART, (uint64_t)t, 0));
+                for (size_t j = 0; j < n_f; ++j)
+                    pids[j] = (int32_t)gpt.integers(0, n_parts_);
+            }
             for (size_t j = 0; j < n_f; ++j) {
                 int32_t sender = firing[j], receiver = peers[j];
                 bool is_pull = proto_ == PULL;
@@ -266,7 +286,7 @@ py::dict NativeScheduler::next_round(int64_t r)
                 if (drop_u[j] >= drop_) {
                     int64_t due = t + delays[j];
                     int32_t rf = (proto_ == PUSH_PULL) ? -2 : -1;
-                    pending_[due].push_back({receiver, slot, rf, is_pull, sender});
+                    pending_[due].push_back({receiver, slot, rf, is_pull, sender, pids[j]});
                 } else {
                     failed += 1;
                     if (slot >= 0) freed.push_back(slot);
@@ -290,7 +310,7 @@ py::dict NativeScheduler::next_round(int64_t r)
         }
         // receiver -> (slot, rslot, sender) in first-appearance order
         std::vector<int32_t> order;
-        std::unordered_map<int32_t, std::vector<std::array<int32_t, 3>>> rmap;
+        std::unordered_map<int32_t, std::vector<std::array<int32_t, 4>>> rmap;
         for (const Msg& m : due) {
             if (online[m.receiver] > online_) {
                 failed += 1;
@@ -301,24 +321,26 @@ py::dict NativeScheduler::next_round(int64_t r)
                 int32_t rslot = alloc_slot(m.receiver);
                 pull_nodes.push_back(m.receiver);
                 pull_slots.push_back(rslot);
-                if (!enqueue_reply(t, m.receiver, m.sender, rslot, sent,
-                                   failed, total_size))
+                if (!enqueue_reply(t, m.receiver, m.sender, rslot,
+                                   reply_pid(t, m.receiver), sent, failed,
+                                   total_size))
                     freed.push_back(rslot);
                 continue;
             }
             int32_t rslot = -1;
             if (m.reply_flag == -2) {
                 rslot = alloc_slot(m.receiver);
-                if (!enqueue_reply(t, m.receiver, m.sender, rslot, sent,
-                                   failed, total_size))
+                if (!enqueue_reply(t, m.receiver, m.sender, rslot,
+                                   reply_pid(t, m.receiver), sent, failed,
+                                   total_size))
                     freed.push_back(rslot);
             }
             auto f = rmap.find(m.receiver);
             if (f == rmap.end()) {
                 order.push_back(m.receiver);
-                rmap[m.receiver] = {{m.slot, rslot, m.sender}};
+                rmap[m.receiver] = {{m.slot, rslot, m.sender, m.pid}};
             } else {
-                f->second.push_back({m.slot, rslot, m.sender});
+                f->second.push_back({m.slot, rslot, m.sender, m.pid});
             }
             freed.push_back(m.slot);
         }
@@ -328,6 +350,7 @@ py::dict NativeScheduler::next_round(int64_t r)
                 del_slots.push_back(p[0]);
                 reply_slots.push_back(p[1]);
                 del_owners.push_back(p[2]);
+                del_pids.push_back(p[3]);
             }
             recv_nptr.push_back((int32_t)del_slots.size());
         }
@@ -344,7 +367,7 @@ py::dict NativeScheduler::next_round(int64_t r)
             }
         }
         std::vector<int32_t> rorder;
-        std::unordered_map<int32_t, std::vector<std::array<int32_t, 2>>> rrmap;
+        std::unordered_map<int32_t, std::vector<std::array<int32_t, 3>>> rrmap;
         for (const Msg& m : rep_due) {
             if (online[m.receiver] > online_) {
                 failed += 1;
@@ -354,9 +377,9 @@ py::dict NativeScheduler::next_round(int64_t r)
             auto f = rrmap.find(m.receiver);
             if (f == rrmap.end()) {
                 rorder.push_back(m.receiver);
-                rrmap[m.receiver] = {{m.slot, m.sender}};
+                rrmap[m.receiver] = {{m.slot, m.sender, m.pid}};
             } else {
-                f->second.push_back({m.slot, m.sender});
+                f->second.push_back({m.slot, m.sender, m.pid});
             }
             freed.push_back(m.slot);
         }
@@ -365,6 +388,7 @@ py::dict NativeScheduler::next_round(int64_t r)
             for (auto& p : rrmap[rn]) {
                 rep_slots.push_back(p[0]);
                 rep_owners.push_back(p[1]);
+                rep_pids.push_back(p[2]);
             }
             rep_nptr.push_back((int32_t)rep_slots.size());
         }
@@ -396,6 +420,8 @@ py::dict NativeScheduler::next_round(int64_t r)
     out["rep_tptr"] = arr(rep_tptr);
     out["rep_slots"] = arr(rep_slots);
     out["rep_owners"] = arr(rep_owners);
+    out["del_pids"] = arr(del_pids);
+    out["rep_pids"] = arr(rep_pids);
     out["sent"] = sent;
     out["failed"] = failed;
     out["total_size"] = total_size;
@@ -418,6 +444,13 @@ PYBIND11_MODULE(_gossip_sched, m)
     py::class_<NativeScheduler>(m, "NativeScheduler")
         .def(py::init<int64_t, int64_t, int, int64_t, double, double, int,
                       int64_t, int64_t, double, int64_t, bool, double,
-                      uint64_t, py::object, py::object>())
+                      uint64_t, py::object, py::object, int64_t>(),
+             py::arg("n_nodes"), py::arg("delta"), py::arg("proto"),
+             py::arg("model_size"), py::arg("drop_prob"), py::arg("online_prob"),
+             py::arg("delay_kind"), py::arg("dmin"), py::arg("dmax"),
+             py::arg("timexunit"), py::arg("overhead"), py::arg("sync"),
+             py::arg("sampling_eval"), py::arg("seed"),
+             py::arg("peers_indptr"), py::arg("peers_indices"),
+             py::arg("n_parts") = 0)
         .def("next_round", &NativeScheduler::next_round);
 }

@@ -38,9 +38,22 @@ class NodeStateArena:
     ``g - node_lo``.
     """
 
-    def __init__(self, n_local: int, D: int, device: torch.device, node_lo: int = 0):
+    def __init__(
+        self,
+        n_local: int,
+        D: int,
+        device: torch.device,
+        node_lo: int = 0,
+        age_width: int = 1,
+    ):
         self.params = torch.zeros(n_local, D, device=device, dtype=torch.float32)
-        self.ages = torch.zeros(n_local, device=device, dtype=torch.int32)
+        # ``age_width > 1``: per-partition age vectors (PartitionedTMH's
+        # ``n_updates`` array, gossipy/model/handler.py:475); else scalar age
+        if age_width > 1:
+            self.ages = torch.zeros(n_local, age_width, device=device, dtype=torch.int32)
+        else:
+            self.ages = torch.zeros(n_local, device=device, dtype=torch.int32)
+        self.age_width = age_width
         self.node_lo = node_lo
         self.n_local = n_local
         self.D = D
@@ -52,11 +65,21 @@ class NodeStateArena:
 class SlotPool:
     """Per-round snapshot slot pool (grown geometrically, never shrunk)."""
 
-    def __init__(self, D: int, device: torch.device, capacity: int = 1024):
+    def __init__(
+        self, D: int, device: torch.device, capacity: int = 1024, age_width: int = 1
+    ):
         self.D = D
         self.device = device
+        self.age_width = age_width
         self.slots = torch.zeros(capacity, D, device=device, dtype=torch.float32)
-        self.slot_ages = torch.zeros(capacity, device=device, dtype=torch.int32)
+        self.slot_ages = self._new_ages(capacity)
+
+    def _new_ages(self, cap: int) -> torch.Tensor:
+        if self.age_width > 1:
+            return torch.zeros(
+                cap, self.age_width, device=self.device, dtype=torch.int32
+            )
+        return torch.zeros(cap, device=self.device, dtype=torch.int32)
 
     def ensure(self, n_slots: int) -> None:
         if n_slots > self.slots.shape[0]:
@@ -64,7 +87,7 @@ class SlotPool:
             self.slots = torch.zeros(
                 cap, self.D, device=self.device, dtype=torch.float32
             )
-            self.slot_ages = torch.zeros(cap, device=self.device, dtype=torch.int32)
+            self.slot_ages = self._new_ages(cap)
 
 
 class DataArena:

@@ -58,6 +58,23 @@ class TorchBackend:
 
     name = "torch"
 
+    def __init__(self):
+        #: per-spec partition tensors: id(spec) -> (perm, ptr, arena_part)
+        self._part_cache = {}
+
+    def _part(self, spec, device):
+        """(perm[D], ptr[P+1], arena_part[D]) on ``device`` for a
+        partitioned spec (cached)."""
+        key = (id(spec), str(device))
+        hit = self._part_cache.get(key)
+        if hit is None:
+            perm = torch.from_numpy(spec.part_perm().astype(np.int64)).to(device)
+            ptr = spec.part_ptr()
+            apart = torch.from_numpy(spec.arena_part().astype(np.int64)).to(device)
+            hit = (perm, ptr, apart)
+            self._part_cache[key] = hit
+        return hit
+
     # -- init ----------------------------------------------------------------
 
     def init_params(
@@ -125,7 +142,9 @@ class TorchBackend:
         delivery loop)."""
         if len(nodes) == 0:
             return
-        if spec.family == "logreg":
+        if getattr(spec, "n_parts", 0) > 0:
+            self._update_part(state.params, state.ages, data, spec, nodes.long())
+        elif spec.family == "logreg":
             self._update_logreg(state.params, state.ages, data, spec, nodes.long())
         elif spec.family == "mlp":
             self._update_mlp(state.params, state.ages, data, spec, nodes.long())
@@ -237,6 +256,95 @@ class TorchBackend:
                 w += spec.lr * err * xs
             ages[idx] += c
 
+    def _update_part(self, params, ages, data, spec, nodes) -> None:
+        """Partitioned local step (PartitionedTMH._local_step +
+        _adjust_gradient, gossipy/model/handler.py:503-520): the whole age
+        vector is incremented once per batch, and each parameter's gradient
+        is divided by the age of its partition before the SGD step."""
+        d, k = spec.d_in, spec.n_classes
+        _, _, apart = self._part(spec, params.device)
+        for idx in nodes.tolist():
+            c = int(data.counts[idx])
+            if c == 0:
+                continue
+            x = data.x[idx, :c]
+            y = data.y[idx, :c].long()
+            row = params[idx]
+            W = row[: k * d].view(k, d)
+            b = row[k * d :]
+            bs = c if spec.batch_size == 0 else spec.batch_size
+            for _ in range(max(1, spec.local_epochs)):
+                for s in range(0, c, bs):
+                    xb, yb = x[s : s + bs], y[s : s + bs]
+                    m = xb.shape[0]
+                    ages[idx] += 1
+                    z = xb @ W.t() + b
+                    a = torch.sigmoid(z)
+                    p = torch.softmax(a, dim=1)
+                    p[torch.arange(m), yb] -= 1.0
+                    dz = (p / m) * a * (1.0 - a)
+                    g = torch.empty_like(row)
+                    g[: k * d] = (dz.t() @ xb).reshape(-1)
+                    g[k * d :] = dz.sum(0)
+                    g /= ages[idx].float()[apart]
+                    if spec.weight_decay:
+                        g += spec.weight_decay * row
+                    row -= spec.lr * g
+
+    def _merge_part(self, state, pool, spec, node: int, slot: int, pid: int) -> None:
+        """Age-weighted single-partition merge
+        (gossipy/model/handler.py:497-501 + gossipy/model/sampling.py:201-234)."""
+        perm, ptr, _ = self._part(spec, state.params.device)
+        idx = perm[int(ptr[pid]) : int(ptr[pid + 1])]
+        w1 = int(state.ages[node, pid])
+        w2 = int(pool.slot_ages[slot, pid])
+        if (w1, w2) == (0, 0):
+            w1 = w2 = 1
+        m1, m2 = w1 / (w1 + w2), w2 / (w1 + w2)
+        state.params[node, idx] = (
+            m1 * state.params[node, idx] + m2 * pool.slots[slot, idx]
+        )
+        state.ages[node, pid] = max(
+            int(state.ages[node, pid]), int(pool.slot_ages[slot, pid])
+        )
+
+    def _deliver_part(
+        self, state, pool, data, spec, recv_nodes, recv_ptr, del_slots,
+        reply_slots, del_pids,
+    ) -> None:
+        """Partitioned mode dispatch (PartitionedTMH.__call__,
+        gossipy/model/handler.py:477-494; PASS is an error there)."""
+        mode = spec.mode
+        ptr = recv_ptr.tolist()
+        for i, node_t in enumerate(recv_nodes.tolist()):
+            node = torch.tensor([node_t])
+            for j in range(ptr[i], ptr[i + 1]):
+                slot = int(del_slots[j])
+                pid = int(del_pids[j])
+                if mode == CreateModelMode.MERGE_UPDATE:
+                    self._merge_part(state, pool, spec, node_t, slot, pid)
+                    self.update(state, data, spec, node)
+                elif mode == CreateModelMode.UPDATE:
+                    # train the received model, then merge its partition
+                    self._train_slot(state, pool, data, spec, node_t, slot)
+                    self._merge_part(state, pool, spec, node_t, slot, pid)
+                elif mode == CreateModelMode.UPDATE_MERGE:
+                    self.update(state, data, spec, node)
+                    self._train_slot(state, pool, data, spec, node_t, slot)
+                    self._merge_part(state, pool, spec, node_t, slot, pid)
+                else:
+                    raise ValueError(
+                        "Mode PASS not allowed for partitioned models."
+                    )
+                r = int(reply_slots[j])
+                if r >= 0:
+                    self.snapshot(
+                        state,
+                        pool,
+                        torch.tensor([node_t], dtype=torch.long),
+                        torch.tensor([r], dtype=torch.long),
+                    )
+
     # -- deliveries ----------------------------------------------------------
 
     def deliver(
@@ -249,10 +357,17 @@ class TorchBackend:
         recv_ptr: torch.Tensor,
         del_slots: torch.Tensor,
         reply_slots: torch.Tensor,
+        del_pids: Optional[torch.Tensor] = None,
     ) -> None:
         """Per receiver, apply its deliveries in order: mode-dispatched
         merge/update per message, then write the reply snapshot if the
         message asked for one (PUSH_PULL)."""
+        if getattr(spec, "n_parts", 0) > 0:
+            self._deliver_part(
+                state, pool, data, spec, recv_nodes, recv_ptr, del_slots,
+                reply_slots, del_pids,
+            )
+            return
         mode = spec.mode
         ptr = recv_ptr.tolist()
         for i, node_t in enumerate(recv_nodes.tolist()):
@@ -346,6 +461,7 @@ class HIPBackend(TorchBackend):
     name = "hip"
 
     def __init__(self):
+        super().__init__()
         from .. import ops
 
         self.ext = ops.load_extension()  # raises if the .so is missing
@@ -360,6 +476,7 @@ class HIPBackend(TorchBackend):
             pool.slot_ages,
             nodes.to(state.params.device, torch.int32),
             slot_ids.to(state.params.device, torch.int32),
+            getattr(state, "age_width", 1),
         )
 
     def update(self, state, data, spec, nodes) -> None:
@@ -367,14 +484,23 @@ class HIPBackend(TorchBackend):
             return
         nodes_dev = nodes.to(state.params.device, torch.int32)
         empty = torch.zeros(0, dtype=torch.int32, device=state.params.device)
-        self._dispatch(state, None, data, spec, nodes_dev, None, empty, empty, update_only=True)
+        self._dispatch(
+            state, None, data, spec, nodes_dev, None, empty, empty, empty,
+            update_only=True,
+        )
 
     def deliver(
-        self, state, pool, data, spec, recv_nodes, recv_ptr, del_slots, reply_slots
+        self, state, pool, data, spec, recv_nodes, recv_ptr, del_slots,
+        reply_slots, del_pids=None,
     ) -> None:
         if len(recv_nodes) == 0:
             return
         dev = state.params.device
+        pids = (
+            del_pids.to(dev, torch.int32)
+            if del_pids is not None
+            else torch.zeros(0, dtype=torch.int32, device=dev)
+        )
         self._dispatch(
             state,
             pool,
@@ -384,22 +510,68 @@ class HIPBackend(TorchBackend):
             recv_ptr.to(dev, torch.int32),
             del_slots.to(dev, torch.int32),
             reply_slots.to(dev, torch.int32),
+            pids,
             update_only=False,
         )
 
+    def _part_dev(self, spec, dev):
+        """(perm, ptr, arena_part) as int32 device tensors (cached)."""
+        key = ("dev32", id(spec), str(dev))
+        hit = self._part_cache.get(key)
+        if hit is None:
+            perm = torch.from_numpy(spec.part_perm()).to(dev)
+            ptr = torch.from_numpy(spec.part_ptr()).to(dev)
+            apart = torch.from_numpy(spec.arena_part()).to(dev)
+            hit = (perm, ptr, apart)
+            self._part_cache[key] = hit
+        return hit
+
     def _dispatch(
-        self, state, pool, data, spec, nodes, recv_ptr, del_slots, reply_slots, update_only
+        self, state, pool, data, spec, nodes, recv_ptr, del_slots, reply_slots,
+        del_pids, update_only
     ):
         dev = state.params.device
+        aw = getattr(state, "age_width", 1)
         if pool is None:
             # update-only call: fabricate an empty pool of the right width
             slots = torch.zeros(1, state.D, device=dev)
-            slot_ages = torch.zeros(1, device=dev, dtype=torch.int32)
+            slot_ages = torch.zeros(
+                (1, aw) if aw > 1 else (1,), device=dev, dtype=torch.int32
+            )
             recv_ptr = torch.zeros(len(nodes) + 1, dtype=torch.int32, device=dev)
         else:
             slots, slot_ages = pool.slots, pool.slot_ages
         mode = _MODE_ID[spec.mode]
-        if spec.family == "logreg":
+        if getattr(spec, "n_parts", 0) > 0:
+            assert spec.family == "logreg", "partitioned HIP path: logreg only"
+            perm, ptr, apart = self._part_dev(spec, dev)
+            self.ext.tick_logreg_part(
+                state.params,
+                state.ages,
+                slots,
+                slot_ages,
+                nodes,
+                recv_ptr,
+                del_slots,
+                reply_slots,
+                del_pids,
+                data.x,
+                data.y,
+                data.counts,
+                perm,
+                ptr,
+                apart,
+                spec.n_parts,
+                spec.d_in,
+                spec.n_classes,
+                spec.lr,
+                spec.weight_decay,
+                max(1, spec.local_epochs),
+                spec.batch_size,
+                mode,
+                bool(update_only),
+            )
+        elif spec.family == "logreg":
             self.ext.tick_logreg(
                 state.params,
                 state.ages,

@@ -21,11 +21,76 @@ which batched kernel trains them:
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import Tuple
+from typing import List, Tuple
+
+import numpy as np
 
 from ..core import CreateModelMode
 
 __all__ = ["PegasosSpec", "AdaLineSpec", "LogRegSpec", "MLPSpec"]
+
+
+def _fortran_to_arena(layers: List[Tuple[int, int, int, int]], D: int) -> np.ndarray:
+    """Permutation mapping *reference partition order* -> arena offset.
+
+    The reference's :class:`TorchModelPartition` splits the model into
+    contiguous ranges of the Fortran-flat (dim-0 fastest) concatenation of
+    its parameter tensors (gossipy/model/sampling.py:144-198). The arena
+    packs each ``W`` row-major. ``perm[g]`` is the arena offset of global
+    F-flat position ``g``, so partition ``p`` is ``perm[lo:hi]`` for the
+    p-th equal split ``[lo, hi)``.
+
+    ``layers``: per-layer ``(w_off, b_off, fin, fout)`` arena offsets.
+    """
+    perm = np.empty(D, dtype=np.int32)
+    g = 0
+    for (w_off, b_off, fin, fout) in layers:
+        n_w = fin * fout
+        idx = np.arange(n_w)
+        # F-order over shape (fout, fin): position i -> (row i%fout, col i//fout)
+        perm[g : g + n_w] = w_off + (idx % fout) * fin + idx // fout
+        g += n_w
+        perm[g : g + fout] = b_off + np.arange(fout)
+        g += fout
+    assert g == D
+    return perm
+
+
+class _PartitionMixin:
+    """Partition cover helpers shared by partitionable specs.
+
+    Active when ``n_parts > 0``. The cover replicates the reference's
+    equal-size F-flat split: partition ``p`` owns ``mu`` scalars (``mu+1``
+    for the first ``D % n_parts``).
+    """
+
+    def part_ptr(self) -> np.ndarray:
+        """``[P+1]`` int32 boundaries of the partition split."""
+        P = self.n_parts
+        mu, rem = divmod(self.D, P)
+        sizes = np.full(P, mu, dtype=np.int32)
+        sizes[:rem] += 1
+        return np.concatenate([[0], np.cumsum(sizes)]).astype(np.int32)
+
+    def part_perm(self) -> np.ndarray:
+        """``[D]`` int32: arena offset of each F-flat position (partition
+        ``p`` = ``part_perm()[ptr[p]:ptr[p+1]]``)."""
+        return _fortran_to_arena(self._arena_layers(), self.D)
+
+    def arena_part(self) -> np.ndarray:
+        """``[D]`` int32: partition id owning each *arena* offset (the
+        inverse view, used by the gradient age-rescale,
+        gossipy/model/handler.py:514-520)."""
+        perm = self.part_perm()
+        ptr = self.part_ptr()
+        out = np.empty(self.D, dtype=np.int32)
+        for p in range(self.n_parts):
+            out[perm[ptr[p] : ptr[p + 1]]] = p
+        return out
+
+    @property
+    def age_width(self) -> int:
+        return max(1, self.n_parts)
 
 
 @dataclass
@@ -38,6 +103,8 @@ class PegasosSpec:
     mode: CreateModelMode = CreateModelMode.MERGE_UPDATE
 
     family = "pegasos"
+    n_parts = 0
+    age_width = 1
 
     @property
     def D(self) -> int:
@@ -54,6 +121,8 @@ class AdaLineSpec:
     mode: CreateModelMode = CreateModelMode.MERGE_UPDATE
 
     family = "adaline"
+    n_parts = 0
+    age_width = 1
 
     @property
     def D(self) -> int:
@@ -61,10 +130,13 @@ class AdaLineSpec:
 
 
 @dataclass
-class LogRegSpec:
+class LogRegSpec(_PartitionMixin):
     """Logistic regression ``softmax-CE(sigmoid(W x + b))`` with plain SGD.
 
     Parameter row layout: ``W`` (k*d, row-major) then ``b`` (k).
+    ``n_parts > 0`` switches the family to partitioned gossip
+    (PartitionedTMH semantics: per-partition ages, age-weighted partition
+    merge, gradient age-rescale — gossipy/model/handler.py:455-525).
     """
 
     d_in: int
@@ -74,6 +146,7 @@ class LogRegSpec:
     local_epochs: int = 1
     batch_size: int = 32  #: 0 = full-batch
     mode: CreateModelMode = CreateModelMode.MERGE_UPDATE
+    n_parts: int = 0
 
     family = "logreg"
 
@@ -81,9 +154,13 @@ class LogRegSpec:
     def D(self) -> int:
         return self.n_classes * self.d_in + self.n_classes
 
+    def _arena_layers(self):
+        kd = self.n_classes * self.d_in
+        return [(0, kd, self.d_in, self.n_classes)]
+
 
 @dataclass
-class MLPSpec:
+class MLPSpec(_PartitionMixin):
     """MLP with ReLU hidden layers and a linear head
     (gossipy/model/nn.py:67-113); CrossEntropyLoss on the raw head outputs.
 
@@ -99,8 +176,12 @@ class MLPSpec:
     local_epochs: int = 1
     batch_size: int = 32
     mode: CreateModelMode = CreateModelMode.MERGE_UPDATE
+    n_parts: int = 0
 
     family = "mlp"
+
+    def _arena_layers(self):
+        return self.layer_offsets()
 
     @property
     def dims(self) -> Tuple[int, ...]:

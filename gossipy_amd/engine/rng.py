@@ -51,6 +51,8 @@ class Purpose(IntEnum):
     DATA = 7  #: data shuffling / minibatch permutations
     TOKEN = 8  #: token-account proactive coin flips
     MISC = 9
+    PART = 10  #: partition-id draws for partitioned gossip (Hegedus 2021)
+    SAMPLE = 11  #: parameter-sample seeds for sampled gossip (Hegedus 2021)
 
 
 class TapeStream:

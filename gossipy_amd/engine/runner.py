@@ -87,8 +87,11 @@ class BatchedGossipSimulator(SimulationEventSender):
             torch.cuda.set_device(device)
 
         self.backend = make_backend(device)
-        self.state = NodeStateArena(self.n_local, spec.D, device, self.node_lo)
-        self.pool = SlotPool(spec.D, device)
+        aw = getattr(spec, "age_width", 1)
+        self.state = NodeStateArena(
+            self.n_local, spec.D, device, self.node_lo, age_width=aw
+        )
+        self.pool = SlotPool(spec.D, device, age_width=aw)
         self.data = data
         self.scheduler = make_scheduler(cfg)
         self.initialized = False
@@ -137,17 +140,18 @@ class BatchedGossipSimulator(SimulationEventSender):
         ops = []
         recv_bufs = []
         D = self.spec.D
+        A = getattr(self.spec, "age_width", 1)
         for src, dst, slot_ids in needed:
             if src == dst:
                 continue
             if src == self.rank:
                 ids = torch.from_numpy(slot_ids.astype(np.int64)).to(self.device)
-                buf = torch.empty(len(slot_ids), D + 1, device=self.device)
+                buf = torch.empty(len(slot_ids), D + A, device=self.device)
                 buf[:, :D] = self.pool.slots[ids]
-                buf[:, D] = self.pool.slot_ages[ids].float()
+                buf[:, D:] = self.pool.slot_ages[ids].reshape(len(ids), A).float()
                 ops.append(dist.P2POp(dist.isend, buf, dst))
             elif dst == self.rank:
-                buf = torch.empty(len(slot_ids), D + 1, device=self.device)
+                buf = torch.empty(len(slot_ids), D + A, device=self.device)
                 ops.append(dist.P2POp(dist.irecv, buf, src))
                 recv_bufs.append((slot_ids, buf))
         if ops:
@@ -156,7 +160,8 @@ class BatchedGossipSimulator(SimulationEventSender):
         for slot_ids, buf in recv_bufs:
             ids = torch.from_numpy(slot_ids.astype(np.int64)).to(self.device)
             self.pool.slots[ids] = buf[:, :D]
-            self.pool.slot_ages[ids] = buf[:, D].int()
+            ages = buf[:, D:].int()
+            self.pool.slot_ages[ids] = ages.reshape(self.pool.slot_ages[ids].shape)
 
     def _plan_exchange(
         self,
@@ -213,6 +218,9 @@ class BatchedGossipSimulator(SimulationEventSender):
             )
             new_ptr = np.zeros(len(counts) + 1, dtype=np.int64)
             np.cumsum(counts, out=new_ptr[1:])
+            pids = None
+            if phase.del_pids is not None and len(phase.del_pids):
+                pids = torch.from_numpy(phase.del_pids[sel].astype(np.int64))
             self.backend.deliver(
                 self.state,
                 self.pool,
@@ -222,6 +230,7 @@ class BatchedGossipSimulator(SimulationEventSender):
                 torch.from_numpy(new_ptr),
                 torch.from_numpy(phase.del_slots[sel].astype(np.int64)),
                 torch.from_numpy(phase.reply_slots[sel].astype(np.int64)),
+                del_pids=pids,
             )
         pmine = self._is_mine(phase.pull_snap_nodes)
         if pmine.any():
@@ -256,6 +265,9 @@ class BatchedGossipSimulator(SimulationEventSender):
                 new_ptr = np.zeros(len(counts) + 1, dtype=np.int64)
                 np.cumsum(counts, out=new_ptr[1:])
                 no_reply = torch.full((len(sel),), -1, dtype=torch.int64)
+                rpids = None
+                if phase.rep_pids is not None and len(phase.rep_pids):
+                    rpids = torch.from_numpy(phase.rep_pids[sel].astype(np.int64))
                 self.backend.deliver(
                     self.state,
                     self.pool,
@@ -265,6 +277,7 @@ class BatchedGossipSimulator(SimulationEventSender):
                     torch.from_numpy(new_ptr),
                     torch.from_numpy(phase.rep_del_slots[sel].astype(np.int64)),
                     no_reply,
+                    del_pids=rpids,
                 )
 
     def _evaluate(self, sched: RoundSchedule, t: int) -> None:
@@ -327,6 +340,7 @@ class BatchedGossipSimulator(SimulationEventSender):
             and isinstance(self.scheduler, NativeSchedulerAdapter)
             and getattr(self.backend, "ext", None) is not None
             and self.spec.family in ("logreg", "pegasos", "adaline")
+            and (self.spec.family == "logreg" or getattr(self.spec, "n_parts", 0) == 0)
         )
 
     def _run_round_fast(self, f: dict) -> None:
@@ -345,6 +359,8 @@ class BatchedGossipSimulator(SimulationEventSender):
             "rep_nodes",
             "rep_nptr",
             "rep_slots",
+            "del_pids",
+            "rep_pids",
         )
         parts = [np.ascontiguousarray(f[n], dtype=np.int32) for n in dev_names]
         lens = [len(p) for p in parts]
@@ -387,7 +403,28 @@ class BatchedGossipSimulator(SimulationEventSender):
         from .backend import _MODE_ID
 
         spec = self.spec
-        if spec.family == "logreg":
+        if getattr(spec, "n_parts", 0) > 0:
+            perm, pptr, apart = self.backend._part_dev(spec, dev)
+            # run_round_logreg_part takes del_pids after reply_slots and
+            # rep_pids after rep_slots
+            c = list(common)
+            c.insert(12, views["del_pids"])   # after reply_slots
+            c.insert(20, views["rep_pids"])   # after rep_slots
+            ext.run_round_logreg_part(
+                *c,
+                perm,
+                pptr,
+                apart,
+                spec.n_parts,
+                spec.d_in,
+                spec.n_classes,
+                spec.lr,
+                spec.weight_decay,
+                max(1, spec.local_epochs),
+                spec.batch_size,
+                _MODE_ID[spec.mode],
+            )
+        elif spec.family == "logreg":
             ext.run_round_logreg(
                 *common,
                 spec.d_in,

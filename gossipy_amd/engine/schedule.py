@@ -63,6 +63,9 @@ class EngineConfig:
     #: adjacency CSR (None = fully connected); built once by the runner
     peers_indptr: Optional[np.ndarray] = None
     peers_indices: Optional[np.ndarray] = None
+    #: >0 = partitioned gossip (PartitioningBasedNode, gossipy/node.py:566-659):
+    #: every PUSH/REPLY carries a uniformly drawn partition id
+    n_parts: int = 0
 
 
 @dataclass
@@ -94,6 +97,10 @@ class TickPhase:
     rep_recv_ptr: np.ndarray = None  # int32
     rep_del_slots: np.ndarray = None  # int32
     rep_del_owners: np.ndarray = None  # int32
+    # partitioned gossip: partition id per delivery (aligned with del_slots /
+    # rep_del_slots; length 0 when cfg.n_parts == 0)
+    del_pids: np.ndarray = None  # int32
+    rep_pids: np.ndarray = None  # int32
 
     @property
     def n_events(self) -> int:
@@ -145,8 +152,8 @@ class Scheduler:
                 1, g.normal(cfg.delta, cfg.delta / 10, size=n).astype(np.int64)
             )
         # in-flight messages carried across round boundaries:
-        # tick -> list of (receiver, slot, reply_flag, is_pull_request, sender)
-        self._pending: Dict[int, List[Tuple[int, int, int, bool, int]]] = {}
+        # tick -> list of (receiver, slot, reply_flag, is_pull_request, sender, pid)
+        self._pending: Dict[int, List[Tuple[int, int, int, bool, int, int]]] = {}
         # tick -> (sent, failed, size) accounting of replies enqueued there
         self._reply_accounting: Dict[int, Tuple[int, int, int]] = {}
         # deterministic slot allocator: ids persist across rounds (a delayed
@@ -232,6 +239,15 @@ class Scheduler:
                 drop_u = self.tape.uniform(Purpose.DROP, t, n_f)
                 sizes = np.full(n_f, cfg.model_size if proto != AntiEntropyProtocol.PULL else 1)
                 delays = self._delays(t, n_f, sizes)
+                # partition ids drawn per send (gossipy/node.py:617,631)
+                if cfg.n_parts > 0:
+                    pids = np.atleast_1d(
+                        self.tape.stream(Purpose.PART, t).integers(
+                            0, cfg.n_parts, size=n_f
+                        )
+                    )
+                else:
+                    pids = np.full(n_f, -1, dtype=np.int64)
                 for j in range(n_f):
                     sender, receiver = int(firing[j]), int(peers[j])
                     is_pull = proto == AntiEntropyProtocol.PULL
@@ -246,7 +262,8 @@ class Scheduler:
                         due = t + int(delays[j])
                         wants_reply = proto == AntiEntropyProtocol.PUSH_PULL
                         self._pending.setdefault(due, []).append(
-                            (receiver, slot, -2 if wants_reply else -1, is_pull, sender)
+                            (receiver, slot, -2 if wants_reply else -1, is_pull,
+                             sender, int(pids[j]))
                         )
                     else:
                         failed += 1
@@ -260,29 +277,32 @@ class Scheduler:
             online_u = self.tape.uniform(Purpose.ONLINE, t, cfg.n_nodes)
             online = online_u <= cfg.online_prob
             due = self._pending.pop(t, [])
-            recv_map: Dict[int, List[Tuple[int, int]]] = {}
+            recv_map: Dict[int, List[Tuple[int, int, int, int]]] = {}
             pull_nodes: List[int] = []
             pull_slots: List[int] = []
-            for receiver, slot, reply_flag, is_pull, sender in due:
+            for receiver, slot, reply_flag, is_pull, sender, pid in due:
                 if not online[receiver]:
                     failed += 1
                     if slot >= 0:
                         freed.append(slot)
                     continue
                 if is_pull:
-                    # PULL request: receiver snapshots and replies
+                    # PULL request: receiver snapshots and replies (with a
+                    # fresh partition id, gossipy/node.py:651-653)
                     rslot = self._alloc_slot(receiver)
                     pull_nodes.append(receiver)
                     pull_slots.append(rslot)
-                    if not self._enqueue_reply(t, receiver, sender, rslot):
+                    rpid = self._reply_pid(t, receiver)
+                    if not self._enqueue_reply(t, receiver, sender, rslot, rpid):
                         freed.append(rslot)
                     continue
                 rslot = -1
                 if reply_flag == -2:  # PUSH_PULL: reply with post-merge model
                     rslot = self._alloc_slot(receiver)
-                    if not self._enqueue_reply(t, receiver, sender, rslot):
+                    rpid = self._reply_pid(t, receiver)
+                    if not self._enqueue_reply(t, receiver, sender, rslot, rpid):
                         freed.append(rslot)
-                recv_map.setdefault(receiver, []).append((slot, rslot, sender))
+                recv_map.setdefault(receiver, []).append((slot, rslot, sender, pid))
                 freed.append(slot)  # consumed by this delivery
 
             recv_nodes = np.fromiter(recv_map.keys(), dtype=np.int32, count=len(recv_map))
@@ -290,32 +310,36 @@ class Scheduler:
             del_slots: List[int] = []
             del_owners: List[int] = []
             reply_slots: List[int] = []
+            del_pids: List[int] = []
             for i, rn in enumerate(recv_nodes):
                 pairs = recv_map[int(rn)]
                 del_slots.extend(p[0] for p in pairs)
                 reply_slots.extend(p[1] for p in pairs)
                 del_owners.extend(p[2] for p in pairs)
+                del_pids.extend(p[3] for p in pairs)
                 recv_ptr[i + 1] = recv_ptr[i] + len(pairs)
 
             # --- sub-phase C: replies that came due in THIS tick (delay 0).
             # They were enqueued by the loop above into _pending[t]; pop
             # again. Replies to replies do not exist (parity).
             rep_due = self._pending.pop(t, [])
-            rep_map: Dict[int, List[Tuple[int, int]]] = {}
-            for receiver, slot, _rf, _ip, sender in rep_due:
+            rep_map: Dict[int, List[Tuple[int, int, int]]] = {}
+            for receiver, slot, _rf, _ip, sender, pid in rep_due:
                 if not online[receiver]:
                     failed += 1
                     freed.append(slot)
                     continue
-                rep_map.setdefault(receiver, []).append((slot, sender))
+                rep_map.setdefault(receiver, []).append((slot, sender, pid))
                 freed.append(slot)  # consumed by this reply delivery
             rep_recv = np.fromiter(rep_map.keys(), dtype=np.int32, count=len(rep_map))
             rep_ptr = np.zeros(len(rep_map) + 1, dtype=np.int32)
             rep_slots: List[int] = []
             rep_owners: List[int] = []
+            rep_pids: List[int] = []
             for i, rn in enumerate(rep_recv):
                 rep_slots.extend(p[0] for p in rep_map[int(rn)])
                 rep_owners.extend(p[1] for p in rep_map[int(rn)])
+                rep_pids.extend(p[2] for p in rep_map[int(rn)])
                 rep_ptr[i + 1] = len(rep_slots)
 
             phase = TickPhase(
@@ -333,6 +357,8 @@ class Scheduler:
                 rep_recv_ptr=rep_ptr,
                 rep_del_slots=np.asarray(rep_slots, dtype=np.int32),
                 rep_del_owners=np.asarray(rep_owners, dtype=np.int32),
+                del_pids=np.asarray(del_pids, dtype=np.int32),
+                rep_pids=np.asarray(rep_pids, dtype=np.int32),
             )
             if phase.n_events:
                 ticks.append(phase)
@@ -365,7 +391,20 @@ class Scheduler:
             eval_nodes=eval_nodes,
         )
 
-    def _enqueue_reply(self, t: int, replier: int, requester: int, slot: int) -> bool:
+    def _reply_pid(self, t: int, replier: int) -> int:
+        """Fresh partition id for a reply (gossipy/node.py:651). Keyed on
+        (t, replier): two replies by one node in one tick share the draw."""
+        if self.cfg.n_parts <= 0:
+            return -1
+        return int(
+            self.tape.stream(Purpose.PART, t, extra=1 + replier).integers(
+                0, self.cfg.n_parts
+            )
+        )
+
+    def _enqueue_reply(
+        self, t: int, replier: int, requester: int, slot: int, pid: int = -1
+    ) -> bool:
         """Queue a REPLY message (drop-tested with the reference's
         ``random() > drop_prob`` variant, gossipy/simul.py:414). Returns
         whether the reply was actually enqueued (False = dropped)."""
@@ -386,7 +425,7 @@ class Scheduler:
             else:
                 dly = int(d.get(None))
             self._pending.setdefault(t + dly, []).append(
-                (requester, slot, -1, False, replier)
+                (requester, slot, -1, False, replier, pid)
             )
             self._reply_accounting[t] = (sent, failed, size)
             return True
@@ -440,6 +479,7 @@ class NativeSchedulerAdapter:
             cfg.seed & 0xFFFFFFFFFFFFFFFF,
             None if ip is None else np.ascontiguousarray(ip, dtype=np.int64),
             None if ix is None else np.ascontiguousarray(ix, dtype=np.int64),
+            cfg.n_parts,
         )
         self.last_flat: Optional[dict] = None
 
@@ -474,6 +514,8 @@ class NativeSchedulerAdapter:
                     rep_recv_ptr=f["rep_nptr"][q0 : q1 + 1] - e0,
                     rep_del_slots=f["rep_slots"][e0:e1],
                     rep_del_owners=f["rep_owners"][e0:e1],
+                    del_pids=f["del_pids"][d0:d1],
+                    rep_pids=f["rep_pids"][e0:e1],
                 )
             )
         return RoundSchedule(

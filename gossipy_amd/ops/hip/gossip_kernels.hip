@@ -54,7 +54,7 @@ __global__ void snapshot_kernel(
     int* __restrict__ slot_ages,
     const int* __restrict__ nodes,
     const int* __restrict__ slot_ids,
-    int n, int D)
+    int n, int D, int A)  // A = age width (1 scalar, P for partitioned)
 {
     long total = (long)n * D;
     for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x; idx < total;
@@ -64,7 +64,7 @@ __global__ void snapshot_kernel(
         int node = nodes[row];
         int slot = slot_ids[row];
         slots[(long)slot * D + col] = params[(long)node * D + col];
-        if (col == 0) slot_ages[slot] = ages[node];
+        if (col < A) slot_ages[(long)slot * A + col] = ages[(long)node * A + col];
     }
 }
 
@@ -218,6 +218,185 @@ tick_logreg_kernel(LogregArgs a)
     for (int e = tid; e < a.D; e += blockDim.x)
         a.params[(long)node * a.D + e] = W[e];
     if (tid == 0) a.ages[node] = age;
+}
+
+// ---------------------------------------------------------------------------
+// partitioned logreg tick (K7/K8, PartitionedTMH semantics)
+//
+// Parity: gossipy/model/handler.py:455-525 + gossipy/model/sampling.py:
+// 201-234. Each node keeps a per-partition age vector; a delivery merges
+// ONE partition (age-weighted), and each local batch increments every
+// partition age then divides each parameter's gradient by its partition's
+// age. The partition cover is the reference's F-flat equal split mapped to
+// arena offsets via the device-resident perm/pptr arrays (engine/models.py
+// _PartitionMixin).
+// ---------------------------------------------------------------------------
+
+struct LogregPartArgs {
+    float* params; int* ages;            // ages: [n_local, P]
+    float* slots; int* slot_ages;        // slot_ages: [cap, P]
+    const int* nodes; const int* ptr;
+    const int* dslots; const int* rslots; const int* dpids;
+    const float* X; const float* Y; const int* counts;
+    const int* perm;   // [D]  partition-ordered -> arena offset
+    const int* pptr;   // [P+1]
+    const int* apart;  // [D]  arena offset -> partition id
+    int P, d, k, Smax, D;
+    float lr, wd;
+    int epochs, bs, mode, update_only;
+};
+
+// Minibatch SGD with whole-age-vector increment per batch and per-partition
+// gradient rescale (gossipy/model/handler.py:503-520).
+DEV_INLINE void logreg_part_update(const LogregPartArgs& a, int node, float* W,
+                                   int* agev, float* xb, float* dz)
+{
+    int tid = threadIdx.x;
+    int c = a.counts[node];
+    if (c == 0) return;
+    int bsz = (a.bs == 0) ? c : min(a.bs, c);
+    const float* Xn = a.X + (long)node * a.Smax * a.d;
+    const float* Yn = a.Y + (long)node * a.Smax;
+    for (int ep = 0; ep < a.epochs; ++ep) {
+        for (int s0 = 0; s0 < c; s0 += bsz) {
+            int m = min(bsz, c - s0);
+            for (int e = tid; e < m * a.d; e += blockDim.x)
+                xb[e] = Xn[(long)s0 * a.d + e];
+            // self.n_updates += 1 happens before the step (handler.py:506)
+            for (int p = tid; p < a.P; p += blockDim.x) agev[p] += 1;
+            __syncthreads();
+            if (tid < m) {
+                float z[KMAX];
+                for (int kk = 0; kk < a.k; ++kk) {
+                    float acc = W[a.k * a.d + kk];
+                    const float* wrow = W + kk * a.d;
+                    const float* xrow = xb + tid * a.d;
+                    for (int dd = 0; dd < a.d; ++dd) acc += wrow[dd] * xrow[dd];
+                    z[kk] = acc;
+                }
+                float amax = -1e30f;
+                for (int kk = 0; kk < a.k; ++kk) {
+                    z[kk] = 1.0f / (1.0f + __expf(-z[kk]));
+                    amax = fmaxf(amax, z[kk]);
+                }
+                float sum = 0.f;
+                float p[KMAX];
+                for (int kk = 0; kk < a.k; ++kk) {
+                    p[kk] = __expf(z[kk] - amax);
+                    sum += p[kk];
+                }
+                int yi = (int)Yn[s0 + tid];
+                float inv = 1.0f / sum;
+                for (int kk = 0; kk < a.k; ++kk) {
+                    float g = p[kk] * inv - (kk == yi ? 1.0f : 0.0f);
+                    dz[tid * a.k + kk] = (g / m) * z[kk] * (1.0f - z[kk]);
+                }
+            }
+            __syncthreads();
+            for (int e = tid; e < a.k * a.d; e += blockDim.x) {
+                int kk = e / a.d, dd = e - kk * a.d;
+                float g = 0.f;
+                for (int s = 0; s < m; ++s) g += dz[s * a.k + kk] * xb[s * a.d + dd];
+                g /= (float)agev[a.apart[e]];  // _adjust_gradient
+                if (a.wd != 0.f) g += a.wd * W[e];
+                W[e] -= a.lr * g;
+            }
+            for (int e = tid; e < a.k; e += blockDim.x) {
+                float g = 0.f;
+                for (int s = 0; s < m; ++s) g += dz[s * a.k + e];
+                g /= (float)agev[a.apart[a.k * a.d + e]];
+                if (a.wd != 0.f) g += a.wd * W[a.k * a.d + e];
+                W[a.k * a.d + e] -= a.lr * g;
+            }
+            __syncthreads();
+        }
+    }
+}
+
+// Age-weighted merge of partition `pid` of src into W
+// (handler.py:497-501; weights (0,0) -> (1,1)). src/src_ages may point to
+// LDS (trained received model) or HBM (slot row).
+DEV_INLINE void logreg_part_merge(const LogregPartArgs& a, float* W, int* agev,
+                                  const float* src, const int* src_ages, int pid)
+{
+    int tid = threadIdx.x;
+    int w1 = agev[pid], w2 = src_ages[pid];
+    float m1, m2;
+    if (w1 == 0 && w2 == 0) { m1 = 0.5f; m2 = 0.5f; }
+    else {
+        float s = (float)(w1 + w2);
+        m1 = (float)w1 / s;
+        m2 = (float)w2 / s;
+    }
+    __syncthreads();
+    for (int e = a.pptr[pid] + tid; e < a.pptr[pid + 1]; e += blockDim.x) {
+        int i = a.perm[e];
+        W[i] = m1 * W[i] + m2 * src[i];
+    }
+    __syncthreads();
+    if (tid == 0) agev[pid] = max(w1, w2);
+    __syncthreads();
+}
+
+__global__ void __launch_bounds__(128)
+tick_logreg_part_kernel(LogregPartArgs a)
+{
+    int i = blockIdx.x;
+    int node = a.nodes[i];
+    int tid = threadIdx.x;
+    extern __shared__ float sm[];
+    float* W = sm;                         // D
+    float* W2 = W + a.D;                   // D (received-model scratch)
+    float* xb = W2 + a.D;                  // bsmax*d
+    int bsmax = (a.bs == 0) ? a.Smax : min(a.bs, a.Smax);
+    float* dz = xb + bsmax * a.d;          // bsmax*k
+    int* agev = (int*)(dz + bsmax * a.k);  // P
+    int* agev2 = agev + a.P;               // P
+
+    for (int e = tid; e < a.D; e += blockDim.x)
+        W[e] = a.params[(long)node * a.D + e];
+    for (int p = tid; p < a.P; p += blockDim.x)
+        agev[p] = a.ages[(long)node * a.P + p];
+    __syncthreads();
+
+    if (a.update_only) {
+        logreg_part_update(a, node, W, agev, xb, dz);
+    } else {
+        for (int j = a.ptr[i]; j < a.ptr[i + 1]; ++j) {
+            int slot = a.dslots[j];
+            int pid = a.dpids ? a.dpids[j] : 0;
+            const float* srow = a.slots + (long)slot * a.D;
+            const int* sages = a.slot_ages + (long)slot * a.P;
+            if (a.mode == MODE_MERGE_UPDATE) {
+                logreg_part_merge(a, W, agev, srow, sages, pid);
+                logreg_part_update(a, node, W, agev, xb, dz);
+            } else {
+                // UPDATE: train the received model, merge its partition
+                // (handler.py:481-483). UPDATE_MERGE: also self-update first
+                // (handler.py:487-490). PASS is rejected host-side.
+                if (a.mode == MODE_UPDATE_MERGE)
+                    logreg_part_update(a, node, W, agev, xb, dz);
+                for (int e = tid; e < a.D; e += blockDim.x) W2[e] = srow[e];
+                for (int p = tid; p < a.P; p += blockDim.x) agev2[p] = sages[p];
+                __syncthreads();
+                logreg_part_update(a, node, W2, agev2, xb, dz);
+                logreg_part_merge(a, W, agev, W2, agev2, pid);
+            }
+            int rs = a.rslots ? a.rslots[j] : -1;
+            if (rs >= 0) {
+                for (int e = tid; e < a.D; e += blockDim.x)
+                    a.slots[(long)rs * a.D + e] = W[e];
+                for (int p = tid; p < a.P; p += blockDim.x)
+                    a.slot_ages[(long)rs * a.P + p] = agev[p];
+                __syncthreads();
+            }
+        }
+    }
+    __syncthreads();
+    for (int e = tid; e < a.D; e += blockDim.x)
+        a.params[(long)node * a.D + e] = W[e];
+    for (int p = tid; p < a.P; p += blockDim.x)
+        a.ages[(long)node * a.P + p] = agev[p];
 }
 
 // ---------------------------------------------------------------------------
@@ -550,7 +729,8 @@ static hipStream_t current_stream()
 }
 
 void snapshot(torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
-              torch::Tensor slot_ages, torch::Tensor nodes, torch::Tensor slot_ids)
+              torch::Tensor slot_ages, torch::Tensor nodes, torch::Tensor slot_ids,
+              int64_t age_width)
 {
     CHECK_DEV(params); CHECK_DEV(slots); CHECK_DEV(nodes); CHECK_DEV(slot_ids);
     int n = nodes.size(0);
@@ -562,7 +742,45 @@ void snapshot(torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
     hipLaunchKernelGGL(snapshot_kernel, dim3(grid), dim3(block), 0, current_stream(),
         params.data_ptr<float>(), ages.data_ptr<int>(),
         slots.data_ptr<float>(), slot_ages.data_ptr<int>(),
-        nodes.data_ptr<int>(), slot_ids.data_ptr<int>(), n, D);
+        nodes.data_ptr<int>(), slot_ids.data_ptr<int>(), n, D, (int)age_width);
+}
+
+void tick_logreg_part(
+    torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
+    torch::Tensor slot_ages, torch::Tensor nodes, torch::Tensor recv_ptr,
+    torch::Tensor del_slots, torch::Tensor reply_slots, torch::Tensor del_pids,
+    torch::Tensor X, torch::Tensor Y, torch::Tensor counts,
+    torch::Tensor perm, torch::Tensor pptr, torch::Tensor apart,
+    int64_t n_parts, int64_t d, int64_t k, double lr, double wd,
+    int64_t epochs, int64_t bs, int64_t mode, bool update_only)
+{
+    CHECK_DEV(params); CHECK_DEV(slots); CHECK_DEV(X); CHECK_DEV(Y);
+    CHECK_DEV(perm); CHECK_DEV(pptr); CHECK_DEV(apart);
+    TORCH_CHECK(k <= KMAX, "n_classes > ", KMAX, " unsupported");
+    TORCH_CHECK(mode != MODE_PASS, "Mode PASS not allowed for partitioned models.");
+    int n = nodes.size(0);
+    if (n == 0) return;
+    LogregPartArgs a;
+    a.params = params.data_ptr<float>(); a.ages = ages.data_ptr<int>();
+    a.slots = slots.data_ptr<float>(); a.slot_ages = slot_ages.data_ptr<int>();
+    a.nodes = nodes.data_ptr<int>(); a.ptr = recv_ptr.data_ptr<int>();
+    a.dslots = del_slots.numel() ? del_slots.data_ptr<int>() : nullptr;
+    a.rslots = reply_slots.numel() ? reply_slots.data_ptr<int>() : nullptr;
+    a.dpids = del_pids.numel() ? del_pids.data_ptr<int>() : nullptr;
+    a.X = X.data_ptr<float>(); a.Y = Y.data_ptr<float>();
+    a.counts = counts.data_ptr<int>();
+    a.perm = perm.data_ptr<int>(); a.pptr = pptr.data_ptr<int>();
+    a.apart = apart.data_ptr<int>();
+    a.P = n_parts; a.d = d; a.k = k; a.Smax = X.size(1); a.D = params.size(1);
+    a.lr = lr; a.wd = wd; a.epochs = epochs; a.bs = bs; a.mode = mode;
+    a.update_only = update_only;
+    int bsmax = (bs == 0) ? a.Smax : std::min<int>(bs, a.Smax);
+    size_t smem = sizeof(float) * (2 * a.D + (size_t)bsmax * a.d +
+                                   (size_t)bsmax * a.k) +
+                  sizeof(int) * 2 * a.P;
+    TORCH_CHECK(smem <= 160 * 1024, "partitioned logreg LDS budget exceeded: ", smem);
+    hipLaunchKernelGGL(tick_logreg_part_kernel, dim3(n), dim3(128), smem,
+                       current_stream(), a);
 }
 
 void tick_logreg(torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
@@ -716,13 +934,13 @@ static RoundArrays unpack_round(
 
 static void launch_snap(const float* params, const int* ages, float* slots,
                         int* slot_ages, const int* nodes, const int* slot_ids,
-                        int n, int D, hipStream_t s)
+                        int n, int D, hipStream_t s, int A = 1)
 {
     long total = (long)n * D;
     int block = 256;
     int grid = (int)std::min<long>((total + block - 1) / block, 2048);
     hipLaunchKernelGGL(snapshot_kernel, dim3(grid), dim3(block), 0, s,
-                       params, ages, slots, slot_ages, nodes, slot_ids, n, D);
+                       params, ages, slots, slot_ages, nodes, slot_ids, n, D, A);
 }
 
 void run_round_logreg(
@@ -841,6 +1059,79 @@ void run_round_linear(
     }
 }
 
+void run_round_logreg_part(
+    torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
+    torch::Tensor slot_ages,
+    torch::Tensor snap_nodes, torch::Tensor snap_slots, torch::Tensor snap_tptr,
+    torch::Tensor recv_nodes, torch::Tensor recv_nptr, torch::Tensor recv_tptr,
+    torch::Tensor del_slots, torch::Tensor reply_slots, torch::Tensor del_pids,
+    torch::Tensor pull_nodes, torch::Tensor pull_slots, torch::Tensor pull_tptr,
+    torch::Tensor rep_nodes, torch::Tensor rep_nptr, torch::Tensor rep_tptr,
+    torch::Tensor rep_slots, torch::Tensor rep_pids,
+    torch::Tensor X, torch::Tensor Y, torch::Tensor counts,
+    torch::Tensor perm, torch::Tensor pptr, torch::Tensor apart,
+    int64_t n_parts, int64_t d, int64_t k, double lr, double wd,
+    int64_t epochs, int64_t bs, int64_t mode)
+{
+    CHECK_DEV(params); CHECK_DEV(slots); CHECK_DEV(X);
+    CHECK_DEV(perm); CHECK_DEV(pptr); CHECK_DEV(apart);
+    TORCH_CHECK(mode != MODE_PASS, "Mode PASS not allowed for partitioned models.");
+    RoundArrays r = unpack_round(snap_nodes, snap_slots, snap_tptr, recv_nodes,
+                                 recv_nptr, recv_tptr, del_slots, reply_slots,
+                                 pull_nodes, pull_slots, pull_tptr, rep_nodes,
+                                 rep_nptr, rep_tptr, rep_slots);
+    const int* d_pids = del_pids.numel() ? del_pids.data_ptr<int>() : nullptr;
+    const int* r_pids = rep_pids.numel() ? rep_pids.data_ptr<int>() : nullptr;
+    hipStream_t s = current_stream();
+    LogregPartArgs a;
+    a.params = params.data_ptr<float>(); a.ages = ages.data_ptr<int>();
+    a.slots = slots.data_ptr<float>(); a.slot_ages = slot_ages.data_ptr<int>();
+    a.X = X.data_ptr<float>(); a.Y = Y.data_ptr<float>();
+    a.counts = counts.data_ptr<int>();
+    a.perm = perm.data_ptr<int>(); a.pptr = pptr.data_ptr<int>();
+    a.apart = apart.data_ptr<int>();
+    a.P = n_parts; a.d = d; a.k = k; a.Smax = X.size(1); a.D = params.size(1);
+    a.lr = lr; a.wd = wd; a.epochs = epochs; a.bs = bs; a.mode = mode;
+    a.update_only = 0;
+    int bsmax = (bs == 0) ? a.Smax : std::min<int>(bs, a.Smax);
+    size_t smem = sizeof(float) * (2 * a.D + (size_t)bsmax * a.d +
+                                   (size_t)bsmax * a.k) +
+                  sizeof(int) * 2 * a.P;
+    TORCH_CHECK(smem <= 160 * 1024, "partitioned logreg LDS budget exceeded");
+    for (int t = 0; t < r.delta; ++t) {
+        int s0 = r.snap_tptr[t], s1 = r.snap_tptr[t + 1];
+        if (s1 > s0)
+            launch_snap(a.params, a.ages, a.slots, a.slot_ages,
+                        r.snap_nodes + s0, r.snap_slots + s0, s1 - s0, a.D, s,
+                        a.P);
+        int r0 = r.recv_tptr[t], r1 = r.recv_tptr[t + 1];
+        if (r1 > r0) {
+            a.nodes = r.recv_nodes + r0;
+            a.ptr = r.recv_nptr + r0;
+            a.dslots = r.del_slots;
+            a.rslots = r.reply_slots;
+            a.dpids = d_pids;
+            hipLaunchKernelGGL(tick_logreg_part_kernel, dim3(r1 - r0), dim3(128),
+                               smem, s, a);
+        }
+        int p0 = r.pull_tptr[t], p1 = r.pull_tptr[t + 1];
+        if (p1 > p0)
+            launch_snap(a.params, a.ages, a.slots, a.slot_ages,
+                        r.pull_nodes + p0, r.pull_slots + p0, p1 - p0, a.D, s,
+                        a.P);
+        int q0 = r.rep_tptr[t], q1 = r.rep_tptr[t + 1];
+        if (q1 > q0) {
+            a.nodes = r.rep_nodes + q0;
+            a.ptr = r.rep_nptr + q0;
+            a.dslots = r.rep_slots;
+            a.rslots = nullptr;
+            a.dpids = r_pids;
+            hipLaunchKernelGGL(tick_logreg_part_kernel, dim3(q1 - q0), dim3(128),
+                               smem, s, a);
+        }
+    }
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m)
 {
     m.def("snapshot", &snapshot, "batched model snapshot (arena row copy)");
@@ -851,4 +1142,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m)
           "whole-round executor, logreg family");
     m.def("run_round_linear", &run_round_linear,
           "whole-round executor, pegasos/adaline family");
+    m.def("tick_logreg_part", &tick_logreg_part,
+          "fused partition-merge + age-rescaled logreg SGD tick (K7/K8)");
+    m.def("run_round_logreg_part", &run_round_logreg_part,
+          "whole-round executor, partitioned logreg family");
 }

@@ -290,3 +290,223 @@ def test_two_rank_matches_single_rank():
     assert np.array_equal(single, multi), (
         "2-rank run must be bit-identical to 1-rank run"
     )
+
+
+# ---------------------------------------------------------------------------
+# partitioned gossip (PartitionedTMH / PartitioningBasedNode semantics)
+# ---------------------------------------------------------------------------
+
+
+class TestPartitioned:
+    """Batched partitioned engine vs the object layer (which itself mirrors
+    gossipy/model/handler.py:455-525 + gossipy/model/sampling.py:110-234)."""
+
+    def _obj_handler(self, d=11, k=3, n_parts=4, lr=0.1, mode=None):
+        from gossipy_amd.core import CreateModelMode
+        from gossipy_amd.model.handler import PartitionedTMH
+        from gossipy_amd.model.nn import LogisticRegression
+        from gossipy_amd.model.sampling import TorchModelPartition
+
+        net = LogisticRegression(d, k)
+        part = TorchModelPartition(net, n_parts)
+        h = PartitionedTMH(
+            net,
+            part,
+            torch.optim.SGD,
+            {"lr": lr},
+            torch.nn.CrossEntropyLoss(),
+            local_epochs=1,
+            batch_size=0,
+            create_model_mode=mode or CreateModelMode.MERGE_UPDATE,
+        )
+        return h
+
+    def _spec(self, d=11, k=3, n_parts=4, lr=0.1):
+        return LogRegSpec(
+            d_in=d, n_classes=k, lr=lr, local_epochs=1, batch_size=0,
+            n_parts=n_parts,
+        )
+
+    @staticmethod
+    def _row_from_model(model) -> torch.Tensor:
+        W = model.model.weight.detach().reshape(-1)
+        b = model.model.bias.detach()
+        return torch.cat([W, b]).clone()
+
+    def test_partition_cover_matches_object_layer(self):
+        """spec.part_perm/ptr must reproduce TorchModelPartition's cover
+        translated to arena offsets."""
+        d, k, P = 11, 3, 4
+        h = self._obj_handler(d, k, P)
+        spec = self._spec(d, k, P)
+        perm, ptr = spec.part_perm(), spec.part_ptr()
+        for p in range(P):
+            arena = []
+            ids_w = h.tm_partition.partitions[p][0]
+            if ids_w is not None:
+                arena.extend((ids_w[0] * d + ids_w[1]).tolist())
+            ids_b = h.tm_partition.partitions[p][1]
+            if ids_b is not None:
+                arena.extend((k * d + ids_b[0]).tolist())
+            assert sorted(arena) == sorted(perm[ptr[p] : ptr[p + 1]].tolist())
+
+    def test_update_matches_object_layer(self):
+        """One full-batch partitioned local step: engine oracle ==
+        object-layer autograd path (age increment + grad age-rescale)."""
+        from gossipy_amd.engine.arena import NodeStateArena
+        from gossipy_amd.engine.backend import TorchBackend
+
+        d, k, P = 11, 3, 4
+        h = self._obj_handler(d, k, P)
+        h.n_updates[:] = [2, 0, 5, 1]
+        spec = self._spec(d, k, P)
+
+        X, y = make_synthetic_classification((20, d, k), seed=3)
+        state = NodeStateArena(1, spec.D, torch.device("cpu"), age_width=P)
+        state.params[0] = self._row_from_model(h.model)
+        state.ages[0] = torch.tensor([2, 0, 5, 1], dtype=torch.int32)
+        data = DataArena.from_shards([(X, y)], torch.device("cpu"))
+
+        TorchBackend().update(state, data, spec, torch.tensor([0]))
+        h._update((X, y))
+
+        assert np.array_equal(state.ages[0].numpy(), h.n_updates)
+        assert torch.allclose(
+            state.params[0], self._row_from_model(h.model), atol=1e-6
+        )
+
+    def test_merge_matches_object_layer(self):
+        from gossipy_amd.engine.arena import NodeStateArena, SlotPool
+        from gossipy_amd.engine.backend import TorchBackend
+
+        d, k, P = 11, 3, 4
+        h1 = self._obj_handler(d, k, P)
+        h2 = self._obj_handler(d, k, P)
+        with torch.no_grad():
+            for p_ in h2.model.parameters():
+                p_.add_(torch.randn_like(p_))
+        h1.n_updates[:] = [3, 0, 2, 7]
+        h2.n_updates[:] = [1, 0, 4, 7]
+        spec = self._spec(d, k, P)
+
+        state = NodeStateArena(1, spec.D, torch.device("cpu"), age_width=P)
+        state.params[0] = self._row_from_model(h1.model)
+        state.ages[0] = torch.tensor(h1.n_updates, dtype=torch.int32)
+        pool = SlotPool(spec.D, torch.device("cpu"), 4, age_width=P)
+        pool.slots[2] = self._row_from_model(h2.model)
+        pool.slot_ages[2] = torch.tensor(h2.n_updates, dtype=torch.int32)
+
+        be = TorchBackend()
+        for pid in (0, 1, 3):
+            be._merge_part(state, pool, spec, 0, 2, pid)
+            h1._merge(h2, pid)
+            assert torch.allclose(
+                state.params[0], self._row_from_model(h1.model), atol=1e-6
+            ), f"partition {pid}"
+            assert np.array_equal(state.ages[0].numpy(), h1.n_updates)
+
+    def _run_part(self, mode=None, rounds=15, n_nodes=40, **cfg_kw):
+        from gossipy_amd.core import CreateModelMode
+
+        shards, geval = _make_data(n_nodes, seed=1)
+        data = DataArena.from_shards(shards, torch.device("cpu"), global_eval=geval)
+        base = dict(
+            n_nodes=n_nodes,
+            delta=10,
+            protocol=AntiEntropyProtocol.PUSH,
+            model_size=116,
+            sampling_eval=0.25,
+            seed=7,
+            n_parts=4,
+        )
+        base.update(cfg_kw)
+        cfg = EngineConfig(**base)
+        spec = LogRegSpec(
+            d_in=57, n_classes=2, lr=0.1, n_parts=4,
+            mode=mode or CreateModelMode.MERGE_UPDATE,
+        )
+        sim = BatchedGossipSimulator(cfg, spec, data)
+        rep = SimulationReport()
+        sim.add_receiver(rep)
+        sim.init_nodes()
+        sim.start(n_rounds=rounds)
+        return sim, rep
+
+    def test_partitioned_engine_learns(self):
+        sim, rep = self._run_part()
+        assert rep.get_evaluation(False)[-1][1]["accuracy"] > 0.9
+
+    def test_partitioned_push_pull(self):
+        sim, rep = self._run_part(
+            rounds=10, protocol=AntiEntropyProtocol.PUSH_PULL
+        )
+        assert rep.get_evaluation(False)[-1][1]["accuracy"] > 0.85
+
+    def test_partitioned_deterministic(self):
+        s1, _ = self._run_part(rounds=5)
+        s2, _ = self._run_part(rounds=5)
+        assert torch.equal(s1.local_params(), s2.local_params())
+        assert torch.equal(s1.state.ages, s2.state.ages)
+
+
+def _part_worker(rank, world, port, q):
+    import torch.distributed as dist
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        shards, geval = _make_data(40, seed=1)
+        data = _arena_for_rank(shards, geval, rank, world)
+        cfg = EngineConfig(
+            n_nodes=40,
+            delta=10,
+            protocol=AntiEntropyProtocol.PUSH_PULL,
+            model_size=116,
+            sampling_eval=0.25,
+            seed=7,
+            n_parts=4,
+            delay=UniformDelay(0, 4),
+        )
+        spec = LogRegSpec(d_in=57, n_classes=2, lr=0.1, n_parts=4)
+        sim = BatchedGossipSimulator(cfg, spec, data, device=torch.device("cpu"))
+        sim.init_nodes()
+        sim.start(n_rounds=5)
+        full = sim.gather_params()
+        if rank == 0:
+            q.put(full.numpy())
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_partitioned_two_rank_matches_single_rank():
+    shards, geval = _make_data(40, seed=1)
+    data = DataArena.from_shards(shards, torch.device("cpu"), global_eval=geval)
+    cfg = EngineConfig(
+        n_nodes=40,
+        delta=10,
+        protocol=AntiEntropyProtocol.PUSH_PULL,
+        model_size=116,
+        sampling_eval=0.25,
+        seed=7,
+        n_parts=4,
+        delay=UniformDelay(0, 4),
+    )
+    spec = LogRegSpec(d_in=57, n_classes=2, lr=0.1, n_parts=4)
+    ref = BatchedGossipSimulator(cfg, spec, data, device=torch.device("cpu"))
+    ref.init_nodes()
+    ref.start(n_rounds=5)
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29531
+    procs = [
+        ctx.Process(target=_part_worker, args=(r, 2, port, q)) for r in range(2)
+    ]
+    for p in procs:
+        p.start()
+    got = q.get(timeout=240)
+    for p in procs:
+        p.join(timeout=60)
+    assert np.allclose(ref.local_params().numpy(), got, atol=1e-6)

@@ -271,3 +271,124 @@ class TestFastPath:
         assert torch.allclose(
             fast.local_params(), slow.local_params(), atol=1e-5, rtol=1e-5
         )
+
+
+class TestPartitionedGPU:
+    """tick_logreg_part vs the torch oracle, and the partitioned whole-round
+    executor vs the per-tick path."""
+
+    def _spec(self, mode=CreateModelMode.MERGE_UPDATE):
+        return LogRegSpec(
+            d_in=57, n_classes=2, lr=0.1, local_epochs=1, batch_size=0,
+            n_parts=4, mode=mode,
+        )
+
+    def _pair_part(self, n, spec, seed=3):
+        P = spec.n_parts
+        cs = NodeStateArena(n, spec.D, CPU, age_width=P)
+        tape = RandomTape(seed)
+        TorchBackend().init_params(cs, spec, tape, n)
+        cs.ages += (
+            torch.arange(n * P, dtype=torch.int32).reshape(n, P) % 5
+        )
+        gs = NodeStateArena(n, spec.D, CUDA, age_width=P)
+        gs.params.copy_(cs.params)
+        gs.ages.copy_(cs.ages)
+        cd = _mk_data(n, 57, CPU, seed)
+        gd = DataArena(
+            cd.x.to(CUDA), cd.y.to(CUDA), cd.counts.to(CUDA),
+            gx=cd.gx.to(CUDA), gy=cd.gy.to(CUDA),
+        )
+        return cs, gs, cd, gd
+
+    @pytest.mark.parametrize("mode", [
+        CreateModelMode.MERGE_UPDATE,
+        CreateModelMode.UPDATE,
+        CreateModelMode.UPDATE_MERGE,
+    ])
+    def test_partitioned_deliver_matches_oracle(self, mode):
+        spec = self._spec(mode)
+        P = spec.n_parts
+        cs, gs, cd, gd = self._pair_part(12, spec)
+        cpool = SlotPool(spec.D, CPU, 16, age_width=P)
+        gpool = SlotPool(spec.D, CUDA, 16, age_width=P)
+        cpool.slots.normal_(generator=torch.Generator().manual_seed(5))
+        cpool.slot_ages.copy_(
+            torch.arange(16 * P, dtype=torch.int32).reshape(16, P) % 7
+        )
+        gpool.slots.copy_(cpool.slots)
+        gpool.slot_ages.copy_(cpool.slot_ages)
+        recv = torch.tensor([2, 5, 9], dtype=torch.int64)
+        ptr = torch.tensor([0, 2, 3, 4], dtype=torch.int64)
+        slots = torch.tensor([1, 4, 7, 10], dtype=torch.int64)
+        reply = torch.tensor([12, -1, 13, -1], dtype=torch.int64)
+        pids = torch.tensor([0, 3, 1, 2], dtype=torch.int64)
+        TorchBackend().deliver(cs, cpool, cd, spec, recv, ptr, slots, reply, pids)
+        HIPBackend().deliver(gs, gpool, gd, spec, recv, ptr, slots, reply, pids)
+        torch.cuda.synchronize()
+        assert _close(cs.params, gs.params)
+        assert torch.equal(cs.ages.cpu(), gs.ages.cpu())
+        assert _close(cpool.slots[12:14], gpool.slots[12:14])
+        assert torch.equal(
+            cpool.slot_ages[12:14].cpu(), gpool.slot_ages[12:14].cpu()
+        )
+
+    def test_partitioned_update_only(self):
+        spec = self._spec()
+        cs, gs, cd, gd = self._pair_part(10, spec)
+        nodes = torch.arange(10)
+        TorchBackend().update(cs, cd, spec, nodes)
+        HIPBackend().update(gs, gd, spec, nodes)
+        torch.cuda.synchronize()
+        assert _close(cs.params, gs.params)
+        assert torch.equal(cs.ages.cpu(), gs.ages.cpu())
+
+    def test_partitioned_fast_matches_tick_path(self):
+        X, y = make_synthetic_classification((640, 57, 2), seed=0, margin=2.0)
+        idx = np.random.default_rng(0).permutation(640)
+        shards = [(X[s], y[s]) for s in np.array_split(idx[:576], 64)]
+        data = DataArena.from_shards(
+            shards, CUDA, global_eval=(X[idx[576:]], y[idx[576:]])
+        )
+        cfg = EngineConfig(
+            n_nodes=64, delta=10, protocol=AntiEntropyProtocol.PUSH_PULL,
+            model_size=116, sampling_eval=0.0, seed=3, n_parts=4,
+        )
+        spec = LogRegSpec(d_in=57, n_classes=2, lr=0.1, n_parts=4)
+
+        fast = BatchedGossipSimulator(cfg, spec, data, device=CUDA)
+        assert fast._fast_path_ok()
+        fast.init_nodes()
+        fast.start(n_rounds=4)
+
+        slow = BatchedGossipSimulator(cfg, spec, data, device=CUDA)
+        slow._fast_path_ok = lambda: False
+        slow.init_nodes()
+        slow.start(n_rounds=4)
+        torch.cuda.synchronize()
+        assert torch.allclose(
+            fast.local_params(), slow.local_params(), atol=1e-5, rtol=1e-5
+        )
+        assert torch.equal(fast.state.ages, slow.state.ages)
+
+    def test_partitioned_gpu_learns(self):
+        X, y = make_synthetic_classification((640, 57, 2), seed=0, margin=2.0)
+        idx = np.random.default_rng(0).permutation(640)
+        shards = [(X[s], y[s]) for s in np.array_split(idx[:576], 64)]
+        data = DataArena.from_shards(
+            shards, CUDA, global_eval=(X[idx[576:]], y[idx[576:]])
+        )
+        cfg = EngineConfig(
+            n_nodes=64, delta=10, protocol=AntiEntropyProtocol.PUSH,
+            model_size=116, sampling_eval=0.25, seed=1, n_parts=4,
+        )
+        sim = BatchedGossipSimulator(
+            cfg, LogRegSpec(d_in=57, n_classes=2, lr=0.1, n_parts=4),
+            data, device=CUDA,
+        )
+        rep = SimulationReport()
+        sim.add_receiver(rep)
+        sim.init_nodes()
+        sim.start(n_rounds=15)
+        torch.cuda.synchronize()
+        assert rep.get_evaluation(False)[-1][1]["accuracy"] > 0.9

@@ -58,9 +58,18 @@ def _cfgs():
         peers_indices=indices,
         **base,
     )
+    # partitioned gossip: partition ids ride along with every delivery
+    yield EngineConfig(protocol=AntiEntropyProtocol.PUSH, n_parts=4, **base)
+    yield EngineConfig(
+        protocol=AntiEntropyProtocol.PUSH_PULL,
+        n_parts=7,
+        delay=UniformDelay(0, 6),
+        drop_prob=0.15,
+        **base,
+    )
 
 
-@pytest.mark.parametrize("cfg_i", range(9))
+@pytest.mark.parametrize("cfg_i", range(11))
 def test_native_matches_python(cfg_i):
     cfg = list(_cfgs())[cfg_i]
     py_s = Scheduler(cfg)
@@ -89,6 +98,8 @@ def test_native_matches_python(cfg_i):
                 "rep_recv_ptr",
                 "rep_del_slots",
                 "rep_del_owners",
+                "del_pids",
+                "rep_pids",
             ):
                 va, vb = getattr(pa, f), getattr(pb, f)
                 assert np.array_equal(
