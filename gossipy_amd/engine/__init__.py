@@ -13,11 +13,20 @@ from .arena import DataArena, NodeStateArena, SlotPool
 from .backend import HIPBackend, TorchBackend, make_backend
 from .models import AdaLineSpec, LogRegSpec, MLPSpec, PegasosSpec
 from .rng import Purpose, RandomTape
-from .runner import BatchedGossipSimulator
-from .schedule import EngineConfig, RoundSchedule, Scheduler, TickPhase, make_scheduler
+from .runner import BatchedGossipSimulator, BatchedTokenizedGossipSimulator
+from .schedule import (
+    EngineConfig,
+    RoundSchedule,
+    Scheduler,
+    TickPhase,
+    TokenizedScheduler,
+    make_scheduler,
+)
 
 __all__ = [
     "BatchedGossipSimulator",
+    "BatchedTokenizedGossipSimulator",
+    "TokenizedScheduler",
     "EngineConfig",
     "Scheduler",
     "RoundSchedule",

@@ -37,9 +37,16 @@ from .arena import DataArena, NodeStateArena, SlotPool
 from .backend import make_backend
 from .metrics import binary_margin_metrics, classification_metrics_shared
 from .rng import Purpose, RandomTape
-from .schedule import EngineConfig, RoundSchedule, Scheduler, TickPhase, make_scheduler
+from .schedule import (
+    EngineConfig,
+    RoundSchedule,
+    Scheduler,
+    TickPhase,
+    TokenizedScheduler,
+    make_scheduler,
+)
 
-__all__ = ["BatchedGossipSimulator"]
+__all__ = ["BatchedGossipSimulator", "BatchedTokenizedGossipSimulator"]
 
 
 class BatchedGossipSimulator(SimulationEventSender):
@@ -481,6 +488,13 @@ class BatchedGossipSimulator(SimulationEventSender):
     def local_params(self) -> torch.Tensor:
         return self.state.params
 
+    def gather_ages(self) -> Optional[torch.Tensor]:
+        if self.world == 1:
+            return self.state.ages
+        out = [torch.empty_like(self.state.ages) for _ in range(self.world)]
+        dist.all_gather(out, self.state.ages.contiguous())
+        return torch.cat(out, dim=0)
+
     def gather_params(self) -> Optional[torch.Tensor]:
         """Full ``[n_nodes, D]`` parameter matrix on rank 0 (None elsewhere)."""
         if self.world == 1:
@@ -488,3 +502,40 @@ class BatchedGossipSimulator(SimulationEventSender):
         out = [torch.empty_like(self.state.params) for _ in range(self.world)]
         dist.all_gather(out, self.state.params.contiguous())
         return torch.cat(out, dim=0)
+
+
+class BatchedTokenizedGossipSimulator(BatchedGossipSimulator):
+    """Flow-controlled batched simulator (TokenizedGossipSimulator,
+    gossipy/simul.py:506-689, on the batched engine).
+
+    Token accounts and the utility function live host-side in the
+    :class:`~gossipy_amd.engine.schedule.TokenizedScheduler` (pure python —
+    the schedule's burst waves are data-dependent on the account state, so
+    the C++ scheduler is not used); the per-wave batched kernels are the
+    same ones the plain simulator launches.
+
+    ``utility_fun(receiver, sender, t) -> int`` is evaluated on node ids
+    (the reference's flagship config uses a constant,
+    main_hegedus_2021.py:57); utilities that inspect model state belong on
+    the object layer.
+    """
+
+    def __init__(
+        self,
+        cfg: EngineConfig,
+        spec,
+        data: DataArena,
+        token_account,
+        utility_fun=None,
+        device: Optional[torch.device] = None,
+    ):
+        super().__init__(cfg, spec, data, device=device)
+        self.scheduler = TokenizedScheduler(cfg, token_account, utility_fun)
+
+    def _fast_path_ok(self) -> bool:
+        # burst waves are python-scheduled; kernels still run per wave
+        return False
+
+    @property
+    def accounts(self):
+        return self.scheduler.accounts

@@ -41,6 +41,7 @@ __all__ = [
     "TickPhase",
     "RoundSchedule",
     "Scheduler",
+    "TokenizedScheduler",
     "NativeSchedulerAdapter",
     "make_scheduler",
 ]
@@ -424,14 +425,336 @@ class Scheduler:
                 dly = int(d._timexunit * self.cfg.model_size) + d._overhead
             else:
                 dly = int(d.get(None))
+            # -3 marks a REPLY payload: processed like a plain delivery, but
+            # it never triggers a reactive burst (the reference delivers
+            # replies outside the reacting loop, gossipy/simul.py:623-648)
             self._pending.setdefault(t + dly, []).append(
-                (requester, slot, -1, False, replier, pid)
+                (requester, slot, -3, False, replier, pid)
             )
             self._reply_accounting[t] = (sent, failed, size)
             return True
         failed += 1
         self._reply_accounting[t] = (sent, failed, size)
         return False
+
+
+class TokenizedScheduler(Scheduler):
+    """Token-account flow-controlled schedule (TokenizedGossipSimulator,
+    gossipy/simul.py:506-689 — with the receiver, not a stale loop variable,
+    sending the reactive burst; see gossipy_amd/simul.py for the same fix).
+
+    Differences from the base schedule:
+
+    * a firing node sends only with probability ``proactive()`` (one tape
+      draw per firing node, ascending node order); otherwise banks a token;
+    * a delivered *push* (a message that generates no reply) triggers a
+      reactive burst from its receiver: ``reactive(utility)`` extra sends,
+      with tokens spent, snapshot taken *after* the receiving merge/update
+      (the burst-sender snapshot lands in the next same-tick wave's phase A);
+    * zero-delay burst messages are delivered in the same tick as additional
+      waves (each wave is one more batched deliver launch), matching the
+      reference's grow-while-iterating message loop (gossipy/simul.py:
+      634-648).
+
+    The utility function is host-side: ``utility_fun(receiver, sender, t)``
+    over node ids (the reference's experiments use a constant — e.g.
+    main_hegedus_2021.py:57; model-state-dependent utilities stay on the
+    object layer).
+    """
+
+    def __init__(self, cfg: EngineConfig, token_account, utility_fun=None):
+        super().__init__(cfg)
+        import copy as _copy
+        import inspect as _inspect
+
+        self.accounts = [
+            _copy.deepcopy(token_account) for _ in range(cfg.n_nodes)
+        ]
+        self.utility_fun = utility_fun or (lambda recv, sender, t: 1)
+        self._react_takes_u = (
+            "u" in _inspect.signature(token_account.reactive).parameters
+        )
+
+    def _react(self, node: int, utility: int, u: float) -> int:
+        acct = self.accounts[node]
+        if self._react_takes_u:
+            return acct.reactive(utility, u=u)
+        return acct.reactive(utility)
+
+    def next_round(self, r: int) -> RoundSchedule:
+        cfg = self.cfg
+        proto = cfg.protocol
+        t0, t1 = r * cfg.delta, (r + 1) * cfg.delta
+        sent = failed = total_size = 0
+        ticks: List[TickPhase] = []
+
+        for t in range(t0, t1):
+            firing = self._firing(t)
+            freed: List[int] = []
+
+            # --- proactive-gated sends (gossipy/simul.py:602-615)
+            snap_nodes: List[int] = []
+            snap_slots: List[int] = []
+            n_f = len(firing)
+            if n_f:
+                pro_u = self.tape.uniform(Purpose.TOKEN, t, n_f)
+                go = [
+                    pro_u[j] < self.accounts[int(firing[j])].proactive()
+                    for j in range(n_f)
+                ]
+                senders = firing[np.asarray(go, dtype=bool)]
+                for j in range(n_f):
+                    if not go[j]:
+                        self.accounts[int(firing[j])].add(1)
+                n_s = len(senders)
+                if n_s:
+                    peers = self._peers_of(senders, t)
+                    drop_u = self.tape.uniform(Purpose.DROP, t, n_s)
+                    sizes = np.full(
+                        n_s,
+                        cfg.model_size if proto != AntiEntropyProtocol.PULL else 1,
+                    )
+                    delays = self._delays(t, n_s, sizes)
+                    if cfg.n_parts > 0:
+                        pids = np.atleast_1d(
+                            self.tape.stream(Purpose.PART, t).integers(
+                                0, cfg.n_parts, size=n_s
+                            )
+                        )
+                    else:
+                        pids = np.full(n_s, -1, dtype=np.int64)
+                    for j in range(n_s):
+                        sender, receiver = int(senders[j]), int(peers[j])
+                        is_pull = proto == AntiEntropyProtocol.PULL
+                        slot = -1
+                        if not is_pull:
+                            slot = self._alloc_slot(sender)
+                            snap_nodes.append(sender)
+                            snap_slots.append(slot)
+                        sent += 1
+                        total_size += int(sizes[j])
+                        if drop_u[j] >= cfg.drop_prob:
+                            due = t + int(delays[j])
+                            wants = proto == AntiEntropyProtocol.PUSH_PULL
+                            self._pending.setdefault(due, []).append(
+                                (receiver, slot, -2 if wants else -1, is_pull,
+                                 sender, int(pids[j]))
+                            )
+                        else:
+                            failed += 1
+                            if slot >= 0:
+                                freed.append(slot)
+
+            online_u = self.tape.uniform(Purpose.ONLINE, t, cfg.n_nodes)
+            online = online_u <= cfg.online_prob
+
+            # --- delivery waves: wave 0 = scheduled messages due at t;
+            # each wave's reactive bursts with zero delay feed the next wave
+            wave_due = self._pending.pop(t, [])
+            #: per-(receiver) sequential tape streams for this tick's bursts
+            burst_streams: Dict[Tuple[int, int], object] = {}
+
+            def tick_stream(purpose, node):
+                key = (int(purpose), node)
+                s = burst_streams.get(key)
+                if s is None:
+                    s = self.tape.stream(purpose, t, extra=1 + node)
+                    burst_streams[key] = s
+                return s
+
+            while wave_due or snap_nodes:
+                recv_map: Dict[int, List[Tuple[int, int, int, int]]] = {}
+                pull_nodes: List[int] = []
+                pull_slots: List[int] = []
+                next_due: List[Tuple[int, int, int, bool, int, int]] = []
+                burst_snap_nodes: List[int] = []
+                burst_snap_slots: List[int] = []
+
+                for receiver, slot, reply_flag, is_pull, sender, pid in wave_due:
+                    if not online[receiver]:
+                        failed += 1
+                        if slot >= 0:
+                            freed.append(slot)
+                        continue
+                    if is_pull:
+                        rslot = self._alloc_slot(receiver)
+                        pull_nodes.append(receiver)
+                        pull_slots.append(rslot)
+                        rpid = self._reply_pid(t, receiver)
+                        if not self._enqueue_reply(t, receiver, sender, rslot, rpid):
+                            freed.append(rslot)
+                        continue
+                    rslot = -1
+                    if reply_flag == -2:
+                        rslot = self._alloc_slot(receiver)
+                        rpid = self._reply_pid(t, receiver)
+                        if not self._enqueue_reply(t, receiver, sender, rslot, rpid):
+                            freed.append(rslot)
+                    recv_map.setdefault(receiver, []).append((slot, rslot, pid, sender))
+                    freed.append(slot)
+                    # reactive burst: only deliveries that generate no reply
+                    # (gossipy/simul.py:631-648); REPLY deliveries are marked
+                    # -3 by _enqueue_reply and excluded
+                    if reply_flag == -1:
+                        utility = int(self.utility_fun(receiver, sender, t))
+                        ru = float(tick_stream(Purpose.TOKEN, receiver).random())
+                        reaction = self._react(receiver, utility, ru)
+                        if reaction <= 0:
+                            continue
+                        self.accounts[receiver].sub(reaction)
+                        gp = tick_stream(Purpose.PEER, receiver)
+                        gd = tick_stream(Purpose.DROP, receiver)
+                        gdl = tick_stream(Purpose.DELAY, receiver)
+                        for _ in range(reaction):
+                            peer = self._burst_peer(receiver, gp)
+                            bslot = self._alloc_slot(receiver)
+                            burst_snap_nodes.append(receiver)
+                            burst_snap_slots.append(bslot)
+                            bpid = (
+                                int(
+                                    tick_stream(Purpose.PART, receiver).integers(
+                                        0, cfg.n_parts
+                                    )
+                                )
+                                if cfg.n_parts > 0
+                                else -1
+                            )
+                            sent += 1
+                            total_size += cfg.model_size
+                            if float(gd.random()) >= cfg.drop_prob:
+                                dly = self._burst_delay(gdl)
+                                wants = proto == AntiEntropyProtocol.PUSH_PULL
+                                m = (peer, bslot, -2 if wants else -1, False,
+                                     receiver, bpid)
+                                if dly == 0:
+                                    next_due.append(m)
+                                else:
+                                    self._pending.setdefault(t + dly, []).append(m)
+                            else:
+                                failed += 1
+                                freed.append(bslot)
+
+                # emit this wave's phase
+                recv_nodes = np.fromiter(
+                    recv_map.keys(), dtype=np.int32, count=len(recv_map)
+                )
+                recv_ptr = np.zeros(len(recv_map) + 1, dtype=np.int32)
+                del_slots: List[int] = []
+                del_owners: List[int] = []
+                reply_slots: List[int] = []
+                del_pids: List[int] = []
+                for i, rn in enumerate(recv_nodes):
+                    quads = recv_map[int(rn)]
+                    del_slots.extend(q[0] for q in quads)
+                    reply_slots.extend(q[1] for q in quads)
+                    del_pids.extend(q[2] for q in quads)
+                    del_owners.extend(q[3] for q in quads)
+                    recv_ptr[i + 1] = recv_ptr[i] + len(quads)
+                phase = TickPhase(
+                    t=t,
+                    snap_nodes=np.asarray(snap_nodes, dtype=np.int32),
+                    snap_slots=np.asarray(snap_slots, dtype=np.int32),
+                    recv_nodes=recv_nodes,
+                    recv_ptr=recv_ptr,
+                    del_slots=np.asarray(del_slots, dtype=np.int32),
+                    del_owners=np.asarray(del_owners, dtype=np.int32),
+                    reply_slots=np.asarray(reply_slots, dtype=np.int32),
+                    pull_snap_nodes=np.asarray(pull_nodes, dtype=np.int32),
+                    pull_snap_slots=np.asarray(pull_slots, dtype=np.int32),
+                    del_pids=np.asarray(del_pids, dtype=np.int32),
+                    rep_pids=np.zeros(0, dtype=np.int32),
+                )
+                if phase.n_events:
+                    ticks.append(phase)
+                # next wave: burst snapshots become phase-A of the next wave
+                snap_nodes, snap_slots = burst_snap_nodes, burst_snap_slots
+                wave_due = next_due
+
+            # --- same-tick replies (sub-phase C)
+            rep_due = self._pending.pop(t, [])
+            if rep_due:
+                rep_map: Dict[int, List[Tuple[int, int, int]]] = {}
+                for receiver, slot, _rf, _ip, sender, pid in rep_due:
+                    if not online[receiver]:
+                        failed += 1
+                        freed.append(slot)
+                        continue
+                    rep_map.setdefault(receiver, []).append((slot, sender, pid))
+                    freed.append(slot)
+                rep_recv = np.fromiter(
+                    rep_map.keys(), dtype=np.int32, count=len(rep_map)
+                )
+                rep_ptr = np.zeros(len(rep_map) + 1, dtype=np.int32)
+                rep_slots: List[int] = []
+                rep_owners: List[int] = []
+                rep_pids: List[int] = []
+                for i, rn in enumerate(rep_recv):
+                    rep_slots.extend(p[0] for p in rep_map[int(rn)])
+                    rep_owners.extend(p[1] for p in rep_map[int(rn)])
+                    rep_pids.extend(p[2] for p in rep_map[int(rn)])
+                    rep_ptr[i + 1] = len(rep_slots)
+                ticks.append(
+                    TickPhase(
+                        t=t,
+                        snap_nodes=np.zeros(0, dtype=np.int32),
+                        snap_slots=np.zeros(0, dtype=np.int32),
+                        recv_nodes=np.zeros(0, dtype=np.int32),
+                        recv_ptr=np.zeros(1, dtype=np.int32),
+                        del_slots=np.zeros(0, dtype=np.int32),
+                        del_owners=np.zeros(0, dtype=np.int32),
+                        reply_slots=np.zeros(0, dtype=np.int32),
+                        pull_snap_nodes=np.zeros(0, dtype=np.int32),
+                        pull_snap_slots=np.zeros(0, dtype=np.int32),
+                        rep_recv_nodes=rep_recv,
+                        rep_recv_ptr=rep_ptr,
+                        rep_del_slots=np.asarray(rep_slots, dtype=np.int32),
+                        rep_del_owners=np.asarray(rep_owners, dtype=np.int32),
+                        rep_pids=np.asarray(rep_pids, dtype=np.int32),
+                        del_pids=np.zeros(0, dtype=np.int32),
+                    )
+                )
+
+            sent_r, failed_r, size_r = self._reply_accounting.pop(t, (0, 0, 0))
+            sent += sent_r
+            failed += failed_r
+            total_size += size_r
+            self._free_slots.extend(freed)
+
+        eval_nodes = None
+        if cfg.sampling_eval > 0:
+            g = self.tape.stream(Purpose.EVAL, t1 - 1)
+            k = max(int(cfg.n_nodes * cfg.sampling_eval), 1)
+            eval_nodes = np.atleast_1d(g.integers(0, cfg.n_nodes, size=k))
+
+        return RoundSchedule(
+            round_idx=r,
+            ticks=ticks,
+            n_slots=self._next_slot,
+            slot_owner=self.slot_owner[: self._next_slot].copy(),
+            sent_messages=sent,
+            failed_messages=failed,
+            total_size=total_size,
+            eval_nodes=eval_nodes,
+        )
+
+    def _burst_peer(self, node: int, gp) -> int:
+        cfg = self.cfg
+        if cfg.peers_indptr is None:
+            draw = int(gp.integers(0, cfg.n_nodes - 1))
+            return draw + (1 if draw >= node else 0)
+        s = int(cfg.peers_indptr[node])
+        deg = int(cfg.peers_indptr[node + 1]) - s
+        return int(cfg.peers_indices[s + int(np.floor(float(gp.random()) * deg))])
+
+    def _burst_delay(self, gdl) -> int:
+        d = self.cfg.delay
+        if isinstance(d, ConstantDelay):
+            return d._delay
+        if isinstance(d, UniformDelay):
+            return int(gdl.integers(d._min_delay, d._max_delay + 1))
+        if isinstance(d, LinearDelay):
+            return int(d._timexunit * self.cfg.model_size) + d._overhead
+        return int(d.get(None))
 
 
 class NativeSchedulerAdapter:

@@ -122,8 +122,14 @@ class RandomizedTokenAccount(GeneralizedTokenAccount):
             )
         return 1
 
-    def reactive(self, utility: int) -> int:
+    def reactive(self, utility: int, u: float = None) -> int:
+        """``u``: optional uniform in [0,1) for the randomized rounding —
+        the batched engine passes a tape draw so reactions are deterministic
+        and residency-invariant; ``None`` falls back to numpy's global RNG
+        (object-layer behavior, gossipy/flow_control.py:232-236)."""
         if utility > 0:
             r = self.n_tokens / self.reactivity
-            return int(r) + binomial(1, r - int(r))  # randomized rounding
+            frac = r - int(r)
+            rounded = (u < frac) if u is not None else binomial(1, frac)
+            return int(r) + int(rounded)  # randomized rounding
         return 0
