@@ -13,7 +13,11 @@ from .arena import DataArena, NodeStateArena, SlotPool
 from .backend import HIPBackend, TorchBackend, make_backend
 from .models import AdaLineSpec, LogRegSpec, MLPSpec, PegasosSpec
 from .rng import Purpose, RandomTape
-from .runner import BatchedGossipSimulator, BatchedTokenizedGossipSimulator
+from .runner import (
+    BatchedAll2AllGossipSimulator,
+    BatchedGossipSimulator,
+    BatchedTokenizedGossipSimulator,
+)
 from .schedule import (
     EngineConfig,
     RoundSchedule,
@@ -26,6 +30,7 @@ from .schedule import (
 __all__ = [
     "BatchedGossipSimulator",
     "BatchedTokenizedGossipSimulator",
+    "BatchedAll2AllGossipSimulator",
     "TokenizedScheduler",
     "EngineConfig",
     "Scheduler",

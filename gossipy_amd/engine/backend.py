@@ -420,6 +420,42 @@ class TorchBackend:
         state.params[node] = saved_p
         state.ages[node] = saved_a
 
+    # -- all2all weighted merge ----------------------------------------------
+
+    def deliver_weighted(
+        self,
+        state: NodeStateArena,
+        pool: SlotPool,
+        data: DataArena,
+        spec,
+        wm_nodes: torch.Tensor,
+        wm_ptr: torch.Tensor,
+        wm_slots: torch.Tensor,
+        wm_weights: torch.Tensor,
+        wm_self_w: torch.Tensor,
+    ) -> None:
+        """Koloskova-style weighted k-way merge then local update
+        (WeightedTMH MERGE_UPDATE, gossipy/model/handler.py:652-654 +
+        666-688): ``theta_i = w0*theta_i + sum_j w_j*theta_recv_j``; age =
+        max over all merged models."""
+        assert spec.mode == CreateModelMode.MERGE_UPDATE, (
+            "all2all engine supports MERGE_UPDATE (the reference's"
+            " main_all2all configuration)"
+        )
+        if len(wm_nodes) == 0:
+            return
+        ptr = wm_ptr.tolist()
+        for i, node in enumerate(wm_nodes.tolist()):
+            acc = state.params[node] * float(wm_self_w[i])
+            age = int(state.ages[node])
+            for j in range(ptr[i], ptr[i + 1]):
+                s = int(wm_slots[j])
+                acc += float(wm_weights[j]) * pool.slots[s]
+                age = max(age, int(pool.slot_ages[s]))
+            state.params[node] = acc
+            state.ages[node] = age
+        self.update(state, data, spec, wm_nodes)
+
     # -- evaluation ----------------------------------------------------------
 
     def scores(
@@ -513,6 +549,28 @@ class HIPBackend(TorchBackend):
             pids,
             update_only=False,
         )
+
+    def deliver_weighted(
+        self, state, pool, data, spec, wm_nodes, wm_ptr, wm_slots, wm_weights,
+        wm_self_w,
+    ) -> None:
+        assert spec.mode == CreateModelMode.MERGE_UPDATE
+        if len(wm_nodes) == 0:
+            return
+        dev = state.params.device
+        nodes = wm_nodes.to(dev, torch.int32)
+        self.ext.wmerge(
+            state.params,
+            state.ages,
+            pool.slots,
+            pool.slot_ages,
+            nodes,
+            wm_ptr.to(dev, torch.int32),
+            wm_slots.to(dev, torch.int32),
+            wm_weights.to(dev, torch.float32),
+            wm_self_w.to(dev, torch.float32),
+        )
+        self.update(state, data, spec, nodes)
 
     def _part_dev(self, spec, dev):
         """(perm, ptr, arena_part) as int32 device tensors (cached)."""

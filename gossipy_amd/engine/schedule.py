@@ -102,15 +102,27 @@ class TickPhase:
     # rep_del_slots; length 0 when cfg.n_parts == 0)
     del_pids: np.ndarray = None  # int32
     rep_pids: np.ndarray = None  # int32
+    # all2all weighted merge (Koloskova 2020): nodes merging their
+    # accumulated neighbor models *before* this tick's snapshots
+    # (gossipy/node.py:833-843). CSR over wm_nodes; wm_self_w = the node's
+    # own mixing weight (weights[0]).
+    wm_nodes: np.ndarray = None  # int32
+    wm_ptr: np.ndarray = None  # int32
+    wm_slots: np.ndarray = None  # int32
+    wm_weights: np.ndarray = None  # float32 (aligned with wm_slots)
+    wm_owners: np.ndarray = None  # int32 (slot writers, for the comm plan)
+    wm_self_w: np.ndarray = None  # float32 [len(wm_nodes)]
 
     @property
     def n_events(self) -> int:
         n_rep = 0 if self.rep_del_slots is None else len(self.rep_del_slots)
+        n_wm = 0 if self.wm_nodes is None else len(self.wm_nodes)
         return (
             len(self.snap_nodes)
             + len(self.del_slots)
             + len(self.pull_snap_nodes)
             + n_rep
+            + n_wm
         )
 
 
@@ -747,6 +759,203 @@ class TokenizedScheduler(Scheduler):
         return int(cfg.peers_indices[s + int(np.floor(float(gp.random()) * deg))])
 
     def _burst_delay(self, gdl) -> int:
+        d = self.cfg.delay
+        if isinstance(d, ConstantDelay):
+            return d._delay
+        if isinstance(d, UniformDelay):
+            return int(gdl.integers(d._min_delay, d._max_delay + 1))
+        if isinstance(d, LinearDelay):
+            return int(d._timexunit * self.cfg.model_size) + d._overhead
+        return int(d.get(None))
+
+
+class All2AllScheduler(Scheduler):
+    """Schedule for decentralized weighted averaging (Koloskova 2020;
+    All2AllGossipSimulator, gossipy/simul.py:720-852).
+
+    Per tick: every timed-out node first merges the neighbor models it has
+    accumulated since its last timeout — a *weighted* k-way merge with
+    mixing weights, followed by a local update — and then pushes its
+    post-merge model to **all** its peers (gossipy/node.py:833-846). A
+    receiver only stores the incoming model (replacing any previous one
+    from the same sender, gossipy/node.py:860-869); no learning happens at
+    delivery time.
+
+    The mixing weights replicate the reference contract: ``weights[0]``
+    scales the node's own model and ``weights[1+j]`` the j-th accumulated
+    model *in arrival order* (gossipy/model/handler.py:666-688 applies the
+    vector positionally). With fewer arrivals than peers the unused tail is
+    dropped — so partial participation shrinks the average exactly as the
+    reference does.
+
+    One snapshot slot serves all of a node's peers (the reference caches
+    once under one key and pushes the same CacheKey to every peer,
+    gossipy/node.py:845-846 + handler.py:160-176), so slots are
+    ref-counted: freed when the last referencing delivery is consumed,
+    dropped, or replaced.
+    """
+
+    def __init__(self, cfg: EngineConfig, mixing=None):
+        super().__init__(cfg)
+        n = cfg.n_nodes
+        if cfg.peers_indptr is None:
+            self._degs = np.full(n, n - 1, dtype=np.int64)
+        else:
+            self._degs = (cfg.peers_indptr[1:] - cfg.peers_indptr[:-1]).astype(
+                np.int64
+            )
+        if mixing is None:
+            # UniformMixing (gossipy/core.py:419-434)
+            self._W = [
+                np.ones(self._degs[i] + 1) / (self._degs[i] + 1) for i in range(n)
+            ]
+        elif callable(getattr(mixing, "get", None)):
+            self._W = [np.asarray(mixing.get(i), dtype=np.float64) for i in range(n)]
+        else:
+            self._W = [np.asarray(mixing(i), dtype=np.float64) for i in range(n)]
+        #: per-node accumulated (sender, slot) in arrival order
+        self._acc: List[List[Tuple[int, int]]] = [[] for _ in range(n)]
+        #: outstanding references per slot
+        self._refs: Dict[int, int] = {}
+
+    def _deref(self, slot: int, freed: List[int]) -> None:
+        r = self._refs.get(slot, 0) - 1
+        if r <= 0:
+            self._refs.pop(slot, None)
+            freed.append(slot)
+        else:
+            self._refs[slot] = r
+
+    def _all_peers(self, node: int) -> np.ndarray:
+        cfg = self.cfg
+        if cfg.peers_indptr is None:
+            return np.concatenate(
+                [np.arange(node), np.arange(node + 1, cfg.n_nodes)]
+            )
+        s = cfg.peers_indptr[node]
+        return cfg.peers_indices[s : s + self._degs[node]]
+
+    def next_round(self, r: int) -> RoundSchedule:
+        cfg = self.cfg
+        t0, t1 = r * cfg.delta, (r + 1) * cfg.delta
+        sent = failed = total_size = 0
+        ticks: List[TickPhase] = []
+
+        for t in range(t0, t1):
+            freed: List[int] = []
+            firing = self._firing(t)
+
+            # --- weighted merges of accumulated models (before sends)
+            wm_nodes: List[int] = []
+            wm_ptr = [0]
+            wm_slots: List[int] = []
+            wm_weights: List[float] = []
+            wm_owners: List[int] = []
+            wm_self_w: List[float] = []
+            for i in firing:
+                acc = self._acc[int(i)]
+                if not acc:
+                    continue
+                W = self._W[int(i)]
+                wm_nodes.append(int(i))
+                wm_self_w.append(float(W[0]))
+                for j, (sender, slot) in enumerate(acc):
+                    wm_slots.append(slot)
+                    # positional weight; tail beyond the vector repeats the
+                    # last entry defensively (cannot happen on static nets)
+                    wm_weights.append(float(W[min(1 + j, len(W) - 1)]))
+                    wm_owners.append(sender)
+                    self._deref(slot, freed)
+                wm_ptr.append(len(wm_slots))
+                self._acc[int(i)] = []
+
+            # --- broadcast sends: one snapshot slot, one message per peer
+            snap_nodes: List[int] = []
+            snap_slots: List[int] = []
+            n_f = len(firing)
+            if n_f:
+                drop_stream = self.tape.stream(Purpose.DROP, t)
+                delay_stream = self.tape.stream(Purpose.DELAY, t)
+                for i in firing:
+                    node = int(i)
+                    peers = self._all_peers(node)
+                    slot = self._alloc_slot(node)
+                    snap_nodes.append(node)
+                    snap_slots.append(slot)
+                    refs = 0
+                    for peer in peers:
+                        sent += 1
+                        total_size += cfg.model_size
+                        if float(drop_stream.random()) >= cfg.drop_prob:
+                            dly = self._wm_delay(delay_stream)
+                            self._pending.setdefault(t + dly, []).append(
+                                (int(peer), slot, -1, False, node, -1)
+                            )
+                            refs += 1
+                        else:
+                            failed += 1
+                    if refs == 0:
+                        freed.append(slot)
+                    else:
+                        self._refs[slot] = refs
+
+            # --- deliveries: store into the receiver's accumulator
+            online_u = self.tape.uniform(Purpose.ONLINE, t, cfg.n_nodes)
+            online = online_u <= cfg.online_prob
+            for receiver, slot, _rf, _ip, sender, _pid in self._pending.pop(t, []):
+                if not online[receiver]:
+                    failed += 1
+                    self._deref(slot, freed)
+                    continue
+                acc = self._acc[receiver]
+                for j, (s0, slot0) in enumerate(acc):
+                    if s0 == sender:  # replace stale model from this sender
+                        self._deref(slot0, freed)
+                        acc[j] = (sender, slot)
+                        break
+                else:
+                    acc.append((sender, slot))
+
+            phase = TickPhase(
+                t=t,
+                snap_nodes=np.asarray(snap_nodes, dtype=np.int32),
+                snap_slots=np.asarray(snap_slots, dtype=np.int32),
+                recv_nodes=np.zeros(0, dtype=np.int32),
+                recv_ptr=np.zeros(1, dtype=np.int32),
+                del_slots=np.zeros(0, dtype=np.int32),
+                del_owners=np.zeros(0, dtype=np.int32),
+                reply_slots=np.zeros(0, dtype=np.int32),
+                pull_snap_nodes=np.zeros(0, dtype=np.int32),
+                pull_snap_slots=np.zeros(0, dtype=np.int32),
+                wm_nodes=np.asarray(wm_nodes, dtype=np.int32),
+                wm_ptr=np.asarray(wm_ptr, dtype=np.int32),
+                wm_slots=np.asarray(wm_slots, dtype=np.int32),
+                wm_weights=np.asarray(wm_weights, dtype=np.float32),
+                wm_owners=np.asarray(wm_owners, dtype=np.int32),
+                wm_self_w=np.asarray(wm_self_w, dtype=np.float32),
+            )
+            if phase.n_events:
+                ticks.append(phase)
+            self._free_slots.extend(freed)
+
+        eval_nodes = None
+        if cfg.sampling_eval > 0:
+            g = self.tape.stream(Purpose.EVAL, t1 - 1)
+            k = max(int(cfg.n_nodes * cfg.sampling_eval), 1)
+            eval_nodes = np.atleast_1d(g.integers(0, cfg.n_nodes, size=k))
+
+        return RoundSchedule(
+            round_idx=r,
+            ticks=ticks,
+            n_slots=self._next_slot,
+            slot_owner=self.slot_owner[: self._next_slot].copy(),
+            sent_messages=sent,
+            failed_messages=failed,
+            total_size=total_size,
+            eval_nodes=eval_nodes,
+        )
+
+    def _wm_delay(self, gdl) -> int:
         d = self.cfg.delay
         if isinstance(d, ConstantDelay):
             return d._delay

@@ -221,6 +221,43 @@ tick_logreg_kernel(LogregArgs a)
 }
 
 // ---------------------------------------------------------------------------
+// all2all weighted k-way merge (K5 weighted variant; WeightedTMH._merge,
+// gossipy/model/handler.py:666-688): params[n] = w0*params[n] + sum_j
+// w_j*slots[s_j]; age = max over merged. Family-agnostic (elementwise over
+// D); the per-family update-only kernel runs right after it.
+// ---------------------------------------------------------------------------
+
+__global__ void wmerge_kernel(
+    float* __restrict__ params,
+    int* __restrict__ ages,
+    const float* __restrict__ slots,
+    const int* __restrict__ slot_ages,
+    const int* __restrict__ nodes,
+    const int* __restrict__ ptr,
+    const int* __restrict__ wslots,
+    const float* __restrict__ wweights,
+    const float* __restrict__ selfw,
+    int D)
+{
+    int i = blockIdx.x;
+    int node = nodes[i];
+    int tid = threadIdx.x;
+    int lo = ptr[i], hi = ptr[i + 1];
+    float w0 = selfw[i];
+    for (int e = tid; e < D; e += blockDim.x) {
+        float acc = w0 * params[(long)node * D + e];
+        for (int j = lo; j < hi; ++j)
+            acc += wweights[j] * slots[(long)wslots[j] * D + e];
+        params[(long)node * D + e] = acc;
+    }
+    if (tid == 0) {
+        int age = ages[node];
+        for (int j = lo; j < hi; ++j) age = max(age, slot_ages[wslots[j]]);
+        ages[node] = age;
+    }
+}
+
+// ---------------------------------------------------------------------------
 // partitioned logreg tick (K7/K8, PartitionedTMH semantics)
 //
 // Parity: gossipy/model/handler.py:455-525 + gossipy/model/sampling.py:
@@ -745,6 +782,23 @@ void snapshot(torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
         nodes.data_ptr<int>(), slot_ids.data_ptr<int>(), n, D, (int)age_width);
 }
 
+void wmerge(torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
+            torch::Tensor slot_ages, torch::Tensor nodes, torch::Tensor ptr,
+            torch::Tensor wslots, torch::Tensor wweights, torch::Tensor selfw)
+{
+    CHECK_DEV(params); CHECK_DEV(slots); CHECK_DEV(nodes); CHECK_DEV(ptr);
+    int n = nodes.size(0);
+    if (n == 0) return;
+    int D = params.size(1);
+    hipLaunchKernelGGL(wmerge_kernel, dim3(n), dim3(128), 0, current_stream(),
+        params.data_ptr<float>(), ages.data_ptr<int>(),
+        slots.data_ptr<float>(), slot_ages.data_ptr<int>(),
+        nodes.data_ptr<int>(), ptr.data_ptr<int>(),
+        wslots.numel() ? wslots.data_ptr<int>() : nullptr,
+        wweights.numel() ? wweights.data_ptr<float>() : nullptr,
+        selfw.data_ptr<float>(), D);
+}
+
 void tick_logreg_part(
     torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
     torch::Tensor slot_ages, torch::Tensor nodes, torch::Tensor recv_ptr,
@@ -1146,4 +1200,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m)
           "fused partition-merge + age-rescaled logreg SGD tick (K7/K8)");
     m.def("run_round_logreg_part", &run_round_logreg_part,
           "whole-round executor, partitioned logreg family");
+    m.def("wmerge", &wmerge, "all2all weighted k-way merge (K5 weighted)");
 }

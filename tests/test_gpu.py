@@ -392,3 +392,74 @@ class TestPartitionedGPU:
         sim.start(n_rounds=15)
         torch.cuda.synchronize()
         assert rep.get_evaluation(False)[-1][1]["accuracy"] > 0.9
+
+
+class TestAll2AllGPU:
+    def test_wmerge_matches_oracle(self):
+        spec = LogRegSpec(d_in=57, n_classes=2, lr=0.1)
+        cs, gs, cd, gd = _pair(10, spec, 57)
+        cpool, gpool = SlotPool(spec.D, CPU, 8), SlotPool(spec.D, CUDA, 8)
+        cpool.slots.normal_(generator=torch.Generator().manual_seed(6))
+        cpool.slot_ages.copy_(torch.arange(8, dtype=torch.int32) * 3)
+        gpool.slots.copy_(cpool.slots)
+        gpool.slot_ages.copy_(cpool.slot_ages)
+        nodes = torch.tensor([1, 4, 7], dtype=torch.int64)
+        ptr = torch.tensor([0, 2, 5, 6], dtype=torch.int64)
+        slots = torch.tensor([0, 3, 1, 2, 5, 7], dtype=torch.int64)
+        w = torch.tensor([0.3, 0.2, 0.25, 0.25, 0.25, 0.5], dtype=torch.float32)
+        sw = torch.tensor([0.5, 0.25, 0.5], dtype=torch.float32)
+        TorchBackend().deliver_weighted(cs, cpool, cd, spec, nodes, ptr, slots, w, sw)
+        HIPBackend().deliver_weighted(gs, gpool, gd, spec, nodes, ptr, slots, w, sw)
+        torch.cuda.synchronize()
+        assert _close(cs.params, gs.params, 1e-4)
+        assert torch.equal(cs.ages.cpu(), gs.ages.cpu())
+
+    def test_all2all_gpu_learns(self):
+        from gossipy_amd.engine import BatchedAll2AllGossipSimulator
+
+        X, y = make_synthetic_classification((640, 57, 2), seed=0, margin=2.0)
+        idx = np.random.default_rng(0).permutation(640)
+        shards = [(X[s], y[s]) for s in np.array_split(idx[:576], 32)]
+        data = DataArena.from_shards(
+            shards, CUDA, global_eval=(X[idx[576:]], y[idx[576:]])
+        )
+        cfg = EngineConfig(
+            n_nodes=32, delta=10, protocol=AntiEntropyProtocol.PUSH,
+            model_size=116, sampling_eval=0.25, seed=13,
+        )
+        sim = BatchedAll2AllGossipSimulator(
+            cfg, LogRegSpec(d_in=57, n_classes=2, lr=0.1), data, device=CUDA
+        )
+        rep = SimulationReport()
+        sim.add_receiver(rep)
+        sim.init_nodes()
+        sim.start(n_rounds=8)
+        torch.cuda.synchronize()
+        assert rep.get_evaluation(False)[-1][1]["accuracy"] > 0.9
+
+
+class TestTokenizedGPU:
+    def test_tokenized_gpu_learns(self):
+        from gossipy_amd.engine import BatchedTokenizedGossipSimulator
+        from gossipy_amd.flow_control import RandomizedTokenAccount
+
+        X, y = make_synthetic_classification((640, 57, 2), seed=0, margin=2.0)
+        idx = np.random.default_rng(0).permutation(640)
+        shards = [(X[s], y[s]) for s in np.array_split(idx[:576], 64)]
+        data = DataArena.from_shards(
+            shards, CUDA, global_eval=(X[idx[576:]], y[idx[576:]])
+        )
+        cfg = EngineConfig(
+            n_nodes=64, delta=10, protocol=AntiEntropyProtocol.PUSH,
+            model_size=116, sampling_eval=0.25, seed=11, n_parts=4,
+        )
+        sim = BatchedTokenizedGossipSimulator(
+            cfg, LogRegSpec(d_in=57, n_classes=2, lr=0.1, n_parts=4), data,
+            token_account=RandomizedTokenAccount(C=20, A=10), device=CUDA,
+        )
+        rep = SimulationReport()
+        sim.add_receiver(rep)
+        sim.init_nodes()
+        sim.start(n_rounds=12)
+        torch.cuda.synchronize()
+        assert rep.get_evaluation(False)[-1][1]["accuracy"] > 0.85
