@@ -499,6 +499,85 @@ class BatchedGossipSimulator(SimulationEventSender):
         dist.all_gather(out, self.state.ages.contiguous())
         return torch.cat(out, dim=0)
 
+    # -- checkpointing --------------------------------------------------------
+
+    def save(self, filename: str) -> None:
+        """Engine-native checkpoint (per rank: ``filename`` should embed the
+        rank for multi-GPU runs).
+
+        The object layer keeps the reference's two-slot dill format
+        (gossipy/simul.py:460-494); the engine materializes its arenas
+        host-side instead of pickling Python object graphs. Scheduler
+        carry-over state (in-flight messages, slot free list, token
+        accounts) is NOT stored: the schedule is a pure function of
+        ``(seed, config, round)``, so :meth:`load` replays rounds
+        ``0..rounds_done`` on a fresh scheduler to restore it exactly —
+        this is what makes the checkpoint valid for any scheduler
+        implementation (python or native C++).
+        """
+        import dill
+
+        blob = {
+            "cfg": self.cfg,
+            "spec": self.spec,
+            "rounds_done": self.rounds_done,
+            "params": self.state.params.cpu(),
+            "ages": self.state.ages.cpu(),
+            "pool_slots": self.pool.slots.cpu(),
+            "pool_ages": self.pool.slot_ages.cpu(),
+            "data": {
+                k: (getattr(self.data, k).cpu() if getattr(self.data, k) is not None else None)
+                for k in ("x", "y", "counts", "tx", "ty", "tcounts", "gx", "gy")
+            },
+            "kind": type(self).__name__,
+            "extra": self._checkpoint_extra(),
+        }
+        with open(filename, "wb") as f:
+            dill.dump(blob, f)
+
+    def _checkpoint_extra(self) -> dict:
+        return {}
+
+    def _restore_extra(self, extra: dict) -> None:
+        pass
+
+    @classmethod
+    def load(cls, filename: str, device: Optional[torch.device] = None, **kw):
+        """Restore a checkpoint written by :meth:`save` (same world size)."""
+        import dill
+
+        with open(filename, "rb") as f:
+            blob = dill.load(f)
+        if device is None:
+            device = torch.device("cuda:0" if torch.cuda.is_available() else "cpu")
+        d = blob["data"]
+        data = DataArena(
+            *[d[k].to(device) if d[k] is not None else None for k in ("x", "y", "counts")],
+            *[d[k].to(device) if d[k] is not None else None for k in ("tx", "ty", "tcounts")],
+            *[d[k].to(device) if d[k] is not None else None for k in ("gx", "gy")],
+        )
+        sim = cls(blob["cfg"], blob["spec"], data, device=device, **kw)
+        sim.state.params.copy_(blob["params"].to(device))
+        sim.state.ages.copy_(blob["ages"].to(device))
+        sim.pool.ensure(blob["pool_slots"].shape[0])
+        sim.pool.slots[: blob["pool_slots"].shape[0]].copy_(
+            blob["pool_slots"].to(device)
+        )
+        sim.pool.slot_ages[: blob["pool_ages"].shape[0]].copy_(
+            blob["pool_ages"].to(device)
+        )
+        # replay the deterministic schedule to restore carry-over state
+        # (in-flight messages, slot allocator, token accounts)
+        for r in range(blob["rounds_done"]):
+            if hasattr(sim.scheduler, "next_round_flat"):
+                sim.scheduler.next_round_flat(r)
+            else:
+                sim.scheduler.next_round(r)
+        sim.rounds_done = blob["rounds_done"]
+        sim._restore_extra(blob["extra"])
+        sim.initialized = True
+        return sim
+
     def gather_params(self) -> Optional[torch.Tensor]:
         """Full ``[n_nodes, D]`` parameter matrix on rank 0 (None elsewhere)."""
         if self.world == 1:
