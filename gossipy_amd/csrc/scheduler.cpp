@@ -103,12 +103,12 @@ public:
                     double timexunit, int64_t overhead, bool sync,
                     double sampling_eval, uint64_t seed,
                     py::object peers_indptr, py::object peers_indices,
-                    int64_t n_parts = 0)
+                    int64_t n_parts = 0, bool sampled = false)
         : n_(n_nodes), delta_(delta), proto_(proto), model_size_(model_size),
           drop_(drop_prob), online_(online_prob), dkind_(delay_kind),
           dmin_(dmin), dmax_(dmax), timexunit_(timexunit), overhead_(overhead),
           sync_(sync), sampling_eval_(sampling_eval), seed_(seed),
-          n_parts_(n_parts)
+          n_parts_(n_parts), sampled_(sampled)
     {
         Stream g(stream_key(seed_, TIMEOUT, 0, 0));
         deltas_.resize(n_);
@@ -144,6 +144,7 @@ private:
     double sampling_eval_;
     uint64_t seed_;
     int64_t n_parts_ = 0;
+    bool sampled_ = false;
     std::vector<int64_t> deltas_;
     std::vector<int64_t> indptr_, indices_;
     std::unordered_map<int64_t, std::vector<Msg>> pending_;
@@ -179,9 +180,17 @@ private:
     // python scheduler's _reply_pid)
     int32_t reply_pid(int64_t t, int32_t replier)
     {
-        if (n_parts_ <= 0) return -1;
-        Stream g(stream_key(seed_, PART, (uint64_t)t, (uint64_t)(1 + replier)));
-        return (int32_t)g.integers(0, n_parts_);
+        if (n_parts_ > 0) {
+            Stream g(stream_key(seed_, PART, (uint64_t)t,
+                                (uint64_t)(1 + replier)));
+            return (int32_t)g.integers(0, n_parts_);
+        }
+        if (sampled_) {
+            Stream g(stream_key(seed_, SAMPLE, (uint64_t)t,
+                                (uint64_t)(1 + replier)));
+            return (int32_t)g.integers(0, (int64_t)1 << 31);
+        }
+        return -1;
     }
 
     // returns true if the reply was enqueued (false = dropped)
@@ -271,6 +280,10 @@ py::dict NativeScheduler::next_round(int64_t r)
                 Stream gpt(stream_key(seed_, PART, (uint64_t)t, 0));
                 for (size_t j = 0; j < n_f; ++j)
                     pids[j] = (int32_t)gpt.integers(0, n_parts_);
+            } else if (sampled_) {
+                Stream gsm(stream_key(seed_, SAMPLE, (uint64_t)t, 0));
+                for (size_t j = 0; j < n_f; ++j)
+                    pids[j] = (int32_t)gsm.integers(0, (int64_t)1 << 31);
             }
             for (size_t j = 0; j < n_f; ++j) {
                 int32_t sender = firing[j], receiver = peers[j];
@@ -444,13 +457,13 @@ PYBIND11_MODULE(_gossip_sched, m)
     py::class_<NativeScheduler>(m, "NativeScheduler")
         .def(py::init<int64_t, int64_t, int, int64_t, double, double, int,
                       int64_t, int64_t, double, int64_t, bool, double,
-                      uint64_t, py::object, py::object, int64_t>(),
+                      uint64_t, py::object, py::object, int64_t, bool>(),
              py::arg("n_nodes"), py::arg("delta"), py::arg("proto"),
              py::arg("model_size"), py::arg("drop_prob"), py::arg("online_prob"),
              py::arg("delay_kind"), py::arg("dmin"), py::arg("dmax"),
              py::arg("timexunit"), py::arg("overhead"), py::arg("sync"),
              py::arg("sampling_eval"), py::arg("seed"),
              py::arg("peers_indptr"), py::arg("peers_indices"),
-             py::arg("n_parts") = 0)
+             py::arg("n_parts") = 0, py::arg("sampled") = false)
         .def("next_round", &NativeScheduler::next_round);
 }

@@ -41,7 +41,7 @@ import torch
 from ..core import CreateModelMode
 from .arena import DataArena, NodeStateArena, SlotPool
 from .models import AdaLineSpec, LogRegSpec, MLPSpec, PegasosSpec
-from .rng import Purpose, RandomTape
+from .rng import Purpose, RandomTape, sample_indices
 
 __all__ = ["TorchBackend", "HIPBackend", "make_backend"]
 
@@ -345,6 +345,51 @@ class TorchBackend:
                         torch.tensor([r], dtype=torch.long),
                     )
 
+    def _merge_samp(self, state, pool, spec, node: int, slot: int, seed: int) -> None:
+        """Sampled-coordinate mean merge (TorchModelSampling.merge,
+        gossipy/model/sampling.py:75-107): averages only the seeded random
+        index subset; ages untouched (SamplingTMH._merge never merges
+        n_updates, gossipy/model/handler.py:431-433)."""
+        idx = torch.from_numpy(
+            sample_indices(seed, spec.samp_count(), spec.D)
+        ).to(state.params.device)
+        vals = 0.5 * (state.params[node, idx] + pool.slots[slot, idx])
+        state.params[node, idx] = vals
+
+    def _deliver_samp(
+        self, state, pool, data, spec, recv_nodes, recv_ptr, del_slots,
+        reply_slots, del_seeds,
+    ) -> None:
+        """Sampled mode dispatch (SamplingTMH.__call__,
+        gossipy/model/handler.py:435-452; PASS is an error there)."""
+        mode = spec.mode
+        ptr = recv_ptr.tolist()
+        for i, node_t in enumerate(recv_nodes.tolist()):
+            node = torch.tensor([node_t])
+            for j in range(ptr[i], ptr[i + 1]):
+                slot = int(del_slots[j])
+                seed = int(del_seeds[j])
+                if mode == CreateModelMode.MERGE_UPDATE:
+                    self._merge_samp(state, pool, spec, node_t, slot, seed)
+                    self.update(state, data, spec, node)
+                elif mode == CreateModelMode.UPDATE:
+                    self._train_slot(state, pool, data, spec, node_t, slot)
+                    self._merge_samp(state, pool, spec, node_t, slot, seed)
+                elif mode == CreateModelMode.UPDATE_MERGE:
+                    self.update(state, data, spec, node)
+                    self._train_slot(state, pool, data, spec, node_t, slot)
+                    self._merge_samp(state, pool, spec, node_t, slot, seed)
+                else:
+                    raise ValueError("Mode PASS not allowed for sampled models.")
+                r = int(reply_slots[j])
+                if r >= 0:
+                    self.snapshot(
+                        state,
+                        pool,
+                        torch.tensor([node_t], dtype=torch.long),
+                        torch.tensor([r], dtype=torch.long),
+                    )
+
     # -- deliveries ----------------------------------------------------------
 
     def deliver(
@@ -364,6 +409,12 @@ class TorchBackend:
         message asked for one (PUSH_PULL)."""
         if getattr(spec, "n_parts", 0) > 0:
             self._deliver_part(
+                state, pool, data, spec, recv_nodes, recv_ptr, del_slots,
+                reply_slots, del_pids,
+            )
+            return
+        if getattr(spec, "sample_size", 0) > 0:
+            self._deliver_samp(
                 state, pool, data, spec, recv_nodes, recv_ptr, del_slots,
                 reply_slots, del_pids,
             )
@@ -620,6 +671,31 @@ class HIPBackend(TorchBackend):
                 ptr,
                 apart,
                 spec.n_parts,
+                spec.d_in,
+                spec.n_classes,
+                spec.lr,
+                spec.weight_decay,
+                max(1, spec.local_epochs),
+                spec.batch_size,
+                mode,
+                bool(update_only),
+            )
+        elif getattr(spec, "sample_size", 0) > 0:
+            assert spec.family == "logreg", "sampled HIP path: logreg only"
+            self.ext.tick_logreg_samp(
+                state.params,
+                state.ages,
+                slots,
+                slot_ages,
+                nodes,
+                recv_ptr,
+                del_slots,
+                reply_slots,
+                del_pids,
+                data.x,
+                data.y,
+                data.counts,
+                spec.samp_count(),
                 spec.d_in,
                 spec.n_classes,
                 spec.lr,

@@ -147,8 +147,20 @@ class LogRegSpec(_PartitionMixin):
     batch_size: int = 32  #: 0 = full-batch
     mode: CreateModelMode = CreateModelMode.MERGE_UPDATE
     n_parts: int = 0
+    #: >0 = sampled gossip (SamplingBasedNode, gossipy/node.py:499-562):
+    #: each delivery merges only a random ``sample_size`` fraction of the
+    #: coordinates. Engine-native sampling: uniform over the flat parameter
+    #: row with replacement — the same distribution as the reference's
+    #: numel-weighted layer multinomial + per-dim uniform draw
+    #: (gossipy/model/sampling.py:37-72), derived from a per-delivery tape
+    #: seed instead of global RNG. Mutually exclusive with ``n_parts``.
+    sample_size: float = 0.0
 
     family = "logreg"
+
+    def samp_count(self) -> int:
+        """Coordinates per sampled merge (gossipy/model/sampling.py:57)."""
+        return max(1, int(round(self.sample_size * self.D)))
 
     @property
     def D(self) -> int:

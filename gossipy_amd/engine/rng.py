@@ -97,6 +97,15 @@ class TapeStream:
         return mu + sigma * z
 
 
+def sample_indices(seed: int, count: int, D: int) -> np.ndarray:
+    """``count`` coordinate indices in [0, D) with replacement for a sampled
+    merge (K6): draw ``j`` is ``splitmix64(seed + j) % D``. The HIP kernel
+    computes the identical sequence device-side (gossip_kernels.hip,
+    samp_merge)."""
+    raw = _splitmix64(np.uint64(seed) + np.arange(count, dtype=np.uint64))
+    return (raw % np.uint64(D)).astype(np.int64)
+
+
 class RandomTape:
     """Deterministic per-(purpose, timestep, extra) streams."""
 

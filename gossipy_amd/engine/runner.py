@@ -352,6 +352,7 @@ class BatchedGossipSimulator(SimulationEventSender):
             and getattr(self.backend, "ext", None) is not None
             and self.spec.family in ("logreg", "pegasos", "adaline")
             and (self.spec.family == "logreg" or getattr(self.spec, "n_parts", 0) == 0)
+            and getattr(self.spec, "sample_size", 0) == 0
         )
 
     def _run_round_fast(self, f: dict) -> None:

@@ -67,6 +67,10 @@ class EngineConfig:
     #: >0 = partitioned gossip (PartitioningBasedNode, gossipy/node.py:566-659):
     #: every PUSH/REPLY carries a uniformly drawn partition id
     n_parts: int = 0
+    #: sampled gossip (SamplingBasedNode, gossipy/node.py:499-562): every
+    #: PUSH/REPLY carries a sample seed (rides the del_pids channel;
+    #: mutually exclusive with n_parts)
+    sampled: bool = False
 
 
 @dataclass
@@ -252,15 +256,9 @@ class Scheduler:
                 drop_u = self.tape.uniform(Purpose.DROP, t, n_f)
                 sizes = np.full(n_f, cfg.model_size if proto != AntiEntropyProtocol.PULL else 1)
                 delays = self._delays(t, n_f, sizes)
-                # partition ids drawn per send (gossipy/node.py:617,631)
-                if cfg.n_parts > 0:
-                    pids = np.atleast_1d(
-                        self.tape.stream(Purpose.PART, t).integers(
-                            0, cfg.n_parts, size=n_f
-                        )
-                    )
-                else:
-                    pids = np.full(n_f, -1, dtype=np.int64)
+                # partition ids / sample seeds drawn per send
+                # (gossipy/node.py:617,631 / :520-531)
+                pids = self._send_extras(t, n_f)
                 for j in range(n_f):
                     sender, receiver = int(firing[j]), int(peers[j])
                     is_pull = proto == AntiEntropyProtocol.PULL
@@ -404,16 +402,37 @@ class Scheduler:
             eval_nodes=eval_nodes,
         )
 
-    def _reply_pid(self, t: int, replier: int) -> int:
-        """Fresh partition id for a reply (gossipy/node.py:651). Keyed on
-        (t, replier): two replies by one node in one tick share the draw."""
-        if self.cfg.n_parts <= 0:
-            return -1
-        return int(
-            self.tape.stream(Purpose.PART, t, extra=1 + replier).integers(
-                0, self.cfg.n_parts
+    def _send_extras(self, t: int, n: int) -> np.ndarray:
+        """Per-send extra int riding with each message: partition id
+        (partitioned gossip), sample seed (sampled gossip), or -1."""
+        cfg = self.cfg
+        if cfg.n_parts > 0:
+            return np.atleast_1d(
+                self.tape.stream(Purpose.PART, t).integers(0, cfg.n_parts, size=n)
             )
-        )
+        if cfg.sampled:
+            return np.atleast_1d(
+                self.tape.stream(Purpose.SAMPLE, t).integers(0, 2**31, size=n)
+            )
+        return np.full(n, -1, dtype=np.int64)
+
+    def _reply_pid(self, t: int, replier: int) -> int:
+        """Fresh partition id / sample seed for a reply
+        (gossipy/node.py:651 / :551-557). Keyed on (t, replier): two
+        replies by one node in one tick share the draw."""
+        if self.cfg.n_parts > 0:
+            return int(
+                self.tape.stream(Purpose.PART, t, extra=1 + replier).integers(
+                    0, self.cfg.n_parts
+                )
+            )
+        if self.cfg.sampled:
+            return int(
+                self.tape.stream(Purpose.SAMPLE, t, extra=1 + replier).integers(
+                    0, 2**31
+                )
+            )
+        return -1
 
     def _enqueue_reply(
         self, t: int, replier: int, requester: int, slot: int, pid: int = -1
@@ -527,14 +546,7 @@ class TokenizedScheduler(Scheduler):
                         cfg.model_size if proto != AntiEntropyProtocol.PULL else 1,
                     )
                     delays = self._delays(t, n_s, sizes)
-                    if cfg.n_parts > 0:
-                        pids = np.atleast_1d(
-                            self.tape.stream(Purpose.PART, t).integers(
-                                0, cfg.n_parts, size=n_s
-                            )
-                        )
-                    else:
-                        pids = np.full(n_s, -1, dtype=np.int64)
+                    pids = self._send_extras(t, n_s)
                     for j in range(n_s):
                         sender, receiver = int(senders[j]), int(peers[j])
                         is_pull = proto == AntiEntropyProtocol.PULL
@@ -622,15 +634,20 @@ class TokenizedScheduler(Scheduler):
                             bslot = self._alloc_slot(receiver)
                             burst_snap_nodes.append(receiver)
                             burst_snap_slots.append(bslot)
-                            bpid = (
-                                int(
+                            if cfg.n_parts > 0:
+                                bpid = int(
                                     tick_stream(Purpose.PART, receiver).integers(
                                         0, cfg.n_parts
                                     )
                                 )
-                                if cfg.n_parts > 0
-                                else -1
-                            )
+                            elif cfg.sampled:
+                                bpid = int(
+                                    tick_stream(Purpose.SAMPLE, receiver).integers(
+                                        0, 2**31
+                                    )
+                                )
+                            else:
+                                bpid = -1
                             sent += 1
                             total_size += cfg.model_size
                             if float(gd.random()) >= cfg.drop_prob:
@@ -1012,6 +1029,7 @@ class NativeSchedulerAdapter:
             None if ip is None else np.ascontiguousarray(ip, dtype=np.int64),
             None if ix is None else np.ascontiguousarray(ix, dtype=np.int64),
             cfg.n_parts,
+            cfg.sampled,
         )
         self.last_flat: Optional[dict] = None
 

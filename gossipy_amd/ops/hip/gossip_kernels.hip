@@ -437,6 +437,115 @@ tick_logreg_part_kernel(LogregPartArgs a)
 }
 
 // ---------------------------------------------------------------------------
+// sampled logreg tick (K6, SamplingTMH semantics)
+//
+// Parity: gossipy/model/handler.py:426-452 + gossipy/model/sampling.py:
+// 75-107. Each delivery averages only a seeded random coordinate subset
+// (with replacement); the index sequence is splitmix64(seed + j) % D —
+// bit-identical to engine/rng.py sample_indices, so the torch oracle and
+// this kernel draw the same coordinates. Ages are not merged.
+// ---------------------------------------------------------------------------
+
+DEV_INLINE unsigned long long splitmix64_dev(unsigned long long x)
+{
+    x += 0x9E3779B97F4A7C15ULL;
+    x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ULL;
+    x = (x ^ (x >> 27)) * 0x94D049BB133111EBULL;
+    return x ^ (x >> 31);
+}
+
+struct LogregSampArgs {
+    float* params; int* ages;
+    float* slots; int* slot_ages;
+    const int* nodes; const int* ptr;
+    const int* dslots; const int* rslots; const int* dseeds;
+    const float* X; const float* Y; const int* counts;
+    int samp_c, d, k, Smax, D;
+    float lr, wd;
+    int epochs, bs, mode, update_only;
+};
+
+// W3 holds the pre-merge W (duplicate sampled indices must all read the
+// ORIGINAL value — torch advanced-index assignment semantics).
+DEV_INLINE void logreg_samp_merge(const LogregSampArgs& a, float* W, float* W3,
+                                  const float* src, int seed)
+{
+    int tid = threadIdx.x;
+    for (int e = tid; e < a.D; e += blockDim.x) W3[e] = W[e];
+    __syncthreads();
+    for (int q = tid; q < a.samp_c; q += blockDim.x) {
+        int i = (int)(splitmix64_dev((unsigned long long)seed +
+                                     (unsigned long long)q) %
+                      (unsigned long long)a.D);
+        W[i] = 0.5f * (W3[i] + src[i]);
+    }
+    __syncthreads();
+}
+
+__global__ void __launch_bounds__(128)
+tick_logreg_samp_kernel(LogregSampArgs a)
+{
+    int i = blockIdx.x;
+    int node = a.nodes[i];
+    int tid = threadIdx.x;
+    extern __shared__ float sm[];
+    float* W = sm;                         // D
+    float* W2 = W + a.D;                   // D (received-model scratch)
+    float* W3 = W2 + a.D;                  // D (pre-merge snapshot)
+    float* xb = W3 + a.D;                  // bsmax*d
+    int bsmax = (a.bs == 0) ? a.Smax : min(a.bs, a.Smax);
+    float* dz = xb + bsmax * a.d;          // bsmax*k
+
+    for (int e = tid; e < a.D; e += blockDim.x)
+        W[e] = a.params[(long)node * a.D + e];
+    __syncthreads();
+    int age = a.ages[node];
+
+    // reuse the plain logreg update math: wrap args in a LogregArgs view
+    LogregArgs u;
+    u.X = a.X; u.Y = a.Y; u.counts = a.counts;
+    u.d = a.d; u.k = a.k; u.Smax = a.Smax; u.D = a.D;
+    u.lr = a.lr; u.wd = a.wd; u.epochs = a.epochs; u.bs = a.bs;
+
+    if (a.update_only) {
+        logreg_update(u, node, W, xb, dz, age);
+    } else {
+        for (int j = a.ptr[i]; j < a.ptr[i + 1]; ++j) {
+            int slot = a.dslots[j];
+            int seed = a.dseeds ? a.dseeds[j] : 0;
+            const float* srow = a.slots + (long)slot * a.D;
+            int sage = a.slot_ages[slot];
+            if (a.mode == MODE_MERGE_UPDATE) {
+                logreg_samp_merge(a, W, W3, srow, seed);
+                logreg_update(u, node, W, xb, dz, age);
+            } else {
+                // UPDATE: train received, merge its sample into self
+                // (handler.py:440-442); UPDATE_MERGE: self-update first
+                // (handler.py:445-448)
+                if (a.mode == MODE_UPDATE_MERGE)
+                    logreg_update(u, node, W, xb, dz, age);
+                for (int e = tid; e < a.D; e += blockDim.x) W2[e] = srow[e];
+                __syncthreads();
+                int age2 = sage;
+                logreg_update(u, node, W2, xb, dz, age2);
+                logreg_samp_merge(a, W, W3, W2, seed);
+            }
+            int rs = a.rslots ? a.rslots[j] : -1;
+            if (rs >= 0) {
+                for (int e = tid; e < a.D; e += blockDim.x)
+                    a.slots[(long)rs * a.D + e] = W[e];
+                if (tid == 0) a.slot_ages[rs] = age;
+                __syncthreads();
+            }
+        }
+    }
+    __syncthreads();
+    for (int e = tid; e < a.D; e += blockDim.x)
+        a.params[(long)node * a.D + e] = W[e];
+    if (tid == 0) a.ages[node] = age;
+}
+
+// ---------------------------------------------------------------------------
 // pegasos / adaline tick: one WAVE per node, weights in registers
 // ---------------------------------------------------------------------------
 
@@ -1113,6 +1222,40 @@ void run_round_linear(
     }
 }
 
+void tick_logreg_samp(
+    torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
+    torch::Tensor slot_ages, torch::Tensor nodes, torch::Tensor recv_ptr,
+    torch::Tensor del_slots, torch::Tensor reply_slots, torch::Tensor del_seeds,
+    torch::Tensor X, torch::Tensor Y, torch::Tensor counts,
+    int64_t samp_c, int64_t d, int64_t k, double lr, double wd,
+    int64_t epochs, int64_t bs, int64_t mode, bool update_only)
+{
+    CHECK_DEV(params); CHECK_DEV(slots); CHECK_DEV(X); CHECK_DEV(Y);
+    TORCH_CHECK(k <= KMAX, "n_classes > ", KMAX, " unsupported");
+    TORCH_CHECK(mode != MODE_PASS, "Mode PASS not allowed for sampled models.");
+    int n = nodes.size(0);
+    if (n == 0) return;
+    LogregSampArgs a;
+    a.params = params.data_ptr<float>(); a.ages = ages.data_ptr<int>();
+    a.slots = slots.data_ptr<float>(); a.slot_ages = slot_ages.data_ptr<int>();
+    a.nodes = nodes.data_ptr<int>(); a.ptr = recv_ptr.data_ptr<int>();
+    a.dslots = del_slots.numel() ? del_slots.data_ptr<int>() : nullptr;
+    a.rslots = reply_slots.numel() ? reply_slots.data_ptr<int>() : nullptr;
+    a.dseeds = del_seeds.numel() ? del_seeds.data_ptr<int>() : nullptr;
+    a.X = X.data_ptr<float>(); a.Y = Y.data_ptr<float>();
+    a.counts = counts.data_ptr<int>();
+    a.samp_c = samp_c; a.d = d; a.k = k; a.Smax = X.size(1);
+    a.D = params.size(1);
+    a.lr = lr; a.wd = wd; a.epochs = epochs; a.bs = bs; a.mode = mode;
+    a.update_only = update_only;
+    int bsmax = (bs == 0) ? a.Smax : std::min<int>(bs, a.Smax);
+    size_t smem = sizeof(float) * (3 * a.D + (size_t)bsmax * a.d +
+                                   (size_t)bsmax * a.k);
+    TORCH_CHECK(smem <= 160 * 1024, "sampled logreg LDS budget exceeded: ", smem);
+    hipLaunchKernelGGL(tick_logreg_samp_kernel, dim3(n), dim3(128), smem,
+                       current_stream(), a);
+}
+
 void run_round_logreg_part(
     torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
     torch::Tensor slot_ages,
@@ -1201,4 +1344,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m)
     m.def("run_round_logreg_part", &run_round_logreg_part,
           "whole-round executor, partitioned logreg family");
     m.def("wmerge", &wmerge, "all2all weighted k-way merge (K5 weighted)");
+    m.def("tick_logreg_samp", &tick_logreg_samp,
+          "fused sampled-merge + logreg SGD tick (K6)");
 }

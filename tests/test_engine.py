@@ -814,3 +814,57 @@ class TestEngineCheckpoint:
         ]
         restored.start(n_rounds=3)
         assert torch.equal(ref.local_params(), restored.local_params())
+
+
+# ---------------------------------------------------------------------------
+# sampled gossip (SamplingBasedNode / SamplingTMH)
+# ---------------------------------------------------------------------------
+
+
+class TestSampledEngine:
+    def _run(self, rounds=15, n_nodes=40, sample=0.3, **cfg_kw):
+        shards, geval = _make_data(n_nodes, seed=1)
+        data = DataArena.from_shards(shards, torch.device("cpu"), global_eval=geval)
+        base = dict(
+            n_nodes=n_nodes, delta=10, protocol=AntiEntropyProtocol.PUSH,
+            model_size=116, sampling_eval=0.25, seed=17, sampled=True,
+        )
+        base.update(cfg_kw)
+        cfg = EngineConfig(**base)
+        spec = LogRegSpec(d_in=57, n_classes=2, lr=0.1, sample_size=sample)
+        sim = BatchedGossipSimulator(cfg, spec, data)
+        rep = SimulationReport()
+        sim.add_receiver(rep)
+        sim.init_nodes()
+        sim.start(n_rounds=rounds)
+        return sim, rep
+
+    def test_learns(self):
+        sim, rep = self._run()
+        assert rep.get_evaluation(False)[-1][1]["accuracy"] > 0.9
+
+    def test_deterministic(self):
+        s1, _ = self._run(rounds=4)
+        s2, _ = self._run(rounds=4)
+        assert torch.equal(s1.local_params(), s2.local_params())
+
+    def test_sample_touches_subset(self):
+        """A single sampled merge must change at most samp_count coords."""
+        from gossipy_amd.engine.arena import NodeStateArena, SlotPool
+        from gossipy_amd.engine.backend import TorchBackend
+
+        spec = LogRegSpec(d_in=57, n_classes=2, sample_size=0.2)
+        state = NodeStateArena(1, spec.D, torch.device("cpu"))
+        state.params.normal_(generator=torch.Generator().manual_seed(0))
+        before = state.params.clone()
+        pool = SlotPool(spec.D, torch.device("cpu"), 2)
+        pool.slots.normal_(generator=torch.Generator().manual_seed(1))
+        TorchBackend()._merge_samp(state, pool, spec, 0, 1, seed=1234)
+        changed = (state.params[0] != before[0]).sum()
+        assert 0 < changed <= spec.samp_count()
+
+    def test_matches_object_layer_distribution(self):
+        """samp_count matches the reference's round(size * net_size)
+        (gossipy/model/sampling.py:57)."""
+        spec = LogRegSpec(d_in=57, n_classes=2, sample_size=0.25)
+        assert spec.samp_count() == max(1, int(round(0.25 * 116)))

@@ -463,3 +463,59 @@ class TestTokenizedGPU:
         sim.start(n_rounds=12)
         torch.cuda.synchronize()
         assert rep.get_evaluation(False)[-1][1]["accuracy"] > 0.85
+
+
+class TestSampledGPU:
+    """tick_logreg_samp must draw the same splitmix64 coordinate sequence as
+    the torch oracle and match it bitwise-closely."""
+
+    @pytest.mark.parametrize("mode", [
+        CreateModelMode.MERGE_UPDATE,
+        CreateModelMode.UPDATE,
+        CreateModelMode.UPDATE_MERGE,
+    ])
+    def test_sampled_deliver_matches_oracle(self, mode):
+        spec = LogRegSpec(
+            d_in=57, n_classes=2, lr=0.1, batch_size=0, sample_size=0.3,
+            mode=mode,
+        )
+        cs, gs, cd, gd = _pair(12, spec, 57)
+        cpool, gpool = SlotPool(spec.D, CPU, 16), SlotPool(spec.D, CUDA, 16)
+        cpool.slots.normal_(generator=torch.Generator().manual_seed(5))
+        cpool.slot_ages.copy_(torch.arange(16, dtype=torch.int32) * 2)
+        gpool.slots.copy_(cpool.slots)
+        gpool.slot_ages.copy_(cpool.slot_ages)
+        recv = torch.tensor([2, 5, 9], dtype=torch.int64)
+        ptr = torch.tensor([0, 2, 3, 4], dtype=torch.int64)
+        slots = torch.tensor([1, 4, 7, 10], dtype=torch.int64)
+        reply = torch.tensor([12, -1, 13, -1], dtype=torch.int64)
+        seeds = torch.tensor(
+            [123456, 777, 2**30 + 5, 42], dtype=torch.int64
+        )
+        TorchBackend().deliver(cs, cpool, cd, spec, recv, ptr, slots, reply, seeds)
+        HIPBackend().deliver(gs, gpool, gd, spec, recv, ptr, slots, reply, seeds)
+        torch.cuda.synchronize()
+        assert _close(cs.params, gs.params)
+        assert torch.equal(cs.ages.cpu(), gs.ages.cpu())
+
+    def test_sampled_gpu_learns(self):
+        X, y = make_synthetic_classification((640, 57, 2), seed=0, margin=2.0)
+        idx = np.random.default_rng(0).permutation(640)
+        shards = [(X[s], y[s]) for s in np.array_split(idx[:576], 64)]
+        data = DataArena.from_shards(
+            shards, CUDA, global_eval=(X[idx[576:]], y[idx[576:]])
+        )
+        cfg = EngineConfig(
+            n_nodes=64, delta=10, protocol=AntiEntropyProtocol.PUSH,
+            model_size=116, sampling_eval=0.25, seed=17, sampled=True,
+        )
+        sim = BatchedGossipSimulator(
+            cfg, LogRegSpec(d_in=57, n_classes=2, lr=0.1, sample_size=0.3),
+            data, device=CUDA,
+        )
+        rep = SimulationReport()
+        sim.add_receiver(rep)
+        sim.init_nodes()
+        sim.start(n_rounds=15)
+        torch.cuda.synchronize()
+        assert rep.get_evaluation(False)[-1][1]["accuracy"] > 0.9
