@@ -11,7 +11,7 @@ the reference.
 
 from .arena import DataArena, NodeStateArena, SlotPool
 from .backend import HIPBackend, TorchBackend, make_backend
-from .models import AdaLineSpec, LogRegSpec, MLPSpec, PegasosSpec
+from .models import AdaLineSpec, LogRegSpec, MFSpec, MLPSpec, PegasosSpec
 from .rng import Purpose, RandomTape
 from .runner import (
     BatchedAll2AllGossipSimulator,
@@ -48,4 +48,5 @@ __all__ = [
     "PegasosSpec",
     "LogRegSpec",
     "MLPSpec",
+    "MFSpec",
 ]

@@ -92,6 +92,26 @@ class TorchBackend:
         if spec.family in ("pegasos", "adaline"):
             state.params.zero_()
             return
+        if spec.family == "mf":
+            # MFModelHandler.init (gossipy/model/handler.py:542-548):
+            # X, Y ~ U(0,1)*sqrt((r_max-r_min)/k); b = c = r_min/2; age = 1.
+            # Per-node tape streams keyed on the GLOBAL node id make the
+            # init residency-invariant without drawing the whole population.
+            mul = np.sqrt((spec.r_max - spec.r_min) / spec.k)
+            n = state.params.shape[0]
+            k, ni = spec.k, spec.n_items
+            rows = np.empty((n, spec.D), dtype=np.float32)
+            for li in range(n):
+                g = tape.stream(Purpose.INIT, t=state.node_lo + li)
+                rows[li, :k] = g.uniform(0, 1, size=k) * mul  # X
+                rows[li, k] = spec.r_min / 2.0  # b
+                rows[li, k + 1 : k + 1 + ni * k] = (
+                    g.uniform(0, 1, size=ni * k) * mul
+                )  # Y
+                rows[li, k + 1 + ni * k :] = spec.r_min / 2.0  # c
+            state.params.copy_(torch.from_numpy(rows).to(state.params.device))
+            state.ages.fill_(1)  # n_updates starts at 1 (handler.py:540)
+            return
         g = tape.stream(Purpose.INIT)
         n, D = state.params.shape
         n_total = n_total or n
@@ -124,9 +144,16 @@ class TorchBackend:
         pool: SlotPool,
         nodes: torch.Tensor,
         slot_ids: torch.Tensor,
+        src_off: int = 0,
     ) -> None:
-        """Arena row copy — the batched ``ModelHandler.caching``."""
-        pool.slots[slot_ids.long()] = state.params[nodes.long()]
+        """Arena row copy — the batched ``ModelHandler.caching``.
+
+        ``src_off`` > 0 snapshots a sub-block of the row (MF ships only the
+        item block, gossipy/model/handler.py:562-568)."""
+        W = pool.slots.shape[1]
+        pool.slots[slot_ids.long()] = state.params[
+            nodes.long(), src_off : src_off + W
+        ]
         pool.slot_ages[slot_ids.long()] = state.ages[nodes.long()]
 
     # -- local updates -------------------------------------------------------
@@ -144,6 +171,8 @@ class TorchBackend:
             return
         if getattr(spec, "n_parts", 0) > 0:
             self._update_part(state.params, state.ages, data, spec, nodes.long())
+        elif spec.family == "mf":
+            self._update_mf(state.params, state.ages, data, spec, nodes.long())
         elif spec.family == "logreg":
             self._update_logreg(state.params, state.ages, data, spec, nodes.long())
         elif spec.family == "mlp":
@@ -390,6 +419,90 @@ class TorchBackend:
                         torch.tensor([r], dtype=torch.long),
                     )
 
+    # -- matrix factorization (K9/K10) ---------------------------------------
+
+    def _update_mf(self, params, ages, data, spec, nodes) -> None:
+        """Per-rating SGD (MFModelHandler._update,
+        gossipy/model/handler.py:550-560). Order-dependent: ``Y[i]`` uses
+        the OLD ``X``; ``X`` then uses the NEW ``Y[i]``."""
+        k, ni = spec.k, spec.n_items
+        shrink = 1.0 - spec.reg * spec.lr
+        for idx in nodes.tolist():
+            c_n = int(data.counts[idx])
+            if c_n == 0:
+                continue
+            row = params[idx]
+            X = row[:k]
+            Yoff, coff = k + 1, k + 1 + ni * k
+            t_age = int(ages[idx])
+            for s in range(c_n):
+                item = int(data.x[idx, s, 0])
+                r = float(data.y[idx, s])
+                Yi = row[Yoff + item * k : Yoff + (item + 1) * k]
+                err = r - float(X @ Yi) - float(row[k]) - float(row[coff + item])
+                Yi.mul_(shrink).add_(spec.lr * err * X)
+                X.mul_(shrink).add_(spec.lr * err * Yi)
+                row[k] += spec.lr * err
+                row[coff + item] += spec.lr * err
+                t_age += 1
+            ages[idx] = t_age
+
+    def _merge_mf(self, state, pool, spec, node: int, slot: int) -> None:
+        """Item-side age-weighted merge with the reference's extra /2
+        (gossipy/model/handler.py:562-568); ``n_updates`` is NOT merged."""
+        off = spec.item_off
+        w1 = int(state.ages[node])
+        w2 = int(pool.slot_ages[slot])
+        den = 2.0 * (w1 + w2)
+        blk = state.params[node, off:]
+        blk.mul_(w1 / den).add_(pool.slots[slot] * (w2 / den))
+
+    def _deliver_mf(
+        self, state, pool, data, spec, recv_nodes, recv_ptr, del_slots,
+        reply_slots,
+    ) -> None:
+        if spec.mode != CreateModelMode.MERGE_UPDATE:
+            raise ValueError(
+                "MF engine supports MERGE_UPDATE only (item-block transport"
+                " cannot adopt the sender's user factors)"
+            )
+        ptr = recv_ptr.tolist()
+        for i, node_t in enumerate(recv_nodes.tolist()):
+            node = torch.tensor([node_t])
+            for j in range(ptr[i], ptr[i + 1]):
+                self._merge_mf(state, pool, spec, node_t, int(del_slots[j]))
+                self.update(state, data, spec, node)
+                r = int(reply_slots[j])
+                if r >= 0:
+                    self.snapshot(
+                        state,
+                        pool,
+                        torch.tensor([node_t], dtype=torch.long),
+                        torch.tensor([r], dtype=torch.long),
+                        src_off=spec.item_off,
+                    )
+
+    def mf_rmse(self, state, data, spec, nodes: torch.Tensor) -> list:
+        """Per-node RMSE on its test ratings (MFModelHandler.evaluate,
+        gossipy/model/handler.py:570-573)."""
+        k, ni = spec.k, spec.n_items
+        out = []
+        Yoff, coff = k + 1, k + 1 + ni * k
+        for idx in nodes.tolist():
+            c_n = int(data.tcounts[idx]) if data.tcounts is not None else 0
+            if c_n == 0:
+                continue
+            row = state.params[idx]
+            X = row[:k]
+            Y = row[Yoff:coff].view(ni, k)
+            R = Y @ X + row[k] + row[coff:]
+            items = data.tx[idx, :c_n, 0].long()
+            ratings = data.ty[idx, :c_n]
+            out.append(
+                {"rmse": float(torch.sqrt(torch.mean((ratings - R[items]) ** 2)))}
+            )
+        return out
+
     # -- deliveries ----------------------------------------------------------
 
     def deliver(
@@ -417,6 +530,12 @@ class TorchBackend:
             self._deliver_samp(
                 state, pool, data, spec, recv_nodes, recv_ptr, del_slots,
                 reply_slots, del_pids,
+            )
+            return
+        if spec.family == "mf":
+            self._deliver_mf(
+                state, pool, data, spec, recv_nodes, recv_ptr, del_slots,
+                reply_slots,
             )
             return
         mode = spec.mode
@@ -553,7 +672,7 @@ class HIPBackend(TorchBackend):
 
         self.ext = ops.load_extension()  # raises if the .so is missing
 
-    def snapshot(self, state, pool, nodes, slot_ids) -> None:
+    def snapshot(self, state, pool, nodes, slot_ids, src_off: int = 0) -> None:
         if len(nodes) == 0:
             return
         self.ext.snapshot(
@@ -564,6 +683,7 @@ class HIPBackend(TorchBackend):
             nodes.to(state.params.device, torch.int32),
             slot_ids.to(state.params.device, torch.int32),
             getattr(state, "age_width", 1),
+            src_off,
         )
 
     def update(self, state, data, spec, nodes) -> None:
@@ -678,6 +798,30 @@ class HIPBackend(TorchBackend):
                 max(1, spec.local_epochs),
                 spec.batch_size,
                 mode,
+                bool(update_only),
+            )
+        elif spec.family == "mf":
+            if not update_only and spec.mode != CreateModelMode.MERGE_UPDATE:
+                raise ValueError(
+                    "MF engine supports MERGE_UPDATE only (item-block"
+                    " transport cannot adopt the sender's user factors)"
+                )
+            self.ext.tick_mf(
+                state.params,
+                state.ages,
+                slots,
+                slot_ages,
+                nodes,
+                recv_ptr,
+                del_slots,
+                reply_slots,
+                data.x,
+                data.y,
+                data.counts,
+                spec.k,
+                spec.n_items,
+                spec.reg,
+                spec.lr,
                 bool(update_only),
             )
         elif getattr(spec, "sample_size", 0) > 0:

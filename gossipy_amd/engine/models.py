@@ -27,7 +27,7 @@ import numpy as np
 
 from ..core import CreateModelMode
 
-__all__ = ["PegasosSpec", "AdaLineSpec", "LogRegSpec", "MLPSpec"]
+__all__ = ["PegasosSpec", "AdaLineSpec", "LogRegSpec", "MLPSpec", "MFSpec"]
 
 
 def _fortran_to_arena(layers: List[Tuple[int, int, int, int]], D: int) -> np.ndarray:
@@ -215,3 +215,47 @@ class MLPSpec(_PartitionMixin):
             off = b_off + dims[i + 1]
             out.append((w_off, b_off, dims[i], dims[i + 1]))
         return out
+
+
+@dataclass
+class MFSpec:
+    """Low-rank matrix-factorization recommender (MFModelHandler,
+    gossipy/model/handler.py:528-576).
+
+    Arena row layout: ``X`` (k user factors), ``b`` (1), ``Y``
+    (n_items*k, row-major), ``c`` (n_items). Messages carry only the item
+    block ``(Y, c)`` — the merge touches nothing else
+    (gossipy/model/handler.py:562-568) — so ``slot_width`` is the item
+    block size, not ``D``. Quirk parity: the merge divides by ``2*(w1+w2)``
+    (the reference's extra factor of two, handler.py:566-567) and does NOT
+    update ``n_updates``; ages start at 1 (handler.py:540).
+
+    Only MERGE_UPDATE is supported on the engine: UPDATE-style adoption
+    would need the sender's *user* factors, which the item-block transport
+    deliberately omits (use the object layer for those modes).
+    """
+
+    k: int  #: latent dimension (``dim``)
+    n_items: int
+    reg: float = 0.1  #: ``lam_reg``
+    lr: float = 0.001
+    r_min: int = 1
+    r_max: int = 5
+    mode: CreateModelMode = CreateModelMode.MERGE_UPDATE
+
+    family = "mf"
+    n_parts = 0
+    sample_size = 0.0
+    age_width = 1
+
+    @property
+    def D(self) -> int:
+        return self.k + 1 + self.n_items * self.k + self.n_items
+
+    @property
+    def item_off(self) -> int:
+        return self.k + 1
+
+    @property
+    def slot_width(self) -> int:
+        return self.n_items * self.k + self.n_items

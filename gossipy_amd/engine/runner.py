@@ -102,7 +102,11 @@ class BatchedGossipSimulator(SimulationEventSender):
         self.state = NodeStateArena(
             self.n_local, spec.D, device, self.node_lo, age_width=aw
         )
-        self.pool = SlotPool(spec.D, device, age_width=aw)
+        # MF ships only the item block; other families ship the full row
+        self.pool = SlotPool(
+            getattr(spec, "slot_width", spec.D), device, age_width=aw
+        )
+        self._snap_off = getattr(spec, "item_off", 0)
         self.data = data
         self.scheduler = make_scheduler(cfg)
         self.initialized = False
@@ -150,7 +154,7 @@ class BatchedGossipSimulator(SimulationEventSender):
             return
         ops = []
         recv_bufs = []
-        D = self.spec.D
+        D = self.pool.slots.shape[1]
         A = getattr(self.spec, "age_width", 1)
         for src, dst, slot_ids in needed:
             if src == dst:
@@ -208,6 +212,7 @@ class BatchedGossipSimulator(SimulationEventSender):
                 self.pool,
                 self._to_local_t(phase.snap_nodes[mine]),
                 torch.from_numpy(phase.snap_slots[mine].astype(np.int64)).to(self.device),
+                src_off=self._snap_off,
             )
         # move cross-GPU slots needed by this tick's deliveries
         self._exchange(
@@ -252,6 +257,7 @@ class BatchedGossipSimulator(SimulationEventSender):
                 torch.from_numpy(phase.pull_snap_slots[pmine].astype(np.int64)).to(
                     self.device
                 ),
+                src_off=self._snap_off,
             )
 
         # C: same-tick replies
@@ -301,6 +307,20 @@ class BatchedGossipSimulator(SimulationEventSender):
             nodes = np.arange(self.cfg.n_nodes)
         mine = nodes[self._is_mine(nodes)]
         local_ids = torch.from_numpy((mine - self.node_lo).astype(np.int64))
+
+        if self.spec.family == "mf":
+            # RecSys eval: per-user RMSE on the held-out ratings; there is
+            # no global test set (gossipy/data/__init__.py:550-555)
+            results_local = self.backend.mf_rmse(
+                self.state, self.data, self.spec, local_ids
+            )
+            if self.world > 1:
+                gathered: List[List[dict]] = [None] * self.world
+                dist.all_gather_object(gathered, results_local)
+                results_local = [d for part in gathered for d in part]
+            if self.rank == 0 and results_local:
+                self.notify_evaluation(t, True, results_local)
+            return
 
         results_global: List[dict] = []
         if self.data.gx is not None and len(mine):

@@ -43,6 +43,14 @@ constexpr int MODE_MERGE_UPDATE = 1;
 constexpr int MODE_UPDATE_MERGE = 2;
 constexpr int MODE_PASS = 3;
 
+DEV_INLINE float wave_sum(float v)
+{
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+        v += __shfl_down(v, off, WAVE);
+    return __shfl(v, 0, WAVE);
+}
+
+
 // ---------------------------------------------------------------------------
 // snapshot: slots[slot_ids[i]] = params[nodes[i]]  (+ age)
 // ---------------------------------------------------------------------------
@@ -54,16 +62,18 @@ __global__ void snapshot_kernel(
     int* __restrict__ slot_ages,
     const int* __restrict__ nodes,
     const int* __restrict__ slot_ids,
-    int n, int D, int A)  // A = age width (1 scalar, P for partitioned)
+    int n, int W, int Dp, int src_off, int A)
+    // W = slot width, Dp = full row width, src_off = sub-block offset
+    // (MF snapshots only the item block), A = age width
 {
-    long total = (long)n * D;
+    long total = (long)n * W;
     for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x; idx < total;
          idx += (long)gridDim.x * blockDim.x) {
-        int row = idx / D;
-        int col = idx - (long)row * D;
+        int row = idx / W;
+        int col = idx - (long)row * W;
         int node = nodes[row];
         int slot = slot_ids[row];
-        slots[(long)slot * D + col] = params[(long)node * D + col];
+        slots[(long)slot * W + col] = params[(long)node * Dp + src_off + col];
         if (col < A) slot_ages[(long)slot * A + col] = ages[(long)node * A + col];
     }
 }
@@ -546,15 +556,95 @@ tick_logreg_samp_kernel(LogregSampArgs a)
 }
 
 // ---------------------------------------------------------------------------
-// pegasos / adaline tick: one WAVE per node, weights in registers
+// matrix-factorization tick (K9/K10, MFModelHandler semantics)
+//
+// Parity: gossipy/model/handler.py:528-576. Per delivery: item-block
+// age-weighted merge with the reference's extra /2 (n_updates NOT merged),
+// then per-rating SGD where Y[i] uses the OLD X and X uses the NEW Y[i].
+// One wavefront per receiver: lanes cover the k latent dims in the rating
+// loop and stride the item block in the merge.
 // ---------------------------------------------------------------------------
 
-DEV_INLINE float wave_sum(float v)
+struct MFArgs {
+    float* params; int* ages;
+    float* slots; int* slot_ages;
+    const int* nodes; const int* ptr;
+    const int* dslots; const int* rslots;
+    const float* X; const float* Y; const int* counts;  // items ride X[:,:,0]
+    int k, n_items, Smax, D, item_off, Wslot;
+    float reg, lr;
+    int update_only;
+};
+
+DEV_INLINE void mf_update(const MFArgs& a, int node, float* row, int& age)
 {
-    for (int off = WAVE / 2; off > 0; off >>= 1)
-        v += __shfl_down(v, off, WAVE);
-    return __shfl(v, 0, WAVE);
+    int lane = threadIdx.x;
+    int c = a.counts[node];
+    const float* items = a.X + (long)node * a.Smax;
+    const float* ratings = a.Y + (long)node * a.Smax;
+    float shrink = 1.0f - a.reg * a.lr;
+    int coff = a.item_off + a.n_items * a.k;
+    for (int s = 0; s < c; ++s) {
+        int item = (int)items[s];
+        float r = ratings[s];
+        float* Yi = row + a.item_off + item * a.k;
+        float part = (lane < a.k) ? row[lane] * Yi[lane] : 0.f;
+        float dot = wave_sum(part);
+        float err = r - dot - row[a.k] - row[coff + item];
+        if (lane < a.k) {
+            float yi_new = shrink * Yi[lane] + a.lr * err * row[lane];
+            Yi[lane] = yi_new;
+            row[lane] = shrink * row[lane] + a.lr * err * yi_new;
+        }
+        if (lane == 0) {
+            row[a.k] += a.lr * err;
+            row[coff + item] += a.lr * err;
+        }
+        age += 1;
+    }
 }
+
+DEV_INLINE void mf_merge(const MFArgs& a, float* row, int age,
+                         const float* srow, int sage)
+{
+    float den = 2.0f * (age + sage);
+    float w1 = (float)age / den, w2 = (float)sage / den;
+    for (int e = threadIdx.x; e < a.Wslot; e += blockDim.x)
+        row[a.item_off + e] = w1 * row[a.item_off + e] + w2 * srow[e];
+}
+
+__global__ void __launch_bounds__(WAVE)
+tick_mf_kernel(MFArgs a)
+{
+    int i = blockIdx.x;
+    int node = a.nodes[i];
+    float* row = a.params + (long)node * a.D;
+    int age = a.ages[node];
+    if (a.update_only) {
+        mf_update(a, node, row, age);
+    } else {
+        for (int j = a.ptr[i]; j < a.ptr[i + 1]; ++j) {
+            int slot = a.dslots[j];
+            mf_merge(a, row, age, a.slots + (long)slot * a.Wslot,
+                     a.slot_ages[slot]);
+            __syncthreads();
+            mf_update(a, node, row, age);
+            __syncthreads();
+            int rs = a.rslots ? a.rslots[j] : -1;
+            if (rs >= 0) {
+                for (int e = threadIdx.x; e < a.Wslot; e += blockDim.x)
+                    a.slots[(long)rs * a.Wslot + e] = row[a.item_off + e];
+                if (threadIdx.x == 0) a.slot_ages[rs] = age;
+                __syncthreads();
+            }
+        }
+    }
+    if (threadIdx.x == 0) a.ages[node] = age;
+}
+
+// ---------------------------------------------------------------------------
+// pegasos / adaline tick: one WAVE per node, weights in registers
+// ---------------------------------------------------------------------------
 
 struct LinearArgs {
     float* params; int* ages;
@@ -876,19 +966,21 @@ static hipStream_t current_stream()
 
 void snapshot(torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
               torch::Tensor slot_ages, torch::Tensor nodes, torch::Tensor slot_ids,
-              int64_t age_width)
+              int64_t age_width, int64_t src_off)
 {
     CHECK_DEV(params); CHECK_DEV(slots); CHECK_DEV(nodes); CHECK_DEV(slot_ids);
     int n = nodes.size(0);
     if (n == 0) return;
-    int D = params.size(1);
-    long total = (long)n * D;
+    int W = slots.size(1);
+    int Dp = params.size(1);
+    long total = (long)n * W;
     int block = 256;
     int grid = (int)std::min<long>((total + block - 1) / block, 2048);
     hipLaunchKernelGGL(snapshot_kernel, dim3(grid), dim3(block), 0, current_stream(),
         params.data_ptr<float>(), ages.data_ptr<int>(),
         slots.data_ptr<float>(), slot_ages.data_ptr<int>(),
-        nodes.data_ptr<int>(), slot_ids.data_ptr<int>(), n, D, (int)age_width);
+        nodes.data_ptr<int>(), slot_ids.data_ptr<int>(), n, W, Dp,
+        (int)src_off, (int)age_width);
 }
 
 void wmerge(torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
@@ -1103,7 +1195,8 @@ static void launch_snap(const float* params, const int* ages, float* slots,
     int block = 256;
     int grid = (int)std::min<long>((total + block - 1) / block, 2048);
     hipLaunchKernelGGL(snapshot_kernel, dim3(grid), dim3(block), 0, s,
-                       params, ages, slots, slot_ages, nodes, slot_ids, n, D, A);
+                       params, ages, slots, slot_ages, nodes, slot_ids, n, D,
+                       D, 0, A);
 }
 
 void run_round_logreg(
@@ -1256,6 +1349,32 @@ void tick_logreg_samp(
                        current_stream(), a);
 }
 
+void tick_mf(torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
+             torch::Tensor slot_ages, torch::Tensor nodes, torch::Tensor recv_ptr,
+             torch::Tensor del_slots, torch::Tensor reply_slots,
+             torch::Tensor X, torch::Tensor Y, torch::Tensor counts,
+             int64_t k, int64_t n_items, double reg, double lr,
+             bool update_only)
+{
+    CHECK_DEV(params); CHECK_DEV(slots); CHECK_DEV(X); CHECK_DEV(Y);
+    TORCH_CHECK(k <= WAVE, "MF latent dim > ", WAVE, " unsupported");
+    int n = nodes.size(0);
+    if (n == 0) return;
+    MFArgs a;
+    a.params = params.data_ptr<float>(); a.ages = ages.data_ptr<int>();
+    a.slots = slots.data_ptr<float>(); a.slot_ages = slot_ages.data_ptr<int>();
+    a.nodes = nodes.data_ptr<int>(); a.ptr = recv_ptr.data_ptr<int>();
+    a.dslots = del_slots.numel() ? del_slots.data_ptr<int>() : nullptr;
+    a.rslots = reply_slots.numel() ? reply_slots.data_ptr<int>() : nullptr;
+    a.X = X.data_ptr<float>(); a.Y = Y.data_ptr<float>();
+    a.counts = counts.data_ptr<int>();
+    a.k = k; a.n_items = n_items; a.Smax = X.size(1); a.D = params.size(1);
+    a.item_off = k + 1; a.Wslot = n_items * (k + 1);
+    a.reg = reg; a.lr = lr; a.update_only = update_only;
+    hipLaunchKernelGGL(tick_mf_kernel, dim3(n), dim3(WAVE), 0,
+                       current_stream(), a);
+}
+
 void run_round_logreg_part(
     torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
     torch::Tensor slot_ages,
@@ -1346,4 +1465,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m)
     m.def("wmerge", &wmerge, "all2all weighted k-way merge (K5 weighted)");
     m.def("tick_logreg_samp", &tick_logreg_samp,
           "fused sampled-merge + logreg SGD tick (K6)");
+    m.def("tick_mf", &tick_mf,
+          "fused item-block merge + per-rating MF SGD tick (K9/K10)");
 }

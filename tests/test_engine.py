@@ -868,3 +868,131 @@ class TestSampledEngine:
         (gossipy/model/sampling.py:57)."""
         spec = LogRegSpec(d_in=57, n_classes=2, sample_size=0.25)
         assert spec.samp_count() == max(1, int(round(0.25 * 116)))
+
+
+# ---------------------------------------------------------------------------
+# matrix-factorization recommender (K9/K10)
+# ---------------------------------------------------------------------------
+
+
+def _mf_arena(n_users=20, n_items=50, rpu=30, device=torch.device("cpu"), seed=3):
+    from gossipy_amd.data import make_synthetic_recsys
+
+    ratings, _, _ = make_synthetic_recsys(n_users, n_items, rpu, seed=seed)
+    shards, tests = [], []
+    for u in range(n_users):
+        rs = ratings[u]
+        cut = max(1, int(0.8 * len(rs)))
+        tr, te = rs[:cut], rs[cut:]
+        shards.append(
+            (torch.tensor([[i] for i, _ in tr], dtype=torch.float32),
+             torch.tensor([r for _, r in tr]))
+        )
+        tests.append(
+            (torch.tensor([[i] for i, _ in te], dtype=torch.float32),
+             torch.tensor([r for _, r in te]))
+        )
+    return DataArena.from_shards(shards, device, test_shards=tests)
+
+
+class TestMFEngine:
+    def _spec(self, n_items=50):
+        from gossipy_amd.engine import MFSpec
+
+        return MFSpec(k=5, n_items=n_items, reg=0.1, lr=0.01)
+
+    def test_update_matches_object_layer(self):
+        """One per-rating SGD pass: engine oracle == MFModelHandler."""
+        from gossipy_amd.engine.arena import NodeStateArena
+        from gossipy_amd.engine.backend import TorchBackend
+        from gossipy_amd.model.handler import MFModelHandler
+
+        spec = self._spec()
+        h = MFModelHandler(dim=5, n_items=50, lam_reg=0.1, learning_rate=0.01)
+        h.init()
+        (X, b), (Y, c) = h.model
+
+        state = NodeStateArena(1, spec.D, torch.device("cpu"))
+        state.params[0, :5] = torch.from_numpy(X[0]).float()
+        state.params[0, 5] = b
+        state.params[0, 6 : 6 + 250] = torch.from_numpy(Y.reshape(-1)).float()
+        state.params[0, 256:] = torch.from_numpy(c).float()
+        state.ages[0] = h.n_updates
+
+        data = _mf_arena(n_users=1, n_items=50, rpu=25)
+        TorchBackend().update(state, data, spec, torch.tensor([0]))
+
+        ratings = [
+            (int(data.x[0, s, 0]), float(data.y[0, s]))
+            for s in range(int(data.counts[0]))
+        ]
+        h._update(ratings)
+        (X2, b2), (Y2, c2) = h.model
+        assert np.allclose(state.params[0, :5].numpy(), X2[0], atol=1e-5)
+        assert abs(float(state.params[0, 5]) - b2) < 1e-6
+        assert np.allclose(
+            state.params[0, 6:256].numpy(), Y2.reshape(-1), atol=1e-5
+        )
+        assert np.allclose(state.params[0, 256:].numpy(), c2, atol=1e-5)
+        assert int(state.ages[0]) == h.n_updates
+
+    def test_merge_matches_object_layer(self):
+        from gossipy_amd.engine.arena import NodeStateArena, SlotPool
+        from gossipy_amd.engine.backend import TorchBackend
+        from gossipy_amd.model.handler import MFModelHandler
+
+        spec = self._spec()
+        h1 = MFModelHandler(dim=5, n_items=50, lam_reg=0.1, learning_rate=0.01)
+        h2 = MFModelHandler(dim=5, n_items=50, lam_reg=0.1, learning_rate=0.01)
+        h1.init(); h2.init()
+        h1.n_updates, h2.n_updates = 7, 3
+
+        state = NodeStateArena(1, spec.D, torch.device("cpu"))
+        (X, b), (Y, c) = h1.model
+        state.params[0, :5] = torch.from_numpy(X[0]).float()
+        state.params[0, 5] = b
+        state.params[0, 6:256] = torch.from_numpy(Y.reshape(-1)).float()
+        state.params[0, 256:] = torch.from_numpy(c).float()
+        state.ages[0] = 7
+        pool = SlotPool(spec.slot_width, torch.device("cpu"), 2)
+        (_, _), (Y2, c2) = h2.model
+        pool.slots[1, :250] = torch.from_numpy(Y2.reshape(-1)).float()
+        pool.slots[1, 250:] = torch.from_numpy(c2).float()
+        pool.slot_ages[1] = 3
+
+        TorchBackend()._merge_mf(state, pool, spec, 0, 1)
+        h1._merge(h2)
+        (_, _), (Ym, cm) = h1.model
+        assert np.allclose(state.params[0, 6:256].numpy(), Ym.reshape(-1), atol=1e-6)
+        assert np.allclose(state.params[0, 256:].numpy(), cm, atol=1e-6)
+        assert int(state.ages[0]) == h1.n_updates  # merge leaves ages alone
+
+    def _run(self, rounds=10, n_users=20, **cfg_kw):
+        from gossipy_amd.engine import MFSpec
+
+        data = _mf_arena(n_users=n_users)
+        base = dict(
+            n_nodes=n_users, delta=10, protocol=AntiEntropyProtocol.PUSH,
+            model_size=300, sampling_eval=0.0, seed=23,
+        )
+        base.update(cfg_kw)
+        cfg = EngineConfig(**base)
+        sim = BatchedGossipSimulator(cfg, self._spec(), data)
+        rep = SimulationReport()
+        sim.add_receiver(rep)
+        sim.init_nodes()
+        sim.start(n_rounds=rounds)
+        return sim, rep
+
+    def test_rmse_improves(self):
+        sim, rep = self._run()
+        evals = rep.get_evaluation(True)
+        assert evals, "local RMSE evaluations must be reported"
+        first = evals[0][1]["rmse"]
+        last = evals[-1][1]["rmse"]
+        assert last < first
+
+    def test_deterministic(self):
+        s1, _ = self._run(rounds=3)
+        s2, _ = self._run(rounds=3)
+        assert torch.equal(s1.local_params(), s2.local_params())

@@ -23,6 +23,7 @@ from gossipy_amd.engine import (
 )
 from gossipy_amd.engine.backend import HIPBackend, TorchBackend
 from gossipy_amd.simul import SimulationReport
+from gossipy_amd.data import make_synthetic_classification
 
 CUDA = torch.device("cuda:0")
 CPU = torch.device("cpu")
@@ -519,3 +520,58 @@ class TestSampledGPU:
         sim.start(n_rounds=15)
         torch.cuda.synchronize()
         assert rep.get_evaluation(False)[-1][1]["accuracy"] > 0.9
+
+
+class TestMFGPU:
+    def test_mf_deliver_matches_oracle(self):
+        from gossipy_amd.engine import MFSpec
+        from tests.test_engine import _mf_arena
+
+        spec = MFSpec(k=5, n_items=50, reg=0.1, lr=0.01)
+        cd = _mf_arena(n_users=12, n_items=50, rpu=25)
+        gd = DataArena(
+            cd.x.to(CUDA), cd.y.to(CUDA), cd.counts.to(CUDA),
+            tx=cd.tx.to(CUDA), ty=cd.ty.to(CUDA), tcounts=cd.tcounts.to(CUDA),
+        )
+        cs = NodeStateArena(12, spec.D, CPU)
+        tape = RandomTape(5)
+        TorchBackend().init_params(cs, spec, tape)
+        gs = NodeStateArena(12, spec.D, CUDA)
+        gs.params.copy_(cs.params)
+        gs.ages.copy_(cs.ages)
+        cpool = SlotPool(spec.slot_width, CPU, 8)
+        gpool = SlotPool(spec.slot_width, CUDA, 8)
+        cpool.slots.uniform_(0, 1, generator=torch.Generator().manual_seed(2))
+        cpool.slot_ages.copy_(torch.arange(8, dtype=torch.int32) + 1)
+        gpool.slots.copy_(cpool.slots)
+        gpool.slot_ages.copy_(cpool.slot_ages)
+        recv = torch.tensor([2, 7], dtype=torch.int64)
+        ptr = torch.tensor([0, 2, 3], dtype=torch.int64)
+        slots = torch.tensor([1, 4, 6], dtype=torch.int64)
+        reply = torch.tensor([7, -1, -1], dtype=torch.int64)
+        TorchBackend().deliver(cs, cpool, cd, spec, recv, ptr, slots, reply)
+        HIPBackend().deliver(gs, gpool, gd, spec, recv, ptr, slots, reply)
+        torch.cuda.synchronize()
+        assert _close(cs.params, gs.params, 1e-4)
+        assert torch.equal(cs.ages.cpu(), gs.ages.cpu())
+        assert _close(cpool.slots[7], gpool.slots[7], 1e-4)
+
+    def test_mf_gpu_rmse_improves(self):
+        from gossipy_amd.engine import MFSpec
+        from tests.test_engine import _mf_arena
+
+        data = _mf_arena(n_users=20, device=CUDA)
+        cfg = EngineConfig(
+            n_nodes=20, delta=10, protocol=AntiEntropyProtocol.PUSH,
+            model_size=300, sampling_eval=0.0, seed=23,
+        )
+        sim = BatchedGossipSimulator(
+            cfg, MFSpec(k=5, n_items=50, reg=0.1, lr=0.01), data, device=CUDA
+        )
+        rep = SimulationReport()
+        sim.add_receiver(rep)
+        sim.init_nodes()
+        sim.start(n_rounds=10)
+        torch.cuda.synchronize()
+        evals = rep.get_evaluation(True)
+        assert evals[-1][1]["rmse"] < evals[0][1]["rmse"]
