@@ -11,7 +11,7 @@ the reference.
 
 from .arena import DataArena, NodeStateArena, SlotPool
 from .backend import HIPBackend, TorchBackend, make_backend
-from .models import AdaLineSpec, LogRegSpec, MFSpec, MLPSpec, PegasosSpec
+from .models import AdaLineSpec, KMeansSpec, LogRegSpec, MFSpec, MLPSpec, PegasosSpec
 from .rng import Purpose, RandomTape
 from .runner import (
     BatchedAll2AllGossipSimulator,
@@ -49,4 +49,5 @@ __all__ = [
     "LogRegSpec",
     "MLPSpec",
     "MFSpec",
+    "KMeansSpec",
 ]

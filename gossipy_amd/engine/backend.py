@@ -92,6 +92,17 @@ class TorchBackend:
         if spec.family in ("pegasos", "adaline"):
             state.params.zero_()
             return
+        if spec.family == "kmeans":
+            # KMeansHandler.init = torch.rand(k, dim)
+            # (gossipy/model/handler.py:594-595), tape-driven per node
+            n = state.params.shape[0]
+            rows = np.empty((n, spec.D), dtype=np.float32)
+            for li in range(n):
+                g = tape.stream(Purpose.INIT, t=state.node_lo + li)
+                rows[li] = g.uniform(0, 1, size=spec.D)
+            state.params.copy_(torch.from_numpy(rows).to(state.params.device))
+            state.ages.zero_()
+            return
         if spec.family == "mf":
             # MFModelHandler.init (gossipy/model/handler.py:542-548):
             # X, Y ~ U(0,1)*sqrt((r_max-r_min)/k); b = c = r_min/2; age = 1.
@@ -173,6 +184,8 @@ class TorchBackend:
             self._update_part(state.params, state.ages, data, spec, nodes.long())
         elif spec.family == "mf":
             self._update_mf(state.params, state.ages, data, spec, nodes.long())
+        elif spec.family == "kmeans":
+            self._update_kmeans(state.params, state.ages, data, spec, nodes.long())
         elif spec.family == "logreg":
             self._update_logreg(state.params, state.ages, data, spec, nodes.long())
         elif spec.family == "mlp":
@@ -503,6 +516,74 @@ class TorchBackend:
             )
         return out
 
+    # -- k-means (K11/K12) ---------------------------------------------------
+
+    def _update_kmeans(self, params, ages, data, spec, nodes) -> None:
+        """Assign + EMA with last-write-wins duplicates
+        (KMeansHandler._update, gossipy/model/handler.py:608-615)."""
+        for idx in nodes.tolist():
+            c_n = int(data.counts[idx])
+            if c_n == 0:
+                continue
+            C = params[idx].view(spec.k, spec.dim)
+            x = data.x[idx, :c_n]
+            assign = torch.argmin(torch.cdist(x, C, p=2), dim=1)
+            C[assign] = C[assign] * (1 - spec.alpha) + spec.alpha * x
+            ages[idx] += 1
+
+    def _merge_kmeans(self, state, pool, spec, node: int, slot: int) -> None:
+        """Naive mean or (bug-fixed) Hungarian-matched mean
+        (gossipy/model/handler.py:617-630); ages untouched."""
+        C = state.params[node].view(spec.k, spec.dim)
+        other = pool.slots[slot].view(spec.k, spec.dim)
+        if spec.matching == "hungarian":
+            from scipy.optimize import linear_sum_assignment
+
+            cost = torch.cdist(C, other).cpu().numpy()
+            perm = linear_sum_assignment(cost)[1]
+            other = other[torch.from_numpy(perm).to(other.device)]
+        state.params[node] = ((C + other) / 2).reshape(-1)
+
+    def _deliver_kmeans(
+        self, state, pool, data, spec, recv_nodes, recv_ptr, del_slots,
+        reply_slots,
+    ) -> None:
+        mode = spec.mode
+        ptr = recv_ptr.tolist()
+        for i, node_t in enumerate(recv_nodes.tolist()):
+            node = torch.tensor([node_t])
+            for j in range(ptr[i], ptr[i + 1]):
+                slot = int(del_slots[j])
+                if mode == CreateModelMode.MERGE_UPDATE:
+                    self._merge_kmeans(state, pool, spec, node_t, slot)
+                    self.update(state, data, spec, node)
+                elif mode == CreateModelMode.UPDATE:
+                    self._adopt_slot(state, pool, node_t, slot)
+                    self.update(state, data, spec, node)
+                elif mode == CreateModelMode.UPDATE_MERGE:
+                    self.update(state, data, spec, node)
+                    self._train_slot(state, pool, data, spec, node_t, slot)
+                    self._merge_kmeans(state, pool, spec, node_t, slot)
+                elif mode == CreateModelMode.PASS:
+                    self._adopt_slot(state, pool, node_t, slot)
+                else:
+                    raise ValueError(mode)
+                r = int(reply_slots[j])
+                if r >= 0:
+                    self.snapshot(
+                        state,
+                        pool,
+                        torch.tensor([node_t], dtype=torch.long),
+                        torch.tensor([r], dtype=torch.long),
+                    )
+
+    def kmeans_assign(self, state, spec, nodes: torch.Tensor, X) -> torch.Tensor:
+        """Cluster assignments ``[len(nodes), len(X)]`` for NMI evaluation
+        (gossipy/model/handler.py:632-636)."""
+        C = state.params[nodes.long()].view(-1, spec.k, spec.dim)
+        d = torch.cdist(X.unsqueeze(0).expand(C.shape[0], -1, -1), C, p=2)
+        return torch.argmin(d, dim=2)
+
     # -- deliveries ----------------------------------------------------------
 
     def deliver(
@@ -534,6 +615,12 @@ class TorchBackend:
             return
         if spec.family == "mf":
             self._deliver_mf(
+                state, pool, data, spec, recv_nodes, recv_ptr, del_slots,
+                reply_slots,
+            )
+            return
+        if spec.family == "kmeans":
+            self._deliver_kmeans(
                 state, pool, data, spec, recv_nodes, recv_ptr, del_slots,
                 reply_slots,
             )
@@ -822,6 +909,31 @@ class HIPBackend(TorchBackend):
                 spec.n_items,
                 spec.reg,
                 spec.lr,
+                bool(update_only),
+            )
+        elif spec.family == "kmeans":
+            if not update_only and spec.matching != "naive":
+                raise ValueError(
+                    "kmeans HIP path implements naive matching (the"
+                    " reference's hungarian path is an identity no-op,"
+                    " gossipy/model/handler.py:629-630); use the torch"
+                    " backend for the bug-fixed hungarian merge"
+                )
+            self.ext.tick_kmeans(
+                state.params,
+                state.ages,
+                slots,
+                slot_ages,
+                nodes,
+                recv_ptr,
+                del_slots,
+                reply_slots,
+                data.x,
+                data.counts,
+                spec.k,
+                spec.dim,
+                spec.alpha,
+                mode,
                 bool(update_only),
             )
         elif getattr(spec, "sample_size", 0) > 0:

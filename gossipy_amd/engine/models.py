@@ -27,7 +27,7 @@ import numpy as np
 
 from ..core import CreateModelMode
 
-__all__ = ["PegasosSpec", "AdaLineSpec", "LogRegSpec", "MLPSpec", "MFSpec"]
+__all__ = ["PegasosSpec", "AdaLineSpec", "LogRegSpec", "MLPSpec", "MFSpec", "KMeansSpec"]
 
 
 def _fortran_to_arena(layers: List[Tuple[int, int, int, int]], D: int) -> np.ndarray:
@@ -259,3 +259,38 @@ class MFSpec:
     @property
     def slot_width(self) -> int:
         return self.n_items * self.k + self.n_items
+
+
+@dataclass
+class KMeansSpec:
+    """Online k-means with EMA centroid updates (KMeansHandler,
+    gossipy/model/handler.py:579-639).
+
+    Arena row layout: centroids ``C`` (k*dim, row-major). Quirk parity: the
+    per-call update assigns every sample to its nearest centroid and applies
+    ``C[idx] = (1-alpha)*C[idx] + alpha*x`` with torch indexed-assignment
+    semantics — duplicate assignments collapse to the LAST sample per
+    centroid (handler.py:613-615); ``n_updates`` bumps once per update call,
+    and merges never touch it.
+
+    ``matching="hungarian"`` (bug-fixed column-index matching, see
+    gossipy_amd/model/handler.py) runs on the torch backend only — the
+    reference's own hungarian path is a no-op identity permutation
+    (handler.py:629-630), so the HIP kernel implements ``naive``, the
+    behavior the reference actually ships.
+    """
+
+    k: int
+    dim: int
+    alpha: float = 0.1
+    matching: str = "naive"
+    mode: CreateModelMode = CreateModelMode.UPDATE
+
+    family = "kmeans"
+    n_parts = 0
+    sample_size = 0.0
+    age_width = 1
+
+    @property
+    def D(self) -> int:
+        return self.k * self.dim

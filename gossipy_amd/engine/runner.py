@@ -308,6 +308,30 @@ class BatchedGossipSimulator(SimulationEventSender):
         mine = nodes[self._is_mine(nodes)]
         local_ids = torch.from_numpy((mine - self.node_lo).astype(np.int64))
 
+        if self.spec.family == "kmeans":
+            # NMI of each node's clustering of the global eval set
+            # (gossipy/model/handler.py:632-636; ClusteringDataHandler's
+            # eval set IS the train set, gossipy/data/handler.py:156-161)
+            results = []
+            if self.data.gx is not None and len(mine):
+                from sklearn.metrics import normalized_mutual_info_score as nmi
+
+                assign = self.backend.kmeans_assign(
+                    self.state, self.spec, local_ids, self.data.gx
+                ).cpu().numpy()
+                y_true = self.data.gy.cpu().numpy()
+                results = [
+                    {"nmi": float(nmi(y_true, assign[r]))}
+                    for r in range(assign.shape[0])
+                ]
+            if self.world > 1:
+                gathered = [None] * self.world
+                dist.all_gather_object(gathered, results)
+                results = [d for part in gathered for d in part]
+            if self.rank == 0 and results:
+                self.notify_evaluation(t, False, results)
+            return
+
         if self.spec.family == "mf":
             # RecSys eval: per-user RMSE on the held-out ratings; there is
             # no global test set (gossipy/data/__init__.py:550-555)

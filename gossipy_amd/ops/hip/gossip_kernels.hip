@@ -643,6 +643,129 @@ tick_mf_kernel(MFArgs a)
 }
 
 // ---------------------------------------------------------------------------
+// k-means tick (K11, KMeansHandler semantics, naive matching)
+//
+// Parity: gossipy/model/handler.py:579-639. Per update call: every sample
+// assigns to its nearest centroid; the EMA applies with torch
+// indexed-assignment semantics — per centroid only the LAST assigned
+// sample lands (handler.py:613-615); n_updates bumps once per call and is
+// never merged. Centroids live in LDS.
+// ---------------------------------------------------------------------------
+
+struct KMeansArgs {
+    float* params; int* ages;
+    float* slots; int* slot_ages;
+    const int* nodes; const int* ptr;
+    const int* dslots; const int* rslots;
+    const float* X; const int* counts;
+    int k, dim, Smax, D;
+    float alpha;
+    int mode, update_only;
+};
+
+DEV_INLINE void kmeans_update(const KMeansArgs& a, int node, float* C,
+                              int* assign, int& age)
+{
+    int tid = threadIdx.x;
+    int c = a.counts[node];
+    if (c == 0) return;
+    const float* Xn = a.X + (long)node * a.Smax * a.dim;
+    // assignment: thread = sample
+    for (int s = tid; s < c; s += blockDim.x) {
+        const float* x = Xn + (long)s * a.dim;
+        float best = 1e30f;
+        int bj = 0;
+        for (int j = 0; j < a.k; ++j) {
+            const float* cj = C + j * a.dim;
+            float d2 = 0.f;
+            for (int e = 0; e < a.dim; ++e) {
+                float t = x[e] - cj[e];
+                d2 += t * t;
+            }
+            if (d2 < best) { best = d2; bj = j; }
+        }
+        assign[s] = bj;
+    }
+    __syncthreads();
+    // EMA, last-write-wins: thread = centroid, scan for its last sample
+    for (int j = tid; j < a.k; j += blockDim.x) {
+        int last = -1;
+        for (int s = 0; s < c; ++s)
+            if (assign[s] == j) last = s;
+        if (last >= 0) {
+            const float* x = Xn + (long)last * a.dim;
+            float* cj = C + j * a.dim;
+            for (int e = 0; e < a.dim; ++e)
+                cj[e] = cj[e] * (1.0f - a.alpha) + a.alpha * x[e];
+        }
+    }
+    __syncthreads();
+    age += 1;
+}
+
+__global__ void __launch_bounds__(128)
+tick_kmeans_kernel(KMeansArgs a)
+{
+    int i = blockIdx.x;
+    int node = a.nodes[i];
+    int tid = threadIdx.x;
+    extern __shared__ float sm[];
+    float* C = sm;                       // D = k*dim
+    float* C2 = C + a.D;                 // received-model scratch
+    int* assign = (int*)(C2 + a.D);      // Smax
+
+    for (int e = tid; e < a.D; e += blockDim.x)
+        C[e] = a.params[(long)node * a.D + e];
+    __syncthreads();
+    int age = a.ages[node];
+
+    if (a.update_only) {
+        kmeans_update(a, node, C, assign, age);
+    } else {
+        for (int j = a.ptr[i]; j < a.ptr[i + 1]; ++j) {
+            int slot = a.dslots[j];
+            const float* srow = a.slots + (long)slot * a.D;
+            int sage = a.slot_ages[slot];
+            if (a.mode == MODE_MERGE_UPDATE) {
+                for (int e = tid; e < a.D; e += blockDim.x)
+                    C[e] = 0.5f * (C[e] + srow[e]);
+                __syncthreads();
+                kmeans_update(a, node, C, assign, age);
+            } else if (a.mode == MODE_UPDATE) {
+                for (int e = tid; e < a.D; e += blockDim.x) C[e] = srow[e];
+                age = sage;
+                __syncthreads();
+                kmeans_update(a, node, C, assign, age);
+            } else if (a.mode == MODE_UPDATE_MERGE) {
+                kmeans_update(a, node, C, assign, age);
+                for (int e = tid; e < a.D; e += blockDim.x) C2[e] = srow[e];
+                __syncthreads();
+                int age2 = sage;
+                kmeans_update(a, node, C2, assign, age2);
+                for (int e = tid; e < a.D; e += blockDim.x)
+                    C[e] = 0.5f * (C[e] + C2[e]);
+                __syncthreads();
+            } else {  // PASS
+                for (int e = tid; e < a.D; e += blockDim.x) C[e] = srow[e];
+                age = sage;
+                __syncthreads();
+            }
+            int rs = a.rslots ? a.rslots[j] : -1;
+            if (rs >= 0) {
+                for (int e = tid; e < a.D; e += blockDim.x)
+                    a.slots[(long)rs * a.D + e] = C[e];
+                if (tid == 0) a.slot_ages[rs] = age;
+                __syncthreads();
+            }
+        }
+    }
+    __syncthreads();
+    for (int e = tid; e < a.D; e += blockDim.x)
+        a.params[(long)node * a.D + e] = C[e];
+    if (tid == 0) a.ages[node] = age;
+}
+
+// ---------------------------------------------------------------------------
 // pegasos / adaline tick: one WAVE per node, weights in registers
 // ---------------------------------------------------------------------------
 
@@ -1375,6 +1498,31 @@ void tick_mf(torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
                        current_stream(), a);
 }
 
+void tick_kmeans(torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
+                 torch::Tensor slot_ages, torch::Tensor nodes,
+                 torch::Tensor recv_ptr, torch::Tensor del_slots,
+                 torch::Tensor reply_slots, torch::Tensor X,
+                 torch::Tensor counts, int64_t k, int64_t dim, double alpha,
+                 int64_t mode, bool update_only)
+{
+    CHECK_DEV(params); CHECK_DEV(slots); CHECK_DEV(X);
+    int n = nodes.size(0);
+    if (n == 0) return;
+    KMeansArgs a;
+    a.params = params.data_ptr<float>(); a.ages = ages.data_ptr<int>();
+    a.slots = slots.data_ptr<float>(); a.slot_ages = slot_ages.data_ptr<int>();
+    a.nodes = nodes.data_ptr<int>(); a.ptr = recv_ptr.data_ptr<int>();
+    a.dslots = del_slots.numel() ? del_slots.data_ptr<int>() : nullptr;
+    a.rslots = reply_slots.numel() ? reply_slots.data_ptr<int>() : nullptr;
+    a.X = X.data_ptr<float>(); a.counts = counts.data_ptr<int>();
+    a.k = k; a.dim = dim; a.Smax = X.size(1); a.D = params.size(1);
+    a.alpha = alpha; a.mode = mode; a.update_only = update_only;
+    size_t smem = sizeof(float) * 2 * a.D + sizeof(int) * a.Smax;
+    TORCH_CHECK(smem <= 160 * 1024, "kmeans LDS budget exceeded: ", smem);
+    hipLaunchKernelGGL(tick_kmeans_kernel, dim3(n), dim3(128), smem,
+                       current_stream(), a);
+}
+
 void run_round_logreg_part(
     torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
     torch::Tensor slot_ages,
@@ -1467,4 +1615,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m)
           "fused sampled-merge + logreg SGD tick (K6)");
     m.def("tick_mf", &tick_mf,
           "fused item-block merge + per-rating MF SGD tick (K9/K10)");
+    m.def("tick_kmeans", &tick_kmeans,
+          "fused centroid merge + assign/EMA k-means tick (K11)");
 }

@@ -996,3 +996,94 @@ class TestMFEngine:
         s1, _ = self._run(rounds=3)
         s2, _ = self._run(rounds=3)
         assert torch.equal(s1.local_params(), s2.local_params())
+
+
+# ---------------------------------------------------------------------------
+# k-means clustering (K11/K12)
+# ---------------------------------------------------------------------------
+
+
+def _blob_arena(n_nodes=16, k=3, dim=8, per_node=30, device=torch.device("cpu"), seed=4):
+    rng = np.random.default_rng(seed)
+    centers = rng.normal(0, 4, size=(k, dim))
+    shards, allx, ally = [], [], []
+    for _ in range(n_nodes):
+        labels = rng.integers(0, k, size=per_node)
+        x = centers[labels] + rng.normal(0, 0.4, size=(per_node, dim))
+        shards.append(
+            (torch.from_numpy(x).float(), torch.from_numpy(labels).float())
+        )
+        allx.append(x); ally.append(labels)
+    geval = (
+        torch.from_numpy(np.concatenate(allx)).float(),
+        torch.from_numpy(np.concatenate(ally)).float(),
+    )
+    return DataArena.from_shards(shards, device, global_eval=geval)
+
+
+class TestKMeansEngine:
+    def _spec(self, matching="naive"):
+        from gossipy_amd.engine import KMeansSpec
+
+        return KMeansSpec(k=3, dim=8, alpha=0.1, matching=matching)
+
+    def test_update_matches_object_layer(self):
+        from gossipy_amd.engine.arena import NodeStateArena
+        from gossipy_amd.engine.backend import TorchBackend
+        from gossipy_amd.model.handler import KMeansHandler
+
+        spec = self._spec()
+        h = KMeansHandler(k=3, dim=8, alpha=0.1)
+        h.init()
+        state = NodeStateArena(1, spec.D, torch.device("cpu"))
+        state.params[0] = h.model.reshape(-1)
+        data = _blob_arena(n_nodes=1)
+        TorchBackend().update(state, data, spec, torch.tensor([0]))
+        c_n = int(data.counts[0])
+        h._update((data.x[0, :c_n], None))
+        assert torch.allclose(
+            state.params[0].view(3, 8), h.model, atol=1e-6
+        )
+        assert int(state.ages[0]) == h.n_updates
+
+    def test_merge_matches_object_layer(self):
+        from gossipy_amd.engine.arena import NodeStateArena, SlotPool
+        from gossipy_amd.engine.backend import TorchBackend
+        from gossipy_amd.model.handler import KMeansHandler
+
+        for matching in ("naive", "hungarian"):
+            spec = self._spec(matching)
+            h1 = KMeansHandler(k=3, dim=8, alpha=0.1, matching=matching)
+            h2 = KMeansHandler(k=3, dim=8, alpha=0.1, matching=matching)
+            torch.manual_seed(0)
+            h1.init(); h2.init()
+            state = NodeStateArena(1, spec.D, torch.device("cpu"))
+            state.params[0] = h1.model.reshape(-1)
+            pool = SlotPool(spec.D, torch.device("cpu"), 2)
+            pool.slots[0] = h2.model.reshape(-1)
+            TorchBackend()._merge_kmeans(state, pool, spec, 0, 0)
+            h1._merge(h2)
+            assert torch.allclose(
+                state.params[0].view(3, 8), h1.model, atol=1e-6
+            ), matching
+
+    def test_engine_clusters(self):
+        from gossipy_amd.engine import KMeansSpec
+
+        data = _blob_arena()
+        cfg = EngineConfig(
+            n_nodes=16, delta=10, protocol=AntiEntropyProtocol.PUSH,
+            model_size=24, sampling_eval=0.0, seed=29,
+        )
+        from gossipy_amd.core import CreateModelMode
+
+        spec = KMeansSpec(
+            k=3, dim=8, alpha=0.1, mode=CreateModelMode.MERGE_UPDATE
+        )
+        sim = BatchedGossipSimulator(cfg, spec, data)
+        rep = SimulationReport()
+        sim.add_receiver(rep)
+        sim.init_nodes()
+        sim.start(n_rounds=10)
+        evals = rep.get_evaluation(False)
+        assert evals[-1][1]["nmi"] > 0.6

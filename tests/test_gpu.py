@@ -575,3 +575,56 @@ class TestMFGPU:
         torch.cuda.synchronize()
         evals = rep.get_evaluation(True)
         assert evals[-1][1]["rmse"] < evals[0][1]["rmse"]
+
+
+class TestKMeansGPU:
+    def test_kmeans_deliver_matches_oracle(self):
+        from gossipy_amd.engine import KMeansSpec
+        from tests.test_engine import _blob_arena
+
+        spec = KMeansSpec(
+            k=3, dim=8, alpha=0.1, mode=CreateModelMode.MERGE_UPDATE
+        )
+        cd = _blob_arena(n_nodes=10)
+        gd = DataArena(
+            cd.x.to(CUDA), cd.y.to(CUDA), cd.counts.to(CUDA),
+            gx=cd.gx.to(CUDA), gy=cd.gy.to(CUDA),
+        )
+        cs = NodeStateArena(10, spec.D, CPU)
+        TorchBackend().init_params(cs, spec, RandomTape(5))
+        gs = NodeStateArena(10, spec.D, CUDA)
+        gs.params.copy_(cs.params)
+        gs.ages.copy_(cs.ages)
+        cpool, gpool = SlotPool(spec.D, CPU, 6), SlotPool(spec.D, CUDA, 6)
+        cpool.slots.uniform_(0, 1, generator=torch.Generator().manual_seed(3))
+        gpool.slots.copy_(cpool.slots)
+        recv = torch.tensor([1, 6], dtype=torch.int64)
+        ptr = torch.tensor([0, 2, 3], dtype=torch.int64)
+        slots = torch.tensor([0, 2, 4], dtype=torch.int64)
+        reply = torch.tensor([5, -1, -1], dtype=torch.int64)
+        TorchBackend().deliver(cs, cpool, cd, spec, recv, ptr, slots, reply)
+        HIPBackend().deliver(gs, gpool, gd, spec, recv, ptr, slots, reply)
+        torch.cuda.synchronize()
+        assert _close(cs.params, gs.params, 1e-4)
+        assert torch.equal(cs.ages.cpu(), gs.ages.cpu())
+
+    def test_kmeans_gpu_clusters(self):
+        from gossipy_amd.engine import KMeansSpec
+        from tests.test_engine import _blob_arena
+
+        data = _blob_arena(device=CUDA)
+        cfg = EngineConfig(
+            n_nodes=16, delta=10, protocol=AntiEntropyProtocol.PUSH,
+            model_size=24, sampling_eval=0.0, seed=29,
+        )
+        sim = BatchedGossipSimulator(
+            cfg,
+            KMeansSpec(k=3, dim=8, alpha=0.1, mode=CreateModelMode.MERGE_UPDATE),
+            data, device=CUDA,
+        )
+        rep = SimulationReport()
+        sim.add_receiver(rep)
+        sim.init_nodes()
+        sim.start(n_rounds=10)
+        torch.cuda.synchronize()
+        assert rep.get_evaluation(False)[-1][1]["nmi"] > 0.6
