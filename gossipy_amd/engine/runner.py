@@ -394,7 +394,7 @@ class BatchedGossipSimulator(SimulationEventSender):
             self.world == 1
             and isinstance(self.scheduler, NativeSchedulerAdapter)
             and getattr(self.backend, "ext", None) is not None
-            and self.spec.family in ("logreg", "pegasos", "adaline")
+            and self.spec.family in ("logreg", "pegasos", "adaline", "mlp")
             and (self.spec.family == "logreg" or getattr(self.spec, "n_parts", 0) == 0)
             and getattr(self.spec, "sample_size", 0) == 0
         )
@@ -485,6 +485,22 @@ class BatchedGossipSimulator(SimulationEventSender):
                 *common,
                 spec.d_in,
                 spec.n_classes,
+                spec.lr,
+                spec.weight_decay,
+                max(1, spec.local_epochs),
+                spec.batch_size,
+                _MODE_ID[spec.mode],
+            )
+        elif spec.family == "mlp":
+            layout = torch.tensor(
+                [x for tup in spec.layer_offsets() for x in tup],
+                dtype=torch.int32,
+                device=dev,
+            )
+            ext.run_round_mlp(
+                *common,
+                layout,
+                len(spec.layer_offsets()),
                 spec.lr,
                 spec.weight_decay,
                 max(1, spec.local_epochs),

@@ -1596,6 +1596,79 @@ void run_round_logreg_part(
     }
 }
 
+void run_round_mlp(
+    torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
+    torch::Tensor slot_ages,
+    torch::Tensor snap_nodes, torch::Tensor snap_slots, torch::Tensor snap_tptr,
+    torch::Tensor recv_nodes, torch::Tensor recv_nptr, torch::Tensor recv_tptr,
+    torch::Tensor del_slots, torch::Tensor reply_slots,
+    torch::Tensor pull_nodes, torch::Tensor pull_slots, torch::Tensor pull_tptr,
+    torch::Tensor rep_nodes, torch::Tensor rep_nptr, torch::Tensor rep_tptr,
+    torch::Tensor rep_slots,
+    torch::Tensor X, torch::Tensor Y, torch::Tensor counts,
+    torch::Tensor layout, int64_t n_layers, double lr, double wd,
+    int64_t epochs, int64_t bs, int64_t mode)
+{
+    CHECK_DEV(params); CHECK_DEV(slots); CHECK_DEV(X); CHECK_DEV(layout);
+    RoundArrays r = unpack_round(snap_nodes, snap_slots, snap_tptr, recv_nodes,
+                                 recv_nptr, recv_tptr, del_slots, reply_slots,
+                                 pull_nodes, pull_slots, pull_tptr, rep_nodes,
+                                 rep_nptr, rep_tptr, rep_slots);
+    hipStream_t s = current_stream();
+    MlpArgs a;
+    a.params = params.data_ptr<float>(); a.ages = ages.data_ptr<int>();
+    a.slots = slots.data_ptr<float>(); a.slot_ages = slot_ages.data_ptr<int>();
+    a.X = X.data_ptr<float>(); a.Y = Y.data_ptr<float>();
+    a.counts = counts.data_ptr<int>();
+    a.layout = layout.data_ptr<int>();
+    a.n_layers = n_layers; a.Smax = X.size(1); a.D = params.size(1);
+    a.lr = lr; a.wd = wd; a.epochs = epochs; a.bs = bs; a.mode = mode;
+    a.update_only = 0;
+    auto lay = layout.cpu();
+    const int* L = lay.data_ptr<int>();
+    int act_sum = 0, act_max = L[2];
+    for (int l = 0; l < n_layers; ++l) {
+        act_sum += L[4 * l + 3];
+        act_max = std::max(act_max, L[4 * l + 3]);
+        act_max = std::max(act_max, L[4 * l + 2]);
+    }
+    a.act_max = act_max;
+    a.d_in = L[2];
+    int bsmax = (bs == 0) ? a.Smax : std::min<int>(bs, a.Smax);
+    size_t smem = sizeof(float) *
+        (2 * (size_t)a.D + (size_t)bsmax * (a.d_in + act_sum) +
+         2 * (size_t)bsmax * act_max);
+    TORCH_CHECK(smem <= 160 * 1024, "mlp LDS budget exceeded");
+    for (int t = 0; t < r.delta; ++t) {
+        int s0 = r.snap_tptr[t], s1 = r.snap_tptr[t + 1];
+        if (s1 > s0)
+            launch_snap(a.params, a.ages, a.slots, a.slot_ages,
+                        r.snap_nodes + s0, r.snap_slots + s0, s1 - s0, a.D, s);
+        int r0 = r.recv_tptr[t], r1 = r.recv_tptr[t + 1];
+        if (r1 > r0) {
+            a.nodes = r.recv_nodes + r0;
+            a.ptr = r.recv_nptr + r0;
+            a.dslots = r.del_slots;
+            a.rslots = r.reply_slots;
+            hipLaunchKernelGGL(tick_mlp_kernel, dim3(r1 - r0), dim3(256),
+                               smem, s, a);
+        }
+        int p0 = r.pull_tptr[t], p1 = r.pull_tptr[t + 1];
+        if (p1 > p0)
+            launch_snap(a.params, a.ages, a.slots, a.slot_ages,
+                        r.pull_nodes + p0, r.pull_slots + p0, p1 - p0, a.D, s);
+        int q0 = r.rep_tptr[t], q1 = r.rep_tptr[t + 1];
+        if (q1 > q0) {
+            a.nodes = r.rep_nodes + q0;
+            a.ptr = r.rep_nptr + q0;
+            a.dslots = r.rep_slots;
+            a.rslots = nullptr;
+            hipLaunchKernelGGL(tick_mlp_kernel, dim3(q1 - q0), dim3(256),
+                               smem, s, a);
+        }
+    }
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m)
 {
     m.def("snapshot", &snapshot, "batched model snapshot (arena row copy)");
@@ -1615,6 +1688,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m)
           "fused sampled-merge + logreg SGD tick (K6)");
     m.def("tick_mf", &tick_mf,
           "fused item-block merge + per-rating MF SGD tick (K9/K10)");
+    m.def("run_round_mlp", &run_round_mlp,
+          "whole-round executor, MLP family");
     m.def("tick_kmeans", &tick_kmeans,
           "fused centroid merge + assign/EMA k-means tick (K11)");
 }

@@ -628,3 +628,32 @@ class TestKMeansGPU:
         sim.start(n_rounds=10)
         torch.cuda.synchronize()
         assert rep.get_evaluation(False)[-1][1]["nmi"] > 0.6
+
+
+class TestMLPFastPath:
+    def test_mlp_fast_matches_tick_path(self):
+        X, y = make_synthetic_classification((480, 20, 3), seed=0, margin=2.0)
+        idx = np.random.default_rng(0).permutation(480)
+        shards = [(X[s], y[s]) for s in np.array_split(idx[:432], 48)]
+        data = DataArena.from_shards(
+            shards, CUDA, global_eval=(X[idx[432:]], y[idx[432:]])
+        )
+        spec = MLPSpec(d_in=20, n_classes=3, hidden=(32,), lr=0.05)
+        cfg = EngineConfig(
+            n_nodes=48, delta=10, protocol=AntiEntropyProtocol.PUSH_PULL,
+            model_size=spec.D, sampling_eval=0.0, seed=3,
+        )
+        fast = BatchedGossipSimulator(cfg, spec, data, device=CUDA)
+        assert fast._fast_path_ok()
+        fast.init_nodes()
+        fast.start(n_rounds=3)
+
+        slow = BatchedGossipSimulator(cfg, spec, data, device=CUDA)
+        slow._fast_path_ok = lambda: False
+        slow.init_nodes()
+        slow.start(n_rounds=3)
+        torch.cuda.synchronize()
+        assert torch.allclose(
+            fast.local_params(), slow.local_params(), atol=1e-4, rtol=1e-4
+        )
+        assert torch.equal(fast.state.ages, slow.state.ages)
