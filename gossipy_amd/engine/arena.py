@@ -84,10 +84,15 @@ class SlotPool:
     def ensure(self, n_slots: int) -> None:
         if n_slots > self.slots.shape[0]:
             cap = max(n_slots, 2 * self.slots.shape[0])
+            old_s, old_a = self.slots, self.slot_ages
             self.slots = torch.zeros(
                 cap, self.D, device=self.device, dtype=torch.float32
             )
             self.slot_ages = self._new_ages(cap)
+            # slots of in-flight (delayed) messages survive the growth;
+            # the copy is stream-ordered after any kernel still reading them
+            self.slots[: old_s.shape[0]].copy_(old_s)
+            self.slot_ages[: old_a.shape[0]].copy_(old_a)
 
 
 class DataArena:

@@ -349,10 +349,15 @@ class BatchedGossipSimulator(SimulationEventSender):
         results_global: List[dict] = []
         if self.data.gx is not None and len(mine):
             scores = self.backend.scores(self.state, self.spec, local_ids, self.data.gx)
+            gy = self.data.gy
+            if scores.numel() < (1 << 22) and scores.device.type != "cpu":
+                # small eval batches: one D2H copy, then metrics on host —
+                # far fewer launches/syncs than ~15 tiny GPU kernels
+                scores, gy = scores.cpu(), gy.cpu()
             if self.spec.family in ("pegasos", "adaline"):
-                results_global = binary_margin_metrics(scores[:, :, 0], self.data.gy)
+                results_global = binary_margin_metrics(scores[:, :, 0], gy)
             else:
-                results_global = classification_metrics_shared(scores, self.data.gy)
+                results_global = classification_metrics_shared(scores, gy)
 
         results_local: List[dict] = []
         if self.data.tx is not None and len(mine):
@@ -523,9 +528,22 @@ class BatchedGossipSimulator(SimulationEventSender):
         for _ in range(n_rounds):
             r = self.rounds_done
             if fast:
-                sched = self.scheduler.next_round_flat(r)
+                # the scheduler is stateful (rounds must be generated exactly
+                # once, in order), so the prefetch lives on self and is keyed
+                # by round index — it survives across start() calls
+                pre = getattr(self, "_prefetched", None)
+                if pre is not None and pre[0] == r:
+                    _, sched, flat = pre
+                else:
+                    sched = self.scheduler.next_round_flat(r)
+                    flat = self.scheduler.last_flat
+                self._prefetched = None
                 self.pool.ensure(sched.n_slots)
-                self._run_round_fast(self.scheduler.last_flat)
+                self._run_round_fast(flat)
+                # overlap: derive round r+1's schedule on the host while the
+                # GPU executes round r's launches
+                s2 = self.scheduler.next_round_flat(r + 1)
+                self._prefetched = (r + 1, s2, self.scheduler.last_flat)
             else:
                 sched = self.scheduler.next_round(r)
                 self.pool.ensure(sched.n_slots)

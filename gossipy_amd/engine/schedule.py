@@ -983,6 +983,177 @@ class All2AllScheduler(Scheduler):
         return int(d.get(None))
 
 
+class CacheNeighScheduler(Scheduler):
+    """Cache-neighborhood gossip (Giaretta 2019; CacheNeighNode,
+    gossipy/node.py:395-496).
+
+    A receiver stores each incoming model in a per-sender slot (replacing
+    any previous one); when a node times out it first pops a *random*
+    cached slot and runs the normal merge+update with it, then snapshots
+    and pushes its model. The reference's slot-pick
+    (``random.choice(set(...))``, gossipy/node.py:449) crashes as shipped
+    (SURVEY.md §2.3 quirk 7); like the object layer, the engine draws
+    uniformly over the senders present (sorted order, tape-driven).
+
+    Emitted as two phases per tick: phase 1 delivers the popped-cache
+    merges (ordinary delivery CSR — no new kernels), phase 2 carries the
+    post-merge snapshots; stores at delivery time are host-side only.
+    """
+
+    def __init__(self, cfg: EngineConfig):
+        super().__init__(cfg)
+        #: per-node cache: sender -> slot (insertion-ordered)
+        self._acc: List[Dict[int, int]] = [dict() for _ in range(cfg.n_nodes)]
+        self._refs: Dict[int, int] = {}
+
+    def _deref(self, slot: int, freed: List[int]) -> None:
+        r = self._refs.get(slot, 1) - 1
+        if r <= 0:
+            self._refs.pop(slot, None)
+            freed.append(slot)
+        else:
+            self._refs[slot] = r
+
+    def next_round(self, r: int) -> RoundSchedule:
+        cfg = self.cfg
+        proto = cfg.protocol
+        assert proto != AntiEntropyProtocol.PULL, (
+            "CacheNeigh PULL requests carry no model; use PUSH or PUSH_PULL"
+        )
+        t0, t1 = r * cfg.delta, (r + 1) * cfg.delta
+        sent = failed = total_size = 0
+        ticks: List[TickPhase] = []
+
+        for t in range(t0, t1):
+            freed: List[int] = []
+            firing = self._firing(t)
+
+            # --- phase 1: pop-a-random-cached-model merges (node.py:442-474)
+            m_nodes: List[int] = []
+            m_slots: List[int] = []
+            m_owners: List[int] = []
+            pick = self.tape.stream(Purpose.MISC, t)
+            for i in firing:
+                acc = self._acc[int(i)]
+                if not acc:
+                    continue
+                senders = sorted(acc)
+                kpick = senders[int(pick.integers(0, len(senders)))]
+                slot = acc.pop(kpick)
+                m_nodes.append(int(i))
+                m_slots.append(slot)
+                m_owners.append(kpick)
+                self._deref(slot, freed)
+            if m_nodes:
+                ptr = np.arange(len(m_nodes) + 1, dtype=np.int32)
+                ticks.append(
+                    TickPhase(
+                        t=t,
+                        snap_nodes=np.zeros(0, dtype=np.int32),
+                        snap_slots=np.zeros(0, dtype=np.int32),
+                        recv_nodes=np.asarray(m_nodes, dtype=np.int32),
+                        recv_ptr=ptr,
+                        del_slots=np.asarray(m_slots, dtype=np.int32),
+                        del_owners=np.asarray(m_owners, dtype=np.int32),
+                        reply_slots=np.full(len(m_nodes), -1, dtype=np.int32),
+                        pull_snap_nodes=np.zeros(0, dtype=np.int32),
+                        pull_snap_slots=np.zeros(0, dtype=np.int32),
+                        del_pids=np.full(len(m_nodes), -1, dtype=np.int32),
+                    )
+                )
+
+            # --- phase 2: post-merge snapshots + sends (node.py:445-460)
+            snap_nodes: List[int] = []
+            snap_slots: List[int] = []
+            reply_snap_nodes: List[int] = []
+            reply_snap_slots: List[int] = []
+            n_f = len(firing)
+            if n_f:
+                peers = self._peers_of(firing, t)
+                drop_u = self.tape.uniform(Purpose.DROP, t, n_f)
+                sizes = np.full(n_f, cfg.model_size)
+                delays = self._delays(t, n_f, sizes)
+                for j in range(n_f):
+                    sender, receiver = int(firing[j]), int(peers[j])
+                    slot = self._alloc_slot(sender)
+                    snap_nodes.append(sender)
+                    snap_slots.append(slot)
+                    sent += 1
+                    total_size += int(sizes[j])
+                    if drop_u[j] >= cfg.drop_prob:
+                        wants = proto == AntiEntropyProtocol.PUSH_PULL
+                        self._pending.setdefault(t + int(delays[j]), []).append(
+                            (receiver, slot, -2 if wants else -1, False, sender, -1)
+                        )
+                        self._refs[slot] = 1
+                    else:
+                        failed += 1
+                        freed.append(slot)
+
+            # --- deliveries: store into the receiver's per-sender slot
+            # (node.py:477-496); PUSH_PULL triggers a reply snapshot of the
+            # receiver's CURRENT model (no merge)
+            online_u = self.tape.uniform(Purpose.ONLINE, t, cfg.n_nodes)
+            online = online_u <= cfg.online_prob
+            for receiver, slot, rf, _ip, sender, _pid in self._pending.pop(t, []):
+                if not online[receiver]:
+                    failed += 1
+                    self._deref(slot, freed)
+                    continue
+                if rf == -2:  # reply with own snapshot, delivered like a push
+                    rslot = self._alloc_slot(receiver)
+                    reply_snap_nodes.append(receiver)
+                    reply_snap_slots.append(rslot)
+                    if self._enqueue_reply(t, receiver, sender, rslot):
+                        self._refs[rslot] = 1
+                    else:
+                        freed.append(rslot)
+                acc = self._acc[receiver]
+                old = acc.get(sender)
+                if old is not None:
+                    self._deref(old, freed)
+                acc[sender] = slot
+
+            phase2 = TickPhase(
+                t=t,
+                snap_nodes=np.asarray(snap_nodes, dtype=np.int32),
+                snap_slots=np.asarray(snap_slots, dtype=np.int32),
+                recv_nodes=np.zeros(0, dtype=np.int32),
+                recv_ptr=np.zeros(1, dtype=np.int32),
+                del_slots=np.zeros(0, dtype=np.int32),
+                del_owners=np.zeros(0, dtype=np.int32),
+                reply_slots=np.zeros(0, dtype=np.int32),
+                pull_snap_nodes=np.asarray(reply_snap_nodes, dtype=np.int32),
+                pull_snap_slots=np.asarray(reply_snap_slots, dtype=np.int32),
+                del_pids=np.zeros(0, dtype=np.int32),
+            )
+            if phase2.n_events:
+                ticks.append(phase2)
+
+            sent_r, failed_r, size_r = self._reply_accounting.pop(t, (0, 0, 0))
+            sent += sent_r
+            failed += failed_r
+            total_size += size_r
+            self._free_slots.extend(freed)
+
+        eval_nodes = None
+        if cfg.sampling_eval > 0:
+            g = self.tape.stream(Purpose.EVAL, t1 - 1)
+            k = max(int(cfg.n_nodes * cfg.sampling_eval), 1)
+            eval_nodes = np.atleast_1d(g.integers(0, cfg.n_nodes, size=k))
+
+        return RoundSchedule(
+            round_idx=r,
+            ticks=ticks,
+            n_slots=self._next_slot,
+            slot_owner=self.slot_owner[: self._next_slot].copy(),
+            sent_messages=sent,
+            failed_messages=failed,
+            total_size=total_size,
+            eval_nodes=eval_nodes,
+        )
+
+
 class NativeSchedulerAdapter:
     """Adapter over the C++ scheduler (``csrc/scheduler.cpp``).
 
