@@ -127,6 +127,11 @@ public:
             indices_.assign(ix.data(), ix.data() + ix.size());
         }
         slot_owner_.resize(64);
+        if (sync_) {  // static offsets: bucket nodes by firing phase once
+            fire_buckets_.resize(delta_);
+            for (int64_t i = 0; i < n_; ++i)
+                fire_buckets_[deltas_[i]].push_back((int32_t)i);
+        }
     }
 
     py::dict next_round(int64_t r);
@@ -146,6 +151,7 @@ private:
     int64_t n_parts_ = 0;
     bool sampled_ = false;
     std::vector<int64_t> deltas_;
+    std::vector<std::vector<int32_t>> fire_buckets_;
     std::vector<int64_t> indptr_, indices_;
     std::unordered_map<int64_t, std::vector<Msg>> pending_;
     std::vector<int32_t> free_slots_;
@@ -237,9 +243,7 @@ py::dict NativeScheduler::next_round(int64_t r)
         // --- firing set (ascending node id, like np.where)
         firing.clear();
         if (sync_) {
-            int64_t phase = t % delta_;
-            for (int64_t i = 0; i < n_; ++i)
-                if (deltas_[i] == phase) firing.push_back((int32_t)i);
+            firing = fire_buckets_[t % delta_];
         } else {
             for (int64_t i = 0; i < n_; ++i)
                 if (t % deltas_[i] == 0) firing.push_back((int32_t)i);
@@ -266,9 +270,11 @@ py::dict NativeScheduler::next_round(int64_t r)
                                                           u[j] * (double)deg)];
                 }
             }
-            Stream gdrop(stream_key(seed_, DROP, (uint64_t)t, 0));
-            std::vector<double> drop_u(n_f);
-            for (size_t j = 0; j < n_f; ++j) drop_u[j] = gdrop.rnd();
+            std::vector<double> drop_u(n_f, 1.0);
+            if (drop_ > 0.0) {
+                Stream gdrop(stream_key(seed_, DROP, (uint64_t)t, 0));
+                for (size_t j = 0; j < n_f; ++j) drop_u[j] = gdrop.rnd();
+            }
             int64_t msize = (proto_ == PULL) ? 1 : model_size_;
             std::vector<int64_t> delays(n_f);
             {
@@ -309,9 +315,11 @@ py::dict NativeScheduler::next_round(int64_t r)
         snap_tptr.push_back((int32_t)snap_nodes.size());
 
         // --- deliveries due this tick (sub-phase B)
-        {
+        if (online_ < 1.0) {
             Stream go(stream_key(seed_, ONLINE, (uint64_t)t, 0));
             for (int64_t i = 0; i < n_; ++i) online[i] = go.rnd();
+        } else {
+            std::fill(online.begin(), online.end(), 0.0);
         }
         std::vector<Msg> due;
         {

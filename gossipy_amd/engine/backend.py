@@ -625,12 +625,17 @@ class TorchBackend:
                 reply_slots,
             )
             return
-        mode = spec.mode
+        base_mode = spec.mode
+        pt = getattr(spec, "pass_through", False)
         ptr = recv_ptr.tolist()
         for i, node_t in enumerate(recv_nodes.tolist()):
             node = torch.tensor([node_t])
             for j in range(ptr[i], ptr[i + 1]):
                 slot = int(del_slots[j])
+                mode = base_mode
+                if pt and del_pids is not None and int(del_pids[j]) == 1:
+                    # pass-through adoption (gossipy/node.py:386-392)
+                    mode = CreateModelMode.PASS
                 if mode == CreateModelMode.MERGE_UPDATE:
                     self._merge_mean(state, pool, node_t, slot)
                     self.update(state, data, spec, node)
@@ -961,75 +966,86 @@ class HIPBackend(TorchBackend):
                 mode,
                 bool(update_only),
             )
-        elif spec.family == "logreg":
-            self.ext.tick_logreg(
-                state.params,
-                state.ages,
-                slots,
-                slot_ages,
-                nodes,
-                recv_ptr,
-                del_slots,
-                reply_slots,
-                data.x,
-                data.y,
-                data.counts,
-                spec.d_in,
-                spec.n_classes,
-                spec.lr,
-                spec.weight_decay,
-                max(1, spec.local_epochs),
-                spec.batch_size,
-                mode,
-                bool(update_only),
-            )
-        elif spec.family in ("pegasos", "adaline"):
-            self.ext.tick_linear(
-                state.params,
-                state.ages,
-                slots,
-                slot_ages,
-                nodes,
-                recv_ptr,
-                del_slots,
-                reply_slots,
-                data.x,
-                data.y,
-                data.counts,
-                spec.d_in,
-                spec.lam if spec.family == "pegasos" else spec.lr,
-                1 if spec.family == "pegasos" else 0,
-                mode,
-                bool(update_only),
-            )
-        elif spec.family == "mlp":
-            self.ext.tick_mlp(
-                state.params,
-                state.ages,
-                slots,
-                slot_ages,
-                nodes,
-                recv_ptr,
-                del_slots,
-                reply_slots,
-                data.x,
-                data.y,
-                data.counts,
-                torch.tensor(
-                    [x for t in spec.layer_offsets() for x in t],
-                    dtype=torch.int32,
-                    device=dev,
-                ),
-                len(spec.layer_offsets()),
-                spec.lr,
-                spec.weight_decay,
-                max(1, spec.local_epochs),
-                spec.batch_size,
-                mode,
-                bool(update_only),
-            )
         else:
-            raise ValueError(spec.family)
+            # plain families: pass-through resolves deliveries to PASS via
+            # the per-delivery dmodes channel (del_pids carries the coin)
+            dmodes = (
+                del_pids
+                if getattr(spec, "pass_through", False) and del_pids.numel()
+                else torch.zeros(0, dtype=torch.int32, device=dev)
+            )
+            if spec.family == "logreg":
+                self.ext.tick_logreg(
+                    state.params,
+                    state.ages,
+                    slots,
+                    slot_ages,
+                    nodes,
+                    recv_ptr,
+                    del_slots,
+                    reply_slots,
+                    data.x,
+                    data.y,
+                    data.counts,
+                    spec.d_in,
+                    spec.n_classes,
+                    spec.lr,
+                    spec.weight_decay,
+                    max(1, spec.local_epochs),
+                    spec.batch_size,
+                    mode,
+                    bool(update_only),
+                    dmodes,
+                )
+            elif spec.family in ("pegasos", "adaline"):
+                self.ext.tick_linear(
+                    state.params,
+                    state.ages,
+                    slots,
+                    slot_ages,
+                    nodes,
+                    recv_ptr,
+                    del_slots,
+                    reply_slots,
+                    data.x,
+                    data.y,
+                    data.counts,
+                    spec.d_in,
+                    spec.lam if spec.family == "pegasos" else spec.lr,
+                    1 if spec.family == "pegasos" else 0,
+                    mode,
+                    bool(update_only),
+                    dmodes,
+                )
+            elif spec.family == "mlp":
+                self.ext.tick_mlp(
+                    state.params,
+                    state.ages,
+                    slots,
+                    slot_ages,
+                    nodes,
+                    recv_ptr,
+                    del_slots,
+                    reply_slots,
+                    data.x,
+                    data.y,
+                    data.counts,
+                    torch.tensor(
+                        [x for t in spec.layer_offsets() for x in t],
+                        dtype=torch.int32,
+                        device=dev,
+                    ),
+                    len(spec.layer_offsets()),
+                    spec.lr,
+                    spec.weight_decay,
+                    max(1, spec.local_epochs),
+                    spec.batch_size,
+                    mode,
+                    bool(update_only),
+                    dmodes,
+                )
+            else:
+                raise ValueError(spec.family)
 
 
 def make_backend(device: torch.device):

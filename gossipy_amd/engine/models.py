@@ -101,6 +101,8 @@ class PegasosSpec:
     d_in: int
     lam: float = 0.01  #: the handler's ``learning_rate`` (lambda)
     mode: CreateModelMode = CreateModelMode.MERGE_UPDATE
+    #: pass-through gossip (PassThroughNode): deliveries may resolve to PASS
+    pass_through: bool = False
 
     family = "pegasos"
     n_parts = 0
@@ -119,6 +121,7 @@ class AdaLineSpec:
     d_in: int
     lr: float = 0.01
     mode: CreateModelMode = CreateModelMode.MERGE_UPDATE
+    pass_through: bool = False
 
     family = "adaline"
     n_parts = 0
@@ -155,6 +158,8 @@ class LogRegSpec(_PartitionMixin):
     #: (gossipy/model/sampling.py:37-72), derived from a per-delivery tape
     #: seed instead of global RNG. Mutually exclusive with ``n_parts``.
     sample_size: float = 0.0
+    #: pass-through gossip (PassThroughNode): deliveries may resolve to PASS
+    pass_through: bool = False
 
     family = "logreg"
 
@@ -189,6 +194,7 @@ class MLPSpec(_PartitionMixin):
     batch_size: int = 32
     mode: CreateModelMode = CreateModelMode.MERGE_UPDATE
     n_parts: int = 0
+    pass_through: bool = False
 
     family = "mlp"
 

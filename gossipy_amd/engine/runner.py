@@ -350,10 +350,6 @@ class BatchedGossipSimulator(SimulationEventSender):
         if self.data.gx is not None and len(mine):
             scores = self.backend.scores(self.state, self.spec, local_ids, self.data.gx)
             gy = self.data.gy
-            if scores.numel() < (1 << 22) and scores.device.type != "cpu":
-                # small eval batches: one D2H copy, then metrics on host —
-                # far fewer launches/syncs than ~15 tiny GPU kernels
-                scores, gy = scores.cpu(), gy.cpu()
             if self.spec.family in ("pegasos", "adaline"):
                 results_global = binary_margin_metrics(scores[:, :, 0], gy)
             else:

@@ -71,6 +71,13 @@ class EngineConfig:
     #: PUSH/REPLY carries a sample seed (rides the del_pids channel;
     #: mutually exclusive with n_parts)
     sampled: bool = False
+    #: pass-through gossip (Giaretta 2019; PassThroughNode,
+    #: gossipy/node.py:289-392): each delivery merges normally with
+    #: probability min(1, deg_sender/deg_receiver) and otherwise ADOPTS the
+    #: received model (PASS). The del_pids channel carries the resolved
+    #: coin (0 = normal, 1 = pass). Mutually exclusive with
+    #: n_parts/sampled.
+    pass_through: bool = False
 
 
 @dataclass
@@ -199,7 +206,15 @@ class Scheduler:
 
     def _firing(self, t: int) -> np.ndarray:
         if self.cfg.sync:
-            return np.where((t % self.cfg.delta) == self.deltas)[0]
+            # static timeout offsets: bucket once, O(1) per tick
+            buckets = getattr(self, "_fire_buckets", None)
+            if buckets is None:
+                buckets = [
+                    np.where(self.deltas == ph)[0].astype(np.int64)
+                    for ph in range(self.cfg.delta)
+                ]
+                self._fire_buckets = buckets
+            return buckets[t % self.cfg.delta]
         return np.where((t % self.deltas) == 0)[0]
 
     def _peers_of(self, nodes: np.ndarray, t: int) -> np.ndarray:
@@ -253,7 +268,10 @@ class Scheduler:
             n_f = len(firing)
             if n_f:
                 peers = self._peers_of(firing, t)
-                drop_u = self.tape.uniform(Purpose.DROP, t, n_f)
+                if cfg.drop_prob <= 0.0:
+                    drop_u = np.ones(n_f)
+                else:
+                    drop_u = self.tape.uniform(Purpose.DROP, t, n_f)
                 sizes = np.full(n_f, cfg.model_size if proto != AntiEntropyProtocol.PULL else 1)
                 delays = self._delays(t, n_f, sizes)
                 # partition ids / sample seeds drawn per send
@@ -285,18 +303,28 @@ class Scheduler:
             #     gossipy/simul.py:409-430). Reply messages generated here are
             #     enqueued for their own delivery tick; replies to replies are
             #     discarded (reference parity).
-            online_u = self.tape.uniform(Purpose.ONLINE, t, cfg.n_nodes)
-            online = online_u <= cfg.online_prob
+            if cfg.online_prob >= 1.0:
+                online = np.ones(cfg.n_nodes, dtype=bool)
+            else:
+                online_u = self.tape.uniform(Purpose.ONLINE, t, cfg.n_nodes)
+                online = online_u <= cfg.online_prob
             due = self._pending.pop(t, [])
             recv_map: Dict[int, List[Tuple[int, int, int, int]]] = {}
             pull_nodes: List[int] = []
             pull_slots: List[int] = []
+            pt_stream = (
+                self.tape.stream(Purpose.MISC, t) if cfg.pass_through else None
+            )
             for receiver, slot, reply_flag, is_pull, sender, pid in due:
                 if not online[receiver]:
                     failed += 1
                     if slot >= 0:
                         freed.append(slot)
                     continue
+                if pt_stream is not None and not is_pull:
+                    # degree-aware accept coin (gossipy/node.py:380-386)
+                    p = min(1.0, self._deg(sender) / max(1, self._deg(receiver)))
+                    pid = 0 if float(pt_stream.random()) < p else 1
                 if is_pull:
                     # PULL request: receiver snapshots and replies (with a
                     # fresh partition id, gossipy/node.py:651-653)
@@ -340,6 +368,9 @@ class Scheduler:
                     failed += 1
                     freed.append(slot)
                     continue
+                if pt_stream is not None:
+                    p = min(1.0, self._deg(sender) / max(1, self._deg(receiver)))
+                    pid = 0 if float(pt_stream.random()) < p else 1
                 rep_map.setdefault(receiver, []).append((slot, sender, pid))
                 freed.append(slot)  # consumed by this reply delivery
             rep_recv = np.fromiter(rep_map.keys(), dtype=np.int32, count=len(rep_map))
@@ -400,6 +431,13 @@ class Scheduler:
             failed_messages=failed,
             total_size=total_size,
             eval_nodes=eval_nodes,
+        )
+
+    def _deg(self, node: int) -> int:
+        if self.cfg.peers_indptr is None:
+            return self.cfg.n_nodes - 1
+        return int(
+            self.cfg.peers_indptr[node + 1] - self.cfg.peers_indptr[node]
         )
 
     def _send_extras(self, t: int, n: int) -> np.ndarray:
@@ -569,8 +607,11 @@ class TokenizedScheduler(Scheduler):
                             if slot >= 0:
                                 freed.append(slot)
 
-            online_u = self.tape.uniform(Purpose.ONLINE, t, cfg.n_nodes)
-            online = online_u <= cfg.online_prob
+            if cfg.online_prob >= 1.0:
+                online = np.ones(cfg.n_nodes, dtype=bool)
+            else:
+                online_u = self.tape.uniform(Purpose.ONLINE, t, cfg.n_nodes)
+                online = online_u <= cfg.online_prob
 
             # --- delivery waves: wave 0 = scheduled messages due at t;
             # each wave's reactive bursts with zero delay feed the next wave
@@ -917,8 +958,11 @@ class All2AllScheduler(Scheduler):
                         self._refs[slot] = refs
 
             # --- deliveries: store into the receiver's accumulator
-            online_u = self.tape.uniform(Purpose.ONLINE, t, cfg.n_nodes)
-            online = online_u <= cfg.online_prob
+            if cfg.online_prob >= 1.0:
+                online = np.ones(cfg.n_nodes, dtype=bool)
+            else:
+                online_u = self.tape.uniform(Purpose.ONLINE, t, cfg.n_nodes)
+                online = online_u <= cfg.online_prob
             for receiver, slot, _rf, _ip, sender, _pid in self._pending.pop(t, []):
                 if not online[receiver]:
                     failed += 1
@@ -1093,26 +1137,35 @@ class CacheNeighScheduler(Scheduler):
             # --- deliveries: store into the receiver's per-sender slot
             # (node.py:477-496); PUSH_PULL triggers a reply snapshot of the
             # receiver's CURRENT model (no merge)
-            online_u = self.tape.uniform(Purpose.ONLINE, t, cfg.n_nodes)
-            online = online_u <= cfg.online_prob
-            for receiver, slot, rf, _ip, sender, _pid in self._pending.pop(t, []):
-                if not online[receiver]:
-                    failed += 1
-                    self._deref(slot, freed)
-                    continue
-                if rf == -2:  # reply with own snapshot, delivered like a push
-                    rslot = self._alloc_slot(receiver)
-                    reply_snap_nodes.append(receiver)
-                    reply_snap_slots.append(rslot)
-                    if self._enqueue_reply(t, receiver, sender, rslot):
-                        self._refs[rslot] = 1
-                    else:
-                        freed.append(rslot)
-                acc = self._acc[receiver]
-                old = acc.get(sender)
-                if old is not None:
-                    self._deref(old, freed)
-                acc[sender] = slot
+            if cfg.online_prob >= 1.0:
+                online = np.ones(cfg.n_nodes, dtype=bool)
+            else:
+                online_u = self.tape.uniform(Purpose.ONLINE, t, cfg.n_nodes)
+                online = online_u <= cfg.online_prob
+            # loop: zero-delay PUSH_PULL replies land back in _pending[t]
+            # and are stored the same tick (the reference's rep_queues[t])
+            while True:
+                due = self._pending.pop(t, [])
+                if not due:
+                    break
+                for receiver, slot, rf, _ip, sender, _pid in due:
+                    if not online[receiver]:
+                        failed += 1
+                        self._deref(slot, freed)
+                        continue
+                    if rf == -2:  # reply with own snapshot (no merge)
+                        rslot = self._alloc_slot(receiver)
+                        reply_snap_nodes.append(receiver)
+                        reply_snap_slots.append(rslot)
+                        if self._enqueue_reply(t, receiver, sender, rslot):
+                            self._refs[rslot] = 1
+                        else:
+                            freed.append(rslot)
+                    acc = self._acc[receiver]
+                    old = acc.get(sender)
+                    if old is not None:
+                        self._deref(old, freed)
+                    acc[sender] = slot
 
             phase2 = TickPhase(
                 t=t,
@@ -1180,6 +1233,8 @@ class NativeSchedulerAdapter:
             kind, dmin, dmax, tx, ov = 2, 0, 0, d._timexunit, d._overhead
         else:
             raise TypeError("custom Delay subclasses need the python Scheduler")
+        if cfg.pass_through:
+            raise TypeError("pass-through gossip needs the python Scheduler")
         ip = cfg.peers_indptr
         ix = cfg.peers_indices
         self._native = mod.NativeScheduler(

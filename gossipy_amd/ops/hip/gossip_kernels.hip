@@ -87,6 +87,7 @@ struct LogregArgs {
     float* slots; int* slot_ages;
     const int* nodes; const int* ptr;
     const int* dslots; const int* rslots;
+    const int* dmodes;  // per-delivery pass-through flag (1 = PASS), or null
     const float* X; const float* Y; const int* counts;
     int d, k, Smax, D;
     float lr, wd;
@@ -186,19 +187,22 @@ tick_logreg_kernel(LogregArgs a)
             int slot = a.dslots[j];
             const float* srow = a.slots + (long)slot * a.D;
             int sage = a.slot_ages[slot];
-            if (a.mode == MODE_MERGE_UPDATE) {
+            // pass-through gossip resolves some deliveries to PASS
+            // (gossipy/node.py:380-392)
+            int md = (a.dmodes && a.dmodes[j]) ? MODE_PASS : a.mode;
+            if (md == MODE_MERGE_UPDATE) {
                 for (int e = tid; e < a.D; e += blockDim.x)
                     W[e] = 0.5f * (W[e] + srow[e]);
                 age = max(age, sage);
                 __syncthreads();
                 logreg_update(a, node, W, xb, dz, age);
-            } else if (a.mode == MODE_UPDATE) {
+            } else if (md == MODE_UPDATE) {
                 // adopt the received model, then train it
                 for (int e = tid; e < a.D; e += blockDim.x) W[e] = srow[e];
                 age = sage;
                 __syncthreads();
                 logreg_update(a, node, W, xb, dz, age);
-            } else if (a.mode == MODE_UPDATE_MERGE) {
+            } else if (md == MODE_UPDATE_MERGE) {
                 logreg_update(a, node, W, xb, dz, age);
                 for (int e = tid; e < a.D; e += blockDim.x) W2[e] = srow[e];
                 __syncthreads();
@@ -774,6 +778,7 @@ struct LinearArgs {
     float* slots; int* slot_ages;
     const int* nodes; const int* ptr;
     const int* dslots; const int* rslots;
+    const int* dmodes;
     const float* X; const float* Y; const int* counts;
     int d, Smax;
     float lrlam;       // pegasos lambda or adaline lr
@@ -839,21 +844,22 @@ tick_linear_kernel(LinearArgs a)
             int slot = a.dslots[j];
             const float* srow = a.slots + (long)slot * a.d;
             int sage = a.slot_ages[slot];
-            if (a.mode == MODE_MERGE_UPDATE) {
+            int md = (a.dmodes && a.dmodes[j]) ? MODE_PASS : a.mode;
+            if (md == MODE_MERGE_UPDATE) {
                 for (int r = 0; r < nreg; ++r) {
                     int e = lane + r * WAVE;
                     if (e < a.d) w[r] = 0.5f * (w[r] + srow[e]);
                 }
                 age = max(age, sage);
                 linear_update(a, node, w, nreg, age);
-            } else if (a.mode == MODE_UPDATE) {
+            } else if (md == MODE_UPDATE) {
                 for (int r = 0; r < nreg; ++r) {
                     int e = lane + r * WAVE;
                     if (e < a.d) w[r] = srow[e];
                 }
                 age = sage;
                 linear_update(a, node, w, nreg, age);
-            } else if (a.mode == MODE_UPDATE_MERGE) {
+            } else if (md == MODE_UPDATE_MERGE) {
                 linear_update(a, node, w, nreg, age);
                 float w2[LIN_MAX_REGS];
                 for (int r = 0; r < nreg; ++r) {
@@ -897,6 +903,7 @@ struct MlpArgs {
     float* slots; int* slot_ages;
     const int* nodes; const int* ptr;
     const int* dslots; const int* rslots;
+    const int* dmodes;
     const float* X; const float* Y; const int* counts;
     const int* layout;  // [n_layers][4] = (w_off, b_off, in, out)
     int n_layers, Smax, D, act_max, d_in;
@@ -1035,18 +1042,19 @@ tick_mlp_kernel(MlpArgs a)
             int slot = a.dslots[j];
             const float* srow = a.slots + (long)slot * a.D;
             int sage = a.slot_ages[slot];
-            if (a.mode == MODE_MERGE_UPDATE) {
+            int md = (a.dmodes && a.dmodes[j]) ? MODE_PASS : a.mode;
+            if (md == MODE_MERGE_UPDATE) {
                 for (int e = tid; e < a.D; e += blockDim.x)
                     W[e] = 0.5f * (W[e] + srow[e]);
                 age = max(age, sage);
                 __syncthreads();
                 mlp_update(a, node, W, act, grad, age);
-            } else if (a.mode == MODE_UPDATE) {
+            } else if (md == MODE_UPDATE) {
                 for (int e = tid; e < a.D; e += blockDim.x) W[e] = srow[e];
                 age = sage;
                 __syncthreads();
                 mlp_update(a, node, W, act, grad, age);
-            } else if (a.mode == MODE_UPDATE_MERGE) {
+            } else if (md == MODE_UPDATE_MERGE) {
                 mlp_update(a, node, W, act, grad, age);
                 for (int e = tid; e < a.D; e += blockDim.x) W2[e] = srow[e];
                 __syncthreads();
@@ -1166,7 +1174,8 @@ void tick_logreg(torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
                  torch::Tensor del_slots, torch::Tensor reply_slots,
                  torch::Tensor X, torch::Tensor Y, torch::Tensor counts,
                  int64_t d, int64_t k, double lr, double wd, int64_t epochs,
-                 int64_t bs, int64_t mode, bool update_only)
+                 int64_t bs, int64_t mode, bool update_only,
+                 torch::Tensor dmodes)
 {
     CHECK_DEV(params); CHECK_DEV(slots); CHECK_DEV(X); CHECK_DEV(Y);
     TORCH_CHECK(k <= KMAX, "n_classes > ", KMAX, " unsupported");
@@ -1180,6 +1189,7 @@ void tick_logreg(torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
     a.rslots = reply_slots.numel() ? reply_slots.data_ptr<int>() : nullptr;
     a.X = X.data_ptr<float>(); a.Y = Y.data_ptr<float>();
     a.counts = counts.data_ptr<int>();
+    a.dmodes = dmodes.numel() ? dmodes.data_ptr<int>() : nullptr;
     a.d = d; a.k = k; a.Smax = X.size(1); a.D = params.size(1);
     a.lr = lr; a.wd = wd; a.epochs = epochs; a.bs = bs; a.mode = mode;
     a.update_only = update_only;
@@ -1195,7 +1205,7 @@ void tick_linear(torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
                  torch::Tensor del_slots, torch::Tensor reply_slots,
                  torch::Tensor X, torch::Tensor Y, torch::Tensor counts,
                  int64_t d, double lrlam, int64_t is_pegasos, int64_t mode,
-                 bool update_only)
+                 bool update_only, torch::Tensor dmodes)
 {
     CHECK_DEV(params); CHECK_DEV(slots); CHECK_DEV(X); CHECK_DEV(Y);
     TORCH_CHECK(d <= LIN_MAX_REGS * WAVE, "d > ", LIN_MAX_REGS * WAVE, " unsupported");
@@ -1209,6 +1219,7 @@ void tick_linear(torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
     a.rslots = reply_slots.numel() ? reply_slots.data_ptr<int>() : nullptr;
     a.X = X.data_ptr<float>(); a.Y = Y.data_ptr<float>();
     a.counts = counts.data_ptr<int>();
+    a.dmodes = dmodes.numel() ? dmodes.data_ptr<int>() : nullptr;
     a.d = d; a.Smax = X.size(1);
     a.lrlam = lrlam; a.is_pegasos = is_pegasos; a.mode = mode;
     a.update_only = update_only;
@@ -1221,7 +1232,8 @@ void tick_mlp(torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
               torch::Tensor del_slots, torch::Tensor reply_slots,
               torch::Tensor X, torch::Tensor Y, torch::Tensor counts,
               torch::Tensor layout, int64_t n_layers, double lr, double wd,
-              int64_t epochs, int64_t bs, int64_t mode, bool update_only)
+              int64_t epochs, int64_t bs, int64_t mode, bool update_only,
+              torch::Tensor dmodes)
 {
     CHECK_DEV(params); CHECK_DEV(slots); CHECK_DEV(X); CHECK_DEV(Y);
     CHECK_DEV(layout);
@@ -1236,6 +1248,7 @@ void tick_mlp(torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
     a.X = X.data_ptr<float>(); a.Y = Y.data_ptr<float>();
     a.counts = counts.data_ptr<int>();
     a.layout = layout.data_ptr<int>();
+    a.dmodes = dmodes.numel() ? dmodes.data_ptr<int>() : nullptr;
     a.n_layers = n_layers; a.Smax = X.size(1); a.D = params.size(1);
     a.d_in = layout[2].item<int>();
     a.lr = lr; a.wd = wd; a.epochs = epochs; a.bs = bs; a.mode = mode;
@@ -1346,6 +1359,7 @@ void run_round_logreg(
     a.slots = slots.data_ptr<float>(); a.slot_ages = slot_ages.data_ptr<int>();
     a.X = X.data_ptr<float>(); a.Y = Y.data_ptr<float>();
     a.counts = counts.data_ptr<int>();
+    a.dmodes = nullptr;
     a.d = d; a.k = k; a.Smax = X.size(1); a.D = params.size(1);
     a.lr = lr; a.wd = wd; a.epochs = epochs; a.bs = bs; a.mode = mode;
     a.update_only = 0;
@@ -1405,6 +1419,7 @@ void run_round_linear(
     a.slots = slots.data_ptr<float>(); a.slot_ages = slot_ages.data_ptr<int>();
     a.X = X.data_ptr<float>(); a.Y = Y.data_ptr<float>();
     a.counts = counts.data_ptr<int>();
+    a.dmodes = nullptr;
     a.d = d; a.Smax = X.size(1);
     a.lrlam = lrlam; a.is_pegasos = is_pegasos; a.mode = mode;
     a.update_only = 0;
@@ -1621,6 +1636,7 @@ void run_round_mlp(
     a.X = X.data_ptr<float>(); a.Y = Y.data_ptr<float>();
     a.counts = counts.data_ptr<int>();
     a.layout = layout.data_ptr<int>();
+    a.dmodes = nullptr;
     a.n_layers = n_layers; a.Smax = X.size(1); a.D = params.size(1);
     a.lr = lr; a.wd = wd; a.epochs = epochs; a.bs = bs; a.mode = mode;
     a.update_only = 0;
