@@ -15,6 +15,7 @@ from .models import AdaLineSpec, KMeansSpec, LogRegSpec, MFSpec, MLPSpec, Pegaso
 from .rng import Purpose, RandomTape
 from .runner import (
     BatchedAll2AllGossipSimulator,
+    BatchedCacheNeighGossipSimulator,
     BatchedGossipSimulator,
     BatchedTokenizedGossipSimulator,
 )
@@ -31,6 +32,7 @@ __all__ = [
     "BatchedGossipSimulator",
     "BatchedTokenizedGossipSimulator",
     "BatchedAll2AllGossipSimulator",
+    "BatchedCacheNeighGossipSimulator",
     "TokenizedScheduler",
     "EngineConfig",
     "Scheduler",

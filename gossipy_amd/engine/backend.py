@@ -835,6 +835,38 @@ class HIPBackend(TorchBackend):
         )
         self.update(state, data, spec, nodes)
 
+    def eval_metrics_fast(self, state, spec, local_ids, gx, gy):
+        """One-launch K13 evaluation for affine/margin families; returns a
+        list of metric dicts, or ``None`` if the family needs the generic
+        path. Metrics match gossipy_amd.engine.metrics (sklearn
+        conventions); argmax/AUC on raw affine scores — the reference's
+        sigmoid is monotonic so predictions and ranks are identical."""
+        if spec.family not in ("logreg", "pegasos", "adaline"):
+            return None
+        is_margin = spec.family in ("pegasos", "adaline")
+        out = self.ext.eval_metrics(
+            state.params,
+            local_ids.to(state.params.device, torch.int32),
+            gx,
+            gy,
+            spec.d_in,
+            1 if is_margin else spec.n_classes,
+            is_margin,
+        )
+        vals = out.cpu().numpy()
+        res = []
+        for row in vals:
+            d = {
+                "accuracy": float(row[0]),
+                "precision": float(row[1]),
+                "recall": float(row[2]),
+                "f1_score": float(row[3]),
+            }
+            if row[4] >= 0:
+                d["auc"] = float(row[4])
+            res.append(d)
+        return res
+
     def _part_dev(self, spec, dev):
         """(perm, ptr, arena_part) as int32 device tensors (cached)."""
         key = ("dev32", id(spec), str(dev))

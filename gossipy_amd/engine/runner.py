@@ -50,6 +50,7 @@ __all__ = [
     "BatchedGossipSimulator",
     "BatchedTokenizedGossipSimulator",
     "BatchedAll2AllGossipSimulator",
+    "BatchedCacheNeighGossipSimulator",
 ]
 
 
@@ -348,12 +349,23 @@ class BatchedGossipSimulator(SimulationEventSender):
 
         results_global: List[dict] = []
         if self.data.gx is not None and len(mine):
-            scores = self.backend.scores(self.state, self.spec, local_ids, self.data.gx)
-            gy = self.data.gy
-            if self.spec.family in ("pegasos", "adaline"):
-                results_global = binary_margin_metrics(scores[:, :, 0], gy)
+            fast_eval = getattr(self.backend, "eval_metrics_fast", None)
+            got = None
+            if fast_eval is not None:
+                got = fast_eval(
+                    self.state, self.spec, local_ids, self.data.gx, self.data.gy
+                )
+            if got is not None:
+                results_global = got
             else:
-                results_global = classification_metrics_shared(scores, gy)
+                scores = self.backend.scores(
+                    self.state, self.spec, local_ids, self.data.gx
+                )
+                gy = self.data.gy
+                if self.spec.family in ("pegasos", "adaline"):
+                    results_global = binary_margin_metrics(scores[:, :, 0], gy)
+                else:
+                    results_global = classification_metrics_shared(scores, gy)
 
         results_local: List[dict] = []
         if self.data.tx is not None and len(mine):
@@ -777,3 +789,26 @@ class BatchedAll2AllGossipSimulator(BatchedGossipSimulator):
                     self.device
                 ),
             )
+
+
+class BatchedCacheNeighGossipSimulator(BatchedGossipSimulator):
+    """Cache-neighborhood gossip on the batched engine (CacheNeighNode,
+    gossipy/node.py:395-496): deliveries only store the model; a timed-out
+    node merges one randomly chosen cached neighbor model before sending.
+    Scheduling is host-side (CacheNeighScheduler); the kernels are the
+    ordinary per-family tick kernels."""
+
+    def __init__(
+        self,
+        cfg: EngineConfig,
+        spec,
+        data: DataArena,
+        device: Optional[torch.device] = None,
+    ):
+        super().__init__(cfg, spec, data, device=device)
+        from .schedule import CacheNeighScheduler
+
+        self.scheduler = CacheNeighScheduler(cfg)
+
+    def _fast_path_ok(self) -> bool:
+        return False

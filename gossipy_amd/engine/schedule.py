@@ -42,6 +42,8 @@ __all__ = [
     "RoundSchedule",
     "Scheduler",
     "TokenizedScheduler",
+    "All2AllScheduler",
+    "CacheNeighScheduler",
     "NativeSchedulerAdapter",
     "make_scheduler",
 ]

@@ -1085,6 +1085,133 @@ tick_mlp_kernel(MlpArgs a)
 }
 
 // ---------------------------------------------------------------------------
+// fused evaluation metrics (K13): accuracy / macro precision / recall / F1
+// (+ pairwise ROC-AUC for binary) of R node models on one shared eval set,
+// in a single launch. Replaces the reference's per-node sklearn calls
+// (gossipy/model/handler.py:282-334, 375-391) and the engine's previous
+// ~15-small-torch-ops path. argmax/AUC are computed on raw affine scores —
+// the sigmoid is monotonic, so predictions and ranks are unchanged.
+// ---------------------------------------------------------------------------
+
+struct EvalArgs {
+    const float* params;  // [n_local, D]
+    const int* nodes;     // [R] local row ids
+    const float* X;       // [n_eval, d]
+    const float* Y;       // [n_eval] class index (or +-1 when is_margin)
+    float* out;           // [R, 5] acc, prec, rec, f1, auc
+    int d, k, n_eval, D;
+    int is_margin;        // pegasos/adaline: score = <w, x>, label +-1
+};
+
+constexpr int EVAL_KMAX = 16;
+
+__global__ void __launch_bounds__(256)
+eval_metrics_kernel(EvalArgs a)
+{
+    int r = blockIdx.x;
+    int node = a.nodes[r];
+    int tid = threadIdx.x;
+    extern __shared__ float sm[];
+    float* W = sm;                       // D
+    float* s1 = W + a.D;                 // n_eval: positive-class score
+    int* conf = (int*)(s1 + a.n_eval);   // k*k confusion (true x pred)
+    int kk = a.is_margin ? 2 : a.k;
+
+    for (int e = tid; e < a.D; e += blockDim.x)
+        W[e] = a.params[(long)node * a.D + e];
+    for (int e = tid; e < kk * kk; e += blockDim.x) conf[e] = 0;
+    __syncthreads();
+
+    for (int sidx = tid; sidx < a.n_eval; sidx += blockDim.x) {
+        const float* x = a.X + (long)sidx * a.d;
+        int pred, yt;
+        float sc1;
+        if (a.is_margin) {
+            float acc = 0.f;
+            for (int e = 0; e < a.d; ++e) acc += W[e] * x[e];
+            sc1 = acc;
+            pred = acc >= 0.f ? 1 : 0;
+            yt = a.Y[sidx] > 0.f ? 1 : 0;
+        } else {
+            float best = -1e30f;
+            int bj = 0;
+            float z1 = 0.f;
+            for (int j = 0; j < a.k; ++j) {
+                float acc = W[a.k * a.d + j];
+                const float* wrow = W + j * a.d;
+                for (int e = 0; e < a.d; ++e) acc += wrow[e] * x[e];
+                if (acc > best) { best = acc; bj = j; }
+                if (j == 1) z1 = acc;
+            }
+            sc1 = z1;
+            pred = bj;
+            yt = (int)a.Y[sidx];
+        }
+        s1[sidx] = sc1;
+        atomicAdd(&conf[yt * kk + pred], 1);
+    }
+    __syncthreads();
+
+    // pairwise AUC (binary only): wins over (pos, neg) pairs, ties 0.5 —
+    // the Mann-Whitney statistic (== average-rank AUC)
+    __shared__ float auc_wins;
+    __shared__ int npos_s;
+    if (tid == 0) { auc_wins = 0.f; npos_s = 0; }
+    __syncthreads();
+    if (kk == 2) {
+        float wins = 0.f;
+        int npos_l = 0;
+        for (int i = tid; i < a.n_eval; i += blockDim.x) {
+            bool pos_i = a.is_margin ? (a.Y[i] > 0.f) : ((int)a.Y[i] == 1);
+            if (!pos_i) continue;
+            npos_l += 1;
+            for (int j = 0; j < a.n_eval; ++j) {
+                bool pos_j = a.is_margin ? (a.Y[j] > 0.f) : ((int)a.Y[j] == 1);
+                if (pos_j) continue;
+                if (s1[i] > s1[j]) wins += 1.f;
+                else if (s1[i] == s1[j]) wins += 0.5f;
+            }
+        }
+        atomicAdd(&auc_wins, wins);
+        atomicAdd(&npos_s, npos_l);
+    }
+    __syncthreads();
+
+    if (tid == 0) {
+        int n = a.n_eval;
+        int correct = 0;
+        float prec = 0.f, rec = 0.f, f1 = 0.f;
+        for (int c = 0; c < kk; ++c) {
+            int tp = conf[c * kk + c];
+            correct += tp;
+            int pred_tot = 0, true_tot = 0;
+            for (int t2 = 0; t2 < kk; ++t2) {
+                pred_tot += conf[t2 * kk + c];
+                true_tot += conf[c * kk + t2];
+            }
+            float p = pred_tot > 0 ? (float)tp / pred_tot : 0.f;
+            float q = true_tot > 0 ? (float)tp / true_tot : 0.f;
+            prec += p;
+            rec += q;
+            f1 += (p + q) > 0.f ? 2.f * p * q / (p + q) : 0.f;
+        }
+        float* o = a.out + (long)r * 5;
+        o[0] = (float)correct / n;
+        o[1] = prec / kk;
+        o[2] = rec / kk;
+        o[3] = f1 / kk;
+        if (kk == 2) {
+            int npos = npos_s, nneg = n - npos_s;
+            o[4] = (npos == 0 || nneg == 0)
+                       ? 0.5f
+                       : auc_wins / ((float)npos * nneg);
+        } else {
+            o[4] = -1.f;  // AUC undefined for k > 2 (reference parity)
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
 // host wrappers
 // ---------------------------------------------------------------------------
 
@@ -1685,6 +1812,32 @@ void run_round_mlp(
     }
 }
 
+torch::Tensor eval_metrics(torch::Tensor params, torch::Tensor nodes,
+                           torch::Tensor X, torch::Tensor Y, int64_t d,
+                           int64_t k, bool is_margin)
+{
+    CHECK_DEV(params); CHECK_DEV(nodes); CHECK_DEV(X); CHECK_DEV(Y);
+    TORCH_CHECK(k <= EVAL_KMAX, "n_classes > ", EVAL_KMAX, " unsupported");
+    int R = nodes.size(0);
+    int n_eval = X.size(0);
+    auto out = torch::empty({R, 5}, params.options());
+    if (R == 0) return out;
+    EvalArgs a;
+    a.params = params.data_ptr<float>();
+    a.nodes = nodes.data_ptr<int>();
+    a.X = X.data_ptr<float>();
+    a.Y = Y.data_ptr<float>();
+    a.out = out.data_ptr<float>();
+    a.d = d; a.k = k; a.n_eval = n_eval; a.D = params.size(1);
+    a.is_margin = is_margin;
+    int kk = is_margin ? 2 : (int)k;
+    size_t smem = sizeof(float) * (a.D + n_eval) + sizeof(int) * kk * kk;
+    TORCH_CHECK(smem <= 160 * 1024, "eval LDS budget exceeded: ", smem);
+    hipLaunchKernelGGL(eval_metrics_kernel, dim3(R), dim3(256), smem,
+                       current_stream(), a);
+    return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m)
 {
     m.def("snapshot", &snapshot, "batched model snapshot (arena row copy)");
@@ -1706,6 +1859,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m)
           "fused item-block merge + per-rating MF SGD tick (K9/K10)");
     m.def("run_round_mlp", &run_round_mlp,
           "whole-round executor, MLP family");
+    m.def("eval_metrics", &eval_metrics,
+          "fused per-node eval metrics on a shared eval set (K13)");
     m.def("tick_kmeans", &tick_kmeans,
           "fused centroid merge + assign/EMA k-means tick (K11)");
 }

@@ -1087,3 +1087,121 @@ class TestKMeansEngine:
         sim.start(n_rounds=10)
         evals = rep.get_evaluation(False)
         assert evals[-1][1]["nmi"] > 0.6
+
+
+# ---------------------------------------------------------------------------
+# pass-through and cache-neighborhood gossip (Giaretta 2019)
+# ---------------------------------------------------------------------------
+
+
+def _star_csr(n):
+    """Star topology: node 0 is the hub (degree n-1), others are leaves."""
+    deg = [n - 1] + [1] * (n - 1)
+    indptr = np.concatenate([[0], np.cumsum(deg)]).astype(np.int64)
+    indices = np.concatenate(
+        [np.arange(1, n), np.zeros(n - 1)]
+    ).astype(np.int64)
+    return indptr, indices
+
+
+class TestPassThroughEngine:
+    def _run(self, rounds=10, n_nodes=30, **cfg_kw):
+        shards, geval = _make_data(n_nodes, seed=1)
+        data = DataArena.from_shards(shards, torch.device("cpu"), global_eval=geval)
+        base = dict(
+            n_nodes=n_nodes, delta=10, protocol=AntiEntropyProtocol.PUSH,
+            model_size=116, sampling_eval=0.25, seed=31, pass_through=True,
+        )
+        base.update(cfg_kw)
+        cfg = EngineConfig(**base)
+        spec = LogRegSpec(d_in=57, n_classes=2, lr=0.1, pass_through=True)
+        sim = BatchedGossipSimulator(cfg, spec, data)
+        rep = SimulationReport()
+        sim.add_receiver(rep)
+        sim.init_nodes()
+        sim.start(n_rounds=rounds)
+        return sim, rep
+
+    def test_learns_full_mesh(self):
+        # full mesh: equal degrees -> accept prob 1 -> behaves like plain
+        sim, rep = self._run()
+        assert rep.get_evaluation(False)[-1][1]["accuracy"] > 0.9
+
+    def test_deterministic(self):
+        s1, _ = self._run(rounds=4)
+        s2, _ = self._run(rounds=4)
+        assert torch.equal(s1.local_params(), s2.local_params())
+
+    def test_star_topology_hub_passes(self):
+        """On a star, leaf->hub deliveries should mostly resolve to PASS
+        (accept prob = 1/(n-1), gossipy/node.py:380-386)."""
+        from gossipy_amd.engine import Scheduler
+
+        n = 20
+        indptr, indices = _star_csr(n)
+        cfg = EngineConfig(
+            n_nodes=n, delta=10, protocol=AntiEntropyProtocol.PUSH,
+            model_size=116, seed=31, pass_through=True,
+            peers_indptr=indptr, peers_indices=indices,
+        )
+        s = Scheduler(cfg)
+        n_pass = n_norm = 0
+        for r in range(30):
+            sched = s.next_round(r)
+            for ph in sched.ticks:
+                for j, owner in enumerate(ph.del_owners):
+                    recv = None
+                    # find receiver of delivery j
+                    for i, rn in enumerate(ph.recv_nodes):
+                        if ph.recv_ptr[i] <= j < ph.recv_ptr[i + 1]:
+                            recv = int(rn)
+                    if recv == 0:  # leaf -> hub
+                        if ph.del_pids[j] == 1:
+                            n_pass += 1
+                        else:
+                            n_norm += 1
+        assert n_pass + n_norm > 100
+        frac = n_pass / (n_pass + n_norm)
+        assert 0.85 < frac < 1.0, frac  # expect ~ 1 - 1/19 = 0.947
+
+    def test_learns_star(self):
+        n = 30
+        indptr, indices = _star_csr(n)
+        sim, rep = self._run(
+            rounds=15, n_nodes=n, peers_indptr=indptr, peers_indices=indices
+        )
+        assert rep.get_evaluation(False)[-1][1]["accuracy"] > 0.8
+
+
+class TestCacheNeighEngine:
+    def _run(self, rounds=12, n_nodes=30, **cfg_kw):
+        from gossipy_amd.engine import BatchedCacheNeighGossipSimulator
+
+        shards, geval = _make_data(n_nodes, seed=1)
+        data = DataArena.from_shards(shards, torch.device("cpu"), global_eval=geval)
+        base = dict(
+            n_nodes=n_nodes, delta=10, protocol=AntiEntropyProtocol.PUSH,
+            model_size=116, sampling_eval=0.25, seed=33,
+        )
+        base.update(cfg_kw)
+        cfg = EngineConfig(**base)
+        spec = LogRegSpec(d_in=57, n_classes=2, lr=0.1)
+        sim = BatchedCacheNeighGossipSimulator(cfg, spec, data)
+        rep = SimulationReport()
+        sim.add_receiver(rep)
+        sim.init_nodes()
+        sim.start(n_rounds=rounds)
+        return sim, rep
+
+    def test_learns(self):
+        sim, rep = self._run()
+        assert rep.get_evaluation(False)[-1][1]["accuracy"] > 0.9
+
+    def test_deterministic(self):
+        s1, _ = self._run(rounds=4)
+        s2, _ = self._run(rounds=4)
+        assert torch.equal(s1.local_params(), s2.local_params())
+
+    def test_push_pull(self):
+        sim, rep = self._run(rounds=10, protocol=AntiEntropyProtocol.PUSH_PULL)
+        assert rep.get_evaluation(False)[-1][1]["accuracy"] > 0.85

@@ -657,3 +657,37 @@ class TestMLPFastPath:
             fast.local_params(), slow.local_params(), atol=1e-4, rtol=1e-4
         )
         assert torch.equal(fast.state.ages, slow.state.ages)
+
+
+class TestEvalKernel:
+    """Fused K13 eval kernel vs the torch metric implementation."""
+
+    def test_logreg_metrics_match(self):
+        spec = LogRegSpec(d_in=57, n_classes=2, lr=0.1)
+        cs, gs, cd, gd = _pair(10, spec, 57)
+        from gossipy_amd.engine.metrics import classification_metrics_shared
+
+        scores = TorchBackend().scores(cs, spec, torch.arange(10), cd.gx)
+        want = classification_metrics_shared(scores, cd.gy)
+        got = HIPBackend().eval_metrics_fast(
+            gs, spec, torch.arange(10), gd.gx, gd.gy
+        )
+        for w, g in zip(want, got):
+            for key in ("accuracy", "precision", "recall", "f1_score", "auc"):
+                assert abs(w[key] - g[key]) < 1e-3, (key, w[key], g[key])
+
+    def test_margin_metrics_match(self):
+        spec = PegasosSpec(d_in=57, lam=0.01)
+        cs, gs, cd, gd = _pair(8, spec, 57, pm1=True)
+        cs.params.normal_(generator=torch.Generator().manual_seed(4))
+        gs.params.copy_(cs.params)
+        from gossipy_amd.engine.metrics import binary_margin_metrics
+
+        scores = TorchBackend().scores(cs, spec, torch.arange(8), cd.gx)
+        want = binary_margin_metrics(scores[:, :, 0], cd.gy)
+        got = HIPBackend().eval_metrics_fast(
+            gs, spec, torch.arange(8), gd.gx, gd.gy
+        )
+        for w, g in zip(want, got):
+            for key in ("accuracy", "precision", "recall", "f1_score", "auc"):
+                assert abs(w[key] - g[key]) < 1e-3, (key, w[key], g[key])
