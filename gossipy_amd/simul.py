@@ -53,6 +53,7 @@ __all__ = [
     "GossipSimulator",
     "TokenizedGossipSimulator",
     "All2AllGossipSimulator",
+    "ThroughputTracer",
 ]
 
 
@@ -543,3 +544,60 @@ class All2AllGossipSimulator(GossipSimulator):
             LOG.warning("Simulation interrupted by user.")
 
         self.notify_end()
+
+
+class ThroughputTracer(SimulationEventReceiver):
+    """Tracing receiver: wall-clock per round and running rounds/sec.
+
+    The reference has no tracing; this receiver hangs off
+    the same observer interface the report uses (gossipy/simul.py:37-177)
+    and works with both the object layer and the batched engine. Attach
+    with ``sim.add_receiver(ThroughputTracer())``; read ``rounds_per_sec``
+    / ``round_times`` afterwards, or pass ``log_every`` to emit LOG lines
+    while running.
+    """
+
+    def __init__(self, delta: Optional[int] = None, log_every: int = 0):
+        self._delta = delta
+        self._log_every = log_every
+        self._t_prev: Optional[float] = None
+        self._last_t = -1
+        self.round_times: List[float] = []
+
+    def update_message(self, failed: bool, msg: Optional[Message] = None) -> None:
+        pass
+
+    def update_evaluation(self, round: int, on_user: bool, evals) -> None:
+        pass
+
+    def update_timestep(self, t: int) -> None:
+        import time as _time
+
+        if self._delta is None:
+            # first two timesteps reveal the cadence (the engine notifies
+            # once per round; the object layer once per tick)
+            if self._last_t >= 0 and self._delta is None:
+                self._delta = t - self._last_t
+            self._last_t = t
+        now = _time.perf_counter()
+        if self._t_prev is not None:
+            self.round_times.append(now - self._t_prev)
+            if self._log_every and len(self.round_times) % self._log_every == 0:
+                LOG.info(
+                    "round %d: %.1f rounds/sec",
+                    len(self.round_times),
+                    self.rounds_per_sec,
+                )
+        self._t_prev = now
+
+    def update_end(self) -> None:
+        self._t_prev = None
+
+    @property
+    def rounds_per_sec(self) -> float:
+        """Mean throughput over the recorded windows (each window is one
+        notify interval: a round on the engine, a tick on the object
+        layer)."""
+        if not self.round_times:
+            return 0.0
+        return len(self.round_times) / sum(self.round_times)

@@ -232,3 +232,17 @@ class TestAll2AllSimulator:
         evals = report.get_evaluation(False)
         assert len(evals) == 5
         assert evals[-1][1]["accuracy"] > 0.6
+
+
+def test_throughput_tracer_records_windows():
+    import time
+
+    from gossipy_amd.simul import ThroughputTracer
+
+    tr = ThroughputTracer()
+    for t in (99, 199, 299):
+        tr.update_timestep(t)
+        time.sleep(0.01)
+    assert len(tr.round_times) == 2
+    assert tr.rounds_per_sec > 0
+    tr.update_end()
