@@ -125,17 +125,26 @@ class TickPhase:
     wm_weights: np.ndarray = None  # float32 (aligned with wm_slots)
     wm_owners: np.ndarray = None  # int32 (slot writers, for the comm plan)
     wm_self_w: np.ndarray = None  # float32 [len(wm_nodes)]
+    # PENS step-1 events (Onoszko 2021): a full candidate cache is scored on
+    # the receiver's train data; the top-m are merged (the selection happens
+    # on-device — the schedule only lists the candidates)
+    pens_nodes: np.ndarray = None  # int32
+    pens_ptr: np.ndarray = None  # int32
+    pens_slots: np.ndarray = None  # int32 (candidate slots, arrival order)
+    pens_owners: np.ndarray = None  # int32 (candidate senders)
 
     @property
     def n_events(self) -> int:
         n_rep = 0 if self.rep_del_slots is None else len(self.rep_del_slots)
         n_wm = 0 if self.wm_nodes is None else len(self.wm_nodes)
+        n_pens = 0 if self.pens_nodes is None else len(self.pens_nodes)
         return (
             len(self.snap_nodes)
             + len(self.del_slots)
             + len(self.pull_snap_nodes)
             + n_rep
             + n_wm
+            + n_pens
         )
 
 
