@@ -17,6 +17,7 @@ from .runner import (
     BatchedAll2AllGossipSimulator,
     BatchedCacheNeighGossipSimulator,
     BatchedGossipSimulator,
+    BatchedPENSGossipSimulator,
     BatchedTokenizedGossipSimulator,
 )
 from .schedule import (
@@ -33,6 +34,7 @@ __all__ = [
     "BatchedTokenizedGossipSimulator",
     "BatchedAll2AllGossipSimulator",
     "BatchedCacheNeighGossipSimulator",
+    "BatchedPENSGossipSimulator",
     "TokenizedScheduler",
     "EngineConfig",
     "Scheduler",

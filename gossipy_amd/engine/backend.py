@@ -682,6 +682,45 @@ class TorchBackend:
         state.params[node] = saved_p
         state.ages[node] = saved_a
 
+    # -- PENS step-1 events (Onoszko 2021) -----------------------------------
+
+    def deliver_pens(
+        self, state, pool, data, spec, pens_nodes, pens_ptr, pens_slots,
+        pens_owners, counts, m_top: int,
+    ) -> None:
+        """Score every cached candidate on the receiver's train shard,
+        merge the top-m (mean including the own model,
+        gossipy/model/handler.py:260-280) and bump the winner counters
+        (gossipy/node.py:771-782). ``counts`` is ``[n_local, n_nodes]``
+        int32 (device-resident on the HIP backend)."""
+        assert spec.family == "logreg", "PENS engine path: logreg family"
+        ptr = pens_ptr.tolist()
+        d, k = spec.d_in, spec.n_classes
+        for i, node_t in enumerate(pens_nodes.tolist()):
+            c_n = int(data.counts[node_t])
+            x = data.x[node_t, :c_n]
+            y = data.y[node_t, :c_n].long()
+            cands = [int(pens_slots[j]) for j in range(ptr[i], ptr[i + 1])]
+            owners = [int(pens_owners[j]) for j in range(ptr[i], ptr[i + 1])]
+            accs = []
+            for slot in cands:
+                W = pool.slots[slot, : k * d].view(k, d)
+                b = pool.slots[slot, k * d :]
+                pred = (x @ W.t() + b).argmax(dim=1)
+                accs.append(float((pred == y).float().mean()))
+            # top-m by accuracy, stable in arrival order (node.py:776-777
+            # sorts on -accuracy; python sort is stable)
+            order = sorted(range(len(cands)), key=lambda j: -accs[j])[:m_top]
+            merged = state.params[node_t].clone()
+            age = int(state.ages[node_t])
+            for j in order:
+                merged += pool.slots[cands[j]]
+                age = max(age, int(pool.slot_ages[cands[j]]))
+                counts[node_t, owners[j]] += 1
+            state.params[node_t] = merged / (len(order) + 1)
+            state.ages[node_t] = age
+            self.update(state, data, spec, torch.tensor([node_t]))
+
     # -- all2all weighted merge ----------------------------------------------
 
     def deliver_weighted(
@@ -834,6 +873,36 @@ class HIPBackend(TorchBackend):
             wm_self_w.to(dev, torch.float32),
         )
         self.update(state, data, spec, nodes)
+
+    def deliver_pens(
+        self, state, pool, data, spec, pens_nodes, pens_ptr, pens_slots,
+        pens_owners, counts, m_top: int,
+    ) -> None:
+        assert spec.family == "logreg", "PENS engine path: logreg family"
+        if len(pens_nodes) == 0:
+            return
+        dev = state.params.device
+        self.ext.tick_pens(
+            state.params,
+            state.ages,
+            pool.slots,
+            pool.slot_ages,
+            pens_nodes.to(dev, torch.int32),
+            pens_ptr.to(dev, torch.int32),
+            pens_slots.to(dev, torch.int32),
+            pens_owners.to(dev, torch.int32),
+            counts,
+            data.x,
+            data.y,
+            data.counts,
+            spec.d_in,
+            spec.n_classes,
+            m_top,
+            spec.lr,
+            spec.weight_decay,
+            max(1, spec.local_epochs),
+            spec.batch_size,
+        )
 
     def eval_metrics_fast(self, state, spec, local_ids, gx, gy):
         """One-launch K13 evaluation for affine/margin families; returns a

@@ -51,6 +51,7 @@ __all__ = [
     "BatchedTokenizedGossipSimulator",
     "BatchedAll2AllGossipSimulator",
     "BatchedCacheNeighGossipSimulator",
+    "BatchedPENSGossipSimulator",
 ]
 
 
@@ -812,3 +813,108 @@ class BatchedCacheNeighGossipSimulator(BatchedGossipSimulator):
 
     def _fast_path_ok(self) -> bool:
         return False
+
+
+class BatchedPENSGossipSimulator(BatchedGossipSimulator):
+    """PENS on the batched engine (PENSNode, gossipy/node.py:663-785).
+
+    Step-1 scoring/merging runs as one kernel per tick (tick_pens) with
+    winner counts accumulated device-side; the only host<->device
+    dependency is ONE counts read at the step boundary, where
+    ``best_nodes`` is computed and the scheduler's step-2 peer draws are
+    restricted to it. PUSH + MERGE_UPDATE + logreg family.
+    """
+
+    def __init__(
+        self,
+        cfg: EngineConfig,
+        spec,
+        data: DataArena,
+        n_sampled: int = 10,
+        m_top: int = 2,
+        step1_rounds: int = 10,
+        device: Optional[torch.device] = None,
+    ):
+        super().__init__(cfg, spec, data, device=device)
+        from ..core import CreateModelMode
+        from .schedule import PENSScheduler
+
+        assert spec.mode == CreateModelMode.MERGE_UPDATE, (
+            "PENSNode can only be used with MERGE_UPDATE mode."
+        )
+        self.scheduler = PENSScheduler(cfg, n_sampled, m_top, step1_rounds)
+        self.m_top = m_top
+        #: device-side winner counters [n_local, n_nodes]
+        self.counts = torch.zeros(
+            self.n_local, cfg.n_nodes, dtype=torch.int32, device=self.device
+        )
+
+    def _fast_path_ok(self) -> bool:
+        return False
+
+    def _run_tick(self, phase: TickPhase) -> None:
+        super()._run_tick(phase)
+        if phase.pens_nodes is not None and len(phase.pens_nodes):
+            self._exchange(
+                self._plan_exchange(
+                    phase.pens_nodes, phase.pens_ptr, phase.pens_slots,
+                    phase.pens_owners,
+                )
+            )
+            mine = self._is_mine(phase.pens_nodes)
+            if mine.any():
+                cnt = np.diff(phase.pens_ptr)[mine]
+                sel = np.concatenate(
+                    [
+                        np.arange(phase.pens_ptr[i], phase.pens_ptr[i + 1])
+                        for i in np.where(mine)[0]
+                    ]
+                )
+                new_ptr = np.zeros(len(cnt) + 1, dtype=np.int64)
+                np.cumsum(cnt, out=new_ptr[1:])
+                self.backend.deliver_pens(
+                    self.state,
+                    self.pool,
+                    self.data,
+                    self.spec,
+                    self._to_local_t(phase.pens_nodes[mine]),
+                    torch.from_numpy(new_ptr),
+                    torch.from_numpy(phase.pens_slots[sel].astype(np.int64)),
+                    torch.from_numpy(phase.pens_owners[sel].astype(np.int64)),
+                    self.counts,
+                    self.m_top,
+                )
+
+    def _select_neighbors(self) -> None:
+        """Step boundary: read the device counters, assemble the full
+        [n_nodes, n_nodes] matrix on every rank, and fix step-2 topology."""
+        local = self.counts.cpu().numpy()
+        if self.world > 1:
+            gathered = [
+                torch.empty_like(self.counts) for _ in range(self.world)
+            ]
+            dist.all_gather(gathered, self.counts.contiguous())
+            full = torch.cat(gathered, dim=0).cpu().numpy()
+        else:
+            full = local
+        self.scheduler.select_neighbors(full.astype(np.int64))
+
+    def start(self, n_rounds: int = 100) -> None:
+        assert self.initialized, "call init_nodes() first"
+        for _ in range(n_rounds):
+            r = self.rounds_done
+            if (
+                r >= self.scheduler.step1_rounds
+                and self.scheduler.best_nodes is None
+            ):
+                self._select_neighbors()
+            sched = self.scheduler.next_round(r)
+            self.pool.ensure(sched.n_slots)
+            for phase in sched.ticks:
+                self._run_tick(phase)
+            if self.rank == 0:
+                self.notify_message_counts(sched)
+            self._evaluate(sched, (r + 1) * self.cfg.delta - 1)
+            self.rounds_done += 1
+            self.notify_timestep((r + 1) * self.cfg.delta - 1)
+        self.notify_end()

@@ -1218,6 +1218,212 @@ class CacheNeighScheduler(Scheduler):
         )
 
 
+class PENSScheduler(Scheduler):
+    """PENS — performance-based neighbor selection (Onoszko 2021;
+    PENSNode, gossipy/node.py:663-785). PUSH only, MERGE_UPDATE only.
+
+    Step 1 (rounds ``< step1_rounds``): a receiver caches one model slot
+    per sender; when the cache reaches ``n_sampled`` the schedule emits a
+    *PENS event* listing all candidates — the kernel scores each candidate
+    on the receiver's train data, merges the top-``m_top`` (mean incl. own
+    model) and counts the winners device-side; the schedule itself never
+    needs the accuracies. The node-side ``selected`` counters (how often a
+    peer was drawn as a send target, gossipy/node.py:746-749) are pure
+    host state.
+
+    At the step boundary the runner reads the device-side winner counts
+    and calls :meth:`select_neighbors`; step-2 peer draws are restricted
+    to each node's ``best_nodes`` (count > selected * m/n — node.py:
+    726-730), falling back to uniform when empty.
+    """
+
+    def __init__(self, cfg: EngineConfig, n_sampled: int = 10, m_top: int = 2,
+                 step1_rounds: int = 10):
+        super().__init__(cfg)
+        assert cfg.protocol == AntiEntropyProtocol.PUSH, (
+            "PENSNode only supports PUSH (gossipy/node.py:758-763)"
+        )
+        self.n_sampled = n_sampled
+        self.m_top = m_top
+        self.step1_rounds = step1_rounds
+        #: per-node candidate cache: sender -> slot, arrival-ordered
+        self._acc: List[Dict[int, int]] = [dict() for _ in range(cfg.n_nodes)]
+        self._refs: Dict[int, int] = {}
+        #: host-side target-draw counters (gossipy/node.py:746-749)
+        self.selected = np.zeros((cfg.n_nodes, cfg.n_nodes), dtype=np.int64)
+        self.best_nodes: Optional[List[np.ndarray]] = None
+
+    def select_neighbors(self, counts: np.ndarray) -> None:
+        """``counts[i, j]``: how often j's model made i's top-m (from the
+        device counter buffer). Computes each node's ``best_nodes``
+        (gossipy/node.py:728-730)."""
+        ratio = self.m_top / self.n_sampled
+        self.best_nodes = []
+        for i in range(self.cfg.n_nodes):
+            peers = self._peer_candidates(i)
+            best = peers[counts[i, peers] > self.selected[i, peers] * ratio]
+            self.best_nodes.append(best.astype(np.int64))
+
+    def _peer_candidates(self, i: int) -> np.ndarray:
+        cfg = self.cfg
+        if cfg.peers_indptr is None:
+            return np.concatenate([np.arange(i), np.arange(i + 1, cfg.n_nodes)])
+        s = int(cfg.peers_indptr[i])
+        e = int(cfg.peers_indptr[i + 1])
+        return np.asarray(cfg.peers_indices[s:e])
+
+    def _deref(self, slot: int, freed: List[int]) -> None:
+        r = self._refs.get(slot, 1) - 1
+        if r <= 0:
+            self._refs.pop(slot, None)
+            freed.append(slot)
+        else:
+            self._refs[slot] = r
+
+    def _peers_of(self, nodes: np.ndarray, t: int) -> np.ndarray:
+        """Step-2 peer draws come from best_nodes (node.py:751-755);
+        step 1 uses the base draw and bumps the selected counters."""
+        in_step2 = self.best_nodes is not None
+        if not in_step2:
+            peers = super()._peers_of(nodes, t)
+            for i, p in zip(nodes, peers):
+                self.selected[int(i), int(p)] += 1
+            return peers
+        g = self.tape.stream(Purpose.PEER, t)
+        n = len(nodes)
+        if n == 0:
+            return np.empty(0, dtype=np.int64)
+        u = np.atleast_1d(g.random(n))
+        out = np.empty(n, dtype=np.int64)
+        for j, i in enumerate(nodes):
+            best = self.best_nodes[int(i)]
+            cands = best if len(best) else self._peer_candidates(int(i))
+            out[j] = int(cands[int(np.floor(u[j] * len(cands)))])
+        return out
+
+    def next_round(self, r: int) -> RoundSchedule:
+        cfg = self.cfg
+        t0, t1 = r * cfg.delta, (r + 1) * cfg.delta
+        in_step1 = r < self.step1_rounds
+        sent = failed = total_size = 0
+        ticks: List[TickPhase] = []
+
+        for t in range(t0, t1):
+            freed: List[int] = []
+            firing = self._firing(t)
+
+            snap_nodes: List[int] = []
+            snap_slots: List[int] = []
+            n_f = len(firing)
+            if n_f:
+                peers = self._peers_of(firing, t)
+                if cfg.drop_prob <= 0.0:
+                    drop_u = np.ones(n_f)
+                else:
+                    drop_u = self.tape.uniform(Purpose.DROP, t, n_f)
+                delays = self._delays(t, n_f, np.full(n_f, cfg.model_size))
+                for j in range(n_f):
+                    sender, receiver = int(firing[j]), int(peers[j])
+                    slot = self._alloc_slot(sender)
+                    snap_nodes.append(sender)
+                    snap_slots.append(slot)
+                    sent += 1
+                    total_size += cfg.model_size
+                    if drop_u[j] >= cfg.drop_prob:
+                        self._pending.setdefault(t + int(delays[j]), []).append(
+                            (receiver, slot, -1, False, sender, -1)
+                        )
+                        self._refs[slot] = 1
+                    else:
+                        failed += 1
+                        freed.append(slot)
+
+            if cfg.online_prob >= 1.0:
+                online = np.ones(cfg.n_nodes, dtype=bool)
+            else:
+                online_u = self.tape.uniform(Purpose.ONLINE, t, cfg.n_nodes)
+                online = online_u <= cfg.online_prob
+
+            # deliveries
+            pens_nodes: List[int] = []
+            pens_ptr = [0]
+            pens_slots: List[int] = []
+            pens_owners: List[int] = []
+            recv_map: Dict[int, List[Tuple[int, int]]] = {}
+            for receiver, slot, _rf, _ip, sender, _pid in self._pending.pop(t, []):
+                if not online[receiver]:
+                    failed += 1
+                    self._deref(slot, freed)
+                    continue
+                if in_step1:
+                    acc = self._acc[receiver]
+                    old = acc.get(sender)
+                    if old is not None:
+                        self._deref(old, freed)
+                    acc[sender] = slot
+                    if len(acc) >= self.n_sampled:
+                        # emit the scoring/merge event with every candidate
+                        pens_nodes.append(receiver)
+                        for snd, sl in acc.items():
+                            pens_slots.append(sl)
+                            pens_owners.append(snd)
+                            self._deref(sl, freed)
+                        pens_ptr.append(len(pens_slots))
+                        acc.clear()
+                else:
+                    # step 2: ordinary MERGE_UPDATE delivery
+                    recv_map.setdefault(receiver, []).append((slot, sender))
+                    self._deref(slot, freed)
+
+            recv_nodes = np.fromiter(recv_map.keys(), dtype=np.int32, count=len(recv_map))
+            recv_ptr = np.zeros(len(recv_map) + 1, dtype=np.int32)
+            del_slots: List[int] = []
+            del_owners: List[int] = []
+            for i, rn in enumerate(recv_nodes):
+                pairs = recv_map[int(rn)]
+                del_slots.extend(p[0] for p in pairs)
+                del_owners.extend(p[1] for p in pairs)
+                recv_ptr[i + 1] = recv_ptr[i] + len(pairs)
+
+            phase = TickPhase(
+                t=t,
+                snap_nodes=np.asarray(snap_nodes, dtype=np.int32),
+                snap_slots=np.asarray(snap_slots, dtype=np.int32),
+                recv_nodes=recv_nodes,
+                recv_ptr=recv_ptr,
+                del_slots=np.asarray(del_slots, dtype=np.int32),
+                del_owners=np.asarray(del_owners, dtype=np.int32),
+                reply_slots=np.full(len(del_slots), -1, dtype=np.int32),
+                pull_snap_nodes=np.zeros(0, dtype=np.int32),
+                pull_snap_slots=np.zeros(0, dtype=np.int32),
+                del_pids=np.full(len(del_slots), -1, dtype=np.int32),
+                pens_nodes=np.asarray(pens_nodes, dtype=np.int32),
+                pens_ptr=np.asarray(pens_ptr, dtype=np.int32),
+                pens_slots=np.asarray(pens_slots, dtype=np.int32),
+                pens_owners=np.asarray(pens_owners, dtype=np.int32),
+            )
+            if phase.n_events:
+                ticks.append(phase)
+            self._free_slots.extend(freed)
+
+        eval_nodes = None
+        if cfg.sampling_eval > 0:
+            g = self.tape.stream(Purpose.EVAL, t1 - 1)
+            k = max(int(cfg.n_nodes * cfg.sampling_eval), 1)
+            eval_nodes = np.atleast_1d(g.integers(0, cfg.n_nodes, size=k))
+
+        return RoundSchedule(
+            round_idx=r,
+            ticks=ticks,
+            n_slots=self._next_slot,
+            slot_owner=self.slot_owner[: self._next_slot].copy(),
+            sent_messages=sent,
+            failed_messages=failed,
+            total_size=total_size,
+            eval_nodes=eval_nodes,
+        )
+
+
 class NativeSchedulerAdapter:
     """Adapter over the C++ scheduler (``csrc/scheduler.cpp``).
 

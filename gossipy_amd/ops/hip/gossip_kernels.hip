@@ -1212,6 +1212,120 @@ eval_metrics_kernel(EvalArgs a)
 }
 
 // ---------------------------------------------------------------------------
+// PENS step-1 event kernel (Onoszko 2021; gossipy/node.py:767-782): one
+// workgroup per event — score all cached candidate models on the
+// receiver's train shard (argmax accuracy; the sigmoid is monotonic so
+// raw affine scores give the reference's predictions), select the top-m
+// (stable in arrival order), merge them as a (m+1)-way mean with the own
+// model, run the local update, and count the winning senders.
+// ---------------------------------------------------------------------------
+
+constexpr int PENS_MAX_CAND = 64;
+
+struct PensArgs {
+    float* params; int* ages;
+    const float* slots; const int* slot_ages;
+    const int* nodes;       // [E] event receivers (unique within a launch)
+    const int* ptr;         // [E+1] candidate ranges
+    const int* cslots;      // candidate slots
+    const int* cowners;     // candidate senders (global node ids)
+    int* counts;            // [n_local, n_total] winner counters
+    const float* X; const float* Y; const int* dcounts;
+    int d, k, Smax, D, m_top, n_total;
+    float lr, wd;
+    int epochs, bs;
+};
+
+__global__ void __launch_bounds__(128)
+tick_pens_kernel(PensArgs a)
+{
+    int i = blockIdx.x;
+    int node = a.nodes[i];
+    int tid = threadIdx.x;
+    int lo = a.ptr[i], hi = a.ptr[i + 1];
+    int n_cand = hi - lo;
+    extern __shared__ float sm[];
+    float* W = sm;                      // D
+    float* xb = W + a.D;                // bsmax*d (update scratch)
+    int bsmax = (a.bs == 0) ? a.Smax : min(a.bs, a.Smax);
+    float* dz = xb + bsmax * a.d;       // bsmax*k
+    __shared__ int correct[PENS_MAX_CAND];
+    __shared__ int sel[PENS_MAX_CAND];
+
+    int c_n = a.dcounts[node];
+    const float* Xn = a.X + (long)node * a.Smax * a.d;
+    const float* Yn = a.Y + (long)node * a.Smax;
+
+    // --- score candidates: thread = sample, loop candidates
+    for (int c = tid; c < n_cand; c += blockDim.x) correct[c] = 0;
+    __syncthreads();
+    for (int c = 0; c < n_cand; ++c) {
+        const float* Wc = a.slots + (long)a.cslots[lo + c] * a.D;
+        int my = 0;
+        for (int s = tid; s < c_n; s += blockDim.x) {
+            const float* x = Xn + (long)s * a.d;
+            float best = -1e30f;
+            int bj = 0;
+            for (int j = 0; j < a.k; ++j) {
+                float acc = Wc[a.k * a.d + j];
+                const float* wrow = Wc + j * a.d;
+                for (int e = 0; e < a.d; ++e) acc += wrow[e] * x[e];
+                if (acc > best) { best = acc; bj = j; }
+            }
+            if (bj == (int)Yn[s]) my += 1;
+        }
+        if (my) atomicAdd(&correct[c], my);
+    }
+    __syncthreads();
+
+    // --- stable top-m selection (thread 0; n_cand small)
+    if (tid == 0) {
+        for (int c = 0; c < n_cand; ++c) sel[c] = 0;
+        int m = min(a.m_top, n_cand);
+        for (int pick = 0; pick < m; ++pick) {
+            int best_c = -1, best_v = -1;
+            for (int c = 0; c < n_cand; ++c) {
+                if (!sel[c] && correct[c] > best_v) {
+                    best_v = correct[c];
+                    best_c = c;
+                }
+            }
+            sel[best_c] = 1;
+            atomicAdd(&a.counts[(long)node * a.n_total + a.cowners[lo + best_c]], 1);
+        }
+    }
+    __syncthreads();
+
+    // --- merge: (own + sum selected) / (m+1), age = max
+    for (int e = tid; e < a.D; e += blockDim.x)
+        W[e] = a.params[(long)node * a.D + e];
+    __syncthreads();
+    int n_sel = 0;
+    int age = a.ages[node];
+    for (int c = 0; c < n_cand; ++c) {
+        if (!sel[c]) continue;
+        n_sel += 1;
+        const float* Wc = a.slots + (long)a.cslots[lo + c] * a.D;
+        for (int e = tid; e < a.D; e += blockDim.x) W[e] += Wc[e];
+        age = max(age, a.slot_ages[a.cslots[lo + c]]);
+    }
+    float inv = 1.0f / (n_sel + 1);
+    for (int e = tid; e < a.D; e += blockDim.x) W[e] *= inv;
+    __syncthreads();
+
+    // --- local update
+    LogregArgs u;
+    u.X = a.X; u.Y = a.Y; u.counts = a.dcounts;
+    u.d = a.d; u.k = a.k; u.Smax = a.Smax; u.D = a.D;
+    u.lr = a.lr; u.wd = a.wd; u.epochs = a.epochs; u.bs = a.bs;
+    logreg_update(u, node, W, xb, dz, age);
+    __syncthreads();
+    for (int e = tid; e < a.D; e += blockDim.x)
+        a.params[(long)node * a.D + e] = W[e];
+    if (tid == 0) a.ages[node] = age;
+}
+
+// ---------------------------------------------------------------------------
 // host wrappers
 // ---------------------------------------------------------------------------
 
@@ -1812,6 +1926,36 @@ void run_round_mlp(
     }
 }
 
+void tick_pens(torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
+               torch::Tensor slot_ages, torch::Tensor nodes, torch::Tensor ptr,
+               torch::Tensor cslots, torch::Tensor cowners,
+               torch::Tensor counts, torch::Tensor X, torch::Tensor Y,
+               torch::Tensor dcounts, int64_t d, int64_t k, int64_t m_top,
+               double lr, double wd, int64_t epochs, int64_t bs)
+{
+    CHECK_DEV(params); CHECK_DEV(slots); CHECK_DEV(X); CHECK_DEV(counts);
+    TORCH_CHECK(k <= KMAX, "n_classes > ", KMAX, " unsupported");
+    int n = nodes.size(0);
+    if (n == 0) return;
+    PensArgs a;
+    a.params = params.data_ptr<float>(); a.ages = ages.data_ptr<int>();
+    a.slots = slots.data_ptr<float>(); a.slot_ages = slot_ages.data_ptr<int>();
+    a.nodes = nodes.data_ptr<int>(); a.ptr = ptr.data_ptr<int>();
+    a.cslots = cslots.data_ptr<int>(); a.cowners = cowners.data_ptr<int>();
+    a.counts = counts.data_ptr<int>();
+    a.X = X.data_ptr<float>(); a.Y = Y.data_ptr<float>();
+    a.dcounts = dcounts.data_ptr<int>();
+    a.d = d; a.k = k; a.Smax = X.size(1); a.D = params.size(1);
+    a.m_top = m_top; a.n_total = counts.size(1);
+    a.lr = lr; a.wd = wd; a.epochs = epochs; a.bs = bs;
+    int bsmax = (bs == 0) ? a.Smax : std::min<int>(bs, a.Smax);
+    size_t smem = sizeof(float) * (a.D + (size_t)bsmax * a.d +
+                                   (size_t)bsmax * a.k);
+    TORCH_CHECK(smem <= 160 * 1024, "pens LDS budget exceeded: ", smem);
+    hipLaunchKernelGGL(tick_pens_kernel, dim3(n), dim3(128), smem,
+                       current_stream(), a);
+}
+
 torch::Tensor eval_metrics(torch::Tensor params, torch::Tensor nodes,
                            torch::Tensor X, torch::Tensor Y, int64_t d,
                            int64_t k, bool is_margin)
@@ -1859,6 +2003,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m)
           "fused item-block merge + per-rating MF SGD tick (K9/K10)");
     m.def("run_round_mlp", &run_round_mlp,
           "whole-round executor, MLP family");
+    m.def("tick_pens", &tick_pens,
+          "PENS step-1 event: score candidates, merge top-m, count winners");
     m.def("eval_metrics", &eval_metrics,
           "fused per-node eval metrics on a shared eval set (K13)");
     m.def("tick_kmeans", &tick_kmeans,
