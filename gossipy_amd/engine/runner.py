@@ -152,10 +152,16 @@ class BatchedGossipSimulator(SimulationEventSender):
         point-to-point sends, which RCCL maps onto the direct xGMI link of
         each GPU pair.
         """
+        self._exchange_finish(self._exchange_start(needed))
+
+    def _exchange_start(self, needed):
+        """Issue the grouped P2P ops; returns the in-flight state for
+        :meth:`_exchange_finish` (None when there is nothing to move)."""
         if self.world == 1 or not needed:
-            return
+            return None
         ops = []
         recv_bufs = []
+        send_bufs = []  # keep alive until the waits complete
         D = self.pool.slots.shape[1]
         A = getattr(self.spec, "age_width", 1)
         for src, dst, slot_ids in needed:
@@ -167,13 +173,20 @@ class BatchedGossipSimulator(SimulationEventSender):
                 buf[:, :D] = self.pool.slots[ids]
                 buf[:, D:] = self.pool.slot_ages[ids].reshape(len(ids), A).float()
                 ops.append(dist.P2POp(dist.isend, buf, dst))
+                send_bufs.append(buf)
             elif dst == self.rank:
                 buf = torch.empty(len(slot_ids), D + A, device=self.device)
                 ops.append(dist.P2POp(dist.irecv, buf, src))
                 recv_bufs.append((slot_ids, buf))
-        if ops:
-            for w in dist.batch_isend_irecv(ops):
-                w.wait()
+        works = dist.batch_isend_irecv(ops) if ops else []
+        return (works, recv_bufs, send_bufs, D)
+
+    def _exchange_finish(self, pending) -> None:
+        if pending is None:
+            return
+        works, recv_bufs, _send_bufs, D = pending
+        for w in works:
+            w.wait()
         for slot_ids, buf in recv_bufs:
             ids = torch.from_numpy(slot_ids.astype(np.int64)).to(self.device)
             self.pool.slots[ids] = buf[:, :D]
@@ -205,6 +218,35 @@ class BatchedGossipSimulator(SimulationEventSender):
 
     # -- round execution -----------------------------------------------------
 
+    def _deliver_group(self, phase: TickPhase, recv_idx: np.ndarray) -> None:
+        """Launch the delivery kernel for the receiver rows ``recv_idx``
+        (indices into phase.recv_nodes)."""
+        if len(recv_idx) == 0:
+            return
+        counts = np.diff(phase.recv_ptr)[recv_idx]
+        sel = np.concatenate(
+            [
+                np.arange(phase.recv_ptr[i], phase.recv_ptr[i + 1])
+                for i in recv_idx
+            ]
+        )
+        new_ptr = np.zeros(len(counts) + 1, dtype=np.int64)
+        np.cumsum(counts, out=new_ptr[1:])
+        pids = None
+        if phase.del_pids is not None and len(phase.del_pids):
+            pids = torch.from_numpy(phase.del_pids[sel].astype(np.int64))
+        self.backend.deliver(
+            self.state,
+            self.pool,
+            self.data,
+            self.spec,
+            self._to_local_t(phase.recv_nodes[recv_idx]),
+            torch.from_numpy(new_ptr),
+            torch.from_numpy(phase.del_slots[sel].astype(np.int64)),
+            torch.from_numpy(phase.reply_slots[sel].astype(np.int64)),
+            del_pids=pids,
+        )
+
     def _run_tick(self, phase: TickPhase) -> None:
         # A: snapshots of firing nodes
         mine = self._is_mine(phase.snap_nodes)
@@ -216,40 +258,29 @@ class BatchedGossipSimulator(SimulationEventSender):
                 torch.from_numpy(phase.snap_slots[mine].astype(np.int64)).to(self.device),
                 src_off=self._snap_off,
             )
-        # move cross-GPU slots needed by this tick's deliveries
-        self._exchange(
-            self._plan_exchange(
-                phase.recv_nodes, phase.recv_ptr, phase.del_slots, phase.del_owners
-            )
-        )
 
-        # B: deliveries (merge + update [+ reply snapshot]) and PULL snapshots
+        # B: deliveries (merge + update [+ reply snapshot]) and PULL snapshots.
+        # Comm/compute overlap: receivers whose deliveries are all locally
+        # sourced launch while the cross-GPU slot exchange is in flight;
+        # receivers touching remote slots launch after it completes.
         rmine = self._is_mine(phase.recv_nodes)
-        if rmine.any():
-            # compress the CSR to my receivers
-            counts = np.diff(phase.recv_ptr)[rmine]
-            sel = np.concatenate(
-                [
-                    np.arange(phase.recv_ptr[i], phase.recv_ptr[i + 1])
-                    for i in np.where(rmine)[0]
-                ]
-            )
-            new_ptr = np.zeros(len(counts) + 1, dtype=np.int64)
-            np.cumsum(counts, out=new_ptr[1:])
-            pids = None
-            if phase.del_pids is not None and len(phase.del_pids):
-                pids = torch.from_numpy(phase.del_pids[sel].astype(np.int64))
-            self.backend.deliver(
-                self.state,
-                self.pool,
-                self.data,
-                self.spec,
-                self._to_local_t(phase.recv_nodes[rmine]),
-                torch.from_numpy(new_ptr),
-                torch.from_numpy(phase.del_slots[sel].astype(np.int64)),
-                torch.from_numpy(phase.reply_slots[sel].astype(np.int64)),
-                del_pids=pids,
-            )
+        plan = self._plan_exchange(
+            phase.recv_nodes, phase.recv_ptr, phase.del_slots, phase.del_owners
+        )
+        if not plan:
+            self._exchange(plan)
+            if rmine.any():
+                self._deliver_group(phase, np.where(rmine)[0])
+        else:
+            src_rank = self._rank_of(phase.del_owners)
+            has_remote = np.zeros(len(phase.recv_nodes), dtype=bool)
+            for i in range(len(phase.recv_nodes)):
+                seg = src_rank[phase.recv_ptr[i] : phase.recv_ptr[i + 1]]
+                has_remote[i] = (seg != self.rank).any()
+            pending = self._exchange_start(plan)
+            self._deliver_group(phase, np.where(rmine & ~has_remote)[0])
+            self._exchange_finish(pending)
+            self._deliver_group(phase, np.where(rmine & has_remote)[0])
         pmine = self._is_mine(phase.pull_snap_nodes)
         if pmine.any():
             self.backend.snapshot(
