@@ -1,8 +1,13 @@
 """Giaretta & Girdzijauskas 2019 — gossip learning on skewed topologies.
 
 Engine-first equivalent of the reference's main_giaretta_2019.py (Pegasos
-on a Barabasi-Albert scale-free graph). The BA graph is generated with a
-plain numpy preferential-attachment loop (networkx is not available here).
+on a Barabasi-Albert scale-free graph), plus the paper's two protocol
+remedies for power-law topologies: ``--variant passthrough`` (hubs adopt
+low-degree models with prob 1 - deg_s/deg_r and relay them,
+gossipy/node.py:289-392) and ``--variant cacheneigh`` (per-neighbor model
+slots merged lazily at send time, gossipy/node.py:395-496). The BA graph
+is generated with a plain numpy preferential-attachment loop (networkx is
+not available here).
 """
 
 import os
@@ -18,6 +23,7 @@ import torch
 from gossipy_amd.core import AntiEntropyProtocol
 from gossipy_amd.data import make_synthetic_classification
 from gossipy_amd.engine import (
+    BatchedCacheNeighGossipSimulator,
     BatchedGossipSimulator,
     DataArena,
     EngineConfig,
@@ -60,6 +66,11 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--nodes", type=int, default=100)
     ap.add_argument("--rounds", type=int, default=100)
+    ap.add_argument(
+        "--variant",
+        choices=["plain", "passthrough", "cacheneigh"],
+        default="plain",
+    )
     args = ap.parse_args()
 
     device = torch.device("cuda:0" if torch.cuda.is_available() else "cpu")
@@ -83,8 +94,15 @@ def main():
         seed=42,
         peers_indptr=indptr,
         peers_indices=indices,
+        pass_through=args.variant == "passthrough",
     )
-    sim = BatchedGossipSimulator(cfg, PegasosSpec(d_in=d, lam=0.01), data, device=device)
+    spec = PegasosSpec(
+        d_in=d, lam=0.01, pass_through=args.variant == "passthrough"
+    )
+    if args.variant == "cacheneigh":
+        sim = BatchedCacheNeighGossipSimulator(cfg, spec, data, device=device)
+    else:
+        sim = BatchedGossipSimulator(cfg, spec, data, device=device)
     report = SimulationReport()
     sim.add_receiver(report)
     sim.init_nodes()

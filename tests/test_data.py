@@ -175,3 +175,14 @@ class TestSynthetic:
         assert set(r.keys()) == set(range(5))
         for u, lst in r.items():
             assert all(0 <= i < 12 and 1 <= v <= 5 for i, v in lst)
+
+
+def test_sklearn_builtin_loaders_offline():
+    """The sklearn-bundled datasets need no network (parity with
+    gossipy/data/__init__.py:561-624 for the offline subset)."""
+    from gossipy_amd.data import load_classification_dataset
+
+    for name, shape in (("iris", (150, 4)), ("wine", (178, 13))):
+        X, y = load_classification_dataset(name, as_tensor=True)
+        assert tuple(X.shape) == shape
+        assert len(y) == shape[0]
