@@ -88,8 +88,6 @@ def main():
     print(f"final local eval: {report.get_evaluation(True)[-1][1]}")
 
 
-if __name__ == "__main__":
-    main()
 
 
 def main_engine(args):
@@ -128,3 +126,7 @@ def main_engine(args):
     print(f"final global eval: {report.get_evaluation(False)[-1][1]}")
     n_best = sum(len(b) for b in sim.scheduler.best_nodes or [])
     print(f"selected neighbors across nodes: {n_best}")
+
+
+if __name__ == "__main__":
+    main()
