@@ -113,6 +113,9 @@ class BatchedGossipSimulator(SimulationEventSender):
         self.scheduler = make_scheduler(cfg)
         self.initialized = False
         self.rounds_done = 0
+        #: subclasses with non-standard phases (wm/pens) opt out of the
+        #: flatten-to-round-executor path
+        self._flat_schedulable = True
         #: host mirror of per-round slot owners (set per round)
         self._slot_owner: Optional[np.ndarray] = None
 
@@ -438,11 +441,75 @@ class BatchedGossipSimulator(SimulationEventSender):
         return (
             self.world == 1
             and isinstance(self.scheduler, NativeSchedulerAdapter)
+            and self._flat_exec_ok()
+        )
+
+    def _flat_exec_ok(self) -> bool:
+        """Whether this spec can run through the whole-round C++ executor
+        (python-scheduled rounds are flattened into the same format)."""
+        return (
+            self.world == 1
             and getattr(self.backend, "ext", None) is not None
             and self.spec.family in ("logreg", "pegasos", "adaline", "mlp")
             and (self.spec.family == "logreg" or getattr(self.spec, "n_parts", 0) == 0)
             and getattr(self.spec, "sample_size", 0) == 0
+            and not getattr(self.spec, "pass_through", False)
         )
+
+    @staticmethod
+    def _flatten_phases(ticks) -> dict:
+        """Concatenate a python-scheduled round's TickPhases into the flat
+        per-launch-group arrays the C++ round executors consume (each phase
+        becomes one group; CSR pointers become globally cumulative)."""
+        f = {
+            "snap_nodes": [], "snap_slots": [], "snap_tptr": [0],
+            "recv_nodes": [], "recv_nptr": [0], "recv_tptr": [0],
+            "del_slots": [], "reply_slots": [], "del_pids": [],
+            "pull_nodes": [], "pull_slots": [], "pull_tptr": [0],
+            "rep_nodes": [], "rep_nptr": [0], "rep_tptr": [0],
+            "rep_slots": [], "rep_pids": [],
+        }
+        d_off = 0
+        e_off = 0
+        for ph in ticks:
+            f["snap_nodes"].append(ph.snap_nodes)
+            f["snap_slots"].append(ph.snap_slots)
+            f["snap_tptr"].append(f["snap_tptr"][-1] + len(ph.snap_nodes))
+            f["recv_nodes"].append(ph.recv_nodes)
+            if len(ph.recv_nodes):
+                f["recv_nptr"].extend(
+                    (np.asarray(ph.recv_ptr[1:]) + d_off).tolist()
+                )
+            d_off += len(ph.del_slots)
+            f["recv_tptr"].append(f["recv_tptr"][-1] + len(ph.recv_nodes))
+            f["del_slots"].append(ph.del_slots)
+            f["reply_slots"].append(ph.reply_slots)
+            if ph.del_pids is not None and len(ph.del_pids):
+                f["del_pids"].append(ph.del_pids)
+            f["pull_nodes"].append(ph.pull_snap_nodes)
+            f["pull_slots"].append(ph.pull_snap_slots)
+            f["pull_tptr"].append(f["pull_tptr"][-1] + len(ph.pull_snap_nodes))
+            rep_n = ph.rep_recv_nodes if ph.rep_recv_nodes is not None else np.zeros(0, np.int32)
+            f["rep_nodes"].append(rep_n)
+            if len(rep_n):
+                f["rep_nptr"].extend(
+                    (np.asarray(ph.rep_recv_ptr[1:]) + e_off).tolist()
+                )
+                e_off += len(ph.rep_del_slots)
+                f["rep_slots"].append(ph.rep_del_slots)
+                if ph.rep_pids is not None and len(ph.rep_pids):
+                    f["rep_pids"].append(ph.rep_pids)
+            f["rep_tptr"].append(f["rep_tptr"][-1] + len(rep_n))
+        out = {}
+        for k, v in f.items():
+            if k.endswith("tptr") or k in ("recv_nptr", "rep_nptr"):
+                out[k] = np.asarray(v, dtype=np.int32)
+            else:
+                out[k] = (
+                    np.concatenate(v).astype(np.int32)
+                    if v else np.zeros(0, dtype=np.int32)
+                )
+        return out
 
     def _run_round_fast(self, f: dict) -> None:
         """Upload the round's flat event arrays in one H2D copy and hand
@@ -587,8 +654,17 @@ class BatchedGossipSimulator(SimulationEventSender):
             else:
                 sched = self.scheduler.next_round(r)
                 self.pool.ensure(sched.n_slots)
-                for phase in sched.ticks:
-                    self._run_tick(phase)
+                if (
+                    self._flat_schedulable
+                    and self._flat_exec_ok()
+                    and sched.ticks
+                ):
+                    # python-scheduled round through the C++ round executor
+                    # (tokenized / cache-neigh / python-scheduler fallback)
+                    self._run_round_fast(self._flatten_phases(sched.ticks))
+                else:
+                    for phase in sched.ticks:
+                        self._run_tick(phase)
             if self.rank == 0:
                 # report accounting comes from the schedule (host-side)
                 self.notify_message_counts(sched)
@@ -775,6 +851,7 @@ class BatchedAll2AllGossipSimulator(BatchedGossipSimulator):
         from .schedule import All2AllScheduler
 
         self.scheduler = All2AllScheduler(cfg, mixing)
+        self._flat_schedulable = False
 
     def _fast_path_ok(self) -> bool:
         return False

@@ -742,3 +742,66 @@ class TestPENSGPU:
         torch.cuda.synchronize()
         assert rep.get_evaluation(False)[-1][1]["accuracy"] > 0.9
         assert int(sim.counts.sum()) > 0
+
+
+class TestFlattenedExecutor:
+    """Python-scheduled rounds (tokenized / cache-neigh) flattened into the
+    C++ round executor must match per-tick dispatch."""
+
+    def _tok(self, flat: bool):
+        from gossipy_amd.engine import BatchedTokenizedGossipSimulator
+        from gossipy_amd.flow_control import RandomizedTokenAccount
+
+        X, y = make_synthetic_classification((640, 57, 2), seed=0, margin=2.0)
+        idx = np.random.default_rng(0).permutation(640)
+        shards = [(X[s], y[s]) for s in np.array_split(idx[:576], 64)]
+        data = DataArena.from_shards(
+            shards, CUDA, global_eval=(X[idx[576:]], y[idx[576:]])
+        )
+        cfg = EngineConfig(
+            n_nodes=64, delta=10, protocol=AntiEntropyProtocol.PUSH,
+            model_size=116, sampling_eval=0.0, seed=11, n_parts=4,
+        )
+        sim = BatchedTokenizedGossipSimulator(
+            cfg, LogRegSpec(d_in=57, n_classes=2, lr=0.1, n_parts=4), data,
+            token_account=RandomizedTokenAccount(C=20, A=10), device=CUDA,
+        )
+        sim._flat_schedulable = flat
+        sim.init_nodes()
+        sim.start(n_rounds=5)
+        torch.cuda.synchronize()
+        return sim
+
+    def test_tokenized_flat_matches_tick(self):
+        a = self._tok(True)
+        b = self._tok(False)
+        assert torch.allclose(a.local_params(), b.local_params(), atol=1e-5)
+        assert torch.equal(a.state.ages, b.state.ages)
+
+    def _cn(self, flat: bool):
+        from gossipy_amd.engine import BatchedCacheNeighGossipSimulator
+
+        X, y = make_synthetic_classification((640, 57, 2), seed=0, margin=2.0)
+        idx = np.random.default_rng(0).permutation(640)
+        shards = [(X[s], y[s]) for s in np.array_split(idx[:576], 48)]
+        data = DataArena.from_shards(
+            shards, CUDA, global_eval=(X[idx[576:]], y[idx[576:]])
+        )
+        cfg = EngineConfig(
+            n_nodes=48, delta=10, protocol=AntiEntropyProtocol.PUSH_PULL,
+            model_size=116, sampling_eval=0.0, seed=33,
+        )
+        sim = BatchedCacheNeighGossipSimulator(
+            cfg, LogRegSpec(d_in=57, n_classes=2, lr=0.1), data, device=CUDA
+        )
+        sim._flat_schedulable = flat
+        sim.init_nodes()
+        sim.start(n_rounds=5)
+        torch.cuda.synchronize()
+        return sim
+
+    def test_cacheneigh_flat_matches_tick(self):
+        a = self._cn(True)
+        b = self._cn(False)
+        assert torch.allclose(a.local_params(), b.local_params(), atol=1e-5)
+        assert torch.equal(a.state.ages, b.state.ages)
