@@ -28,6 +28,17 @@ _GOLDEN = np.uint64(0x9E3779B97F4A7C15)
 _MIX1 = np.uint64(0xBF58476D1CE4E5B9)
 _MIX2 = np.uint64(0x94D049BB133111EB)
 _U53_INV = 1.0 / 9007199254740992.0  # 2^-53
+_M64 = (1 << 64) - 1
+
+
+def _splitmix64_int(x: int) -> int:
+    """Scalar splitmix64 on python ints — bit-identical to the vectorized
+    numpy version but ~10x faster for single draws (numpy scalar uint64
+    ops dominate the tokenized scheduler otherwise)."""
+    x = (x + 0x9E3779B97F4A7C15) & _M64
+    x = ((x ^ (x >> 30)) * 0xBF58476D1CE4E5B9) & _M64
+    x = ((x ^ (x >> 27)) * 0x94D049BB133111EB) & _M64
+    return x ^ (x >> 31)
 
 
 def _splitmix64(x: np.ndarray) -> np.ndarray:
@@ -70,17 +81,27 @@ class TapeStream:
         self._off += n
         return _splitmix64(self.key + idx)
 
+    def _raw1(self) -> int:
+        v = _splitmix64_int((int(self.key) + self._off) & _M64)
+        self._off += 1
+        return v
+
     def random(self, n: int = 1):
         """``n`` float64 uniforms in [0, 1) (53-bit)."""
+        if n == 1:
+            return float(self._raw1() >> 11) * _U53_INV
         u = (self._raw(n) >> np.uint64(11)).astype(np.float64) * _U53_INV
-        return u if n != 1 else float(u[0])
+        return u
 
     def integers(self, low: int, high: int, size: int = 1):
         """``size`` ints in [low, high) (scaled-double method — tiny bias,
         chosen for trivial cross-language reproducibility)."""
+        if size == 1:
+            u = float(self._raw1() >> 11) * _U53_INV
+            return low + int(np.floor(u * (high - low)))
         u = (self._raw(size) >> np.uint64(11)).astype(np.float64) * _U53_INV
         v = low + np.floor(u * (high - low)).astype(np.int64)
-        return v if size != 1 else int(v[0])
+        return v
 
     def uniform(self, low: float, high: float, size) -> np.ndarray:
         """Uniform floats in [low, high); ``size`` may be a tuple."""
@@ -113,11 +134,10 @@ class RandomTape:
         self.seed = int(seed) & 0xFFFFFFFFFFFFFFFF
 
     def stream_key(self, purpose: Purpose, t: int = 0, extra: int = 0) -> int:
-        k = np.uint64(self.seed)
-        k = _splitmix64(np.uint64([k ^ np.uint64(int(purpose))]))[0]
-        k = _splitmix64(np.uint64([k ^ np.uint64(int(t))]))[0]
-        k = _splitmix64(np.uint64([k ^ np.uint64(int(extra))]))[0]
-        return int(k)
+        k = _splitmix64_int(self.seed ^ int(purpose))
+        k = _splitmix64_int(k ^ int(t))
+        k = _splitmix64_int(k ^ int(extra))
+        return k
 
     def stream(self, purpose: Purpose, t: int = 0, extra: int = 0) -> TapeStream:
         return TapeStream(self.stream_key(purpose, t, extra))
