@@ -136,7 +136,7 @@ public:
 
     py::dict next_round(int64_t r);
 
-private:
+protected:
     int64_t n_, delta_;
     int proto_;
     int64_t model_size_;
@@ -460,6 +460,464 @@ py::dict NativeScheduler::next_round(int64_t r)
     return out;
 }
 
+
+// ---------------------------------------------------------------------------
+// Tokenized (flow-controlled) scheduler — C++ replica of
+// engine/schedule.py TokenizedScheduler (bit-exact; equivalence enforced by
+// tests/test_native_sched.py). Emits flat arrays whose "tick" groups are
+// the python scheduler's phases: one group per delivery WAVE (reactive
+// bursts with zero delay cascade within a tick) plus one group per
+// same-tick reply batch. Constant utility only (the reference experiments
+// use utility == 1, main_hegedus_2021.py:57); callables stay in python.
+// ---------------------------------------------------------------------------
+
+enum AccountKind {
+    ACC_PURELY_PROACTIVE = 0,
+    ACC_PURELY_REACTIVE = 1,
+    ACC_SIMPLE = 2,
+    ACC_GENERALIZED = 3,
+    ACC_RANDOMIZED = 4
+};
+
+struct Account {  // matches gossipy_amd/flow_control.py
+    int kind;
+    double capacity;   // C
+    double reactivity; // A
+    double k;          // purely-reactive multiplier
+    int64_t n_tokens = 0;
+
+    void add(int64_t n) { n_tokens += n; }
+    void sub(int64_t n) { n_tokens = std::max<int64_t>(0, n_tokens - n); }
+
+    double proactive() const
+    {
+        switch (kind) {
+        case ACC_PURELY_PROACTIVE: return 1.0;
+        case ACC_PURELY_REACTIVE: return 0.0;
+        case ACC_SIMPLE: return n_tokens >= capacity ? 1.0 : 0.0;
+        case ACC_GENERALIZED: return n_tokens >= capacity ? 1.0 : 0.0;
+        default:  // randomized: linear ramp (flow_control.py:116-123)
+            if ((double)n_tokens < reactivity - 1) return 0.0;
+            if ((double)n_tokens <= capacity)
+                return ((double)n_tokens - reactivity + 1) /
+                       (capacity - reactivity + 1);
+            return 1.0;
+        }
+    }
+
+    int64_t reactive(int64_t utility, double u) const
+    {
+        switch (kind) {
+        case ACC_PURELY_PROACTIVE: return 0;
+        case ACC_PURELY_REACTIVE: return (int64_t)(utility * k);
+        case ACC_SIMPLE: return n_tokens > 0 ? 1 : 0;
+        case ACC_GENERALIZED: {
+            double num = reactivity + (double)n_tokens - 1;
+            return (int64_t)(utility > 0 ? num / reactivity
+                                         : num / (2 * reactivity));
+        }
+        default: {  // randomized rounding with the tape draw
+            if (utility <= 0) return 0;
+            double r = (double)n_tokens / reactivity;
+            double frac = r - std::floor(r);
+            return (int64_t)r + (u < frac ? 1 : 0);
+        }
+        }
+    }
+};
+
+class NativeTokenizedScheduler : public NativeScheduler {
+public:
+    NativeTokenizedScheduler(int64_t n_nodes, int64_t delta, int proto,
+                             int64_t model_size, double drop_prob,
+                             double online_prob, int delay_kind, int64_t dmin,
+                             int64_t dmax, double timexunit, int64_t overhead,
+                             bool sync, double sampling_eval, uint64_t seed,
+                             py::object peers_indptr, py::object peers_indices,
+                             int64_t n_parts, bool sampled, int account_kind,
+                             double acc_C, double acc_A, double acc_k,
+                             int64_t utility)
+        : NativeScheduler(n_nodes, delta, proto, model_size, drop_prob,
+                          online_prob, delay_kind, dmin, dmax, timexunit,
+                          overhead, sync, sampling_eval, seed, peers_indptr,
+                          peers_indices, n_parts, sampled),
+          utility_(utility)
+    {
+        Account proto_acc{account_kind, acc_C, acc_A, acc_k, 0};
+        accounts_.assign(n_, proto_acc);
+    }
+
+    py::dict next_round(int64_t r);
+
+    py::list token_balances() const
+    {
+        py::list out;
+        for (const auto& a : accounts_) out.append(a.n_tokens);
+        return out;
+    }
+
+private:
+    std::vector<Account> accounts_;
+    int64_t utility_;
+
+    int32_t reply_extra(int64_t t, int32_t replier)
+    {
+        if (n_parts_ > 0) {
+            Stream g(stream_key(seed_, PART, (uint64_t)t,
+                                (uint64_t)(1 + replier)));
+            return (int32_t)g.integers(0, n_parts_);
+        }
+        if (sampled_) {
+            Stream g(stream_key(seed_, SAMPLE, (uint64_t)t,
+                                (uint64_t)(1 + replier)));
+            return (int32_t)g.integers(0, (int64_t)1 << 31);
+        }
+        return -1;
+    }
+
+    int32_t burst_peer(int32_t node, Stream& gp)
+    {
+        if (indptr_.empty()) {
+            int64_t draw = gp.integers(0, n_ - 1);
+            return (int32_t)(draw + (draw >= node ? 1 : 0));
+        }
+        int64_t s = indptr_[node];
+        int64_t deg = indptr_[node + 1] - s;
+        return (int32_t)indices_[s + (int64_t)std::floor(gp.rnd() * (double)deg)];
+    }
+
+    int64_t burst_delay(Stream& gdl)
+    {
+        switch (dkind_) {
+        case D_CONST: return dmin_;
+        case D_UNIFORM: return gdl.integers(dmin_, dmax_ + 1);
+        default: return (int64_t)(timexunit_ * (double)model_size_) + overhead_;
+        }
+    }
+};
+
+py::dict NativeTokenizedScheduler::next_round(int64_t r)
+{
+    const int64_t t0 = r * delta_, t1 = (r + 1) * delta_;
+    int64_t sent = 0, failed = 0, total_size = 0;
+
+    std::vector<int32_t> snap_nodes, snap_slots, snap_tptr{0};
+    std::vector<int32_t> recv_nodes, recv_tptr{0}, recv_nptr{0};
+    std::vector<int32_t> del_slots, del_owners, reply_slots, del_pids;
+    std::vector<int32_t> pull_nodes, pull_slots, pull_tptr{0};
+    std::vector<int32_t> rep_nodes, rep_tptr{0}, rep_nptr{0};
+    std::vector<int32_t> rep_slots, rep_owners, rep_pids;
+    std::vector<double> online(n_);
+
+    // a "group" boundary closes every per-kind tick pointer
+    auto close_group = [&]() {
+        snap_tptr.push_back((int32_t)snap_nodes.size());
+        recv_tptr.push_back((int32_t)recv_nodes.size());
+        pull_tptr.push_back((int32_t)pull_nodes.size());
+        rep_tptr.push_back((int32_t)rep_nodes.size());
+    };
+
+    std::vector<int32_t> firing;
+    for (int64_t t = t0; t < t1; ++t) {
+        std::vector<int32_t> freed;
+        firing.clear();
+        if (sync_) {
+            firing = fire_buckets_[t % delta_];
+        } else {
+            for (int64_t i = 0; i < n_; ++i)
+                if (t % deltas_[i] == 0) firing.push_back((int32_t)i);
+        }
+
+        // --- proactive-gated sends
+        std::vector<int32_t> wave_snap_nodes, wave_snap_slots;
+        size_t n_f = firing.size();
+        if (n_f) {
+            Stream gpro(stream_key(seed_, TOKEN, (uint64_t)t, 0));
+            std::vector<double> pro_u(n_f);
+            for (size_t j = 0; j < n_f; ++j) pro_u[j] = gpro.rnd();
+            std::vector<int32_t> senders;
+            for (size_t j = 0; j < n_f; ++j) {
+                if (pro_u[j] < accounts_[firing[j]].proactive())
+                    senders.push_back(firing[j]);
+                else
+                    accounts_[firing[j]].add(1);
+            }
+            size_t n_s = senders.size();
+            if (n_s) {
+                // peers (python _peers_of draw order)
+                Stream gp(stream_key(seed_, PEER, (uint64_t)t, 0));
+                std::vector<int32_t> peers(n_s);
+                if (indptr_.empty()) {
+                    for (size_t j = 0; j < n_s; ++j) {
+                        int64_t draw = gp.integers(0, n_ - 1);
+                        peers[j] = (int32_t)(draw + (draw >= senders[j] ? 1 : 0));
+                    }
+                } else {
+                    std::vector<double> u(n_s);
+                    for (size_t j = 0; j < n_s; ++j) u[j] = gp.rnd();
+                    for (size_t j = 0; j < n_s; ++j) {
+                        int64_t st = indptr_[senders[j]];
+                        int64_t deg = indptr_[senders[j] + 1] - st;
+                        peers[j] = (int32_t)indices_[st + (int64_t)std::floor(
+                                                             u[j] * (double)deg)];
+                    }
+                }
+                Stream gdrop(stream_key(seed_, DROP, (uint64_t)t, 0));
+                std::vector<double> drop_u(n_s);
+                for (size_t j = 0; j < n_s; ++j) drop_u[j] = gdrop.rnd();
+                int64_t msize =
+                    (proto_ == PULL) ? 1 : model_size_;
+                std::vector<int64_t> delays(n_s);
+                {
+                    Stream gdl(stream_key(seed_, DELAY, (uint64_t)t, 0));
+                    for (size_t j = 0; j < n_s; ++j)
+                        delays[j] = delay_for(gdl, msize);
+                }
+                std::vector<int32_t> pids(n_s, -1);
+                if (n_parts_ > 0) {
+                    Stream gpt(stream_key(seed_, PART, (uint64_t)t, 0));
+                    for (size_t j = 0; j < n_s; ++j)
+                        pids[j] = (int32_t)gpt.integers(0, n_parts_);
+                } else if (sampled_) {
+                    Stream gsm(stream_key(seed_, SAMPLE, (uint64_t)t, 0));
+                    for (size_t j = 0; j < n_s; ++j)
+                        pids[j] = (int32_t)gsm.integers(0, (int64_t)1 << 31);
+                }
+                for (size_t j = 0; j < n_s; ++j) {
+                    int32_t sender = senders[j], receiver = peers[j];
+                    bool is_pull = proto_ == PULL;
+                    int32_t slot = -1;
+                    if (!is_pull) {
+                        slot = alloc_slot(sender);
+                        wave_snap_nodes.push_back(sender);
+                        wave_snap_slots.push_back(slot);
+                    }
+                    sent += 1;
+                    total_size += msize;
+                    if (drop_u[j] >= drop_) {
+                        int32_t rf = (proto_ == PUSH_PULL) ? -2 : -1;
+                        pending_[t + delays[j]].push_back(
+                            {receiver, slot, rf, is_pull, sender, pids[j]});
+                    } else {
+                        failed += 1;
+                        if (slot >= 0) freed.push_back(slot);
+                    }
+                }
+            }
+        }
+
+        if (online_ < 1.0) {
+            Stream go(stream_key(seed_, ONLINE, (uint64_t)t, 0));
+            for (int64_t i = 0; i < n_; ++i) online[i] = go.rnd();
+        } else {
+            std::fill(online.begin(), online.end(), 0.0);
+        }
+
+        // --- delivery waves with reactive bursts
+        std::vector<Msg> wave_due;
+        {
+            auto it = pending_.find(t);
+            if (it != pending_.end()) {
+                wave_due = std::move(it->second);
+                pending_.erase(it);
+            }
+        }
+        // per-(purpose, node) streams for this tick's bursts
+        std::map<std::pair<int, int32_t>, Stream> tstreams;
+        auto tick_stream = [&](int purpose, int32_t node) -> Stream& {
+            auto key = std::make_pair(purpose, node);
+            auto it = tstreams.find(key);
+            if (it == tstreams.end())
+                it = tstreams
+                         .emplace(key, Stream(stream_key(seed_, purpose,
+                                                         (uint64_t)t,
+                                                         (uint64_t)(1 + node))))
+                         .first;
+            return it->second;
+        };
+
+        while (!wave_due.empty() || !wave_snap_nodes.empty()) {
+            // group start: this wave's snapshots
+            for (size_t j = 0; j < wave_snap_nodes.size(); ++j) {
+                snap_nodes.push_back(wave_snap_nodes[j]);
+                snap_slots.push_back(wave_snap_slots[j]);
+            }
+            wave_snap_nodes.clear();
+            wave_snap_slots.clear();
+
+            std::vector<Msg> next_due;
+            std::vector<int32_t> order;
+            std::unordered_map<int32_t, std::vector<std::array<int32_t, 4>>> rmap;
+            for (const Msg& m : wave_due) {
+                if (online[m.receiver] > online_) {
+                    failed += 1;
+                    if (m.slot >= 0) freed.push_back(m.slot);
+                    continue;
+                }
+                if (m.is_pull) {
+                    int32_t rslot = alloc_slot(m.receiver);
+                    pull_nodes.push_back(m.receiver);
+                    pull_slots.push_back(rslot);
+                    if (!enqueue_reply(t, m.receiver, m.sender, rslot,
+                                       reply_extra(t, m.receiver), sent,
+                                       failed, total_size))
+                        freed.push_back(rslot);
+                    continue;
+                }
+                int32_t rslot = -1;
+                if (m.reply_flag == -2) {
+                    rslot = alloc_slot(m.receiver);
+                    if (!enqueue_reply(t, m.receiver, m.sender, rslot,
+                                       reply_extra(t, m.receiver), sent,
+                                       failed, total_size))
+                        freed.push_back(rslot);
+                }
+                auto f = rmap.find(m.receiver);
+                if (f == rmap.end()) {
+                    order.push_back(m.receiver);
+                    rmap[m.receiver] = {{m.slot, rslot, m.pid, m.sender}};
+                } else {
+                    f->second.push_back({m.slot, rslot, m.pid, m.sender});
+                }
+                freed.push_back(m.slot);
+                // reactive burst on reply-free deliveries (flag -1; REPLY
+                // payloads carry -3 and never react)
+                if (m.reply_flag == -1) {
+                    double ru = tick_stream(TOKEN, m.receiver).rnd();
+                    int64_t reaction =
+                        accounts_[m.receiver].reactive(utility_, ru);
+                    if (reaction <= 0) continue;
+                    accounts_[m.receiver].sub(reaction);
+                    Stream& gp = tick_stream(PEER, m.receiver);
+                    Stream& gd = tick_stream(DROP, m.receiver);
+                    Stream& gdl = tick_stream(DELAY, m.receiver);
+                    for (int64_t b = 0; b < reaction; ++b) {
+                        int32_t peer = burst_peer(m.receiver, gp);
+                        int32_t bslot = alloc_slot(m.receiver);
+                        wave_snap_nodes.push_back(m.receiver);
+                        wave_snap_slots.push_back(bslot);
+                        int32_t bpid = -1;
+                        if (n_parts_ > 0)
+                            bpid = (int32_t)tick_stream(PART, m.receiver)
+                                       .integers(0, n_parts_);
+                        else if (sampled_)
+                            bpid = (int32_t)tick_stream(SAMPLE, m.receiver)
+                                       .integers(0, (int64_t)1 << 31);
+                        sent += 1;
+                        total_size += model_size_;
+                        if (gd.rnd() >= drop_) {
+                            int64_t dly = burst_delay(gdl);
+                            int32_t rf = (proto_ == PUSH_PULL) ? -2 : -1;
+                            Msg bm{peer, bslot, rf, false, m.receiver, bpid};
+                            if (dly == 0)
+                                next_due.push_back(bm);
+                            else
+                                pending_[t + dly].push_back(bm);
+                        } else {
+                            failed += 1;
+                            freed.push_back(bslot);
+                        }
+                    }
+                }
+            }
+            for (int32_t rn : order) {
+                recv_nodes.push_back(rn);
+                for (auto& p : rmap[rn]) {
+                    del_slots.push_back(p[0]);
+                    reply_slots.push_back(p[1]);
+                    del_pids.push_back(p[2]);
+                    del_owners.push_back(p[3]);
+                }
+                recv_nptr.push_back((int32_t)del_slots.size());
+            }
+            close_group();
+            wave_due = std::move(next_due);
+        }
+
+        // --- same-tick replies: their own group
+        std::vector<Msg> rep_due;
+        {
+            auto it = pending_.find(t);
+            if (it != pending_.end()) {
+                rep_due = std::move(it->second);
+                pending_.erase(it);
+            }
+        }
+        if (!rep_due.empty()) {
+            std::vector<int32_t> rorder;
+            std::unordered_map<int32_t, std::vector<std::array<int32_t, 3>>> rrmap;
+            for (const Msg& m : rep_due) {
+                if (online[m.receiver] > online_) {
+                    failed += 1;
+                    freed.push_back(m.slot);
+                    continue;
+                }
+                auto f = rrmap.find(m.receiver);
+                if (f == rrmap.end()) {
+                    rorder.push_back(m.receiver);
+                    rrmap[m.receiver] = {{m.slot, m.sender, m.pid}};
+                } else {
+                    f->second.push_back({m.slot, m.sender, m.pid});
+                }
+                freed.push_back(m.slot);
+            }
+            for (int32_t rn : rorder) {
+                rep_nodes.push_back(rn);
+                for (auto& p : rrmap[rn]) {
+                    rep_slots.push_back(p[0]);
+                    rep_owners.push_back(p[1]);
+                    rep_pids.push_back(p[2]);
+                }
+                rep_nptr.push_back((int32_t)rep_slots.size());
+            }
+            close_group();
+        }
+
+        for (int32_t sfree : freed) free_slots_.push_back(sfree);
+    }
+
+    py::dict out;
+    auto arr = [](std::vector<int32_t>& v) {
+        auto a = py::array_t<int32_t>((py::ssize_t)v.size());
+        std::copy(v.begin(), v.end(), a.mutable_data());
+        return a;
+    };
+    out["snap_nodes"] = arr(snap_nodes);
+    out["snap_slots"] = arr(snap_slots);
+    out["snap_tptr"] = arr(snap_tptr);
+    out["recv_nodes"] = arr(recv_nodes);
+    out["recv_nptr"] = arr(recv_nptr);
+    out["recv_tptr"] = arr(recv_tptr);
+    out["del_slots"] = arr(del_slots);
+    out["del_owners"] = arr(del_owners);
+    out["reply_slots"] = arr(reply_slots);
+    out["del_pids"] = arr(del_pids);
+    out["pull_nodes"] = arr(pull_nodes);
+    out["pull_slots"] = arr(pull_slots);
+    out["pull_tptr"] = arr(pull_tptr);
+    out["rep_nodes"] = arr(rep_nodes);
+    out["rep_nptr"] = arr(rep_nptr);
+    out["rep_tptr"] = arr(rep_tptr);
+    out["rep_slots"] = arr(rep_slots);
+    out["rep_owners"] = arr(rep_owners);
+    out["rep_pids"] = arr(rep_pids);
+    out["sent"] = sent;
+    out["failed"] = failed;
+    out["total_size"] = total_size;
+    out["n_slots"] = next_slot_;
+    if (sampling_eval_ > 0) {
+        Stream g(stream_key(seed_, EVAL, (uint64_t)(t1 - 1), 0));
+        int64_t k = std::max<int64_t>((int64_t)(n_ * sampling_eval_), 1);
+        auto ev = py::array_t<int64_t>(k);
+        for (int64_t i = 0; i < k; ++i)
+            ev.mutable_data()[i] = g.integers(0, n_);
+        out["eval_nodes"] = ev;
+    } else {
+        out["eval_nodes"] = py::none();
+    }
+    return out;
+}
+
 PYBIND11_MODULE(_gossip_sched, m)
 {
     py::class_<NativeScheduler>(m, "NativeScheduler")
@@ -474,4 +932,11 @@ PYBIND11_MODULE(_gossip_sched, m)
              py::arg("peers_indptr"), py::arg("peers_indices"),
              py::arg("n_parts") = 0, py::arg("sampled") = false)
         .def("next_round", &NativeScheduler::next_round);
+    py::class_<NativeTokenizedScheduler>(m, "NativeTokenizedScheduler")
+        .def(py::init<int64_t, int64_t, int, int64_t, double, double, int,
+                      int64_t, int64_t, double, int64_t, bool, double,
+                      uint64_t, py::object, py::object, int64_t, bool, int,
+                      double, double, double, int64_t>())
+        .def("next_round", &NativeTokenizedScheduler::next_round)
+        .def("token_balances", &NativeTokenizedScheduler::token_balances);
 }

@@ -450,9 +450,8 @@ class BatchedGossipSimulator(SimulationEventSender):
         return (
             self.world == 1
             and getattr(self.backend, "ext", None) is not None
-            and self.spec.family in ("logreg", "pegasos", "adaline", "mlp")
+            and self.spec.family in ("logreg", "pegasos", "adaline", "mlp", "mf")
             and (self.spec.family == "logreg" or getattr(self.spec, "n_parts", 0) == 0)
-            and getattr(self.spec, "sample_size", 0) == 0
             and not getattr(self.spec, "pass_through", False)
         )
 
@@ -571,7 +570,30 @@ class BatchedGossipSimulator(SimulationEventSender):
         from .backend import _MODE_ID
 
         spec = self.spec
-        if getattr(spec, "n_parts", 0) > 0:
+        if getattr(spec, "sample_size", 0) > 0:
+            c = list(common)
+            c.insert(12, views["del_pids"])   # sample seeds ride the pid slot
+            c.insert(20, views["rep_pids"])
+            ext.run_round_logreg_samp(
+                *c,
+                spec.samp_count(),
+                spec.d_in,
+                spec.n_classes,
+                spec.lr,
+                spec.weight_decay,
+                max(1, spec.local_epochs),
+                spec.batch_size,
+                _MODE_ID[spec.mode],
+            )
+        elif spec.family == "mf":
+            ext.run_round_mf(
+                *common,
+                spec.k,
+                spec.n_items,
+                spec.reg,
+                spec.lr,
+            )
+        elif getattr(spec, "n_parts", 0) > 0:
             perm, pptr, apart = self.backend._part_dev(spec, dev)
             # run_round_logreg_part takes del_pids after reply_slots and
             # rep_pids after rep_slots
@@ -808,14 +830,39 @@ class BatchedTokenizedGossipSimulator(BatchedGossipSimulator):
         device: Optional[torch.device] = None,
     ):
         super().__init__(cfg, spec, data, device=device)
-        self.scheduler = TokenizedScheduler(cfg, token_account, utility_fun)
+        from .schedule import NativeTokenizedAdapter
+
+        self.scheduler = None
+        if (utility_fun is None or isinstance(utility_fun, int)) and self._flat_exec_ok():
+            try:
+                self.scheduler = NativeTokenizedAdapter(
+                    cfg,
+                    token_account,
+                    1 if utility_fun is None else utility_fun,
+                )
+            except (ImportError, TypeError):
+                self.scheduler = None
+        if self.scheduler is None:
+            self.scheduler = TokenizedScheduler(cfg, token_account, utility_fun)
 
     def _fast_path_ok(self) -> bool:
-        # burst waves are python-scheduled; kernels still run per wave
-        return False
+        from .schedule import NativeTokenizedAdapter
+
+        # native tokenized schedule feeds the whole-round executor; the
+        # python scheduler's waves go through the flatten path instead
+        return isinstance(self.scheduler, NativeTokenizedAdapter) and self._flat_exec_ok()
 
     @property
     def accounts(self):
+        from types import SimpleNamespace
+
+        from .schedule import NativeTokenizedAdapter
+
+        if isinstance(self.scheduler, NativeTokenizedAdapter):
+            return [
+                SimpleNamespace(n_tokens=b)
+                for b in self.scheduler.token_balances()
+            ]
         return self.scheduler.accounts
 
 

@@ -44,6 +44,7 @@ __all__ = [
     "TokenizedScheduler",
     "All2AllScheduler",
     "CacheNeighScheduler",
+    "NativeTokenizedAdapter",
     "NativeSchedulerAdapter",
     "make_scheduler",
 ]
@@ -1521,6 +1522,87 @@ class NativeSchedulerAdapter:
             total_size=int(f["total_size"]),
             eval_nodes=f["eval_nodes"],
         )
+
+
+class NativeTokenizedAdapter:
+    """Adapter over the C++ tokenized scheduler (``NativeTokenizedScheduler``
+    in csrc/scheduler.cpp) — bit-exact with :class:`TokenizedScheduler`
+    (tests/test_native_sched.py) and emitting flat per-wave launch groups
+    for the GPU round executor directly.
+
+    Covers the 5 in-tree token-account strategies and constant utilities
+    (the reference experiments use ``utility == 1``,
+    main_hegedus_2021.py:57); custom accounts or callable utilities fall
+    back to the python scheduler.
+    """
+
+    ACCOUNT_KINDS = {
+        "PurelyProactiveTokenAccount": 0,
+        "PurelyReactiveTokenAccount": 1,
+        "SimpleTokenAccount": 2,
+        "GeneralizedTokenAccount": 3,
+        "RandomizedTokenAccount": 4,
+    }
+
+    def __init__(self, cfg: EngineConfig, token_account, utility: int = 1):
+        from .. import ops
+
+        mod = ops.load_sched()
+        if mod is None or not hasattr(mod, "NativeTokenizedScheduler"):
+            raise ImportError("_gossip_sched.so missing NativeTokenizedScheduler")
+        kind = self.ACCOUNT_KINDS.get(type(token_account).__name__)
+        if kind is None:
+            raise TypeError("custom token accounts need the python scheduler")
+        if cfg.pass_through:
+            raise TypeError("pass-through needs the python scheduler")
+        d = cfg.delay
+        if isinstance(d, ConstantDelay):
+            dk, dmin, dmax, tx, ov = 0, d._delay, d._delay, 0.0, 0
+        elif isinstance(d, UniformDelay):
+            dk, dmin, dmax, tx, ov = 1, d._min_delay, d._max_delay, 0.0, 0
+        elif isinstance(d, LinearDelay):
+            dk, dmin, dmax, tx, ov = 2, 0, 0, d._timexunit, d._overhead
+        else:
+            raise TypeError("custom Delay subclasses need the python Scheduler")
+        self.cfg = cfg
+        ip, ix = cfg.peers_indptr, cfg.peers_indices
+        self._native = mod.NativeTokenizedScheduler(
+            cfg.n_nodes,
+            cfg.delta,
+            int(cfg.protocol.value),
+            cfg.model_size,
+            cfg.drop_prob,
+            cfg.online_prob,
+            dk,
+            dmin,
+            dmax,
+            tx,
+            ov,
+            cfg.sync,
+            cfg.sampling_eval,
+            cfg.seed & 0xFFFFFFFFFFFFFFFF,
+            None if ip is None else np.ascontiguousarray(ip, dtype=np.int64),
+            None if ix is None else np.ascontiguousarray(ix, dtype=np.int64),
+            cfg.n_parts,
+            cfg.sampled,
+            kind,
+            float(getattr(token_account, "capacity", 0.0)),
+            float(getattr(token_account, "reactivity", 0.0)),
+            float(getattr(token_account, "k", 0.0)),
+            int(utility),
+        )
+        self.last_flat: Optional[dict] = None
+
+    def token_balances(self):
+        return list(self._native.token_balances())
+
+    def next_round_flat(self, r: int) -> RoundSchedule:
+        f = self._native.next_round(r)
+        self.last_flat = f
+        return _flat_round_summary(f, r)
+
+    # the per-tick path is not used with this adapter (flat-exec only)
+    next_round = next_round_flat
 
 
 def make_scheduler(cfg: EngineConfig):

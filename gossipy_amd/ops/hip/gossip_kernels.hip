@@ -1982,6 +1982,142 @@ torch::Tensor eval_metrics(torch::Tensor params, torch::Tensor nodes,
     return out;
 }
 
+void run_round_logreg_samp(
+    torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
+    torch::Tensor slot_ages,
+    torch::Tensor snap_nodes, torch::Tensor snap_slots, torch::Tensor snap_tptr,
+    torch::Tensor recv_nodes, torch::Tensor recv_nptr, torch::Tensor recv_tptr,
+    torch::Tensor del_slots, torch::Tensor reply_slots, torch::Tensor del_pids,
+    torch::Tensor pull_nodes, torch::Tensor pull_slots, torch::Tensor pull_tptr,
+    torch::Tensor rep_nodes, torch::Tensor rep_nptr, torch::Tensor rep_tptr,
+    torch::Tensor rep_slots, torch::Tensor rep_pids,
+    torch::Tensor X, torch::Tensor Y, torch::Tensor counts,
+    int64_t samp_c, int64_t d, int64_t k, double lr, double wd,
+    int64_t epochs, int64_t bs, int64_t mode)
+{
+    CHECK_DEV(params); CHECK_DEV(slots); CHECK_DEV(X);
+    TORCH_CHECK(mode != MODE_PASS, "Mode PASS not allowed for sampled models.");
+    RoundArrays r = unpack_round(snap_nodes, snap_slots, snap_tptr, recv_nodes,
+                                 recv_nptr, recv_tptr, del_slots, reply_slots,
+                                 pull_nodes, pull_slots, pull_tptr, rep_nodes,
+                                 rep_nptr, rep_tptr, rep_slots);
+    const int* d_pids = del_pids.numel() ? del_pids.data_ptr<int>() : nullptr;
+    const int* r_pids = rep_pids.numel() ? rep_pids.data_ptr<int>() : nullptr;
+    hipStream_t s = current_stream();
+    LogregSampArgs a;
+    a.params = params.data_ptr<float>(); a.ages = ages.data_ptr<int>();
+    a.slots = slots.data_ptr<float>(); a.slot_ages = slot_ages.data_ptr<int>();
+    a.X = X.data_ptr<float>(); a.Y = Y.data_ptr<float>();
+    a.counts = counts.data_ptr<int>();
+    a.samp_c = samp_c; a.d = d; a.k = k; a.Smax = X.size(1);
+    a.D = params.size(1);
+    a.lr = lr; a.wd = wd; a.epochs = epochs; a.bs = bs; a.mode = mode;
+    a.update_only = 0;
+    int bsmax = (bs == 0) ? a.Smax : std::min<int>(bs, a.Smax);
+    size_t smem = sizeof(float) * (3 * a.D + (size_t)bsmax * a.d +
+                                   (size_t)bsmax * a.k);
+    TORCH_CHECK(smem <= 160 * 1024, "sampled logreg LDS budget exceeded");
+    for (int t = 0; t < r.delta; ++t) {
+        int s0 = r.snap_tptr[t], s1 = r.snap_tptr[t + 1];
+        if (s1 > s0)
+            launch_snap(a.params, a.ages, a.slots, a.slot_ages,
+                        r.snap_nodes + s0, r.snap_slots + s0, s1 - s0, a.D, s);
+        int r0 = r.recv_tptr[t], r1 = r.recv_tptr[t + 1];
+        if (r1 > r0) {
+            a.nodes = r.recv_nodes + r0;
+            a.ptr = r.recv_nptr + r0;
+            a.dslots = r.del_slots;
+            a.rslots = r.reply_slots;
+            a.dseeds = d_pids;
+            hipLaunchKernelGGL(tick_logreg_samp_kernel, dim3(r1 - r0),
+                               dim3(128), smem, s, a);
+        }
+        int p0 = r.pull_tptr[t], p1 = r.pull_tptr[t + 1];
+        if (p1 > p0)
+            launch_snap(a.params, a.ages, a.slots, a.slot_ages,
+                        r.pull_nodes + p0, r.pull_slots + p0, p1 - p0, a.D, s);
+        int q0 = r.rep_tptr[t], q1 = r.rep_tptr[t + 1];
+        if (q1 > q0) {
+            a.nodes = r.rep_nodes + q0;
+            a.ptr = r.rep_nptr + q0;
+            a.dslots = r.rep_slots;
+            a.rslots = nullptr;
+            a.dseeds = r_pids;
+            hipLaunchKernelGGL(tick_logreg_samp_kernel, dim3(q1 - q0),
+                               dim3(128), smem, s, a);
+        }
+    }
+}
+
+void run_round_mf(
+    torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
+    torch::Tensor slot_ages,
+    torch::Tensor snap_nodes, torch::Tensor snap_slots, torch::Tensor snap_tptr,
+    torch::Tensor recv_nodes, torch::Tensor recv_nptr, torch::Tensor recv_tptr,
+    torch::Tensor del_slots, torch::Tensor reply_slots,
+    torch::Tensor pull_nodes, torch::Tensor pull_slots, torch::Tensor pull_tptr,
+    torch::Tensor rep_nodes, torch::Tensor rep_nptr, torch::Tensor rep_tptr,
+    torch::Tensor rep_slots,
+    torch::Tensor X, torch::Tensor Y, torch::Tensor counts,
+    int64_t k, int64_t n_items, double reg, double lr)
+{
+    CHECK_DEV(params); CHECK_DEV(slots); CHECK_DEV(X);
+    RoundArrays r = unpack_round(snap_nodes, snap_slots, snap_tptr, recv_nodes,
+                                 recv_nptr, recv_tptr, del_slots, reply_slots,
+                                 pull_nodes, pull_slots, pull_tptr, rep_nodes,
+                                 rep_nptr, rep_tptr, rep_slots);
+    hipStream_t s = current_stream();
+    MFArgs a;
+    a.params = params.data_ptr<float>(); a.ages = ages.data_ptr<int>();
+    a.slots = slots.data_ptr<float>(); a.slot_ages = slot_ages.data_ptr<int>();
+    a.X = X.data_ptr<float>(); a.Y = Y.data_ptr<float>();
+    a.counts = counts.data_ptr<int>();
+    a.k = k; a.n_items = n_items; a.Smax = X.size(1); a.D = params.size(1);
+    a.item_off = k + 1; a.Wslot = n_items * (k + 1);
+    a.reg = reg; a.lr = lr; a.update_only = 0;
+    // MF snapshots carry only the item block (src_off = item_off)
+    for (int t = 0; t < r.delta; ++t) {
+        int s0 = r.snap_tptr[t], s1 = r.snap_tptr[t + 1];
+        if (s1 > s0) {
+            long total = (long)(s1 - s0) * a.Wslot;
+            int grid = (int)std::min<long>((total + 255) / 256, 2048);
+            hipLaunchKernelGGL(snapshot_kernel, dim3(grid), dim3(256), 0, s,
+                               a.params, a.ages, a.slots,
+                               slot_ages.data_ptr<int>(), r.snap_nodes + s0,
+                               r.snap_slots + s0, s1 - s0, a.Wslot, a.D,
+                               a.item_off, 1);
+        }
+        int r0 = r.recv_tptr[t], r1 = r.recv_tptr[t + 1];
+        if (r1 > r0) {
+            a.nodes = r.recv_nodes + r0;
+            a.ptr = r.recv_nptr + r0;
+            a.dslots = r.del_slots;
+            a.rslots = r.reply_slots;
+            hipLaunchKernelGGL(tick_mf_kernel, dim3(r1 - r0), dim3(WAVE), 0,
+                               s, a);
+        }
+        int p0 = r.pull_tptr[t], p1 = r.pull_tptr[t + 1];
+        if (p1 > p0) {
+            long total = (long)(p1 - p0) * a.Wslot;
+            int grid = (int)std::min<long>((total + 255) / 256, 2048);
+            hipLaunchKernelGGL(snapshot_kernel, dim3(grid), dim3(256), 0, s,
+                               a.params, a.ages, a.slots,
+                               slot_ages.data_ptr<int>(), r.pull_nodes + p0,
+                               r.pull_slots + p0, p1 - p0, a.Wslot, a.D,
+                               a.item_off, 1);
+        }
+        int q0 = r.rep_tptr[t], q1 = r.rep_tptr[t + 1];
+        if (q1 > q0) {
+            a.nodes = r.rep_nodes + q0;
+            a.ptr = r.rep_nptr + q0;
+            a.dslots = r.rep_slots;
+            a.rslots = nullptr;
+            hipLaunchKernelGGL(tick_mf_kernel, dim3(q1 - q0), dim3(WAVE), 0,
+                               s, a);
+        }
+    }
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m)
 {
     m.def("snapshot", &snapshot, "batched model snapshot (arena row copy)");
@@ -2003,6 +2139,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m)
           "fused item-block merge + per-rating MF SGD tick (K9/K10)");
     m.def("run_round_mlp", &run_round_mlp,
           "whole-round executor, MLP family");
+    m.def("run_round_logreg_samp", &run_round_logreg_samp,
+          "whole-round executor, sampled logreg family");
+    m.def("run_round_mf", &run_round_mf,
+          "whole-round executor, MF recommender family");
     m.def("tick_pens", &tick_pens,
           "PENS step-1 event: score candidates, merge top-m, count winners");
     m.def("eval_metrics", &eval_metrics,

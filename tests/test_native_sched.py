@@ -107,3 +107,81 @@ def test_native_matches_python(cfg_i):
                 ), f"round {r} tick {pa.t} field {f}: {va} vs {vb}"
         if a.eval_nodes is not None:
             assert np.array_equal(a.eval_nodes, b.eval_nodes)
+
+
+# ---------------------------------------------------------------------------
+# native tokenized scheduler: bit-exact vs the python TokenizedScheduler
+# (compared on flattened per-wave launch groups)
+# ---------------------------------------------------------------------------
+
+
+def _tok_cfgs():
+    base = dict(n_nodes=50, delta=10, model_size=116, seed=13)
+    yield EngineConfig(protocol=AntiEntropyProtocol.PUSH, **base), "randomized"
+    yield EngineConfig(
+        protocol=AntiEntropyProtocol.PUSH, n_parts=4, **base
+    ), "randomized"
+    yield EngineConfig(
+        protocol=AntiEntropyProtocol.PUSH, drop_prob=0.2, online_prob=0.8, **base
+    ), "randomized"
+    yield EngineConfig(
+        protocol=AntiEntropyProtocol.PUSH_PULL, delay=UniformDelay(0, 4), **base
+    ), "randomized"
+    yield EngineConfig(protocol=AntiEntropyProtocol.PUSH, **base), "proactive"
+    yield EngineConfig(protocol=AntiEntropyProtocol.PUSH, **base), "simple"
+    yield EngineConfig(protocol=AntiEntropyProtocol.PUSH, **base), "generalized"
+
+
+def _account(name):
+    from gossipy_amd.flow_control import (
+        GeneralizedTokenAccount,
+        PurelyProactiveTokenAccount,
+        RandomizedTokenAccount,
+        SimpleTokenAccount,
+    )
+
+    return {
+        "randomized": lambda: RandomizedTokenAccount(C=20, A=10),
+        "proactive": lambda: PurelyProactiveTokenAccount(),
+        "simple": lambda: SimpleTokenAccount(C=3),
+        "generalized": lambda: GeneralizedTokenAccount(C=8, A=4),
+    }[name]()
+
+
+@pytest.mark.parametrize("cfg_i", range(7))
+def test_native_tokenized_matches_python(cfg_i):
+    from gossipy_amd.engine.runner import BatchedGossipSimulator
+    from gossipy_amd.engine.schedule import (
+        NativeTokenizedAdapter,
+        TokenizedScheduler,
+    )
+
+    cfg, acc_name = list(_tok_cfgs())[cfg_i]
+    py_s = TokenizedScheduler(cfg, _account(acc_name))
+    nat_s = NativeTokenizedAdapter(cfg, _account(acc_name), utility=1)
+    for r in range(4):
+        a = py_s.next_round(r)
+        nat = nat_s.next_round_flat(r)
+        f = nat_s.last_flat
+        assert a.sent_messages == f["sent"], f"round {r} sent"
+        assert a.failed_messages == f["failed"], f"round {r} failed"
+        assert a.total_size == f["total_size"], f"round {r} size"
+        assert a.n_slots == f["n_slots"], f"round {r} slots"
+        flat_py = BatchedGossipSimulator._flatten_phases(a.ticks)
+        for key in (
+            "snap_nodes", "snap_slots", "recv_nodes", "recv_nptr",
+            "del_slots", "reply_slots", "del_pids", "pull_nodes",
+            "pull_slots", "rep_nodes", "rep_nptr", "rep_slots", "rep_pids",
+        ):
+            va = np.asarray(flat_py[key], dtype=np.int64)
+            vb = np.asarray(f[key], dtype=np.int64)
+            if key in ("del_pids", "rep_pids") and len(va) != len(vb):
+                # python flatten omits all -1 pid blocks; native always emits
+                if len(va) == 0:
+                    assert (vb == -1).all()
+                    continue
+            assert np.array_equal(va, vb), f"round {r} {key}: {va} vs {vb}"
+        # token balances must track exactly
+        assert [acct.n_tokens for acct in py_s.accounts] == list(
+            nat_s.token_balances()
+        ), f"round {r} balances"
