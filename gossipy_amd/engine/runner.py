@@ -31,7 +31,7 @@ import torch
 import torch.distributed as dist
 
 from .. import LOG
-from ..core import AntiEntropyProtocol
+from ..core import AntiEntropyProtocol, CreateModelMode
 from ..simul import SimulationEventSender
 from .arena import DataArena, NodeStateArena, SlotPool
 from .backend import make_backend
@@ -453,6 +453,7 @@ class BatchedGossipSimulator(SimulationEventSender):
             and self.spec.family in ("logreg", "pegasos", "adaline", "mlp", "mf")
             and (self.spec.family == "logreg" or getattr(self.spec, "n_parts", 0) == 0)
             and not getattr(self.spec, "pass_through", False)
+            and (self.spec.family != "mf" or self.spec.mode == CreateModelMode.MERGE_UPDATE)
         )
 
     @staticmethod

@@ -805,3 +805,57 @@ class TestFlattenedExecutor:
         b = self._cn(False)
         assert torch.allclose(a.local_params(), b.local_params(), atol=1e-5)
         assert torch.equal(a.state.ages, b.state.ages)
+
+
+class TestMoreRoundExecutors:
+    def test_sampled_fast_matches_tick(self):
+        X, y = make_synthetic_classification((640, 57, 2), seed=0, margin=2.0)
+        idx = np.random.default_rng(0).permutation(640)
+        shards = [(X[s], y[s]) for s in np.array_split(idx[:576], 64)]
+        data = DataArena.from_shards(
+            shards, CUDA, global_eval=(X[idx[576:]], y[idx[576:]])
+        )
+        cfg = EngineConfig(
+            n_nodes=64, delta=10, protocol=AntiEntropyProtocol.PUSH_PULL,
+            model_size=116, sampling_eval=0.0, seed=17, sampled=True,
+        )
+        spec = LogRegSpec(d_in=57, n_classes=2, lr=0.1, sample_size=0.3)
+        fast = BatchedGossipSimulator(cfg, spec, data, device=CUDA)
+        assert fast._fast_path_ok()
+        fast.init_nodes()
+        fast.start(n_rounds=4)
+        slow = BatchedGossipSimulator(cfg, spec, data, device=CUDA)
+        slow._fast_path_ok = lambda: False
+        slow._flat_schedulable = False
+        slow.init_nodes()
+        slow.start(n_rounds=4)
+        torch.cuda.synchronize()
+        assert torch.allclose(
+            fast.local_params(), slow.local_params(), atol=1e-5, rtol=1e-5
+        )
+        assert torch.equal(fast.state.ages, slow.state.ages)
+
+    def test_mf_fast_matches_tick(self):
+        from gossipy_amd.engine import MFSpec
+        from tests.test_engine import _mf_arena
+
+        data = _mf_arena(n_users=20, device=CUDA)
+        cfg = EngineConfig(
+            n_nodes=20, delta=10, protocol=AntiEntropyProtocol.PUSH,
+            model_size=300, sampling_eval=0.0, seed=23,
+        )
+        spec = MFSpec(k=5, n_items=50, reg=0.1, lr=0.01)
+        fast = BatchedGossipSimulator(cfg, spec, data, device=CUDA)
+        assert fast._fast_path_ok()
+        fast.init_nodes()
+        fast.start(n_rounds=4)
+        slow = BatchedGossipSimulator(cfg, spec, data, device=CUDA)
+        slow._fast_path_ok = lambda: False
+        slow._flat_schedulable = False
+        slow.init_nodes()
+        slow.start(n_rounds=4)
+        torch.cuda.synchronize()
+        assert torch.allclose(
+            fast.local_params(), slow.local_params(), atol=1e-4, rtol=1e-4
+        )
+        assert torch.equal(fast.state.ages, slow.state.ages)
