@@ -571,6 +571,68 @@ class BatchedGossipSimulator(SimulationEventSender):
         from .backend import _MODE_ID
 
         spec = self.spec
+        if (
+            spec.family == "logreg"
+            and getattr(spec, "n_parts", 0) == 0
+            and getattr(spec, "sample_size", 0) == 0
+            and not getattr(spec, "pass_through", False)
+            and getattr(self, "_coop_enabled", True)
+            and os.environ.get("GOSSIPY_NO_COOP") != "1"
+        ):
+            # single-launch cooperative round (plain logreg): grid sized to
+            # the largest per-tick batch; falls back to the stream executor
+            # if the cooperative launch is unavailable
+            per_tick = []
+            for name in ("snap_tptr", "recv_tptr", "pull_tptr", "rep_tptr"):
+                dif = np.diff(f[name])
+                if len(dif):
+                    per_tick.append(int(dif.max()))
+            max_batch = max(per_tick) if per_tick else 1
+            tpd = {
+                n: torch.from_numpy(
+                    np.ascontiguousarray(f[n], dtype=np.int32)
+                ).to(dev)
+                for n in ("snap_tptr", "recv_tptr", "pull_tptr", "rep_tptr")
+            }
+            try:
+                ext.run_round_coop_logreg(
+                    self.state.params,
+                    self.state.ages,
+                    self.pool.slots,
+                    self.pool.slot_ages,
+                    views["snap_nodes"],
+                    views["snap_slots"],
+                    tpd["snap_tptr"],
+                    views["recv_nodes"],
+                    views["recv_nptr"],
+                    tpd["recv_tptr"],
+                    views["del_slots"],
+                    views["reply_slots"],
+                    views["pull_nodes"],
+                    views["pull_slots"],
+                    tpd["pull_tptr"],
+                    views["rep_nodes"],
+                    views["rep_nptr"],
+                    tpd["rep_tptr"],
+                    views["rep_slots"],
+                    self.data.x,
+                    self.data.y,
+                    self.data.counts,
+                    spec.d_in,
+                    spec.n_classes,
+                    spec.lr,
+                    spec.weight_decay,
+                    max(1, spec.local_epochs),
+                    spec.batch_size,
+                    _MODE_ID[spec.mode],
+                    max_batch,
+                )
+                return
+            except RuntimeError as e:
+                LOG.warning("cooperative round launch unavailable (%s); "
+                            "falling back to the stream executor", e)
+                self._coop_enabled = False
+
         if getattr(spec, "sample_size", 0) > 0:
             c = list(common)
             c.insert(12, views["del_pids"])   # sample seeds ride the pid slot

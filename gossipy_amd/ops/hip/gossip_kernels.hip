@@ -31,6 +31,7 @@
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 #include <hip/hip_runtime.h>
+#include <hip/hip_cooperative_groups.h>
 
 #define DEV_INLINE __device__ __forceinline__
 
@@ -162,10 +163,8 @@ DEV_INLINE void logreg_update(const LogregArgs& a, int node, float* W,
     }
 }
 
-__global__ void __launch_bounds__(128)
-tick_logreg_kernel(LogregArgs a)
+DEV_INLINE void logreg_process_node(const LogregArgs& a, int i)
 {
-    int i = blockIdx.x;
     int node = a.nodes[i];
     int tid = threadIdx.x;
     extern __shared__ float sm[];
@@ -232,6 +231,12 @@ tick_logreg_kernel(LogregArgs a)
     for (int e = tid; e < a.D; e += blockDim.x)
         a.params[(long)node * a.D + e] = W[e];
     if (tid == 0) a.ages[node] = age;
+}
+
+__global__ void __launch_bounds__(128)
+tick_logreg_kernel(LogregArgs a)
+{
+    logreg_process_node(a, blockIdx.x);
 }
 
 // ---------------------------------------------------------------------------
@@ -2118,6 +2123,161 @@ void run_round_mf(
     }
 }
 
+// ---------------------------------------------------------------------------
+// cooperative whole-round kernel (logreg family): the entire round — every
+// tick's snapshot and delivery batch — runs inside ONE kernel launch, with
+// grid.sync() as the tick barrier. Grids here are tiny (max batch ≈ a few
+// dozen blocks on 256 CUs), so cooperative residency is guaranteed and the
+// launch is REJECTED (not deadlocked) if ever oversubscribed. Cuts the
+// ~200 per-round launch overheads of the stream executor to one.
+// ---------------------------------------------------------------------------
+
+struct CoopRoundArgs {
+    LogregArgs base;  // nodes/ptr/dslots/rslots are set per group inside
+    const int* snap_nodes; const int* snap_slots; const int* snap_tptr;
+    const int* recv_nodes; const int* recv_nptr; const int* recv_tptr;
+    const int* del_slots; const int* reply_slots;
+    const int* pull_nodes; const int* pull_slots; const int* pull_tptr;
+    const int* rep_nodes; const int* rep_nptr; const int* rep_tptr;
+    const int* rep_slots;
+    int delta;
+};
+
+__global__ void __launch_bounds__(128)
+coop_round_logreg_kernel(CoopRoundArgs c)
+{
+    namespace cg = cooperative_groups;
+    cg::grid_group grid = cg::this_grid();
+    LogregArgs a = c.base;
+    const int nb = gridDim.x;
+    const int bid = blockIdx.x;
+    const int tid = threadIdx.x;
+    bool dirty = false;  // writes since the last grid sync
+
+    for (int t = 0; t < c.delta; ++t) {
+        int s0 = c.snap_tptr[t], s1 = c.snap_tptr[t + 1];
+        int r0 = c.recv_tptr[t], r1 = c.recv_tptr[t + 1];
+        int p0 = c.pull_tptr[t], p1 = c.pull_tptr[t + 1];
+        int q0 = c.rep_tptr[t], q1 = c.rep_tptr[t + 1];
+
+        if (s1 > s0) {
+            if (dirty) { grid.sync(); dirty = false; }
+            // block-strided row copies (the snapshot sub-phase)
+            for (int i = s0 + bid; i < s1; i += nb) {
+                int node = c.snap_nodes[i];
+                int slot = c.snap_slots[i];
+                for (int e = tid; e < a.D; e += blockDim.x)
+                    a.slots[(long)slot * a.D + e] =
+                        a.params[(long)node * a.D + e];
+                if (tid == 0) a.slot_ages[slot] = a.ages[node];
+            }
+            dirty = true;
+        }
+        if (r1 > r0) {
+            if (dirty) { grid.sync(); dirty = false; }
+            a.nodes = c.recv_nodes;
+            a.ptr = c.recv_nptr;
+            a.dslots = c.del_slots;
+            a.rslots = c.reply_slots;
+            for (int i = r0 + bid; i < r1; i += nb) {
+                logreg_process_node(a, i);
+                __syncthreads();
+            }
+            dirty = true;
+        }
+        if (p1 > p0) {
+            if (dirty) { grid.sync(); dirty = false; }
+            for (int i = p0 + bid; i < p1; i += nb) {
+                int node = c.pull_nodes[i];
+                int slot = c.pull_slots[i];
+                for (int e = tid; e < a.D; e += blockDim.x)
+                    a.slots[(long)slot * a.D + e] =
+                        a.params[(long)node * a.D + e];
+                if (tid == 0) a.slot_ages[slot] = a.ages[node];
+            }
+            dirty = true;
+        }
+        if (q1 > q0) {
+            if (dirty) { grid.sync(); dirty = false; }
+            a.nodes = c.rep_nodes;
+            a.ptr = c.rep_nptr;
+            a.dslots = c.rep_slots;
+            a.rslots = nullptr;
+            for (int i = q0 + bid; i < q1; i += nb) {
+                logreg_process_node(a, i);
+                __syncthreads();
+            }
+            dirty = true;
+        }
+    }
+}
+
+void run_round_coop_logreg(
+    torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
+    torch::Tensor slot_ages,
+    torch::Tensor snap_nodes, torch::Tensor snap_slots, torch::Tensor snap_tptr,
+    torch::Tensor recv_nodes, torch::Tensor recv_nptr, torch::Tensor recv_tptr,
+    torch::Tensor del_slots, torch::Tensor reply_slots,
+    torch::Tensor pull_nodes, torch::Tensor pull_slots, torch::Tensor pull_tptr,
+    torch::Tensor rep_nodes, torch::Tensor rep_nptr, torch::Tensor rep_tptr,
+    torch::Tensor rep_slots,
+    torch::Tensor X, torch::Tensor Y, torch::Tensor counts,
+    int64_t d, int64_t k, double lr, double wd, int64_t epochs, int64_t bs,
+    int64_t mode, int64_t max_batch)
+{
+    // NOTE: unlike the stream executor, ALL arrays (tick ptrs included)
+    // must be device-resident here.
+    CHECK_DEV(params); CHECK_DEV(slots); CHECK_DEV(X);
+    CHECK_DEV(snap_tptr); CHECK_DEV(recv_tptr); CHECK_DEV(pull_tptr);
+    CHECK_DEV(rep_tptr);
+    CoopRoundArgs c;
+    LogregArgs& a = c.base;
+    a.params = params.data_ptr<float>(); a.ages = ages.data_ptr<int>();
+    a.slots = slots.data_ptr<float>(); a.slot_ages = slot_ages.data_ptr<int>();
+    a.X = X.data_ptr<float>(); a.Y = Y.data_ptr<float>();
+    a.counts = counts.data_ptr<int>();
+    a.dmodes = nullptr;
+    a.d = d; a.k = k; a.Smax = X.size(1); a.D = params.size(1);
+    a.lr = lr; a.wd = wd; a.epochs = epochs; a.bs = bs; a.mode = mode;
+    a.update_only = 0;
+    auto dp = [](torch::Tensor& t) {
+        return t.numel() ? t.data_ptr<int>() : nullptr;
+    };
+    c.snap_nodes = dp(snap_nodes); c.snap_slots = dp(snap_slots);
+    c.snap_tptr = snap_tptr.data_ptr<int>();
+    c.recv_nodes = dp(recv_nodes); c.recv_nptr = recv_nptr.data_ptr<int>();
+    c.recv_tptr = recv_tptr.data_ptr<int>();
+    c.del_slots = dp(del_slots); c.reply_slots = dp(reply_slots);
+    c.pull_nodes = dp(pull_nodes); c.pull_slots = dp(pull_slots);
+    c.pull_tptr = pull_tptr.data_ptr<int>();
+    c.rep_nodes = dp(rep_nodes); c.rep_nptr = rep_nptr.data_ptr<int>();
+    c.rep_tptr = rep_tptr.data_ptr<int>();
+    c.rep_slots = dp(rep_slots);
+    c.delta = (int)snap_tptr.numel() - 1;
+
+    int bsmax = (bs == 0) ? a.Smax : std::min<int>(bs, a.Smax);
+    size_t smem = sizeof(float) * (2 * a.D + (size_t)bsmax * a.d +
+                                   (size_t)bsmax * a.k);
+    TORCH_CHECK(smem <= 160 * 1024, "logreg LDS budget exceeded");
+    int grid = std::max<int>(1, (int)max_batch);
+    // stay within guaranteed cooperative residency
+    int max_blocks = 0;
+    hipError_t oerr = hipOccupancyMaxActiveBlocksPerMultiprocessor(
+        &max_blocks, (const void*)coop_round_logreg_kernel, 128, smem);
+    TORCH_CHECK(oerr == hipSuccess, "occupancy query failed");
+    hipDeviceProp_t prop;
+    (void)hipGetDeviceProperties(&prop, 0);
+    int limit = max_blocks * prop.multiProcessorCount;
+    TORCH_CHECK(limit > 0, "cooperative launch not supported");
+    if (grid > limit) grid = limit;
+    void* kargs[] = {&c};
+    hipError_t err = hipLaunchCooperativeKernel(
+        (const void*)coop_round_logreg_kernel, dim3(grid), dim3(128), kargs,
+        smem, current_stream());
+    TORCH_CHECK(err == hipSuccess, "cooperative launch failed: ",
+                hipGetErrorString(err));
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m)
 {
     m.def("snapshot", &snapshot, "batched model snapshot (arena row copy)");
@@ -2143,6 +2303,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m)
           "whole-round executor, sampled logreg family");
     m.def("run_round_mf", &run_round_mf,
           "whole-round executor, MF recommender family");
+    m.def("run_round_coop_logreg", &run_round_coop_logreg,
+          "single-launch cooperative whole-round executor (logreg)");
     m.def("tick_pens", &tick_pens,
           "PENS step-1 event: score candidates, merge top-m, count winners");
     m.def("eval_metrics", &eval_metrics,

@@ -859,3 +859,44 @@ class TestMoreRoundExecutors:
             fast.local_params(), slow.local_params(), atol=1e-4, rtol=1e-4
         )
         assert torch.equal(fast.state.ages, slow.state.ages)
+
+
+class TestCoopRound:
+    """Single-launch cooperative round executor vs the stream executor."""
+
+    @pytest.mark.parametrize("proto", [
+        AntiEntropyProtocol.PUSH, AntiEntropyProtocol.PUSH_PULL,
+    ])
+    def test_coop_matches_stream(self, proto):
+        import os
+
+        X, y = make_synthetic_classification((640, 57, 2), seed=0, margin=2.0)
+        idx = np.random.default_rng(0).permutation(640)
+        shards = [(X[s], y[s]) for s in np.array_split(idx[:576], 64)]
+        data = DataArena.from_shards(
+            shards, CUDA, global_eval=(X[idx[576:]], y[idx[576:]])
+        )
+        cfg = EngineConfig(
+            n_nodes=64, delta=10, protocol=proto,
+            model_size=116, sampling_eval=0.0, seed=3,
+        )
+        spec = LogRegSpec(d_in=57, n_classes=2, lr=0.1)
+
+        coop = BatchedGossipSimulator(cfg, spec, data, device=CUDA)
+        assert coop._fast_path_ok()
+        coop.init_nodes()
+        coop.start(n_rounds=4)
+        assert getattr(coop, "_coop_enabled", True), "coop launch must work"
+
+        os.environ["GOSSIPY_NO_COOP"] = "1"
+        try:
+            stream = BatchedGossipSimulator(cfg, spec, data, device=CUDA)
+            stream.init_nodes()
+            stream.start(n_rounds=4)
+        finally:
+            del os.environ["GOSSIPY_NO_COOP"]
+        torch.cuda.synchronize()
+        assert torch.allclose(
+            coop.local_params(), stream.local_params(), atol=1e-5, rtol=1e-5
+        )
+        assert torch.equal(coop.state.ages, stream.state.ages)
