@@ -577,11 +577,13 @@ class BatchedGossipSimulator(SimulationEventSender):
             and getattr(spec, "sample_size", 0) == 0
             and not getattr(spec, "pass_through", False)
             and getattr(self, "_coop_enabled", True)
-            and os.environ.get("GOSSIPY_NO_COOP") != "1"
+            and os.environ.get("GOSSIPY_COOP") == "1"
         ):
-            # single-launch cooperative round (plain logreg): grid sized to
-            # the largest per-tick batch; falls back to the stream executor
-            # if the cooperative launch is unavailable
+            # single-launch cooperative round (plain logreg) — measured
+            # SLOWER than the stream executor on MI355X (1.68 vs 1.41
+            # ms/round at the flagship config: ~200 grid.sync barriers cost
+            # more than ~200 stream launch gaps), so opt-in only
+            # (GOSSIPY_COOP=1); kept as the measured alternative
             per_tick = []
             for name in ("snap_tptr", "recv_tptr", "pull_tptr", "rep_tptr"):
                 dif = np.diff(f[name])

@@ -882,19 +882,19 @@ class TestCoopRound:
         )
         spec = LogRegSpec(d_in=57, n_classes=2, lr=0.1)
 
-        coop = BatchedGossipSimulator(cfg, spec, data, device=CUDA)
-        assert coop._fast_path_ok()
-        coop.init_nodes()
-        coop.start(n_rounds=4)
-        assert getattr(coop, "_coop_enabled", True), "coop launch must work"
-
-        os.environ["GOSSIPY_NO_COOP"] = "1"
+        os.environ["GOSSIPY_COOP"] = "1"
         try:
-            stream = BatchedGossipSimulator(cfg, spec, data, device=CUDA)
-            stream.init_nodes()
-            stream.start(n_rounds=4)
+            coop = BatchedGossipSimulator(cfg, spec, data, device=CUDA)
+            assert coop._fast_path_ok()
+            coop.init_nodes()
+            coop.start(n_rounds=4)
+            assert getattr(coop, "_coop_enabled", True), "coop launch must work"
         finally:
-            del os.environ["GOSSIPY_NO_COOP"]
+            del os.environ["GOSSIPY_COOP"]
+
+        stream = BatchedGossipSimulator(cfg, spec, data, device=CUDA)
+        stream.init_nodes()
+        stream.start(n_rounds=4)
         torch.cuda.synchronize()
         assert torch.allclose(
             coop.local_params(), stream.local_params(), atol=1e-5, rtol=1e-5
