@@ -55,6 +55,18 @@ __all__ = [
 ]
 
 
+def _csr_gather(ptr: np.ndarray, rows: np.ndarray):
+    """Select CSR rows: returns (flat element indices, new local ptr)."""
+    ptr = np.asarray(ptr)
+    counts = np.diff(ptr)[rows]
+    new_ptr = np.zeros(len(counts) + 1, dtype=np.int64)
+    np.cumsum(counts, out=new_ptr[1:])
+    sel = np.repeat(ptr[rows] - new_ptr[:-1], counts) + np.arange(
+        int(new_ptr[-1])
+    )
+    return sel, new_ptr
+
+
 class BatchedGossipSimulator(SimulationEventSender):
     """Node-batched gossip simulator for one or many GPUs (or CPU).
 
@@ -210,14 +222,24 @@ class BatchedGossipSimulator(SimulationEventSender):
             return []
         src = self._rank_of(owners)
         dst = self._rank_of(np.repeat(recv_nodes, np.diff(recv_ptr)))
-        plan: Dict[Tuple[int, int], List[int]] = {}
         cross = src != dst
-        for s, d, slot in zip(src[cross], dst[cross], slots[cross]):
-            plan.setdefault((int(s), int(d)), []).append(int(slot))
-        return [
-            (s, d, np.unique(np.asarray(v, dtype=np.int64)))
-            for (s, d), v in sorted(plan.items())
-        ]
+        if not cross.any():
+            return []
+        cs, cd, csl = src[cross], dst[cross], slots[cross]
+        # group by (src, dst) and dedupe slots, fully vectorized
+        order = np.lexsort((csl, cd, cs))
+        cs, cd, csl = cs[order], cd[order], csl[order]
+        new_grp = np.empty(len(cs), dtype=bool)
+        new_grp[0] = True
+        new_grp[1:] = (cs[1:] != cs[:-1]) | (cd[1:] != cd[:-1])
+        starts = np.flatnonzero(new_grp)
+        ends = np.append(starts[1:], len(cs))
+        out = []
+        for a, b in zip(starts, ends):
+            sl = csl[a:b]
+            uniq = np.unique(sl)
+            out.append((int(cs[a]), int(cd[a]), uniq.astype(np.int64)))
+        return out
 
     # -- round execution -----------------------------------------------------
 
@@ -227,14 +249,17 @@ class BatchedGossipSimulator(SimulationEventSender):
         if len(recv_idx) == 0:
             return
         counts = np.diff(phase.recv_ptr)[recv_idx]
-        sel = np.concatenate(
-            [
-                np.arange(phase.recv_ptr[i], phase.recv_ptr[i + 1])
-                for i in recv_idx
-            ]
-        )
+        # vectorized gather of each kept receiver's delivery range
         new_ptr = np.zeros(len(counts) + 1, dtype=np.int64)
         np.cumsum(counts, out=new_ptr[1:])
+        sel = (
+            np.repeat(
+                np.asarray(phase.recv_ptr)[recv_idx]
+                - new_ptr[:-1],
+                counts,
+            )
+            + np.arange(int(new_ptr[-1]))
+        )
         pids = None
         if phase.del_pids is not None and len(phase.del_pids):
             pids = torch.from_numpy(phase.del_pids[sel].astype(np.int64))
@@ -308,15 +333,9 @@ class BatchedGossipSimulator(SimulationEventSender):
             )
             cmine = self._is_mine(phase.rep_recv_nodes)
             if cmine.any():
-                counts = np.diff(phase.rep_recv_ptr)[cmine]
-                sel = np.concatenate(
-                    [
-                        np.arange(phase.rep_recv_ptr[i], phase.rep_recv_ptr[i + 1])
-                        for i in np.where(cmine)[0]
-                    ]
+                sel, new_ptr = _csr_gather(
+                    phase.rep_recv_ptr, np.where(cmine)[0]
                 )
-                new_ptr = np.zeros(len(counts) + 1, dtype=np.int64)
-                np.cumsum(counts, out=new_ptr[1:])
                 no_reply = torch.full((len(sel),), -1, dtype=torch.int64)
                 rpids = None
                 if phase.rep_pids is not None and len(phase.rep_pids):
@@ -980,15 +999,7 @@ class BatchedAll2AllGossipSimulator(BatchedGossipSimulator):
             )
             mine = self._is_mine(phase.wm_nodes)
             if mine.any():
-                counts = np.diff(phase.wm_ptr)[mine]
-                sel = np.concatenate(
-                    [
-                        np.arange(phase.wm_ptr[i], phase.wm_ptr[i + 1])
-                        for i in np.where(mine)[0]
-                    ]
-                )
-                new_ptr = np.zeros(len(counts) + 1, dtype=np.int64)
-                np.cumsum(counts, out=new_ptr[1:])
+                sel, new_ptr = _csr_gather(phase.wm_ptr, np.where(mine)[0])
                 self.backend.deliver_weighted(
                     self.state,
                     self.pool,
@@ -1083,15 +1094,7 @@ class BatchedPENSGossipSimulator(BatchedGossipSimulator):
             )
             mine = self._is_mine(phase.pens_nodes)
             if mine.any():
-                cnt = np.diff(phase.pens_ptr)[mine]
-                sel = np.concatenate(
-                    [
-                        np.arange(phase.pens_ptr[i], phase.pens_ptr[i + 1])
-                        for i in np.where(mine)[0]
-                    ]
-                )
-                new_ptr = np.zeros(len(cnt) + 1, dtype=np.int64)
-                np.cumsum(cnt, out=new_ptr[1:])
+                sel, new_ptr = _csr_gather(phase.pens_ptr, np.where(mine)[0])
                 self.backend.deliver_pens(
                     self.state,
                     self.pool,
