@@ -275,6 +275,171 @@ class BatchedGossipSimulator(SimulationEventSender):
             del_pids=pids,
         )
 
+    def _run_round_multi(self, f: dict) -> None:
+        """Multi-rank whole-round execution from the flat schedule arrays:
+        all per-rank filtering and index math happens ONCE per round
+        (vectorized numpy + one device upload); the per-tick loop issues
+        only kernel launches and the RCCL slot exchanges. Local-sourced
+        deliveries launch while the exchange for remote-sourced ones is in
+        flight."""
+        dev = self.device
+        delta = len(f["snap_tptr"]) - 1
+        lo, n_local = self.node_lo, self.n_local
+
+        def _tickify(nodes, tptr):
+            """(mask of mine, tick id per kept entry, per-tick ptr)."""
+            nodes = np.asarray(nodes)
+            mask = (nodes >= lo) & (nodes < lo + n_local)
+            tick_of = np.repeat(np.arange(delta), np.diff(tptr))[mask]
+            ptr = np.zeros(delta + 1, dtype=np.int64)
+            np.cumsum(np.bincount(tick_of, minlength=delta), out=ptr[1:])
+            return mask, ptr
+
+        def _dev32(a):
+            return torch.from_numpy(np.ascontiguousarray(a, dtype=np.int32)).to(
+                dev, non_blocking=True
+            )
+
+        # --- snapshots / pull snapshots
+        s_mask, s_ptr = _tickify(f["snap_nodes"], f["snap_tptr"])
+        snap_nodes_d = _dev32(np.asarray(f["snap_nodes"])[s_mask] - lo)
+        snap_slots_d = _dev32(np.asarray(f["snap_slots"])[s_mask])
+        p_mask, p_ptr = _tickify(f["pull_nodes"], f["pull_tptr"])
+        pull_nodes_d = _dev32(np.asarray(f["pull_nodes"])[p_mask] - lo)
+        pull_slots_d = _dev32(np.asarray(f["pull_slots"])[p_mask])
+
+        def _deliver_prep(nodes_key, nptr_key, tptr_key, slots_key, reply_key,
+                          pids_key, owners_key):
+            """Per-round prep of one delivery CSR family. Returns per-tick
+            launch slices for local-sourced and remote-sourced receiver
+            groups plus the per-tick exchange plans."""
+            recv_nodes = np.asarray(f[nodes_key])
+            nptr = np.asarray(f[nptr_key])
+            tptr = np.asarray(f[tptr_key])
+            dslots = np.asarray(f[slots_key])
+            owners = np.asarray(f[owners_key])
+            n_rows = len(recv_nodes)
+            if n_rows == 0:
+                return None
+            row_tick = np.repeat(np.arange(delta), np.diff(tptr))
+            row_mine = (recv_nodes >= lo) & (recv_nodes < lo + n_local)
+            src_rank = owners // n_local
+            has_remote = (
+                np.maximum.reduceat(src_rank != self.rank, nptr[:-1])
+                if len(dslots)
+                else np.zeros(n_rows, dtype=bool)
+            ).astype(bool)
+
+            # exchange plans, grouped per (tick, src, dst) in one lexsort
+            d_tick = np.repeat(row_tick, np.diff(nptr))
+            d_dst = np.repeat(recv_nodes // n_local, np.diff(nptr))
+            cross = src_rank != d_dst
+            plans = [[] for _ in range(delta)]
+            if cross.any():
+                ct, cs, cd2, csl = (
+                    d_tick[cross], src_rank[cross], d_dst[cross], dslots[cross]
+                )
+                order = np.lexsort((csl, cd2, cs, ct))
+                ct, cs, cd2, csl = ct[order], cs[order], cd2[order], csl[order]
+                new_g = np.empty(len(ct), dtype=bool)
+                new_g[0] = True
+                new_g[1:] = (
+                    (ct[1:] != ct[:-1]) | (cs[1:] != cs[:-1]) | (cd2[1:] != cd2[:-1])
+                )
+                starts = np.flatnonzero(new_g)
+                ends = np.append(starts[1:], len(ct))
+                for a, b in zip(starts, ends):
+                    plans[int(ct[a])].append(
+                        (int(cs[a]), int(cd2[a]), np.unique(csl[a:b]).astype(np.int64))
+                    )
+
+            groups = {}
+            for name, rows_mask in (
+                ("local", row_mine & ~has_remote),
+                ("remote", row_mine & has_remote),
+            ):
+                rows = np.flatnonzero(rows_mask)
+                sel, _ = _csr_gather(nptr, rows)
+                # absolute ptr into the gathered arrays
+                lens = np.diff(nptr)[rows]
+                abs_ptr = np.zeros(len(rows) + 1, dtype=np.int64)
+                np.cumsum(lens, out=abs_ptr[1:])
+                rt = row_tick[rows]
+                row_tptr = np.zeros(delta + 1, dtype=np.int64)
+                np.cumsum(np.bincount(rt, minlength=delta), out=row_tptr[1:])
+                pids = np.asarray(f[pids_key]) if pids_key else None
+                groups[name] = {
+                    "nodes": _dev32(recv_nodes[rows] - lo),
+                    "ptr": _dev32(abs_ptr),
+                    "slots": _dev32(dslots[sel]) if len(sel) else _dev32([]),
+                    "reply": _dev32(np.asarray(f[reply_key])[sel])
+                    if reply_key
+                    else None,
+                    "pids": _dev32(pids[sel])
+                    if pids is not None and len(pids)
+                    else None,
+                    "row_tptr": row_tptr,
+                }
+            return {"groups": groups, "plans": plans}
+
+        dmain = _deliver_prep(
+            "recv_nodes", "recv_nptr", "recv_tptr", "del_slots",
+            "reply_slots", "del_pids", "del_owners",
+        )
+        drep = _deliver_prep(
+            "rep_nodes", "rep_nptr", "rep_tptr", "rep_slots",
+            None, "rep_pids", "rep_owners",
+        )
+
+        def _launch(g, t, reply_default=-1):
+            a, b = int(g["row_tptr"][t]), int(g["row_tptr"][t + 1])
+            if a == b:
+                return
+            lo_d = int(g["ptr"][a])
+            hi_d = int(g["ptr"][b])
+            reply = (
+                g["reply"][lo_d:hi_d]
+                if g["reply"] is not None
+                else torch.full(
+                    (hi_d - lo_d,), reply_default, dtype=torch.int32, device=dev
+                )
+            )
+            self.backend.deliver(
+                self.state,
+                self.pool,
+                self.data,
+                self.spec,
+                g["nodes"][a:b],
+                g["ptr"][a : b + 1] - lo_d,
+                g["slots"][:hi_d][lo_d:] if lo_d else g["slots"][:hi_d],
+                reply,
+                del_pids=g["pids"][lo_d:hi_d] if g["pids"] is not None else None,
+            )
+
+        for t in range(delta):
+            a, b = int(s_ptr[t]), int(s_ptr[t + 1])
+            if b > a:
+                self.backend.snapshot(
+                    self.state, self.pool, snap_nodes_d[a:b], snap_slots_d[a:b],
+                    src_off=self._snap_off,
+                )
+            if dmain is not None:
+                pending = self._exchange_start(dmain["plans"][t])
+                _launch(dmain["groups"]["local"], t)
+                self._exchange_finish(pending)
+                _launch(dmain["groups"]["remote"], t)
+            a, b = int(p_ptr[t]), int(p_ptr[t + 1])
+            if b > a:
+                self.backend.snapshot(
+                    self.state, self.pool, pull_nodes_d[a:b], pull_slots_d[a:b],
+                    src_off=self._snap_off,
+                )
+            if drep is not None:
+                pending = self._exchange_start(drep["plans"][t])
+                _launch(drep["groups"]["local"], t)
+                self._exchange_finish(pending)
+                _launch(drep["groups"]["remote"], t)
+
     def _run_tick(self, phase: TickPhase) -> None:
         # A: snapshots of firing nodes
         mine = self._is_mine(phase.snap_nodes)
@@ -487,6 +652,7 @@ class BatchedGossipSimulator(SimulationEventSender):
             "pull_nodes": [], "pull_slots": [], "pull_tptr": [0],
             "rep_nodes": [], "rep_nptr": [0], "rep_tptr": [0],
             "rep_slots": [], "rep_pids": [],
+            "del_owners": [], "rep_owners": [],
         }
         d_off = 0
         e_off = 0
@@ -502,6 +668,7 @@ class BatchedGossipSimulator(SimulationEventSender):
             d_off += len(ph.del_slots)
             f["recv_tptr"].append(f["recv_tptr"][-1] + len(ph.recv_nodes))
             f["del_slots"].append(ph.del_slots)
+            f["del_owners"].append(ph.del_owners)
             f["reply_slots"].append(ph.reply_slots)
             if ph.del_pids is not None and len(ph.del_pids):
                 f["del_pids"].append(ph.del_pids)
@@ -516,6 +683,7 @@ class BatchedGossipSimulator(SimulationEventSender):
                 )
                 e_off += len(ph.rep_del_slots)
                 f["rep_slots"].append(ph.rep_del_slots)
+                f["rep_owners"].append(ph.rep_del_owners)
                 if ph.rep_pids is not None and len(ph.rep_pids):
                     f["rep_pids"].append(ph.rep_pids)
             f["rep_tptr"].append(f["rep_tptr"][-1] + len(rep_n))
@@ -761,6 +929,17 @@ class BatchedGossipSimulator(SimulationEventSender):
                 sched = self.scheduler.next_round(r)
                 self.pool.ensure(sched.n_slots)
                 if (
+                    self.world > 1
+                    and self._flat_schedulable
+                    and os.environ.get("GOSSIPY_NO_MULTIFAST") != "1"
+                ):
+                    # multi-rank: one vectorized per-round prep + upload,
+                    # per-tick loop of launches and RCCL exchanges only
+                    flat = getattr(self.scheduler, "last_flat", None)
+                    if flat is None:
+                        flat = self._flatten_phases(sched.ticks)
+                    self._run_round_multi(flat)
+                elif (
                     self._flat_schedulable
                     and self._flat_exec_ok()
                     and sched.ticks

@@ -1262,3 +1262,123 @@ class TestPENSEngine:
             for ph in rs.ticks:
                 for nd, sl in zip(ph.snap_nodes, ph.snap_slots):
                     pass  # schedule generation alone must not crash
+
+
+def _tok_worker(rank, world, port, q):
+    import torch.distributed as dist
+
+    from gossipy_amd.engine import BatchedTokenizedGossipSimulator
+    from gossipy_amd.flow_control import RandomizedTokenAccount
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        shards, geval = _make_data(40, seed=1)
+        data = _arena_for_rank(shards, geval, rank, world)
+        cfg = EngineConfig(
+            n_nodes=40, delta=10, protocol=AntiEntropyProtocol.PUSH,
+            model_size=116, sampling_eval=0.0, seed=11, n_parts=4,
+        )
+        spec = LogRegSpec(d_in=57, n_classes=2, lr=0.1, n_parts=4)
+        sim = BatchedTokenizedGossipSimulator(
+            cfg, spec, data, token_account=RandomizedTokenAccount(C=20, A=10),
+            device=torch.device("cpu"),
+        )
+        sim.init_nodes()
+        sim.start(n_rounds=4)
+        full = sim.gather_params()
+        if rank == 0:
+            q.put(full.numpy())
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_tokenized_two_rank_matches_single():
+    from gossipy_amd.engine import BatchedTokenizedGossipSimulator
+    from gossipy_amd.flow_control import RandomizedTokenAccount
+
+    shards, geval = _make_data(40, seed=1)
+    data = DataArena.from_shards(shards, torch.device("cpu"), global_eval=geval)
+    cfg = EngineConfig(
+        n_nodes=40, delta=10, protocol=AntiEntropyProtocol.PUSH,
+        model_size=116, sampling_eval=0.0, seed=11, n_parts=4,
+    )
+    spec = LogRegSpec(d_in=57, n_classes=2, lr=0.1, n_parts=4)
+    ref = BatchedTokenizedGossipSimulator(
+        cfg, spec, data, token_account=RandomizedTokenAccount(C=20, A=10),
+        device=torch.device("cpu"),
+    )
+    ref.init_nodes()
+    ref.start(n_rounds=4)
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [
+        ctx.Process(target=_tok_worker, args=(r, 2, 29537, q)) for r in range(2)
+    ]
+    for p in procs:
+        p.start()
+    got = q.get(timeout=240)
+    for p in procs:
+        p.join(timeout=60)
+    assert np.allclose(ref.local_params().numpy(), got, atol=1e-6)
+
+
+def _cn_worker(rank, world, port, q):
+    import torch.distributed as dist
+
+    from gossipy_amd.engine import BatchedCacheNeighGossipSimulator
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        shards, geval = _make_data(40, seed=1)
+        data = _arena_for_rank(shards, geval, rank, world)
+        cfg = EngineConfig(
+            n_nodes=40, delta=10, protocol=AntiEntropyProtocol.PUSH_PULL,
+            model_size=116, sampling_eval=0.0, seed=33,
+        )
+        sim = BatchedCacheNeighGossipSimulator(
+            cfg, LogRegSpec(d_in=57, n_classes=2, lr=0.1), data,
+            device=torch.device("cpu"),
+        )
+        sim.init_nodes()
+        sim.start(n_rounds=4)
+        full = sim.gather_params()
+        if rank == 0:
+            q.put(full.numpy())
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_cacheneigh_two_rank_matches_single():
+    from gossipy_amd.engine import BatchedCacheNeighGossipSimulator
+
+    shards, geval = _make_data(40, seed=1)
+    data = DataArena.from_shards(shards, torch.device("cpu"), global_eval=geval)
+    cfg = EngineConfig(
+        n_nodes=40, delta=10, protocol=AntiEntropyProtocol.PUSH_PULL,
+        model_size=116, sampling_eval=0.0, seed=33,
+    )
+    ref = BatchedCacheNeighGossipSimulator(
+        cfg, LogRegSpec(d_in=57, n_classes=2, lr=0.1), data,
+        device=torch.device("cpu"),
+    )
+    ref.init_nodes()
+    ref.start(n_rounds=4)
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [
+        ctx.Process(target=_cn_worker, args=(r, 2, 29539, q)) for r in range(2)
+    ]
+    for p in procs:
+        p.start()
+    got = q.get(timeout=240)
+    for p in procs:
+        p.join(timeout=60)
+    assert np.allclose(ref.local_params().numpy(), got, atol=1e-6)
