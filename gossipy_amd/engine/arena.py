@@ -24,7 +24,6 @@ from __future__ import annotations
 
 from typing import Optional, Tuple
 
-import numpy as np
 import torch
 
 __all__ = ["NodeStateArena", "SlotPool", "DataArena"]

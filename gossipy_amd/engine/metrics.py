@@ -11,7 +11,7 @@ resident sample and rank 0 gathers the dicts.
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 import torch
 

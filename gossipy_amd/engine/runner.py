@@ -24,19 +24,19 @@ works unchanged.
 from __future__ import annotations
 
 import os
-from typing import Dict, List, Optional, Tuple
+from typing import List, Optional, Tuple
 
 import numpy as np
 import torch
 import torch.distributed as dist
 
 from .. import LOG
-from ..core import AntiEntropyProtocol, CreateModelMode
+from ..core import CreateModelMode
 from ..simul import SimulationEventSender
 from .arena import DataArena, NodeStateArena, SlotPool
 from .backend import make_backend
 from .metrics import binary_margin_metrics, classification_metrics_shared
-from .rng import Purpose, RandomTape
+from .rng import RandomTape
 from .schedule import (
     EngineConfig,
     RoundSchedule,
