@@ -28,6 +28,7 @@ from gossipy_amd.engine import (
     DataArena,
     EngineConfig,
     LogRegSpec,
+    MLPSpec,
     PegasosSpec,
 )
 from gossipy_amd.flow_control import RandomizedTokenAccount
@@ -107,5 +108,24 @@ def main():
     }))
 
 
+def mlp_bench(steps=20, warmup=3):
+    """Giaretta-shaped MLP gossip at 1000 nodes (BASELINE.json config 3
+    shape, 1 GPU): 57 -> 100 -> 2 MLP, every tick's GEMMs on MFMA."""
+    device = torch.device("cuda:0" if torch.cuda.is_available() else "cpu")
+    spec = MLPSpec(d_in=57, n_classes=2, hidden=(100,), lr=0.05, batch_size=32)
+    cfg = EngineConfig(
+        n_nodes=1000, delta=100, protocol=AntiEntropyProtocol.PUSH,
+        model_size=spec.D, sampling_eval=0.01, seed=42,
+    )
+    sim = BatchedGossipSimulator(cfg, spec, _data(1000, device), device=device)
+    r = timed(sim, steps, warmup, device)
+    print(json.dumps({
+        "config": "giaretta-shaped-mlp-57-100-2-1000n",
+        "rounds_per_sec": round(r, 2),
+        "node_rounds_per_sec": round(r * 1000, 0),
+    }))
+
+
 if __name__ == "__main__":
     main()
+    mlp_bench()

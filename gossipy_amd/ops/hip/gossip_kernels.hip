@@ -918,6 +918,17 @@ struct MlpArgs {
 
 // Fused fwd+bwd+SGD over the node's shard. LDS holds the model row plus
 // per-batch activations of every layer (act) and their gradients (grad).
+//
+// All three GEMMs (forward X·Wᵀ, input-grad G·W, weight-grad Gᵀ·X) run on
+// the matrix cores via v_mfma_f32_16x16x4_f32 (exact f32 — bitwise an fmaf
+// chain, so the torch oracle tolerance is unchanged). Fragment maps per
+// the CDNA4 ISA: lane l supplies A[l&15][l>>4] and B[l>>4][l&15]; the
+// accumulator lane l, register r holds D[(l>>4)*4 + r][l&15]. Ragged tile
+// edges are zero-padded in registers (exact for f32). One 16x16 output
+// tile per wave per iteration; the block's 4 waves stride the tile grid.
+
+typedef float f32x4_t __attribute__((ext_vector_type(4)));
+
 DEV_INLINE void mlp_update(const MlpArgs& a, int node, float* W,
                            float* act, float* grad, int& age)
 {
@@ -927,6 +938,11 @@ DEV_INLINE void mlp_update(const MlpArgs& a, int node, float* W,
     int bsz = (a.bs == 0) ? c : min(a.bs, c);
     const float* Xn = a.X + (long)node * a.Smax * a.d_in;
     const float* Yn = a.Y + (long)node * a.Smax;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int fr = lane & 15;       // A-row / B-col fragment index
+    const int fk = lane >> 4;       // K fragment index (0..3)
+    const int nwaves = blockDim.x >> 6;
     for (int ep = 0; ep < a.epochs; ++ep) {
         for (int s0 = 0; s0 < c; s0 += bsz) {
             int m = min(bsz, c - s0);
@@ -936,20 +952,33 @@ DEV_INLINE void mlp_update(const MlpArgs& a, int node, float* W,
             for (int e = tid; e < m * a.d_in; e += blockDim.x)
                 in[e] = Xn[(long)s0 * a.d_in + e];
             __syncthreads();
-            // ---- forward
+            // ---- forward: H = X · Wᵀ + b on MFMA
             float* cur = in;
             float* nxt = act + m * a.d_in;
             for (int l = 0; l < a.n_layers; ++l) {
                 int w_off = a.layout[4 * l], b_off = a.layout[4 * l + 1];
                 int fin = a.layout[4 * l + 2], fout = a.layout[4 * l + 3];
-                for (int e = tid; e < m * fout; e += blockDim.x) {
-                    int s = e / fout, o = e - s * fout;
-                    const float* wrow = W + w_off + o * fin;
-                    const float* xrow = cur + s * fin;
-                    float acc = W[b_off + o];
-                    for (int q = 0; q < fin; ++q) acc += wrow[q] * xrow[q];
-                    if (l < a.n_layers - 1) acc = fmaxf(acc, 0.f);  // ReLU
-                    nxt[e] = acc;
+                int m16 = (m + 15) >> 4, o16 = (fout + 15) >> 4;
+                for (int tile = wave; tile < m16 * o16; tile += nwaves) {
+                    int tr = (tile / o16) << 4, tc = (tile % o16) << 4;
+                    f32x4_t acc = {0.f, 0.f, 0.f, 0.f};
+                    for (int k0 = 0; k0 < fin; k0 += 4) {
+                        int q = k0 + fk;
+                        float av = (tr + fr < m && q < fin)
+                                       ? cur[(tr + fr) * fin + q] : 0.f;
+                        float bv = (tc + fr < fout && q < fin)
+                                       ? W[w_off + (tc + fr) * fin + q] : 0.f;
+                        acc = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                            av, bv, acc, 0, 0, 0);
+                    }
+                    for (int r = 0; r < 4; ++r) {
+                        int row = tr + (fk << 2) + r, col = tc + fr;
+                        if (row < m && col < fout) {
+                            float v = acc[r] + W[b_off + col];
+                            if (l < a.n_layers - 1) v = fmaxf(v, 0.f);  // ReLU
+                            nxt[row * fout + col] = v;
+                        }
+                    }
                 }
                 __syncthreads();
                 cur = nxt;
@@ -972,7 +1001,8 @@ DEV_INLINE void mlp_update(const MlpArgs& a, int node, float* W,
                 }
             }
             __syncthreads();
-            // ---- backward + SGD, layer by layer (grad buffers ping-pong)
+            // ---- backward + SGD, layer by layer (grad buffers ping-pong);
+            //      both GEMMs (dX = G·W, dW = Gᵀ·X) on MFMA
             float* gnext = grad + m * a.act_max;
             for (int l = a.n_layers - 1; l >= 0; --l) {
                 int w_off = a.layout[4 * l], b_off = a.layout[4 * l + 1];
@@ -982,27 +1012,59 @@ DEV_INLINE void mlp_update(const MlpArgs& a, int node, float* W,
                 for (int q = 0; q < l; ++q) ain += m * a.layout[4 * q + 2];
                 // (ain now points at act of layer l's input: layers are
                 //  packed input-first, so offset = m*(d_in + hidden_0 + ...))
-                // grad wrt input (needed before W is updated)
+                // grad wrt input (needed before W is updated):
+                // dX[m, fin] = G[m, fout] · W[fout, fin], ReLU-masked
                 if (l > 0) {
-                    for (int e = tid; e < m * fin; e += blockDim.x) {
-                        int s = e / fin, q = e - s * fin;
-                        float acc = 0.f;
-                        for (int o = 0; o < fout; ++o)
-                            acc += gcur[s * fout + o] * W[w_off + o * fin + q];
-                        // ReLU mask of the layer-(l-1) activation
-                        acc *= (ain[s * fin + q] > 0.f) ? 1.0f : 0.0f;
-                        gnext[e] = acc;
+                    int m16 = (m + 15) >> 4, q16 = (fin + 15) >> 4;
+                    for (int tile = wave; tile < m16 * q16; tile += nwaves) {
+                        int tr = (tile / q16) << 4, tc = (tile % q16) << 4;
+                        f32x4_t acc = {0.f, 0.f, 0.f, 0.f};
+                        for (int k0 = 0; k0 < fout; k0 += 4) {
+                            int o = k0 + fk;
+                            float av = (tr + fr < m && o < fout)
+                                           ? gcur[(tr + fr) * fout + o] : 0.f;
+                            float bv = (o < fout && tc + fr < fin)
+                                           ? W[w_off + o * fin + tc + fr] : 0.f;
+                            acc = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                                av, bv, acc, 0, 0, 0);
+                        }
+                        for (int r = 0; r < 4; ++r) {
+                            int row = tr + (fk << 2) + r, col = tc + fr;
+                            if (row < m && col < fin) {
+                                float v = acc[r];
+                                // ReLU mask of the layer-(l-1) activation
+                                v *= (ain[row * fin + col] > 0.f) ? 1.0f : 0.0f;
+                                gnext[row * fin + col] = v;
+                            }
+                        }
                     }
                     __syncthreads();
                 }
-                // weight/bias SGD
-                for (int e = tid; e < fout * fin; e += blockDim.x) {
-                    int o = e / fin, q = e - o * fin;
-                    float g = 0.f;
-                    for (int s = 0; s < m; ++s)
-                        g += gcur[s * fout + o] * ain[s * fin + q];
-                    if (a.wd != 0.f) g += a.wd * W[w_off + e];
-                    W[w_off + e] -= a.lr * g;
+                // weight SGD: dW[fout, fin] = Gᵀ[fout, m] · X[m, fin]
+                {
+                    int o16 = (fout + 15) >> 4, q16 = (fin + 15) >> 4;
+                    for (int tile = wave; tile < o16 * q16; tile += nwaves) {
+                        int tr = (tile / q16) << 4, tc = (tile % q16) << 4;
+                        f32x4_t acc = {0.f, 0.f, 0.f, 0.f};
+                        for (int k0 = 0; k0 < m; k0 += 4) {
+                            int sidx = k0 + fk;
+                            float av = (tr + fr < fout && sidx < m)
+                                           ? gcur[sidx * fout + tr + fr] : 0.f;
+                            float bv = (sidx < m && tc + fr < fin)
+                                           ? ain[sidx * fin + tc + fr] : 0.f;
+                            acc = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                                av, bv, acc, 0, 0, 0);
+                        }
+                        for (int r = 0; r < 4; ++r) {
+                            int row = tr + (fk << 2) + r, col = tc + fr;
+                            if (row < fout && col < fin) {
+                                float g = acc[r];
+                                int e = row * fin + col;
+                                if (a.wd != 0.f) g += a.wd * W[w_off + e];
+                                W[w_off + e] -= a.lr * g;
+                            }
+                        }
+                    }
                 }
                 for (int e = tid; e < fout; e += blockDim.x) {
                     float g = 0.f;
