@@ -11,7 +11,15 @@ the reference.
 
 from .arena import DataArena, NodeStateArena, SlotPool
 from .backend import HIPBackend, TorchBackend, make_backend
-from .models import AdaLineSpec, KMeansSpec, LogRegSpec, MFSpec, MLPSpec, PegasosSpec
+from .models import (
+    AdaLineSpec,
+    KMeansSpec,
+    LogRegSpec,
+    MFSpec,
+    MLPSpec,
+    PegasosSpec,
+    TorchModuleSpec,
+)
 from .rng import Purpose, RandomTape
 from .runner import (
     BatchedAll2AllGossipSimulator,
@@ -54,4 +62,5 @@ __all__ = [
     "MLPSpec",
     "MFSpec",
     "KMeansSpec",
+    "TorchModuleSpec",
 ]

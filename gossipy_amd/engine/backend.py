@@ -92,6 +92,20 @@ class TorchBackend:
         if spec.family in ("pegasos", "adaline"):
             state.params.zero_()
             return
+        if spec.family == "torchmod":
+            # ONE xavier-uniform init shared by every node: the reference's
+            # CIFAR10Net.init_weights is a no-op (main_onoszko_2021.py:
+            # 38-44), so all nodes start from the same deep-copied
+            # prototype — which is what makes merging independently trained
+            # CNNs meaningful (feature alignment)
+            row = spec.init_row(tape.stream(Purpose.INIT))
+            state.params.copy_(
+                torch.from_numpy(row).to(state.params.device).expand_as(
+                    state.params
+                )
+            )
+            state.ages.zero_()
+            return
         if spec.family == "kmeans":
             # KMeansHandler.init = torch.rand(k, dim)
             # (gossipy/model/handler.py:594-595), tape-driven per node
@@ -186,6 +200,10 @@ class TorchBackend:
             self._update_mf(state.params, state.ages, data, spec, nodes.long())
         elif spec.family == "kmeans":
             self._update_kmeans(state.params, state.ages, data, spec, nodes.long())
+        elif spec.family == "torchmod":
+            self._update_torchmod(
+                state.params, state.ages, data, spec, nodes.long()
+            )
         elif spec.family == "logreg":
             self._update_logreg(state.params, state.ages, data, spec, nodes.long())
         elif spec.family == "mlp":
@@ -431,6 +449,46 @@ class TorchBackend:
                         torch.tensor([node_t], dtype=torch.long),
                         torch.tensor([r], dtype=torch.long),
                     )
+
+    # -- arbitrary nn.Module family (CNN path) ----------
+
+    def _update_torchmod(self, params, ages, data, spec, nodes) -> None:
+        """Autograd local SGD on a per-node view of the arena row
+        (TorchModelHandler._update/_local_step semantics,
+        gossipy/model/handler.py:235-258). Convolutions run on MIOpen via
+        torch — the measured-right vendor path for small convs."""
+        module = spec.template().to(params.device)
+        crit = torch.nn.CrossEntropyLoss()
+        for idx in nodes.tolist():
+            c = int(data.counts[idx])
+            if c == 0:
+                continue
+            spec.load_row(module, params[idx])
+            opt = torch.optim.SGD(
+                module.parameters(), lr=spec.lr,
+                weight_decay=spec.weight_decay,
+            )
+            x = data.x[idx, :c].view(c, *spec.input_shape)
+            yv = data.y[idx, :c].long()
+            bs = c if spec.batch_size == 0 else spec.batch_size
+            for _ in range(max(1, spec.local_epochs)):
+                for s in range(0, c, bs):
+                    opt.zero_grad()
+                    crit(module(x[s : s + bs]), yv[s : s + bs]).backward()
+                    opt.step()
+                    ages[idx] += 1
+            spec.store_row(module, params[idx])
+
+    def torchmod_scores(self, state, spec, nodes, X) -> torch.Tensor:
+        """Class scores ``[R, n, k]`` for the eval sweep."""
+        module = spec.template().to(state.params.device)
+        outs = []
+        with torch.no_grad():
+            xin = X.view(X.shape[0], *spec.input_shape)
+            for idx in nodes.tolist():
+                spec.load_row(module, state.params[idx])
+                outs.append(module(xin))
+        return torch.stack(outs)
 
     # -- matrix factorization (K9/K10) ---------------------------------------
 
@@ -765,6 +823,8 @@ class TorchBackend:
         """Class scores ``[len(nodes), n_samples, k]`` of each node's model
         on a shared input matrix (global eval set)."""
         nodes = nodes.long()
+        if spec.family == "torchmod":
+            return self.torchmod_scores(state, spec, nodes, X)
         if spec.family in ("pegasos", "adaline"):
             w = state.params[nodes]  # [R, d]
             s = X @ w.t()  # [n, R]
@@ -820,6 +880,13 @@ class HIPBackend(TorchBackend):
     def update(self, state, data, spec, nodes) -> None:
         if len(nodes) == 0:
             return
+        if spec.family == "torchmod":
+            # autograd step on arena-row views; convs go through MIOpen —
+            # the vendor conv path (SURVEY.md §7 step 7 measured decision)
+            self._update_torchmod(
+                state.params, state.ages, data, spec, nodes.long()
+            )
+            return
         nodes_dev = nodes.to(state.params.device, torch.int32)
         empty = torch.zeros(0, dtype=torch.int32, device=state.params.device)
         self._dispatch(
@@ -832,6 +899,12 @@ class HIPBackend(TorchBackend):
         reply_slots, del_pids=None,
     ) -> None:
         if len(recv_nodes) == 0:
+            return
+        if spec.family == "torchmod":
+            TorchBackend.deliver(
+                self, state, pool, data, spec, recv_nodes, recv_ptr,
+                del_slots, reply_slots, del_pids,
+            )
             return
         dev = state.params.device
         pids = (

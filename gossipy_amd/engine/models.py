@@ -27,7 +27,7 @@ import numpy as np
 
 from ..core import CreateModelMode
 
-__all__ = ["PegasosSpec", "AdaLineSpec", "LogRegSpec", "MLPSpec", "MFSpec", "KMeansSpec"]
+__all__ = ["PegasosSpec", "AdaLineSpec", "LogRegSpec", "MLPSpec", "MFSpec", "KMeansSpec", "TorchModuleSpec"]
 
 
 def _fortran_to_arena(layers: List[Tuple[int, int, int, int]], D: int) -> np.ndarray:
@@ -300,3 +300,100 @@ class KMeansSpec:
     @property
     def D(self) -> int:
         return self.k * self.dim
+
+
+class TorchModuleSpec:
+    """Engine family for arbitrary ``nn.Module`` architectures (the CNN
+    path of SURVEY.md §7 step 7 — Onoszko 2021's CIFAR10Net,
+    main_onoszko_2021.py:31-60 — and any user model).
+
+    The node population's parameters still live in the packed ``[n, D]``
+    arena (flattened in ``parameters()`` order, each tensor C-contiguous),
+    so snapshots, merges and cross-GPU traffic use the same HIP kernels and
+    RCCL paths as every other family. The local SGD step runs through torch
+    autograd on a per-node view of the arena row — convolutions therefore
+    execute on MIOpen, which is the measured-right choice for 3x3 convs
+    ("conv via MIOpen or an im2col+MFMA kernel — decide by
+    measurement").
+
+    ``module_factory`` must build a fresh module (identical architecture)
+    each call; inputs may be N-d (the DataArena stores them flattened and
+    ``input_shape`` restores them).
+    """
+
+    family = "torchmod"
+    n_parts = 0
+    sample_size = 0.0
+    pass_through = False
+    age_width = 1
+
+    def __init__(
+        self,
+        module_factory,
+        input_shape,
+        lr: float = 0.1,
+        weight_decay: float = 0.0,
+        local_epochs: int = 1,
+        batch_size: int = 32,
+        mode: CreateModelMode = CreateModelMode.MERGE_UPDATE,
+    ):
+        self.module_factory = module_factory
+        self.input_shape = tuple(input_shape)
+        self.lr = lr
+        self.weight_decay = weight_decay
+        self.local_epochs = local_epochs
+        self.batch_size = batch_size
+        self.mode = mode
+        proto = module_factory()
+        self._shapes = [tuple(p.shape) for p in proto.parameters()]
+        self._numels = [int(np.prod(s)) for s in self._shapes]
+        self._D = int(sum(self._numels))
+
+    @property
+    def D(self) -> int:
+        return self._D
+
+    @property
+    def d_in(self) -> int:
+        return int(np.prod(self.input_shape))
+
+    def template(self):
+        """A module instance for row_to/row_from round-trips."""
+        if not hasattr(self, "_template"):
+            self._template = self.module_factory()
+        return self._template
+
+    def load_row(self, module, row) -> None:
+        """Copy an arena row into a module's parameters (in place)."""
+        import torch
+
+        off = 0
+        with torch.no_grad():
+            for p, n in zip(module.parameters(), self._numels):
+                p.copy_(row[off : off + n].view(p.shape))
+                off += n
+
+    def store_row(self, module, row) -> None:
+        import torch
+
+        off = 0
+        with torch.no_grad():
+            for p, n in zip(module.parameters(), self._numels):
+                row[off : off + n] = p.reshape(-1)
+                off += n
+
+    def init_row(self, generator) -> "np.ndarray":
+        """Xavier-uniform weights / zero biases per the reference's intent
+        (main_onoszko_2021.py:38-44), drawn from the tape stream."""
+        out = np.empty(self._D, dtype=np.float32)
+        off = 0
+        for shape, n in zip(self._shapes, self._numels):
+            if len(shape) >= 2:
+                fan_out = shape[0] * int(np.prod(shape[2:]))
+                fan_in = shape[1] * int(np.prod(shape[2:]))
+                bound = float(np.sqrt(6.0 / (fan_in + fan_out)))
+                out[off : off + n] = generator.uniform(-bound, bound, size=n)
+            else:
+                out[off : off + n] = 0.0
+            off += n
+        return out
