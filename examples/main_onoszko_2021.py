@@ -39,8 +39,13 @@ def main():
     ap.add_argument("--rounds", type=int, default=30)
     ap.add_argument("--engine", action="store_true",
                     help="batched MI355X PENS instead of the object layer")
+    ap.add_argument("--engine-cnn", action="store_true",
+                    help="batched engine with the reference CIFAR10Net CNN"
+                         " (TorchModuleSpec; convs on MIOpen)")
     args = ap.parse_args()
 
+    if args.engine_cnn:
+        return main_engine_cnn(args)
     if args.engine:
         return main_engine(args)
 
@@ -126,6 +131,67 @@ def main_engine(args):
     print(f"final global eval: {report.get_evaluation(False)[-1][1]}")
     n_best = sum(len(b) for b in sim.scheduler.best_nodes or [])
     print(f"selected neighbors across nodes: {n_best}")
+
+
+
+
+def main_engine_cnn(args):
+    """CIFAR-shaped synthetic gossip with the reference CNN on the batched
+    engine (BASELINE.json config 4 shape)."""
+    from gossipy_amd.core import AntiEntropyProtocol
+    from gossipy_amd.engine import (
+        BatchedGossipSimulator,
+        DataArena,
+        EngineConfig,
+        TorchModuleSpec,
+    )
+    from gossipy_amd.simul import SimulationReport
+    import torch.nn as nn
+    import torch.nn.functional as F
+
+    class CIFAR10Net(nn.Module):
+        """The reference PENS CNN (main_onoszko_2021.py:31-60)."""
+
+        def __init__(self):
+            super().__init__()
+            self.conv1 = nn.Conv2d(3, 32, 3)
+            self.pool = nn.MaxPool2d(2, 2)
+            self.conv2 = nn.Conv2d(32, 64, 3)
+            self.conv3 = nn.Conv2d(64, 64, 3)
+            self.fc1 = nn.Linear(64 * 2 * 2, 64)
+            self.fc2 = nn.Linear(64, 10)
+
+        def forward(self, x):
+            x = self.pool(F.relu(self.conv1(x)))
+            x = self.pool(F.relu(self.conv2(x)))
+            x = self.pool(F.relu(self.conv3(x)))
+            x = x.view(-1, 64 * 2 * 2)
+            x = F.relu(self.fc1(x))
+            return self.fc2(x)
+
+    device = torch.device("cuda:0" if torch.cuda.is_available() else "cpu")
+    n = args.nodes
+    rng = np.random.default_rng(42)
+    per = 40
+    labels = rng.integers(0, 10, size=n * per)
+    x = rng.normal(0, 0.3, size=(len(labels), 3, 32, 32)).astype(np.float32)
+    for c in range(10):
+        x[labels == c, c % 3] += 0.8 + 0.25 * c
+    X = torch.from_numpy(x.reshape(len(labels), -1))
+    y = torch.from_numpy(labels).float()
+    shards = [(X[i * per : (i + 1) * per], y[i * per : (i + 1) * per]) for i in range(n)]
+    data = DataArena.from_shards(shards, device, global_eval=(X, y))
+    spec = TorchModuleSpec(CIFAR10Net, input_shape=(3, 32, 32), lr=0.1, batch_size=32)
+    cfg = EngineConfig(
+        n_nodes=n, delta=10, protocol=AntiEntropyProtocol.PUSH,
+        model_size=spec.D, sampling_eval=0.25, seed=42,
+    )
+    sim = BatchedGossipSimulator(cfg, spec, data, device=device)
+    report = SimulationReport()
+    sim.add_receiver(report)
+    sim.init_nodes()
+    sim.start(n_rounds=args.rounds)
+    print(f"final global eval: {report.get_evaluation(False)[-1][1]}")
 
 
 if __name__ == "__main__":

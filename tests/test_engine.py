@@ -1473,3 +1473,64 @@ class TestTorchModuleEngine:
 
         s1, s2 = run(), run()
         assert torch.allclose(s1.local_params(), s2.local_params(), atol=1e-6)
+
+
+def _cnn_worker(rank, world, port, q):
+    import torch.distributed as dist
+
+    from gossipy_amd.engine import TorchModuleSpec
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        spec = TorchModuleSpec(
+            _cifar10net, input_shape=(3, 32, 32), lr=0.1, batch_size=0
+        )
+        full = _cnn_data()
+        lo = rank * 4
+        data = DataArena(
+            full.x[lo : lo + 4], full.y[lo : lo + 4], full.counts[lo : lo + 4],
+            gx=full.gx, gy=full.gy,
+        )
+        cfg = EngineConfig(
+            n_nodes=8, delta=5, protocol=AntiEntropyProtocol.PUSH,
+            model_size=spec.D, sampling_eval=0.0, seed=41,
+        )
+        sim = BatchedGossipSimulator(cfg, spec, data, device=torch.device("cpu"))
+        sim.init_nodes()
+        sim.start(n_rounds=2)
+        out = sim.gather_params()
+        if rank == 0:
+            q.put(out.numpy())
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_cnn_two_rank_matches_single():
+    from gossipy_amd.engine import TorchModuleSpec
+
+    spec = TorchModuleSpec(
+        _cifar10net, input_shape=(3, 32, 32), lr=0.1, batch_size=0
+    )
+    data = _cnn_data()
+    cfg = EngineConfig(
+        n_nodes=8, delta=5, protocol=AntiEntropyProtocol.PUSH,
+        model_size=spec.D, sampling_eval=0.0, seed=41,
+    )
+    ref = BatchedGossipSimulator(cfg, spec, data, device=torch.device("cpu"))
+    ref.init_nodes()
+    ref.start(n_rounds=2)
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [
+        ctx.Process(target=_cnn_worker, args=(r, 2, 29547, q)) for r in range(2)
+    ]
+    for p in procs:
+        p.start()
+    got = q.get(timeout=240)
+    for p in procs:
+        p.join(timeout=60)
+    assert np.allclose(ref.local_params().numpy(), got, atol=1e-5)

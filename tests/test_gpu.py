@@ -952,3 +952,27 @@ class TestMFMAMLP:
         HIPBackend().update(gs, gd, spec, nodes)
         torch.cuda.synchronize()
         assert _close(cs.params, gs.params, 2e-3)
+
+
+class TestTorchModuleGPU:
+    def test_cnn_engine_gpu(self):
+        from gossipy_amd.engine import TorchModuleSpec
+        from tests.test_engine import _cifar10net, _cnn_data
+
+        spec = TorchModuleSpec(
+            _cifar10net, input_shape=(3, 32, 32), lr=0.1, batch_size=0
+        )
+        data = _cnn_data(device=CUDA)
+        cfg = EngineConfig(
+            n_nodes=8, delta=5, protocol=AntiEntropyProtocol.PUSH,
+            model_size=spec.D, sampling_eval=0.0, seed=41,
+        )
+        sim = BatchedGossipSimulator(cfg, spec, data, device=CUDA)
+        assert sim.backend.name == "hip"  # snapshots/merges on HIP kernels
+        rep = SimulationReport()
+        sim.add_receiver(rep)
+        sim.init_nodes()
+        sim.start(n_rounds=8)
+        torch.cuda.synchronize()
+        evals = rep.get_evaluation(False)
+        assert evals[-1][1]["accuracy"] > 0.25
