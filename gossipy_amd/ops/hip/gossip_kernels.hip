@@ -394,10 +394,8 @@ DEV_INLINE void logreg_part_merge(const LogregPartArgs& a, float* W, int* agev,
     __syncthreads();
 }
 
-__global__ void __launch_bounds__(128)
-tick_logreg_part_kernel(LogregPartArgs a)
+DEV_INLINE void logreg_part_process_node(const LogregPartArgs& a, int i)
 {
-    int i = blockIdx.x;
     int node = a.nodes[i];
     int tid = threadIdx.x;
     extern __shared__ float sm[];
@@ -453,6 +451,12 @@ tick_logreg_part_kernel(LogregPartArgs a)
         a.params[(long)node * a.D + e] = W[e];
     for (int p = tid; p < a.P; p += blockDim.x)
         a.ages[(long)node * a.P + p] = agev[p];
+}
+
+__global__ void __launch_bounds__(128)
+tick_logreg_part_kernel(LogregPartArgs a)
+{
+    logreg_part_process_node(a, blockIdx.x);
 }
 
 // ---------------------------------------------------------------------------
@@ -828,12 +832,10 @@ DEV_INLINE void linear_update(const LinearArgs& a, int node, float* w,
     if (!a.is_pegasos) age += c;  // AdaLine ages by sample count
 }
 
-__global__ void __launch_bounds__(WAVE)
-tick_linear_kernel(LinearArgs a)
+DEV_INLINE void linear_process_node(const LinearArgs& a, int i)
 {
-    int i = blockIdx.x;
     int node = a.nodes[i];
-    int lane = threadIdx.x;
+    int lane = threadIdx.x & (WAVE - 1);
     int nreg = (a.d + WAVE - 1) / WAVE;
     float w[LIN_MAX_REGS];
     for (int r = 0; r < nreg; ++r) {
@@ -897,6 +899,12 @@ tick_linear_kernel(LinearArgs a)
         if (e < a.d) a.params[(long)node * a.d + e] = w[r];
     }
     if (lane == 0) a.ages[node] = age;
+}
+
+__global__ void __launch_bounds__(WAVE)
+tick_linear_kernel(LinearArgs a)
+{
+    linear_process_node(a, blockIdx.x);
 }
 
 // ---------------------------------------------------------------------------
@@ -1393,6 +1401,129 @@ tick_pens_kernel(PensArgs a)
 }
 
 // ---------------------------------------------------------------------------
+// single-block round kernels (linear / partitioned): for schedules whose
+// per-tick batches are tiny (async gossip, tokenized 100-node configs),
+// the whole round runs as ONE plain launch of ONE workgroup —
+// __syncthreads() is the tick barrier, replacing ~2 launches per tick.
+// ---------------------------------------------------------------------------
+
+struct LinRoundArgs {
+    LinearArgs base;
+    const int* snap_nodes; const int* snap_slots; const int* snap_tptr;
+    const int* recv_nodes; const int* recv_nptr; const int* recv_tptr;
+    const int* del_slots; const int* reply_slots;
+    const int* pull_nodes; const int* pull_slots; const int* pull_tptr;
+    const int* rep_nodes; const int* rep_nptr; const int* rep_tptr;
+    const int* rep_slots;
+    int delta;
+};
+
+__global__ void __launch_bounds__(WAVE)
+sb_round_linear_kernel(LinRoundArgs c)
+{
+    LinearArgs a = c.base;
+    int tid = threadIdx.x;
+    for (int t = 0; t < c.delta; ++t) {
+        int s0 = c.snap_tptr[t], s1 = c.snap_tptr[t + 1];
+        for (int i = s0; i < s1; ++i) {
+            int node = c.snap_nodes[i], slot = c.snap_slots[i];
+            for (int e = tid; e < a.d; e += blockDim.x)
+                a.slots[(long)slot * a.d + e] = a.params[(long)node * a.d + e];
+            if (tid == 0) a.slot_ages[slot] = a.ages[node];
+        }
+        __syncthreads();
+        int r0 = c.recv_tptr[t], r1 = c.recv_tptr[t + 1];
+        if (r1 > r0) {
+            a.nodes = c.recv_nodes;
+            a.ptr = c.recv_nptr;
+            a.dslots = c.del_slots;
+            a.rslots = c.reply_slots;
+            for (int i = r0; i < r1; ++i) linear_process_node(a, i);
+            __syncthreads();
+        }
+        int p0 = c.pull_tptr[t], p1 = c.pull_tptr[t + 1];
+        for (int i = p0; i < p1; ++i) {
+            int node = c.pull_nodes[i], slot = c.pull_slots[i];
+            for (int e = tid; e < a.d; e += blockDim.x)
+                a.slots[(long)slot * a.d + e] = a.params[(long)node * a.d + e];
+            if (tid == 0) a.slot_ages[slot] = a.ages[node];
+        }
+        if (p1 > p0) __syncthreads();
+        int q0 = c.rep_tptr[t], q1 = c.rep_tptr[t + 1];
+        if (q1 > q0) {
+            a.nodes = c.rep_nodes;
+            a.ptr = c.rep_nptr;
+            a.dslots = c.rep_slots;
+            a.rslots = nullptr;
+            for (int i = q0; i < q1; ++i) linear_process_node(a, i);
+            __syncthreads();
+        }
+    }
+}
+
+struct PartRoundArgs {
+    LogregPartArgs base;
+    const int* snap_nodes; const int* snap_slots; const int* snap_tptr;
+    const int* recv_nodes; const int* recv_nptr; const int* recv_tptr;
+    const int* del_slots; const int* reply_slots; const int* del_pids;
+    const int* pull_nodes; const int* pull_slots; const int* pull_tptr;
+    const int* rep_nodes; const int* rep_nptr; const int* rep_tptr;
+    const int* rep_slots; const int* rep_pids;
+    int delta;
+};
+
+__global__ void __launch_bounds__(128)
+sb_round_logreg_part_kernel(PartRoundArgs c)
+{
+    LogregPartArgs a = c.base;
+    int tid = threadIdx.x;
+    for (int t = 0; t < c.delta; ++t) {
+        int s0 = c.snap_tptr[t], s1 = c.snap_tptr[t + 1];
+        for (int i = s0; i < s1; ++i) {
+            int node = c.snap_nodes[i], slot = c.snap_slots[i];
+            for (int e = tid; e < a.D; e += blockDim.x)
+                a.slots[(long)slot * a.D + e] = a.params[(long)node * a.D + e];
+            for (int p = tid; p < a.P; p += blockDim.x)
+                a.slot_ages[(long)slot * a.P + p] = a.ages[(long)node * a.P + p];
+        }
+        __syncthreads();
+        int r0 = c.recv_tptr[t], r1 = c.recv_tptr[t + 1];
+        if (r1 > r0) {
+            a.nodes = c.recv_nodes;
+            a.ptr = c.recv_nptr;
+            a.dslots = c.del_slots;
+            a.rslots = c.reply_slots;
+            a.dpids = c.del_pids;
+            for (int i = r0; i < r1; ++i) {
+                logreg_part_process_node(a, i);
+                __syncthreads();
+            }
+        }
+        int p0 = c.pull_tptr[t], p1 = c.pull_tptr[t + 1];
+        for (int i = p0; i < p1; ++i) {
+            int node = c.pull_nodes[i], slot = c.pull_slots[i];
+            for (int e = tid; e < a.D; e += blockDim.x)
+                a.slots[(long)slot * a.D + e] = a.params[(long)node * a.D + e];
+            for (int p = tid; p < a.P; p += blockDim.x)
+                a.slot_ages[(long)slot * a.P + p] = a.ages[(long)node * a.P + p];
+        }
+        if (p1 > p0) __syncthreads();
+        int q0 = c.rep_tptr[t], q1 = c.rep_tptr[t + 1];
+        if (q1 > q0) {
+            a.nodes = c.rep_nodes;
+            a.ptr = c.rep_nptr;
+            a.dslots = c.rep_slots;
+            a.rslots = nullptr;
+            a.dpids = c.rep_pids;
+            for (int i = q0; i < q1; ++i) {
+                logreg_part_process_node(a, i);
+                __syncthreads();
+            }
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
 // host wrappers
 // ---------------------------------------------------------------------------
 
@@ -1631,6 +1762,25 @@ static RoundArrays unpack_round(
     return r;
 }
 
+static int sb_threshold()
+{
+    const char* v = getenv("GOSSIPY_SB_MAX");
+    return v ? atoi(v) : 4;
+}
+
+// largest per-tick batch across all four event kinds (host-side tptrs)
+static int max_group(const RoundArrays& r)
+{
+    int mx = 0;
+    for (int t = 0; t < r.delta; ++t) {
+        mx = std::max(mx, r.snap_tptr[t + 1] - r.snap_tptr[t]);
+        mx = std::max(mx, r.recv_tptr[t + 1] - r.recv_tptr[t]);
+        mx = std::max(mx, r.pull_tptr[t + 1] - r.pull_tptr[t]);
+        mx = std::max(mx, r.rep_tptr[t + 1] - r.rep_tptr[t]);
+    }
+    return mx;
+}
+
 static void launch_snap(const float* params, const int* ages, float* slots,
                         int* slot_ages, const int* nodes, const int* slot_ids,
                         int n, int D, hipStream_t s, int A = 1)
@@ -1674,6 +1824,32 @@ void run_round_logreg(
     int bsmax = (bs == 0) ? a.Smax : std::min<int>(bs, a.Smax);
     size_t smem = sizeof(float) * (2 * a.D + (size_t)bsmax * a.d + (size_t)bsmax * a.k);
     TORCH_CHECK(smem <= 160 * 1024, "logreg LDS budget exceeded");
+    if (max_group(r) <= sb_threshold()) {
+        // tiny batches: the coop round kernel launched PLAIN with one
+        // workgroup (round_sync degrades to __syncthreads)
+        auto dev = params.device();
+        auto st = snap_tptr.to(dev), rt = recv_tptr.to(dev);
+        auto pt = pull_tptr.to(dev), qt = rep_tptr.to(dev);
+        CoopRoundArgs c;
+        c.base = a;
+        c.snap_nodes = r.snap_nodes; c.snap_slots = r.snap_slots;
+        c.snap_tptr = st.data_ptr<int>();
+        c.recv_nodes = r.recv_nodes; c.recv_nptr = r.recv_nptr;
+        c.recv_tptr = rt.data_ptr<int>();
+        c.del_slots = r.del_slots; c.reply_slots = r.reply_slots;
+        c.pull_nodes = r.pull_nodes; c.pull_slots = r.pull_slots;
+        c.pull_tptr = pt.data_ptr<int>();
+        c.rep_nodes = r.rep_nodes; c.rep_nptr = r.rep_nptr;
+        c.rep_tptr = qt.data_ptr<int>();
+        c.rep_slots = r.rep_slots;
+        c.delta = r.delta;
+        hipLaunchKernelGGL(coop_round_logreg_kernel, dim3(1), dim3(128),
+                           smem, s, c);
+        static thread_local std::vector<torch::Tensor> keep;
+        keep.insert(keep.end(), {st, rt, pt, qt});
+        if (keep.size() > 64) keep.erase(keep.begin(), keep.begin() + 32);
+        return;
+    }
     for (int t = 0; t < r.delta; ++t) {
         int s0 = r.snap_tptr[t], s1 = r.snap_tptr[t + 1];
         if (s1 > s0)
@@ -1731,6 +1907,31 @@ void run_round_linear(
     a.d = d; a.Smax = X.size(1);
     a.lrlam = lrlam; a.is_pegasos = is_pegasos; a.mode = mode;
     a.update_only = 0;
+    if (max_group(r) <= sb_threshold()) {
+        // tiny batches: one plain single-workgroup launch for the round
+        auto dev = params.device();
+        auto st = snap_tptr.to(dev), rt = recv_tptr.to(dev);
+        auto pt = pull_tptr.to(dev), qt = rep_tptr.to(dev);
+        LinRoundArgs c;
+        c.base = a;
+        c.snap_nodes = r.snap_nodes; c.snap_slots = r.snap_slots;
+        c.snap_tptr = st.data_ptr<int>();
+        c.recv_nodes = r.recv_nodes; c.recv_nptr = r.recv_nptr;
+        c.recv_tptr = rt.data_ptr<int>();
+        c.del_slots = r.del_slots; c.reply_slots = r.reply_slots;
+        c.pull_nodes = r.pull_nodes; c.pull_slots = r.pull_slots;
+        c.pull_tptr = pt.data_ptr<int>();
+        c.rep_nodes = r.rep_nodes; c.rep_nptr = r.rep_nptr;
+        c.rep_tptr = qt.data_ptr<int>();
+        c.rep_slots = r.rep_slots;
+        c.delta = r.delta;
+        hipLaunchKernelGGL(sb_round_linear_kernel, dim3(1), dim3(WAVE), 0, s, c);
+        // keep the uploaded tptr tensors alive past the async launch
+        static thread_local std::vector<torch::Tensor> keep;
+        keep.insert(keep.end(), {st, rt, pt, qt});
+        if (keep.size() > 64) keep.erase(keep.begin(), keep.begin() + 32);
+        return;
+    }
     for (int t = 0; t < r.delta; ++t) {
         int s0 = r.snap_tptr[t], s1 = r.snap_tptr[t + 1];
         if (s1 > s0)
@@ -1885,6 +2086,32 @@ void run_round_logreg_part(
                                    (size_t)bsmax * a.k) +
                   sizeof(int) * 2 * a.P;
     TORCH_CHECK(smem <= 160 * 1024, "partitioned logreg LDS budget exceeded");
+    if (max_group(r) <= sb_threshold()) {
+        auto dev = params.device();
+        auto st = snap_tptr.to(dev), rt = recv_tptr.to(dev);
+        auto pt = pull_tptr.to(dev), qt = rep_tptr.to(dev);
+        PartRoundArgs c;
+        c.base = a;
+        c.snap_nodes = r.snap_nodes; c.snap_slots = r.snap_slots;
+        c.snap_tptr = st.data_ptr<int>();
+        c.recv_nodes = r.recv_nodes; c.recv_nptr = r.recv_nptr;
+        c.recv_tptr = rt.data_ptr<int>();
+        c.del_slots = r.del_slots; c.reply_slots = r.reply_slots;
+        c.del_pids = d_pids;
+        c.pull_nodes = r.pull_nodes; c.pull_slots = r.pull_slots;
+        c.pull_tptr = pt.data_ptr<int>();
+        c.rep_nodes = r.rep_nodes; c.rep_nptr = r.rep_nptr;
+        c.rep_tptr = qt.data_ptr<int>();
+        c.rep_slots = r.rep_slots;
+        c.rep_pids = r_pids;
+        c.delta = r.delta;
+        hipLaunchKernelGGL(sb_round_logreg_part_kernel, dim3(1), dim3(128),
+                           smem, s, c);
+        static thread_local std::vector<torch::Tensor> keep;
+        keep.insert(keep.end(), {st, rt, pt, qt});
+        if (keep.size() > 64) keep.erase(keep.begin(), keep.begin() + 32);
+        return;
+    }
     for (int t = 0; t < r.delta; ++t) {
         int s0 = r.snap_tptr[t], s1 = r.snap_tptr[t + 1];
         if (s1 > s0)
@@ -2205,11 +2432,21 @@ struct CoopRoundArgs {
     int delta;
 };
 
+// grid barrier that degrades to a block barrier when the kernel was
+// launched PLAIN with one workgroup (the single-block round path for
+// tiny-batch schedules — one launch per round, no cooperative API)
+DEV_INLINE void round_sync()
+{
+    if (gridDim.x == 1) {
+        __syncthreads();
+    } else {
+        cooperative_groups::this_grid().sync();
+    }
+}
+
 __global__ void __launch_bounds__(128)
 coop_round_logreg_kernel(CoopRoundArgs c)
 {
-    namespace cg = cooperative_groups;
-    cg::grid_group grid = cg::this_grid();
     LogregArgs a = c.base;
     const int nb = gridDim.x;
     const int bid = blockIdx.x;
@@ -2223,7 +2460,7 @@ coop_round_logreg_kernel(CoopRoundArgs c)
         int q0 = c.rep_tptr[t], q1 = c.rep_tptr[t + 1];
 
         if (s1 > s0) {
-            if (dirty) { grid.sync(); dirty = false; }
+            if (dirty) { round_sync(); dirty = false; }
             // block-strided row copies (the snapshot sub-phase)
             for (int i = s0 + bid; i < s1; i += nb) {
                 int node = c.snap_nodes[i];
@@ -2236,7 +2473,7 @@ coop_round_logreg_kernel(CoopRoundArgs c)
             dirty = true;
         }
         if (r1 > r0) {
-            if (dirty) { grid.sync(); dirty = false; }
+            if (dirty) { round_sync(); dirty = false; }
             a.nodes = c.recv_nodes;
             a.ptr = c.recv_nptr;
             a.dslots = c.del_slots;
@@ -2248,7 +2485,7 @@ coop_round_logreg_kernel(CoopRoundArgs c)
             dirty = true;
         }
         if (p1 > p0) {
-            if (dirty) { grid.sync(); dirty = false; }
+            if (dirty) { round_sync(); dirty = false; }
             for (int i = p0 + bid; i < p1; i += nb) {
                 int node = c.pull_nodes[i];
                 int slot = c.pull_slots[i];
@@ -2260,7 +2497,7 @@ coop_round_logreg_kernel(CoopRoundArgs c)
             dirty = true;
         }
         if (q1 > q0) {
-            if (dirty) { grid.sync(); dirty = false; }
+            if (dirty) { round_sync(); dirty = false; }
             a.nodes = c.rep_nodes;
             a.ptr = c.rep_nptr;
             a.dslots = c.rep_slots;

@@ -976,3 +976,63 @@ class TestTorchModuleGPU:
         torch.cuda.synchronize()
         evals = rep.get_evaluation(False)
         assert evals[-1][1]["accuracy"] > 0.25
+
+
+class TestSingleBlockRound:
+    """Tiny-batch rounds run as ONE plain single-workgroup launch; forced
+    here via GOSSIPY_SB_MAX and compared against the stream executor."""
+
+    def _run(self, spec, cfg, sb: bool):
+        import os
+
+        X, y = make_synthetic_classification((640, 57, 2), seed=0, margin=2.0)
+        idx = np.random.default_rng(0).permutation(640)
+        shards = [(X[s], y[s]) for s in np.array_split(idx[:576], cfg.n_nodes)]
+        if spec.family == "pegasos":
+            for i, (xs, ys) in enumerate(shards):
+                shards[i] = (xs, 2 * ys.float() - 1)
+        data = DataArena.from_shards(
+            shards, CUDA, global_eval=(X[idx[576:]], y[idx[576:]])
+        )
+        os.environ["GOSSIPY_SB_MAX"] = "1000" if sb else "0"
+        try:
+            sim = BatchedGossipSimulator(cfg, spec, data, device=CUDA)
+            sim.init_nodes()
+            sim.start(n_rounds=4)
+            torch.cuda.synchronize()
+        finally:
+            del os.environ["GOSSIPY_SB_MAX"]
+        return sim
+
+    def test_sb_logreg_matches_stream(self):
+        cfg = EngineConfig(
+            n_nodes=64, delta=10, protocol=AntiEntropyProtocol.PUSH_PULL,
+            model_size=116, sampling_eval=0.0, seed=3,
+        )
+        spec = LogRegSpec(d_in=57, n_classes=2, lr=0.1)
+        a = self._run(spec, cfg, sb=True)
+        b = self._run(spec, cfg, sb=False)
+        assert torch.allclose(a.local_params(), b.local_params(), atol=1e-5)
+        assert torch.equal(a.state.ages, b.state.ages)
+
+    def test_sb_pegasos_matches_stream(self):
+        cfg = EngineConfig(
+            n_nodes=64, delta=10, protocol=AntiEntropyProtocol.PUSH_PULL,
+            model_size=57, sampling_eval=0.0, seed=5,
+        )
+        spec = PegasosSpec(d_in=57, lam=0.01)
+        a = self._run(spec, cfg, sb=True)
+        b = self._run(spec, cfg, sb=False)
+        assert torch.allclose(a.local_params(), b.local_params(), atol=1e-5)
+        assert torch.equal(a.state.ages, b.state.ages)
+
+    def test_sb_partitioned_matches_stream(self):
+        cfg = EngineConfig(
+            n_nodes=64, delta=10, protocol=AntiEntropyProtocol.PUSH_PULL,
+            model_size=116, sampling_eval=0.0, seed=7, n_parts=4,
+        )
+        spec = LogRegSpec(d_in=57, n_classes=2, lr=0.1, n_parts=4)
+        a = self._run(spec, cfg, sb=True)
+        b = self._run(spec, cfg, sb=False)
+        assert torch.allclose(a.local_params(), b.local_params(), atol=1e-5)
+        assert torch.equal(a.state.ages, b.state.ages)
