@@ -1401,6 +1401,105 @@ tick_pens_kernel(PensArgs a)
 }
 
 // ---------------------------------------------------------------------------
+// cooperative whole-round kernel (logreg family): the entire round — every
+// tick's snapshot and delivery batch — runs inside ONE kernel launch, with
+// grid.sync() as the tick barrier. Grids here are tiny (max batch ≈ a few
+// dozen blocks on 256 CUs), so cooperative residency is guaranteed and the
+// launch is REJECTED (not deadlocked) if ever oversubscribed. Cuts the
+// ~200 per-round launch overheads of the stream executor to one.
+// ---------------------------------------------------------------------------
+
+struct CoopRoundArgs {
+    LogregArgs base;  // nodes/ptr/dslots/rslots are set per group inside
+    const int* snap_nodes; const int* snap_slots; const int* snap_tptr;
+    const int* recv_nodes; const int* recv_nptr; const int* recv_tptr;
+    const int* del_slots; const int* reply_slots;
+    const int* pull_nodes; const int* pull_slots; const int* pull_tptr;
+    const int* rep_nodes; const int* rep_nptr; const int* rep_tptr;
+    const int* rep_slots;
+    int delta;
+};
+
+// grid barrier that degrades to a block barrier when the kernel was
+// launched PLAIN with one workgroup (the single-block round path for
+// tiny-batch schedules — one launch per round, no cooperative API)
+DEV_INLINE void round_sync()
+{
+    if (gridDim.x == 1) {
+        __syncthreads();
+    } else {
+        cooperative_groups::this_grid().sync();
+    }
+}
+
+__global__ void __launch_bounds__(128)
+coop_round_logreg_kernel(CoopRoundArgs c)
+{
+    LogregArgs a = c.base;
+    const int nb = gridDim.x;
+    const int bid = blockIdx.x;
+    const int tid = threadIdx.x;
+    bool dirty = false;  // writes since the last grid sync
+
+    for (int t = 0; t < c.delta; ++t) {
+        int s0 = c.snap_tptr[t], s1 = c.snap_tptr[t + 1];
+        int r0 = c.recv_tptr[t], r1 = c.recv_tptr[t + 1];
+        int p0 = c.pull_tptr[t], p1 = c.pull_tptr[t + 1];
+        int q0 = c.rep_tptr[t], q1 = c.rep_tptr[t + 1];
+
+        if (s1 > s0) {
+            if (dirty) { round_sync(); dirty = false; }
+            // block-strided row copies (the snapshot sub-phase)
+            for (int i = s0 + bid; i < s1; i += nb) {
+                int node = c.snap_nodes[i];
+                int slot = c.snap_slots[i];
+                for (int e = tid; e < a.D; e += blockDim.x)
+                    a.slots[(long)slot * a.D + e] =
+                        a.params[(long)node * a.D + e];
+                if (tid == 0) a.slot_ages[slot] = a.ages[node];
+            }
+            dirty = true;
+        }
+        if (r1 > r0) {
+            if (dirty) { round_sync(); dirty = false; }
+            a.nodes = c.recv_nodes;
+            a.ptr = c.recv_nptr;
+            a.dslots = c.del_slots;
+            a.rslots = c.reply_slots;
+            for (int i = r0 + bid; i < r1; i += nb) {
+                logreg_process_node(a, i);
+                __syncthreads();
+            }
+            dirty = true;
+        }
+        if (p1 > p0) {
+            if (dirty) { round_sync(); dirty = false; }
+            for (int i = p0 + bid; i < p1; i += nb) {
+                int node = c.pull_nodes[i];
+                int slot = c.pull_slots[i];
+                for (int e = tid; e < a.D; e += blockDim.x)
+                    a.slots[(long)slot * a.D + e] =
+                        a.params[(long)node * a.D + e];
+                if (tid == 0) a.slot_ages[slot] = a.ages[node];
+            }
+            dirty = true;
+        }
+        if (q1 > q0) {
+            if (dirty) { round_sync(); dirty = false; }
+            a.nodes = c.rep_nodes;
+            a.ptr = c.rep_nptr;
+            a.dslots = c.rep_slots;
+            a.rslots = nullptr;
+            for (int i = q0 + bid; i < q1; i += nb) {
+                logreg_process_node(a, i);
+                __syncthreads();
+            }
+            dirty = true;
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
 // single-block round kernels (linear / partitioned): for schedules whose
 // per-tick batches are tiny (async gossip, tokenized 100-node configs),
 // the whole round runs as ONE plain launch of ONE workgroup —
@@ -2412,104 +2511,7 @@ void run_round_mf(
     }
 }
 
-// ---------------------------------------------------------------------------
-// cooperative whole-round kernel (logreg family): the entire round — every
-// tick's snapshot and delivery batch — runs inside ONE kernel launch, with
-// grid.sync() as the tick barrier. Grids here are tiny (max batch ≈ a few
-// dozen blocks on 256 CUs), so cooperative residency is guaranteed and the
-// launch is REJECTED (not deadlocked) if ever oversubscribed. Cuts the
-// ~200 per-round launch overheads of the stream executor to one.
-// ---------------------------------------------------------------------------
 
-struct CoopRoundArgs {
-    LogregArgs base;  // nodes/ptr/dslots/rslots are set per group inside
-    const int* snap_nodes; const int* snap_slots; const int* snap_tptr;
-    const int* recv_nodes; const int* recv_nptr; const int* recv_tptr;
-    const int* del_slots; const int* reply_slots;
-    const int* pull_nodes; const int* pull_slots; const int* pull_tptr;
-    const int* rep_nodes; const int* rep_nptr; const int* rep_tptr;
-    const int* rep_slots;
-    int delta;
-};
-
-// grid barrier that degrades to a block barrier when the kernel was
-// launched PLAIN with one workgroup (the single-block round path for
-// tiny-batch schedules — one launch per round, no cooperative API)
-DEV_INLINE void round_sync()
-{
-    if (gridDim.x == 1) {
-        __syncthreads();
-    } else {
-        cooperative_groups::this_grid().sync();
-    }
-}
-
-__global__ void __launch_bounds__(128)
-coop_round_logreg_kernel(CoopRoundArgs c)
-{
-    LogregArgs a = c.base;
-    const int nb = gridDim.x;
-    const int bid = blockIdx.x;
-    const int tid = threadIdx.x;
-    bool dirty = false;  // writes since the last grid sync
-
-    for (int t = 0; t < c.delta; ++t) {
-        int s0 = c.snap_tptr[t], s1 = c.snap_tptr[t + 1];
-        int r0 = c.recv_tptr[t], r1 = c.recv_tptr[t + 1];
-        int p0 = c.pull_tptr[t], p1 = c.pull_tptr[t + 1];
-        int q0 = c.rep_tptr[t], q1 = c.rep_tptr[t + 1];
-
-        if (s1 > s0) {
-            if (dirty) { round_sync(); dirty = false; }
-            // block-strided row copies (the snapshot sub-phase)
-            for (int i = s0 + bid; i < s1; i += nb) {
-                int node = c.snap_nodes[i];
-                int slot = c.snap_slots[i];
-                for (int e = tid; e < a.D; e += blockDim.x)
-                    a.slots[(long)slot * a.D + e] =
-                        a.params[(long)node * a.D + e];
-                if (tid == 0) a.slot_ages[slot] = a.ages[node];
-            }
-            dirty = true;
-        }
-        if (r1 > r0) {
-            if (dirty) { round_sync(); dirty = false; }
-            a.nodes = c.recv_nodes;
-            a.ptr = c.recv_nptr;
-            a.dslots = c.del_slots;
-            a.rslots = c.reply_slots;
-            for (int i = r0 + bid; i < r1; i += nb) {
-                logreg_process_node(a, i);
-                __syncthreads();
-            }
-            dirty = true;
-        }
-        if (p1 > p0) {
-            if (dirty) { round_sync(); dirty = false; }
-            for (int i = p0 + bid; i < p1; i += nb) {
-                int node = c.pull_nodes[i];
-                int slot = c.pull_slots[i];
-                for (int e = tid; e < a.D; e += blockDim.x)
-                    a.slots[(long)slot * a.D + e] =
-                        a.params[(long)node * a.D + e];
-                if (tid == 0) a.slot_ages[slot] = a.ages[node];
-            }
-            dirty = true;
-        }
-        if (q1 > q0) {
-            if (dirty) { round_sync(); dirty = false; }
-            a.nodes = c.rep_nodes;
-            a.ptr = c.rep_nptr;
-            a.dslots = c.rep_slots;
-            a.rslots = nullptr;
-            for (int i = q0 + bid; i < q1; i += nb) {
-                logreg_process_node(a, i);
-                __syncthreads();
-            }
-            dirty = true;
-        }
-    }
-}
 
 void run_round_coop_logreg(
     torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
