@@ -1863,8 +1863,12 @@ static RoundArrays unpack_round(
 
 static int sb_threshold()
 {
+    // measured on MI355X: the single-workgroup round is ~5% SLOWER than
+    // back-to-back stream launches even at 1-3-block batches (the stream
+    // front-end pipelines launches well on gfx950), so the path is
+    // opt-in via GOSSIPY_SB_MAX (profiles/r01_coop_ab.txt)
     const char* v = getenv("GOSSIPY_SB_MAX");
-    return v ? atoi(v) : 4;
+    return v ? atoi(v) : 0;
 }
 
 // largest per-tick batch across all four event kinds (host-side tptrs)
