@@ -288,3 +288,17 @@ class Cache:
 
 CACHE = Cache()
 """Global model cache — the in-process transport for exchanged models."""
+
+
+#: reference-name alias (gossipy/__init__.py:94-115)
+DuplicateFilter = _OncePerMessageFilter
+
+# make the layer modules reachable as package attributes, like the
+# reference's ``import gossipy; gossipy.node`` usage
+from . import core  # noqa: E402,F401
+from . import data  # noqa: E402,F401
+from . import flow_control  # noqa: E402,F401
+from . import model  # noqa: E402,F401
+from . import node  # noqa: E402,F401
+from . import simul  # noqa: E402,F401
+from . import utils  # noqa: E402,F401
