@@ -27,6 +27,7 @@ from .runner import (
     BatchedGossipSimulator,
     BatchedPENSGossipSimulator,
     BatchedTokenizedGossipSimulator,
+    RoundTimer,
 )
 from .schedule import (
     EngineConfig,
@@ -63,4 +64,5 @@ __all__ = [
     "MFSpec",
     "KMeansSpec",
     "TorchModuleSpec",
+    "RoundTimer",
 ]

@@ -52,6 +52,7 @@ __all__ = [
     "BatchedAll2AllGossipSimulator",
     "BatchedCacheNeighGossipSimulator",
     "BatchedPENSGossipSimulator",
+    "RoundTimer",
 ]
 
 
@@ -1320,3 +1321,55 @@ class BatchedPENSGossipSimulator(BatchedGossipSimulator):
             self.rounds_done += 1
             self.notify_timestep((r + 1) * self.cfg.delta - 1)
         self.notify_end()
+
+
+class RoundTimer:
+    """Opt-in GPU round timing via HIP events (SURVEY.md §5's tracing gap).
+
+    ``with RoundTimer(sim) as rt: sim.start(...)`` brackets every round's
+    device work with ``torch.cuda.Event`` pairs (hipEvent under ROCm) and
+    reports per-round GPU milliseconds — kernel + comm time as the GPU saw
+    it, independent of host overlap. On CPU devices it falls back to
+    wall-clock.
+    """
+
+    def __init__(self, sim: BatchedGossipSimulator):
+        self.sim = sim
+        self.gpu_ms: List[float] = []
+        self._orig = None
+
+    def __enter__(self):
+        sim = self.sim
+        orig = sim._run_round_fast if hasattr(sim, "_run_round_fast") else None
+        timer = self
+        use_cuda = sim.device.type == "cuda"
+        orig_start = sim.start
+
+        def timed_start(n_rounds: int = 100):
+            import time as _t
+
+            for _ in range(n_rounds):
+                if use_cuda:
+                    e0 = torch.cuda.Event(enable_timing=True)
+                    e1 = torch.cuda.Event(enable_timing=True)
+                    e0.record()
+                    orig_start(n_rounds=1)
+                    e1.record()
+                    e1.synchronize()
+                    timer.gpu_ms.append(e0.elapsed_time(e1))
+                else:
+                    t0 = _t.perf_counter()
+                    orig_start(n_rounds=1)
+                    timer.gpu_ms.append(1000.0 * (_t.perf_counter() - t0))
+
+        self._orig = orig_start
+        sim.start = timed_start
+        return self
+
+    def __exit__(self, *exc):
+        self.sim.start = self._orig
+        return False
+
+    @property
+    def mean_ms(self) -> float:
+        return sum(self.gpu_ms) / max(1, len(self.gpu_ms))

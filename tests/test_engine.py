@@ -1534,3 +1534,21 @@ def test_cnn_two_rank_matches_single():
     for p in procs:
         p.join(timeout=60)
     assert np.allclose(ref.local_params().numpy(), got, atol=1e-5)
+
+
+def test_round_timer_records_rounds():
+    from gossipy_amd.engine import RoundTimer
+
+    shards, geval = _make_data(20, seed=1)
+    data = DataArena.from_shards(shards, torch.device("cpu"), global_eval=geval)
+    cfg = EngineConfig(
+        n_nodes=20, delta=5, protocol=AntiEntropyProtocol.PUSH,
+        model_size=116, sampling_eval=0.0, seed=3,
+    )
+    sim = BatchedGossipSimulator(cfg, LogRegSpec(d_in=57, n_classes=2), data)
+    sim.init_nodes()
+    with RoundTimer(sim) as rt:
+        sim.start(n_rounds=3)
+    assert len(rt.gpu_ms) == 3
+    assert rt.mean_ms > 0
+    sim.start(n_rounds=1)  # un-patched after exit
