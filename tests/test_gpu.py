@@ -1036,3 +1036,50 @@ class TestSingleBlockRound:
         b = self._run(spec, cfg, sb=False)
         assert torch.allclose(a.local_params(), b.local_params(), atol=1e-5)
         assert torch.equal(a.state.ages, b.state.ages)
+
+
+class TestObjectLayerGPU:
+    """The reference-parity object layer also runs on the MI355X (its
+    per-update device round-trips are the reference's own 'GPU support'
+    pattern, gossipy/model/handler.py:236-248)."""
+
+    def test_object_layer_simulator_on_gpu(self):
+        from gossipy_amd import GlobalSettings, set_seed
+        from gossipy_amd.core import CreateModelMode, StaticP2PNetwork
+        from gossipy_amd.data import DataDispatcher
+        from gossipy_amd.data.handler import ClassificationDataHandler
+        from gossipy_amd.model.handler import TorchModelHandler
+        from gossipy_amd.model.nn import LogisticRegression
+        from gossipy_amd.node import GossipNode
+        from gossipy_amd.simul import GossipSimulator, SimulationReport
+
+        GlobalSettings().set_device("cuda")
+        try:
+            set_seed(98765)
+            X, y = make_synthetic_classification((460, 57, 2), seed=42, margin=2.0)
+            handler = ClassificationDataHandler(X, y, test_size=0.1, seed=42)
+            dispatcher = DataDispatcher(handler, n=10, eval_on_user=False)
+            nodes = GossipNode.generate(
+                data_dispatcher=dispatcher,
+                p2p_net=StaticP2PNetwork(10),
+                model_proto=TorchModelHandler(
+                    net=LogisticRegression(57, 2),
+                    optimizer=torch.optim.SGD,
+                    optimizer_params={"lr": 0.1},
+                    criterion=torch.nn.CrossEntropyLoss(),
+                    create_model_mode=CreateModelMode.MERGE_UPDATE,
+                ),
+                round_len=5,
+                sync=True,
+            )
+            sim = GossipSimulator(
+                nodes=nodes, data_dispatcher=dispatcher, delta=5,
+                protocol=AntiEntropyProtocol.PUSH, sampling_eval=0.5,
+            )
+            rep = SimulationReport()
+            sim.add_receiver(rep)
+            sim.init_nodes(seed=42)
+            sim.start(n_rounds=5)
+            assert rep.get_evaluation(False)[-1][1]["accuracy"] > 0.8
+        finally:
+            GlobalSettings().set_device("cpu")
