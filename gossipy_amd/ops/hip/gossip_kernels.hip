@@ -591,7 +591,10 @@ struct MFArgs {
 
 DEV_INLINE void mf_update(const MFArgs& a, int node, float* row, int& age)
 {
+    // per-rating chain is order-dependent: wave 0 runs it; other waves of
+    // the (wider) block idle here and rejoin at the caller's barrier
     int lane = threadIdx.x;
+    if (lane >= WAVE) return;
     int c = a.counts[node];
     const float* items = a.X + (long)node * a.Smax;
     const float* ratings = a.Y + (long)node * a.Smax;
@@ -626,15 +629,18 @@ DEV_INLINE void mf_merge(const MFArgs& a, float* row, int age,
         row[a.item_off + e] = w1 * row[a.item_off + e] + w2 * srow[e];
 }
 
-__global__ void __launch_bounds__(WAVE)
+__global__ void __launch_bounds__(256)
 tick_mf_kernel(MFArgs a)
 {
+    // 256 threads: the item-block merge/copy loops (tens of KB per row)
+    // use the full block; the order-dependent rating chain runs on wave 0
     int i = blockIdx.x;
     int node = a.nodes[i];
     float* row = a.params + (long)node * a.D;
     int age = a.ages[node];
     if (a.update_only) {
         mf_update(a, node, row, age);
+        __syncthreads();
     } else {
         for (int j = a.ptr[i]; j < a.ptr[i + 1]; ++j) {
             int slot = a.dslots[j];
@@ -647,6 +653,7 @@ tick_mf_kernel(MFArgs a)
             if (rs >= 0) {
                 for (int e = threadIdx.x; e < a.Wslot; e += blockDim.x)
                     a.slots[(long)rs * a.Wslot + e] = row[a.item_off + e];
+                // thread 0 (wave 0) holds the post-update age
                 if (threadIdx.x == 0) a.slot_ages[rs] = age;
                 __syncthreads();
             }
@@ -2121,7 +2128,7 @@ void tick_mf(torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
     a.k = k; a.n_items = n_items; a.Smax = X.size(1); a.D = params.size(1);
     a.item_off = k + 1; a.Wslot = n_items * (k + 1);
     a.reg = reg; a.lr = lr; a.update_only = update_only;
-    hipLaunchKernelGGL(tick_mf_kernel, dim3(n), dim3(WAVE), 0,
+    hipLaunchKernelGGL(tick_mf_kernel, dim3(n), dim3(256), 0,
                        current_stream(), a);
 }
 
@@ -2490,7 +2497,7 @@ void run_round_mf(
             a.ptr = r.recv_nptr + r0;
             a.dslots = r.del_slots;
             a.rslots = r.reply_slots;
-            hipLaunchKernelGGL(tick_mf_kernel, dim3(r1 - r0), dim3(WAVE), 0,
+            hipLaunchKernelGGL(tick_mf_kernel, dim3(r1 - r0), dim3(256), 0,
                                s, a);
         }
         int p0 = r.pull_tptr[t], p1 = r.pull_tptr[t + 1];
@@ -2509,7 +2516,7 @@ void run_round_mf(
             a.ptr = r.rep_nptr + q0;
             a.dslots = r.rep_slots;
             a.rslots = nullptr;
-            hipLaunchKernelGGL(tick_mf_kernel, dim3(q1 - q0), dim3(WAVE), 0,
+            hipLaunchKernelGGL(tick_mf_kernel, dim3(q1 - q0), dim3(256), 0,
                                s, a);
         }
     }
