@@ -642,6 +642,7 @@ tick_mf_kernel(MFArgs a)
         mf_update(a, node, row, age);
         __syncthreads();
     } else {
+        __shared__ int age_sh;
         for (int j = a.ptr[i]; j < a.ptr[i + 1]; ++j) {
             int slot = a.dslots[j];
             mf_merge(a, row, age, a.slots + (long)slot * a.Wslot,
@@ -649,6 +650,11 @@ tick_mf_kernel(MFArgs a)
             __syncthreads();
             mf_update(a, node, row, age);
             __syncthreads();
+            // the rating chain ran on wave 0 only — rebroadcast the age so
+            // a second delivery's merge weights agree across all waves
+            if (threadIdx.x == 0) age_sh = age;
+            __syncthreads();
+            age = age_sh;
             int rs = a.rslots ? a.rslots[j] : -1;
             if (rs >= 0) {
                 for (int e = threadIdx.x; e < a.Wslot; e += blockDim.x)
