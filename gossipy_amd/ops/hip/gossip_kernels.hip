@@ -98,7 +98,8 @@ struct LogregArgs {
 // Minibatch SGD epochs over the node's shard, on the LDS-resident model W.
 // xb/dz are LDS scratch; age is a wave-uniform register.
 DEV_INLINE void logreg_update(const LogregArgs& a, int node, float* W,
-                              float* xb, float* dz, int& age)
+                              float* xb, float* dz, int& age,
+                              int prestaged = 0)
 {
     int tid = threadIdx.x;
     int c = a.counts[node];
@@ -110,9 +111,14 @@ DEV_INLINE void logreg_update(const LogregArgs& a, int node, float* W,
         for (int s0 = 0; s0 < c; s0 += bsz) {
             int m = min(bsz, c - s0);
             // stage the batch in LDS (coalesced: consecutive threads read
-            // consecutive floats of the shard)
-            for (int e = tid; e < m * a.d; e += blockDim.x)
-                xb[e] = Xn[(long)s0 * a.d + e];
+            // consecutive floats of the shard); the caller may have staged
+            // the very first batch already (fused with the W load)
+            if (prestaged) {
+                prestaged = 0;
+            } else {
+                for (int e = tid; e < m * a.d; e += blockDim.x)
+                    xb[e] = Xn[(long)s0 * a.d + e];
+            }
             __syncthreads();
             // per-sample forward + dLoss/dz (thread = sample)
             if (tid < m) {
@@ -173,6 +179,63 @@ DEV_INLINE void logreg_process_node(const LogregArgs& a, int i)
     float* xb = W2 + a.D;                  // bsmax*d
     int bsmax = (a.bs == 0) ? a.Smax : min(a.bs, a.Smax);
     float* dz = xb + bsmax * a.d;          // bsmax*k
+
+    // Fast path for the common MERGE_UPDATE single-sequence case: fuse the
+    // model load with the FIRST delivery's merge and prefetch the first
+    // minibatch in the same phase — three independent HBM streams behind
+    // one barrier instead of three serial chains.
+    int j0 = a.update_only ? 0 : a.ptr[i];
+    int j1 = a.update_only ? 0 : a.ptr[i + 1];
+    if (!a.update_only && j1 > j0 && a.mode == MODE_MERGE_UPDATE &&
+        !(a.dmodes && a.dmodes[j0])) {
+        int slot = a.dslots[j0];
+        const float* srow = a.slots + (long)slot * a.D;
+        for (int e = tid; e < a.D; e += blockDim.x)
+            W[e] = 0.5f * (a.params[(long)node * a.D + e] + srow[e]);
+        int c = a.counts[node];
+        int m0 = min((a.bs == 0) ? max(c, 1) : a.bs, c);
+        for (int e = tid; e < m0 * a.d; e += blockDim.x)
+            xb[e] = a.X[(long)node * a.Smax * a.d + e];
+        __syncthreads();
+        int age = max(a.ages[node], a.slot_ages[slot]);
+        logreg_update(a, node, W, xb, dz, age, /*prestaged=*/1);
+        int rs0 = a.rslots ? a.rslots[j0] : -1;
+        if (rs0 >= 0) {  // first delivery's PUSH_PULL reply snapshot
+            for (int e = tid; e < a.D; e += blockDim.x)
+                a.slots[(long)rs0 * a.D + e] = W[e];
+            if (tid == 0) a.slot_ages[rs0] = age;
+            __syncthreads();
+        }
+        for (int j = j0 + 1; j < j1; ++j) {
+            int slot2 = a.dslots[j];
+            const float* srow2 = a.slots + (long)slot2 * a.D;
+            int sage2 = a.slot_ages[slot2];
+            int md = (a.dmodes && a.dmodes[j]) ? MODE_PASS : a.mode;
+            if (md == MODE_MERGE_UPDATE) {
+                for (int e = tid; e < a.D; e += blockDim.x)
+                    W[e] = 0.5f * (W[e] + srow2[e]);
+                age = max(age, sage2);
+                __syncthreads();
+                logreg_update(a, node, W, xb, dz, age);
+            } else {  // PASS (pass-through coin resolved to adopt)
+                for (int e = tid; e < a.D; e += blockDim.x) W[e] = srow2[e];
+                age = sage2;
+                __syncthreads();
+            }
+            int rs2 = a.rslots ? a.rslots[j] : -1;
+            if (rs2 >= 0) {
+                for (int e = tid; e < a.D; e += blockDim.x)
+                    a.slots[(long)rs2 * a.D + e] = W[e];
+                if (tid == 0) a.slot_ages[rs2] = age;
+                __syncthreads();
+            }
+        }
+        __syncthreads();
+        for (int e = tid; e < a.D; e += blockDim.x)
+            a.params[(long)node * a.D + e] = W[e];
+        if (tid == 0) a.ages[node] = age;
+        return;
+    }
 
     for (int e = tid; e < a.D; e += blockDim.x)
         W[e] = a.params[(long)node * a.D + e];
