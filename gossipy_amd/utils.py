@@ -85,6 +85,11 @@ def plot_evaluation(evals: List[List[Dict]], title: str = "Untitled plot") -> No
     """
     if not evals or not evals[0] or not evals[0][0]:
         return
+    # accept either the reference's list-of-dicts runs or
+    # SimulationReport.get_evaluation's (round, dict) tuples
+    evals = [
+        [p[1] if isinstance(p, tuple) else p for p in run] for run in evals
+    ]
     import matplotlib.pyplot as plt
 
     fig = plt.figure()
