@@ -244,6 +244,8 @@ def _worker(rank, world, port, q):
             sampling_eval=0.25,
             seed=5,
             delay=UniformDelay(0, 4),
+            drop_prob=0.15,
+            online_prob=0.85,
         )
         spec = LogRegSpec(d_in=57, n_classes=2, lr=0.1)
         sim = BatchedGossipSimulator(cfg, spec, data, device=torch.device("cpu"))
@@ -269,6 +271,8 @@ def test_two_rank_matches_single_rank():
         sampling_eval=0.25,
         seed=5,
         delay=UniformDelay(0, 4),
+        drop_prob=0.15,
+        online_prob=0.85,
     )
     spec = LogRegSpec(d_in=57, n_classes=2, lr=0.1)
     sim = BatchedGossipSimulator(cfg, spec, data, device=torch.device("cpu"))
