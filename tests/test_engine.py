@@ -1556,3 +1556,49 @@ def test_round_timer_records_rounds():
     assert len(rt.gpu_ms) == 3
     assert rt.mean_ms > 0
     sim.start(n_rounds=1)  # un-patched after exit
+
+
+def test_pass_mode_adopts_model():
+    """CreateModelMode.PASS: the receiver adopts the received model
+    verbatim (gossipy/model/handler.py:134-136)."""
+    from gossipy_amd.core import CreateModelMode
+    from gossipy_amd.engine.arena import NodeStateArena, SlotPool
+    from gossipy_amd.engine.backend import TorchBackend
+
+    spec = LogRegSpec(d_in=5, n_classes=2, mode=CreateModelMode.PASS)
+    state = NodeStateArena(2, spec.D, torch.device("cpu"))
+    state.params.normal_(generator=torch.Generator().manual_seed(0))
+    pool = SlotPool(spec.D, torch.device("cpu"), 2)
+    pool.slots.normal_(generator=torch.Generator().manual_seed(1))
+    pool.slot_ages[1] = 7
+    X, y = make_synthetic_classification((10, 5, 2), seed=0)
+    data = DataArena.from_shards([(X[:5], y[:5]), (X[5:], y[5:])], torch.device("cpu"))
+    TorchBackend().deliver(
+        state, pool, data, spec,
+        torch.tensor([0]), torch.tensor([0, 1]), torch.tensor([1]),
+        torch.tensor([-1]),
+    )
+    assert torch.equal(state.params[0], pool.slots[1])
+    assert int(state.ages[0]) == 7
+
+
+def test_custom_delay_subclass_falls_back_to_python_scheduler():
+    from gossipy_amd.core import Delay
+
+    class JitterDelay(Delay):
+        def get(self, msg):
+            return 2
+
+    shards, geval = _make_data(20, seed=1)
+    data = DataArena.from_shards(shards, torch.device("cpu"), global_eval=geval)
+    cfg = EngineConfig(
+        n_nodes=20, delta=10, protocol=AntiEntropyProtocol.PUSH,
+        model_size=116, sampling_eval=0.0, seed=3, delay=JitterDelay(),
+    )
+    sim = BatchedGossipSimulator(cfg, LogRegSpec(d_in=57, n_classes=2), data)
+    from gossipy_amd.engine import Scheduler
+
+    assert type(sim.scheduler) is Scheduler  # python fallback
+    sim.init_nodes()
+    sim.start(n_rounds=3)
+    assert torch.isfinite(sim.local_params()).all()
