@@ -24,7 +24,6 @@ from .core import AntiEntropyProtocol, CreateModelMode, Message, MessageType, P2
 from .data import DataDispatcher
 from .model.handler import ModelHandler, PartitionedTMH, SamplingTMH, WeightedTMH
 from .model.sampling import TorchModelSampling
-from .utils import choice_not_n
 
 __all__ = [
     "GossipNode",

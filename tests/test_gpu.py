@@ -87,6 +87,7 @@ class TestKernelNumerics:
         CreateModelMode.MERGE_UPDATE,
         CreateModelMode.UPDATE,
         CreateModelMode.UPDATE_MERGE,
+        CreateModelMode.PASS,
     ])
     def test_logreg_deliver_matches_oracle(self, mode):
         spec = LogRegSpec(d_in=57, n_classes=2, lr=0.1, mode=mode)
