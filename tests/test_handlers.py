@@ -444,3 +444,30 @@ class TestSamplingTMH:
         sample = TorchModelSampling.sample(0.5, h2.model)
         h1(h2, (X, y), sample)
         assert h1.n_updates > 0
+
+
+def test_perceptron_and_linear_regression_models():
+    """Model-zoo parity: TorchPerceptron (gossipy/model/nn.py:26-64) and
+    LinearRegression (:176-198) forward shapes + trainability."""
+    import torch
+
+    from gossipy_amd.model.nn import LinearRegression, TorchPerceptron
+
+    p = TorchPerceptron(8)
+    p.init_weights()
+    out = p(torch.randn(5, 8))
+    assert out.shape == (5, 1)
+    assert (out >= 0).all() and (out <= 1).all()  # sigmoid head
+
+    lr = LinearRegression(8, 1)
+    x = torch.randn(64, 8)
+    w = torch.randn(8, 1)
+    y = x @ w
+    opt = torch.optim.SGD(lr.parameters(), lr=0.05)
+    for _ in range(300):
+        opt.zero_grad()
+        loss = torch.nn.functional.mse_loss(lr(x), y)
+        loss.backward()
+        opt.step()
+    assert loss.item() < 0.05
+    assert lr.get_size() == 9
