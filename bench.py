@@ -135,7 +135,7 @@ def main() -> None:
             "scaling": "weak",
             "vs_baseline": round(value / BASELINE_NODE_ROUNDS_PER_SEC, 2),
             "dtype": "fp32",
-            "data": "synthetic spambase-shaped (4601x57 per 1000 nodes), random-init weights",
+            "data": f"synthetic spambase-shaped ({SAMPLES_PER_SHARD_SET}x{D_IN} per {args.nodes_per_gpu} nodes/GPU), random-init weights",
             "config": {
                 "model": "logreg-57x2 (Hegedus-2021)",
                 "global_batch": None,
