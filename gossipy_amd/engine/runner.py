@@ -172,42 +172,68 @@ class BatchedGossipSimulator(SimulationEventSender):
 
     def _exchange_start(self, needed):
         """Issue the grouped P2P ops; returns the in-flight state for
-        :meth:`_exchange_finish` (None when there is nothing to move)."""
+        :meth:`_exchange_finish` (None when there is nothing to move).
+
+        All of this rank's outgoing rows gather into ONE staging buffer
+        (one fancy-index per tick, per-destination views feed the isends);
+        incoming rows land in one buffer scattered back with one
+        index-copy — at 8 GPUs this replaces up to 7+7 small gathers per
+        tick with 2."""
         if self.world == 1 or not needed:
             return None
-        ops = []
-        recv_bufs = []
-        send_bufs = []  # keep alive until the waits complete
         D = self.pool.slots.shape[1]
         A = getattr(self.spec, "age_width", 1)
+        send_parts = []  # (dst, count)
+        send_ids = []
+        recv_parts = []  # (src, count)
+        recv_ids = []
         for src, dst, slot_ids in needed:
             if src == dst:
                 continue
             if src == self.rank:
-                ids = torch.from_numpy(slot_ids.astype(np.int64)).to(self.device)
-                buf = torch.empty(len(slot_ids), D + A, device=self.device)
-                buf[:, :D] = self.pool.slots[ids]
-                buf[:, D:] = self.pool.slot_ages[ids].reshape(len(ids), A).float()
-                ops.append(dist.P2POp(dist.isend, buf, dst))
-                send_bufs.append(buf)
+                send_parts.append((dst, len(slot_ids)))
+                send_ids.append(slot_ids)
             elif dst == self.rank:
-                buf = torch.empty(len(slot_ids), D + A, device=self.device)
-                ops.append(dist.P2POp(dist.irecv, buf, src))
-                recv_bufs.append((slot_ids, buf))
+                recv_parts.append((src, len(slot_ids)))
+                recv_ids.append(slot_ids)
+        ops = []
+        send_buf = recv_buf = None
+        if send_ids:
+            ids = torch.from_numpy(
+                np.concatenate(send_ids).astype(np.int64)
+            ).to(self.device)
+            send_buf = torch.empty(len(ids), D + A, device=self.device)
+            send_buf[:, :D] = self.pool.slots[ids]
+            send_buf[:, D:] = self.pool.slot_ages[ids].reshape(len(ids), A).float()
+            off = 0
+            for dst, n in send_parts:
+                ops.append(dist.P2POp(dist.isend, send_buf[off : off + n], dst))
+                off += n
+        if recv_ids:
+            total = sum(n for _, n in recv_parts)
+            recv_buf = torch.empty(total, D + A, device=self.device)
+            off = 0
+            for src, n in recv_parts:
+                ops.append(dist.P2POp(dist.irecv, recv_buf[off : off + n], src))
+                off += n
         works = dist.batch_isend_irecv(ops) if ops else []
-        return (works, recv_bufs, send_bufs, D)
+        return (works, recv_ids, recv_buf, send_buf, D)
 
     def _exchange_finish(self, pending) -> None:
         if pending is None:
             return
-        works, recv_bufs, _send_bufs, D = pending
+        works, recv_ids, recv_buf, _send_buf, D = pending
         for w in works:
             w.wait()
-        for slot_ids, buf in recv_bufs:
-            ids = torch.from_numpy(slot_ids.astype(np.int64)).to(self.device)
-            self.pool.slots[ids] = buf[:, :D]
-            ages = buf[:, D:].int()
-            self.pool.slot_ages[ids] = ages.reshape(self.pool.slot_ages[ids].shape)
+        if recv_buf is not None:
+            ids = torch.from_numpy(
+                np.concatenate(recv_ids).astype(np.int64)
+            ).to(self.device)
+            self.pool.slots[ids] = recv_buf[:, :D]
+            ages = recv_buf[:, D:].int()
+            self.pool.slot_ages[ids] = ages.reshape(
+                self.pool.slot_ages[ids].shape
+            )
 
     def _plan_exchange(
         self,
