@@ -127,6 +127,7 @@ def _tok_cfgs():
     yield EngineConfig(
         protocol=AntiEntropyProtocol.PUSH_PULL, delay=UniformDelay(0, 4), **base
     ), "randomized"
+    yield EngineConfig(protocol=AntiEntropyProtocol.PULL, **base), "randomized"
     yield EngineConfig(protocol=AntiEntropyProtocol.PUSH, **base), "proactive"
     yield EngineConfig(protocol=AntiEntropyProtocol.PUSH, **base), "simple"
     yield EngineConfig(protocol=AntiEntropyProtocol.PUSH, **base), "generalized"
@@ -148,7 +149,7 @@ def _account(name):
     }[name]()
 
 
-@pytest.mark.parametrize("cfg_i", range(7))
+@pytest.mark.parametrize("cfg_i", range(8))
 def test_native_tokenized_matches_python(cfg_i):
     from gossipy_amd.engine.runner import BatchedGossipSimulator
     from gossipy_amd.engine.schedule import (
