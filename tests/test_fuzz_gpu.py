@@ -38,6 +38,7 @@ def _rand_case(rng):
         CreateModelMode.MERGE_UPDATE,
         CreateModelMode.UPDATE,
         CreateModelMode.UPDATE_MERGE,
+        CreateModelMode.PASS,
     ])
     n_nodes = int(rng.integers(8, 48))
     kw = dict(
@@ -52,6 +53,8 @@ def _rand_case(rng):
         seed=int(rng.integers(0, 10**6)),
     )
     pm1 = False
+    if fam in ("logreg_part", "logreg_samp") and mode == CreateModelMode.PASS:
+        mode = CreateModelMode.MERGE_UPDATE  # PASS rejected by those handlers
     if fam == "logreg":
         spec = LogRegSpec(
             d_in=d, n_classes=k, lr=0.1, mode=mode,
@@ -102,7 +105,7 @@ def _run(cfg, spec, k, pm1, device):
     return sim
 
 
-@pytest.mark.parametrize("case", range(24))
+@pytest.mark.parametrize("case", range(40))
 def test_fuzz_hip_matches_oracle(case):
     rng = np.random.default_rng(1000 + case)
     cfg, spec, k, pm1 = _rand_case(rng)
