@@ -398,6 +398,8 @@ class BatchedGossipSimulator(SimulationEventSender):
                 groups[name] = {
                     "nodes": _dev32(recv_nodes[rows] - lo),
                     "ptr": _dev32(abs_ptr),
+                    # host mirror: per-tick slicing must not sync the GPU
+                    "ptr_host": abs_ptr,
                     "slots": _dev32(dslots[sel]) if len(sel) else _dev32([]),
                     "reply": _dev32(np.asarray(f[reply_key])[sel])
                     if reply_key
@@ -422,8 +424,8 @@ class BatchedGossipSimulator(SimulationEventSender):
             a, b = int(g["row_tptr"][t]), int(g["row_tptr"][t + 1])
             if a == b:
                 return
-            lo_d = int(g["ptr"][a])
-            hi_d = int(g["ptr"][b])
+            lo_d = int(g["ptr_host"][a])
+            hi_d = int(g["ptr_host"][b])
             reply = (
                 g["reply"][lo_d:hi_d]
                 if g["reply"] is not None
