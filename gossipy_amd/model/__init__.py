@@ -39,3 +39,8 @@ class TorchModel(torch.nn.Module, Sizeable, ABC):
 
     def __str__(self) -> str:
         return "%s(size=%d)" % (self.__class__.__name__, self.get_size())
+
+    def _get_n_params(self) -> int:
+        """Reference-private alias (gossipy/model/__init__.py): parameter
+        count — same value as :meth:`get_size`."""
+        return sum(p.numel() for p in self.parameters())

@@ -155,6 +155,12 @@ class TorchModelPartition:
             parts[p] = per_layer
         return parts
 
+    def _partition(self, net_size: int):
+        """Reference-private alias (gossipy/model/sampling.py:144-198):
+        builds the per-partition index cover — same output as the
+        range-based :meth:`_build`."""
+        return self._build(net_size)
+
     def merge(
         self,
         id_part: int,
