@@ -115,3 +115,75 @@ def test_fuzz_hip_matches_oracle(case):
         g.local_params().cpu(), c.local_params(), atol=2e-3, rtol=2e-3
     ), f"params diverge: {cfg} {spec}"
     assert torch.equal(g.state.ages.cpu(), c.state.ages), f"ages: {cfg} {spec}"
+
+
+def _rand_runner_case(rng):
+    """Random protocol-runner config (tokenized / cacheneigh / all2all /
+    pens) over logreg."""
+    kind = rng.choice(["tokenized", "cacheneigh", "all2all", "pens"])
+    n_nodes = int(rng.integers(8, 32))
+    d = int(rng.integers(8, 60))
+    proto = AntiEntropyProtocol.PUSH
+    if kind == "cacheneigh" and rng.random() < 0.5:
+        proto = AntiEntropyProtocol.PUSH_PULL
+    kw = dict(
+        n_nodes=n_nodes,
+        delta=int(rng.integers(3, 9)),
+        protocol=proto,
+        model_size=d,
+        drop_prob=float(rng.choice([0.0, 0.2])),
+        online_prob=float(rng.choice([1.0, 0.85])),
+        sync=bool(rng.random() < 0.7),
+        sampling_eval=0.0,
+        seed=int(rng.integers(0, 10**6)),
+    )
+    spec = LogRegSpec(d_in=d, n_classes=2, lr=0.1)
+    return kind, EngineConfig(**kw), spec
+
+
+def _run_runner(kind, cfg, spec, device):
+    from gossipy_amd.engine import (
+        BatchedAll2AllGossipSimulator,
+        BatchedCacheNeighGossipSimulator,
+        BatchedPENSGossipSimulator,
+        BatchedTokenizedGossipSimulator,
+    )
+    from gossipy_amd.flow_control import RandomizedTokenAccount
+
+    n_samp = cfg.n_nodes * 8
+    X, y = make_synthetic_classification((n_samp, spec.d_in, 2), seed=cfg.seed)
+    shards = [
+        (X[s], y[s]) for s in np.array_split(np.arange(n_samp), cfg.n_nodes)
+    ]
+    data = DataArena.from_shards(shards, device, global_eval=(X, y))
+    if kind == "tokenized":
+        sim = BatchedTokenizedGossipSimulator(
+            cfg, spec, data, token_account=RandomizedTokenAccount(C=10, A=4),
+            device=device,
+        )
+    elif kind == "cacheneigh":
+        sim = BatchedCacheNeighGossipSimulator(cfg, spec, data, device=device)
+    elif kind == "all2all":
+        sim = BatchedAll2AllGossipSimulator(cfg, spec, data, device=device)
+    else:
+        sim = BatchedPENSGossipSimulator(
+            cfg, spec, data, n_sampled=3, m_top=1, step1_rounds=2,
+            device=device,
+        )
+    sim.init_nodes()
+    sim.start(n_rounds=4)
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    return sim
+
+
+@pytest.mark.parametrize("case", range(16))
+def test_fuzz_protocol_runners(case):
+    rng = np.random.default_rng(70_000 + case)
+    kind, cfg, spec = _rand_runner_case(rng)
+    g = _run_runner(kind, cfg, spec, CUDA)
+    c = _run_runner(kind, cfg, spec, CPU)
+    assert torch.allclose(
+        g.local_params().cpu(), c.local_params(), atol=2e-3, rtol=2e-3
+    ), f"{kind}: {cfg}"
+    assert torch.equal(g.state.ages.cpu(), c.state.ages), f"{kind}: {cfg}"
