@@ -24,6 +24,7 @@
 #include <cmath>
 #include <cstdint>
 #include <map>
+#include <deque>
 #include <unordered_map>
 #include <vector>
 
@@ -95,6 +96,97 @@ struct Msg {  // matches the python pending tuple
     int32_t pid;        // partition id (-1 when n_parts == 0)
 };
 
+
+// ---------------------------------------------------------------------------
+// Launch-group merging (mirrors BatchedGossipSimulator._merge_flat_groups):
+// fuse conflict-free adjacent ticks into single launch groups. Returns the
+// tick-index boundaries [0, ..., delta]. Epoch-stamped membership arrays give
+// O(total events) with no per-group clearing.
+// ---------------------------------------------------------------------------
+static std::vector<int32_t> compute_merge_bounds(
+    int64_t n_nodes, int64_t n_slots,
+    const std::vector<int32_t>& snap_nodes, const std::vector<int32_t>& snap_slots,
+    const std::vector<int32_t>& snap_tptr,
+    const std::vector<int32_t>& recv_nodes, const std::vector<int32_t>& recv_nptr,
+    const std::vector<int32_t>& recv_tptr,
+    const std::vector<int32_t>& del_slots, const std::vector<int32_t>& reply_slots,
+    const std::vector<int32_t>& pull_nodes, const std::vector<int32_t>& pull_slots,
+    const std::vector<int32_t>& pull_tptr,
+    const std::vector<int32_t>& rep_nodes, const std::vector<int32_t>& rep_nptr,
+    const std::vector<int32_t>& rep_tptr, const std::vector<int32_t>& rep_slots)
+{
+    int64_t delta = (int64_t)snap_tptr.size() - 1;
+    std::vector<int32_t> bounds{0};
+    if (delta <= 1) {
+        if (delta == 1) bounds.push_back(1);
+        return bounds;
+    }
+    std::vector<int32_t> node_mut(n_nodes, -1), node_pull(n_nodes, -1);
+    std::vector<int32_t> slot_read(n_slots, -1), slot_write(n_slots, -1),
+        slot_late(n_slots, -1);
+    int32_t gid = 0;
+    auto tick_ok = [&](int64_t t) -> bool {
+        // candidate tick's reads/mutations vs the group's accumulated sets
+        for (int32_t i = snap_tptr[t]; i < snap_tptr[t + 1]; ++i) {
+            if (node_mut[snap_nodes[i]] == gid) return false;        // touched & g_mut
+            if (slot_read[snap_slots[i]] == gid) return false;       // writes & g_reads
+            if (slot_write[snap_slots[i]] == gid) return false;      // writes & g_writes
+        }
+        for (int32_t i = pull_tptr[t]; i < pull_tptr[t + 1]; ++i) {
+            if (node_mut[pull_nodes[i]] == gid) return false;
+            if (slot_read[pull_slots[i]] == gid) return false;
+            if (slot_write[pull_slots[i]] == gid) return false;
+        }
+        for (int32_t r = recv_tptr[t]; r < recv_tptr[t + 1]; ++r) {
+            int32_t x = recv_nodes[r];
+            if (node_mut[x] == gid) return false;                    // touched & g_mut
+            if (node_pull[x] == gid) return false;                   // recv & g_pulls
+            for (int32_t d = recv_nptr[r]; d < recv_nptr[r + 1]; ++d) {
+                if (slot_late[del_slots[d]] == gid) return false;    // dreads & g_late
+                int32_t rw = reply_slots[d];
+                if (rw >= 0 && (slot_read[rw] == gid || slot_write[rw] == gid))
+                    return false;                                    // writes & ...
+            }
+        }
+        for (int32_t r = rep_tptr[t]; r < rep_tptr[t + 1]; ++r) {
+            if (node_mut[rep_nodes[r]] == gid) return false;
+        }
+        return true;
+    };
+    auto tick_add = [&](int64_t t) {
+        for (int32_t i = snap_tptr[t]; i < snap_tptr[t + 1]; ++i)
+            slot_write[snap_slots[i]] = gid;
+        for (int32_t i = pull_tptr[t]; i < pull_tptr[t + 1]; ++i) {
+            node_pull[pull_nodes[i]] = gid;
+            slot_write[pull_slots[i]] = gid;
+            slot_late[pull_slots[i]] = gid;
+        }
+        for (int32_t r = recv_tptr[t]; r < recv_tptr[t + 1]; ++r) {
+            node_mut[recv_nodes[r]] = gid;
+            for (int32_t d = recv_nptr[r]; d < recv_nptr[r + 1]; ++d) {
+                slot_read[del_slots[d]] = gid;
+                int32_t rw = reply_slots[d];
+                if (rw >= 0) { slot_write[rw] = gid; slot_late[rw] = gid; }
+            }
+        }
+        for (int32_t r = rep_tptr[t]; r < rep_tptr[t + 1]; ++r) {
+            node_mut[rep_nodes[r]] = gid;
+            for (int32_t d = rep_nptr[r]; d < rep_nptr[r + 1]; ++d)
+                slot_read[rep_slots[d]] = gid;
+        }
+    };
+    tick_add(0);
+    for (int64_t t = 1; t < delta; ++t) {
+        if (!tick_ok(t)) {
+            bounds.push_back((int32_t)t);
+            ++gid;
+        }
+        tick_add(t);
+    }
+    bounds.push_back((int32_t)delta);
+    return bounds;
+}
+
 class NativeScheduler {
 public:
     NativeScheduler(int64_t n_nodes, int64_t delta, int proto,
@@ -154,16 +246,21 @@ protected:
     std::vector<std::vector<int32_t>> fire_buckets_;
     std::vector<int64_t> indptr_, indices_;
     std::unordered_map<int64_t, std::vector<Msg>> pending_;
-    std::vector<int32_t> free_slots_;
+    // consumed slots re-enter circulation only kSlotReuseLag ticks after
+    // the tick that freed them (FIFO), mirroring Scheduler.SLOT_REUSE_LAG:
+    // adjacent ticks never alias slot ids, so the runner can fuse
+    // conflict-free ticks into single launch groups.
+    static constexpr int64_t kSlotReuseLag = 16;
+    std::deque<std::pair<int64_t, int32_t>> reuse_q_;
     int64_t next_slot_ = 0;
     std::vector<int32_t> slot_owner_;
 
-    int32_t alloc_slot(int32_t owner)
+    int32_t alloc_slot(int32_t owner, int64_t t)
     {
         int32_t s;
-        if (!free_slots_.empty()) {
-            s = free_slots_.back();
-            free_slots_.pop_back();
+        if (!reuse_q_.empty() && reuse_q_.front().first + kSlotReuseLag <= t) {
+            s = reuse_q_.front().second;
+            reuse_q_.pop_front();
         } else {
             s = (int32_t)next_slot_++;
             if (next_slot_ > (int64_t)slot_owner_.size())
@@ -296,7 +393,7 @@ py::dict NativeScheduler::next_round(int64_t r)
                 bool is_pull = proto_ == PULL;
                 int32_t slot = -1;
                 if (!is_pull) {
-                    slot = alloc_slot(sender);
+                    slot = alloc_slot(sender, t);
                     snap_nodes.push_back(sender);
                     snap_slots.push_back(slot);
                 }
@@ -339,7 +436,7 @@ py::dict NativeScheduler::next_round(int64_t r)
                 continue;
             }
             if (m.is_pull) {
-                int32_t rslot = alloc_slot(m.receiver);
+                int32_t rslot = alloc_slot(m.receiver, t);
                 pull_nodes.push_back(m.receiver);
                 pull_slots.push_back(rslot);
                 if (!enqueue_reply(t, m.receiver, m.sender, rslot,
@@ -350,7 +447,7 @@ py::dict NativeScheduler::next_round(int64_t r)
             }
             int32_t rslot = -1;
             if (m.reply_flag == -2) {
-                rslot = alloc_slot(m.receiver);
+                rslot = alloc_slot(m.receiver, t);
                 if (!enqueue_reply(t, m.receiver, m.sender, rslot,
                                    reply_pid(t, m.receiver), sent, failed,
                                    total_size))
@@ -415,7 +512,7 @@ py::dict NativeScheduler::next_round(int64_t r)
         }
         rep_tptr.push_back((int32_t)rep_nodes.size());
 
-        for (int32_t s : freed) free_slots_.push_back(s);
+        for (int32_t s : freed) reuse_q_.emplace_back(t, s);
     }
 
     py::dict out;
@@ -447,6 +544,13 @@ py::dict NativeScheduler::next_round(int64_t r)
     out["failed"] = failed;
     out["total_size"] = total_size;
     out["n_slots"] = next_slot_;
+    {
+        auto mb = compute_merge_bounds(
+            n_, next_slot_, snap_nodes, snap_slots, snap_tptr, recv_nodes,
+            recv_nptr, recv_tptr, del_slots, reply_slots, pull_nodes,
+            pull_slots, pull_tptr, rep_nodes, rep_nptr, rep_tptr, rep_slots);
+        out["merge_bounds"] = arr(mb);
+    }
     if (sampling_eval_ > 0) {
         Stream g(stream_key(seed_, EVAL, (uint64_t)(t1 - 1), 0));
         int64_t k = std::max<int64_t>((int64_t)(n_ * sampling_eval_), 1);
@@ -688,7 +792,7 @@ py::dict NativeTokenizedScheduler::next_round(int64_t r)
                     bool is_pull = proto_ == PULL;
                     int32_t slot = -1;
                     if (!is_pull) {
-                        slot = alloc_slot(sender);
+                        slot = alloc_slot(sender, t);
                         wave_snap_nodes.push_back(sender);
                         wave_snap_slots.push_back(slot);
                     }
@@ -755,7 +859,7 @@ py::dict NativeTokenizedScheduler::next_round(int64_t r)
                     continue;
                 }
                 if (m.is_pull) {
-                    int32_t rslot = alloc_slot(m.receiver);
+                    int32_t rslot = alloc_slot(m.receiver, t);
                     pull_nodes.push_back(m.receiver);
                     pull_slots.push_back(rslot);
                     if (!enqueue_reply(t, m.receiver, m.sender, rslot,
@@ -766,7 +870,7 @@ py::dict NativeTokenizedScheduler::next_round(int64_t r)
                 }
                 int32_t rslot = -1;
                 if (m.reply_flag == -2) {
-                    rslot = alloc_slot(m.receiver);
+                    rslot = alloc_slot(m.receiver, t);
                     if (!enqueue_reply(t, m.receiver, m.sender, rslot,
                                        reply_extra(t, m.receiver), sent,
                                        failed, total_size))
@@ -793,7 +897,7 @@ py::dict NativeTokenizedScheduler::next_round(int64_t r)
                     Stream& gdl = tick_stream(DELAY, m.receiver);
                     for (int64_t b = 0; b < reaction; ++b) {
                         int32_t peer = burst_peer(m.receiver, gp);
-                        int32_t bslot = alloc_slot(m.receiver);
+                        int32_t bslot = alloc_slot(m.receiver, t);
                         wave_snap_nodes.push_back(m.receiver);
                         wave_snap_slots.push_back(bslot);
                         int32_t bpid = -1;
@@ -873,7 +977,7 @@ py::dict NativeTokenizedScheduler::next_round(int64_t r)
             close_group();
         }
 
-        for (int32_t sfree : freed) free_slots_.push_back(sfree);
+        for (int32_t sfree : freed) reuse_q_.emplace_back(t, sfree);
     }
 
     py::dict out;
@@ -905,6 +1009,13 @@ py::dict NativeTokenizedScheduler::next_round(int64_t r)
     out["failed"] = failed;
     out["total_size"] = total_size;
     out["n_slots"] = next_slot_;
+    {
+        auto mb = compute_merge_bounds(
+            n_, next_slot_, snap_nodes, snap_slots, snap_tptr, recv_nodes,
+            recv_nptr, recv_tptr, del_slots, reply_slots, pull_nodes,
+            pull_slots, pull_tptr, rep_nodes, rep_nptr, rep_tptr, rep_slots);
+        out["merge_bounds"] = arr(mb);
+    }
     if (sampling_eval_ > 0) {
         Stream g(stream_key(seed_, EVAL, (uint64_t)(t1 - 1), 0));
         int64_t k = std::max<int64_t>((int64_t)(n_ * sampling_eval_), 1);

@@ -727,6 +727,127 @@ class BatchedGossipSimulator(SimulationEventSender):
                 )
         return out
 
+    def _maybe_merge(self, f: dict) -> dict:
+        """Apply launch-group merging unless disabled (``GOSSIPY_NO_MERGE=1``).
+        Tracks a cumulative (ticks, groups) counter for the perf probes."""
+        if os.environ.get("GOSSIPY_NO_MERGE") == "1":
+            return f
+        mb = f.get("merge_bounds")
+        if mb is not None and len(mb) - 1 < len(f["snap_tptr"]) - 1:
+            # native scheduler pre-computed the group boundaries in C++
+            out = dict(f)
+            for k in ("snap_tptr", "recv_tptr", "pull_tptr", "rep_tptr"):
+                out[k] = f[k][mb]
+        elif mb is not None:
+            out = f
+        else:
+            out = self._merge_flat_groups(f)
+        t = getattr(self, "merge_stats", (0, 0))
+        self.merge_stats = (
+            t[0] + len(f["snap_tptr"]) - 1,
+            t[1] + len(out["snap_tptr"]) - 1,
+        )
+        return out
+
+    @staticmethod
+    def _merge_flat_groups(f: dict) -> dict:
+        """Merge conflict-free adjacent ticks into single launch groups
+        (SURVEY.md §7 hard-part 2: "batch independent tick-groups").
+
+        Two ticks can share one snapshot/deliver/pull/reply launch
+        quadruple iff no ordering hazard exists between them:
+
+        * node hazard — a later tick may not touch (snapshot, receive)
+          any node the group already MUTATES (deliver/reply receivers),
+          and may not mutate a node the group touches;
+        * slot hazards — a later tick's written slots (snap/pull/reply)
+          must not alias anything the group reads or writes (recycled slot
+          ids!), and its read slots must not alias slots the group writes
+          in the deliver/pull launches (reply and pull writes land after /
+          inside launches a merged read could precede).
+
+        Only the four tick-pointer arrays change; the event arrays and the
+        global CSRs are untouched, so every round executor consumes the
+        merged schedule unmodified.
+        """
+        delta = len(f["snap_tptr"]) - 1
+        if delta <= 1:
+            return f
+        st, rt = np.asarray(f["snap_tptr"]), np.asarray(f["recv_tptr"])
+        pt, qt = np.asarray(f["pull_tptr"]), np.asarray(f["rep_tptr"])
+        nptr, rep_nptr = np.asarray(f["recv_nptr"]), np.asarray(f["rep_nptr"])
+        recv_nodes = np.asarray(f["recv_nodes"])
+        rep_nodes = np.asarray(f["rep_nodes"])
+        snap_nodes = np.asarray(f["snap_nodes"])
+        pull_nodes = np.asarray(f["pull_nodes"])
+        snap_slots = np.asarray(f["snap_slots"])
+        pull_slots = np.asarray(f["pull_slots"])
+        del_slots = np.asarray(f["del_slots"])
+        rep_slots = np.asarray(f["rep_slots"])
+        reply_slots = np.asarray(f["reply_slots"])
+
+        def tick_sets(t):
+            r0, r1 = rt[t], rt[t + 1]
+            q0, q1 = qt[t], qt[t + 1]
+            d0, d1 = nptr[r0], nptr[r1]
+            e0, e1 = rep_nptr[q0], rep_nptr[q1]
+            recv = set(recv_nodes[r0:r1])
+            mut = recv | set(rep_nodes[q0:q1])
+            pulls = set(pull_nodes[pt[t] : pt[t + 1]])
+            touched = mut | pulls | set(snap_nodes[st[t] : st[t + 1]])
+            dreads = set(del_slots[d0:d1])
+            reads = dreads | set(rep_slots[e0:e1])
+            rw = set(x for x in reply_slots[d0:d1] if x >= 0)
+            pw = set(pull_slots[pt[t] : pt[t + 1]])
+            writes = set(snap_slots[st[t] : st[t + 1]]) | pw | rw
+            # reply/pull writes land inside or after the deliver launch, so a
+            # merged tick's *deliver* reads must not alias them (its reply
+            # reads run in the final launch and are ordered fine)
+            late_writes = rw | pw
+            return recv, mut, pulls, touched, dreads, reads, writes, late_writes
+
+        bounds = [0]
+        (g_recv, g_mut, g_pulls, g_touched,
+         g_dreads, g_reads, g_writes, g_late) = tick_sets(0)
+        for t in range(1, delta):
+            recv, mut, pulls, touched, dreads, reads, writes, late = tick_sets(t)
+            ok = (
+                # later tick may not read/mutate a node the group mutates
+                not (touched & g_mut)
+                # deliver launch precedes the pull-snapshot launch: a merged
+                # delivery may not hit a node the group pull-snapshots
+                and not (recv & g_pulls)
+                # recycled slot ids: new writes may not alias live reads/writes
+                and not (writes & (g_reads | g_writes))
+                # merged deliver reads vs the group's late (reply/pull) writes
+                and not (dreads & g_late)
+            )
+            if ok:
+                g_recv |= recv
+                g_mut |= mut
+                g_pulls |= pulls
+                g_touched |= touched
+                g_dreads |= dreads
+                g_reads |= reads
+                g_writes |= writes
+                g_late |= late
+            else:
+                bounds.append(t)
+                (g_recv, g_mut, g_pulls, g_touched,
+                 g_dreads, g_reads, g_writes, g_late) = (
+                    recv, mut, pulls, touched, dreads, reads, writes, late,
+                )
+        bounds.append(delta)
+        if len(bounds) - 1 == delta:
+            return f  # nothing merged
+        sel = np.asarray(bounds)
+        out = dict(f)
+        out["snap_tptr"] = st[sel]
+        out["recv_tptr"] = rt[sel]
+        out["pull_tptr"] = pt[sel]
+        out["rep_tptr"] = qt[sel]
+        return out
+
     def _run_round_fast(self, f: dict) -> None:
         """Upload the round's flat event arrays in one H2D copy and hand
         the whole round to the C++ executor (ops/hip round executors)."""
@@ -946,14 +1067,16 @@ class BatchedGossipSimulator(SimulationEventSender):
                     _, sched, flat = pre
                 else:
                     sched = self.scheduler.next_round_flat(r)
-                    flat = self.scheduler.last_flat
+                    flat = self._maybe_merge(self.scheduler.last_flat)
                 self._prefetched = None
                 self.pool.ensure(sched.n_slots)
                 self._run_round_fast(flat)
-                # overlap: derive round r+1's schedule on the host while the
-                # GPU executes round r's launches
+                # overlap: derive round r+1's schedule (and its launch-group
+                # merge) on the host while the GPU executes round r's launches
                 s2 = self.scheduler.next_round_flat(r + 1)
-                self._prefetched = (r + 1, s2, self.scheduler.last_flat)
+                self._prefetched = (
+                    r + 1, s2, self._maybe_merge(self.scheduler.last_flat)
+                )
             else:
                 sched = self.scheduler.next_round(r)
                 self.pool.ensure(sched.n_slots)
@@ -975,7 +1098,9 @@ class BatchedGossipSimulator(SimulationEventSender):
                 ):
                     # python-scheduled round through the C++ round executor
                     # (tokenized / cache-neigh / python-scheduler fallback)
-                    self._run_round_fast(self._flatten_phases(sched.ticks))
+                    self._run_round_fast(
+                        self._maybe_merge(self._flatten_phases(sched.ticks))
+                    )
                 else:
                     for phase in sched.ticks:
                         self._run_tick(phase)

@@ -29,6 +29,7 @@ is a gather/`ncclSend` per (src GPU, dst GPU) pair.
 from __future__ import annotations
 
 from dataclasses import dataclass, field
+from collections import deque
 from typing import Dict, List, Optional, Tuple
 
 import numpy as np
@@ -199,13 +200,22 @@ class Scheduler:
         # except slots of messages lost to drops/offline receivers are
         # reclaimed instead of leaked (the reference leaks them,
         # SURVEY.md §5 'failure detection').
-        self._free_slots: List[int] = []
+        self._reuse_q: deque = deque()  # (free_tick, slot), tick-ordered
         self._next_slot = 0
         self.slot_owner = np.zeros(64, dtype=np.int32)
 
-    def _alloc_slot(self, owner: int) -> int:
-        if self._free_slots:
-            s = self._free_slots.pop()
+    #: a consumed slot id is only reused SLOT_REUSE_LAG ticks after the tick
+    #: that freed it. Adjacent ticks therefore never alias slot ids, which is
+    #: what lets BatchedGossipSimulator._merge_flat_groups fuse conflict-free
+    #: ticks into single launch groups (the old LIFO free list re-issued a
+    #: slot on the very next tick, making every tick pair conflict). Costs a
+    #: few hundred extra pool rows at the flagship scale.
+    SLOT_REUSE_LAG = 16
+
+    def _alloc_slot(self, owner: int, t: int) -> int:
+        q = self._reuse_q
+        if q and q[0][0] + self.SLOT_REUSE_LAG <= t:
+            s = q.popleft()[1]
         else:
             s = self._next_slot
             self._next_slot += 1
@@ -213,6 +223,9 @@ class Scheduler:
                 self.slot_owner = np.resize(self.slot_owner, 2 * len(self.slot_owner))
         self.slot_owner[s] = owner
         return s
+
+    def _recycle(self, t: int, freed) -> None:
+        self._reuse_q.extend((t, s) for s in freed)
 
     # -- internals -----------------------------------------------------------
 
@@ -294,7 +307,7 @@ class Scheduler:
                     is_pull = proto == AntiEntropyProtocol.PULL
                     slot = -1
                     if not is_pull:
-                        slot = self._alloc_slot(sender)
+                        slot = self._alloc_slot(sender, t)
                         snap_nodes.append(sender)
                         snap_slots.append(slot)
                     sent += 1
@@ -340,7 +353,7 @@ class Scheduler:
                 if is_pull:
                     # PULL request: receiver snapshots and replies (with a
                     # fresh partition id, gossipy/node.py:651-653)
-                    rslot = self._alloc_slot(receiver)
+                    rslot = self._alloc_slot(receiver, t)
                     pull_nodes.append(receiver)
                     pull_slots.append(rslot)
                     rpid = self._reply_pid(t, receiver)
@@ -349,7 +362,7 @@ class Scheduler:
                     continue
                 rslot = -1
                 if reply_flag == -2:  # PUSH_PULL: reply with post-merge model
-                    rslot = self._alloc_slot(receiver)
+                    rslot = self._alloc_slot(receiver, t)
                     rpid = self._reply_pid(t, receiver)
                     if not self._enqueue_reply(t, receiver, sender, rslot, rpid):
                         freed.append(rslot)
@@ -426,7 +439,7 @@ class Scheduler:
             # recycle this tick's consumed/dropped slots (safe: any reuse
             # happens in a later tick's launch, stream-ordered after the
             # consuming kernel)
-            self._free_slots.extend(freed)
+            self._recycle(t, freed)
 
         eval_nodes = None
         if cfg.sampling_eval > 0:
@@ -602,7 +615,7 @@ class TokenizedScheduler(Scheduler):
                         is_pull = proto == AntiEntropyProtocol.PULL
                         slot = -1
                         if not is_pull:
-                            slot = self._alloc_slot(sender)
+                            slot = self._alloc_slot(sender, t)
                             snap_nodes.append(sender)
                             snap_slots.append(slot)
                         sent += 1
@@ -654,7 +667,7 @@ class TokenizedScheduler(Scheduler):
                             freed.append(slot)
                         continue
                     if is_pull:
-                        rslot = self._alloc_slot(receiver)
+                        rslot = self._alloc_slot(receiver, t)
                         pull_nodes.append(receiver)
                         pull_slots.append(rslot)
                         rpid = self._reply_pid(t, receiver)
@@ -663,7 +676,7 @@ class TokenizedScheduler(Scheduler):
                         continue
                     rslot = -1
                     if reply_flag == -2:
-                        rslot = self._alloc_slot(receiver)
+                        rslot = self._alloc_slot(receiver, t)
                         rpid = self._reply_pid(t, receiver)
                         if not self._enqueue_reply(t, receiver, sender, rslot, rpid):
                             freed.append(rslot)
@@ -684,7 +697,7 @@ class TokenizedScheduler(Scheduler):
                         gdl = tick_stream(Purpose.DELAY, receiver)
                         for _ in range(reaction):
                             peer = self._burst_peer(receiver, gp)
-                            bslot = self._alloc_slot(receiver)
+                            bslot = self._alloc_slot(receiver, t)
                             burst_snap_nodes.append(receiver)
                             burst_snap_slots.append(bslot)
                             if cfg.n_parts > 0:
@@ -800,7 +813,7 @@ class TokenizedScheduler(Scheduler):
             sent += sent_r
             failed += failed_r
             total_size += size_r
-            self._free_slots.extend(freed)
+            self._recycle(t, freed)
 
         eval_nodes = None
         if cfg.sampling_eval > 0:
@@ -949,7 +962,7 @@ class All2AllScheduler(Scheduler):
                 for i in firing:
                     node = int(i)
                     peers = self._all_peers(node)
-                    slot = self._alloc_slot(node)
+                    slot = self._alloc_slot(node, t)
                     snap_nodes.append(node)
                     snap_slots.append(slot)
                     refs = 0
@@ -1009,7 +1022,7 @@ class All2AllScheduler(Scheduler):
             )
             if phase.n_events:
                 ticks.append(phase)
-            self._free_slots.extend(freed)
+            self._recycle(t, freed)
 
         eval_nodes = None
         if cfg.sampling_eval > 0:
@@ -1131,7 +1144,7 @@ class CacheNeighScheduler(Scheduler):
                 delays = self._delays(t, n_f, sizes)
                 for j in range(n_f):
                     sender, receiver = int(firing[j]), int(peers[j])
-                    slot = self._alloc_slot(sender)
+                    slot = self._alloc_slot(sender, t)
                     snap_nodes.append(sender)
                     snap_slots.append(slot)
                     sent += 1
@@ -1166,7 +1179,7 @@ class CacheNeighScheduler(Scheduler):
                         self._deref(slot, freed)
                         continue
                     if rf == -2:  # reply with own snapshot (no merge)
-                        rslot = self._alloc_slot(receiver)
+                        rslot = self._alloc_slot(receiver, t)
                         reply_snap_nodes.append(receiver)
                         reply_snap_slots.append(rslot)
                         if self._enqueue_reply(t, receiver, sender, rslot):
@@ -1199,7 +1212,7 @@ class CacheNeighScheduler(Scheduler):
             sent += sent_r
             failed += failed_r
             total_size += size_r
-            self._free_slots.extend(freed)
+            self._recycle(t, freed)
 
         eval_nodes = None
         if cfg.sampling_eval > 0:
@@ -1325,7 +1338,7 @@ class PENSScheduler(Scheduler):
                 delays = self._delays(t, n_f, np.full(n_f, cfg.model_size))
                 for j in range(n_f):
                     sender, receiver = int(firing[j]), int(peers[j])
-                    slot = self._alloc_slot(sender)
+                    slot = self._alloc_slot(sender, t)
                     snap_nodes.append(sender)
                     snap_slots.append(slot)
                     sent += 1
@@ -1405,7 +1418,7 @@ class PENSScheduler(Scheduler):
             )
             if phase.n_events:
                 ticks.append(phase)
-            self._free_slots.extend(freed)
+            self._recycle(t, freed)
 
         eval_nodes = None
         if cfg.sampling_eval > 0:
