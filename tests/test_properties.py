@@ -53,9 +53,12 @@ def test_slot_allocator_never_double_frees(proto, drop, online, dmax, seed):
     s = Scheduler(cfg)
     for r in range(6):
         sched = s.next_round(r)
-        # free list must never contain duplicates (double free)
-        assert len(s._free_slots) == len(set(s._free_slots))
-        # high-water bounded: in-flight <= messages within the delay horizon
-        assert sched.n_slots <= 40 * (dmax + 2) * 2 + 64
+        # reuse queue must never contain duplicates (double free)
+        q = [slot for _, slot in s._reuse_q]
+        assert len(q) == len(set(q))
+        # high-water bounded: in-flight + the reuse-lag window's worth of
+        # retired slots (slots sit out SLOT_REUSE_LAG ticks before reuse)
+        lag = Scheduler.SLOT_REUSE_LAG
+        assert sched.n_slots <= 40 * (dmax + 2 + lag) * 2 + 64
     # delivered + failed == sent accounting closes over rounds with no
     # in-flight messages remaining after a drain round
