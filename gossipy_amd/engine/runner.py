@@ -1086,11 +1086,15 @@ class BatchedGossipSimulator(SimulationEventSender):
                     and os.environ.get("GOSSIPY_NO_MULTIFAST") != "1"
                 ):
                     # multi-rank: one vectorized per-round prep + upload,
-                    # per-tick loop of launches and RCCL exchanges only
+                    # per-group loop of launches and RCCL exchanges only.
+                    # Launch-group merging applies here too — the hazard scan
+                    # runs on global node/slot ids, so the merged bounds are
+                    # identical on every rank, and each group moves all its
+                    # cross-rank slots in one batched exchange.
                     flat = getattr(self.scheduler, "last_flat", None)
                     if flat is None:
                         flat = self._flatten_phases(sched.ticks)
-                    self._run_round_multi(flat)
+                    self._run_round_multi(self._maybe_merge(flat))
                 elif (
                     self._flat_schedulable
                     and self._flat_exec_ok()
