@@ -1,4 +1,7 @@
 """Per-phase host timing of the flagship bench loop (debug utility)."""
+import os
+import sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import time
 import numpy as np
 import torch
@@ -38,14 +41,14 @@ for i in range(N):
         _, sched, flat = pre
     else:
         sched = sim.scheduler.next_round_flat(r)
-        flat = sim.scheduler.last_flat
+        flat = sim._maybe_merge(sim.scheduler.last_flat)
     sim._prefetched = None
     sim.pool.ensure(sched.n_slots)
     t1 = time.perf_counter()
     sim._run_round_fast(flat)
     t2 = time.perf_counter()
     s2 = sim.scheduler.next_round_flat(r + 1)
-    sim._prefetched = (r + 1, s2, sim.scheduler.last_flat)
+    sim._prefetched = (r + 1, s2, sim._maybe_merge(sim.scheduler.last_flat))
     t3 = time.perf_counter()
     sim._evaluate(sched, (r + 1) * cfg.delta - 1)
     sim.rounds_done += 1

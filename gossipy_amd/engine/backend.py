@@ -61,6 +61,8 @@ class TorchBackend:
     def __init__(self):
         #: per-spec partition tensors: id(spec) -> (perm, ptr, arena_part)
         self._part_cache = {}
+        #: pinned staging buffers for the pipelined eval fetch, per shape
+        self._eval_bufs = {}
 
     def _part(self, spec, device):
         """(perm[D], ptr[P+1], arena_part[D]) on ``device`` for a
@@ -976,6 +978,55 @@ class HIPBackend(TorchBackend):
             max(1, spec.local_epochs),
             spec.batch_size,
         )
+
+    def eval_metrics_launch(self, state, spec, local_ids, gx, gy):
+        """Asynchronous half of :meth:`eval_metrics_fast`: enqueue the K13
+        kernel and a non-blocking D2H into a pinned staging buffer, record
+        an event, and return a handle for :meth:`eval_metrics_collect`.
+        Lets the runner overlap a round's evaluation fetch with the next
+        round's compute (the collect usually finds the event already
+        signalled). Returns ``None`` for families on the generic path."""
+        if spec.family not in ("logreg", "pegasos", "adaline"):
+            return None
+        is_margin = spec.family in ("pegasos", "adaline")
+        out = self.ext.eval_metrics(
+            state.params,
+            local_ids.to(state.params.device, torch.int32),
+            gx,
+            gy,
+            spec.d_in,
+            1 if is_margin else spec.n_classes,
+            is_margin,
+        )
+        # double-buffered pinned staging: at most one handle is in flight
+        bufs = self._eval_bufs.setdefault(tuple(out.shape), [None, None])
+        self._eval_flip = flip = getattr(self, "_eval_flip", 0) ^ 1
+        if bufs[flip] is None or bufs[flip].shape != out.shape:
+            bufs[flip] = torch.empty(out.shape, dtype=out.dtype,
+                                     pin_memory=True)
+        bufs[flip].copy_(out, non_blocking=True)
+        ev = torch.cuda.Event()
+        ev.record()
+        return (bufs[flip], ev)
+
+    @staticmethod
+    def eval_metrics_collect(handle):
+        """Blocking half: wait for the staged copy and build metric dicts
+        (same values as :meth:`eval_metrics_fast`)."""
+        buf, ev = handle
+        ev.synchronize()
+        res = []
+        for row in buf.numpy():
+            d = {
+                "accuracy": float(row[0]),
+                "precision": float(row[1]),
+                "recall": float(row[2]),
+                "f1_score": float(row[3]),
+            }
+            if row[4] >= 0:
+                d["auc"] = float(row[4])
+            res.append(d)
+        return res
 
     def eval_metrics_fast(self, state, spec, local_ids, gx, gy):
         """One-launch K13 evaluation for affine/margin families; returns a

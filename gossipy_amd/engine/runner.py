@@ -595,6 +595,30 @@ class BatchedGossipSimulator(SimulationEventSender):
                 self.notify_evaluation(t, True, results_local)
             return
 
+        # pipelined eval (single rank, global eval set only): enqueue the
+        # K13 kernel + async D2H now, collect the metrics one round later —
+        # the fetch overlaps the next round's compute instead of syncing
+        # the stream mid-loop. Values are identical; report entries keep
+        # their own round timestamps and stay chronological.
+        if (
+            self.world == 1
+            and self.data.tx is None
+            and self.data.gx is not None
+            and os.environ.get("GOSSIPY_NO_EVAL_PIPE") != "1"
+            and hasattr(self.backend, "eval_metrics_launch")
+            and self.spec.family in ("logreg", "pegasos", "adaline")
+        ):
+            self._drain_eval()
+            if len(mine):
+                h = self.backend.eval_metrics_launch(
+                    self.state, self.spec, local_ids, self.data.gx, self.data.gy
+                )
+                if h is not None:
+                    self._eval_pending = (t, h)
+                    return
+            else:
+                return
+
         results_global: List[dict] = []
         if self.data.gx is not None and len(mine):
             fast_eval = getattr(self.backend, "eval_metrics_fast", None)
@@ -645,6 +669,17 @@ class BatchedGossipSimulator(SimulationEventSender):
                 self.notify_evaluation(t, True, results_local)
             if results_global:
                 self.notify_evaluation(t, False, results_global)
+
+    def _drain_eval(self) -> None:
+        """Collect an in-flight pipelined evaluation, if any."""
+        pend = getattr(self, "_eval_pending", None)
+        if pend is None:
+            return
+        self._eval_pending = None
+        t_prev, h = pend
+        results = self.backend.eval_metrics_collect(h)
+        if results:
+            self.notify_evaluation(t_prev, False, results)
 
     def _fast_path_ok(self) -> bool:
         """Single-GPU fast path: native scheduler + HIP round executor —
@@ -1114,6 +1149,7 @@ class BatchedGossipSimulator(SimulationEventSender):
             self._evaluate(sched, (r + 1) * self.cfg.delta - 1)
             self.rounds_done += 1
             self.notify_timestep((r + 1) * self.cfg.delta - 1)
+        self._drain_eval()
         self.notify_end()
 
     def notify_message_counts(self, sched: RoundSchedule) -> None:
@@ -1153,6 +1189,7 @@ class BatchedGossipSimulator(SimulationEventSender):
         this is what makes the checkpoint valid for any scheduler
         implementation (python or native C++).
         """
+        self._drain_eval()
         import dill
 
         blob = {
@@ -1477,6 +1514,7 @@ class BatchedPENSGossipSimulator(BatchedGossipSimulator):
             self._evaluate(sched, (r + 1) * self.cfg.delta - 1)
             self.rounds_done += 1
             self.notify_timestep((r + 1) * self.cfg.delta - 1)
+        self._drain_eval()
         self.notify_end()
 
 

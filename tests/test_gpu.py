@@ -1136,3 +1136,36 @@ class TestEngineCheckpointGPU:
         assert torch.allclose(
             ref.local_params(), restored.local_params(), atol=1e-6
         )
+
+
+@pytest.mark.gpu
+def test_pipelined_eval_matches_sync(monkeypatch):
+    """The lag-1 pipelined evaluation (async D2H, collected next round)
+    must produce byte-identical report entries to the synchronous path."""
+    def run(pipe: bool):
+        if pipe:
+            monkeypatch.delenv("GOSSIPY_NO_EVAL_PIPE", raising=False)
+        else:
+            monkeypatch.setenv("GOSSIPY_NO_EVAL_PIPE", "1")
+        data = _mk_data(200, 57, CUDA)
+        cfg = EngineConfig(
+            n_nodes=200, delta=60, protocol=AntiEntropyProtocol.PUSH,
+            model_size=116, sampling_eval=0.1, seed=11,
+        )
+        sim = BatchedGossipSimulator(
+            cfg, LogRegSpec(d_in=57, n_classes=2, lr=0.1), data, device=CUDA
+        )
+        rep = SimulationReport()
+        sim.add_receiver(rep)
+        sim.init_nodes()
+        sim.start(n_rounds=3)
+        sim.start(n_rounds=2)  # drain/refill across start() boundaries
+        torch.cuda.synchronize()
+        return rep.get_evaluation(False)
+
+    ev_pipe = run(True)
+    ev_sync = run(False)
+    assert len(ev_pipe) == len(ev_sync) > 0
+    for (t1, d1), (t2, d2) in zip(ev_pipe, ev_sync):
+        assert t1 == t2
+        assert d1 == d2
