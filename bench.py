@@ -64,7 +64,10 @@ def main() -> None:
     from gossipy_amd.simul import SimulationReport
 
     n_nodes = args.nodes_per_gpu * world
-    n_samples = SAMPLES_PER_SHARD_SET * world
+    # dataset size tracks the TOTAL population (4601 samples per 1000
+    # nodes), so a given (n_nodes, seed) pair sees identical data at any
+    # GPU count — results are residency-invariant, only faster
+    n_samples = max(1, SAMPLES_PER_SHARD_SET * n_nodes // NODES_PER_GPU)
 
     # deterministic global dataset; every rank generates it and slices its
     # residency block's shards (no network, random-init weights)
@@ -135,7 +138,7 @@ def main() -> None:
             "scaling": "weak",
             "vs_baseline": round(value / BASELINE_NODE_ROUNDS_PER_SEC, 2),
             "dtype": "fp32",
-            "data": f"synthetic spambase-shaped ({SAMPLES_PER_SHARD_SET}x{D_IN} per {args.nodes_per_gpu} nodes/GPU), random-init weights",
+            "data": f"synthetic spambase-shaped ({SAMPLES_PER_SHARD_SET}x{D_IN} per {NODES_PER_GPU} nodes), random-init weights",
             "config": {
                 "model": "logreg-57x2 (Hegedus-2021)",
                 "global_batch": None,
