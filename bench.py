@@ -51,7 +51,13 @@ def main() -> None:
     import torch.distributed as dist
 
     if world > 1:
-        dist.init_process_group("nccl" if use_cuda else "gloo")
+        # device_id => eager (collective) communicator init; without it the
+        # first NCCL op could be a pairwise batch_isend_irecv, whose lazy
+        # world-comm init deadlocks ranks that have no op in that group
+        if use_cuda:
+            dist.init_process_group("nccl", device_id=device)
+        else:
+            dist.init_process_group("gloo")
 
     from gossipy_amd.core import AntiEntropyProtocol
     from gossipy_amd.data import make_synthetic_classification

@@ -152,6 +152,11 @@ class BatchedGossipSimulator(SimulationEventSender):
         self.backend.init_params(self.state, self.spec, tape, self.cfg.n_nodes)
         all_local = torch.arange(self.n_local)
         self.backend.update(self.state, self.data, self.spec, all_local)
+        if self.world > 1:
+            # collective world-communicator init before any pairwise
+            # batch_isend_irecv (lazy NCCL init on a p2p group deadlocks
+            # ranks that have no op in it)
+            dist.barrier()
         self.initialized = True
 
     # -- cross-GPU slot exchange ---------------------------------------------
