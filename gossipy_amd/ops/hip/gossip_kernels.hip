@@ -1163,7 +1163,7 @@ DEV_INLINE void mlp_update(const MlpArgs& a, int node, float* W,
     }
 }
 
-__global__ void __launch_bounds__(256)
+__global__ void __launch_bounds__(512)
 tick_mlp_kernel(MlpArgs a)
 {
     int i = blockIdx.x;
@@ -1884,7 +1884,22 @@ void tick_mlp(torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
          2 * (size_t)bsmax * act_max);
     TORCH_CHECK(smem <= 160 * 1024,
         "mlp LDS budget exceeded (", smem, " B); shrink batch_size/hidden");
-    hipLaunchKernelGGL(tick_mlp_kernel, dim3(n), dim3(256), smem,
+    // block size: 8 waves when any GEMM stage has >= 8 MFMA tiles (halves
+    // the sequential tile passes of the widest stage — e.g. dW [100,57] is
+    // 28 tiles), else the default 4 waves
+    int max_tiles = 0;
+    {
+        int m16 = (bsmax + 15) >> 4;
+        for (int l = 0; l < n_layers; ++l) {
+            int fin16 = (L[4 * l + 2] + 15) >> 4;
+            int fout16 = (L[4 * l + 3] + 15) >> 4;
+            max_tiles = std::max(max_tiles, m16 * fout16);   // forward
+            max_tiles = std::max(max_tiles, m16 * fin16);    // dX
+            max_tiles = std::max(max_tiles, fout16 * fin16); // dW
+        }
+    }
+    int threads = (max_tiles >= 8) ? 512 : 256;
+    hipLaunchKernelGGL(tick_mlp_kernel, dim3(n), dim3(threads), smem,
                        current_stream(), a);
 }
 
