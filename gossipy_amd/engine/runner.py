@@ -131,6 +131,9 @@ class BatchedGossipSimulator(SimulationEventSender):
         self._flat_schedulable = True
         #: host mirror of per-round slot owners (set per round)
         self._slot_owner: Optional[np.ndarray] = None
+        #: single worker thread that drives the (GIL-releasing) C++ round
+        #: executors so schedule prefetch overlaps the launch queue
+        self._exec_pool = None
 
     # -- residency helpers ---------------------------------------------------
 
@@ -1096,6 +1099,16 @@ class BatchedGossipSimulator(SimulationEventSender):
         """Run ``n_rounds`` rounds."""
         assert self.initialized, "call init_nodes() first"
         fast = self._fast_path_ok()
+        if (
+            fast
+            and self._exec_pool is None
+            and os.environ.get("GOSSIPY_NO_THREAD") != "1"
+        ):
+            from concurrent.futures import ThreadPoolExecutor
+
+            self._exec_pool = ThreadPoolExecutor(
+                1, thread_name_prefix="gossipy-exec"
+            )
         for _ in range(n_rounds):
             r = self.rounds_done
             if fast:
@@ -1110,13 +1123,21 @@ class BatchedGossipSimulator(SimulationEventSender):
                     flat = self._maybe_merge(self.scheduler.last_flat)
                 self._prefetched = None
                 self.pool.ensure(sched.n_slots)
-                self._run_round_fast(flat)
-                # overlap: derive round r+1's schedule (and its launch-group
-                # merge) on the host while the GPU executes round r's launches
+                # the C++ round executors release the GIL, so a worker
+                # thread can drive round r's launch queue while this thread
+                # derives round r+1's schedule + merge. The join before
+                # _evaluate keeps the stream's enqueue order deterministic.
+                fut = None
+                if self._exec_pool is not None:
+                    fut = self._exec_pool.submit(self._run_round_fast, flat)
+                else:
+                    self._run_round_fast(flat)
                 s2 = self.scheduler.next_round_flat(r + 1)
                 self._prefetched = (
                     r + 1, s2, self._maybe_merge(self.scheduler.last_flat)
                 )
+                if fut is not None:
+                    fut.result()
             else:
                 sched = self.scheduler.next_round(r)
                 self.pool.ensure(sched.n_slots)

@@ -2681,24 +2681,30 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m)
     m.def("tick_linear", &tick_linear, "fused merge + pegasos/adaline tick");
     m.def("tick_mlp", &tick_mlp, "fused merge + MLP SGD tick");
     m.def("run_round_logreg", &run_round_logreg,
-          "whole-round executor, logreg family");
+          "whole-round executor, logreg family",
+          py::call_guard<py::gil_scoped_release>());
     m.def("run_round_linear", &run_round_linear,
-          "whole-round executor, pegasos/adaline family");
+          "whole-round executor, pegasos/adaline family",
+          py::call_guard<py::gil_scoped_release>());
     m.def("tick_logreg_part", &tick_logreg_part,
           "fused partition-merge + age-rescaled logreg SGD tick (K7/K8)");
     m.def("run_round_logreg_part", &run_round_logreg_part,
-          "whole-round executor, partitioned logreg family");
+          "whole-round executor, partitioned logreg family",
+          py::call_guard<py::gil_scoped_release>());
     m.def("wmerge", &wmerge, "all2all weighted k-way merge (K5 weighted)");
     m.def("tick_logreg_samp", &tick_logreg_samp,
           "fused sampled-merge + logreg SGD tick (K6)");
     m.def("tick_mf", &tick_mf,
           "fused item-block merge + per-rating MF SGD tick (K9/K10)");
     m.def("run_round_mlp", &run_round_mlp,
-          "whole-round executor, MLP family");
+          "whole-round executor, MLP family",
+          py::call_guard<py::gil_scoped_release>());
     m.def("run_round_logreg_samp", &run_round_logreg_samp,
-          "whole-round executor, sampled logreg family");
+          "whole-round executor, sampled logreg family",
+          py::call_guard<py::gil_scoped_release>());
     m.def("run_round_mf", &run_round_mf,
-          "whole-round executor, MF recommender family");
+          "whole-round executor, MF recommender family",
+          py::call_guard<py::gil_scoped_release>());
     m.def("run_round_coop_logreg", &run_round_coop_logreg,
           "single-launch cooperative whole-round executor (logreg)");
     m.def("tick_pens", &tick_pens,
