@@ -1169,3 +1169,31 @@ def test_pipelined_eval_matches_sync(monkeypatch):
     for (t1, d1), (t2, d2) in zip(ev_pipe, ev_sync):
         assert t1 == t2
         assert d1 == d2
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("knob", ["GOSSIPY_THREAD", "GOSSIPY_NO_MERGE"])
+def test_perf_knobs_preserve_results(monkeypatch, knob):
+    """The opt-in executor-thread path and the merge opt-out must produce
+    bit-identical parameters to the default fast path."""
+    def run(with_knob: bool):
+        for k in ("GOSSIPY_THREAD", "GOSSIPY_NO_MERGE"):
+            monkeypatch.delenv(k, raising=False)
+        if with_knob:
+            monkeypatch.setenv(knob, "1")
+        data = _mk_data(300, 57, CUDA)
+        cfg = EngineConfig(
+            n_nodes=300, delta=50, protocol=AntiEntropyProtocol.PUSH_PULL,
+            model_size=116, drop_prob=0.1, delay=UniformDelay(0, 5), seed=21,
+        )
+        sim = BatchedGossipSimulator(
+            cfg, LogRegSpec(d_in=57, n_classes=2, lr=0.1), data, device=CUDA
+        )
+        sim.init_nodes()
+        sim.start(n_rounds=4)
+        torch.cuda.synchronize()
+        return sim.local_params().cpu().clone()
+
+    base = run(False)
+    alt = run(True)
+    assert torch.equal(base, alt)
