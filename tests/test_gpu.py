@@ -7,7 +7,7 @@ import torch
 
 pytestmark = pytest.mark.gpu
 
-from gossipy_amd.core import AntiEntropyProtocol, CreateModelMode
+from gossipy_amd.core import AntiEntropyProtocol, CreateModelMode, UniformDelay
 from gossipy_amd.data import make_synthetic_classification
 from gossipy_amd.engine import (
     BatchedGossipSimulator,
