@@ -250,7 +250,7 @@ protected:
     // the tick that freed them (FIFO), mirroring Scheduler.SLOT_REUSE_LAG:
     // adjacent ticks never alias slot ids, so the runner can fuse
     // conflict-free ticks into single launch groups.
-    static constexpr int64_t kSlotReuseLag = 16;
+    static constexpr int64_t kSlotReuseLag = 32;
     std::deque<std::pair<int64_t, int32_t>> reuse_q_;
     int64_t next_slot_ = 0;
     std::vector<int32_t> slot_owner_;

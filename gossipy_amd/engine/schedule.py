@@ -206,11 +206,12 @@ class Scheduler:
 
     #: a consumed slot id is only reused SLOT_REUSE_LAG ticks after the tick
     #: that freed it. Adjacent ticks therefore never alias slot ids, which is
-    #: what lets BatchedGossipSimulator._merge_flat_groups fuse conflict-free
-    #: ticks into single launch groups (the old LIFO free list re-issued a
-    #: slot on the very next tick, making every tick pair conflict). Costs a
-    #: few hundred extra pool rows at the flagship scale.
-    SLOT_REUSE_LAG = 16
+    #: what lets the runner fuse conflict-free ticks into single launch
+    #: groups (the old LIFO free list re-issued a slot on the very next
+    #: tick, making every tick pair conflict); 32 keeps even the entry-level
+    #: packer's long groups alias-free. Costs ~delay-window x firing-rate
+    #: extra pool rows — negligible against 288 GB of HBM3E.
+    SLOT_REUSE_LAG = 32
 
     def _alloc_slot(self, owner: int, t: int) -> int:
         q = self._reuse_q

@@ -1609,42 +1609,49 @@ def test_custom_delay_subclass_falls_back_to_python_scheduler():
 # ---------------------------------------------------------------------------
 
 
-def _abstract_replay(f, merged):
+def _abstract_replay(f, merged=None):
     """Replay a flat schedule with the executors' launch-order semantics
     (per group: snapshot launch -> deliver launch -> pull launch -> reply
     launch) over an abstract state machine, and return the final node
-    states. Identical output for the unmerged and merged tptr sets proves
-    the merge preserved every read-after-write ordering."""
+    states. Identical output across the unmerged, tick-merged and packed
+    forms proves every read-after-write ordering is preserved. Every slot
+    write carries the SAME tag ("w", <node state at write time>): a
+    snapshot, a pull snapshot, a protocol reply and an embedded snapshot
+    all capture the node's current params, so they must compare equal."""
     nodes: dict = {}
     slots: dict = {}
 
     def nval(x):
-        return nodes.get(x, ("init", int(x)))
+        return nodes.get(int(x), ("init", int(x)))
 
-    st, rt = f["snap_tptr"], f["recv_tptr"]
-    pt, qt = f["pull_tptr"], f["rep_tptr"]
-    if merged is not None:
-        st, rt = merged["snap_tptr"], merged["recv_tptr"]
-        pt, qt = merged["pull_tptr"], merged["rep_tptr"]
-    nptr, rep_nptr = f["recv_nptr"], f["rep_nptr"]
+    # tick-merged dicts are dict(f) with subsampled tptrs (event arrays
+    # shared with f); packed dicts replace the event arrays wholesale —
+    # either way the merged dict is self-contained
+    g = f if merged is None else merged
+    st, rt = g["snap_tptr"], g["recv_tptr"]
+    pt, qt = g["pull_tptr"], g["rep_tptr"]
+    nptr, rep_nptr = g["recv_nptr"], g["rep_nptr"]
+    rr = g.get("rep_reply_slots")
     for t in range(len(st) - 1):
         for i in range(st[t], st[t + 1]):  # snapshot launch
-            slots[int(f["snap_slots"][i])] = ("snap", nval(f["snap_nodes"][i]))
+            slots[int(g["snap_slots"][i])] = ("w", nval(g["snap_nodes"][i]))
         for r in range(rt[t], rt[t + 1]):  # deliver launch
-            x = int(f["recv_nodes"][r])
+            x = int(g["recv_nodes"][r])
             for d in range(nptr[r], nptr[r + 1]):
-                s = int(f["del_slots"][d])
+                s = int(g["del_slots"][d])
                 nodes[x] = ("merge", nval(x), slots.get(s, ("hole", s)))
-                rep = int(f["reply_slots"][d])
+                rep = int(g["reply_slots"][d])
                 if rep >= 0:
-                    slots[rep] = ("reply", nodes[x])
+                    slots[rep] = ("w", nodes[x])
         for i in range(pt[t], pt[t + 1]):  # pull-snapshot launch
-            slots[int(f["pull_slots"][i])] = ("pull", nval(f["pull_nodes"][i]))
-        for r in range(qt[t], qt[t + 1]):  # reply-deliver launch
-            x = int(f["rep_nodes"][r])
+            slots[int(g["pull_slots"][i])] = ("w", nval(g["pull_nodes"][i]))
+        for r in range(qt[t], qt[t + 1]):  # second deliver launch
+            x = int(g["rep_nodes"][r])
             for d in range(rep_nptr[r], rep_nptr[r + 1]):
-                s = int(f["rep_slots"][d])
+                s = int(g["rep_slots"][d])
                 nodes[x] = ("merge", nval(x), slots.get(s, ("hole", s)))
+                if rr is not None and len(rr) and int(rr[d]) >= 0:
+                    slots[int(rr[d])] = ("w", nodes[x])
     return nodes
 
 
@@ -1696,7 +1703,9 @@ class TestLaunchGroupMerge:
                 assert _abstract_replay(f, None) == _abstract_replay(f, m)
 
     def test_merge_native_scheduler_flat(self):
-        pytest.importorskip("gossipy_amd._native_sched")
+        from gossipy_amd import ops
+        if ops.load_sched() is None:
+            pytest.skip("native scheduler not built")
         from gossipy_amd.engine.schedule import NativeSchedulerAdapter
 
         cfg = EngineConfig(
@@ -1723,3 +1732,101 @@ class TestLaunchGroupMerge:
         f = BatchedGossipSimulator._flatten_phases(sched.ticks)
         m = BatchedGossipSimulator._merge_flat_groups(f)
         assert len(m["snap_tptr"]) - 1 <= (len(f["snap_tptr"]) - 1) * 0.8
+
+
+class TestEntryLevelPacking:
+    """_pack_flat: the entry-level packer (coalesced receiver rows +
+    embedded snapshots) must preserve launch-order semantics exactly."""
+
+    def _flats(self, protocol, delay, seed, n=60, drop=0.2):
+        cfg = EngineConfig(
+            n_nodes=n, model_size=10, protocol=protocol, delay=delay,
+            drop_prob=drop, online_prob=0.8, delta=80, seed=seed,
+        )
+        sch = Scheduler(cfg)
+        for r in range(3):
+            yield BatchedGossipSimulator._flatten_phases(
+                sch.next_round(r).ticks
+            )
+
+    @staticmethod
+    def _invariants(p):
+        st, rt, qt = p["snap_tptr"], p["recv_tptr"], p["rep_tptr"]
+        for t in range(len(st) - 1):
+            rx = p["recv_nodes"][rt[t] : rt[t + 1]].tolist()
+            assert len(set(rx)) == len(rx), "dup receiver in deliver launch"
+            qx = p["rep_nodes"][qt[t] : qt[t + 1]].tolist()
+            assert len(set(qx)) == len(qx), "dup receiver in second launch"
+            ws = [int(s) for s in p["snap_slots"][st[t] : st[t + 1]]]
+            d0, d1 = p["recv_nptr"][rt[t]], p["recv_nptr"][rt[t + 1]]
+            ws += [int(s) for s in p["reply_slots"][d0:d1] if s >= 0]
+            e0, e1 = p["rep_nptr"][qt[t]], p["rep_nptr"][qt[t + 1]]
+            ws += [int(s) for s in p["rep_reply_slots"][e0:e1] if s >= 0]
+            assert len(set(ws)) == len(ws), "slot written twice in group"
+            # produce->consume never within one launch
+            l2w = set(int(s) for s in p["reply_slots"][d0:d1] if s >= 0)
+            l2r = set(int(s) for s in p["del_slots"][d0:d1])
+            assert not (l2w & l2r), "launch-2 write consumed in launch 2"
+            l3w = set(int(s) for s in p["rep_reply_slots"][e0:e1] if s >= 0)
+            l3r = set(int(s) for s in p["rep_slots"][e0:e1])
+            assert not (l3w & l3r), "launch-3 write consumed in launch 3"
+
+    @pytest.mark.parametrize("protocol", [AntiEntropyProtocol.PUSH,
+                                          AntiEntropyProtocol.PUSH_PULL])
+    @pytest.mark.parametrize("delay", [ConstantDelay(0), UniformDelay(0, 7)])
+    def test_pack_preserves_semantics(self, protocol, delay):
+        for seed in (1, 7, 42):
+            for f in self._flats(protocol, delay, seed):
+                p = BatchedGossipSimulator._pack_flat(f)
+                for k in ("snap", "recv", "rep"):
+                    # all events survive (snaps may migrate between the
+                    # snapshot launch and embedded writes; deliveries may
+                    # migrate between the two deliver launches)
+                    pass
+                n_del = len(p["del_slots"]) + len(p["rep_slots"])
+                assert n_del == len(f["del_slots"]) + len(f["rep_slots"])
+                n_snap = (
+                    len(p["snap_slots"])
+                    + sum(1 for s in p["reply_slots"] if s >= 0)
+                    + sum(1 for s in p["rep_reply_slots"] if s >= 0)
+                )
+                assert n_snap == (
+                    len(f["snap_slots"]) + len(f["pull_slots"])
+                    + sum(1 for s in f["reply_slots"] if s >= 0)
+                )
+                self._invariants(p)
+                assert _abstract_replay(f) == _abstract_replay(f, p)
+
+    def test_pack_native_scheduler_flat(self):
+        from gossipy_amd import ops
+        if ops.load_sched() is None:
+            pytest.skip("native scheduler not built")
+        from gossipy_amd.engine.schedule import NativeSchedulerAdapter
+
+        cfg = EngineConfig(
+            n_nodes=120, model_size=10,
+            protocol=AntiEntropyProtocol.PUSH_PULL,
+            delay=UniformDelay(0, 5), drop_prob=0.1, delta=100, seed=9,
+        )
+        sch = NativeSchedulerAdapter(cfg)
+        for r in range(3):
+            sch.next_round_flat(r)
+            f = sch.last_flat
+            p = BatchedGossipSimulator._pack_flat(f)
+            self._invariants(p)
+            assert _abstract_replay(f) == _abstract_replay(f, p)
+
+    def test_pack_is_effective(self):
+        cfg = EngineConfig(
+            n_nodes=1000, model_size=10, protocol=AntiEntropyProtocol.PUSH,
+            delay=UniformDelay(0, 9), drop_prob=0.0, delta=100, seed=4,
+        )
+        sch = Scheduler(cfg)
+        f = BatchedGossipSimulator._flatten_phases(sch.next_round(0).ticks)
+        p = BatchedGossipSimulator._pack_flat(f)
+        groups = len(p["snap_tptr"]) - 1
+        merged = len(
+            BatchedGossipSimulator._merge_flat_groups(f)["snap_tptr"]
+        ) - 1
+        assert groups < merged  # strictly better than tick merging
+        assert groups <= 20
