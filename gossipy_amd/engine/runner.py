@@ -775,6 +775,19 @@ class BatchedGossipSimulator(SimulationEventSender):
         Tracks a cumulative (ticks, groups) counter for the perf probes."""
         if os.environ.get("GOSSIPY_NO_MERGE") == "1":
             return f
+        if (
+            os.environ.get("GOSSIPY_PACK") == "1"
+            and os.environ.get("GOSSIPY_COOP") != "1"
+        ):
+            # entry-level packing (python reference packer — A/B only; the
+            # C++ packer in the scheduler is the production path)
+            out = self._pack_flat(f)
+            t = getattr(self, "merge_stats", (0, 0))
+            self.merge_stats = (
+                t[0] + len(f["snap_tptr"]) - 1,
+                t[1] + len(out["snap_tptr"]) - 1,
+            )
+            return out
         mb = f.get("merge_bounds")
         if mb is not None and len(mb) - 1 < len(f["snap_tptr"]) - 1:
             # native scheduler pre-computed the group boundaries in C++
@@ -1074,8 +1087,13 @@ class BatchedGossipSimulator(SimulationEventSender):
             "rep_slots",
             "del_pids",
             "rep_pids",
+            "rep_reply_slots",
         )
-        parts = [np.ascontiguousarray(f[n], dtype=np.int32) for n in dev_names]
+        _empty = np.zeros(0, np.int32)
+        parts = [
+            np.ascontiguousarray(f.get(n, _empty), dtype=np.int32)
+            for n in dev_names
+        ]
         lens = [len(p) for p in parts]
         host = np.concatenate(parts) if sum(lens) else np.zeros(1, np.int32)
         dbuf = torch.from_numpy(host).to(dev, non_blocking=True)
@@ -1194,6 +1212,7 @@ class BatchedGossipSimulator(SimulationEventSender):
                 max(1, spec.local_epochs),
                 spec.batch_size,
                 _MODE_ID[spec.mode],
+                views["rep_reply_slots"],
             )
         elif spec.family == "mf":
             ext.run_round_mf(
@@ -1202,6 +1221,7 @@ class BatchedGossipSimulator(SimulationEventSender):
                 spec.n_items,
                 spec.reg,
                 spec.lr,
+                views["rep_reply_slots"],
             )
         elif getattr(spec, "n_parts", 0) > 0:
             perm, pptr, apart = self.backend._part_dev(spec, dev)
@@ -1223,6 +1243,7 @@ class BatchedGossipSimulator(SimulationEventSender):
                 max(1, spec.local_epochs),
                 spec.batch_size,
                 _MODE_ID[spec.mode],
+                views["rep_reply_slots"],
             )
         elif spec.family == "logreg":
             ext.run_round_logreg(
@@ -1234,6 +1255,7 @@ class BatchedGossipSimulator(SimulationEventSender):
                 max(1, spec.local_epochs),
                 spec.batch_size,
                 _MODE_ID[spec.mode],
+                views["rep_reply_slots"],
             )
         elif spec.family == "mlp":
             layout = torch.tensor(
@@ -1250,6 +1272,7 @@ class BatchedGossipSimulator(SimulationEventSender):
                 max(1, spec.local_epochs),
                 spec.batch_size,
                 _MODE_ID[spec.mode],
+                views["rep_reply_slots"],
             )
         else:
             ext.run_round_linear(
@@ -1258,6 +1281,7 @@ class BatchedGossipSimulator(SimulationEventSender):
                 spec.lam if spec.family == "pegasos" else spec.lr,
                 1 if spec.family == "pegasos" else 0,
                 _MODE_ID[spec.mode],
+                views["rep_reply_slots"],
             )
 
     def start(self, n_rounds: int = 100) -> None:

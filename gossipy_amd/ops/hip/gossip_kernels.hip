@@ -1998,13 +1998,17 @@ void run_round_logreg(
     torch::Tensor rep_slots,
     torch::Tensor X, torch::Tensor Y, torch::Tensor counts,
     int64_t d, int64_t k, double lr, double wd, int64_t epochs, int64_t bs,
-    int64_t mode)
+    int64_t mode,
+    torch::Tensor rep_reply_slots)
 {
     CHECK_DEV(params); CHECK_DEV(slots); CHECK_DEV(X);
     RoundArrays r = unpack_round(snap_nodes, snap_slots, snap_tptr, recv_nodes,
                                  recv_nptr, recv_tptr, del_slots, reply_slots,
                                  pull_nodes, pull_slots, pull_tptr, rep_nodes,
                                  rep_nptr, rep_tptr, rep_slots);
+    const int* rep_rr =
+        (rep_reply_slots.defined() && rep_reply_slots.numel())
+            ? rep_reply_slots.data_ptr<int>() : nullptr;
     hipStream_t s = current_stream();
     LogregArgs a;
     a.params = params.data_ptr<float>(); a.ages = ages.data_ptr<int>();
@@ -2018,7 +2022,7 @@ void run_round_logreg(
     int bsmax = (bs == 0) ? a.Smax : std::min<int>(bs, a.Smax);
     size_t smem = sizeof(float) * (2 * a.D + (size_t)bsmax * a.d + (size_t)bsmax * a.k);
     TORCH_CHECK(smem <= 160 * 1024, "logreg LDS budget exceeded");
-    if (max_group(r) <= sb_threshold()) {
+    if (max_group(r) <= sb_threshold() && rep_rr == nullptr) {
         // tiny batches: the coop round kernel launched PLAIN with one
         // workgroup (round_sync degrades to __syncthreads)
         auto dev = params.device();
@@ -2067,7 +2071,7 @@ void run_round_logreg(
             a.nodes = r.rep_nodes + q0;
             a.ptr = r.rep_nptr + q0;
             a.dslots = r.rep_slots;
-            a.rslots = nullptr;
+            a.rslots = rep_rr;
             hipLaunchKernelGGL(tick_logreg_kernel, dim3(q1 - q0), dim3(128),
                                smem, s, a);
         }
@@ -2084,13 +2088,17 @@ void run_round_linear(
     torch::Tensor rep_nodes, torch::Tensor rep_nptr, torch::Tensor rep_tptr,
     torch::Tensor rep_slots,
     torch::Tensor X, torch::Tensor Y, torch::Tensor counts,
-    int64_t d, double lrlam, int64_t is_pegasos, int64_t mode)
+    int64_t d, double lrlam, int64_t is_pegasos, int64_t mode,
+    torch::Tensor rep_reply_slots)
 {
     CHECK_DEV(params); CHECK_DEV(slots); CHECK_DEV(X);
     RoundArrays r = unpack_round(snap_nodes, snap_slots, snap_tptr, recv_nodes,
                                  recv_nptr, recv_tptr, del_slots, reply_slots,
                                  pull_nodes, pull_slots, pull_tptr, rep_nodes,
                                  rep_nptr, rep_tptr, rep_slots);
+    const int* rep_rr =
+        (rep_reply_slots.defined() && rep_reply_slots.numel())
+            ? rep_reply_slots.data_ptr<int>() : nullptr;
     hipStream_t s = current_stream();
     LinearArgs a;
     a.params = params.data_ptr<float>(); a.ages = ages.data_ptr<int>();
@@ -2101,7 +2109,7 @@ void run_round_linear(
     a.d = d; a.Smax = X.size(1);
     a.lrlam = lrlam; a.is_pegasos = is_pegasos; a.mode = mode;
     a.update_only = 0;
-    if (max_group(r) <= sb_threshold()) {
+    if (max_group(r) <= sb_threshold() && rep_rr == nullptr) {
         // tiny batches: one plain single-workgroup launch for the round
         auto dev = params.device();
         auto st = snap_tptr.to(dev), rt = recv_tptr.to(dev);
@@ -2149,7 +2157,7 @@ void run_round_linear(
             a.nodes = r.rep_nodes + q0;
             a.ptr = r.rep_nptr + q0;
             a.dslots = r.rep_slots;
-            a.rslots = nullptr;
+            a.rslots = rep_rr;
             hipLaunchKernelGGL(tick_linear_kernel, dim3(q1 - q0), dim3(WAVE),
                                0, s, a);
         }
@@ -2253,7 +2261,8 @@ void run_round_logreg_part(
     torch::Tensor X, torch::Tensor Y, torch::Tensor counts,
     torch::Tensor perm, torch::Tensor pptr, torch::Tensor apart,
     int64_t n_parts, int64_t d, int64_t k, double lr, double wd,
-    int64_t epochs, int64_t bs, int64_t mode)
+    int64_t epochs, int64_t bs, int64_t mode,
+    torch::Tensor rep_reply_slots)
 {
     CHECK_DEV(params); CHECK_DEV(slots); CHECK_DEV(X);
     CHECK_DEV(perm); CHECK_DEV(pptr); CHECK_DEV(apart);
@@ -2264,6 +2273,9 @@ void run_round_logreg_part(
                                  rep_nptr, rep_tptr, rep_slots);
     const int* d_pids = del_pids.numel() ? del_pids.data_ptr<int>() : nullptr;
     const int* r_pids = rep_pids.numel() ? rep_pids.data_ptr<int>() : nullptr;
+    const int* rep_rr =
+        (rep_reply_slots.defined() && rep_reply_slots.numel())
+            ? rep_reply_slots.data_ptr<int>() : nullptr;
     hipStream_t s = current_stream();
     LogregPartArgs a;
     a.params = params.data_ptr<float>(); a.ages = ages.data_ptr<int>();
@@ -2280,7 +2292,7 @@ void run_round_logreg_part(
                                    (size_t)bsmax * a.k) +
                   sizeof(int) * 2 * a.P;
     TORCH_CHECK(smem <= 160 * 1024, "partitioned logreg LDS budget exceeded");
-    if (max_group(r) <= sb_threshold()) {
+    if (max_group(r) <= sb_threshold() && rep_rr == nullptr) {
         auto dev = params.device();
         auto st = snap_tptr.to(dev), rt = recv_tptr.to(dev);
         auto pt = pull_tptr.to(dev), qt = rep_tptr.to(dev);
@@ -2332,7 +2344,7 @@ void run_round_logreg_part(
             a.nodes = r.rep_nodes + q0;
             a.ptr = r.rep_nptr + q0;
             a.dslots = r.rep_slots;
-            a.rslots = nullptr;
+            a.rslots = rep_rr;
             a.dpids = r_pids;
             hipLaunchKernelGGL(tick_logreg_part_kernel, dim3(q1 - q0), dim3(128),
                                smem, s, a);
@@ -2351,13 +2363,17 @@ void run_round_mlp(
     torch::Tensor rep_slots,
     torch::Tensor X, torch::Tensor Y, torch::Tensor counts,
     torch::Tensor layout, int64_t n_layers, double lr, double wd,
-    int64_t epochs, int64_t bs, int64_t mode)
+    int64_t epochs, int64_t bs, int64_t mode,
+    torch::Tensor rep_reply_slots)
 {
     CHECK_DEV(params); CHECK_DEV(slots); CHECK_DEV(X); CHECK_DEV(layout);
     RoundArrays r = unpack_round(snap_nodes, snap_slots, snap_tptr, recv_nodes,
                                  recv_nptr, recv_tptr, del_slots, reply_slots,
                                  pull_nodes, pull_slots, pull_tptr, rep_nodes,
                                  rep_nptr, rep_tptr, rep_slots);
+    const int* rep_rr =
+        (rep_reply_slots.defined() && rep_reply_slots.numel())
+            ? rep_reply_slots.data_ptr<int>() : nullptr;
     hipStream_t s = current_stream();
     MlpArgs a;
     a.params = params.data_ptr<float>(); a.ages = ages.data_ptr<int>();
@@ -2407,7 +2423,7 @@ void run_round_mlp(
             a.nodes = r.rep_nodes + q0;
             a.ptr = r.rep_nptr + q0;
             a.dslots = r.rep_slots;
-            a.rslots = nullptr;
+            a.rslots = rep_rr;
             hipLaunchKernelGGL(tick_mlp_kernel, dim3(q1 - q0), dim3(256),
                                smem, s, a);
         }
@@ -2481,7 +2497,8 @@ void run_round_logreg_samp(
     torch::Tensor rep_slots, torch::Tensor rep_pids,
     torch::Tensor X, torch::Tensor Y, torch::Tensor counts,
     int64_t samp_c, int64_t d, int64_t k, double lr, double wd,
-    int64_t epochs, int64_t bs, int64_t mode)
+    int64_t epochs, int64_t bs, int64_t mode,
+    torch::Tensor rep_reply_slots)
 {
     CHECK_DEV(params); CHECK_DEV(slots); CHECK_DEV(X);
     TORCH_CHECK(mode != MODE_PASS, "Mode PASS not allowed for sampled models.");
@@ -2491,6 +2508,9 @@ void run_round_logreg_samp(
                                  rep_nptr, rep_tptr, rep_slots);
     const int* d_pids = del_pids.numel() ? del_pids.data_ptr<int>() : nullptr;
     const int* r_pids = rep_pids.numel() ? rep_pids.data_ptr<int>() : nullptr;
+    const int* rep_rr =
+        (rep_reply_slots.defined() && rep_reply_slots.numel())
+            ? rep_reply_slots.data_ptr<int>() : nullptr;
     hipStream_t s = current_stream();
     LogregSampArgs a;
     a.params = params.data_ptr<float>(); a.ages = ages.data_ptr<int>();
@@ -2529,7 +2549,7 @@ void run_round_logreg_samp(
             a.nodes = r.rep_nodes + q0;
             a.ptr = r.rep_nptr + q0;
             a.dslots = r.rep_slots;
-            a.rslots = nullptr;
+            a.rslots = rep_rr;
             a.dseeds = r_pids;
             hipLaunchKernelGGL(tick_logreg_samp_kernel, dim3(q1 - q0),
                                dim3(128), smem, s, a);
@@ -2547,13 +2567,17 @@ void run_round_mf(
     torch::Tensor rep_nodes, torch::Tensor rep_nptr, torch::Tensor rep_tptr,
     torch::Tensor rep_slots,
     torch::Tensor X, torch::Tensor Y, torch::Tensor counts,
-    int64_t k, int64_t n_items, double reg, double lr)
+    int64_t k, int64_t n_items, double reg, double lr,
+    torch::Tensor rep_reply_slots)
 {
     CHECK_DEV(params); CHECK_DEV(slots); CHECK_DEV(X);
     RoundArrays r = unpack_round(snap_nodes, snap_slots, snap_tptr, recv_nodes,
                                  recv_nptr, recv_tptr, del_slots, reply_slots,
                                  pull_nodes, pull_slots, pull_tptr, rep_nodes,
                                  rep_nptr, rep_tptr, rep_slots);
+    const int* rep_rr =
+        (rep_reply_slots.defined() && rep_reply_slots.numel())
+            ? rep_reply_slots.data_ptr<int>() : nullptr;
     hipStream_t s = current_stream();
     MFArgs a;
     a.params = params.data_ptr<float>(); a.ages = ages.data_ptr<int>();
@@ -2599,7 +2623,7 @@ void run_round_mf(
             a.nodes = r.rep_nodes + q0;
             a.ptr = r.rep_nptr + q0;
             a.dslots = r.rep_slots;
-            a.rslots = nullptr;
+            a.rslots = rep_rr;
             hipLaunchKernelGGL(tick_mf_kernel, dim3(q1 - q0), dim3(256), 0,
                                s, a);
         }
