@@ -149,7 +149,10 @@ class TestScheduler:
         s = Scheduler(self._cfg())
         for r in range(20):
             sched = s.next_round(r)
-        assert sched.n_slots < 3 * 50, "slots must be recycled across rounds"
+        # bounded by live messages + the reuse-lag window's retirements
+        # (slots sit out SLOT_REUSE_LAG ticks before re-issue), NOT by the
+        # cumulative send count (50 msgs/round x 20 rounds = 1000)
+        assert sched.n_slots < 6 * 50, "slots must be recycled across rounds"
 
 
 class TestEngineCPU:
