@@ -26,6 +26,7 @@
 #include <map>
 #include <deque>
 #include <unordered_map>
+#include <unordered_set>
 #include <vector>
 
 namespace py = pybind11;
@@ -185,6 +186,203 @@ static std::vector<int32_t> compute_merge_bounds(
     }
     bounds.push_back((int32_t)delta);
     return bounds;
+}
+
+
+// ---------------------------------------------------------------------------
+// Entry-level launch packing (bit-exact port of
+// BatchedGossipSimulator._pack_flat — parity enforced by
+// tests/test_native_sched.py). Coalesces deliveries to one receiver from
+// many ticks into one CSR row and embeds snapshots as per-delivery reply
+// writes; deliveries whose source slot is written in the group's first
+// deliver launch move to the second (rep_*) launch. Returns a py::dict of
+// packed arrays in the executors' format.
+// ---------------------------------------------------------------------------
+static py::dict pack_round(
+    const std::vector<int32_t>& snap_nodes, const std::vector<int32_t>& snap_slots,
+    const std::vector<int32_t>& snap_tptr,
+    const std::vector<int32_t>& recv_nodes, const std::vector<int32_t>& recv_nptr,
+    const std::vector<int32_t>& recv_tptr,
+    const std::vector<int32_t>& del_slots, const std::vector<int32_t>& reply_slots,
+    const std::vector<int32_t>& del_pids,
+    const std::vector<int32_t>& pull_nodes, const std::vector<int32_t>& pull_slots,
+    const std::vector<int32_t>& pull_tptr,
+    const std::vector<int32_t>& rep_nodes, const std::vector<int32_t>& rep_nptr,
+    const std::vector<int32_t>& rep_tptr, const std::vector<int32_t>& rep_slots,
+    const std::vector<int32_t>& rep_pids)
+{
+    int64_t delta = (int64_t)snap_tptr.size() - 1;
+    bool has_pid = !del_pids.empty() || !rep_pids.empty();
+
+    struct Ev { int32_t slot, reply, pid; };
+    struct Out {
+        std::vector<int32_t> snap_nodes, snap_slots, snap_tptr{0};
+        std::vector<int32_t> recv_nodes, recv_nptr{0}, recv_tptr{0};
+        std::vector<int32_t> del_slots, reply_slots, del_pids;
+        std::vector<int32_t> rep_nodes, rep_nptr{0}, rep_tptr{0};
+        std::vector<int32_t> rep_slots, rep_reply_slots, rep_pids;
+        std::vector<int32_t> pull_tptr{};
+    } o;
+
+    std::vector<std::pair<int32_t, int32_t>> g_snap;
+    std::unordered_map<int32_t, int32_t> rows2, rows3;
+    std::vector<std::pair<int32_t, std::vector<Ev>>> l2, l3;
+    std::unordered_map<int32_t, std::pair<char, int32_t>> latest;
+    std::unordered_map<int32_t, int32_t> written;
+    std::unordered_set<int32_t> touched;
+
+    auto close = [&]() {
+        for (auto& p : g_snap) {
+            o.snap_nodes.push_back(p.first);
+            o.snap_slots.push_back(p.second);
+        }
+        o.snap_tptr.push_back((int32_t)o.snap_nodes.size());
+        for (auto& row : l2) {
+            o.recv_nodes.push_back(row.first);
+            for (auto& e : row.second) {
+                o.del_slots.push_back(e.slot);
+                o.reply_slots.push_back(e.reply);
+                if (has_pid) o.del_pids.push_back(e.pid);
+            }
+            o.recv_nptr.push_back((int32_t)o.del_slots.size());
+        }
+        o.recv_tptr.push_back((int32_t)o.recv_nodes.size());
+        for (auto& row : l3) {
+            o.rep_nodes.push_back(row.first);
+            for (auto& e : row.second) {
+                o.rep_slots.push_back(e.slot);
+                o.rep_reply_slots.push_back(e.reply);
+                if (has_pid) o.rep_pids.push_back(e.pid);
+            }
+            o.rep_nptr.push_back((int32_t)o.rep_slots.size());
+        }
+        o.rep_tptr.push_back((int32_t)o.rep_nodes.size());
+        g_snap.clear(); rows2.clear(); rows3.clear();
+        l2.clear(); l3.clear(); latest.clear();
+        written.clear(); touched.clear();
+    };
+
+    auto place_snap = [&](int32_t node, int32_t slot) {
+        if (touched.count(slot)) close();
+        auto it = latest.find(node);
+        if (it == latest.end()) {
+            g_snap.emplace_back(node, slot);
+            written[slot] = 1;
+        } else {
+            auto& lst = (it->second.first == 2) ? l2 : l3;
+            auto& evs = lst[it->second.second].second;
+            if (evs.back().reply >= 0) {
+                close();
+                g_snap.emplace_back(node, slot);
+                written[slot] = 1;
+            } else {
+                evs.back().reply = slot;
+                written[slot] = (it->second.first == 2) ? 2 : 3;
+            }
+        }
+        touched.insert(slot);
+    };
+
+    auto place_delivery = [&](int32_t node, int32_t slot, int32_t reply,
+                              int32_t pid) {
+        auto wit = written.find(slot);
+        int lvl = (wit == written.end()) ? 0 : wit->second;
+        if (lvl >= 3 || (reply >= 0 && touched.count(reply))) {
+            close();
+            lvl = 0;
+        }
+        auto it = latest.find(node);
+        bool in_l3 = (it != latest.end() && it->second.first == 3);
+        int wl;
+        if (in_l3) {
+            l3[it->second.second].second.push_back({slot, reply, pid});
+            wl = 3;
+        } else if (lvl >= 2) {
+            auto rit = rows3.find(node);
+            int32_t ri;
+            if (rit == rows3.end()) {
+                ri = (int32_t)l3.size();
+                rows3[node] = ri;
+                l3.emplace_back(node, std::vector<Ev>{});
+            } else {
+                ri = rit->second;
+            }
+            l3[ri].second.push_back({slot, reply, pid});
+            latest[node] = {3, ri};
+            wl = 3;
+        } else {
+            auto rit = rows2.find(node);
+            int32_t ri;
+            if (rit == rows2.end()) {
+                ri = (int32_t)l2.size();
+                rows2[node] = ri;
+                l2.emplace_back(node, std::vector<Ev>{});
+                latest[node] = {2, ri};
+            } else {
+                ri = rit->second;
+            }
+            l2[ri].second.push_back({slot, reply, pid});
+            wl = 2;
+        }
+        touched.insert(slot);
+        if (reply >= 0) {
+            written[reply] = wl;
+            touched.insert(reply);
+        }
+    };
+
+    for (int64_t t = 0; t < delta; ++t) {
+        for (int32_t i = snap_tptr[t]; i < snap_tptr[t + 1]; ++i)
+            place_snap(snap_nodes[i], snap_slots[i]);
+        for (int32_t r = recv_tptr[t]; r < recv_tptr[t + 1]; ++r) {
+            int32_t x = recv_nodes[r];
+            for (int32_t d = recv_nptr[r]; d < recv_nptr[r + 1]; ++d) {
+                int32_t pid =
+                    (has_pid && !del_pids.empty()) ? del_pids[d] : -1;
+                place_delivery(x, del_slots[d], reply_slots[d], pid);
+            }
+        }
+        for (int32_t i = pull_tptr[t]; i < pull_tptr[t + 1]; ++i)
+            place_snap(pull_nodes[i], pull_slots[i]);
+        for (int32_t r = rep_tptr[t]; r < rep_tptr[t + 1]; ++r) {
+            int32_t x = rep_nodes[r];
+            for (int32_t d = rep_nptr[r]; d < rep_nptr[r + 1]; ++d) {
+                int32_t pid =
+                    (has_pid && !rep_pids.empty()) ? rep_pids[d] : -1;
+                place_delivery(x, rep_slots[d], -1, pid);
+            }
+        }
+    }
+    if (!g_snap.empty() || !l2.empty() || !l3.empty()) close();
+
+    int32_t n_groups = (int32_t)o.snap_tptr.size() - 1;
+    o.pull_tptr.assign((size_t)n_groups + 1, 0);
+    py::dict pd;
+    auto arr = [](std::vector<int32_t>& v) {
+        auto a = py::array_t<int32_t>((py::ssize_t)v.size());
+        std::copy(v.begin(), v.end(), a.mutable_data());
+        return a;
+    };
+    std::vector<int32_t> empty;
+    pd["snap_nodes"] = arr(o.snap_nodes);
+    pd["snap_slots"] = arr(o.snap_slots);
+    pd["snap_tptr"] = arr(o.snap_tptr);
+    pd["recv_nodes"] = arr(o.recv_nodes);
+    pd["recv_nptr"] = arr(o.recv_nptr);
+    pd["recv_tptr"] = arr(o.recv_tptr);
+    pd["del_slots"] = arr(o.del_slots);
+    pd["reply_slots"] = arr(o.reply_slots);
+    pd["del_pids"] = arr(o.del_pids);
+    pd["pull_nodes"] = arr(empty);
+    pd["pull_slots"] = arr(empty);
+    pd["pull_tptr"] = arr(o.pull_tptr);
+    pd["rep_nodes"] = arr(o.rep_nodes);
+    pd["rep_nptr"] = arr(o.rep_nptr);
+    pd["rep_tptr"] = arr(o.rep_tptr);
+    pd["rep_slots"] = arr(o.rep_slots);
+    pd["rep_reply_slots"] = arr(o.rep_reply_slots);
+    pd["rep_pids"] = arr(o.rep_pids);
+    return pd;
 }
 
 class NativeScheduler {
@@ -550,6 +748,11 @@ py::dict NativeScheduler::next_round(int64_t r)
             recv_nptr, recv_tptr, del_slots, reply_slots, pull_nodes,
             pull_slots, pull_tptr, rep_nodes, rep_nptr, rep_tptr, rep_slots);
         out["merge_bounds"] = arr(mb);
+        out["packed"] = pack_round(
+            snap_nodes, snap_slots, snap_tptr, recv_nodes, recv_nptr,
+            recv_tptr, del_slots, reply_slots, del_pids, pull_nodes,
+            pull_slots, pull_tptr, rep_nodes, rep_nptr, rep_tptr, rep_slots,
+            rep_pids);
     }
     if (sampling_eval_ > 0) {
         Stream g(stream_key(seed_, EVAL, (uint64_t)(t1 - 1), 0));
@@ -1015,6 +1218,11 @@ py::dict NativeTokenizedScheduler::next_round(int64_t r)
             recv_nptr, recv_tptr, del_slots, reply_slots, pull_nodes,
             pull_slots, pull_tptr, rep_nodes, rep_nptr, rep_tptr, rep_slots);
         out["merge_bounds"] = arr(mb);
+        out["packed"] = pack_round(
+            snap_nodes, snap_slots, snap_tptr, recv_nodes, recv_nptr,
+            recv_tptr, del_slots, reply_slots, del_pids, pull_nodes,
+            pull_slots, pull_tptr, rep_nodes, rep_nptr, rep_tptr, rep_slots,
+            rep_pids);
     }
     if (sampling_eval_ > 0) {
         Stream g(stream_key(seed_, EVAL, (uint64_t)(t1 - 1), 0));

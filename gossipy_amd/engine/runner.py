@@ -770,24 +770,32 @@ class BatchedGossipSimulator(SimulationEventSender):
                 )
         return out
 
-    def _maybe_merge(self, f: dict) -> dict:
-        """Apply launch-group merging unless disabled (``GOSSIPY_NO_MERGE=1``).
-        Tracks a cumulative (ticks, groups) counter for the perf probes."""
+    def _maybe_merge(self, f: dict, pack: bool = True) -> dict:
+        """Apply launch-group packing/merging unless disabled
+        (``GOSSIPY_NO_MERGE=1`` turns everything off, ``GOSSIPY_NO_PACK=1``
+        falls back to tick-level merging). ``pack=False`` callers (the
+        multi-rank path, which needs the owner arrays packing drops) get
+        tick-level merging only. Tracks a cumulative (ticks, groups)
+        counter for the perf probes."""
         if os.environ.get("GOSSIPY_NO_MERGE") == "1":
             return f
-        if (
-            os.environ.get("GOSSIPY_PACK") == "1"
+        pack = (
+            pack
+            and os.environ.get("GOSSIPY_NO_PACK") != "1"
             and os.environ.get("GOSSIPY_COOP") != "1"
-        ):
-            # entry-level packing (python reference packer — A/B only; the
-            # C++ packer in the scheduler is the production path)
-            out = self._pack_flat(f)
+            and not os.environ.get("GOSSIPY_SB_MAX")
+        )
+        packed = f.get("packed") if pack else None
+        if packed is None and pack and os.environ.get("GOSSIPY_PACK") == "1":
+            # python reference packer (A/B + flats without the C++ packer)
+            packed = self._pack_flat(f)
+        if packed is not None:
             t = getattr(self, "merge_stats", (0, 0))
             self.merge_stats = (
                 t[0] + len(f["snap_tptr"]) - 1,
-                t[1] + len(out["snap_tptr"]) - 1,
+                t[1] + len(packed["snap_tptr"]) - 1,
             )
-            return out
+            return packed
         mb = f.get("merge_bounds")
         if mb is not None and len(mb) - 1 < len(f["snap_tptr"]) - 1:
             # native scheduler pre-computed the group boundaries in C++
@@ -1347,7 +1355,7 @@ class BatchedGossipSimulator(SimulationEventSender):
                     flat = getattr(self.scheduler, "last_flat", None)
                     if flat is None:
                         flat = self._flatten_phases(sched.ticks)
-                    self._run_round_multi(self._maybe_merge(flat))
+                    self._run_round_multi(self._maybe_merge(flat, pack=False))
                 elif (
                     self._flat_schedulable
                     and self._flat_exec_ok()

@@ -214,3 +214,27 @@ def test_merge_bounds_match_python(protocol):
         # repeat (empty ticks), so compare the subsampled tptrs instead
         for k in ("snap_tptr", "recv_tptr", "pull_tptr", "rep_tptr"):
             np.testing.assert_array_equal(f[k][got], m[k])
+
+
+@pytest.mark.parametrize("protocol", [AntiEntropyProtocol.PUSH,
+                                      AntiEntropyProtocol.PUSH_PULL])
+def test_packed_matches_python_packer(protocol):
+    """The C++ entry-level packer must emit exactly what the python
+    reference packer emits."""
+    from gossipy_amd.engine.runner import BatchedGossipSimulator
+
+    cfg = EngineConfig(
+        n_nodes=120, model_size=10, protocol=protocol,
+        delay=UniformDelay(0, 7), drop_prob=0.15, online_prob=0.9,
+        delta=90, seed=13,
+    )
+    nat = NativeSchedulerAdapter(cfg)
+    for r in range(4):
+        nat.next_round_flat(r)
+        f = nat.last_flat
+        want = BatchedGossipSimulator._pack_flat(f)
+        got = f["packed"]
+        for k, v in want.items():
+            if k == "eval_nodes":
+                continue
+            np.testing.assert_array_equal(np.asarray(got[k]), v, err_msg=k)
