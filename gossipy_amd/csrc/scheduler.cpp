@@ -209,12 +209,16 @@ static py::dict pack_round(
     const std::vector<int32_t>& pull_tptr,
     const std::vector<int32_t>& rep_nodes, const std::vector<int32_t>& rep_nptr,
     const std::vector<int32_t>& rep_tptr, const std::vector<int32_t>& rep_slots,
-    const std::vector<int32_t>& rep_pids)
+    const std::vector<int32_t>& rep_pids, int64_t n_nodes, int64_t n_slots)
 {
     int64_t delta = (int64_t)snap_tptr.size() - 1;
     bool has_pid = !del_pids.empty() || !rep_pids.empty();
 
-    struct Ev { int32_t slot, reply, pid; };
+    // events live in ONE growing pool chained per row (no per-row heap
+    // allocations — at 10k nodes the old vector<Ev>-per-row design cost
+    // ~1 ms/round in allocator traffic)
+    struct Ev { int32_t slot, reply, pid, next; };
+    struct Row { int32_t node, head, tail; };
     struct Out {
         std::vector<int32_t> snap_nodes, snap_slots, snap_tptr{0};
         std::vector<int32_t> recv_nodes, recv_nptr{0}, recv_tptr{0};
@@ -223,13 +227,37 @@ static py::dict pack_round(
         std::vector<int32_t> rep_slots, rep_reply_slots, rep_pids;
         std::vector<int32_t> pull_tptr{};
     } o;
+    o.snap_nodes.reserve(snap_nodes.size() + pull_nodes.size());
+    o.snap_slots.reserve(snap_slots.size() + pull_slots.size());
+    o.recv_nodes.reserve(recv_nodes.size());
+    o.del_slots.reserve(del_slots.size() + rep_slots.size());
+    o.reply_slots.reserve(del_slots.size() + rep_slots.size());
+
+    // group-local state in epoch-stamped flat arrays (stamp = group id):
+    // no per-group clearing, no hashing — O(1) lookups at ~20k events/ms
+    int32_t n_slots_cap = (int32_t)n_slots;
+    int32_t n_nodes_cap = (int32_t)n_nodes;
+    std::vector<int32_t> t_stamp(n_slots_cap, -1);           // touched
+    std::vector<int32_t> w_stamp(n_slots_cap, -1);           // written
+    std::vector<int8_t> w_lvl(n_slots_cap, 0);
+    std::vector<int32_t> l_stamp(n_nodes_cap, -1);           // latest row
+    std::vector<int8_t> l_kind(n_nodes_cap, 0);
+    std::vector<int32_t> l_idx(n_nodes_cap, 0);
+    std::vector<int32_t> r2_stamp(n_nodes_cap, -1), r2_idx(n_nodes_cap, 0);
+    std::vector<int32_t> r3_stamp(n_nodes_cap, -1), r3_idx(n_nodes_cap, 0);
+    int32_t gid = 0;
 
     std::vector<std::pair<int32_t, int32_t>> g_snap;
-    std::unordered_map<int32_t, int32_t> rows2, rows3;
-    std::vector<std::pair<int32_t, std::vector<Ev>>> l2, l3;
-    std::unordered_map<int32_t, std::pair<char, int32_t>> latest;
-    std::unordered_map<int32_t, int32_t> written;
-    std::unordered_set<int32_t> touched;
+    std::vector<Ev> pool;
+    pool.reserve(del_slots.size() + rep_slots.size() + 16);
+    std::vector<Row> l2, l3;
+    auto append_ev = [&](Row& row, int32_t slot, int32_t reply, int32_t pid) {
+        int32_t idx = (int32_t)pool.size();
+        pool.push_back({slot, reply, pid, -1});
+        if (row.tail >= 0) pool[row.tail].next = idx;
+        else row.head = idx;
+        row.tail = idx;
+    };
 
     auto close = [&]() {
         for (auto& p : g_snap) {
@@ -238,96 +266,91 @@ static py::dict pack_round(
         }
         o.snap_tptr.push_back((int32_t)o.snap_nodes.size());
         for (auto& row : l2) {
-            o.recv_nodes.push_back(row.first);
-            for (auto& e : row.second) {
-                o.del_slots.push_back(e.slot);
-                o.reply_slots.push_back(e.reply);
-                if (has_pid) o.del_pids.push_back(e.pid);
+            o.recv_nodes.push_back(row.node);
+            for (int32_t e = row.head; e >= 0; e = pool[e].next) {
+                o.del_slots.push_back(pool[e].slot);
+                o.reply_slots.push_back(pool[e].reply);
+                if (has_pid) o.del_pids.push_back(pool[e].pid);
             }
             o.recv_nptr.push_back((int32_t)o.del_slots.size());
         }
         o.recv_tptr.push_back((int32_t)o.recv_nodes.size());
         for (auto& row : l3) {
-            o.rep_nodes.push_back(row.first);
-            for (auto& e : row.second) {
-                o.rep_slots.push_back(e.slot);
-                o.rep_reply_slots.push_back(e.reply);
-                if (has_pid) o.rep_pids.push_back(e.pid);
+            o.rep_nodes.push_back(row.node);
+            for (int32_t e = row.head; e >= 0; e = pool[e].next) {
+                o.rep_slots.push_back(pool[e].slot);
+                o.rep_reply_slots.push_back(pool[e].reply);
+                if (has_pid) o.rep_pids.push_back(pool[e].pid);
             }
             o.rep_nptr.push_back((int32_t)o.rep_slots.size());
         }
         o.rep_tptr.push_back((int32_t)o.rep_nodes.size());
-        g_snap.clear(); rows2.clear(); rows3.clear();
-        l2.clear(); l3.clear(); latest.clear();
-        written.clear(); touched.clear();
+        g_snap.clear(); l2.clear(); l3.clear();
+        ++gid;  // invalidates every stamped entry at once
     };
 
     auto place_snap = [&](int32_t node, int32_t slot) {
-        if (touched.count(slot)) close();
-        auto it = latest.find(node);
-        if (it == latest.end()) {
+        if (t_stamp[slot] == gid) close();
+        if (l_stamp[node] != gid) {
             g_snap.emplace_back(node, slot);
-            written[slot] = 1;
+            w_stamp[slot] = gid; w_lvl[slot] = 1;
         } else {
-            auto& lst = (it->second.first == 2) ? l2 : l3;
-            auto& evs = lst[it->second.second].second;
-            if (evs.back().reply >= 0) {
+            auto& lst = (l_kind[node] == 2) ? l2 : l3;
+            Ev& last = pool[lst[l_idx[node]].tail];
+            if (last.reply >= 0) {
                 close();
                 g_snap.emplace_back(node, slot);
-                written[slot] = 1;
+                w_stamp[slot] = gid; w_lvl[slot] = 1;
             } else {
-                evs.back().reply = slot;
-                written[slot] = (it->second.first == 2) ? 2 : 3;
+                last.reply = slot;
+                w_stamp[slot] = gid;
+                w_lvl[slot] = (l_kind[node] == 2) ? 2 : 3;
             }
         }
-        touched.insert(slot);
+        t_stamp[slot] = gid;
     };
 
     auto place_delivery = [&](int32_t node, int32_t slot, int32_t reply,
                               int32_t pid) {
-        auto wit = written.find(slot);
-        int lvl = (wit == written.end()) ? 0 : wit->second;
-        if (lvl >= 3 || (reply >= 0 && touched.count(reply))) {
+        int lvl = (w_stamp[slot] == gid) ? w_lvl[slot] : 0;
+        if (lvl >= 3 || (reply >= 0 && t_stamp[reply] == gid)) {
             close();
             lvl = 0;
         }
-        auto it = latest.find(node);
-        bool in_l3 = (it != latest.end() && it->second.first == 3);
+        bool in_l3 = (l_stamp[node] == gid && l_kind[node] == 3);
         int wl;
         if (in_l3) {
-            l3[it->second.second].second.push_back({slot, reply, pid});
+            append_ev(l3[l_idx[node]], slot, reply, pid);
             wl = 3;
         } else if (lvl >= 2) {
-            auto rit = rows3.find(node);
             int32_t ri;
-            if (rit == rows3.end()) {
+            if (r3_stamp[node] != gid) {
                 ri = (int32_t)l3.size();
-                rows3[node] = ri;
-                l3.emplace_back(node, std::vector<Ev>{});
+                r3_stamp[node] = gid; r3_idx[node] = ri;
+                l3.push_back({node, -1, -1});
             } else {
-                ri = rit->second;
+                ri = r3_idx[node];
             }
-            l3[ri].second.push_back({slot, reply, pid});
-            latest[node] = {3, ri};
+            append_ev(l3[ri], slot, reply, pid);
+            l_stamp[node] = gid; l_kind[node] = 3; l_idx[node] = ri;
             wl = 3;
         } else {
-            auto rit = rows2.find(node);
             int32_t ri;
-            if (rit == rows2.end()) {
+            if (r2_stamp[node] != gid) {
                 ri = (int32_t)l2.size();
-                rows2[node] = ri;
-                l2.emplace_back(node, std::vector<Ev>{});
-                latest[node] = {2, ri};
+                r2_stamp[node] = gid; r2_idx[node] = ri;
+                l2.push_back({node, -1, -1});
+                l_stamp[node] = gid; l_kind[node] = 2; l_idx[node] = ri;
             } else {
-                ri = rit->second;
+                ri = r2_idx[node];
             }
-            l2[ri].second.push_back({slot, reply, pid});
+            append_ev(l2[ri], slot, reply, pid);
             wl = 2;
         }
-        touched.insert(slot);
+        t_stamp[slot] = gid;
         if (reply >= 0) {
-            written[reply] = wl;
-            touched.insert(reply);
+            w_stamp[reply] = gid; w_lvl[reply] = (int8_t)wl;
+            t_stamp[reply] = gid;
         }
     };
 
@@ -752,7 +775,7 @@ py::dict NativeScheduler::next_round(int64_t r)
             snap_nodes, snap_slots, snap_tptr, recv_nodes, recv_nptr,
             recv_tptr, del_slots, reply_slots, del_pids, pull_nodes,
             pull_slots, pull_tptr, rep_nodes, rep_nptr, rep_tptr, rep_slots,
-            rep_pids);
+            rep_pids, n_, next_slot_);
     }
     if (sampling_eval_ > 0) {
         Stream g(stream_key(seed_, EVAL, (uint64_t)(t1 - 1), 0));
@@ -1222,7 +1245,7 @@ py::dict NativeTokenizedScheduler::next_round(int64_t r)
             snap_nodes, snap_slots, snap_tptr, recv_nodes, recv_nptr,
             recv_tptr, del_slots, reply_slots, del_pids, pull_nodes,
             pull_slots, pull_tptr, rep_nodes, rep_nptr, rep_tptr, rep_slots,
-            rep_pids);
+            rep_pids, n_, next_slot_);
     }
     if (sampling_eval_ > 0) {
         Stream g(stream_key(seed_, EVAL, (uint64_t)(t1 - 1), 0));
