@@ -209,22 +209,26 @@ static py::dict pack_round(
     const std::vector<int32_t>& pull_tptr,
     const std::vector<int32_t>& rep_nodes, const std::vector<int32_t>& rep_nptr,
     const std::vector<int32_t>& rep_tptr, const std::vector<int32_t>& rep_slots,
-    const std::vector<int32_t>& rep_pids, int64_t n_nodes, int64_t n_slots)
+    const std::vector<int32_t>& rep_pids,
+    const std::vector<int32_t>& del_owners,
+    const std::vector<int32_t>& rep_owners,
+    int64_t n_nodes, int64_t n_slots)
 {
     int64_t delta = (int64_t)snap_tptr.size() - 1;
     bool has_pid = !del_pids.empty() || !rep_pids.empty();
+    bool has_own = !del_owners.empty() || !rep_owners.empty();
 
     // events live in ONE growing pool chained per row (no per-row heap
     // allocations — at 10k nodes the old vector<Ev>-per-row design cost
     // ~1 ms/round in allocator traffic)
-    struct Ev { int32_t slot, reply, pid, next; };
+    struct Ev { int32_t slot, reply, pid, own, next; };
     struct Row { int32_t node, head, tail; };
     struct Out {
         std::vector<int32_t> snap_nodes, snap_slots, snap_tptr{0};
         std::vector<int32_t> recv_nodes, recv_nptr{0}, recv_tptr{0};
-        std::vector<int32_t> del_slots, reply_slots, del_pids;
+        std::vector<int32_t> del_slots, reply_slots, del_pids, del_owners;
         std::vector<int32_t> rep_nodes, rep_nptr{0}, rep_tptr{0};
-        std::vector<int32_t> rep_slots, rep_reply_slots, rep_pids;
+        std::vector<int32_t> rep_slots, rep_reply_slots, rep_pids, rep_owners;
         std::vector<int32_t> pull_tptr{};
     } o;
     o.snap_nodes.reserve(snap_nodes.size() + pull_nodes.size());
@@ -251,9 +255,10 @@ static py::dict pack_round(
     std::vector<Ev> pool;
     pool.reserve(del_slots.size() + rep_slots.size() + 16);
     std::vector<Row> l2, l3;
-    auto append_ev = [&](Row& row, int32_t slot, int32_t reply, int32_t pid) {
+    auto append_ev = [&](Row& row, int32_t slot, int32_t reply, int32_t pid,
+                         int32_t own) {
         int32_t idx = (int32_t)pool.size();
-        pool.push_back({slot, reply, pid, -1});
+        pool.push_back({slot, reply, pid, own, -1});
         if (row.tail >= 0) pool[row.tail].next = idx;
         else row.head = idx;
         row.tail = idx;
@@ -271,6 +276,7 @@ static py::dict pack_round(
                 o.del_slots.push_back(pool[e].slot);
                 o.reply_slots.push_back(pool[e].reply);
                 if (has_pid) o.del_pids.push_back(pool[e].pid);
+                if (has_own) o.del_owners.push_back(pool[e].own);
             }
             o.recv_nptr.push_back((int32_t)o.del_slots.size());
         }
@@ -281,6 +287,7 @@ static py::dict pack_round(
                 o.rep_slots.push_back(pool[e].slot);
                 o.rep_reply_slots.push_back(pool[e].reply);
                 if (has_pid) o.rep_pids.push_back(pool[e].pid);
+                if (has_own) o.rep_owners.push_back(pool[e].own);
             }
             o.rep_nptr.push_back((int32_t)o.rep_slots.size());
         }
@@ -311,7 +318,7 @@ static py::dict pack_round(
     };
 
     auto place_delivery = [&](int32_t node, int32_t slot, int32_t reply,
-                              int32_t pid) {
+                              int32_t pid, int32_t own) {
         int lvl = (w_stamp[slot] == gid) ? w_lvl[slot] : 0;
         if (lvl >= 3 || (reply >= 0 && t_stamp[reply] == gid)) {
             close();
@@ -320,7 +327,7 @@ static py::dict pack_round(
         bool in_l3 = (l_stamp[node] == gid && l_kind[node] == 3);
         int wl;
         if (in_l3) {
-            append_ev(l3[l_idx[node]], slot, reply, pid);
+            append_ev(l3[l_idx[node]], slot, reply, pid, own);
             wl = 3;
         } else if (lvl >= 2) {
             int32_t ri;
@@ -331,7 +338,7 @@ static py::dict pack_round(
             } else {
                 ri = r3_idx[node];
             }
-            append_ev(l3[ri], slot, reply, pid);
+            append_ev(l3[ri], slot, reply, pid, own);
             l_stamp[node] = gid; l_kind[node] = 3; l_idx[node] = ri;
             wl = 3;
         } else {
@@ -344,7 +351,7 @@ static py::dict pack_round(
             } else {
                 ri = r2_idx[node];
             }
-            append_ev(l2[ri], slot, reply, pid);
+            append_ev(l2[ri], slot, reply, pid, own);
             wl = 2;
         }
         t_stamp[slot] = gid;
@@ -362,7 +369,9 @@ static py::dict pack_round(
             for (int32_t d = recv_nptr[r]; d < recv_nptr[r + 1]; ++d) {
                 int32_t pid =
                     (has_pid && !del_pids.empty()) ? del_pids[d] : -1;
-                place_delivery(x, del_slots[d], reply_slots[d], pid);
+                int32_t own =
+                    (has_own && !del_owners.empty()) ? del_owners[d] : -1;
+                place_delivery(x, del_slots[d], reply_slots[d], pid, own);
             }
         }
         for (int32_t i = pull_tptr[t]; i < pull_tptr[t + 1]; ++i)
@@ -372,7 +381,9 @@ static py::dict pack_round(
             for (int32_t d = rep_nptr[r]; d < rep_nptr[r + 1]; ++d) {
                 int32_t pid =
                     (has_pid && !rep_pids.empty()) ? rep_pids[d] : -1;
-                place_delivery(x, rep_slots[d], -1, pid);
+                int32_t own =
+                    (has_own && !rep_owners.empty()) ? rep_owners[d] : -1;
+                place_delivery(x, rep_slots[d], -1, pid, own);
             }
         }
     }
@@ -396,6 +407,7 @@ static py::dict pack_round(
     pd["del_slots"] = arr(o.del_slots);
     pd["reply_slots"] = arr(o.reply_slots);
     pd["del_pids"] = arr(o.del_pids);
+    pd["del_owners"] = arr(o.del_owners);
     pd["pull_nodes"] = arr(empty);
     pd["pull_slots"] = arr(empty);
     pd["pull_tptr"] = arr(o.pull_tptr);
@@ -405,6 +417,7 @@ static py::dict pack_round(
     pd["rep_slots"] = arr(o.rep_slots);
     pd["rep_reply_slots"] = arr(o.rep_reply_slots);
     pd["rep_pids"] = arr(o.rep_pids);
+    pd["rep_owners"] = arr(o.rep_owners);
     return pd;
 }
 
@@ -775,7 +788,7 @@ py::dict NativeScheduler::next_round(int64_t r)
             snap_nodes, snap_slots, snap_tptr, recv_nodes, recv_nptr,
             recv_tptr, del_slots, reply_slots, del_pids, pull_nodes,
             pull_slots, pull_tptr, rep_nodes, rep_nptr, rep_tptr, rep_slots,
-            rep_pids, n_, next_slot_);
+            rep_pids, del_owners, rep_owners, n_, next_slot_);
     }
     if (sampling_eval_ > 0) {
         Stream g(stream_key(seed_, EVAL, (uint64_t)(t1 - 1), 0));
@@ -1245,7 +1258,7 @@ py::dict NativeTokenizedScheduler::next_round(int64_t r)
             snap_nodes, snap_slots, snap_tptr, recv_nodes, recv_nptr,
             recv_tptr, del_slots, reply_slots, del_pids, pull_nodes,
             pull_slots, pull_tptr, rep_nodes, rep_nptr, rep_tptr, rep_slots,
-            rep_pids, n_, next_slot_);
+            rep_pids, del_owners, rep_owners, n_, next_slot_);
     }
     if (sampling_eval_ > 0) {
         Stream g(stream_key(seed_, EVAL, (uint64_t)(t1 - 1), 0));

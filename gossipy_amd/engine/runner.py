@@ -425,7 +425,8 @@ class BatchedGossipSimulator(SimulationEventSender):
         )
         drep = _deliver_prep(
             "rep_nodes", "rep_nptr", "rep_tptr", "rep_slots",
-            None, "rep_pids", "rep_owners",
+            "rep_reply_slots" if len(f.get("rep_reply_slots", ())) else None,
+            "rep_pids", "rep_owners",
         )
 
         def _launch(g, t, reply_default=-1):
@@ -773,10 +774,10 @@ class BatchedGossipSimulator(SimulationEventSender):
     def _maybe_merge(self, f: dict, pack: bool = True) -> dict:
         """Apply launch-group packing/merging unless disabled
         (``GOSSIPY_NO_MERGE=1`` turns everything off, ``GOSSIPY_NO_PACK=1``
-        falls back to tick-level merging). ``pack=False`` callers (the
-        multi-rank path, which needs the owner arrays packing drops) get
-        tick-level merging only. Tracks a cumulative (ticks, groups)
-        counter for the perf probes."""
+        falls back to tick-level merging). Packed schedules carry the
+        owner arrays, so the multi-rank path consumes them too — one
+        batched slot exchange per packed group instead of per tick.
+        Tracks a cumulative (ticks, groups) counter for the perf probes."""
         if os.environ.get("GOSSIPY_NO_MERGE") == "1":
             return f
         pack = (
@@ -941,14 +942,19 @@ class BatchedGossipSimulator(SimulationEventSender):
         pt, qt = f["pull_tptr"], f["rep_tptr"]
         nptr, rep_nptr = f["recv_nptr"], f["rep_nptr"]
         has_pid = len(f.get("del_pids", ())) > 0 or len(f.get("rep_pids", ())) > 0
+        has_own = (
+            len(f.get("del_owners", ())) > 0 or len(f.get("rep_owners", ())) > 0
+        )
 
         out = {
             "snap_nodes": [], "snap_slots": [], "snap_tptr": [0],
             "recv_nodes": [], "recv_nptr": [0], "recv_tptr": [0],
             "del_slots": [], "reply_slots": [], "del_pids": [],
+            "del_owners": [],
             "pull_nodes": [], "pull_slots": [], "pull_tptr": [0],
             "rep_nodes": [], "rep_nptr": [0], "rep_tptr": [0],
             "rep_slots": [], "rep_reply_slots": [], "rep_pids": [],
+            "rep_owners": [],
         }
 
         # current-group state
@@ -967,20 +973,24 @@ class BatchedGossipSimulator(SimulationEventSender):
             out["snap_tptr"].append(len(out["snap_nodes"]))
             for node, evs in l2:
                 out["recv_nodes"].append(node)
-                for s, rsl, pid in evs:
+                for s, rsl, pid, own in evs:
                     out["del_slots"].append(s)
                     out["reply_slots"].append(rsl)
                     if has_pid:
                         out["del_pids"].append(pid)
+                    if has_own:
+                        out["del_owners"].append(own)
                 out["recv_nptr"].append(len(out["del_slots"]))
             out["recv_tptr"].append(len(out["recv_nodes"]))
             for node, evs in l3:
                 out["rep_nodes"].append(node)
-                for s, rsl, pid in evs:
+                for s, rsl, pid, own in evs:
                     out["rep_slots"].append(s)
                     out["rep_reply_slots"].append(rsl)
                     if has_pid:
                         out["rep_pids"].append(pid)
+                    if has_own:
+                        out["rep_owners"].append(own)
                 out["rep_nptr"].append(len(out["rep_slots"]))
             out["rep_tptr"].append(len(out["rep_nodes"]))
             out["pull_tptr"].append(0)
@@ -1003,11 +1013,11 @@ class BatchedGossipSimulator(SimulationEventSender):
                     g_snap.append((node, slot))
                     written[slot] = 1
                 else:
-                    evs[-1] = (evs[-1][0], slot, evs[-1][2])
+                    evs[-1] = (evs[-1][0], slot, evs[-1][2], evs[-1][3])
                     written[slot] = 2 if pos[0] == "l2" else 3
             touched.add(slot)
 
-        def place_delivery(node, slot, reply, pid):
+        def place_delivery(node, slot, reply, pid, own):
             lvl = written.get(slot, 0)
             if lvl >= 3 or (reply >= 0 and reply in touched):
                 close()
@@ -1018,7 +1028,7 @@ class BatchedGossipSimulator(SimulationEventSender):
                     close()
                     pos = None
             if pos is not None and pos[0] == "l3":
-                l3[pos[1]][1].append((slot, reply, pid))
+                l3[pos[1]][1].append((slot, reply, pid, own))
                 wl = 3
             elif lvl >= 2:
                 # source written in this group's first deliver launch:
@@ -1027,7 +1037,7 @@ class BatchedGossipSimulator(SimulationEventSender):
                 if ri is None:
                     rows3[node] = ri = len(l3)
                     l3.append([node, []])
-                l3[ri][1].append((slot, reply, pid))
+                l3[ri][1].append((slot, reply, pid, own))
                 latest[node] = ("l3", ri)
                 wl = 3
             else:
@@ -1036,7 +1046,7 @@ class BatchedGossipSimulator(SimulationEventSender):
                     rows2[node] = ri = len(l2)
                     l2.append([node, []])
                     latest[node] = ("l2", ri)
-                l2[ri][1].append((slot, reply, pid))
+                l2[ri][1].append((slot, reply, pid, own))
                 wl = 2
             touched.add(slot)
             if reply >= 0:
@@ -1047,9 +1057,11 @@ class BatchedGossipSimulator(SimulationEventSender):
         recv_nodes = f["recv_nodes"]
         del_slots, reply_slots = f["del_slots"], f["reply_slots"]
         del_pids = f.get("del_pids")
+        del_owners = f.get("del_owners")
         pull_nodes, pull_slots = f["pull_nodes"], f["pull_slots"]
         rep_nodes, rep_slots = f["rep_nodes"], f["rep_slots"]
         rep_pids = f.get("rep_pids")
+        rep_owners = f.get("rep_owners")
         for t in range(delta):
             for i in range(st[t], st[t + 1]):
                 place_snap(int(snap_nodes[i]), int(snap_slots[i]))
@@ -1057,15 +1069,23 @@ class BatchedGossipSimulator(SimulationEventSender):
                 x = int(recv_nodes[r])
                 for d in range(nptr[r], nptr[r + 1]):
                     pid = int(del_pids[d]) if has_pid and len(del_pids) else -1
+                    own = (
+                        int(del_owners[d])
+                        if has_own and len(del_owners) else -1
+                    )
                     place_delivery(x, int(del_slots[d]),
-                                   int(reply_slots[d]), pid)
+                                   int(reply_slots[d]), pid, own)
             for i in range(pt[t], pt[t + 1]):
                 place_snap(int(pull_nodes[i]), int(pull_slots[i]))
             for r in range(qt[t], qt[t + 1]):
                 x = int(rep_nodes[r])
                 for d in range(rep_nptr[r], rep_nptr[r + 1]):
                     pid = int(rep_pids[d]) if has_pid and len(rep_pids) else -1
-                    place_delivery(x, int(rep_slots[d]), -1, pid)
+                    own = (
+                        int(rep_owners[d])
+                        if has_own and len(rep_owners) else -1
+                    )
+                    place_delivery(x, int(rep_slots[d]), -1, pid, own)
         if g_snap or l2 or l3:
             close()
         res = {}
@@ -1355,7 +1375,7 @@ class BatchedGossipSimulator(SimulationEventSender):
                     flat = getattr(self.scheduler, "last_flat", None)
                     if flat is None:
                         flat = self._flatten_phases(sched.ticks)
-                    self._run_round_multi(self._maybe_merge(flat, pack=False))
+                    self._run_round_multi(self._maybe_merge(flat))
                 elif (
                     self._flat_schedulable
                     and self._flat_exec_ok()
