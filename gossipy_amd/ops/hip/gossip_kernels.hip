@@ -1252,6 +1252,7 @@ struct EvalArgs {
     float* out;           // [R, 5] acc, prec, rec, f1, auc
     int d, k, n_eval, D;
     int is_margin;        // pegasos/adaline: score = <w, x>, label +-1
+    int tile;             // samples staged per LDS tile (coalesced loads)
 };
 
 constexpr int EVAL_KMAX = 16;
@@ -1265,7 +1266,9 @@ eval_metrics_kernel(EvalArgs a)
     extern __shared__ float sm[];
     float* W = sm;                       // D
     float* s1 = W + a.D;                 // n_eval: positive-class score
-    int* conf = (int*)(s1 + a.n_eval);   // k*k confusion (true x pred)
+    char* yb = (char*)(s1 + a.n_eval);   // n_eval: binarized label
+    int* conf = (int*)(yb + ((a.n_eval + 15) & ~15));  // k*k confusion
+    float* XT = (float*)(conf + EVAL_KMAX * EVAL_KMAX);  // tile x d
     int kk = a.is_margin ? 2 : a.k;
 
     for (int e = tid; e < a.D; e += blockDim.x)
@@ -1273,35 +1276,46 @@ eval_metrics_kernel(EvalArgs a)
     for (int e = tid; e < kk * kk; e += blockDim.x) conf[e] = 0;
     __syncthreads();
 
-    for (int sidx = tid; sidx < a.n_eval; sidx += blockDim.x) {
-        const float* x = a.X + (long)sidx * a.d;
-        int pred, yt;
-        float sc1;
-        if (a.is_margin) {
-            float acc = 0.f;
-            for (int e = 0; e < a.d; ++e) acc += W[e] * x[e];
-            sc1 = acc;
-            pred = acc >= 0.f ? 1 : 0;
-            yt = a.Y[sidx] > 0.f ? 1 : 0;
-        } else {
-            float best = -1e30f;
-            int bj = 0;
-            float z1 = 0.f;
-            for (int j = 0; j < a.k; ++j) {
-                float acc = W[a.k * a.d + j];
-                const float* wrow = W + j * a.d;
-                for (int e = 0; e < a.d; ++e) acc += wrow[e] * x[e];
-                if (acc > best) { best = acc; bj = j; }
-                if (j == 1) z1 = acc;
+    // samples staged through LDS in coalesced tiles (the old thread-per-
+    // sample global gather was a stride-d access pattern and dominated
+    // the kernel's 44 us)
+    for (int s0 = 0; s0 < a.n_eval; s0 += a.tile) {
+        int m = min(a.tile, a.n_eval - s0);
+        for (int e = tid; e < m * a.d; e += blockDim.x)
+            XT[e] = a.X[(long)s0 * a.d + e];
+        __syncthreads();
+        for (int sidx = s0 + tid; sidx < s0 + m; sidx += blockDim.x) {
+            const float* x = XT + (long)(sidx - s0) * a.d;
+            int pred, yt;
+            float sc1;
+            if (a.is_margin) {
+                float acc = 0.f;
+                for (int e = 0; e < a.d; ++e) acc += W[e] * x[e];
+                sc1 = acc;
+                pred = acc >= 0.f ? 1 : 0;
+                yt = a.Y[sidx] > 0.f ? 1 : 0;
+            } else {
+                float best = -1e30f;
+                int bj = 0;
+                float z1 = 0.f;
+                for (int j = 0; j < a.k; ++j) {
+                    float acc = W[a.k * a.d + j];
+                    const float* wrow = W + j * a.d;
+                    for (int e = 0; e < a.d; ++e) acc += wrow[e] * x[e];
+                    if (acc > best) { best = acc; bj = j; }
+                    if (j == 1) z1 = acc;
+                }
+                sc1 = z1;
+                pred = bj;
+                yt = (int)a.Y[sidx];
             }
-            sc1 = z1;
-            pred = bj;
-            yt = (int)a.Y[sidx];
+            s1[sidx] = sc1;
+            yb[sidx] = (char)(a.is_margin ? (a.Y[sidx] > 0.f ? 1 : 0)
+                                          : ((int)a.Y[sidx] == 1 ? 1 : 0));
+            atomicAdd(&conf[yt * kk + pred], 1);
         }
-        s1[sidx] = sc1;
-        atomicAdd(&conf[yt * kk + pred], 1);
+        __syncthreads();
     }
-    __syncthreads();
 
     // pairwise AUC (binary only): wins over (pos, neg) pairs, ties 0.5 —
     // the Mann-Whitney statistic (== average-rank AUC)
@@ -1313,12 +1327,10 @@ eval_metrics_kernel(EvalArgs a)
         float wins = 0.f;
         int npos_l = 0;
         for (int i = tid; i < a.n_eval; i += blockDim.x) {
-            bool pos_i = a.is_margin ? (a.Y[i] > 0.f) : ((int)a.Y[i] == 1);
-            if (!pos_i) continue;
+            if (!yb[i]) continue;
             npos_l += 1;
             for (int j = 0; j < a.n_eval; ++j) {
-                bool pos_j = a.is_margin ? (a.Y[j] > 0.f) : ((int)a.Y[j] == 1);
-                if (pos_j) continue;
+                if (yb[j]) continue;
                 if (s1[i] > s1[j]) wins += 1.f;
                 else if (s1[i] == s1[j]) wins += 0.5f;
             }
@@ -2478,9 +2490,16 @@ torch::Tensor eval_metrics(torch::Tensor params, torch::Tensor nodes,
     a.out = out.data_ptr<float>();
     a.d = d; a.k = k; a.n_eval = n_eval; a.D = params.size(1);
     a.is_margin = is_margin;
-    int kk = is_margin ? 2 : (int)k;
-    size_t smem = sizeof(float) * (a.D + n_eval) + sizeof(int) * kk * kk;
-    TORCH_CHECK(smem <= 160 * 1024, "eval LDS budget exceeded: ", smem);
+    // LDS: model + scores + binarized labels + confusion + one X tile.
+    // The tile size adapts so everything fits the 160 KB budget.
+    size_t fixed = sizeof(float) * (a.D + n_eval)
+        + ((n_eval + 15) & ~15) * sizeof(char)
+        + sizeof(int) * EVAL_KMAX * EVAL_KMAX;
+    long room = (long)(160 * 1024 - fixed) / (long)(sizeof(float) * d);
+    TORCH_CHECK(room >= 16, "eval LDS budget exceeded (n_eval=", n_eval,
+                ", d=", d, ")");
+    a.tile = (int)std::min<long>({room, (long)n_eval, 512});
+    size_t smem = fixed + sizeof(float) * (size_t)a.tile * d;
     hipLaunchKernelGGL(eval_metrics_kernel, dim3(R), dim3(256), smem,
                        current_stream(), a);
     return out;
