@@ -1335,8 +1335,14 @@ eval_metrics_kernel(EvalArgs a)
                 else if (s1[i] == s1[j]) wins += 0.5f;
             }
         }
-        atomicAdd(&auc_wins, wins);
-        atomicAdd(&npos_s, npos_l);
+        // block reduce via wave shuffles: 256 serialized LDS atomics were
+        // ~1/3 of this kernel's time
+        wins = wave_sum(wins);
+        float np_f = wave_sum((float)npos_l);
+        if ((tid & 63) == 0) {
+            atomicAdd(&auc_wins, wins);
+            atomicAdd(&npos_s, (int)np_f);
+        }
     }
     __syncthreads();
 
