@@ -43,6 +43,8 @@ from .arena import DataArena, NodeStateArena, SlotPool
 from .models import AdaLineSpec, LogRegSpec, MLPSpec, PegasosSpec
 from .rng import Purpose, RandomTape, sample_indices
 
+_EMPTY_I32 = torch.zeros(0, dtype=torch.int32)
+
 __all__ = ["TorchBackend", "HIPBackend", "make_backend"]
 
 _MODE_ID = {
@@ -979,6 +981,42 @@ class HIPBackend(TorchBackend):
             spec.batch_size,
         )
 
+    def eval_local_fast(self, state, spec, local_ids, tx, ty, tcounts):
+        """Per-node test-shard evaluation (the reference's ``evaluate``
+        on the node's own split, gossipy/node.py:206-224) in ONE launch:
+        block r scores node r's model on its own padded shard rows.
+        Returns (metric dicts, kept local_ids) skipping empty shards, or
+        ``None`` for families on the generic path."""
+        if spec.family not in ("logreg", "pegasos", "adaline"):
+            return None
+        is_margin = spec.family in ("pegasos", "adaline")
+        ids32 = local_ids.to(state.params.device, torch.int32)
+        out = self.ext.eval_metrics(
+            state.params,
+            ids32,
+            tx,
+            ty,
+            spec.d_in,
+            1 if is_margin else spec.n_classes,
+            is_margin,
+            tcounts,
+        )
+        vals = out.cpu().numpy()
+        res = []
+        for row in vals:
+            if row[0] <= -2.0:  # empty shard sentinel
+                continue
+            d = {
+                "accuracy": float(row[0]),
+                "precision": float(row[1]),
+                "recall": float(row[2]),
+                "f1_score": float(row[3]),
+            }
+            if row[4] >= 0:
+                d["auc"] = float(row[4])
+            res.append(d)
+        return res
+
     def eval_metrics_launch(self, state, spec, local_ids, gx, gy):
         """Asynchronous half of :meth:`eval_metrics_fast`: enqueue the K13
         kernel and a non-blocking D2H into a pinned staging buffer, record
@@ -997,6 +1035,7 @@ class HIPBackend(TorchBackend):
             spec.d_in,
             1 if is_margin else spec.n_classes,
             is_margin,
+            _EMPTY_I32,
         )
         # double-buffered pinned staging: at most one handle is in flight
         bufs = self._eval_bufs.setdefault(tuple(out.shape), [None, None])
@@ -1045,6 +1084,7 @@ class HIPBackend(TorchBackend):
             spec.d_in,
             1 if is_margin else spec.n_classes,
             is_margin,
+            _EMPTY_I32,
         )
         vals = out.cpu().numpy()
         res = []

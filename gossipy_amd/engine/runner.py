@@ -650,6 +650,17 @@ class BatchedGossipSimulator(SimulationEventSender):
 
         results_local: List[dict] = []
         if self.data.tx is not None and len(mine):
+            fast_local = getattr(self.backend, "eval_local_fast", None)
+            got_local = None
+            if fast_local is not None:
+                # one launch: block r scores node r on its own shard
+                got_local = fast_local(
+                    self.state, self.spec, local_ids,
+                    self.data.tx, self.data.ty, self.data.tcounts,
+                )
+            if got_local is not None:
+                results_local = got_local
+        if self.data.tx is not None and len(mine) and not results_local:
             # per-node test shards: evaluate each node on its own shard
             for li in local_ids.tolist():
                 c = int(self.data.tcounts[li])

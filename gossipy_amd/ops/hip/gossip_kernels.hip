@@ -1253,6 +1253,9 @@ struct EvalArgs {
     int d, k, n_eval, D;
     int is_margin;        // pegasos/adaline: score = <w, x>, label +-1
     int tile;             // samples staged per LDS tile (coalesced loads)
+    // per-node shard mode (local test sets): X/Y are [n_local, n_eval(,d)]
+    // arenas, eval_counts[node] gives the node's row count
+    const int* eval_counts;
 };
 
 constexpr int EVAL_KMAX = 16;
@@ -1263,11 +1266,24 @@ eval_metrics_kernel(EvalArgs a)
     int r = blockIdx.x;
     int node = a.nodes[r];
     int tid = threadIdx.x;
+    int n_e = a.n_eval;
+    const float* Xb = a.X;
+    const float* Yb = a.Y;
+    if (a.eval_counts) {  // per-node shard mode: a.n_eval is the row stride
+        Xb = a.X + (long)node * a.n_eval * a.d;
+        Yb = a.Y + (long)node * a.n_eval;
+        n_e = min(a.eval_counts[node], a.n_eval);
+        if (n_e == 0) {
+            if (tid == 0)
+                for (int e = 0; e < 5; ++e) a.out[(long)r * 5 + e] = -2.f;
+            return;
+        }
+    }
     extern __shared__ float sm[];
     float* W = sm;                       // D
     float* s1 = W + a.D;                 // n_eval: positive-class score
-    char* yb = (char*)(s1 + a.n_eval);   // n_eval: binarized label
-    int* conf = (int*)(yb + ((a.n_eval + 15) & ~15));  // k*k confusion
+    char* yb = (char*)(s1 + n_e);   // n_eval: binarized label
+    int* conf = (int*)(yb + ((n_e + 15) & ~15));  // k*k confusion
     float* XT = (float*)(conf + EVAL_KMAX * EVAL_KMAX);  // tile x d
     int kk = a.is_margin ? 2 : a.k;
 
@@ -1279,10 +1295,10 @@ eval_metrics_kernel(EvalArgs a)
     // samples staged through LDS in coalesced tiles (the old thread-per-
     // sample global gather was a stride-d access pattern and dominated
     // the kernel's 44 us)
-    for (int s0 = 0; s0 < a.n_eval; s0 += a.tile) {
-        int m = min(a.tile, a.n_eval - s0);
+    for (int s0 = 0; s0 < n_e; s0 += a.tile) {
+        int m = min(a.tile, n_e - s0);
         for (int e = tid; e < m * a.d; e += blockDim.x)
-            XT[e] = a.X[(long)s0 * a.d + e];
+            XT[e] = Xb[(long)s0 * a.d + e];
         __syncthreads();
         for (int sidx = s0 + tid; sidx < s0 + m; sidx += blockDim.x) {
             const float* x = XT + (long)(sidx - s0) * a.d;
@@ -1293,7 +1309,7 @@ eval_metrics_kernel(EvalArgs a)
                 for (int e = 0; e < a.d; ++e) acc += W[e] * x[e];
                 sc1 = acc;
                 pred = acc >= 0.f ? 1 : 0;
-                yt = a.Y[sidx] > 0.f ? 1 : 0;
+                yt = Yb[sidx] > 0.f ? 1 : 0;
             } else {
                 float best = -1e30f;
                 int bj = 0;
@@ -1307,11 +1323,11 @@ eval_metrics_kernel(EvalArgs a)
                 }
                 sc1 = z1;
                 pred = bj;
-                yt = (int)a.Y[sidx];
+                yt = (int)Yb[sidx];
             }
             s1[sidx] = sc1;
-            yb[sidx] = (char)(a.is_margin ? (a.Y[sidx] > 0.f ? 1 : 0)
-                                          : ((int)a.Y[sidx] == 1 ? 1 : 0));
+            yb[sidx] = (char)(a.is_margin ? (Yb[sidx] > 0.f ? 1 : 0)
+                                          : ((int)Yb[sidx] == 1 ? 1 : 0));
             atomicAdd(&conf[yt * kk + pred], 1);
         }
         __syncthreads();
@@ -1326,10 +1342,10 @@ eval_metrics_kernel(EvalArgs a)
     if (kk == 2) {
         float wins = 0.f;
         int npos_l = 0;
-        for (int i = tid; i < a.n_eval; i += blockDim.x) {
+        for (int i = tid; i < n_e; i += blockDim.x) {
             if (!yb[i]) continue;
             npos_l += 1;
-            for (int j = 0; j < a.n_eval; ++j) {
+            for (int j = 0; j < n_e; ++j) {
                 if (yb[j]) continue;
                 if (s1[i] > s1[j]) wins += 1.f;
                 else if (s1[i] == s1[j]) wins += 0.5f;
@@ -1347,7 +1363,7 @@ eval_metrics_kernel(EvalArgs a)
     __syncthreads();
 
     if (tid == 0) {
-        int n = a.n_eval;
+        int n = n_e;
         int correct = 0;
         float prec = 0.f, rec = 0.f, f1 = 0.f;
         for (int c = 0; c < kk; ++c) {
@@ -2480,12 +2496,14 @@ void tick_pens(torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
 
 torch::Tensor eval_metrics(torch::Tensor params, torch::Tensor nodes,
                            torch::Tensor X, torch::Tensor Y, int64_t d,
-                           int64_t k, bool is_margin)
+                           int64_t k, bool is_margin,
+                           torch::Tensor eval_counts)
 {
     CHECK_DEV(params); CHECK_DEV(nodes); CHECK_DEV(X); CHECK_DEV(Y);
     TORCH_CHECK(k <= EVAL_KMAX, "n_classes > ", EVAL_KMAX, " unsupported");
     int R = nodes.size(0);
-    int n_eval = X.size(0);
+    bool per_node = eval_counts.defined() && eval_counts.numel();
+    int n_eval = per_node ? (int)X.size(1) : (int)X.size(0);
     auto out = torch::empty({R, 5}, params.options());
     if (R == 0) return out;
     EvalArgs a;
@@ -2496,6 +2514,7 @@ torch::Tensor eval_metrics(torch::Tensor params, torch::Tensor nodes,
     a.out = out.data_ptr<float>();
     a.d = d; a.k = k; a.n_eval = n_eval; a.D = params.size(1);
     a.is_margin = is_margin;
+    a.eval_counts = per_node ? eval_counts.data_ptr<int>() : nullptr;
     // LDS: model + scores + binarized labels + confusion + one X tile.
     // The tile size adapts so everything fits the 160 KB budget.
     size_t fixed = sizeof(float) * (a.D + n_eval)

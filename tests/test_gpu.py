@@ -1197,3 +1197,44 @@ def test_perf_knobs_preserve_results(monkeypatch, knob):
     base = run(False)
     alt = run(True)
     assert torch.equal(base, alt)
+
+
+@pytest.mark.gpu
+def test_batched_local_eval_matches_loop():
+    """eval_local_fast (one launch, block-per-node on its own shard) must
+    reproduce the per-node python loop's metrics."""
+    from gossipy_amd.engine.metrics import classification_metrics_shared
+
+    n = 40
+    X, y = make_synthetic_classification((400, 57, 2), seed=5)
+    shards = [(X[s], y[s]) for s in np.array_split(np.arange(400), n)]
+    # ragged test shards incl. one empty
+    tshards = [
+        (X[s][: max(0, len(s) - (i % 4))], y[s][: max(0, len(s) - (i % 4))])
+        for i, s in enumerate(np.array_split(np.arange(400), n))
+    ]
+    data = DataArena.from_shards(shards, CUDA, test_shards=tshards)
+    spec = LogRegSpec(d_in=57, n_classes=2, lr=0.1)
+    cfg = EngineConfig(
+        n_nodes=n, delta=20, protocol=AntiEntropyProtocol.PUSH,
+        model_size=116, seed=3,
+    )
+    sim = BatchedGossipSimulator(cfg, spec, data, device=CUDA)
+    sim.init_nodes()
+    sim.start(n_rounds=2)
+    local_ids = torch.arange(n)
+    fast = sim.backend.eval_local_fast(
+        sim.state, spec, local_ids, data.tx, data.ty, data.tcounts
+    )
+    slow = []
+    for li in range(n):
+        c = int(data.tcounts[li])
+        if c == 0:
+            continue
+        sc = sim.backend.scores(sim.state, spec, torch.tensor([li]), data.tx[li, :c])
+        slow.extend(classification_metrics_shared(sc, data.ty[li, :c]))
+    assert len(fast) == len(slow)
+    for a, b in zip(fast, slow):
+        assert set(a) == set(b)
+        for k in a:
+            assert abs(a[k] - b[k]) < 1e-4, (k, a[k], b[k])
