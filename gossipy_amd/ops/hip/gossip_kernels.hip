@@ -107,18 +107,20 @@ DEV_INLINE void logreg_update(const LogregArgs& a, int node, float* W,
     int bsz = (a.bs == 0) ? c : min(a.bs, c);
     const float* Xn = a.X + (long)node * a.Smax * a.d;
     const float* Yn = a.Y + (long)node * a.Smax;
+    // single-batch shards (every BASELINE config): xb stays valid across
+    // epochs AND deliveries, so it is staged at most once per receiver row
+    int xb_valid = prestaged;
     for (int ep = 0; ep < a.epochs; ++ep) {
         for (int s0 = 0; s0 < c; s0 += bsz) {
             int m = min(bsz, c - s0);
             // stage the batch in LDS (coalesced: consecutive threads read
             // consecutive floats of the shard); the caller may have staged
             // the very first batch already (fused with the W load)
-            if (prestaged) {
-                prestaged = 0;
-            } else {
+            if (!xb_valid) {
                 for (int e = tid; e < m * a.d; e += blockDim.x)
                     xb[e] = Xn[(long)s0 * a.d + e];
             }
+            xb_valid = (c <= bsz);
             __syncthreads();
             // per-sample forward + dLoss/dz (thread = sample)
             if (tid < m) {
@@ -198,6 +200,27 @@ DEV_INLINE void logreg_process_node(const LogregArgs& a, int i)
             xb[e] = a.X[(long)node * a.Smax * a.d + e];
         __syncthreads();
         int age = max(a.ages[node], a.slot_ages[slot]);
+        // xb holds batch 0 after the fused stage above; it STAYS valid for
+        // later deliveries only when the shard is a single batch
+        int c0 = a.counts[node];
+        int xb_keep = (a.bs == 0) || (c0 <= a.bs);
+        // register double-buffer: issue the NEXT delivery's slot-row loads
+        // before each (long) update so the HBM latency hides under compute
+        constexpr int PRE = 8;
+        float pre[PRE];
+        int pre_age = 0;
+        int npre = min(a.D, PRE * (int)blockDim.x);
+        auto issue_pre = [&](int j) {
+            if (j < j1 && !(a.dmodes && a.dmodes[j])) {
+                const float* srow2 = a.slots + (long)a.dslots[j] * a.D;
+                for (int u = 0; u < PRE; ++u) {
+                    int e = tid + u * (int)blockDim.x;
+                    pre[u] = (e < a.D) ? srow2[e] : 0.f;
+                }
+                pre_age = a.slot_ages[a.dslots[j]];
+            }
+        };
+        issue_pre(j0 + 1);
         logreg_update(a, node, W, xb, dz, age, /*prestaged=*/1);
         int rs0 = a.rslots ? a.rslots[j0] : -1;
         if (rs0 >= 0) {  // first delivery's PUSH_PULL reply snapshot
@@ -209,18 +232,23 @@ DEV_INLINE void logreg_process_node(const LogregArgs& a, int i)
         for (int j = j0 + 1; j < j1; ++j) {
             int slot2 = a.dslots[j];
             const float* srow2 = a.slots + (long)slot2 * a.D;
-            int sage2 = a.slot_ages[slot2];
             int md = (a.dmodes && a.dmodes[j]) ? MODE_PASS : a.mode;
             if (md == MODE_MERGE_UPDATE) {
-                for (int e = tid; e < a.D; e += blockDim.x)
+                for (int u = 0; u < PRE; ++u) {
+                    int e = tid + u * (int)blockDim.x;
+                    if (e < a.D) W[e] = 0.5f * (W[e] + pre[u]);
+                }
+                for (int e = npre + tid; e < a.D; e += blockDim.x)
                     W[e] = 0.5f * (W[e] + srow2[e]);
-                age = max(age, sage2);
+                age = max(age, pre_age);
                 __syncthreads();
-                logreg_update(a, node, W, xb, dz, age);
+                issue_pre(j + 1);
+                logreg_update(a, node, W, xb, dz, age, /*prestaged=*/xb_keep);
             } else {  // PASS (pass-through coin resolved to adopt)
                 for (int e = tid; e < a.D; e += blockDim.x) W[e] = srow2[e];
-                age = sage2;
+                age = a.slot_ages[slot2];
                 __syncthreads();
+                issue_pre(j + 1);
             }
             int rs2 = a.rslots ? a.rslots[j] : -1;
             if (rs2 >= 0) {
