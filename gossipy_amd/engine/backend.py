@@ -1023,20 +1023,29 @@ class HIPBackend(TorchBackend):
         an event, and return a handle for :meth:`eval_metrics_collect`.
         Lets the runner overlap a round's evaluation fetch with the next
         round's compute (the collect usually finds the event already
-        signalled). Returns ``None`` for families on the generic path."""
-        if spec.family not in ("logreg", "pegasos", "adaline"):
+        signalled). Affine/margin families score inside the kernel;
+        MLP/torchmod run their (batched) forward first and feed the
+        precomputed scores to the same metrics epilogue. Returns ``None``
+        for unsupported families."""
+        if spec.family in ("logreg", "pegasos", "adaline"):
+            is_margin = spec.family in ("pegasos", "adaline")
+            out = self.ext.eval_metrics(
+                state.params,
+                local_ids.to(state.params.device, torch.int32),
+                gx,
+                gy,
+                spec.d_in,
+                1 if is_margin else spec.n_classes,
+                is_margin,
+                _EMPTY_I32,
+            )
+        elif spec.family in ("mlp", "torchmod"):
+            sc = self.scores(state, spec, local_ids, gx)
+            out = self.ext.eval_metrics_scores(
+                sc.contiguous(), gy, sc.shape[-1], False
+            )
+        else:
             return None
-        is_margin = spec.family in ("pegasos", "adaline")
-        out = self.ext.eval_metrics(
-            state.params,
-            local_ids.to(state.params.device, torch.int32),
-            gx,
-            gy,
-            spec.d_in,
-            1 if is_margin else spec.n_classes,
-            is_margin,
-            _EMPTY_I32,
-        )
         # double-buffered pinned staging: at most one handle is in flight
         bufs = self._eval_bufs.setdefault(tuple(out.shape), [None, None])
         self._eval_flip = flip = getattr(self, "_eval_flip", 0) ^ 1
@@ -1072,20 +1081,28 @@ class HIPBackend(TorchBackend):
         list of metric dicts, or ``None`` if the family needs the generic
         path. Metrics match gossipy_amd.engine.metrics (sklearn
         conventions); argmax/AUC on raw affine scores — the reference's
-        sigmoid is monotonic so predictions and ranks are identical."""
-        if spec.family not in ("logreg", "pegasos", "adaline"):
+        sigmoid is monotonic so predictions and ranks are identical.
+        MLP/torchmod feed their batched forward's scores to the same
+        kernel epilogue."""
+        if spec.family in ("logreg", "pegasos", "adaline"):
+            is_margin = spec.family in ("pegasos", "adaline")
+            out = self.ext.eval_metrics(
+                state.params,
+                local_ids.to(state.params.device, torch.int32),
+                gx,
+                gy,
+                spec.d_in,
+                1 if is_margin else spec.n_classes,
+                is_margin,
+                _EMPTY_I32,
+            )
+        elif spec.family in ("mlp", "torchmod"):
+            sc = self.scores(state, spec, local_ids, gx)
+            out = self.ext.eval_metrics_scores(
+                sc.contiguous(), gy, sc.shape[-1], False
+            )
+        else:
             return None
-        is_margin = spec.family in ("pegasos", "adaline")
-        out = self.ext.eval_metrics(
-            state.params,
-            local_ids.to(state.params.device, torch.int32),
-            gx,
-            gy,
-            spec.d_in,
-            1 if is_margin else spec.n_classes,
-            is_margin,
-            _EMPTY_I32,
-        )
         vals = out.cpu().numpy()
         res = []
         for row in vals:

@@ -615,7 +615,8 @@ class BatchedGossipSimulator(SimulationEventSender):
             and self.data.gx is not None
             and os.environ.get("GOSSIPY_NO_EVAL_PIPE") != "1"
             and hasattr(self.backend, "eval_metrics_launch")
-            and self.spec.family in ("logreg", "pegasos", "adaline")
+            and self.spec.family in ("logreg", "pegasos", "adaline",
+                                     "mlp", "torchmod")
         ):
             self._drain_eval()
             if len(mine):

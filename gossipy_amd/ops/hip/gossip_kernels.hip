@@ -1284,6 +1284,9 @@ struct EvalArgs {
     // per-node shard mode (local test sets): X/Y are [n_local, n_eval(,d)]
     // arenas, eval_counts[node] gives the node's row count
     const int* eval_counts;
+    // precomputed-scores mode (MLP/torchmod: any family whose forward ran
+    // elsewhere): [R, n_eval, k] raw scores; skips the scoring GEMM stage
+    const float* scores;
 };
 
 constexpr int EVAL_KMAX = 16;
@@ -1292,7 +1295,7 @@ __global__ void __launch_bounds__(256)
 eval_metrics_kernel(EvalArgs a)
 {
     int r = blockIdx.x;
-    int node = a.nodes[r];
+    int node = a.nodes ? a.nodes[r] : r;
     int tid = threadIdx.x;
     int n_e = a.n_eval;
     const float* Xb = a.X;
@@ -1320,6 +1323,33 @@ eval_metrics_kernel(EvalArgs a)
     for (int e = tid; e < kk * kk; e += blockDim.x) conf[e] = 0;
     __syncthreads();
 
+    if (a.scores) {
+        // precomputed scores: just argmax + confusion + score staging
+        const float* zs = a.scores + (long)r * n_e * (a.is_margin ? 1 : a.k);
+        for (int sidx = tid; sidx < n_e; sidx += blockDim.x) {
+            int pred, yt;
+            float sc1;
+            if (a.is_margin) {
+                sc1 = zs[sidx];
+                pred = sc1 >= 0.f ? 1 : 0;
+                yt = Yb[sidx] > 0.f ? 1 : 0;
+            } else {
+                const float* z = zs + (long)sidx * a.k;
+                float best = -1e30f;
+                int bj = 0;
+                for (int j = 0; j < a.k; ++j)
+                    if (z[j] > best) { best = z[j]; bj = j; }
+                sc1 = a.k > 1 ? z[1] : z[0];
+                pred = bj;
+                yt = (int)Yb[sidx];
+            }
+            s1[sidx] = sc1;
+            yb[sidx] = (char)(a.is_margin ? (Yb[sidx] > 0.f ? 1 : 0)
+                                          : ((int)Yb[sidx] == 1 ? 1 : 0));
+            atomicAdd(&conf[yt * kk + pred], 1);
+        }
+        __syncthreads();
+    } else
     // samples staged through LDS in coalesced tiles (the old thread-per-
     // sample global gather was a stride-d access pattern and dominated
     // the kernel's 44 us)
@@ -2542,6 +2572,7 @@ torch::Tensor eval_metrics(torch::Tensor params, torch::Tensor nodes,
     a.out = out.data_ptr<float>();
     a.d = d; a.k = k; a.n_eval = n_eval; a.D = params.size(1);
     a.is_margin = is_margin;
+    a.scores = nullptr;
     a.eval_counts = per_node ? eval_counts.data_ptr<int>() : nullptr;
     // LDS: model + scores + binarized labels + confusion + one X tile.
     // The tile size adapts so everything fits the 160 KB budget.
@@ -2553,6 +2584,37 @@ torch::Tensor eval_metrics(torch::Tensor params, torch::Tensor nodes,
                 ", d=", d, ")");
     a.tile = (int)std::min<long>({room, (long)n_eval, 512});
     size_t smem = fixed + sizeof(float) * (size_t)a.tile * d;
+    hipLaunchKernelGGL(eval_metrics_kernel, dim3(R), dim3(256), smem,
+                       current_stream(), a);
+    return out;
+}
+
+torch::Tensor eval_metrics_scores(torch::Tensor scores, torch::Tensor Y,
+                                  int64_t k, bool is_margin)
+{
+    // K13 on PRECOMPUTED raw scores [R, n_eval, k] ([R, n_eval] margin):
+    // families whose forward runs elsewhere (MLP / torchmod) get the same
+    // one-launch confusion + macro-PRF + pairwise-AUC epilogue instead of
+    // ~40 small torch kernels and a host sync.
+    CHECK_DEV(scores); CHECK_DEV(Y);
+    TORCH_CHECK(k <= EVAL_KMAX, "n_classes > ", EVAL_KMAX, " unsupported");
+    int R = scores.size(0);
+    int n_eval = scores.size(1);
+    auto out = torch::empty({R, 5}, scores.options());
+    if (R == 0) return out;
+    EvalArgs a;
+    a.params = nullptr; a.nodes = nullptr; a.X = nullptr;
+    a.Y = Y.data_ptr<float>();
+    a.out = out.data_ptr<float>();
+    a.d = 0; a.k = k; a.n_eval = n_eval; a.D = 0;
+    a.is_margin = is_margin;
+    a.tile = 1;
+    a.eval_counts = nullptr;
+    a.scores = scores.data_ptr<float>();
+    size_t smem = sizeof(float) * n_eval
+        + ((n_eval + 15) & ~15) * sizeof(char)
+        + sizeof(int) * EVAL_KMAX * EVAL_KMAX;
+    TORCH_CHECK(smem <= 160 * 1024, "eval LDS budget exceeded: ", smem);
     hipLaunchKernelGGL(eval_metrics_kernel, dim3(R), dim3(256), smem,
                        current_stream(), a);
     return out;
@@ -2807,6 +2869,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m)
           "PENS step-1 event: score candidates, merge top-m, count winners");
     m.def("eval_metrics", &eval_metrics,
           "fused per-node eval metrics on a shared eval set (K13)");
+    m.def("eval_metrics_scores", &eval_metrics_scores,
+          "K13 metrics epilogue on precomputed scores (MLP/torchmod)");
     m.def("tick_kmeans", &tick_kmeans,
           "fused centroid merge + assign/EMA k-means tick (K11)");
 }

@@ -1238,3 +1238,32 @@ def test_batched_local_eval_matches_loop():
         assert set(a) == set(b)
         for k in a:
             assert abs(a[k] - b[k]) < 1e-4, (k, a[k], b[k])
+
+
+@pytest.mark.gpu
+def test_mlp_scores_epilogue_matches_torch_metrics():
+    """eval_metrics_scores (K13 epilogue on precomputed MLP scores) must
+    reproduce the torch metrics path."""
+    from gossipy_amd.engine import MLPSpec
+    from gossipy_amd.engine.metrics import classification_metrics_shared
+
+    n = 30
+    data = _mk_data(n, 57, CUDA)
+    spec = MLPSpec(d_in=57, hidden=(32,), n_classes=2, lr=0.1)
+    cfg = EngineConfig(
+        n_nodes=n, delta=20, protocol=AntiEntropyProtocol.PUSH,
+        model_size=spec.D, seed=5,
+    )
+    sim = BatchedGossipSimulator(cfg, spec, data, device=CUDA)
+    sim.init_nodes()
+    sim.start(n_rounds=2)
+    ids = torch.arange(n)
+    fast = sim.backend.eval_metrics_fast(
+        sim.state, spec, ids, data.gx, data.gy
+    )
+    sc = sim.backend.scores(sim.state, spec, ids, data.gx)
+    slow = classification_metrics_shared(sc, data.gy)
+    assert len(fast) == len(slow)
+    for a, b in zip(fast, slow):
+        for k in b:
+            assert abs(a[k] - b[k]) < 1e-4, (k, a[k], b[k])
