@@ -1039,7 +1039,7 @@ class HIPBackend(TorchBackend):
                 is_margin,
                 _EMPTY_I32,
             )
-        elif spec.family in ("mlp", "torchmod"):
+        elif spec.family in ("mlp", "torchmod") and gx.shape[0] <= 2048:
             sc = self.scores(state, spec, local_ids, gx)
             out = self.ext.eval_metrics_scores(
                 sc.contiguous(), gy, sc.shape[-1], False
@@ -1096,7 +1096,10 @@ class HIPBackend(TorchBackend):
                 is_margin,
                 _EMPTY_I32,
             )
-        elif spec.family in ("mlp", "torchmod"):
+        elif spec.family in ("mlp", "torchmod") and gx.shape[0] <= 2048:
+            # the kernel epilogue's pairwise AUC is O(n_eval^2) over R
+            # blocks — a bargain for sampled eval sets, a loss for huge
+            # ones (the torch sort path is O(n log n) there)
             sc = self.scores(state, spec, local_ids, gx)
             out = self.ext.eval_metrics_scores(
                 sc.contiguous(), gy, sc.shape[-1], False
