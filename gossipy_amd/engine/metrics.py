@@ -36,20 +36,25 @@ def _rank_auc(scores: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
     binary labels ``[n]`` via the Mann-Whitney rank statistic (average ranks
     for ties)."""
     R, n = scores.shape
+    dev = scores.device
     order = scores.argsort(dim=1)
-    ranks = torch.empty_like(scores)
-    arange = torch.arange(1, n + 1, dtype=scores.dtype, device=scores.device)
-    ranks.scatter_(1, order, arange.expand(R, n))
-    # average tied ranks: group equal scores
-    sorted_scores = scores.gather(1, order)
-    ties = sorted_scores[:, 1:] == sorted_scores[:, :-1]
-    if ties.any():
-        for r in range(R):  # rare path; exact tie handling
-            s = scores[r]
-            uniq, inv = torch.unique(s, return_inverse=True)
-            sums = torch.zeros_like(uniq).scatter_add_(0, inv, ranks[r])
-            cnts = torch.zeros_like(uniq).scatter_add_(0, inv, torch.ones_like(ranks[r]))
-            ranks[r] = (sums / cnts)[inv]
+    ss = scores.gather(1, order)
+    # average tied ranks, fully vectorized (no host sync, exact): group
+    # equal sorted scores, segment-mean their 1-based positions
+    new_grp = torch.ones(R, n, dtype=torch.bool, device=dev)
+    new_grp[:, 1:] = ss[:, 1:] != ss[:, :-1]
+    gid = new_grp.to(torch.int64).cumsum(dim=1) - 1  # [R, n]
+    pos = (
+        torch.arange(1, n + 1, dtype=torch.float64, device=dev)
+        .expand(R, n)
+    )
+    gsum = torch.zeros(R, n, dtype=torch.float64, device=dev)
+    gsum.scatter_add_(1, gid, pos)
+    gcnt = torch.zeros(R, n, dtype=torch.float64, device=dev)
+    gcnt.scatter_add_(1, gid, torch.ones_like(pos))
+    avg = (gsum / gcnt.clamp(min=1)).gather(1, gid)  # sorted-order ranks
+    ranks = torch.empty(R, n, dtype=scores.dtype, device=dev)
+    ranks.scatter_(1, order, avg.to(scores.dtype))
     pos = y > 0.5 if y.max() <= 1 else y > 0
     npos = int(pos.sum())
     nneg = n - npos
