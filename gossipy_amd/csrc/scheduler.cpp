@@ -26,6 +26,8 @@
 #include <map>
 #include <deque>
 #include <unordered_map>
+#include <chrono>
+#include <cstdio>
 #include <unordered_set>
 #include <vector>
 
@@ -436,6 +438,7 @@ public:
           sync_(sync), sampling_eval_(sampling_eval), seed_(seed),
           n_parts_(n_parts), sampled_(sampled)
     {
+        pend_init();
         Stream g(stream_key(seed_, TIMEOUT, 0, 0));
         deltas_.resize(n_);
         if (sync_) {
@@ -480,6 +483,86 @@ protected:
     std::vector<std::vector<int32_t>> fire_buckets_;
     std::vector<int64_t> indptr_, indices_;
     std::unordered_map<int64_t, std::vector<Msg>> pending_;
+    // delay-bounded ring over in-flight messages: slot = tick & mask.
+    // Enabled when the maximum possible delay is known and small (const /
+    // uniform / linear with the config's model size), which removes the
+    // per-message hash — the dominant host cost at 10k+ nodes. Exact-order
+    // preserving: a slot holds one tick's messages in insertion order, and
+    // take() swaps them out so same-tick re-insertions (replies) land in a
+    // fresh vector just as the map's erase/reinsert did.
+    size_t ring_mask_ = 0;
+    std::vector<int64_t> ring_tick_;
+    std::vector<std::vector<Msg>> ring_;
+    // per-tick receiver-row scratch, epoch-stamped by (2t | phase) so the
+    // 100-tick loop never allocates a map or clears an n-sized array
+    // (the per-tick unordered_map + per-row vectors were ~half the loop
+    // cost at 50k nodes)
+    std::vector<int64_t> row_stamp_;
+    std::vector<int32_t> row_idx_;
+public:
+    // lean emission: the single-rank fast path consumes ONLY the packed
+    // schedule (+ tick pointers for accounting), so the full per-event
+    // arrays and the multi-rank merge bounds are skipped — at 50k nodes
+    // that halves the emit phase
+    void set_lean(bool v) { lean_ = v; }
+protected:
+    bool lean_ = false;
+    struct DEv { int32_t v0, v1, v2, v3, next; };
+    std::vector<DEv> ev_pool_;
+    std::vector<int32_t> row_head_, row_tail_;
+
+    void pend_init()
+    {
+        int64_t bound;
+        switch (dkind_) {
+        case D_CONST: bound = dmin_; break;
+        case D_UNIFORM: bound = dmax_; break;
+        default:
+            bound = (int64_t)(timexunit_ * (double)model_size_) + overhead_;
+        }
+        if (bound >= 0 && bound < 65536) {
+            size_t cap = 1;
+            while (cap < (size_t)bound + 2) cap <<= 1;
+            ring_mask_ = cap - 1;
+            ring_tick_.assign(cap, INT64_MIN);
+            ring_.resize(cap);
+        }
+        row_stamp_.assign((size_t)n_, INT64_MIN);
+        row_idx_.assign((size_t)n_, 0);
+    }
+
+    void pend_push(int64_t t, const Msg& m)
+    {
+        if (ring_mask_) {
+            size_t sl = (size_t)t & ring_mask_;
+            if (ring_tick_[sl] != t) {
+                ring_[sl].clear();
+                ring_tick_[sl] = t;
+            }
+            ring_[sl].push_back(m);
+        } else {
+            pending_[t].push_back(m);
+        }
+    }
+
+    std::vector<Msg> pend_take(int64_t t)
+    {
+        std::vector<Msg> out;
+        if (ring_mask_) {
+            size_t sl = (size_t)t & ring_mask_;
+            if (ring_tick_[sl] == t) {
+                out.swap(ring_[sl]);
+                ring_tick_[sl] = INT64_MIN;
+            }
+        } else {
+            auto it = pending_.find(t);
+            if (it != pending_.end()) {
+                out = std::move(it->second);
+                pending_.erase(it);
+            }
+        }
+        return out;
+    }
     // consumed slots re-enter circulation only kSlotReuseLag ticks after
     // the tick that freed them (FIFO), mirroring Scheduler.SLOT_REUSE_LAG:
     // adjacent ticks never alias slot ids, so the runner can fuse
@@ -543,7 +626,7 @@ protected:
             Stream gd(stream_key(seed_, DELAY, (uint64_t)t,
                                  (uint64_t)(1 + replier)));
             int64_t dly = delay_for(gd, model_size_);
-            pending_[t + dly].push_back({requester, slot, -1, false, replier, pid});
+            pend_push(t + dly, {requester, slot, -1, false, replier, pid});
             return true;
         }
         failed += 1;
@@ -553,6 +636,7 @@ protected:
 
 py::dict NativeScheduler::next_round(int64_t r)
 {
+    auto t_start0 = std::chrono::steady_clock::now();
     const int64_t t0 = r * delta_, t1 = (r + 1) * delta_;
     int64_t sent = 0, failed = 0, total_size = 0;
 
@@ -636,7 +720,7 @@ py::dict NativeScheduler::next_round(int64_t r)
                 if (drop_u[j] >= drop_) {
                     int64_t due = t + delays[j];
                     int32_t rf = (proto_ == PUSH_PULL) ? -2 : -1;
-                    pending_[due].push_back({receiver, slot, rf, is_pull, sender, pids[j]});
+                    pend_push(due, {receiver, slot, rf, is_pull, sender, pids[j]});
                 } else {
                     failed += 1;
                     if (slot >= 0) freed.push_back(slot);
@@ -649,22 +733,31 @@ py::dict NativeScheduler::next_round(int64_t r)
         if (online_ < 1.0) {
             Stream go(stream_key(seed_, ONLINE, (uint64_t)t, 0));
             for (int64_t i = 0; i < n_; ++i) online[i] = go.rnd();
-        } else {
-            std::fill(online.begin(), online.end(), 0.0);
         }
-        std::vector<Msg> due;
-        {
-            auto it = pending_.find(t);
-            if (it != pending_.end()) {
-                due = std::move(it->second);
-                pending_.erase(it);
-            }
-        }
+        // online_ >= 1.0: nobody is ever offline — the comparison below is
+        // guarded, so the n-wide per-tick fill is skipped entirely
+        std::vector<Msg> due = pend_take(t);
         // receiver -> (slot, rslot, sender) in first-appearance order
         std::vector<int32_t> order;
-        std::unordered_map<int32_t, std::vector<std::array<int32_t, 4>>> rmap;
+        ev_pool_.clear(); row_head_.clear(); row_tail_.clear();
+        const int64_t st_main = 2 * t;
+        auto row_append = [&](int64_t stamp, int32_t node, int32_t a,
+                              int32_t b, int32_t c, int32_t d) {
+            int32_t e = (int32_t)ev_pool_.size();
+            ev_pool_.push_back({a, b, c, d, -1});
+            if (row_stamp_[node] != stamp) {
+                row_stamp_[node] = stamp;
+                row_idx_[node] = (int32_t)order.size();
+                order.push_back(node);
+                row_head_.push_back(e);
+                row_tail_.push_back(e);
+            } else {
+                ev_pool_[row_tail_[row_idx_[node]]].next = e;
+                row_tail_[row_idx_[node]] = e;
+            }
+        };
         for (const Msg& m : due) {
-            if (online[m.receiver] > online_) {
+            if (online_ < 1.0 && online[m.receiver] > online_) {
                 failed += 1;
                 if (m.slot >= 0) freed.push_back(m.slot);
                 continue;
@@ -687,22 +780,16 @@ py::dict NativeScheduler::next_round(int64_t r)
                                    total_size))
                     freed.push_back(rslot);
             }
-            auto f = rmap.find(m.receiver);
-            if (f == rmap.end()) {
-                order.push_back(m.receiver);
-                rmap[m.receiver] = {{m.slot, rslot, m.sender, m.pid}};
-            } else {
-                f->second.push_back({m.slot, rslot, m.sender, m.pid});
-            }
+            row_append(st_main, m.receiver, m.slot, rslot, m.sender, m.pid);
             freed.push_back(m.slot);
         }
-        for (int32_t rn : order) {
-            recv_nodes.push_back(rn);
-            for (auto& p : rmap[rn]) {
-                del_slots.push_back(p[0]);
-                reply_slots.push_back(p[1]);
-                del_owners.push_back(p[2]);
-                del_pids.push_back(p[3]);
+        for (size_t ri = 0; ri < order.size(); ++ri) {
+            recv_nodes.push_back(order[ri]);
+            for (int32_t e = row_head_[ri]; e >= 0; e = ev_pool_[e].next) {
+                del_slots.push_back(ev_pool_[e].v0);
+                reply_slots.push_back(ev_pool_[e].v1);
+                del_owners.push_back(ev_pool_[e].v2);
+                del_pids.push_back(ev_pool_[e].v3);
             }
             recv_nptr.push_back((int32_t)del_slots.size());
         }
@@ -710,37 +797,39 @@ py::dict NativeScheduler::next_round(int64_t r)
         pull_tptr.push_back((int32_t)pull_nodes.size());
 
         // --- sub-phase C: same-tick replies
-        std::vector<Msg> rep_due;
-        {
-            auto it = pending_.find(t);
-            if (it != pending_.end()) {
-                rep_due = std::move(it->second);
-                pending_.erase(it);
-            }
-        }
+        std::vector<Msg> rep_due = pend_take(t);
         std::vector<int32_t> rorder;
-        std::unordered_map<int32_t, std::vector<std::array<int32_t, 3>>> rrmap;
+        ev_pool_.clear(); row_head_.clear(); row_tail_.clear();
+        const int64_t st_rep = 2 * t + 1;
+        auto rep_append = [&](int32_t node, int32_t a, int32_t b, int32_t c) {
+            int32_t e = (int32_t)ev_pool_.size();
+            ev_pool_.push_back({a, b, c, 0, -1});
+            if (row_stamp_[node] != st_rep) {
+                row_stamp_[node] = st_rep;
+                row_idx_[node] = (int32_t)rorder.size();
+                rorder.push_back(node);
+                row_head_.push_back(e);
+                row_tail_.push_back(e);
+            } else {
+                ev_pool_[row_tail_[row_idx_[node]]].next = e;
+                row_tail_[row_idx_[node]] = e;
+            }
+        };
         for (const Msg& m : rep_due) {
-            if (online[m.receiver] > online_) {
+            if (online_ < 1.0 && online[m.receiver] > online_) {
                 failed += 1;
                 freed.push_back(m.slot);
                 continue;
             }
-            auto f = rrmap.find(m.receiver);
-            if (f == rrmap.end()) {
-                rorder.push_back(m.receiver);
-                rrmap[m.receiver] = {{m.slot, m.sender, m.pid}};
-            } else {
-                f->second.push_back({m.slot, m.sender, m.pid});
-            }
+            rep_append(m.receiver, m.slot, m.sender, m.pid);
             freed.push_back(m.slot);
         }
-        for (int32_t rn : rorder) {
-            rep_nodes.push_back(rn);
-            for (auto& p : rrmap[rn]) {
-                rep_slots.push_back(p[0]);
-                rep_owners.push_back(p[1]);
-                rep_pids.push_back(p[2]);
+        for (size_t ri = 0; ri < rorder.size(); ++ri) {
+            rep_nodes.push_back(rorder[ri]);
+            for (int32_t e = row_head_[ri]; e >= 0; e = ev_pool_[e].next) {
+                rep_slots.push_back(ev_pool_[e].v0);
+                rep_owners.push_back(ev_pool_[e].v1);
+                rep_pids.push_back(ev_pool_[e].v2);
             }
             rep_nptr.push_back((int32_t)rep_slots.size());
         }
@@ -749,41 +838,48 @@ py::dict NativeScheduler::next_round(int64_t r)
         for (int32_t s : freed) reuse_q_.emplace_back(t, s);
     }
 
+    auto t_mid0 = std::chrono::steady_clock::now();
     py::dict out;
     auto arr = [](std::vector<int32_t>& v) {
         auto a = py::array_t<int32_t>((py::ssize_t)v.size());
         std::copy(v.begin(), v.end(), a.mutable_data());
         return a;
     };
-    out["snap_nodes"] = arr(snap_nodes);
-    out["snap_slots"] = arr(snap_slots);
     out["snap_tptr"] = arr(snap_tptr);
-    out["recv_nodes"] = arr(recv_nodes);
-    out["recv_nptr"] = arr(recv_nptr);
     out["recv_tptr"] = arr(recv_tptr);
-    out["del_slots"] = arr(del_slots);
-    out["del_owners"] = arr(del_owners);
-    out["reply_slots"] = arr(reply_slots);
-    out["pull_nodes"] = arr(pull_nodes);
-    out["pull_slots"] = arr(pull_slots);
     out["pull_tptr"] = arr(pull_tptr);
-    out["rep_nodes"] = arr(rep_nodes);
-    out["rep_nptr"] = arr(rep_nptr);
     out["rep_tptr"] = arr(rep_tptr);
-    out["rep_slots"] = arr(rep_slots);
-    out["rep_owners"] = arr(rep_owners);
-    out["del_pids"] = arr(del_pids);
-    out["rep_pids"] = arr(rep_pids);
+    if (!lean_) {
+        out["snap_nodes"] = arr(snap_nodes);
+        out["snap_slots"] = arr(snap_slots);
+        out["recv_nodes"] = arr(recv_nodes);
+        out["recv_nptr"] = arr(recv_nptr);
+        out["del_slots"] = arr(del_slots);
+        out["del_owners"] = arr(del_owners);
+        out["reply_slots"] = arr(reply_slots);
+        out["pull_nodes"] = arr(pull_nodes);
+        out["pull_slots"] = arr(pull_slots);
+        out["rep_nodes"] = arr(rep_nodes);
+        out["rep_nptr"] = arr(rep_nptr);
+        out["rep_slots"] = arr(rep_slots);
+        out["rep_owners"] = arr(rep_owners);
+        out["del_pids"] = arr(del_pids);
+        out["rep_pids"] = arr(rep_pids);
+    }
+
     out["sent"] = sent;
     out["failed"] = failed;
     out["total_size"] = total_size;
     out["n_slots"] = next_slot_;
     {
-        auto mb = compute_merge_bounds(
-            n_, next_slot_, snap_nodes, snap_slots, snap_tptr, recv_nodes,
-            recv_nptr, recv_tptr, del_slots, reply_slots, pull_nodes,
-            pull_slots, pull_tptr, rep_nodes, rep_nptr, rep_tptr, rep_slots);
-        out["merge_bounds"] = arr(mb);
+        if (!lean_) {
+            auto mb = compute_merge_bounds(
+                n_, next_slot_, snap_nodes, snap_slots, snap_tptr, recv_nodes,
+                recv_nptr, recv_tptr, del_slots, reply_slots, pull_nodes,
+                pull_slots, pull_tptr, rep_nodes, rep_nptr, rep_tptr,
+                rep_slots);
+            out["merge_bounds"] = arr(mb);
+        }
         out["packed"] = pack_round(
             snap_nodes, snap_slots, snap_tptr, recv_nodes, recv_nptr,
             recv_tptr, del_slots, reply_slots, del_pids, pull_nodes,
@@ -799,6 +895,14 @@ py::dict NativeScheduler::next_round(int64_t r)
         out["eval_nodes"] = ev;
     } else {
         out["eval_nodes"] = py::none();
+    }
+    if (getenv("GOSSIPY_SCHED_TIME")) {
+        static double t_loop = 0, t_emit = 0;
+        auto t_end0 = std::chrono::steady_clock::now();
+        t_loop += std::chrono::duration<double>(t_mid0 - t_start0).count();
+        t_emit += std::chrono::duration<double>(t_end0 - t_mid0).count();
+        fprintf(stderr, "[sched] loop=%.3fms emit=%.3fms (cum)\n",
+                t_loop * 1000, t_emit * 1000);
     }
     return out;
 }
@@ -1039,7 +1143,7 @@ py::dict NativeTokenizedScheduler::next_round(int64_t r)
                     total_size += msize;
                     if (drop_u[j] >= drop_) {
                         int32_t rf = (proto_ == PUSH_PULL) ? -2 : -1;
-                        pending_[t + delays[j]].push_back(
+                        pend_push(t + delays[j],
                             {receiver, slot, rf, is_pull, sender, pids[j]});
                     } else {
                         failed += 1;
@@ -1052,19 +1156,12 @@ py::dict NativeTokenizedScheduler::next_round(int64_t r)
         if (online_ < 1.0) {
             Stream go(stream_key(seed_, ONLINE, (uint64_t)t, 0));
             for (int64_t i = 0; i < n_; ++i) online[i] = go.rnd();
-        } else {
-            std::fill(online.begin(), online.end(), 0.0);
         }
+        // online_ >= 1.0: nobody is ever offline — the comparison below is
+        // guarded, so the n-wide per-tick fill is skipped entirely
 
         // --- delivery waves with reactive bursts
-        std::vector<Msg> wave_due;
-        {
-            auto it = pending_.find(t);
-            if (it != pending_.end()) {
-                wave_due = std::move(it->second);
-                pending_.erase(it);
-            }
-        }
+        std::vector<Msg> wave_due = pend_take(t);
         // per-(purpose, node) streams for this tick's bursts
         std::map<std::pair<int, int32_t>, Stream> tstreams;
         auto tick_stream = [&](int purpose, int32_t node) -> Stream& {
@@ -1155,7 +1252,7 @@ py::dict NativeTokenizedScheduler::next_round(int64_t r)
                             if (dly == 0)
                                 next_due.push_back(bm);
                             else
-                                pending_[t + dly].push_back(bm);
+                                pend_push(t + dly, bm);
                         } else {
                             failed += 1;
                             freed.push_back(bslot);
@@ -1178,14 +1275,7 @@ py::dict NativeTokenizedScheduler::next_round(int64_t r)
         }
 
         // --- same-tick replies: their own group
-        std::vector<Msg> rep_due;
-        {
-            auto it = pending_.find(t);
-            if (it != pending_.end()) {
-                rep_due = std::move(it->second);
-                pending_.erase(it);
-            }
-        }
+        std::vector<Msg> rep_due = pend_take(t);
         if (!rep_due.empty()) {
             std::vector<int32_t> rorder;
             std::unordered_map<int32_t, std::vector<std::array<int32_t, 3>>> rrmap;
@@ -1225,35 +1315,41 @@ py::dict NativeTokenizedScheduler::next_round(int64_t r)
         std::copy(v.begin(), v.end(), a.mutable_data());
         return a;
     };
-    out["snap_nodes"] = arr(snap_nodes);
-    out["snap_slots"] = arr(snap_slots);
     out["snap_tptr"] = arr(snap_tptr);
-    out["recv_nodes"] = arr(recv_nodes);
-    out["recv_nptr"] = arr(recv_nptr);
     out["recv_tptr"] = arr(recv_tptr);
-    out["del_slots"] = arr(del_slots);
-    out["del_owners"] = arr(del_owners);
-    out["reply_slots"] = arr(reply_slots);
-    out["del_pids"] = arr(del_pids);
-    out["pull_nodes"] = arr(pull_nodes);
-    out["pull_slots"] = arr(pull_slots);
     out["pull_tptr"] = arr(pull_tptr);
-    out["rep_nodes"] = arr(rep_nodes);
-    out["rep_nptr"] = arr(rep_nptr);
     out["rep_tptr"] = arr(rep_tptr);
-    out["rep_slots"] = arr(rep_slots);
-    out["rep_owners"] = arr(rep_owners);
-    out["rep_pids"] = arr(rep_pids);
+    if (!lean_) {
+        out["snap_nodes"] = arr(snap_nodes);
+        out["snap_slots"] = arr(snap_slots);
+        out["recv_nodes"] = arr(recv_nodes);
+        out["recv_nptr"] = arr(recv_nptr);
+        out["del_slots"] = arr(del_slots);
+        out["del_owners"] = arr(del_owners);
+        out["reply_slots"] = arr(reply_slots);
+        out["del_pids"] = arr(del_pids);
+        out["pull_nodes"] = arr(pull_nodes);
+        out["pull_slots"] = arr(pull_slots);
+        out["rep_nodes"] = arr(rep_nodes);
+        out["rep_nptr"] = arr(rep_nptr);
+        out["rep_slots"] = arr(rep_slots);
+        out["rep_owners"] = arr(rep_owners);
+        out["rep_pids"] = arr(rep_pids);
+    }
+
     out["sent"] = sent;
     out["failed"] = failed;
     out["total_size"] = total_size;
     out["n_slots"] = next_slot_;
     {
-        auto mb = compute_merge_bounds(
-            n_, next_slot_, snap_nodes, snap_slots, snap_tptr, recv_nodes,
-            recv_nptr, recv_tptr, del_slots, reply_slots, pull_nodes,
-            pull_slots, pull_tptr, rep_nodes, rep_nptr, rep_tptr, rep_slots);
-        out["merge_bounds"] = arr(mb);
+        if (!lean_) {
+            auto mb = compute_merge_bounds(
+                n_, next_slot_, snap_nodes, snap_slots, snap_tptr, recv_nodes,
+                recv_nptr, recv_tptr, del_slots, reply_slots, pull_nodes,
+                pull_slots, pull_tptr, rep_nodes, rep_nptr, rep_tptr,
+                rep_slots);
+            out["merge_bounds"] = arr(mb);
+        }
         out["packed"] = pack_round(
             snap_nodes, snap_slots, snap_tptr, recv_nodes, recv_nptr,
             recv_tptr, del_slots, reply_slots, del_pids, pull_nodes,
@@ -1286,7 +1382,8 @@ PYBIND11_MODULE(_gossip_sched, m)
              py::arg("sampling_eval"), py::arg("seed"),
              py::arg("peers_indptr"), py::arg("peers_indices"),
              py::arg("n_parts") = 0, py::arg("sampled") = false)
-        .def("next_round", &NativeScheduler::next_round);
+        .def("next_round", &NativeScheduler::next_round)
+        .def("set_lean", &NativeScheduler::set_lean);
     py::class_<NativeTokenizedScheduler>(m, "NativeTokenizedScheduler")
         .def(py::init<int64_t, int64_t, int, int64_t, double, double, int,
                       int64_t, int64_t, double, int64_t, bool, double,

@@ -1615,6 +1615,14 @@ class NativeTokenizedAdapter:
         self.last_flat = f
         return _flat_round_summary(f, r)
 
+    def set_lean(self, v: bool) -> None:
+        self._native.set_lean(v)
+
+    def set_lean(self, v: bool) -> None:
+        """Emit only the packed schedule + tick pointers (the single-rank
+        fast path consumes nothing else)."""
+        self._native.set_lean(v)
+
     # the per-tick path is not used with this adapter (flat-exec only)
     next_round = next_round_flat
 
@@ -1649,3 +1657,13 @@ def _adapter_next_round_flat(self, r: int) -> RoundSchedule:
 
 
 NativeSchedulerAdapter.next_round_flat = _adapter_next_round_flat
+
+
+def _adapter_set_lean(self, v: bool) -> None:
+    """Emit only the packed schedule + tick pointers (the single-rank fast
+    path consumes nothing else — skips the full per-event arrays and the
+    multi-rank merge scan)."""
+    self._native.set_lean(v)
+
+
+NativeSchedulerAdapter.set_lean = _adapter_set_lean
