@@ -799,6 +799,10 @@ class BatchedGossipSimulator(SimulationEventSender):
             and not os.environ.get("GOSSIPY_SB_MAX")
         )
         packed = f.get("packed") if pack else None
+        if packed is None and "recv_nodes" not in f:
+            # lean flat: only the packed schedule was materialized — any
+            # knob that reached here is overridden (loud beats wrong)
+            packed = f["packed"]
         if packed is None and pack and os.environ.get("GOSSIPY_PACK") == "1":
             # python reference packer (A/B + flats without the C++ packer)
             packed = self._pack_flat(f)
@@ -1328,9 +1332,18 @@ class BatchedGossipSimulator(SimulationEventSender):
         """Run ``n_rounds`` rounds."""
         assert self.initialized, "call init_nodes() first"
         fast = self._fast_path_ok()
-        if fast and hasattr(self.scheduler, "set_lean"):
+        if (
+            fast
+            and hasattr(self.scheduler, "set_lean")
+            and os.environ.get("GOSSIPY_NO_MERGE") != "1"
+            and os.environ.get("GOSSIPY_NO_PACK") != "1"
+            and os.environ.get("GOSSIPY_COOP") != "1"
+            and not os.environ.get("GOSSIPY_SB_MAX")
+        ):
             # fast path consumes only the packed schedule — skip the full
-            # per-event array materialization and the multi-rank merge scan
+            # per-event array materialization and the multi-rank merge scan.
+            # Any knob that routes around the packed path needs the full
+            # arrays, so lean stays off there.
             self.scheduler.set_lean(True)
         # measured SLOWER on the flagship (0.78 vs 0.58 ms/round: the
         # submit/result handoff plus GIL contention with the executor's
