@@ -1390,5 +1390,6 @@ PYBIND11_MODULE(_gossip_sched, m)
                       uint64_t, py::object, py::object, int64_t, bool, int,
                       double, double, double, int64_t>())
         .def("next_round", &NativeTokenizedScheduler::next_round)
+        .def("set_lean", &NativeTokenizedScheduler::set_lean)
         .def("token_balances", &NativeTokenizedScheduler::token_balances);
 }
