@@ -238,3 +238,45 @@ def test_packed_matches_python_packer(protocol):
             if k == "eval_nodes":
                 continue
             np.testing.assert_array_equal(np.asarray(got[k]), v, err_msg=k)
+
+
+def test_parity_with_unbounded_delay_map_fallback():
+    """LinearDelay with a huge bandwidth term disables the scheduler's
+    delay ring (bound >= 65536) — the hash-map fallback must stay
+    bit-exact with the python scheduler."""
+    from gossipy_amd.core import LinearDelay
+
+    cfg = EngineConfig(
+        n_nodes=60, model_size=200, protocol=AntiEntropyProtocol.PUSH_PULL,
+        delay=LinearDelay(timexunit=400.0, overhead=3), drop_prob=0.1,
+        delta=50, seed=7,
+    )
+    py = Scheduler(cfg)
+    nat = NativeSchedulerAdapter(cfg)
+    for r in range(3):
+        a = py.next_round(r)
+        b = nat.next_round(r)
+        assert a.sent_messages == b.sent_messages
+        assert a.failed_messages == b.failed_messages
+        assert len(a.ticks) == len(b.ticks)
+        for pa, pb in zip(a.ticks, b.ticks):
+            np.testing.assert_array_equal(pa.snap_slots, pb.snap_slots)
+            np.testing.assert_array_equal(pa.recv_nodes, pb.recv_nodes)
+            np.testing.assert_array_equal(pa.del_slots, pb.del_slots)
+
+
+def test_parity_at_ring_boundary_delay():
+    """A uniform delay window near the ring capacity still matches."""
+    cfg = EngineConfig(
+        n_nodes=40, model_size=20, protocol=AntiEntropyProtocol.PUSH,
+        delay=UniformDelay(0, 1100), drop_prob=0.0, delta=40, seed=3,
+    )
+    py = Scheduler(cfg)
+    nat = NativeSchedulerAdapter(cfg)
+    for r in range(4):
+        a = py.next_round(r)
+        b = nat.next_round(r)
+        assert a.sent_messages == b.sent_messages
+        for pa, pb in zip(a.ticks, b.ticks):
+            np.testing.assert_array_equal(pa.recv_nodes, pb.recv_nodes)
+            np.testing.assert_array_equal(pa.del_slots, pb.del_slots)
