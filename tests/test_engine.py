@@ -1833,3 +1833,30 @@ class TestEntryLevelPacking:
         ) - 1
         assert groups < merged  # strictly better than tick merging
         assert groups <= 20
+
+
+@pytest.mark.parametrize("case", range(12))
+def test_pack_fuzz_random_configs(case):
+    """Randomized scheduler configs through the packer: every case must
+    preserve launch-order semantics under the abstract replay oracle."""
+    rng = np.random.default_rng(91_000 + case)
+    proto = [AntiEntropyProtocol.PUSH, AntiEntropyProtocol.PULL,
+             AntiEntropyProtocol.PUSH_PULL][int(rng.integers(0, 3))]
+    delay = [ConstantDelay(int(rng.integers(0, 4))),
+             UniformDelay(0, int(rng.integers(1, 12)))][int(rng.integers(0, 2))]
+    cfg = EngineConfig(
+        n_nodes=int(rng.integers(10, 150)),
+        model_size=10,
+        protocol=proto,
+        delay=delay,
+        drop_prob=float(rng.uniform(0, 0.4)),
+        online_prob=float(rng.uniform(0.6, 1.0)),
+        delta=int(rng.integers(20, 120)),
+        seed=int(rng.integers(0, 10_000)),
+        sync=bool(rng.integers(0, 2)),
+    )
+    sch = Scheduler(cfg)
+    for r in range(2):
+        f = BatchedGossipSimulator._flatten_phases(sch.next_round(r).ticks)
+        p = BatchedGossipSimulator._pack_flat(f)
+        assert _abstract_replay(f) == _abstract_replay(f, p)
