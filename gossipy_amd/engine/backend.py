@@ -1027,7 +1027,13 @@ class HIPBackend(TorchBackend):
         MLP/torchmod run their (batched) forward first and feed the
         precomputed scores to the same metrics epilogue. Returns ``None``
         for unsupported families."""
-        if spec.family in ("logreg", "pegasos", "adaline"):
+        if (
+            spec.family in ("logreg", "pegasos", "adaline")
+            and gx.shape[0] <= 2048
+        ):
+            # the kernel's pairwise AUC is O(n_eval^2) per node — a bargain
+            # for sampled eval sets, a loss on huge ones where the torch
+            # O(n log n) sort path takes over
             is_margin = spec.family in ("pegasos", "adaline")
             out = self.ext.eval_metrics(
                 state.params,
@@ -1084,7 +1090,13 @@ class HIPBackend(TorchBackend):
         sigmoid is monotonic so predictions and ranks are identical.
         MLP/torchmod feed their batched forward's scores to the same
         kernel epilogue."""
-        if spec.family in ("logreg", "pegasos", "adaline"):
+        if (
+            spec.family in ("logreg", "pegasos", "adaline")
+            and gx.shape[0] <= 2048
+        ):
+            # the kernel's pairwise AUC is O(n_eval^2) per node — a bargain
+            # for sampled eval sets, a loss on huge ones where the torch
+            # O(n log n) sort path takes over
             is_margin = spec.family in ("pegasos", "adaline")
             out = self.ext.eval_metrics(
                 state.params,
