@@ -1029,11 +1029,13 @@ class HIPBackend(TorchBackend):
         for unsupported families."""
         if (
             spec.family in ("logreg", "pegasos", "adaline")
-            and gx.shape[0] <= 2048
+            and len(local_ids) * gx.shape[0] ** 2 <= 20_000_000
         ):
-            # the kernel's pairwise AUC is O(n_eval^2) per node — a bargain
-            # for sampled eval sets, a loss on huge ones where the torch
-            # O(n log n) sort path takes over
+            # the kernel's pairwise AUC costs ~30 ns per (node, pair)
+            # unit: R x n_eval^2 <= 2e7 keeps it under ~0.6 ms (sampled
+            # eval sets are ~1e6). Beyond that (500 nodes x 2300 samples
+            # was 58% of the 50k-node round) the torch O(n log n) sort
+            # path takes over
             is_margin = spec.family in ("pegasos", "adaline")
             out = self.ext.eval_metrics(
                 state.params,
@@ -1045,7 +1047,10 @@ class HIPBackend(TorchBackend):
                 is_margin,
                 _EMPTY_I32,
             )
-        elif spec.family in ("mlp", "torchmod") and gx.shape[0] <= 2048:
+        elif (
+            spec.family in ("mlp", "torchmod")
+            and len(local_ids) * gx.shape[0] ** 2 <= 20_000_000
+        ):
             sc = self.scores(state, spec, local_ids, gx)
             out = self.ext.eval_metrics_scores(
                 sc.contiguous(), gy, sc.shape[-1], False
@@ -1092,11 +1097,13 @@ class HIPBackend(TorchBackend):
         kernel epilogue."""
         if (
             spec.family in ("logreg", "pegasos", "adaline")
-            and gx.shape[0] <= 2048
+            and len(local_ids) * gx.shape[0] ** 2 <= 20_000_000
         ):
-            # the kernel's pairwise AUC is O(n_eval^2) per node — a bargain
-            # for sampled eval sets, a loss on huge ones where the torch
-            # O(n log n) sort path takes over
+            # the kernel's pairwise AUC costs ~30 ns per (node, pair)
+            # unit: R x n_eval^2 <= 2e7 keeps it under ~0.6 ms (sampled
+            # eval sets are ~1e6). Beyond that (500 nodes x 2300 samples
+            # was 58% of the 50k-node round) the torch O(n log n) sort
+            # path takes over
             is_margin = spec.family in ("pegasos", "adaline")
             out = self.ext.eval_metrics(
                 state.params,
@@ -1108,10 +1115,10 @@ class HIPBackend(TorchBackend):
                 is_margin,
                 _EMPTY_I32,
             )
-        elif spec.family in ("mlp", "torchmod") and gx.shape[0] <= 2048:
-            # the kernel epilogue's pairwise AUC is O(n_eval^2) over R
-            # blocks — a bargain for sampled eval sets, a loss for huge
-            # ones (the torch sort path is O(n log n) there)
+        elif (
+            spec.family in ("mlp", "torchmod")
+            and len(local_ids) * gx.shape[0] ** 2 <= 20_000_000
+        ):
             sc = self.scores(state, spec, local_ids, gx)
             out = self.ext.eval_metrics_scores(
                 sc.contiguous(), gy, sc.shape[-1], False
