@@ -618,13 +618,17 @@ class BatchedGossipSimulator(SimulationEventSender):
             and self.spec.family in ("logreg", "pegasos", "adaline",
                                      "mlp", "torchmod")
         ):
-            self._drain_eval()
+            self._drain_eval(force=False)
             if len(mine):
                 h = self.backend.eval_metrics_launch(
                     self.state, self.spec, local_ids, self.data.gx, self.data.gy
                 )
                 if h is not None:
-                    self._eval_pending = (t, h)
+                    if not hasattr(self, "_eval_pending") or self._eval_pending is None:
+                        from collections import deque
+
+                        self._eval_pending = deque()
+                    self._eval_pending.append((t, h))
                     return
             else:
                 return
@@ -691,16 +695,25 @@ class BatchedGossipSimulator(SimulationEventSender):
             if results_global:
                 self.notify_evaluation(t, False, results_global)
 
-    def _drain_eval(self) -> None:
-        """Collect an in-flight pipelined evaluation, if any."""
+    def _drain_eval(self, force: bool = True) -> None:
+        """Collect in-flight pipelined evaluations. ``force=False`` (the
+        per-round call) collects only those whose D2H event has already
+        signalled — the host never blocks on the GPU mid-loop; boundaries
+        (end of start(), checkpoint save) force-collect everything. At
+        most two evaluations ride in flight (the pinned staging is
+        double-buffered), so the per-round call force-collects the oldest
+        when a third would launch."""
         pend = getattr(self, "_eval_pending", None)
-        if pend is None:
+        if not pend:
             return
-        self._eval_pending = None
-        t_prev, h = pend
-        results = self.backend.eval_metrics_collect(h)
-        if results:
-            self.notify_evaluation(t_prev, False, results)
+        while pend:
+            t_prev, h = pend[0]
+            if not force and len(pend) < 2 and not h[1].query():
+                break
+            pend.popleft()
+            results = self.backend.eval_metrics_collect(h)
+            if results:
+                self.notify_evaluation(t_prev, False, results)
 
     def _fast_path_ok(self) -> bool:
         """Single-GPU fast path: native scheduler + HIP round executor —
