@@ -8,11 +8,14 @@ This is the MI355X replacement of the object layer's per-node event loop
 * every rank derives the *same* round schedule from the deterministic
   random tape, so there is no control-plane traffic at all — only model
   rows move between GPUs;
-* per tick the runner issues at most three batched kernels on its resident
-  nodes: snapshot (sub-phase A), merge+update deliveries (B), same-tick
-  reply deliveries (C). Snapshot slots whose writer and consumer live on
-  different GPUs travel as grouped ``batch_isend_irecv`` transfers (RCCL
-  p2p over xGMI) between the sub-phases;
+* the scheduler packs a round at ENTRY granularity (deliveries to one
+  receiver coalesce across ticks into one CSR row, snapshots of receiving
+  nodes embed as per-delivery reply writes), so the runner issues a
+  handful of hazard-free launch groups per round — each group at most a
+  snapshot launch and two deliver launches. Snapshot slots whose writer
+  and consumer live on different GPUs travel as ONE grouped
+  ``batch_isend_irecv`` transfer per group (RCCL p2p over xGMI) between
+  the launches;
 * the round-end evaluation sweep runs batched on-device, and only metric
   dicts are gathered to rank 0 (C5).
 
