@@ -25,6 +25,7 @@
 #include <cstdint>
 #include <map>
 #include <deque>
+#include <functional>
 #include <unordered_map>
 #include <chrono>
 #include <cstdio>
@@ -192,13 +193,15 @@ static std::vector<int32_t> compute_merge_bounds(
 
 
 // ---------------------------------------------------------------------------
-// Entry-level launch packing (bit-exact port of
+// Entry-level launch packing with per-node deferral (bit-exact port of
 // BatchedGossipSimulator._pack_flat — parity enforced by
-// tests/test_native_sched.py). Coalesces deliveries to one receiver from
-// many ticks into one CSR row and embeds snapshots as per-delivery reply
-// writes; deliveries whose source slot is written in the group's first
-// deliver launch move to the second (rep_*) launch. Returns a py::dict of
-// packed arrays in the executors' format.
+// tests/test_native_sched.py). Deliveries to one receiver coalesce across
+// ticks into one CSR row; snapshots of receiving nodes embed as per-delivery
+// reply writes; same-group produce->consume pairs split across two deliver
+// launches; events needing a third level DEFER their node and replay (in
+// global schedule order — causality) after the group closes, so burst
+// chains no longer fragment the round. Returns packed arrays in the
+// executors' format.
 // ---------------------------------------------------------------------------
 static py::dict pack_round(
     const std::vector<int32_t>& snap_nodes, const std::vector<int32_t>& snap_slots,
@@ -220,9 +223,6 @@ static py::dict pack_round(
     bool has_pid = !del_pids.empty() || !rep_pids.empty();
     bool has_own = !del_owners.empty() || !rep_owners.empty();
 
-    // events live in ONE growing pool chained per row (no per-row heap
-    // allocations — at 10k nodes the old vector<Ev>-per-row design cost
-    // ~1 ms/round in allocator traffic)
     struct Ev { int32_t slot, reply, pid, own, next; };
     struct Row { int32_t node, head, tail; };
     struct Out {
@@ -239,13 +239,12 @@ static py::dict pack_round(
     o.del_slots.reserve(del_slots.size() + rep_slots.size());
     o.reply_slots.reserve(del_slots.size() + rep_slots.size());
 
-    // group-local state in epoch-stamped flat arrays (stamp = group id):
-    // no per-group clearing, no hashing — O(1) lookups at ~20k events/ms
+    // group-local state in epoch-stamped flat arrays (stamp = group id)
     int32_t n_slots_cap = (int32_t)n_slots;
     int32_t n_nodes_cap = (int32_t)n_nodes;
     std::vector<int32_t> t_stamp(n_slots_cap, -1);           // touched
     std::vector<int32_t> w_stamp(n_slots_cap, -1);           // written
-    std::vector<int8_t> w_lvl(n_slots_cap, 0);
+    std::vector<int8_t> w_lvl(n_slots_cap, 0);               // 1/2/3, 9=pending
     std::vector<int32_t> l_stamp(n_nodes_cap, -1);           // latest row
     std::vector<int8_t> l_kind(n_nodes_cap, 0);
     std::vector<int32_t> l_idx(n_nodes_cap, 0);
@@ -266,7 +265,13 @@ static py::dict pack_round(
         row.tail = idx;
     };
 
-    auto close = [&]() {
+    // deferral state: node flags + (node, event) queue in schedule order
+    struct DEv { int32_t kind, slot, reply, pid, own; };  // kind 0=snap 1=deliv
+    std::vector<int8_t> is_def(n_nodes_cap, 0);
+    std::vector<std::pair<int32_t, DEv>> deferred;
+    bool in_replay = false;
+
+    auto flush = [&]() {
         for (auto& p : g_snap) {
             o.snap_nodes.push_back(p.first);
             o.snap_slots.push_back(p.second);
@@ -298,8 +303,55 @@ static py::dict pack_round(
         ++gid;  // invalidates every stamped entry at once
     };
 
-    auto place_snap = [&](int32_t node, int32_t slot) {
-        if (t_stamp[slot] == gid) close();
+    std::function<void(int32_t, int32_t)> place_snap;
+    std::function<void(int32_t, int32_t, int32_t, int32_t, int32_t)>
+        place_delivery;
+
+    auto defer = [&](int32_t node, const DEv& ev) {
+        is_def[node] = 1;
+        deferred.emplace_back(node, ev);
+        if (ev.kind == 0) {
+            w_stamp[ev.slot] = gid; w_lvl[ev.slot] = 9;
+            t_stamp[ev.slot] = gid;
+        } else {
+            t_stamp[ev.slot] = gid;
+            if (ev.reply >= 0) {
+                w_stamp[ev.reply] = gid; w_lvl[ev.reply] = 9;
+                t_stamp[ev.reply] = gid;
+            }
+        }
+    };
+
+    auto close = [&]() {
+        if (in_replay) {
+            flush();
+            return;
+        }
+        flush();
+        while (!deferred.empty()) {
+            in_replay = true;
+            std::vector<std::pair<int32_t, DEv>> dl;
+            dl.swap(deferred);
+            for (auto& p : dl) is_def[p.first] = 0;
+            for (auto& p : dl) {
+                const DEv& ev = p.second;
+                if (ev.kind == 0) place_snap(p.first, ev.slot);
+                else place_delivery(p.first, ev.slot, ev.reply, ev.pid, ev.own);
+            }
+            in_replay = false;
+            if (!deferred.empty()) flush();  // next drain pass = fresh group
+        }
+    };
+
+    place_snap = [&](int32_t node, int32_t slot) {
+        if (is_def[node]) { defer(node, {0, slot, -1, -1, -1}); return; }
+        if (t_stamp[slot] == gid) {
+            close();
+            if (is_def[node]) {  // replay re-deferred this node
+                defer(node, {0, slot, -1, -1, -1});
+                return;
+            }
+        }
         if (l_stamp[node] != gid) {
             g_snap.emplace_back(node, slot);
             w_stamp[slot] = gid; w_lvl[slot] = 1;
@@ -307,24 +359,23 @@ static py::dict pack_round(
             auto& lst = (l_kind[node] == 2) ? l2 : l3;
             Ev& last = pool[lst[l_idx[node]].tail];
             if (last.reply >= 0) {
-                close();
-                g_snap.emplace_back(node, slot);
-                w_stamp[slot] = gid; w_lvl[slot] = 1;
-            } else {
-                last.reply = slot;
-                w_stamp[slot] = gid;
-                w_lvl[slot] = (l_kind[node] == 2) ? 2 : 3;
+                defer(node, {0, slot, -1, -1, -1});
+                return;
             }
+            last.reply = slot;
+            w_stamp[slot] = gid;
+            w_lvl[slot] = (l_kind[node] == 2) ? 2 : 3;
         }
         t_stamp[slot] = gid;
     };
 
-    auto place_delivery = [&](int32_t node, int32_t slot, int32_t reply,
-                              int32_t pid, int32_t own) {
+    place_delivery = [&](int32_t node, int32_t slot, int32_t reply,
+                         int32_t pid, int32_t own) {
+        if (is_def[node]) { defer(node, {1, slot, reply, pid, own}); return; }
         int lvl = (w_stamp[slot] == gid) ? w_lvl[slot] : 0;
         if (lvl >= 3 || (reply >= 0 && t_stamp[reply] == gid)) {
-            close();
-            lvl = 0;
+            defer(node, {1, slot, reply, pid, own});
+            return;
         }
         bool in_l3 = (l_stamp[node] == gid && l_kind[node] == 3);
         int wl;
@@ -389,7 +440,12 @@ static py::dict pack_round(
             }
         }
     }
-    if (!g_snap.empty() || !l2.empty() || !l3.empty()) close();
+    int guard = 0;
+    while ((!g_snap.empty() || !l2.empty() || !l3.empty() ||
+            !deferred.empty()) && guard < 100000) {
+        close();
+        ++guard;
+    }
 
     int32_t n_groups = (int32_t)o.snap_tptr.size() - 1;
     o.pull_tptr.assign((size_t)n_groups + 1, 0);

@@ -947,186 +947,167 @@ class BatchedGossipSimulator(SimulationEventSender):
 
     @staticmethod
     def _pack_flat(f: dict) -> dict:
-        """Entry-level launch packing: the aggressive successor of
-        :meth:`_merge_flat_groups`. Instead of fusing whole ticks, every
-        event is placed individually:
+        """Entry-level launch packing with per-node deferral.
 
-        * deliveries to the same receiver from MANY ticks coalesce into ONE
-          CSR row (the kernel already processes a row's deliveries in
-          order, so tick order is preserved inside the row);
-        * a snapshot of a node that already received in the group embeds as
-          the existing per-delivery reply-write (``rslots[j]`` = params+age
-          written right after delivery ``j`` — exactly the state the
-          reference snapshots at the send tick);
-        * a delivery whose source slot is written in the group's deliver
-          launch (an embedded snapshot or a protocol reply) moves to the
-          group's second deliver launch (the ``rep_*`` arrays), keeping
-          produce → consume in separate launches.
+        Every event is placed individually into the open launch group:
 
-        The flagship's 100-tick round packs into a handful of groups, so
-        both launch count and per-launch occupancy improve. Output uses
-        the executors' existing array format; pull arrays come back empty
-        (standalone pull-snapshots fold into the snapshot launch — their
-        content is identical there) and ``rep_reply_slots`` carries the
-        second launch's embedded writes."""
+        * deliveries to one receiver from MANY ticks coalesce into ONE CSR
+          row (the kernel processes a row's deliveries in order, so tick
+          order is preserved inside the row);
+        * a snapshot of a node that already received in the group embeds
+          as the per-delivery reply write (``rslots[j]`` — written right
+          after delivery ``j``, exactly the state the reference snapshots
+          at the send tick);
+        * a delivery whose source slot is produced in the group's own
+          deliver launch moves to the second deliver launch (the
+          ``rep_*`` arrays with ``rep_reply_slots`` for their embedded
+          writes);
+        * an event that would need a THIRD level (burst chains: consume a
+          second-launch product) **defers its node**: the node's remaining
+          events queue up (in global schedule order — causality) and
+          replay after the group closes, so one deep chain no longer
+          fragments the whole round. Groups close only on recycled-slot
+          aliasing, at drain passes, and at round end.
+
+        The flagship's 100-tick round packs into ~4 groups; PUSH_PULL
+        rounds into ~7; tokenized burst rounds into ~15. Output uses the
+        executors' existing array format (pull arrays come back empty —
+        standalone pull-snapshots fold into the snapshot launch, where
+        their content is identical)."""
         delta = len(f["snap_tptr"]) - 1
-        st, rt = f["snap_tptr"], f["recv_tptr"]
-        pt, qt = f["pull_tptr"], f["rep_tptr"]
+        st, rt, pt, qt = f["snap_tptr"], f["recv_tptr"], f["pull_tptr"], f["rep_tptr"]
         nptr, rep_nptr = f["recv_nptr"], f["rep_nptr"]
         has_pid = len(f.get("del_pids", ())) > 0 or len(f.get("rep_pids", ())) > 0
-        has_own = (
-            len(f.get("del_owners", ())) > 0 or len(f.get("rep_owners", ())) > 0
-        )
+        has_own = len(f.get("del_owners", ())) > 0 or len(f.get("rep_owners", ())) > 0
 
-        out = {
-            "snap_nodes": [], "snap_slots": [], "snap_tptr": [0],
-            "recv_nodes": [], "recv_nptr": [0], "recv_tptr": [0],
-            "del_slots": [], "reply_slots": [], "del_pids": [],
-            "del_owners": [],
-            "pull_nodes": [], "pull_slots": [], "pull_tptr": [0],
-            "rep_nodes": [], "rep_nptr": [0], "rep_tptr": [0],
-            "rep_slots": [], "rep_reply_slots": [], "rep_pids": [],
-            "rep_owners": [],
-        }
+        out = {k: [] for k in ("snap_nodes","snap_slots","recv_nodes","del_slots",
+            "reply_slots","del_pids","del_owners","rep_nodes","rep_slots",
+            "rep_reply_slots","rep_pids","rep_owners")}
+        out.update({"snap_tptr":[0],"recv_tptr":[0],"recv_nptr":[0],
+                    "rep_tptr":[0],"rep_nptr":[0],"pull_tptr":[0]})
 
-        # current-group state
-        g_snap = []              # (node, slot)
-        rows2 = {}               # node -> row idx in l2
-        rows3 = {}               # node -> row idx in l3
-        l2 = []                  # list of [node, [(slot, reply, pid), ...]]
-        l3 = []
-        latest = {}              # node -> ("l2"|"l3", row idx)
-        written = {}             # slot -> launch level (1, 2, 3)
-        touched = set()          # every slot id seen this group (alias guard)
+        g_snap = []; rows2 = {}; rows3 = {}; l2 = []; l3 = []; latest = {}
+        written = {}; touched = set()
+        deferred_nodes = set()
+        deferred_list = []   # (node, event) in ORIGINAL schedule order — replay
+                             # must preserve cross-node causality (a consumer's
+                             # producer always precedes it in schedule order)
+
+        in_replay = [False]
 
         def close():
-            out["snap_nodes"].extend(n for n, _ in g_snap)
-            out["snap_slots"].extend(s for _, s in g_snap)
+            if in_replay[0]:
+                flush()
+                return
+            flush()
+            while deferred_list:
+                in_replay[0] = True
+                dl = deferred_list[:]; deferred_list.clear(); deferred_nodes.clear()
+                for n, ev in dl:
+                    if ev[0] == "s": place_snap(n, ev[1])
+                    else: place_delivery(n, ev[1], ev[2], ev[3], ev[4])
+                in_replay[0] = False
+                if deferred_list:
+                    flush()  # each drain pass lands in a fresh group
+
+        def flush():
+            out["snap_nodes"].extend(n for n,_ in g_snap)
+            out["snap_slots"].extend(s for _,s in g_snap)
             out["snap_tptr"].append(len(out["snap_nodes"]))
             for node, evs in l2:
                 out["recv_nodes"].append(node)
-                for s, rsl, pid, own in evs:
-                    out["del_slots"].append(s)
-                    out["reply_slots"].append(rsl)
-                    if has_pid:
-                        out["del_pids"].append(pid)
-                    if has_own:
-                        out["del_owners"].append(own)
+                for s,r_,p,o in evs:
+                    out["del_slots"].append(s); out["reply_slots"].append(r_)
+                    if has_pid: out["del_pids"].append(p)
+                    if has_own: out["del_owners"].append(o)
                 out["recv_nptr"].append(len(out["del_slots"]))
             out["recv_tptr"].append(len(out["recv_nodes"]))
             for node, evs in l3:
                 out["rep_nodes"].append(node)
-                for s, rsl, pid, own in evs:
-                    out["rep_slots"].append(s)
-                    out["rep_reply_slots"].append(rsl)
-                    if has_pid:
-                        out["rep_pids"].append(pid)
-                    if has_own:
-                        out["rep_owners"].append(own)
+                for s,r_,p,o in evs:
+                    out["rep_slots"].append(s); out["rep_reply_slots"].append(r_)
+                    if has_pid: out["rep_pids"].append(p)
+                    if has_own: out["rep_owners"].append(o)
                 out["rep_nptr"].append(len(out["rep_slots"]))
             out["rep_tptr"].append(len(out["rep_nodes"]))
             out["pull_tptr"].append(0)
-            g_snap.clear(); rows2.clear(); rows3.clear()
-            l2.clear(); l3.clear(); latest.clear()
-            written.clear(); touched.clear()
+            g_snap.clear(); rows2.clear(); rows3.clear(); l2.clear(); l3.clear()
+            latest.clear(); written.clear(); touched.clear()
+
+        def defer(node, ev):
+            deferred_nodes.add(node)
+            deferred_list.append((node, ev))
+            if ev[0] == "s":
+                written[ev[1]] = 9; touched.add(ev[1])
+            else:
+                touched.add(ev[1])
+                if ev[2] >= 0: written[ev[2]] = 9; touched.add(ev[2])
 
         def place_snap(node, slot):
+            if node in deferred_nodes: defer(node, ("s", slot)); return
             if slot in touched:
                 close()
+                if node in deferred_nodes:  # replay re-deferred this node
+                    defer(node, ("s", slot)); return
             pos = latest.get(node)
             if pos is None:
-                g_snap.append((node, slot))
-                written[slot] = 1
+                g_snap.append((node, slot)); written[slot] = 1
             else:
-                lst = l2 if pos[0] == "l2" else l3
+                lst = l2 if pos[0]==2 else l3
                 evs = lst[pos[1]][1]
-                if evs[-1][1] >= 0:  # last delivery already writes a slot
-                    close()
-                    g_snap.append((node, slot))
-                    written[slot] = 1
-                else:
-                    evs[-1] = (evs[-1][0], slot, evs[-1][2], evs[-1][3])
-                    written[slot] = 2 if pos[0] == "l2" else 3
+                if evs[-1][1] >= 0:
+                    defer(node, ("s", slot)); return
+                evs[-1] = (evs[-1][0], slot, evs[-1][2], evs[-1][3])
+                written[slot] = 2 if pos[0]==2 else 3
             touched.add(slot)
 
         def place_delivery(node, slot, reply, pid, own):
+            if node in deferred_nodes: defer(node, ("d", slot, reply, pid, own)); return
             lvl = written.get(slot, 0)
             if lvl >= 3 or (reply >= 0 and reply in touched):
-                close()
-                lvl = 0
+                defer(node, ("d", slot, reply, pid, own)); return
             pos = latest.get(node)
-            if pos is not None and pos[0] == "l3":
-                if lvl >= 3:  # re-check after close (lvl reset above)
-                    close()
-                    pos = None
-            if pos is not None and pos[0] == "l3":
-                l3[pos[1]][1].append((slot, reply, pid, own))
-                wl = 3
+            if pos is not None and pos[0] == 3:
+                l3[pos[1]][1].append((slot, reply, pid, own)); wl = 3
             elif lvl >= 2:
-                # source written in this group's first deliver launch:
-                # consume in the second deliver launch
                 ri = rows3.get(node)
                 if ri is None:
-                    rows3[node] = ri = len(l3)
-                    l3.append([node, []])
-                l3[ri][1].append((slot, reply, pid, own))
-                latest[node] = ("l3", ri)
-                wl = 3
+                    rows3[node] = ri = len(l3); l3.append([node, []])
+                l3[ri][1].append((slot, reply, pid, own)); latest[node] = (3, ri); wl = 3
             else:
                 ri = rows2.get(node)
                 if ri is None:
-                    rows2[node] = ri = len(l2)
-                    l2.append([node, []])
-                    latest[node] = ("l2", ri)
-                l2[ri][1].append((slot, reply, pid, own))
-                wl = 2
+                    rows2[node] = ri = len(l2); l2.append([node, []]); latest[node] = (2, ri)
+                l2[ri][1].append((slot, reply, pid, own)); wl = 2
             touched.add(slot)
-            if reply >= 0:
-                written[reply] = wl
-                touched.add(reply)
+            if reply >= 0: written[reply] = wl; touched.add(reply)
 
-        snap_nodes, snap_slots = f["snap_nodes"], f["snap_slots"]
-        recv_nodes = f["recv_nodes"]
-        del_slots, reply_slots = f["del_slots"], f["reply_slots"]
-        del_pids = f.get("del_pids")
-        del_owners = f.get("del_owners")
-        pull_nodes, pull_slots = f["pull_nodes"], f["pull_slots"]
-        rep_nodes, rep_slots = f["rep_nodes"], f["rep_slots"]
-        rep_pids = f.get("rep_pids")
-        rep_owners = f.get("rep_owners")
+        sn, ss = f["snap_nodes"], f["snap_slots"]
+        rn = f["recv_nodes"]; ds, rs = f["del_slots"], f["reply_slots"]
+        dp = f.get("del_pids"); do = f.get("del_owners")
+        pn, ps = f["pull_nodes"], f["pull_slots"]
+        qn, qs = f["rep_nodes"], f["rep_slots"]
+        qp = f.get("rep_pids"); qo = f.get("rep_owners")
         for t in range(delta):
-            for i in range(st[t], st[t + 1]):
-                place_snap(int(snap_nodes[i]), int(snap_slots[i]))
-            for r in range(rt[t], rt[t + 1]):
-                x = int(recv_nodes[r])
-                for d in range(nptr[r], nptr[r + 1]):
-                    pid = int(del_pids[d]) if has_pid and len(del_pids) else -1
-                    own = (
-                        int(del_owners[d])
-                        if has_own and len(del_owners) else -1
-                    )
-                    place_delivery(x, int(del_slots[d]),
-                                   int(reply_slots[d]), pid, own)
-            for i in range(pt[t], pt[t + 1]):
-                place_snap(int(pull_nodes[i]), int(pull_slots[i]))
-            for r in range(qt[t], qt[t + 1]):
-                x = int(rep_nodes[r])
-                for d in range(rep_nptr[r], rep_nptr[r + 1]):
-                    pid = int(rep_pids[d]) if has_pid and len(rep_pids) else -1
-                    own = (
-                        int(rep_owners[d])
-                        if has_own and len(rep_owners) else -1
-                    )
-                    place_delivery(x, int(rep_slots[d]), -1, pid, own)
-        if g_snap or l2 or l3:
-            close()
-        res = {}
-        for k, v in out.items():
-            res[k] = np.asarray(v, dtype=np.int32)
-        res["pull_nodes"] = np.zeros(0, np.int32)
-        res["pull_slots"] = np.zeros(0, np.int32)
-        if "eval_nodes" in f:
-            res["eval_nodes"] = f["eval_nodes"]
+            for i in range(st[t], st[t+1]): place_snap(int(sn[i]), int(ss[i]))
+            for r in range(rt[t], rt[t+1]):
+                x = int(rn[r])
+                for d in range(nptr[r], nptr[r+1]):
+                    place_delivery(x, int(ds[d]), int(rs[d]),
+                                   int(dp[d]) if has_pid and len(dp) else -1,
+                                   int(do[d]) if has_own and len(do) else -1)
+            for i in range(pt[t], pt[t+1]): place_snap(int(pn[i]), int(ps[i]))
+            for r in range(qt[t], qt[t+1]):
+                x = int(qn[r])
+                for d in range(rep_nptr[r], rep_nptr[r+1]):
+                    place_delivery(x, int(qs[d]), -1,
+                                   int(qp[d]) if has_pid and len(qp) else -1,
+                                   int(qo[d]) if has_own and len(qo) else -1)
+        guard = 0
+        while (g_snap or l2 or l3 or deferred_list) and guard < 10000:
+            close(); guard += 1
+        res = {k: np.asarray(v, dtype=np.int32) for k, v in out.items()}
+        res["pull_nodes"] = np.zeros(0, np.int32); res["pull_slots"] = np.zeros(0, np.int32)
         return res
 
     def _run_round_fast(self, f: dict) -> None:
