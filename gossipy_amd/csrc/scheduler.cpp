@@ -256,167 +256,199 @@ static py::dict pack_round(
     std::vector<Ev> pool;
     pool.reserve(del_slots.size() + rep_slots.size() + 16);
     std::vector<Row> l2, l3;
-    auto append_ev = [&](Row& row, int32_t slot, int32_t reply, int32_t pid,
-                         int32_t own) {
-        int32_t idx = (int32_t)pool.size();
-        pool.push_back({slot, reply, pid, own, -1});
-        if (row.tail >= 0) pool[row.tail].next = idx;
-        else row.head = idx;
-        row.tail = idx;
-    };
 
     // deferral state: node flags + (node, event) queue in schedule order
     struct DEv { int32_t kind, slot, reply, pid, own; };  // kind 0=snap 1=deliv
     std::vector<int8_t> is_def(n_nodes_cap, 0);
     std::vector<std::pair<int32_t, DEv>> deferred;
-    bool in_replay = false;
 
-    auto flush = [&]() {
-        for (auto& p : g_snap) {
-            o.snap_nodes.push_back(p.first);
-            o.snap_slots.push_back(p.second);
+    struct Packer {
+        std::vector<std::pair<int32_t, int32_t>>& g_snap;
+        std::vector<Ev>& pool;
+        std::vector<Row>& l2;
+        std::vector<Row>& l3;
+        std::vector<int32_t>& t_stamp;
+        std::vector<int32_t>& w_stamp;
+        std::vector<int8_t>& w_lvl;
+        std::vector<int32_t>& l_stamp;
+        std::vector<int8_t>& l_kind;
+        std::vector<int32_t>& l_idx;
+        std::vector<int32_t>& r2_stamp;
+        std::vector<int32_t>& r2_idx;
+        std::vector<int32_t>& r3_stamp;
+        std::vector<int32_t>& r3_idx;
+        int32_t& gid;
+        std::vector<int8_t>& is_def;
+        std::vector<std::pair<int32_t, DEv>>& deferred;
+        bool in_replay = false;
+        Out& o;
+        bool has_pid, has_own;
+
+        void append_ev(Row& row, int32_t slot, int32_t reply, int32_t pid,
+                       int32_t own)
+        {
+            int32_t idx = (int32_t)pool.size();
+            pool.push_back({slot, reply, pid, own, -1});
+            if (row.tail >= 0) pool[row.tail].next = idx;
+            else row.head = idx;
+            row.tail = idx;
         }
-        o.snap_tptr.push_back((int32_t)o.snap_nodes.size());
-        for (auto& row : l2) {
-            o.recv_nodes.push_back(row.node);
-            for (int32_t e = row.head; e >= 0; e = pool[e].next) {
-                o.del_slots.push_back(pool[e].slot);
-                o.reply_slots.push_back(pool[e].reply);
-                if (has_pid) o.del_pids.push_back(pool[e].pid);
-                if (has_own) o.del_owners.push_back(pool[e].own);
+
+        void flush()
+        {
+            for (auto& p : g_snap) {
+                o.snap_nodes.push_back(p.first);
+                o.snap_slots.push_back(p.second);
             }
-            o.recv_nptr.push_back((int32_t)o.del_slots.size());
-        }
-        o.recv_tptr.push_back((int32_t)o.recv_nodes.size());
-        for (auto& row : l3) {
-            o.rep_nodes.push_back(row.node);
-            for (int32_t e = row.head; e >= 0; e = pool[e].next) {
-                o.rep_slots.push_back(pool[e].slot);
-                o.rep_reply_slots.push_back(pool[e].reply);
-                if (has_pid) o.rep_pids.push_back(pool[e].pid);
-                if (has_own) o.rep_owners.push_back(pool[e].own);
+            o.snap_tptr.push_back((int32_t)o.snap_nodes.size());
+            for (auto& row : l2) {
+                o.recv_nodes.push_back(row.node);
+                for (int32_t e = row.head; e >= 0; e = pool[e].next) {
+                    o.del_slots.push_back(pool[e].slot);
+                    o.reply_slots.push_back(pool[e].reply);
+                    if (has_pid) o.del_pids.push_back(pool[e].pid);
+                    if (has_own) o.del_owners.push_back(pool[e].own);
+                }
+                o.recv_nptr.push_back((int32_t)o.del_slots.size());
             }
-            o.rep_nptr.push_back((int32_t)o.rep_slots.size());
+            o.recv_tptr.push_back((int32_t)o.recv_nodes.size());
+            for (auto& row : l3) {
+                o.rep_nodes.push_back(row.node);
+                for (int32_t e = row.head; e >= 0; e = pool[e].next) {
+                    o.rep_slots.push_back(pool[e].slot);
+                    o.rep_reply_slots.push_back(pool[e].reply);
+                    if (has_pid) o.rep_pids.push_back(pool[e].pid);
+                    if (has_own) o.rep_owners.push_back(pool[e].own);
+                }
+                o.rep_nptr.push_back((int32_t)o.rep_slots.size());
+            }
+            o.rep_tptr.push_back((int32_t)o.rep_nodes.size());
+            g_snap.clear(); l2.clear(); l3.clear();
+            ++gid;
         }
-        o.rep_tptr.push_back((int32_t)o.rep_nodes.size());
-        g_snap.clear(); l2.clear(); l3.clear();
-        ++gid;  // invalidates every stamped entry at once
-    };
 
-    std::function<void(int32_t, int32_t)> place_snap;
-    std::function<void(int32_t, int32_t, int32_t, int32_t, int32_t)>
-        place_delivery;
-
-    auto defer = [&](int32_t node, const DEv& ev) {
-        is_def[node] = 1;
-        deferred.emplace_back(node, ev);
-        if (ev.kind == 0) {
-            w_stamp[ev.slot] = gid; w_lvl[ev.slot] = 9;
-            t_stamp[ev.slot] = gid;
-        } else {
-            t_stamp[ev.slot] = gid;
-            if (ev.reply >= 0) {
-                w_stamp[ev.reply] = gid; w_lvl[ev.reply] = 9;
-                t_stamp[ev.reply] = gid;
+        void defer(int32_t node, const DEv& ev)
+        {
+            is_def[node] = 1;
+            deferred.emplace_back(node, ev);
+            if (ev.kind == 0) {
+                w_stamp[ev.slot] = gid; w_lvl[ev.slot] = 9;
+                t_stamp[ev.slot] = gid;
+            } else {
+                t_stamp[ev.slot] = gid;
+                if (ev.reply >= 0) {
+                    w_stamp[ev.reply] = gid; w_lvl[ev.reply] = 9;
+                    t_stamp[ev.reply] = gid;
+                }
             }
         }
-    };
 
-    auto close = [&]() {
-        if (in_replay) {
+        void close()
+        {
+            if (in_replay) {
+                flush();
+                return;
+            }
             flush();
-            return;
-        }
-        flush();
-        while (!deferred.empty()) {
-            in_replay = true;
-            std::vector<std::pair<int32_t, DEv>> dl;
-            dl.swap(deferred);
-            for (auto& p : dl) is_def[p.first] = 0;
-            for (auto& p : dl) {
-                const DEv& ev = p.second;
-                if (ev.kind == 0) place_snap(p.first, ev.slot);
-                else place_delivery(p.first, ev.slot, ev.reply, ev.pid, ev.own);
+            while (!deferred.empty()) {
+                in_replay = true;
+                std::vector<std::pair<int32_t, DEv>> dl;
+                dl.swap(deferred);
+                for (auto& p : dl) is_def[p.first] = 0;
+                for (auto& p : dl) {
+                    const DEv& ev = p.second;
+                    if (ev.kind == 0) place_snap(p.first, ev.slot);
+                    else
+                        place_delivery(p.first, ev.slot, ev.reply, ev.pid,
+                                       ev.own);
+                }
+                in_replay = false;
+                if (!deferred.empty()) flush();  // next pass = fresh group
             }
-            in_replay = false;
-            if (!deferred.empty()) flush();  // next drain pass = fresh group
         }
-    };
 
-    place_snap = [&](int32_t node, int32_t slot) {
-        if (is_def[node]) { defer(node, {0, slot, -1, -1, -1}); return; }
-        if (t_stamp[slot] == gid) {
-            close();
-            if (is_def[node]) {  // replay re-deferred this node
-                defer(node, {0, slot, -1, -1, -1});
+        void place_snap(int32_t node, int32_t slot)
+        {
+            if (is_def[node]) { defer(node, {0, slot, -1, -1, -1}); return; }
+            if (t_stamp[slot] == gid) {
+                close();
+                if (is_def[node]) {  // replay re-deferred this node
+                    defer(node, {0, slot, -1, -1, -1});
+                    return;
+                }
+            }
+            if (l_stamp[node] != gid) {
+                g_snap.emplace_back(node, slot);
+                w_stamp[slot] = gid; w_lvl[slot] = 1;
+            } else {
+                auto& lst = (l_kind[node] == 2) ? l2 : l3;
+                Ev& last = pool[lst[l_idx[node]].tail];
+                if (last.reply >= 0) {
+                    defer(node, {0, slot, -1, -1, -1});
+                    return;
+                }
+                last.reply = slot;
+                w_stamp[slot] = gid;
+                w_lvl[slot] = (l_kind[node] == 2) ? 2 : 3;
+            }
+            t_stamp[slot] = gid;
+        }
+
+        void place_delivery(int32_t node, int32_t slot, int32_t reply,
+                            int32_t pid, int32_t own)
+        {
+            if (is_def[node]) {
+                defer(node, {1, slot, reply, pid, own});
                 return;
             }
-        }
-        if (l_stamp[node] != gid) {
-            g_snap.emplace_back(node, slot);
-            w_stamp[slot] = gid; w_lvl[slot] = 1;
-        } else {
-            auto& lst = (l_kind[node] == 2) ? l2 : l3;
-            Ev& last = pool[lst[l_idx[node]].tail];
-            if (last.reply >= 0) {
-                defer(node, {0, slot, -1, -1, -1});
+            int lvl = (w_stamp[slot] == gid) ? w_lvl[slot] : 0;
+            if (lvl >= 3 || (reply >= 0 && t_stamp[reply] == gid)) {
+                defer(node, {1, slot, reply, pid, own});
                 return;
             }
-            last.reply = slot;
-            w_stamp[slot] = gid;
-            w_lvl[slot] = (l_kind[node] == 2) ? 2 : 3;
-        }
-        t_stamp[slot] = gid;
-    };
-
-    place_delivery = [&](int32_t node, int32_t slot, int32_t reply,
-                         int32_t pid, int32_t own) {
-        if (is_def[node]) { defer(node, {1, slot, reply, pid, own}); return; }
-        int lvl = (w_stamp[slot] == gid) ? w_lvl[slot] : 0;
-        if (lvl >= 3 || (reply >= 0 && t_stamp[reply] == gid)) {
-            defer(node, {1, slot, reply, pid, own});
-            return;
-        }
-        bool in_l3 = (l_stamp[node] == gid && l_kind[node] == 3);
-        int wl;
-        if (in_l3) {
-            append_ev(l3[l_idx[node]], slot, reply, pid, own);
-            wl = 3;
-        } else if (lvl >= 2) {
-            int32_t ri;
-            if (r3_stamp[node] != gid) {
-                ri = (int32_t)l3.size();
-                r3_stamp[node] = gid; r3_idx[node] = ri;
-                l3.push_back({node, -1, -1});
+            bool in_l3 = (l_stamp[node] == gid && l_kind[node] == 3);
+            int wl;
+            if (in_l3) {
+                append_ev(l3[l_idx[node]], slot, reply, pid, own);
+                wl = 3;
+            } else if (lvl >= 2) {
+                int32_t ri;
+                if (r3_stamp[node] != gid) {
+                    ri = (int32_t)l3.size();
+                    r3_stamp[node] = gid; r3_idx[node] = ri;
+                    l3.push_back({node, -1, -1});
+                } else {
+                    ri = r3_idx[node];
+                }
+                append_ev(l3[ri], slot, reply, pid, own);
+                l_stamp[node] = gid; l_kind[node] = 3; l_idx[node] = ri;
+                wl = 3;
             } else {
-                ri = r3_idx[node];
+                int32_t ri;
+                if (r2_stamp[node] != gid) {
+                    ri = (int32_t)l2.size();
+                    r2_stamp[node] = gid; r2_idx[node] = ri;
+                    l2.push_back({node, -1, -1});
+                    l_stamp[node] = gid; l_kind[node] = 2; l_idx[node] = ri;
+                } else {
+                    ri = r2_idx[node];
+                }
+                append_ev(l2[ri], slot, reply, pid, own);
+                wl = 2;
             }
-            append_ev(l3[ri], slot, reply, pid, own);
-            l_stamp[node] = gid; l_kind[node] = 3; l_idx[node] = ri;
-            wl = 3;
-        } else {
-            int32_t ri;
-            if (r2_stamp[node] != gid) {
-                ri = (int32_t)l2.size();
-                r2_stamp[node] = gid; r2_idx[node] = ri;
-                l2.push_back({node, -1, -1});
-                l_stamp[node] = gid; l_kind[node] = 2; l_idx[node] = ri;
-            } else {
-                ri = r2_idx[node];
+            t_stamp[slot] = gid;
+            if (reply >= 0) {
+                w_stamp[reply] = gid; w_lvl[reply] = (int8_t)wl;
+                t_stamp[reply] = gid;
             }
-            append_ev(l2[ri], slot, reply, pid, own);
-            wl = 2;
         }
-        t_stamp[slot] = gid;
-        if (reply >= 0) {
-            w_stamp[reply] = gid; w_lvl[reply] = (int8_t)wl;
-            t_stamp[reply] = gid;
-        }
-    };
+    } P{g_snap, pool, l2, l3, t_stamp, w_stamp, w_lvl, l_stamp, l_kind,
+       l_idx, r2_stamp, r2_idx, r3_stamp, r3_idx, gid, is_def, deferred,
+       false, o, has_pid, has_own};
 
     for (int64_t t = 0; t < delta; ++t) {
         for (int32_t i = snap_tptr[t]; i < snap_tptr[t + 1]; ++i)
-            place_snap(snap_nodes[i], snap_slots[i]);
+            P.place_snap(snap_nodes[i], snap_slots[i]);
         for (int32_t r = recv_tptr[t]; r < recv_tptr[t + 1]; ++r) {
             int32_t x = recv_nodes[r];
             for (int32_t d = recv_nptr[r]; d < recv_nptr[r + 1]; ++d) {
@@ -424,11 +456,11 @@ static py::dict pack_round(
                     (has_pid && !del_pids.empty()) ? del_pids[d] : -1;
                 int32_t own =
                     (has_own && !del_owners.empty()) ? del_owners[d] : -1;
-                place_delivery(x, del_slots[d], reply_slots[d], pid, own);
+                P.place_delivery(x, del_slots[d], reply_slots[d], pid, own);
             }
         }
         for (int32_t i = pull_tptr[t]; i < pull_tptr[t + 1]; ++i)
-            place_snap(pull_nodes[i], pull_slots[i]);
+            P.place_snap(pull_nodes[i], pull_slots[i]);
         for (int32_t r = rep_tptr[t]; r < rep_tptr[t + 1]; ++r) {
             int32_t x = rep_nodes[r];
             for (int32_t d = rep_nptr[r]; d < rep_nptr[r + 1]; ++d) {
@@ -436,14 +468,14 @@ static py::dict pack_round(
                     (has_pid && !rep_pids.empty()) ? rep_pids[d] : -1;
                 int32_t own =
                     (has_own && !rep_owners.empty()) ? rep_owners[d] : -1;
-                place_delivery(x, rep_slots[d], -1, pid, own);
+                P.place_delivery(x, rep_slots[d], -1, pid, own);
             }
         }
     }
     int guard = 0;
     while ((!g_snap.empty() || !l2.empty() || !l3.empty() ||
             !deferred.empty()) && guard < 100000) {
-        close();
+        P.close();
         ++guard;
     }
 
