@@ -1029,13 +1029,13 @@ class HIPBackend(TorchBackend):
         for unsupported families."""
         if (
             spec.family in ("logreg", "pegasos", "adaline")
-            and len(local_ids) * gx.shape[0] ** 2 <= 20_000_000
+            and gx.shape[0] <= 2048
         ):
-            # the kernel's pairwise AUC costs ~30 ns per (node, pair)
-            # unit: R x n_eval^2 <= 2e7 keeps it under ~0.6 ms (sampled
-            # eval sets are ~1e6). Beyond that (500 nodes x 2300 samples
-            # was 58% of the 50k-node round) the torch O(n log n) sort
-            # path takes over
+            # blocks run concurrently, so the kernel's wall time tracks the
+            # PER-BLOCK pairwise-AUC work (~n_eval^2): n_eval <= 2048 keeps
+            # a block under ~100 us at any sampled-node count. Bigger eval
+            # sets (2300 samples was 58% of the 50k-node round) take the
+            # torch O(n log n) sort path
             is_margin = spec.family in ("pegasos", "adaline")
             out = self.ext.eval_metrics(
                 state.params,
@@ -1049,7 +1049,7 @@ class HIPBackend(TorchBackend):
             )
         elif (
             spec.family in ("mlp", "torchmod")
-            and len(local_ids) * gx.shape[0] ** 2 <= 20_000_000
+            and gx.shape[0] <= 2048
         ):
             sc = self.scores(state, spec, local_ids, gx)
             out = self.ext.eval_metrics_scores(
@@ -1097,13 +1097,13 @@ class HIPBackend(TorchBackend):
         kernel epilogue."""
         if (
             spec.family in ("logreg", "pegasos", "adaline")
-            and len(local_ids) * gx.shape[0] ** 2 <= 20_000_000
+            and gx.shape[0] <= 2048
         ):
-            # the kernel's pairwise AUC costs ~30 ns per (node, pair)
-            # unit: R x n_eval^2 <= 2e7 keeps it under ~0.6 ms (sampled
-            # eval sets are ~1e6). Beyond that (500 nodes x 2300 samples
-            # was 58% of the 50k-node round) the torch O(n log n) sort
-            # path takes over
+            # blocks run concurrently, so the kernel's wall time tracks the
+            # PER-BLOCK pairwise-AUC work (~n_eval^2): n_eval <= 2048 keeps
+            # a block under ~100 us at any sampled-node count. Bigger eval
+            # sets (2300 samples was 58% of the 50k-node round) take the
+            # torch O(n log n) sort path
             is_margin = spec.family in ("pegasos", "adaline")
             out = self.ext.eval_metrics(
                 state.params,
@@ -1117,7 +1117,7 @@ class HIPBackend(TorchBackend):
             )
         elif (
             spec.family in ("mlp", "torchmod")
-            and len(local_ids) * gx.shape[0] ** 2 <= 20_000_000
+            and gx.shape[0] <= 2048
         ):
             sc = self.scores(state, spec, local_ids, gx)
             out = self.ext.eval_metrics_scores(
