@@ -617,6 +617,7 @@ protected:
         }
         row_stamp_.assign((size_t)n_, INT64_MIN);
         row_idx_.assign((size_t)n_, 0);
+        slot_reuse_lag_ = std::max<int64_t>(32, delta_);
     }
 
     void pend_push(int64_t t, const Msg& m)
@@ -651,11 +652,13 @@ protected:
         }
         return out;
     }
-    // consumed slots re-enter circulation only kSlotReuseLag ticks after
-    // the tick that freed them (FIFO), mirroring Scheduler.SLOT_REUSE_LAG:
-    // adjacent ticks never alias slot ids, so the runner can fuse
-    // conflict-free ticks into single launch groups.
-    static constexpr int64_t kSlotReuseLag = 32;
+    // consumed slots re-enter circulation only slot_reuse_lag_ ticks after
+    // the tick that freed them (FIFO), mirroring Scheduler.SLOT_REUSE_LAG.
+    // The effective lag is max(32, delta): launch groups never span a
+    // round, so a lag of >= one full round makes every packed group
+    // structurally alias-free (the packers' touched-set hazard checks
+    // remain as defense-in-depth). Set in pend_init() once delta_ is known.
+    int64_t slot_reuse_lag_ = 32;
     std::deque<std::pair<int64_t, int32_t>> reuse_q_;
     int64_t next_slot_ = 0;
     std::vector<int32_t> slot_owner_;
@@ -663,7 +666,7 @@ protected:
     int32_t alloc_slot(int32_t owner, int64_t t)
     {
         int32_t s;
-        if (!reuse_q_.empty() && reuse_q_.front().first + kSlotReuseLag <= t) {
+        if (!reuse_q_.empty() && reuse_q_.front().first + slot_reuse_lag_ <= t) {
             s = reuse_q_.front().second;
             reuse_q_.pop_front();
         } else {

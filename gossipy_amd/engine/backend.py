@@ -1058,8 +1058,12 @@ class HIPBackend(TorchBackend):
         else:
             return None
         # double-buffered pinned staging: at most one handle is in flight
-        bufs = self._eval_bufs.setdefault(tuple(out.shape), [None, None])
-        self._eval_flip = flip = getattr(self, "_eval_flip", 0) ^ 1
+        # per output shape. The flip lives WITH the buffer pair (slot 2):
+        # a single instance-global toggle would let two interleaving
+        # shapes reuse a buffer whose handle is still pending in the
+        # runner's 2-deep eval deque (silent metric corruption).
+        bufs = self._eval_bufs.setdefault(tuple(out.shape), [None, None, 0])
+        bufs[2] = flip = bufs[2] ^ 1
         if bufs[flip] is None or bufs[flip].shape != out.shape:
             bufs[flip] = torch.empty(out.shape, dtype=out.dtype,
                                      pin_memory=True)

@@ -1482,15 +1482,42 @@ class BatchedGossipSimulator(SimulationEventSender):
         with open(filename, "wb") as f:
             dill.dump(blob, f)
 
+    def _ctor_fingerprint(self) -> tuple:
+        """Identifies the subclass constructor arguments the checkpoint
+        blob does NOT carry (token accounts, PENS hyperparameters,
+        mixing matrices…). Stored on save and validated on load, so a
+        resume with different arguments fails loudly instead of silently
+        replaying a different schedule. Subclasses set ``self._ctor_desc``
+        in ``__init__``."""
+        return getattr(self, "_ctor_desc", ())
+
     def _checkpoint_extra(self) -> dict:
-        return {}
+        return {"ctor": self._ctor_fingerprint()}
 
     def _restore_extra(self, extra: dict) -> None:
         pass
 
+    def _replay_rounds(self, rounds_done: int, extra: dict) -> None:
+        """Replay the deterministic schedule to restore carry-over state
+        (in-flight messages, slot allocator, token accounts). Subclasses
+        whose schedule depends on device-side results (PENS) override
+        this to re-inject the saved results at the right round."""
+        for r in range(rounds_done):
+            if hasattr(self.scheduler, "next_round_flat"):
+                self.scheduler.next_round_flat(r)
+            else:
+                self.scheduler.next_round(r)
+
     @classmethod
     def load(cls, filename: str, device: Optional[torch.device] = None, **kw):
-        """Restore a checkpoint written by :meth:`save` (same world size)."""
+        """Restore a checkpoint written by :meth:`save` (same world size).
+
+        Subclass constructor arguments (``token_account``, PENS
+        ``n_sampled``/``m_top``/``step1_rounds``, ``mixing``…) are NOT
+        stored in the blob — pass them again via ``**kw`` exactly as in
+        the original run. A fingerprint of those arguments IS stored, and
+        a mismatch raises ``ValueError`` rather than silently replaying a
+        different schedule."""
         import dill
 
         with open(filename, "rb") as f:
@@ -1504,6 +1531,13 @@ class BatchedGossipSimulator(SimulationEventSender):
             *[d[k].to(device) if d[k] is not None else None for k in ("gx", "gy")],
         )
         sim = cls(blob["cfg"], blob["spec"], data, device=device, **kw)
+        saved_ctor = blob["extra"].get("ctor", ())
+        if tuple(saved_ctor) != tuple(sim._ctor_fingerprint()):
+            raise ValueError(
+                "checkpoint constructor mismatch: saved "
+                f"{saved_ctor!r} vs supplied {sim._ctor_fingerprint()!r} — "
+                "pass the same constructor arguments the saved run used"
+            )
         sim.state.params.copy_(blob["params"].to(device))
         sim.state.ages.copy_(blob["ages"].to(device))
         sim.pool.ensure(blob["pool_slots"].shape[0])
@@ -1513,13 +1547,7 @@ class BatchedGossipSimulator(SimulationEventSender):
         sim.pool.slot_ages[: blob["pool_ages"].shape[0]].copy_(
             blob["pool_ages"].to(device)
         )
-        # replay the deterministic schedule to restore carry-over state
-        # (in-flight messages, slot allocator, token accounts)
-        for r in range(blob["rounds_done"]):
-            if hasattr(sim.scheduler, "next_round_flat"):
-                sim.scheduler.next_round_flat(r)
-            else:
-                sim.scheduler.next_round(r)
+        sim._replay_rounds(blob["rounds_done"], blob["extra"])
         sim.rounds_done = blob["rounds_done"]
         sim._restore_extra(blob["extra"])
         sim.initialized = True
@@ -1562,6 +1590,17 @@ class BatchedTokenizedGossipSimulator(BatchedGossipSimulator):
         super().__init__(cfg, spec, data, device=device)
         from .schedule import NativeTokenizedAdapter
 
+        self._ctor_desc = (
+            "tokenized",
+            type(token_account).__name__,
+            tuple(sorted(
+                (k, v) for k, v in vars(token_account).items()
+                if isinstance(v, (int, float, str, bool))
+            )),
+            utility_fun
+            if (utility_fun is None or isinstance(utility_fun, int))
+            else "callable:" + getattr(utility_fun, "__name__", "?"),
+        )
         self.scheduler = None
         if (utility_fun is None or isinstance(utility_fun, int)) and self._flat_exec_ok():
             try:
@@ -1627,6 +1666,9 @@ class BatchedAll2AllGossipSimulator(BatchedGossipSimulator):
         )
         from .schedule import All2AllScheduler
 
+        self._ctor_desc = (
+            "all2all", type(mixing).__name__ if mixing is not None else None
+        )
         self.scheduler = All2AllScheduler(cfg, mixing)
         self._flat_schedulable = False
 
@@ -1719,11 +1761,63 @@ class BatchedPENSGossipSimulator(BatchedGossipSimulator):
         assert spec.mode == CreateModelMode.MERGE_UPDATE, (
             "PENSNode can only be used with MERGE_UPDATE mode."
         )
+        self._ctor_desc = ("pens", n_sampled, m_top, step1_rounds)
         self.scheduler = PENSScheduler(cfg, n_sampled, m_top, step1_rounds)
         self.m_top = m_top
         #: device-side winner counters [n_local, n_nodes]
         self.counts = torch.zeros(
             self.n_local, cfg.n_nodes, dtype=torch.int32, device=self.device
+        )
+
+    def _checkpoint_extra(self) -> dict:
+        extra = super()._checkpoint_extra()
+        extra["pens_counts"] = self.counts.cpu()
+        extra["pens_selected"] = self.scheduler.selected.copy()
+        extra["pens_best_nodes"] = (
+            None
+            if self.scheduler.best_nodes is None
+            else [b.copy() for b in self.scheduler.best_nodes]
+        )
+        return extra
+
+    def _replay_rounds(self, rounds_done: int, extra: dict) -> None:
+        # the PENS schedule is data-dependent: step-2 peer draws read
+        # best_nodes, which came from device-side winner counts. Replay
+        # step-1 rounds with best_nodes=None (as the saved run did), then
+        # re-inject the SAVED best_nodes before replaying step-2 rounds —
+        # otherwise the replay takes the step-1 draw path and diverges.
+        boundary = min(rounds_done, self.scheduler.step1_rounds)
+        for r in range(boundary):
+            self.scheduler.next_round(r)
+        if rounds_done > self.scheduler.step1_rounds or (
+            rounds_done == self.scheduler.step1_rounds
+            and extra.get("pens_best_nodes") is not None
+        ):
+            saved = extra.get("pens_best_nodes")
+            if saved is None:
+                raise ValueError(
+                    "PENS checkpoint past the step-1 boundary has no saved "
+                    "best_nodes — blob predates PENS checkpoint support"
+                )
+            self.scheduler.best_nodes = [
+                np.asarray(b, dtype=np.int64) for b in saved
+            ]
+            for r in range(boundary, rounds_done):
+                self.scheduler.next_round(r)
+
+    def _restore_extra(self, extra: dict) -> None:
+        if "pens_counts" not in extra:
+            raise ValueError(
+                "blob predates PENS checkpoint support (no pens_counts)"
+            )
+        self.counts.copy_(extra["pens_counts"].to(self.device))
+        self.scheduler.selected = np.asarray(
+            extra["pens_selected"], dtype=np.int64
+        )
+        saved = extra.get("pens_best_nodes")
+        self.scheduler.best_nodes = (
+            None if saved is None
+            else [np.asarray(b, dtype=np.int64) for b in saved]
         )
 
     def _fast_path_ok(self) -> bool:

@@ -203,14 +203,23 @@ class Scheduler:
         self._reuse_q: deque = deque()  # (free_tick, slot), tick-ordered
         self._next_slot = 0
         self.slot_owner = np.zeros(64, dtype=np.int32)
+        # effective reuse lag >= one full round: packed launch groups never
+        # span a round boundary, so no slot id can recycle into a
+        # still-open group (see SLOT_REUSE_LAG doc below). Must match the
+        # C++ scheduler (csrc/scheduler.cpp pend_init).
+        self.SLOT_REUSE_LAG = max(type(self).SLOT_REUSE_LAG, cfg.delta)
 
     #: a consumed slot id is only reused SLOT_REUSE_LAG ticks after the tick
     #: that freed it. Adjacent ticks therefore never alias slot ids, which is
     #: what lets the runner fuse conflict-free ticks into single launch
     #: groups (the old LIFO free list re-issued a slot on the very next
-    #: tick, making every tick pair conflict); 32 keeps even the entry-level
-    #: packer's long groups alias-free. Costs ~delay-window x firing-rate
-    #: extra pool rows — negligible against 288 GB of HBM3E.
+    #: tick, making every tick pair conflict). The effective lag is
+    #: ``max(32, delta)`` (set per instance in ``__init__``): launch groups
+    #: never span a round, so a lag of at least one full round makes
+    #: every group structurally alias-free — the packers' touched-set
+    #: checks become defense-in-depth instead of the only guarantee.
+    #: Costs ~delay-window x firing-rate extra pool rows — negligible
+    #: against 288 GB of HBM3E.
     SLOT_REUSE_LAG = 32
 
     def _alloc_slot(self, owner: int, t: int) -> int:

@@ -822,6 +822,91 @@ class TestEngineCheckpoint:
         restored.start(n_rounds=3)
         assert torch.equal(ref.local_params(), restored.local_params())
 
+    def test_load_ctor_mismatch_raises(self, tmp_path):
+        from gossipy_amd.engine import BatchedTokenizedGossipSimulator
+        from gossipy_amd.flow_control import (
+            RandomizedTokenAccount,
+            SimpleTokenAccount,
+        )
+
+        shards, geval = _make_data(30, seed=1)
+        data = DataArena.from_shards(
+            shards, torch.device("cpu"), global_eval=geval
+        )
+        cfg = EngineConfig(
+            n_nodes=30, delta=10, protocol=AntiEntropyProtocol.PUSH,
+            model_size=116, sampling_eval=0.0, seed=21,
+        )
+        sim = BatchedTokenizedGossipSimulator(
+            cfg, LogRegSpec(d_in=57, n_classes=2, lr=0.1), data,
+            token_account=RandomizedTokenAccount(C=20, A=10),
+        )
+        sim.init_nodes()
+        sim.start(n_rounds=2)
+        f = str(tmp_path / "ckpt_mismatch.dill")
+        sim.save(f)
+        with pytest.raises(ValueError, match="constructor mismatch"):
+            BatchedTokenizedGossipSimulator.load(
+                f, device=torch.device("cpu"),
+                token_account=SimpleTokenAccount(C=4),
+            )
+
+    def _pens(self, step1_rounds=3):
+        from gossipy_amd.engine import BatchedPENSGossipSimulator
+
+        shards, geval = _make_data(16, seed=3)
+        data = DataArena.from_shards(
+            shards, torch.device("cpu"), global_eval=geval
+        )
+        cfg = EngineConfig(
+            n_nodes=16, delta=10, protocol=AntiEntropyProtocol.PUSH,
+            model_size=116, sampling_eval=0.0, seed=7,
+        )
+        sim = BatchedPENSGossipSimulator(
+            cfg, LogRegSpec(d_in=57, n_classes=2, lr=0.1), data,
+            n_sampled=4, m_top=2, step1_rounds=step1_rounds,
+        )
+        sim.init_nodes()
+        return sim
+
+    @pytest.mark.parametrize("save_at", [2, 3, 5])
+    def test_save_load_pens(self, tmp_path, save_at):
+        """PENS resume is bit-exact whether the checkpoint lands before,
+        at, or after the step-1 boundary (step1_rounds=3): device winner
+        counts, selected counters and best_nodes all survive."""
+        total = 7
+        ref = self._pens()
+        ref.start(n_rounds=total)
+
+        sim = self._pens()
+        sim.start(n_rounds=save_at)
+        f = str(tmp_path / f"ckpt_pens_{save_at}.dill")
+        sim.save(f)
+        from gossipy_amd.engine import BatchedPENSGossipSimulator
+
+        restored = BatchedPENSGossipSimulator.load(
+            f, device=torch.device("cpu"),
+            n_sampled=4, m_top=2, step1_rounds=3,
+        )
+        assert torch.equal(restored.counts, sim.counts)
+        assert np.array_equal(
+            restored.scheduler.selected, sim.scheduler.selected
+        )
+        if sim.scheduler.best_nodes is None:
+            assert restored.scheduler.best_nodes is None
+        else:
+            for a, b in zip(
+                restored.scheduler.best_nodes, sim.scheduler.best_nodes
+            ):
+                assert np.array_equal(a, b)
+        restored.start(n_rounds=total - save_at)
+        assert torch.equal(ref.local_params(), restored.local_params())
+        assert torch.equal(ref.counts, restored.counts)
+        for a, b in zip(
+            ref.scheduler.best_nodes, restored.scheduler.best_nodes
+        ):
+            assert np.array_equal(a, b)
+
 
 # ---------------------------------------------------------------------------
 # sampled gossip (SamplingBasedNode / SamplingTMH)
