@@ -209,12 +209,13 @@ class TorchModelHandler(ModelHandler):
             self._local_step(x[perm][:batch_size], y[perm][:batch_size])
 
     def _local_step(self, x: torch.Tensor, y: torch.Tensor) -> None:
+        # one optimizer step on one minibatch (semantics of
+        # gossipy/model/handler.py:250-258, pinned by the engine's
+        # bit-exact oracle tests)
         self.model.train()
-        x, y = x.to(self.device), y.to(self.device)
-        y_pred = self.model(x)
-        loss = self.criterion(y_pred, y)
         self.optimizer.zero_grad()
-        loss.backward()
+        out = self.model(x.to(self.device))
+        self.criterion(out, y.to(self.device)).backward()
         self.optimizer.step()
         self.n_updates += 1
 
@@ -484,41 +485,52 @@ class MFModelHandler(ModelHandler):
         self.n_updates = 1
 
     def init(self, r_min: int = 1, r_max: int = 5) -> None:
-        mul = np.sqrt((r_max - r_min) / self.k)
-        X = np.random.rand(1, self.k) * mul
-        Y = np.random.rand(self.n_items, self.k) * mul
-        b = r_min / 2.0
-        c = np.ones(self.n_items) * r_min / 2.0
-        self.model = ((X, b), (Y, c))
+        # factors uniform in [0, sqrt(range/k)) so a user.item dot product
+        # spans the rating range; both biases start at half the minimum
+        # rating (gossipy/model/handler.py:542-548)
+        scale = np.sqrt((r_max - r_min) / self.k)
+        user = np.random.rand(1, self.k) * scale
+        items = np.random.rand(self.n_items, self.k) * scale
+        user_bias = r_min / 2.0
+        item_bias = np.full(self.n_items, r_min / 2.0)
+        self.model = ((user, user_bias), (items, item_bias))
 
     def _update(self, data) -> None:
-        (X, b), (Y, c) = self.model
-        for i, r in data:
-            i = int(i)
-            err = (r - np.dot(X, Y[i].T) - b - c[i])[0]
-            Y[i] = (1.0 - self.reg * self.lr) * Y[i] + self.lr * err * X
-            X = (1.0 - self.reg * self.lr) * X + self.lr * err * Y[i]
-            b += self.lr * err
-            c[i] += self.lr * err
+        """Per-rating SGD sweep. Update order is behavior (pinned by the
+        K9 kernel's bit-exact tests): the item row moves against the OLD
+        user row, the user row against the NEW item row, then both biases
+        (gossipy/model/handler.py:550-560)."""
+        (user, user_bias), (items, item_bias) = self.model
+        decay = 1.0 - self.reg * self.lr
+        for item_id, rating in data:
+            j = int(item_id)
+            err = float(rating - user @ items[j] - user_bias - item_bias[j])
+            step = self.lr * err
+            items[j] = decay * items[j] + step * user
+            user = decay * user + step * items[j]
+            user_bias += step
+            item_bias[j] += step
             self.n_updates += 1
-        self.model = ((X, b), (Y, c))
+        self.model = ((user, user_bias), (items, item_bias))
 
     def _merge(self, other_model_handler: "MFModelHandler") -> None:
-        _, (Y1, c1) = other_model_handler.model
-        (X, b), (Y, c) = self.model
-        den = self.n_updates + other_model_handler.n_updates
-        Y = (Y * self.n_updates + Y1 * other_model_handler.n_updates) / (2.0 * den)
-        c = (c * self.n_updates + c1 * other_model_handler.n_updates) / (2.0 * den)
-        self.model = (X, b), (Y, c)
+        """Age-weighted average of the ITEM side only — the user side is
+        private. The reference's extra factor of 2 in the denominator
+        (gossipy/model/handler.py:566-567) is kept: it changes learning
+        curves, so it counts as behavior, not a bug."""
+        (user, user_bias), (items, item_bias) = self.model
+        their_items, their_item_bias = other_model_handler.model[1]
+        w_own, w_other = self.n_updates, other_model_handler.n_updates
+        halved_total = 2.0 * (w_own + w_other)
+        items = (items * w_own + their_items * w_other) / halved_total
+        item_bias = (item_bias * w_own + their_item_bias * w_other) / halved_total
+        self.model = (user, user_bias), (items, item_bias)
 
     def evaluate(self, ratings) -> Dict[str, float]:
-        (X, b), (Y, c) = self.model
-        R = (np.dot(X, Y.T) + b + c)[0]
-        return {
-            "rmse": float(
-                np.sqrt(np.mean([(r - R[int(i)]) ** 2 for i, r in ratings]))
-            )
-        }
+        (user, user_bias), (items, item_bias) = self.model
+        pred = (user @ items.T + user_bias + item_bias)[0]
+        sq_err = [(r - pred[int(i)]) ** 2 for i, r in ratings]
+        return {"rmse": float(np.sqrt(np.mean(sq_err)))}
 
     def get_size(self) -> int:
         return self.k * (self.n_items + 1)
