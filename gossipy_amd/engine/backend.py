@@ -457,10 +457,18 @@ class TorchBackend:
     # -- arbitrary nn.Module family (CNN path) ----------
 
     def _update_torchmod(self, params, ages, data, spec, nodes) -> None:
-        """Autograd local SGD on a per-node view of the arena row
-        (TorchModelHandler._update/_local_step semantics,
-        gossipy/model/handler.py:235-258). Convolutions run on MIOpen via
-        torch — the measured-right vendor path for small convs."""
+        """Autograd local SGD for arbitrary ``nn.Module`` families.
+
+        Node-BATCHED by default: per-node parameter stacks trained with
+        ``torch.func.vmap(grad(...))`` — one grouped-conv/bmm launch set
+        per minibatch step services every node at once (convolutions land
+        on MIOpen grouped conv, the vendor path). Nodes are grouped by
+        shard length so each vmap batch shares its minibatch split.
+        ``GOSSIPY_TORCHMOD_LOOP=1`` forces the per-node reference loop
+        (the A/B + oracle for the batched path)."""
+        if len(nodes) > 1 and os.environ.get("GOSSIPY_TORCHMOD_LOOP") != "1":
+            self._update_torchmod_batched(params, ages, data, spec, nodes)
+            return
         module = spec.template().to(params.device)
         crit = torch.nn.CrossEntropyLoss()
         for idx in nodes.tolist():
@@ -483,12 +491,71 @@ class TorchBackend:
                     ages[idx] += 1
             spec.store_row(module, params[idx])
 
+    def _update_torchmod_batched(self, params, ages, data, spec, nodes) -> None:
+        """One SGD trajectory per node, all nodes at once (the VERDICT r1
+        CNN item: the per-node loop re-built an optimizer per node per
+        tick and ran ~B times more tiny conv launches)."""
+        import torch.func as tfunc
+
+        module = spec.template().to(params.device)
+        layout = spec.param_layout()
+        counts_all = data.counts[nodes].long()
+        lr, wd = spec.lr, spec.weight_decay
+
+        def loss_fn(pd, xb, yb):
+            out = tfunc.functional_call(module, pd, (xb,))
+            return torch.nn.functional.cross_entropy(out, yb)
+
+        gfn = tfunc.vmap(tfunc.grad(loss_fn))
+        for c in torch.unique(counts_all).tolist():
+            c = int(c)
+            if c == 0:
+                continue
+            sel = nodes[counts_all == c]
+            B = len(sel)
+            rows = params[sel]  # fancy index -> private copy
+            stacked = {
+                name: rows[:, o : o + n].view(B, *shape)
+                for name, shape, o, n in layout
+            }
+            x = data.x[sel, :c].view(B, c, *spec.input_shape)
+            y = data.y[sel, :c].long()
+            bs = c if spec.batch_size == 0 else spec.batch_size
+            for _ in range(max(1, spec.local_epochs)):
+                for s in range(0, c, bs):
+                    grads = gfn(stacked, x[:, s : s + bs], y[:, s : s + bs])
+                    with torch.no_grad():
+                        for name, _, _, _ in layout:
+                            g = grads[name]
+                            p = stacked[name]
+                            if wd:
+                                g = g.add(p, alpha=wd)
+                            p.add_(g, alpha=-lr)
+                    ages[sel] += 1
+            params[sel] = rows
+
     def torchmod_scores(self, state, spec, nodes, X) -> torch.Tensor:
-        """Class scores ``[R, n, k]`` for the eval sweep."""
+        """Class scores ``[R, n, k]`` for the eval sweep (node-batched
+        forward via vmap; ``GOSSIPY_TORCHMOD_LOOP=1`` forces the loop)."""
         module = spec.template().to(state.params.device)
+        xin = X.view(X.shape[0], *spec.input_shape)
+        if len(nodes) > 1 and os.environ.get("GOSSIPY_TORCHMOD_LOOP") != "1":
+            import torch.func as tfunc
+
+            rows = state.params[nodes.long()]
+            R = rows.shape[0]
+            stacked = {
+                name: rows[:, o : o + n].view(R, *shape)
+                for name, shape, o, n in spec.param_layout()
+            }
+
+            def fwd(pd):
+                return tfunc.functional_call(module, pd, (xin,))
+
+            with torch.no_grad():
+                return tfunc.vmap(fwd)(stacked)
         outs = []
         with torch.no_grad():
-            xin = X.view(X.shape[0], *spec.input_shape)
             for idx in nodes.tolist():
                 spec.load_row(module, state.params[idx])
                 outs.append(module(xin))
@@ -687,6 +754,11 @@ class TorchBackend:
                 reply_slots,
             )
             return
+        if spec.family == "torchmod" and self._deliver_torchmod(
+            state, pool, data, spec, recv_nodes, recv_ptr, del_slots,
+            reply_slots, del_pids,
+        ):
+            return
         base_mode = spec.mode
         pt = getattr(spec, "pass_through", False)
         ptr = recv_ptr.tolist()
@@ -722,6 +794,64 @@ class TorchBackend:
                         torch.tensor([node_t], dtype=torch.long),
                         torch.tensor([r], dtype=torch.long),
                     )
+
+    def _deliver_torchmod(
+        self, state, pool, data, spec, recv_nodes, recv_ptr, del_slots,
+        reply_slots, del_pids,
+    ) -> bool:
+        """Node-batched delivery for the torchmod family, wave by wave:
+        wave ``w`` applies every receiver's ``w``-th delivery at once
+        (vectorized merge/adopt + one batched vmap SGD trajectory), which
+        preserves each receiver's in-order semantics while replacing the
+        per-message python loop. Within one deliver launch no delivery
+        reads a slot written by a same-launch reply (packer invariant —
+        the same one the HIP kernels rely on), so cross-receiver
+        reordering is safe. Returns False to fall back to the loop."""
+        mode = spec.mode
+        if mode not in (
+            CreateModelMode.MERGE_UPDATE,
+            CreateModelMode.UPDATE,
+            CreateModelMode.PASS,
+        ):
+            return False
+        if getattr(spec, "pass_through", False):
+            return False
+        if os.environ.get("GOSSIPY_TORCHMOD_LOOP") == "1":
+            return False
+        dev = state.params.device
+        ptr = recv_ptr.long()
+        cnt = ptr[1:] - ptr[:-1]
+        max_w = int(cnt.max()) if len(cnt) else 0
+        nodes_all = recv_nodes.long()
+        dslots = del_slots.long()
+        rslots = reply_slots.long()
+        for w in range(max_w):
+            rows = torch.nonzero(cnt > w).flatten()
+            j = ptr[:-1][rows] + w
+            nodes = nodes_all[rows].to(dev)
+            slots = dslots[j].to(dev)
+            if mode == CreateModelMode.MERGE_UPDATE:
+                state.params[nodes] = (
+                    state.params[nodes] + pool.slots[slots]
+                ) * 0.5
+                state.ages[nodes] = torch.maximum(
+                    state.ages[nodes], pool.slot_ages[slots]
+                )
+                self.update(state, data, spec, nodes)
+            else:
+                # UPDATE trains (and PASS just keeps) the RECEIVED model
+                # (gossipy/model/handler.py:122-135)
+                state.params[nodes] = pool.slots[slots]
+                state.ages[nodes] = pool.slot_ages[slots]
+                if mode == CreateModelMode.UPDATE:
+                    self.update(state, data, spec, nodes)
+            r = rslots[j]
+            has_r = r >= 0
+            if bool(has_r.any()):
+                self.snapshot(
+                    state, pool, nodes[has_r.to(dev)], r[has_r].to(dev)
+                )
+        return True
 
     def _merge_mean(self, state, pool, node: int, slot: int) -> None:
         state.params[node] = (state.params[node] + pool.slots[slot]) * 0.5

@@ -345,9 +345,20 @@ class TorchModuleSpec:
         self.batch_size = batch_size
         self.mode = mode
         proto = module_factory()
+        self._names = [n for n, _ in proto.named_parameters()]
         self._shapes = [tuple(p.shape) for p in proto.parameters()]
         self._numels = [int(np.prod(s)) for s in self._shapes]
         self._D = int(sum(self._numels))
+
+    def param_layout(self):
+        """``(name, shape, offset, numel)`` per parameter tensor, in
+        ``parameters()`` order — the arena row layout."""
+        out = []
+        off = 0
+        for name, shape, n in zip(self._names, self._shapes, self._numels):
+            out.append((name, shape, off, n))
+            off += n
+        return out
 
     @property
     def D(self) -> int:

@@ -9,7 +9,12 @@ import numpy as np
 import pytest
 import torch
 
-from gossipy_amd.core import AntiEntropyProtocol, ConstantDelay, UniformDelay
+from gossipy_amd.core import (
+    AntiEntropyProtocol,
+    ConstantDelay,
+    CreateModelMode,
+    UniformDelay,
+)
 from gossipy_amd.data import make_synthetic_classification
 from gossipy_amd.engine import (
     BatchedGossipSimulator,
@@ -1641,6 +1646,99 @@ class TestTorchModuleEngine:
 
         s1, s2 = run(), run()
         assert torch.allclose(s1.local_params(), s2.local_params(), atol=1e-6)
+
+    def test_batched_update_matches_loop_tight(self, monkeypatch):
+        """One batched vmap update == the per-node loop to float noise
+        (~1e-8); the end-to-end comparisons below allow more because SGD
+        on this task amplifies any conv-algorithm noise ~50x per round."""
+        from gossipy_amd.engine import TorchModuleSpec
+        from gossipy_amd.engine.arena import NodeStateArena
+        from gossipy_amd.engine.backend import TorchBackend
+
+        spec = TorchModuleSpec(
+            _cifar10net, input_shape=(3, 32, 32), lr=0.1, batch_size=8
+        )
+        data = _cnn_data()
+        be = TorchBackend()
+        torch.manual_seed(0)
+        base = torch.randn(8, spec.D) * 0.05
+
+        def run(loop):
+            st = NodeStateArena(8, spec.D, torch.device("cpu"))
+            st.params.copy_(base)
+            if loop:
+                monkeypatch.setenv("GOSSIPY_TORCHMOD_LOOP", "1")
+            else:
+                monkeypatch.delenv("GOSSIPY_TORCHMOD_LOOP", raising=False)
+            be._update_torchmod(
+                st.params, st.ages, data, spec, torch.arange(8)
+            )
+            return st.params.clone(), st.ages.clone()
+
+        pa, aa = run(False)
+        pb, ab = run(True)
+        assert torch.equal(aa, ab)
+        assert float((pa - pb).abs().max()) < 1e-6
+
+    @pytest.mark.parametrize(
+        "protocol,mode,ragged,rounds,atol",
+        [
+            (AntiEntropyProtocol.PUSH, CreateModelMode.MERGE_UPDATE, False, 3, 1e-3),
+            (AntiEntropyProtocol.PUSH_PULL, CreateModelMode.MERGE_UPDATE, False, 1, 1e-3),
+            (AntiEntropyProtocol.PUSH, CreateModelMode.UPDATE, False, 3, 1e-3),
+            (AntiEntropyProtocol.PUSH, CreateModelMode.MERGE_UPDATE, True, 1, 1e-3),
+        ],
+    )
+    def test_batched_matches_loop(
+        self, monkeypatch, protocol, mode, ragged, rounds, atol
+    ):
+        """The vmap-batched torchmod update/deliver path is the per-node
+        loop's oracle twin. Integer ages must match
+        EXACTLY (they count every optimizer step in order); params to a
+        tolerance that covers conv-algorithm float noise amplified by
+        training."""
+        from gossipy_amd.engine import TorchModuleSpec
+
+        def build():
+            spec = TorchModuleSpec(
+                _cifar10net, input_shape=(3, 32, 32), lr=0.1,
+                batch_size=8, mode=mode,
+            )
+            if ragged:
+                rng = np.random.default_rng(5)
+                labels = rng.integers(0, 10, size=8 * 12)
+                x = rng.normal(0, 0.3, size=(len(labels), 3, 32, 32)).astype(
+                    np.float32
+                )
+                X = torch.from_numpy(x.reshape(len(labels), -1))
+                y = torch.from_numpy(labels).float()
+                sizes = [12, 7, 12, 5, 12, 7, 12, 12]
+                off, shards = 0, []
+                for s in sizes:
+                    shards.append((X[off : off + s], y[off : off + s]))
+                    off += s
+                data = DataArena.from_shards(
+                    shards, torch.device("cpu"), global_eval=(X, y)
+                )
+            else:
+                data = _cnn_data()
+            cfg = EngineConfig(
+                n_nodes=8, delta=5, protocol=protocol,
+                model_size=spec.D, sampling_eval=0.0, seed=41,
+            )
+            sim = BatchedGossipSimulator(cfg, spec, data)
+            sim.init_nodes()
+            sim.start(n_rounds=rounds)
+            return sim
+
+        monkeypatch.delenv("GOSSIPY_TORCHMOD_LOOP", raising=False)
+        batched = build()
+        monkeypatch.setenv("GOSSIPY_TORCHMOD_LOOP", "1")
+        loop = build()
+        assert torch.equal(batched.state.ages, loop.state.ages)
+        assert torch.allclose(
+            batched.local_params(), loop.local_params(), atol=atol
+        )
 
 
 def _cnn_worker(rank, world, port, q):
