@@ -499,6 +499,7 @@ class TorchBackend:
 
         module = spec.template().to(params.device)
         layout = spec.param_layout()
+        nodes = nodes.to(params.device)
         counts_all = data.counts[nodes].long()
         lr, wd = spec.lr, spec.weight_decay
 
@@ -542,7 +543,7 @@ class TorchBackend:
         if len(nodes) > 1 and os.environ.get("GOSSIPY_TORCHMOD_LOOP") != "1":
             import torch.func as tfunc
 
-            rows = state.params[nodes.long()]
+            rows = state.params[nodes.long().to(state.params.device)]
             R = rows.shape[0]
             stacked = {
                 name: rows[:, o : o + n].view(R, *shape)
