@@ -508,32 +508,40 @@ class TorchBackend:
             return torch.nn.functional.cross_entropy(out, yb)
 
         gfn = tfunc.vmap(tfunc.grad(loss_fn))
+        # bound grouped-conv activation memory (fwd+bwd graph) per launch
+        target = int(os.environ.get(
+            "GOSSIPY_TORCHMOD_CHUNK",
+            65536 if params.device.type == "cuda" else 4096,
+        ))
         for c in torch.unique(counts_all).tolist():
             c = int(c)
             if c == 0:
                 continue
-            sel = nodes[counts_all == c]
-            B = len(sel)
-            rows = params[sel]  # fancy index -> private copy
-            stacked = {
-                name: rows[:, o : o + n].view(B, *shape)
-                for name, shape, o, n in layout
-            }
-            x = data.x[sel, :c].view(B, c, *spec.input_shape)
-            y = data.y[sel, :c].long()
+            sel_all = nodes[counts_all == c]
             bs = c if spec.batch_size == 0 else spec.batch_size
-            for _ in range(max(1, spec.local_epochs)):
-                for s in range(0, c, bs):
-                    grads = gfn(stacked, x[:, s : s + bs], y[:, s : s + bs])
-                    with torch.no_grad():
-                        for name, _, _, _ in layout:
-                            g = grads[name]
-                            p = stacked[name]
-                            if wd:
-                                g = g.add(p, alpha=wd)
-                            p.add_(g, alpha=-lr)
-                    ages[sel] += 1
-            params[sel] = rows
+            node_chunk = max(1, target // max(1, min(bs, c)))
+            for s0 in range(0, len(sel_all), node_chunk):
+                sel = sel_all[s0 : s0 + node_chunk]
+                B = len(sel)
+                rows = params[sel]  # fancy index -> private copy
+                stacked = {
+                    name: rows[:, o : o + n].view(B, *shape)
+                    for name, shape, o, n in layout
+                }
+                x = data.x[sel, :c].view(B, c, *spec.input_shape)
+                y = data.y[sel, :c].long()
+                for _ in range(max(1, spec.local_epochs)):
+                    for s in range(0, c, bs):
+                        grads = gfn(stacked, x[:, s : s + bs], y[:, s : s + bs])
+                        with torch.no_grad():
+                            for name, _, _, _ in layout:
+                                g = grads[name]
+                                p = stacked[name]
+                                if wd:
+                                    g = g.add(p, alpha=wd)
+                                p.add_(g, alpha=-lr)
+                        ages[sel] += 1
+                params[sel] = rows
 
     def torchmod_scores(self, state, spec, nodes, X) -> torch.Tensor:
         """Class scores ``[R, n, k]`` for the eval sweep (node-batched
@@ -545,16 +553,28 @@ class TorchBackend:
 
             rows = state.params[nodes.long().to(state.params.device)]
             R = rows.shape[0]
-            stacked = {
-                name: rows[:, o : o + n].view(R, *shape)
-                for name, shape, o, n in spec.param_layout()
-            }
+            layout = spec.param_layout()
 
             def fwd(pd):
                 return tfunc.functional_call(module, pd, (xin,))
 
+            # bound grouped-conv activation memory (forward only, so 2x
+            # the train target) per launch
+            target = 2 * int(os.environ.get(
+                "GOSSIPY_TORCHMOD_CHUNK",
+                65536 if state.params.device.type == "cuda" else 4096,
+            ))
+            chunk = max(1, target // max(1, xin.shape[0]))
+            outs = []
             with torch.no_grad():
-                return tfunc.vmap(fwd)(stacked)
+                for s in range(0, R, chunk):
+                    part = rows[s : s + chunk]
+                    stacked = {
+                        name: part[:, o : o + n].view(len(part), *shape)
+                        for name, shape, o, n in layout
+                    }
+                    outs.append(tfunc.vmap(fwd)(stacked))
+            return torch.cat(outs) if len(outs) > 1 else outs[0]
         outs = []
         with torch.no_grad():
             for idx in nodes.tolist():

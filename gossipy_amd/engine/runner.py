@@ -481,6 +481,79 @@ class BatchedGossipSimulator(SimulationEventSender):
                 self._exchange_finish(pending)
                 _launch(drep["groups"]["remote"], t)
 
+    def _run_round_groups(self, f: dict) -> None:
+        """Single-rank whole-round replay of merged/packed launch groups
+        through the PYTHON backend (families without a C++ executor —
+        torchmod). One backend call-set per group gives the node-batched
+        vmap paths their batch: a 100-tick round of 1-receiver ticks
+        becomes a handful of ~20-receiver groups."""
+        snap_nodes = np.asarray(f["snap_nodes"])
+        snap_slots = np.asarray(f["snap_slots"])
+        st = np.asarray(f["snap_tptr"])
+        recv_nodes = np.asarray(f["recv_nodes"])
+        nptr = np.asarray(f["recv_nptr"])
+        rt = np.asarray(f["recv_tptr"])
+        del_slots = np.asarray(f["del_slots"])
+        reply_slots = np.asarray(f["reply_slots"])
+        del_pids = np.asarray(f.get("del_pids", ()))
+        pull_nodes = np.asarray(f["pull_nodes"])
+        pull_slots = np.asarray(f["pull_slots"])
+        pt = np.asarray(f["pull_tptr"])
+        rep_nodes = np.asarray(f["rep_nodes"])
+        rep_nptr = np.asarray(f["rep_nptr"])
+        qt = np.asarray(f["rep_tptr"])
+        rep_slots = np.asarray(f["rep_slots"])
+        rep_pids = np.asarray(f.get("rep_pids", ()))
+        rep_reply = np.asarray(f.get("rep_reply_slots", ()))
+
+        def _t64(a):
+            return torch.from_numpy(np.ascontiguousarray(a, dtype=np.int64))
+
+        for g in range(len(st) - 1):
+            a, b = int(st[g]), int(st[g + 1])
+            if b > a:
+                self.backend.snapshot(
+                    self.state, self.pool,
+                    _t64(snap_nodes[a:b]),
+                    _t64(snap_slots[a:b]).to(self.device),
+                    src_off=self._snap_off,
+                )
+            r0, r1 = int(rt[g]), int(rt[g + 1])
+            if r1 > r0:
+                d0, d1 = int(nptr[r0]), int(nptr[r1])
+                self.backend.deliver(
+                    self.state, self.pool, self.data, self.spec,
+                    _t64(recv_nodes[r0:r1]),
+                    _t64(nptr[r0 : r1 + 1] - d0),
+                    _t64(del_slots[d0:d1]),
+                    _t64(reply_slots[d0:d1]),
+                    del_pids=_t64(del_pids[d0:d1]) if len(del_pids) else None,
+                )
+            a, b = int(pt[g]), int(pt[g + 1])
+            if b > a:
+                self.backend.snapshot(
+                    self.state, self.pool,
+                    _t64(pull_nodes[a:b]),
+                    _t64(pull_slots[a:b]).to(self.device),
+                    src_off=self._snap_off,
+                )
+            q0, q1 = int(qt[g]), int(qt[g + 1])
+            if q1 > q0:
+                e0, e1 = int(rep_nptr[q0]), int(rep_nptr[q1])
+                rr = (
+                    _t64(rep_reply[e0:e1])
+                    if len(rep_reply)
+                    else torch.full((e1 - e0,), -1, dtype=torch.int64)
+                )
+                self.backend.deliver(
+                    self.state, self.pool, self.data, self.spec,
+                    _t64(rep_nodes[q0:q1]),
+                    _t64(rep_nptr[q0 : q1 + 1] - e0),
+                    _t64(rep_slots[e0:e1]),
+                    rr,
+                    del_pids=_t64(rep_pids[e0:e1]) if len(rep_pids) else None,
+                )
+
     def _run_tick(self, phase: TickPhase) -> None:
         # A: snapshots of firing nodes
         mine = self._is_mine(phase.snap_nodes)
@@ -799,7 +872,9 @@ class BatchedGossipSimulator(SimulationEventSender):
                 )
         return out
 
-    def _maybe_merge(self, f: dict, pack: bool = True) -> dict:
+    def _maybe_merge(
+        self, f: dict, pack: bool = True, force_py_pack: bool = False
+    ) -> dict:
         """Apply launch-group packing/merging unless disabled
         (``GOSSIPY_NO_MERGE=1`` turns everything off, ``GOSSIPY_NO_PACK=1``
         falls back to tick-level merging). Packed schedules carry the
@@ -819,8 +894,12 @@ class BatchedGossipSimulator(SimulationEventSender):
             # lean flat: only the packed schedule was materialized — any
             # knob that reached here is overridden (loud beats wrong)
             packed = f["packed"]
-        if packed is None and pack and os.environ.get("GOSSIPY_PACK") == "1":
-            # python reference packer (A/B + flats without the C++ packer)
+        if packed is None and pack and (
+            force_py_pack or os.environ.get("GOSSIPY_PACK") == "1"
+        ):
+            # python reference packer (A/B + flats without the C++ packer;
+            # forced for the torchmod group-replay path, whose vmap
+            # batches live on group width)
             packed = self._pack_flat(f)
         if packed is not None:
             t = getattr(self, "merge_stats", (0, 0))
@@ -1411,6 +1490,22 @@ class BatchedGossipSimulator(SimulationEventSender):
                     # (tokenized / cache-neigh / python-scheduler fallback)
                     self._run_round_fast(
                         self._maybe_merge(self._flatten_phases(sched.ticks))
+                    )
+                elif (
+                    self.world == 1
+                    and self._flat_schedulable
+                    and self.spec.family == "torchmod"
+                    and sched.ticks
+                ):
+                    # torchmod has no C++ executor, but its batched python
+                    # backend wants LARGE receiver groups (vmap SGD over
+                    # all of a group's receivers at once) — replay packed
+                    # launch groups instead of 1-2-receiver ticks
+                    self._run_round_groups(
+                        self._maybe_merge(
+                            self._flatten_phases(sched.ticks),
+                            force_py_pack=True,
+                        )
                     )
                 else:
                     for phase in sched.ticks:
