@@ -35,6 +35,12 @@ from __future__ import annotations
 import os
 from typing import Optional
 
+# MIOpen: heuristic (immediate-mode) conv solver selection instead of a
+# per-new-shape exhaustive search — gossip rounds present many distinct
+# grouped-conv shapes, so find-mode search time is pure overhead here.
+# Must be set before the first conv initializes MIOpen.
+os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
+
 import numpy as np
 import torch
 
@@ -491,10 +497,24 @@ class TorchBackend:
                     ages[idx] += 1
             spec.store_row(module, params[idx])
 
+    @staticmethod
+    def _pow2_bucket(b: int) -> int:
+        n = 1
+        while n < b:
+            n <<= 1
+        return n
+
     def _update_torchmod_batched(self, params, ages, data, spec, nodes) -> None:
         """One SGD trajectory per node, all nodes at once (the VERDICT r1
         CNN item: the per-node loop re-built an optimizer per node per
-        tick and ran ~B times more tiny conv launches)."""
+        tick and ran ~B times more tiny conv launches).
+
+        The vmap batch B is PADDED up to a power of two: a grouped conv
+        with ``groups=B`` is a distinct MIOpen problem per B, and gossip
+        group sizes vary every round — unpadded, every new B paid a
+        ~25 ms kernel-search on its first conv (measured: 438 conv2d
+        calls x 24 ms avg in a 5-round profile). Buckets make the shape
+        set small and stable, so the search happens once per bucket."""
         import torch.func as tfunc
 
         module = spec.template().to(params.device)
@@ -519,17 +539,25 @@ class TorchBackend:
                 continue
             sel_all = nodes[counts_all == c]
             bs = c if spec.batch_size == 0 else spec.batch_size
-            node_chunk = max(1, target // max(1, min(bs, c)))
+            node_chunk = self._pow2_bucket(
+                max(1, target // max(1, min(bs, c)))
+            )
             for s0 in range(0, len(sel_all), node_chunk):
                 sel = sel_all[s0 : s0 + node_chunk]
                 B = len(sel)
-                rows = params[sel]  # fancy index -> private copy
+                Bp = self._pow2_bucket(B)
+                rows = params.new_zeros(Bp, params.shape[1])
+                rows[:B] = params[sel]
                 stacked = {
-                    name: rows[:, o : o + n].view(B, *shape)
+                    name: rows[:, o : o + n].view(Bp, *shape)
                     for name, shape, o, n in layout
                 }
-                x = data.x[sel, :c].view(B, c, *spec.input_shape)
-                y = data.y[sel, :c].long()
+                x = data.x.new_zeros(Bp, c, *spec.input_shape)
+                x[:B] = data.x[sel, :c].view(B, c, *spec.input_shape)
+                y = torch.zeros(
+                    Bp, c, dtype=torch.long, device=params.device
+                )
+                y[:B] = data.y[sel, :c].long()
                 for _ in range(max(1, spec.local_epochs)):
                     for s in range(0, c, bs):
                         grads = gfn(stacked, x[:, s : s + bs], y[:, s : s + bs])
@@ -541,7 +569,7 @@ class TorchBackend:
                                     g = g.add(p, alpha=wd)
                                 p.add_(g, alpha=-lr)
                         ages[sel] += 1
-                params[sel] = rows
+                params[sel] = rows[:B]
 
     def torchmod_scores(self, state, spec, nodes, X) -> torch.Tensor:
         """Class scores ``[R, n, k]`` for the eval sweep (node-batched
@@ -559,21 +587,27 @@ class TorchBackend:
                 return tfunc.functional_call(module, pd, (xin,))
 
             # bound grouped-conv activation memory (forward only, so 2x
-            # the train target) per launch
+            # the train target) per launch; chunk and pad to power-of-two
+            # buckets so MIOpen sees a stable grouped-conv shape set
             target = 2 * int(os.environ.get(
                 "GOSSIPY_TORCHMOD_CHUNK",
                 65536 if state.params.device.type == "cuda" else 4096,
             ))
-            chunk = max(1, target // max(1, xin.shape[0]))
+            chunk = self._pow2_bucket(
+                max(1, target // max(1, xin.shape[0]))
+            )
             outs = []
             with torch.no_grad():
                 for s in range(0, R, chunk):
                     part = rows[s : s + chunk]
+                    Bp = self._pow2_bucket(len(part))
+                    padded = rows.new_zeros(Bp, rows.shape[1])
+                    padded[: len(part)] = part
                     stacked = {
-                        name: part[:, o : o + n].view(len(part), *shape)
+                        name: padded[:, o : o + n].view(Bp, *shape)
                         for name, shape, o, n in layout
                     }
-                    outs.append(tfunc.vmap(fwd)(stacked))
+                    outs.append(tfunc.vmap(fwd)(stacked)[: len(part)])
             return torch.cat(outs) if len(outs) > 1 else outs[0]
         outs = []
         with torch.no_grad():
