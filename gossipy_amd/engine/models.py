@@ -302,6 +302,51 @@ class KMeansSpec:
         return self.k * self.dim
 
 
+def _bmm_convs(module) -> None:
+    """Rewrite every plain (groups=1, 2-d) ``nn.Conv2d`` in ``module`` to
+    compute via im2col (``unfold``) + ``matmul`` instead of MIOpen.
+
+    The node-batched engine paths run the module under ``torch.func.vmap``
+    with per-node weights, which MIOpen sees as a grouped conv with
+    ``groups = vmap-batch`` — a distinct solver problem per batch size,
+    and on gfx950 the grouped solvers fall back to slow kernels (measured:
+    the Onoszko eval forward alone cost ~96 ms/round). unfold+matmul maps
+    the same math onto rocBLAS strided-batched GEMMs: no solver search,
+    no grouped-conv penalty, fully vmappable, autograd = more GEMMs.
+    Parameters and state-dict keys are untouched."""
+    import types
+
+    import torch
+    import torch.nn.functional as F
+
+    def bmm_forward(self, x):
+        kh, kw = self.kernel_size
+        cols = F.unfold(
+            x, self.kernel_size, dilation=self.dilation,
+            padding=self.padding, stride=self.stride,
+        )  # [N, C*kh*kw, L]
+        out = self.weight.reshape(self.out_channels, -1) @ cols
+        if self.bias is not None:
+            out = out + self.bias[:, None]
+        ph, pw = self.padding if isinstance(self.padding, tuple) else (
+            self.padding, self.padding
+        )
+        dh, dw = self.dilation
+        sh, sw = self.stride
+        ho = (x.shape[-2] + 2 * ph - dh * (kh - 1) - 1) // sh + 1
+        wo = (x.shape[-1] + 2 * pw - dw * (kw - 1) - 1) // sw + 1
+        return out.reshape(*out.shape[:-2], self.out_channels, ho, wo)
+
+    for sub in module.modules():
+        if (
+            type(sub) is torch.nn.Conv2d
+            and sub.groups == 1
+            and sub.padding_mode == "zeros"
+            and not isinstance(sub.padding, str)
+        ):
+            sub.forward = types.MethodType(bmm_forward, sub)
+
+
 class TorchModuleSpec:
     """Engine family for arbitrary ``nn.Module`` architectures (the CNN
     path of SURVEY.md §7 step 7 — Onoszko 2021's CIFAR10Net,
@@ -369,9 +414,15 @@ class TorchModuleSpec:
         return int(np.prod(self.input_shape))
 
     def template(self):
-        """A module instance for row_to/row_from round-trips."""
+        """A module instance for row_to/row_from round-trips and for the
+        batched vmap paths. Plain ``nn.Conv2d`` forwards are rewritten to
+        unfold+matmul (see :func:`_bmm_convs`) unless ``GOSSIPY_MIOPEN_CONV=1``."""
         if not hasattr(self, "_template"):
             self._template = self.module_factory()
+            import os
+
+            if os.environ.get("GOSSIPY_MIOPEN_CONV") != "1":
+                _bmm_convs(self._template)
         return self._template
 
     def load_row(self, module, row) -> None:
