@@ -421,7 +421,10 @@ class TorchModuleSpec:
             self._template = self.module_factory()
             import os
 
-            if os.environ.get("GOSSIPY_MIOPEN_CONV") != "1":
+            if os.environ.get("GOSSIPY_BMM_CONV") == "1":
+                # opt-in A/B: measured SLOWER than MIOpen grouped conv
+                # under vmap on gfx950 (unfold's vmap rule materializes
+                # im2col per example) — kept for comparison runs
                 _bmm_convs(self._template)
         return self._template
 
