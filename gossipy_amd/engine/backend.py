@@ -542,10 +542,27 @@ class TorchBackend:
             node_chunk = self._pow2_bucket(
                 max(1, target // max(1, min(bs, c)))
             )
+            epochs = max(1, spec.local_epochs)
+            n_steps = epochs * len(range(0, c, bs))
             for s0 in range(0, len(sel_all), node_chunk):
                 sel = sel_all[s0 : s0 + node_chunk]
                 B = len(sel)
                 Bp = self._pow2_bucket(B)
+                entry = self._tm_graph_entry(
+                    spec, layout, gfn, Bp, c, bs, epochs, params, data
+                )
+                if entry is not None:
+                    graph, srows, sx, sy = entry
+                    srows.zero_()
+                    srows[:B] = params[sel]
+                    sx.zero_()
+                    sx[:B] = data.x[sel, :c].view(B, c, *spec.input_shape)
+                    sy.zero_()
+                    sy[:B] = data.y[sel, :c].long()
+                    graph.replay()
+                    params[sel] = srows[:B]
+                    ages[sel] += n_steps
+                    continue
                 rows = params.new_zeros(Bp, params.shape[1])
                 rows[:B] = params[sel]
                 stacked = {
@@ -558,18 +575,71 @@ class TorchBackend:
                     Bp, c, dtype=torch.long, device=params.device
                 )
                 y[:B] = data.y[sel, :c].long()
-                for _ in range(max(1, spec.local_epochs)):
-                    for s in range(0, c, bs):
-                        grads = gfn(stacked, x[:, s : s + bs], y[:, s : s + bs])
-                        with torch.no_grad():
-                            for name, _, _, _ in layout:
-                                g = grads[name]
-                                p = stacked[name]
-                                if wd:
-                                    g = g.add(p, alpha=wd)
-                                p.add_(g, alpha=-lr)
-                        ages[sel] += 1
+                self._tm_sgd_body(
+                    gfn, layout, stacked, x, y, c, bs, epochs, lr, wd
+                )
+                ages[sel] += n_steps
                 params[sel] = rows[:B]
+
+    @staticmethod
+    def _tm_sgd_body(gfn, layout, stacked, x, y, c, bs, epochs, lr, wd):
+        """The (capturable) local-SGD trajectory: every minibatch step of
+        every epoch, vmap-gradded and applied in place on ``stacked``."""
+        for _ in range(epochs):
+            for s in range(0, c, bs):
+                grads = gfn(stacked, x[:, s : s + bs], y[:, s : s + bs])
+                with torch.no_grad():
+                    for name, _, _, _ in layout:
+                        g = grads[name]
+                        p = stacked[name]
+                        if wd:
+                            g = g.add(p, alpha=wd)
+                        p.add_(g, alpha=-lr)
+
+    def _tm_graph_entry(
+        self, spec, layout, gfn, Bp, c, bs, epochs, params, data
+    ):
+        """hipGraph cache for the batched SGD trajectory, keyed by shape
+        bucket: ~60 host-dispatched aten calls per minibatch step collapse
+        into one graph replay. Falls back to eager (returns None) on CPU,
+        when capture fails, or under GOSSIPY_NO_GRAPH=1."""
+        if (
+            params.device.type != "cuda"
+            or os.environ.get("GOSSIPY_NO_GRAPH") == "1"
+        ):
+            return None
+        cache = getattr(self, "_tm_graphs", None)
+        if cache is None:
+            cache = self._tm_graphs = {}
+        key = (id(spec), Bp, c, bs, epochs)
+        if key in cache:
+            return cache[key]
+        lr, wd = spec.lr, spec.weight_decay
+        try:
+            srows = params.new_zeros(Bp, params.shape[1])
+            sx = data.x.new_zeros(Bp, c, *spec.input_shape)
+            sy = torch.zeros(Bp, c, dtype=torch.long, device=params.device)
+            stacked = {
+                name: srows[:, o : o + n].view(Bp, *shape)
+                for name, shape, o, n in layout
+            }
+            stream = torch.cuda.Stream()
+            stream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(stream):
+                for _ in range(2):  # warmup resolves MIOpen finds + allocs
+                    self._tm_sgd_body(
+                        gfn, layout, stacked, sx, sy, c, bs, epochs, lr, wd
+                    )
+            torch.cuda.current_stream().wait_stream(stream)
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                self._tm_sgd_body(
+                    gfn, layout, stacked, sx, sy, c, bs, epochs, lr, wd
+                )
+            cache[key] = (graph, srows, sx, sy)
+        except Exception:
+            cache[key] = None  # capture unsupported here — stay eager
+        return cache[key]
 
     def torchmod_scores(self, state, spec, nodes, X) -> torch.Tensor:
         """Class scores ``[R, n, k]`` for the eval sweep.
