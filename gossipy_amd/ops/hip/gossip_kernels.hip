@@ -1073,8 +1073,30 @@ DEV_INLINE void mlp_update(const MlpArgs& a, int node, float* W,
                 int m16 = (m + 15) >> 4, o16 = (fout + 15) >> 4;
                 for (int tile = wave; tile < m16 * o16; tile += nwaves) {
                     int tr = (tile / o16) << 4, tc = (tile % o16) << 4;
+                    // two independent accumulator chains: back-to-back
+                    // dependent MFMAs stall on the 16x16x4 pipeline; the
+                    // f32 MFMA is exact, so only the summation order
+                    // changes (tests compare vs torch at tolerance)
                     f32x4_t acc = {0.f, 0.f, 0.f, 0.f};
-                    for (int k0 = 0; k0 < fin; k0 += 4) {
+                    f32x4_t acc2 = {0.f, 0.f, 0.f, 0.f};
+                    int k0 = 0;
+                    for (; k0 + 8 <= fin; k0 += 8) {
+                        int q = k0 + fk, q2 = q + 4;
+                        float av = (tr + fr < m) ? cur[(tr + fr) * fin + q]
+                                                 : 0.f;
+                        float bv = (tc + fr < fout)
+                                       ? W[w_off + (tc + fr) * fin + q] : 0.f;
+                        float av2 = (tr + fr < m && q2 < fin)
+                                        ? cur[(tr + fr) * fin + q2] : 0.f;
+                        float bv2 = (tc + fr < fout && q2 < fin)
+                                        ? W[w_off + (tc + fr) * fin + q2]
+                                        : 0.f;
+                        acc = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                            av, bv, acc, 0, 0, 0);
+                        acc2 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                            av2, bv2, acc2, 0, 0, 0);
+                    }
+                    for (; k0 < fin; k0 += 4) {
                         int q = k0 + fk;
                         float av = (tr + fr < m && q < fin)
                                        ? cur[(tr + fr) * fin + q] : 0.f;
@@ -1083,6 +1105,7 @@ DEV_INLINE void mlp_update(const MlpArgs& a, int node, float* W,
                         acc = __builtin_amdgcn_mfma_f32_16x16x4f32(
                             av, bv, acc, 0, 0, 0);
                     }
+                    for (int r = 0; r < 4; ++r) acc[r] += acc2[r];
                     for (int r = 0; r < 4; ++r) {
                         int row = tr + (fk << 2) + r, col = tc + fr;
                         if (row < m && col < fout) {
@@ -1131,7 +1154,27 @@ DEV_INLINE void mlp_update(const MlpArgs& a, int node, float* W,
                     for (int tile = wave; tile < m16 * q16; tile += nwaves) {
                         int tr = (tile / q16) << 4, tc = (tile % q16) << 4;
                         f32x4_t acc = {0.f, 0.f, 0.f, 0.f};
-                        for (int k0 = 0; k0 < fout; k0 += 4) {
+                        f32x4_t acc2 = {0.f, 0.f, 0.f, 0.f};
+                        int k0 = 0;
+                        for (; k0 + 8 <= fout; k0 += 8) {
+                            int o = k0 + fk, o2 = o + 4;
+                            float av = (tr + fr < m)
+                                           ? gcur[(tr + fr) * fout + o] : 0.f;
+                            float bv = (tc + fr < fin)
+                                           ? W[w_off + o * fin + tc + fr]
+                                           : 0.f;
+                            float av2 = (tr + fr < m)
+                                            ? gcur[(tr + fr) * fout + o2]
+                                            : 0.f;
+                            float bv2 = (tc + fr < fin)
+                                            ? W[w_off + o2 * fin + tc + fr]
+                                            : 0.f;
+                            acc = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                                av, bv, acc, 0, 0, 0);
+                            acc2 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                                av2, bv2, acc2, 0, 0, 0);
+                        }
+                        for (; k0 < fout; k0 += 4) {
                             int o = k0 + fk;
                             float av = (tr + fr < m && o < fout)
                                            ? gcur[(tr + fr) * fout + o] : 0.f;
@@ -1140,6 +1183,7 @@ DEV_INLINE void mlp_update(const MlpArgs& a, int node, float* W,
                             acc = __builtin_amdgcn_mfma_f32_16x16x4f32(
                                 av, bv, acc, 0, 0, 0);
                         }
+                        for (int r = 0; r < 4; ++r) acc[r] += acc2[r];
                         for (int r = 0; r < 4; ++r) {
                             int row = tr + (fk << 2) + r, col = tc + fr;
                             if (row < m && col < fin) {
@@ -1158,7 +1202,27 @@ DEV_INLINE void mlp_update(const MlpArgs& a, int node, float* W,
                     for (int tile = wave; tile < o16 * q16; tile += nwaves) {
                         int tr = (tile / q16) << 4, tc = (tile % q16) << 4;
                         f32x4_t acc = {0.f, 0.f, 0.f, 0.f};
-                        for (int k0 = 0; k0 < m; k0 += 4) {
+                        f32x4_t acc2 = {0.f, 0.f, 0.f, 0.f};
+                        int k0 = 0;
+                        for (; k0 + 8 <= m; k0 += 8) {
+                            int sidx = k0 + fk, sidx2 = sidx + 4;
+                            float av = (tr + fr < fout)
+                                           ? gcur[sidx * fout + tr + fr]
+                                           : 0.f;
+                            float bv = (tc + fr < fin)
+                                           ? ain[sidx * fin + tc + fr] : 0.f;
+                            float av2 = (tr + fr < fout)
+                                            ? gcur[sidx2 * fout + tr + fr]
+                                            : 0.f;
+                            float bv2 = (tc + fr < fin)
+                                            ? ain[sidx2 * fin + tc + fr]
+                                            : 0.f;
+                            acc = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                                av, bv, acc, 0, 0, 0);
+                            acc2 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                                av2, bv2, acc2, 0, 0, 0);
+                        }
+                        for (; k0 < m; k0 += 4) {
                             int sidx = k0 + fk;
                             float av = (tr + fr < fout && sidx < m)
                                            ? gcur[sidx * fout + tr + fr] : 0.f;
@@ -1167,6 +1231,7 @@ DEV_INLINE void mlp_update(const MlpArgs& a, int node, float* W,
                             acc = __builtin_amdgcn_mfma_f32_16x16x4f32(
                                 av, bv, acc, 0, 0, 0);
                         }
+                        for (int r = 0; r < 4; ++r) acc[r] += acc2[r];
                         for (int r = 0; r < 4; ++r) {
                             int row = tr + (fk << 2) + r, col = tc + fr;
                             if (row < fout && col < fin) {
@@ -2492,6 +2557,21 @@ void run_round_mlp(
         (2 * (size_t)a.D + (size_t)bsmax * (a.d_in + act_sum) +
          2 * (size_t)bsmax * act_max);
     TORCH_CHECK(smem <= 160 * 1024, "mlp LDS budget exceeded");
+    // 8 waves when any GEMM stage has >= 8 MFMA tiles (same policy as
+    // tick_mlp — the packed executor previously hardcoded 4 waves, which
+    // left each SIMD with a single wave and every LDS/MFMA stall exposed)
+    int max_tiles = 0;
+    {
+        int m16 = (bsmax + 15) >> 4;
+        for (int l = 0; l < n_layers; ++l) {
+            int fin16 = (L[4 * l + 2] + 15) >> 4;
+            int fout16 = (L[4 * l + 3] + 15) >> 4;
+            max_tiles = std::max(max_tiles, m16 * fout16);
+            max_tiles = std::max(max_tiles, m16 * fin16);
+            max_tiles = std::max(max_tiles, fout16 * fin16);
+        }
+    }
+    int threads = (max_tiles >= 8) ? 512 : 256;
     for (int t = 0; t < r.delta; ++t) {
         int s0 = r.snap_tptr[t], s1 = r.snap_tptr[t + 1];
         if (s1 > s0)
@@ -2503,8 +2583,8 @@ void run_round_mlp(
             a.ptr = r.recv_nptr + r0;
             a.dslots = r.del_slots;
             a.rslots = r.reply_slots;
-            hipLaunchKernelGGL(tick_mlp_kernel, dim3(r1 - r0), dim3(256),
-                               smem, s, a);
+            hipLaunchKernelGGL(tick_mlp_kernel, dim3(r1 - r0),
+                               dim3(threads), smem, s, a);
         }
         int p0 = r.pull_tptr[t], p1 = r.pull_tptr[t + 1];
         if (p1 > p0)
@@ -2516,8 +2596,8 @@ void run_round_mlp(
             a.ptr = r.rep_nptr + q0;
             a.dslots = r.rep_slots;
             a.rslots = rep_rr;
-            hipLaunchKernelGGL(tick_mlp_kernel, dim3(q1 - q0), dim3(256),
-                               smem, s, a);
+            hipLaunchKernelGGL(tick_mlp_kernel, dim3(q1 - q0),
+                               dim3(threads), smem, s, a);
         }
     }
 }
