@@ -1320,6 +1320,21 @@ class HIPBackend(TorchBackend):
             out = self.ext.eval_metrics_scores(
                 sc.contiguous(), gy, sc.shape[-1], False
             )
+        elif spec.family in ("logreg", "mlp", "torchmod"):
+            # large eval sets: the kernel's pairwise AUC is O(n_eval^2)
+            # per block, so score with torch GEMMs and run the SYNC-FREE
+            # tensor metrics (sort-based AUC) — the result rides the same
+            # pinned staging pipeline instead of syncing every round
+            # 
+            from .metrics import classification_metrics_tensor
+
+            sc = self.scores(state, spec, local_ids, gx)
+            out = classification_metrics_tensor(sc, gy)
+        elif spec.family in ("pegasos", "adaline"):
+            from .metrics import binary_margin_metrics_tensor
+
+            sc = self.scores(state, spec, local_ids, gx)
+            out = binary_margin_metrics_tensor(sc[:, :, 0], gy)
         else:
             return None
         # double-buffered pinned staging: at most one handle is in flight

@@ -7,6 +7,13 @@ node's model in one batched pass: scores come from the backend
 with tensor ops on the same device, and only the tiny per-node metric dict
 list returns to the host. In multi-GPU runs each rank evaluates its
 resident sample and rank 0 gathers the dicts.
+
+The ``*_tensor`` variants are the SYNC-FREE cores: every op stays on the
+scores' device and the result is one ``[R, 5]`` float32 tensor
+(accuracy, macro precision/recall/F1, AUC or -1) in the same layout the
+K13 kernel emits — so large eval sets can ride the runner's lag-1 pinned
+staging pipeline instead of syncing the stream mid-round (VERDICT r1
+weak #6). The dict-returning wrappers keep the original host interface.
 """
 
 from __future__ import annotations
@@ -15,7 +22,13 @@ from typing import Dict, List
 
 import torch
 
-__all__ = ["classification_metrics_shared", "binary_margin_metrics"]
+__all__ = [
+    "classification_metrics_shared",
+    "classification_metrics_tensor",
+    "binary_margin_metrics",
+    "binary_margin_metrics_tensor",
+    "metric_rows_to_dicts",
+]
 
 
 def _macro_prf(conf: torch.Tensor):
@@ -34,7 +47,8 @@ def _macro_prf(conf: torch.Tensor):
 def _rank_auc(scores: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
     """Per-node ROC-AUC of positive-class scores ``[R, n]`` against shared
     binary labels ``[n]`` via the Mann-Whitney rank statistic (average ranks
-    for ties)."""
+    for ties). Sync-free: degenerate label sets (all one class) resolve to
+    0.5 via a device-side ``where`` instead of a host branch."""
     R, n = scores.shape
     dev = scores.device
     order = scores.argsort(dim=1)
@@ -44,24 +58,67 @@ def _rank_auc(scores: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
     new_grp = torch.ones(R, n, dtype=torch.bool, device=dev)
     new_grp[:, 1:] = ss[:, 1:] != ss[:, :-1]
     gid = new_grp.to(torch.int64).cumsum(dim=1) - 1  # [R, n]
-    pos = (
+    pos_rank = (
         torch.arange(1, n + 1, dtype=torch.float64, device=dev)
         .expand(R, n)
     )
     gsum = torch.zeros(R, n, dtype=torch.float64, device=dev)
-    gsum.scatter_add_(1, gid, pos)
+    gsum.scatter_add_(1, gid, pos_rank)
     gcnt = torch.zeros(R, n, dtype=torch.float64, device=dev)
-    gcnt.scatter_add_(1, gid, torch.ones_like(pos))
+    gcnt.scatter_add_(1, gid, torch.ones_like(pos_rank))
     avg = (gsum / gcnt.clamp(min=1)).gather(1, gid)  # sorted-order ranks
     ranks = torch.empty(R, n, dtype=scores.dtype, device=dev)
     ranks.scatter_(1, order, avg.to(scores.dtype))
-    pos = y > 0.5 if y.max() <= 1 else y > 0
-    npos = int(pos.sum())
+    # binary positives: {0,1} labels -> y > 0.5; {-1,+1} margins arrive
+    # pre-mapped to {0,1}, so the same threshold covers both
+    pos = (y > 0.5).to(scores.dtype)
+    npos = pos.sum()
     nneg = n - npos
-    if npos == 0 or nneg == 0:
-        return torch.full((R,), 0.5, device=scores.device)
-    rank_sum = ranks[:, pos].sum(dim=1)
-    return (rank_sum - npos * (npos + 1) / 2) / (npos * nneg)
+    rank_sum = (ranks * pos.unsqueeze(0)).sum(dim=1)
+    auc = (rank_sum - npos * (npos + 1) / 2) / (npos * nneg).clamp(min=1)
+    valid = (npos > 0) & (nneg > 0)
+    return torch.where(valid, auc, torch.full_like(auc, 0.5))
+
+
+def classification_metrics_tensor(
+    scores: torch.Tensor, y: torch.Tensor, with_auc: bool = True
+) -> torch.Tensor:
+    """``[R, 5]`` device metrics (accuracy, macro P/R/F1, AUC or -1) of R
+    node models on one shared eval set — no host synchronization."""
+    R, n, k = scores.shape
+    y = y.long()
+    pred = scores.argmax(dim=2)  # [R, n]
+    acc = (pred == y.unsqueeze(0)).float().mean(dim=1)
+    oh_pred = torch.nn.functional.one_hot(pred, k).to(scores.dtype)  # [R,n,k]
+    oh_true = torch.nn.functional.one_hot(y, k).to(scores.dtype)  # [n,k]
+    conf = torch.einsum("nt,rnp->rtp", oh_true, oh_pred)  # [R, k, k]
+    prec, rec, f1 = _macro_prf(conf)
+    if with_auc and k == 2:
+        auc = _rank_auc(scores[:, :, 1], y.to(scores.dtype))
+    else:
+        auc = torch.full((R,), -1.0, device=scores.device)
+    return torch.stack(
+        [acc, prec.float(), rec.float(), f1.float(), auc.float()], dim=1
+    )
+
+
+def metric_rows_to_dicts(rows) -> List[Dict[str, float]]:
+    """Host side: ``[R, 5]`` rows (tensor or ndarray) -> metric dicts;
+    column 4 < 0 means 'no AUC' (same sentinel as the K13 kernel)."""
+    if isinstance(rows, torch.Tensor):
+        rows = rows.cpu().numpy()
+    out = []
+    for row in rows:
+        d = {
+            "accuracy": float(row[0]),
+            "precision": float(row[1]),
+            "recall": float(row[2]),
+            "f1_score": float(row[3]),
+        }
+        if row[4] >= 0:
+            d["auc"] = float(row[4])
+        out.append(d)
+    return out
 
 
 def classification_metrics_shared(
@@ -73,50 +130,23 @@ def classification_metrics_shared(
     Returns one dict per node (accuracy, macro precision/recall/F1, AUC for
     binary k=2 — matching gossipy/model/handler.py:282-334).
     """
-    R, n, k = scores.shape
-    y = y.long()
-    pred = scores.argmax(dim=2)  # [R, n]
-    acc = (pred == y.unsqueeze(0)).float().mean(dim=1)
-    oh_pred = torch.nn.functional.one_hot(pred, k).to(scores.dtype)  # [R,n,k]
-    oh_true = torch.nn.functional.one_hot(y, k).to(scores.dtype)  # [n,k]
-    conf = torch.einsum("nt,rnp->rtp", oh_true, oh_pred)  # [R, k, k]
-    prec, rec, f1 = _macro_prf(conf)
-    out = []
-    auc = None
-    if with_auc and k == 2:
-        classes = torch.unique(y)
-        if len(classes) == 2:
-            auc = _rank_auc(scores[:, :, 1], y.to(scores.dtype))
-        else:
-            auc = torch.full((R,), 0.5, device=scores.device)
-    accs, precs, recs, f1s = acc.tolist(), prec.tolist(), rec.tolist(), f1.tolist()
-    aucs = auc.tolist() if auc is not None else None
-    for r in range(R):
-        d = {
-            "accuracy": accs[r],
-            "precision": precs[r],
-            "recall": recs[r],
-            "f1_score": f1s[r],
-        }
-        if aucs is not None:
-            d["auc"] = aucs[r]
-        out.append(d)
-    return out
+    return metric_rows_to_dicts(
+        classification_metrics_tensor(scores, y, with_auc)
+    )
 
 
-def binary_margin_metrics(
+def binary_margin_metrics_tensor(
     margins: torch.Tensor, y: torch.Tensor
-) -> List[Dict[str, float]]:
-    """Metrics for margin models (AdaLine/Pegasos): predictions are
-    ``sign(margin)`` in {-1,+1}, labels are ±1
-    (gossipy/model/handler.py:375-391)."""
+) -> torch.Tensor:
+    """``[R, 5]`` device metrics for margin models (AdaLine/Pegasos):
+    predictions are ``sign(margin)`` in {-1,+1}, labels are ±1
+    (gossipy/model/handler.py:375-391). No host synchronization."""
     R, n = margins.shape[:2]
     m = margins.reshape(R, n)
     y = y.reshape(n)
     pred = torch.where(m >= 0, 1.0, -1.0)
     acc = (pred == y.unsqueeze(0)).float().mean(dim=1)
     # 2-class macro PRF over classes {-1, +1}
-    conf = torch.zeros(R, 2, 2, device=m.device)
     yt = (y > 0).long()  # 0 = class -1, 1 = class +1
     pt = (pred > 0).long()
     oh_true = torch.nn.functional.one_hot(yt, 2).float()
@@ -124,15 +154,14 @@ def binary_margin_metrics(
     conf = torch.einsum("nt,rnp->rtp", oh_true, oh_pred)
     prec, rec, f1 = _macro_prf(conf)
     auc = _rank_auc(m, (y > 0).to(m.dtype))
-    out = []
-    for r in range(R):
-        out.append(
-            {
-                "accuracy": float(acc[r]),
-                "precision": float(prec[r]),
-                "recall": float(rec[r]),
-                "f1_score": float(f1[r]),
-                "auc": float(auc[r]),
-            }
-        )
-    return out
+    return torch.stack(
+        [acc, prec.float(), rec.float(), f1.float(), auc.float()], dim=1
+    )
+
+
+def binary_margin_metrics(
+    margins: torch.Tensor, y: torch.Tensor
+) -> List[Dict[str, float]]:
+    """Dict interface over :func:`binary_margin_metrics_tensor` (margin
+    models always report AUC)."""
+    return metric_rows_to_dicts(binary_margin_metrics_tensor(margins, y))
