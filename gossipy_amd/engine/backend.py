@@ -1105,17 +1105,18 @@ class TorchBackend:
             z = torch.einsum("nd,rkd->rnk", X, W) + b.unsqueeze(1)
             return torch.sigmoid(z)
         if spec.family == "mlp":
-            outs = []
-            for r in nodes.tolist():
-                h = X
-                for (w_off, b_off, fin, fout) in spec.layer_offsets():
-                    W = state.params[r, w_off:b_off].view(fout, fin)
-                    bb = state.params[r, b_off : b_off + fout]
-                    h = h @ W.t() + bb
-                    if b_off + fout < spec.D:
-                        h = torch.relu(h)
-                outs.append(h)
-            return torch.stack(outs)
+            # node-batched forward: one bmm per layer over all R models
+            # (the per-node loop cost 2 small GEMMs per node per round)
+            rows = state.params[nodes]  # [R, D]
+            R = rows.shape[0]
+            h = X.unsqueeze(0).expand(R, *X.shape)  # [R, n, d_in]
+            for (w_off, b_off, fin, fout) in spec.layer_offsets():
+                W = rows[:, w_off:b_off].view(R, fout, fin)
+                bb = rows[:, b_off : b_off + fout]
+                h = torch.baddbmm(bb.unsqueeze(1), h, W.transpose(1, 2))
+                if b_off + fout < spec.D:
+                    h = torch.relu(h)
+            return h
         raise ValueError(spec.family)
 
 
