@@ -1264,10 +1264,14 @@ tick_mlp_kernel(MlpArgs a)
     int tid = threadIdx.x;
     extern __shared__ float sm[];
     float* W = sm;          // D
-    float* W2 = W + a.D;    // D
+    // W2 (the second model slab) exists only for UPDATE_MERGE — dropping
+    // it elsewhere takes the flagship MLP block from 94 KB to 70 KB of
+    // LDS, which fits TWO workgroups per CU (4 waves/SIMD instead of 2)
+    bool has_w2 = (a.mode == MODE_UPDATE_MERGE);
+    float* W2 = has_w2 ? W + a.D : nullptr;
     int bsmax = (a.bs == 0) ? a.Smax : min(a.bs, a.Smax);
     // act: batch input + every layer's activation; grad: 2 ping-pong buffers
-    float* act = W2 + a.D;
+    float* act = W + (has_w2 ? 2 : 1) * (size_t)a.D;
     int act_total = 0;
     // act layout computed on host side == bs*(d_in + sum(out_l)); the host
     // passes act_max = max layer width for the grad buffers
@@ -2036,8 +2040,9 @@ void tick_mlp(torch::Tensor params, torch::Tensor ages, torch::Tensor slots,
     a.act_max = act_max;
     a.d_in = L[2];
     int bsmax = (bs == 0) ? a.Smax : std::min<int>(bs, a.Smax);
+    size_t w_slabs = (mode == MODE_UPDATE_MERGE) ? 2 : 1;  // W2 only there
     size_t smem = sizeof(float) *
-        (2 * (size_t)a.D + (size_t)bsmax * (a.d_in + act_sum) +
+        (w_slabs * (size_t)a.D + (size_t)bsmax * (a.d_in + act_sum) +
          2 * (size_t)bsmax * act_max);
     TORCH_CHECK(smem <= 160 * 1024,
         "mlp LDS budget exceeded (", smem, " B); shrink batch_size/hidden");
@@ -2553,8 +2558,9 @@ void run_round_mlp(
     a.act_max = act_max;
     a.d_in = L[2];
     int bsmax = (bs == 0) ? a.Smax : std::min<int>(bs, a.Smax);
+    size_t w_slabs = (mode == MODE_UPDATE_MERGE) ? 2 : 1;  // W2 only there
     size_t smem = sizeof(float) *
-        (2 * (size_t)a.D + (size_t)bsmax * (a.d_in + act_sum) +
+        (w_slabs * (size_t)a.D + (size_t)bsmax * (a.d_in + act_sum) +
          2 * (size_t)bsmax * act_max);
     TORCH_CHECK(smem <= 160 * 1024, "mlp LDS budget exceeded");
     // 8 waves when any GEMM stage has >= 8 MFMA tiles (same policy as
