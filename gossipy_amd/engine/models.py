@@ -280,10 +280,12 @@ class KMeansSpec:
     and merges never touch it.
 
     ``matching="hungarian"`` (bug-fixed column-index matching, see
-    gossipy_amd/model/handler.py) runs on the torch backend only — the
-    reference's own hungarian path is a no-op identity permutation
-    (handler.py:629-630), so the HIP kernel implements ``naive``, the
-    behavior the reference actually ships.
+    gossipy_amd/model/handler.py — the reference's own hungarian path is
+    a no-op identity permutation, handler.py:629-630) runs on BOTH
+    backends: the torch loop, and on the engine a wave-batched device
+    cdist + host assignment + batched matched-mean feeding the HIP EMA
+    update (HIPBackend._deliver_kmeans_hungarian). ``naive`` is the
+    behavior the reference actually ships and stays fully in-kernel.
     """
 
     k: int
