@@ -18,6 +18,11 @@ _SCHED = None
 
 
 def extension_path() -> str:
+    # GOSSIPY_HIP_SO: alternate build of the same extension (e.g. the
+    # device-AddressSanitizer build the sanitizer CI pass loads)
+    override = os.environ.get("GOSSIPY_HIP_SO")
+    if override:
+        return override
     return os.path.join(os.path.dirname(__file__), "_gossip_hip.so")
 
 

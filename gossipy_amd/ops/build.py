@@ -36,6 +36,37 @@ def build_sched(verbose: bool = False) -> str:
     return target
 
 
+def build_asan(verbose: bool = False) -> str:
+    """Device-AddressSanitizer build of the kernel extension (gfx950 with
+    xnack+ page migration, which device ASAN requires). Loaded via
+    ``GOSSIPY_HIP_SO`` + ``HSA_XNACK=1`` by the sanitizer pass over the
+    GPU kernel tests (SURVEY.md §5 race-detection/sanitizer gap)."""
+    os.environ["PYTORCH_ROCM_ARCH"] = "gfx950:xnack+"
+    from torch.utils import cpp_extension
+
+    here = os.path.dirname(os.path.abspath(__file__))
+    src = os.path.join(here, "hip", "gossip_kernels.hip")
+    build_dir = os.path.join(here, "hip", "build_asan")
+    os.makedirs(build_dir, exist_ok=True)
+    cpp_extension.load(
+        name="_gossip_hip_asan",
+        sources=[src],
+        build_directory=build_dir,
+        extra_cuda_cflags=[
+            "-O1", "--offload-arch=gfx950:xnack+",
+            "-Xarch_device", "-fsanitize=address", "-fgpu-sanitize",
+        ],
+        verbose=verbose,
+        is_python_module=False,
+        with_cuda=True,
+    )
+    built = os.path.join(build_dir, "_gossip_hip_asan.so")
+    target = os.path.join(here, "_gossip_hip_asan.so")
+    shutil.copy2(built, target)
+    os.environ["PYTORCH_ROCM_ARCH"] = "gfx950"
+    return target
+
+
 def build(verbose: bool = False) -> str:
     os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
     os.environ.setdefault("MAX_JOBS", "8")
