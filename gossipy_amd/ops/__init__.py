@@ -63,7 +63,12 @@ def load_extension():
             "Run `python -m gossipy_amd.ops.build` (needs hipcc; "
             "cross-compiles for gfx950 without a GPU)." % path
         )
-    spec = importlib.util.spec_from_file_location("gossipy_amd.ops._gossip_hip", path)
+    # module name must match the .so's PyInit_<name> (the ASAN build
+    # exports PyInit__gossip_hip_asan)
+    modname = os.path.splitext(os.path.basename(path))[0]
+    spec = importlib.util.spec_from_file_location(
+        "gossipy_amd.ops." + modname, path
+    )
     mod = importlib.util.module_from_spec(spec)
     spec.loader.exec_module(mod)
     _EXT = mod
