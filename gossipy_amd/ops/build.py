@@ -41,29 +41,50 @@ def build_asan(verbose: bool = False) -> str:
     xnack+ page migration, which device ASAN requires). Loaded via
     ``GOSSIPY_HIP_SO`` + ``HSA_XNACK=1`` by the sanitizer pass over the
     GPU kernel tests (SURVEY.md §5 race-detection/sanitizer gap)."""
-    os.environ["PYTORCH_ROCM_ARCH"] = "gfx950:xnack+"
-    from torch.utils import cpp_extension
+    import subprocess
+
+    import torch
+    from torch.utils import cpp_extension as ce
 
     here = os.path.dirname(os.path.abspath(__file__))
     src = os.path.join(here, "hip", "gossip_kernels.hip")
     build_dir = os.path.join(here, "hip", "build_asan")
     os.makedirs(build_dir, exist_ok=True)
-    cpp_extension.load(
-        name="_gossip_hip_asan",
-        sources=[src],
-        build_directory=build_dir,
-        extra_cuda_cflags=[
-            "-O1", "--offload-arch=gfx950:xnack+",
-            "-Xarch_device", "-fsanitize=address", "-fgpu-sanitize",
-        ],
-        verbose=verbose,
-        is_python_module=False,
-        with_cuda=True,
+    torch_lib = os.path.join(os.path.dirname(torch.__file__), "lib")
+    includes = ce.include_paths(device_type="cuda")
+    import sysconfig
+
+    # host AND device instrumented (clang ASAN, shared runtime): the
+    # device-asan runtime reports through the host one, which the test
+    # process preloads (LD_PRELOAD=libclang_rt.asan-x86_64.so
+    # ASAN_OPTIONS=detect_leaks=0). hipcc drives BOTH compile and link so
+    # clang's sanitizer runtime is the one linked (g++ would pull GCC's).
+    obj = os.path.join(build_dir, "gossip_kernels_asan.o")
+    out = os.path.join(build_dir, "_gossip_hip_asan.so")
+    common = [
+        "-DWITH_HIP", "-DTORCH_EXTENSION_NAME=_gossip_hip_asan",
+        "-DTORCH_API_INCLUDE_EXTENSION_H", "-D__HIP_PLATFORM_AMD__=1",
+        "-DUSE_ROCM=1", "-DHIPBLAS_V2", "-fPIC", "-DCUDA_HAS_FP16=1",
+        "-D__HIP_NO_HALF_OPERATORS__=1", "-D__HIP_NO_HALF_CONVERSIONS__=1",
+        "-DHIP_ENABLE_WARP_SYNC_BUILTINS=1", "-std=c++17",
+        "--offload-arch=gfx950:xnack+", "-fno-gpu-rdc", "-O1",
+        "-fsanitize=address", "-fgpu-sanitize", "-shared-libsan",
+    ]
+    inc = [f"-isystem{p}" for p in includes] + [
+        f"-isystem{sysconfig.get_paths()['include']}"
+    ]
+    subprocess.run(
+        ["hipcc", *common, *inc, "-c", src, "-o", obj], check=True,
+        capture_output=not verbose,
     )
-    built = os.path.join(build_dir, "_gossip_hip_asan.so")
+    subprocess.run(
+        ["hipcc", "-shared", *common, obj, f"-L{torch_lib}", "-lc10",
+         "-lc10_hip", "-ltorch_cpu", "-ltorch_hip", "-ltorch",
+         "-ltorch_python", "-L/opt/rocm/lib", "-lamdhip64", "-o", out],
+        check=True, capture_output=not verbose,
+    )
     target = os.path.join(here, "_gossip_hip_asan.so")
-    shutil.copy2(built, target)
-    os.environ["PYTORCH_ROCM_ARCH"] = "gfx950"
+    shutil.copy2(out, target)
     return target
 
 
