@@ -27,6 +27,9 @@ def extension_path() -> str:
 
 
 def sched_path() -> str:
+    override = os.environ.get("GOSSIPY_SCHED_SO")
+    if override:
+        return override
     return os.path.join(os.path.dirname(__file__), "_gossip_sched.so")
 
 
@@ -40,7 +43,10 @@ def load_sched():
     path = sched_path()
     if not os.path.exists(path):
         return None
-    spec = importlib.util.spec_from_file_location("gossipy_amd.ops._gossip_sched", path)
+    modname = os.path.splitext(os.path.basename(path))[0]
+    spec = importlib.util.spec_from_file_location(
+        "gossipy_amd.ops." + modname, path
+    )
     mod = importlib.util.module_from_spec(spec)
     spec.loader.exec_module(mod)
     _SCHED = mod

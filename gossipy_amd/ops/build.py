@@ -36,6 +36,36 @@ def build_sched(verbose: bool = False) -> str:
     return target
 
 
+def build_sched_asan(verbose: bool = False) -> str:
+    """AddressSanitizer build of the native C++ scheduler (pure CPU).
+    The sanitizer pass runs the scheduler parity + fuzz suites under it:
+    ``LD_PRELOAD=$(g++ -print-file-name=libasan.so) ASAN_OPTIONS=detect_leaks=0
+    GOSSIPY_SCHED_SO=.../_gossip_sched_asan.so pytest tests/test_native_sched.py``."""
+    import subprocess
+    import sysconfig
+
+    import pybind11
+
+    here = os.path.dirname(os.path.abspath(__file__))
+    src = os.path.join(here, "..", "csrc", "scheduler.cpp")
+    target = os.path.join(here, "_gossip_sched_asan.so")
+    # direct g++ (the scheduler is pure pybind11, no torch): building via
+    # cpp_extension would dlopen the result immediately, which an
+    # ASAN-linked .so refuses without the runtime preloaded
+    cmd = [
+        "g++", "-O1", "-g", "-std=c++17", "-shared", "-fPIC",
+        "-fsanitize=address", "-fno-omit-frame-pointer",
+        # PYBIND11_MODULE(_gossip_sched, m) -> rename the token so the
+        # PyInit_ symbol matches this .so's filename
+        "-D_gossip_sched=_gossip_sched_asan",
+        f"-I{pybind11.get_include()}",
+        f"-I{sysconfig.get_paths()['include']}",
+        src, "-o", target,
+    ]
+    subprocess.run(cmd, check=True, capture_output=not verbose)
+    return target
+
+
 def build_asan(verbose: bool = False) -> str:
     """Device-AddressSanitizer build of the kernel extension (gfx950 with
     xnack+ page migration, which device ASAN requires). Loaded via
