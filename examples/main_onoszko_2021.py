@@ -42,6 +42,10 @@ def main():
     ap.add_argument("--engine-cnn", action="store_true",
                     help="batched engine with the reference CIFAR10Net CNN"
                          " (TorchModuleSpec; convs on MIOpen)")
+    ap.add_argument("--pens", action="store_true",
+                    help="with --engine-cnn: run the paper's PENS protocol"
+                         " (vmap-scored candidates, top-m merges) instead"
+                         " of plain merge-update gossip")
     args = ap.parse_args()
 
     if args.engine_cnn:
@@ -186,12 +190,24 @@ def main_engine_cnn(args):
         n_nodes=n, delta=10, protocol=AntiEntropyProtocol.PUSH,
         model_size=spec.D, sampling_eval=0.25, seed=42,
     )
-    sim = BatchedGossipSimulator(cfg, spec, data, device=device)
+    if getattr(args, "pens", False):
+        # the paper's actual protocol: PENS neighbor selection with the CNN
+        from gossipy_amd.engine import BatchedPENSGossipSimulator
+
+        sim = BatchedPENSGossipSimulator(
+            cfg, spec, data, n_sampled=6, m_top=2,
+            step1_rounds=max(2, args.rounds // 3), device=device,
+        )
+    else:
+        sim = BatchedGossipSimulator(cfg, spec, data, device=device)
     report = SimulationReport()
     sim.add_receiver(report)
     sim.init_nodes()
     sim.start(n_rounds=args.rounds)
     print(f"final global eval: {report.get_evaluation(False)[-1][1]}")
+    if getattr(args, "pens", False):
+        n_best = sum(len(b) for b in sim.scheduler.best_nodes or [])
+        print(f"selected neighbors across nodes: {n_best}")
 
 
 if __name__ == "__main__":

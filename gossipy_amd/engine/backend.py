@@ -1019,22 +1019,20 @@ class TorchBackend:
         merge the top-m (mean including the own model,
         gossipy/model/handler.py:260-280) and bump the winner counters
         (gossipy/node.py:771-782). ``counts`` is ``[n_local, n_nodes]``
-        int32 (device-resident on the HIP backend)."""
-        assert spec.family == "logreg", "PENS engine path: logreg family"
+        int32 (device-resident on the HIP backend). logreg scores each
+        candidate with one affine map; torchmod (the Onoszko CNN) runs a
+        node-batched vmap forward over the candidate rows."""
+        assert spec.family in ("logreg", "torchmod"), (
+            "PENS engine path: logreg or torchmod family"
+        )
         ptr = pens_ptr.tolist()
-        d, k = spec.d_in, spec.n_classes
         for i, node_t in enumerate(pens_nodes.tolist()):
             c_n = int(data.counts[node_t])
             x = data.x[node_t, :c_n]
             y = data.y[node_t, :c_n].long()
             cands = [int(pens_slots[j]) for j in range(ptr[i], ptr[i + 1])]
             owners = [int(pens_owners[j]) for j in range(ptr[i], ptr[i + 1])]
-            accs = []
-            for slot in cands:
-                W = pool.slots[slot, : k * d].view(k, d)
-                b = pool.slots[slot, k * d :]
-                pred = (x @ W.t() + b).argmax(dim=1)
-                accs.append(float((pred == y).float().mean()))
+            accs = self._pens_candidate_accs(pool, spec, cands, x, y)
             # top-m by accuracy, stable in arrival order (node.py:776-777
             # sorts on -accuracy; python sort is stable)
             order = sorted(range(len(cands)), key=lambda j: -accs[j])[:m_top]
@@ -1047,6 +1045,29 @@ class TorchBackend:
             state.params[node_t] = merged / (len(order) + 1)
             state.ages[node_t] = age
             self.update(state, data, spec, torch.tensor([node_t]))
+
+    def _pens_candidate_accs(self, pool, spec, cands, x, y):
+        """Accuracy of each candidate slot's model on ``(x, y)``."""
+        if spec.family == "logreg":
+            d, k = spec.d_in, spec.n_classes
+            accs = []
+            for slot in cands:
+                W = pool.slots[slot, : k * d].view(k, d)
+                b = pool.slots[slot, k * d :]
+                pred = (x @ W.t() + b).argmax(dim=1)
+                accs.append(float((pred == y).float().mean()))
+            return accs
+        # torchmod: one vmap forward over the m candidate rows
+        from types import SimpleNamespace
+
+        slots_t = torch.tensor(cands, dtype=torch.long,
+                               device=pool.slots.device)
+        state = SimpleNamespace(params=pool.slots)
+        sc = self.torchmod_scores(
+            state, spec, slots_t, x
+        )  # [m, c, k]
+        pred = sc.argmax(dim=2)
+        return (pred == y.unsqueeze(0)).float().mean(dim=1).tolist()
 
     # -- all2all weighted merge ----------------------------------------------
 
@@ -1292,9 +1313,18 @@ class HIPBackend(TorchBackend):
         self, state, pool, data, spec, pens_nodes, pens_ptr, pens_slots,
         pens_owners, counts, m_top: int,
     ) -> None:
-        assert spec.family == "logreg", "PENS engine path: logreg family"
         if len(pens_nodes) == 0:
             return
+        if spec.family == "torchmod":
+            # CNN PENS (the actual Onoszko protocol): vmap candidate
+            # scoring + batched trajectories through the python backend —
+            # counts may live on device, the loop indexes it directly
+            TorchBackend.deliver_pens(
+                self, state, pool, data, spec, pens_nodes, pens_ptr,
+                pens_slots, pens_owners, counts, m_top,
+            )
+            return
+        assert spec.family == "logreg", "PENS engine path: logreg family"
         dev = state.params.device
         self.ext.tick_pens(
             state.params,

@@ -1836,7 +1836,9 @@ class BatchedPENSGossipSimulator(BatchedGossipSimulator):
     winner counts accumulated device-side; the only host<->device
     dependency is ONE counts read at the step boundary, where
     ``best_nodes`` is computed and the scheduler's step-2 peer draws are
-    restricted to it. PUSH + MERGE_UPDATE + logreg family.
+    restricted to it. PUSH + MERGE_UPDATE; logreg scores candidates
+    in-kernel (tick_pens), torchmod (the Onoszko CIFAR10Net CNN) scores
+    them with a node-batched vmap forward.
     """
 
     def __init__(
