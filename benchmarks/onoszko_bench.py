@@ -83,7 +83,10 @@ def main():
         (X[i * per : (i + 1) * per], y[i * per : (i + 1) * per])
         for i in range(n)
     ]
-    data = DataArena.from_shards(shards, device, global_eval=(X, y))
+    # fixed-size global eval set (the 100-node config's 4000 samples) so
+    # the eval sweep doesn't scale quadratically with the population
+    ev = rng.permutation(len(labels))[:4000]
+    data = DataArena.from_shards(shards, device, global_eval=(X[ev], y[ev]))
     spec = TorchModuleSpec(
         cifar10net_factory, input_shape=(3, 32, 32), lr=0.1, batch_size=32
     )

@@ -1080,6 +1080,7 @@ DEV_INLINE void mlp_update(const MlpArgs& a, int node, float* W,
                     f32x4_t acc = {0.f, 0.f, 0.f, 0.f};
                     f32x4_t acc2 = {0.f, 0.f, 0.f, 0.f};
                     int k0 = 0;
+#pragma unroll 2
                     for (; k0 + 8 <= fin; k0 += 8) {
                         int q = k0 + fk, q2 = q + 4;
                         float av = (tr + fr < m) ? cur[(tr + fr) * fin + q]
@@ -1156,6 +1157,7 @@ DEV_INLINE void mlp_update(const MlpArgs& a, int node, float* W,
                         f32x4_t acc = {0.f, 0.f, 0.f, 0.f};
                         f32x4_t acc2 = {0.f, 0.f, 0.f, 0.f};
                         int k0 = 0;
+#pragma unroll 2
                         for (; k0 + 8 <= fout; k0 += 8) {
                             int o = k0 + fk, o2 = o + 4;
                             float av = (tr + fr < m)
@@ -1204,6 +1206,7 @@ DEV_INLINE void mlp_update(const MlpArgs& a, int node, float* W,
                         f32x4_t acc = {0.f, 0.f, 0.f, 0.f};
                         f32x4_t acc2 = {0.f, 0.f, 0.f, 0.f};
                         int k0 = 0;
+#pragma unroll 2
                         for (; k0 + 8 <= m; k0 += 8) {
                             int sidx = k0 + fk, sidx2 = sidx + 4;
                             float av = (tr + fr < fout)
