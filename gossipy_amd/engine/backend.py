@@ -1412,14 +1412,20 @@ class HIPBackend(TorchBackend):
         ):
             return compute(ids_dev)
         R = int(ids_dev.shape[0])
-        key = ("evalg", spec.family, R, tuple(gx.shape))
+        # the sampled-node count varies round to round — pad to a
+        # power-of-two bucket so a handful of graphs serve every round
+        # (keying on exact R re-captured EVERY round at 50k nodes).
+        # Padded rows re-score a stale valid id; the caller slices [:R].
+        Rp = self._pow2_bucket(R)
+        key = ("evalg", spec.family, Rp, tuple(gx.shape))
         cache = getattr(self, "_eval_graphs", None)
         if cache is None:
             cache = self._eval_graphs = {}
         entry = cache.get(key, "miss")
         if entry == "miss":
             try:
-                ids_static = ids_dev.clone()
+                ids_static = torch.zeros(Rp, dtype=torch.long, device=dev)
+                ids_static[:R] = ids_dev
                 stream = torch.cuda.Stream()
                 stream.wait_stream(torch.cuda.current_stream())
                 with torch.cuda.stream(stream):
@@ -1436,7 +1442,7 @@ class HIPBackend(TorchBackend):
         if entry is None:
             return compute(ids_dev)
         graph, ids_static, out_static = entry
-        ids_static.copy_(ids_dev)
+        ids_static[:R] = ids_dev
         graph.replay()
         return out_static
 
@@ -1490,6 +1496,9 @@ class HIPBackend(TorchBackend):
             out = self._eval_metrics_torch(state, spec, local_ids, gx, gy)
         else:
             return None
+        # the torch path may return POW2-PADDED rows (graph-bucketed);
+        # only the first len(local_ids) are real
+        valid = min(len(local_ids), out.shape[0])
         # double-buffered pinned staging: at most one handle is in flight
         # per output shape. The flip lives WITH the buffer pair (slot 2):
         # a single instance-global toggle would let two interleaving
@@ -1503,16 +1512,16 @@ class HIPBackend(TorchBackend):
         bufs[flip].copy_(out, non_blocking=True)
         ev = torch.cuda.Event()
         ev.record()
-        return (bufs[flip], ev)
+        return (bufs[flip], ev, valid)
 
     @staticmethod
     def eval_metrics_collect(handle):
         """Blocking half: wait for the staged copy and build metric dicts
         (same values as :meth:`eval_metrics_fast`)."""
-        buf, ev = handle
+        buf, ev, valid = handle
         ev.synchronize()
         res = []
-        for row in buf.numpy():
+        for row in buf.numpy()[:valid]:
             d = {
                 "accuracy": float(row[0]),
                 "precision": float(row[1]),
