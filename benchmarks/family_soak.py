@@ -113,6 +113,55 @@ run("kmeans", BatchedGossipSimulator(
     KMeansSpec(k=4, dim=16, mode=CreateModelMode.MERGE_UPDATE),
     km_data, device=dev), 200)
 
+run("kmeans hungarian", BatchedGossipSimulator(
+    EngineConfig(protocol=AntiEntropyProtocol.PUSH, n_nodes=N, delta=10,
+                 model_size=64, sampling_eval=0.0, seed=6),
+    KMeansSpec(k=4, dim=16, mode=CreateModelMode.MERGE_UPDATE,
+               matching="hungarian"),
+    km_data, device=dev), 150)
+
+
+# torchmod CNN (vmap-batched + hipGraph trajectories) — r2 paths
+def cnn_factory():
+    import torch.nn as nn
+    import torch.nn.functional as F
+
+    class Tiny(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.conv1 = nn.Conv2d(3, 8, 3)
+            self.pool = nn.MaxPool2d(2, 2)
+            self.fc = nn.Linear(8 * 15 * 15, 10)
+
+        def forward(self, x):
+            x = self.pool(F.relu(self.conv1(x)))
+            return self.fc(x.view(-1, 8 * 15 * 15))
+
+    return Tiny()
+
+
+from gossipy_amd.engine import TorchModuleSpec  # noqa: E402
+
+cnn_rng = np.random.default_rng(11)
+lbl = cnn_rng.integers(0, 10, size=32 * 12)
+cx = cnn_rng.normal(0, 0.3, size=(len(lbl), 3, 32, 32)).astype(np.float32)
+CX = torch.from_numpy(cx.reshape(len(lbl), -1))
+CY = torch.from_numpy(lbl).float()
+cnn_shards = [(CX[i * 12 : (i + 1) * 12], CY[i * 12 : (i + 1) * 12]) for i in range(32)]
+cnn_data = DataArena.from_shards(cnn_shards, dev, global_eval=(CX, CY))
+cnn_spec = TorchModuleSpec(cnn_factory, input_shape=(3, 32, 32), lr=0.05,
+                           batch_size=6)
+run("torchmod cnn", BatchedGossipSimulator(
+    EngineConfig(protocol=AntiEntropyProtocol.PUSH, n_nodes=32, delta=10,
+                 model_size=cnn_spec.D, sampling_eval=0.1, seed=13),
+    cnn_spec, cnn_data, device=dev), 120)
+run("pens cnn", BatchedPENSGossipSimulator(
+    EngineConfig(protocol=AntiEntropyProtocol.PUSH, n_nodes=32, delta=10,
+                 model_size=cnn_spec.D, sampling_eval=0.0, seed=14),
+    TorchModuleSpec(cnn_factory, input_shape=(3, 32, 32), lr=0.05,
+                    batch_size=6),
+    cnn_data, n_sampled=4, m_top=2, step1_rounds=40, device=dev), 100)
+
 if dev.type == "cuda":
     print(f"memory allocated: {torch.cuda.memory_allocated()/1e6:.1f} MB")
 print("ALL FAMILIES STABLE")
