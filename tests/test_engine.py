@@ -1515,6 +1515,82 @@ def test_tokenized_two_rank_matches_single():
     assert np.allclose(ref.local_params().numpy(), got, atol=1e-6)
 
 
+_FUZZ_CONFIGS = [
+    # (protocol, n_parts, drop, online, delay_max, seed)
+    (AntiEntropyProtocol.PUSH, 0, 0.0, 1.0, 0, 101),
+    (AntiEntropyProtocol.PUSH_PULL, 0, 0.2, 0.9, 3, 102),
+    (AntiEntropyProtocol.PULL, 0, 0.0, 1.0, 2, 103),
+    (AntiEntropyProtocol.PUSH, 4, 0.1, 0.85, 5, 104),
+    (AntiEntropyProtocol.PUSH_PULL, 4, 0.15, 1.0, 0, 105),
+    (AntiEntropyProtocol.PUSH, 0, 0.3, 0.7, 7, 106),
+]
+
+
+def _fuzz_sim(ci, rank, world):
+    proto, n_parts, drop, online, dmax, seed = _FUZZ_CONFIGS[ci]
+    shards, geval = _make_data(24, seed=ci)
+    if world == 1:
+        data = DataArena.from_shards(
+            shards, torch.device("cpu"), global_eval=geval
+        )
+    else:
+        data = _arena_for_rank(shards, geval, rank, world)
+    cfg = EngineConfig(
+        n_nodes=24, delta=8, protocol=proto, model_size=116,
+        sampling_eval=0.0, seed=seed, n_parts=n_parts,
+        drop_prob=drop, online_prob=online,
+        delay=UniformDelay(0, dmax) if dmax else ConstantDelay(0),
+    )
+    spec = LogRegSpec(d_in=57, n_classes=2, lr=0.1, n_parts=n_parts)
+    sim = BatchedGossipSimulator(cfg, spec, data, device=torch.device("cpu"))
+    sim.init_nodes()
+    sim.start(n_rounds=4)
+    return sim
+
+
+def _fuzz_worker(rank, world, port, q):
+    import torch.distributed as dist
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        outs = []
+        for ci in range(len(_FUZZ_CONFIGS)):
+            sim = _fuzz_sim(ci, rank, world)
+            full = sim.gather_params()
+            if rank == 0:
+                outs.append(full.numpy())
+        if rank == 0:
+            q.put(outs)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_four_rank_fuzz_matches_single():
+    """Six random protocol/fault configs, each bit-exact between a
+    single-rank run and a 4-rank gloo run (rank-count fuzzing ahead of
+    the 8-GPU SCALE run)."""
+    singles = [
+        _fuzz_sim(ci, 0, 1).local_params().numpy()
+        for ci in range(len(_FUZZ_CONFIGS))
+    ]
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [
+        ctx.Process(target=_fuzz_worker, args=(r, 4, 29583, q))
+        for r in range(4)
+    ]
+    for p in procs:
+        p.start()
+    multis = q.get(timeout=500)
+    for p in procs:
+        p.join(timeout=120)
+    for ci, (a, b) in enumerate(zip(singles, multis)):
+        assert np.array_equal(a, b), f"config {ci} diverged at world=4"
+
+
 @pytest.mark.timeout(600)
 def test_tokenized_eight_rank_matches_single():
     """8-rank rehearsal of the native tokenized scheduler + packed
