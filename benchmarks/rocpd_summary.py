@@ -83,5 +83,42 @@ def main(path):
               f"{100.0*tot/total:5.1f}")
 
 
+def pmc(path):
+    """Aggregate PMC counter sums per kernel (rocprofv3 --pmc runs)."""
+    db = find_db(path)
+    con = sqlite3.connect(db)
+    cur = con.cursor()
+    tables = [r[0] for r in cur.execute(
+        "SELECT name FROM sqlite_master WHERE type IN ('table','view')"
+    )]
+    pe = next((t for t in tables if "pmc_event" in t), None)
+    pi = next((t for t in tables if "info_pmc" in t), None)
+    kd = next((t for t in tables if "kernel_dispatch" in t), None)
+    ki = next((t for t in tables if "kernel" in t and "info" in t), None)
+    if not (pe and pi and kd and ki):
+        print("pmc tables missing:", tables)
+        return
+    for t in (pe, pi):
+        cols = [r[1] for r in cur.execute(f"PRAGMA table_info({t})")]
+        print(t, cols)
+    q = f"""
+        SELECT ki.kernel_name, ip.name, SUM(p.value), COUNT(*)
+        FROM {pe} p
+        JOIN {kd} kd ON p.event_id = kd.event_id
+        JOIN {ki} ki ON kd.kernel_id = ki.id
+        JOIN {pi} ip ON p.pmc_id = ip.id
+        GROUP BY ki.kernel_name, ip.name ORDER BY 3 DESC LIMIT 40
+    """
+    try:
+        for kname, cname, total, cnt in cur.execute(q):
+            print(f"{str(kname)[:56]:56s} {str(cname):34s} "
+                  f"{total:16.0f} n={cnt}")
+    except sqlite3.OperationalError as e:
+        print("query failed:", e)
+
+
 if __name__ == "__main__":
-    main(sys.argv[1] if len(sys.argv) > 1 else ".")
+    if len(sys.argv) > 2 and sys.argv[1] == "--pmc":
+        pmc(sys.argv[2])
+    else:
+        main(sys.argv[1] if len(sys.argv) > 1 else ".")
