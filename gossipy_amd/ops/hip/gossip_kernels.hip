@@ -51,6 +51,16 @@ DEV_INLINE float wave_sum(float v)
     return __shfl(v, 0, WAVE);
 }
 
+// reduction when only the first <=8 lanes carry data (e.g. the MF rank-k
+// dot, k<=8): 3 shuffle steps instead of 6, broadcast to every lane
+DEV_INLINE float wave_sum8(float v)
+{
+    v += __shfl_down(v, 4, WAVE);
+    v += __shfl_down(v, 2, WAVE);
+    v += __shfl_down(v, 1, WAVE);
+    return __shfl(v, 0, WAVE);
+}
+
 
 // ---------------------------------------------------------------------------
 // snapshot: slots[slot_ids[i]] = params[nodes[i]]  (+ age)
@@ -718,7 +728,10 @@ DEV_INLINE void mf_update(const MFArgs& a, int node, float* row, int& age)
             ci_n = row[coff + item_n];
         }
         float r = ratings[s];
-        float dot = wave_sum((lane < a.k) ? x * yi : 0.f);
+        float part = (lane < a.k) ? x * yi : 0.f;
+        // k <= 8: 3-step reduction, bit-identical to the 64-lane tree
+        // (the extra lanes only ever add zeros, so the pairings match)
+        float dot = (a.k <= 8) ? wave_sum8(part) : wave_sum(part);
         float err = r - dot - b - ci;
         float yi_new = shrink * yi + a.lr * err * x;
         float ci_new = ci + a.lr * err;
