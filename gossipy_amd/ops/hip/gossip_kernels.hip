@@ -713,71 +713,44 @@ DEV_INLINE void mf_update(const MFArgs& a, int node, float* row, int& age)
     int coff = a.item_off + a.n_items * a.k;
     float x = (lane < a.k) ? row[lane] : 0.f;
     float b = row[a.k];
-    // 8-deep prefetch ring in registers (full unroll keeps the indexing
-    // compile-time — dynamic indices would spill the ring to scratch).
-    // A one-deep prefetch cannot hide the ~700 ns item-row load behind
-    // ~80 ns of per-rating math; eight in flight can. Every write patches
-    // any pending ring entry for the same item, so re-rated items read
-    // register-fresh values — exact.
-    constexpr int PF = 8;
-    int pf_item[PF];
-    float pf_yi[PF], pf_ci[PF];
-#pragma unroll
-    for (int d = 0; d < PF; ++d) {
-        pf_item[d] = -1;
-        pf_yi[d] = 0.f;
-        pf_ci[d] = 0.f;
-        if (d < c) {
-            int it = (int)items[d];
-            pf_item[d] = it;
-            pf_yi[d] = (lane < a.k)
-                           ? row[a.item_off + (long)it * a.k + lane] : 0.f;
-            pf_ci[d] = row[coff + it];
+    int item = (int)items[0];
+    float* Yi = row + a.item_off + (long)item * a.k;
+    float yi = (lane < a.k) ? Yi[lane] : 0.f;
+    float ci = row[coff + item];
+    for (int s = 0; s < c; ++s) {
+        int item_n = 0;
+        float* Yi_n = nullptr;
+        float yi_n = 0.f, ci_n = 0.f;
+        if (s + 1 < c) {  // prefetch next item row + bias
+            item_n = (int)items[s + 1];
+            Yi_n = row + a.item_off + (long)item_n * a.k;
+            yi_n = (lane < a.k) ? Yi_n[lane] : 0.f;
+            ci_n = row[coff + item_n];
         }
-    }
-    for (int s0 = 0; s0 < c; s0 += PF) {
-#pragma unroll
-        for (int d = 0; d < PF; ++d) {
-            int s = s0 + d;
-            if (s >= c) break;
-            int item = pf_item[d];
-            float yi = pf_yi[d];
-            float ci = pf_ci[d];
-            float r = ratings[s];
-            float part = (lane < a.k) ? x * yi : 0.f;
-            // k <= 8: 3-step reduction, bit-identical to the 64-lane tree
-            // (the extra lanes only ever add zeros — pairings match)
-            float dot = (a.k <= 8) ? wave_sum8(part) : wave_sum(part);
-            float err = r - dot - b - ci;
-            float yi_new = shrink * yi + a.lr * err * x;
-            float ci_new = ci + a.lr * err;
-            if (lane < a.k) {
-                row[a.item_off + (long)item * a.k + lane] = yi_new;
-                x = shrink * x + a.lr * err * yi_new;
+        float r = ratings[s];
+        float part = (lane < a.k) ? x * yi : 0.f;
+        // k <= 8: 3-step reduction, bit-identical to the 64-lane tree
+        // (the extra lanes only ever add zeros, so the pairings match)
+        float dot = (a.k <= 8) ? wave_sum8(part) : wave_sum(part);
+        float err = r - dot - b - ci;
+        float yi_new = shrink * yi + a.lr * err * x;
+        float ci_new = ci + a.lr * err;
+        if (lane < a.k) {
+            Yi[lane] = yi_new;
+            x = shrink * x + a.lr * err * yi_new;
+        }
+        if (lane == 0) row[coff + item] = ci_new;
+        b += a.lr * err;
+        age += 1;
+        if (s + 1 < c) {
+            if (item_n == item) {  // re-rated item: fix up from registers
+                yi_n = yi_new;
+                ci_n = ci_new;
             }
-            if (lane == 0) row[coff + item] = ci_new;
-            b += a.lr * err;
-            age += 1;
-            // patch pending prefetches of the item just written
-#pragma unroll
-            for (int dd = 0; dd < PF; ++dd) {
-                if (pf_item[dd] == item) {
-                    pf_yi[dd] = yi_new;
-                    pf_ci[dd] = ci_new;
-                }
-            }
-            // refill this slot with rating s + PF
-            int sn = s + PF;
-            if (sn < c) {
-                int it = (int)items[sn];
-                pf_item[d] = it;
-                pf_yi[d] = (lane < a.k)
-                               ? row[a.item_off + (long)it * a.k + lane]
-                               : 0.f;
-                pf_ci[d] = row[coff + it];
-            } else {
-                pf_item[d] = -1;
-            }
+            item = item_n;
+            Yi = Yi_n;
+            yi = yi_n;
+            ci = ci_n;
         }
     }
     if (lane < a.k) row[lane] = x;
