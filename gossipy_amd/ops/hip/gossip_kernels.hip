@@ -1087,6 +1087,57 @@ struct MlpArgs {
 
 typedef float f32x4_t __attribute__((ext_vector_type(4)));
 
+// Dual-accumulator MFMA K-loop. For K <= MLP_KPAD the loop is PADDED to a
+// compile-time bound and fully unrolled: every LDS load issues up front
+// (independent) and the two MFMA chains run issue-bound instead of
+// load-latency-bound. Out-of-range chunks contribute exact zeros, and the
+// chunk->accumulator pairing (k0%8==0 -> acc, ==4 -> acc2) matches the
+// dynamic loop, so the result is BIT-IDENTICAL either way. LoadA/LoadB
+// take the chunk base k0 and must return 0 outside [0, Kdim).
+constexpr int MLP_KPAD = 128;
+
+template <int PAD, typename LoadA, typename LoadB>
+DEV_INLINE f32x4_t mfma_kloop_padded(LoadA la, LoadB lb)
+{
+    f32x4_t acc = {0.f, 0.f, 0.f, 0.f};
+    f32x4_t acc2 = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int k0 = 0; k0 < PAD; k0 += 8) {
+        float av = la(k0), bv = lb(k0);
+        float av2 = la(k0 + 4), bv2 = lb(k0 + 4);
+        acc = __builtin_amdgcn_mfma_f32_16x16x4f32(av, bv, acc, 0, 0, 0);
+        acc2 = __builtin_amdgcn_mfma_f32_16x16x4f32(av2, bv2, acc2, 0, 0, 0);
+    }
+    for (int r = 0; r < 4; ++r) acc[r] += acc2[r];
+    return acc;
+}
+
+template <typename LoadA, typename LoadB>
+DEV_INLINE f32x4_t mfma_kloop(int Kdim, LoadA la, LoadB lb)
+{
+    if (Kdim <= 32) return mfma_kloop_padded<32>(la, lb);
+    if (Kdim <= 64) return mfma_kloop_padded<64>(la, lb);
+    if (Kdim <= MLP_KPAD) return mfma_kloop_padded<MLP_KPAD>(la, lb);
+    f32x4_t acc = {0.f, 0.f, 0.f, 0.f};
+    f32x4_t acc2 = {0.f, 0.f, 0.f, 0.f};
+    {
+        int k0 = 0;
+        for (; k0 + 8 <= Kdim; k0 += 8) {
+            float av = la(k0), bv = lb(k0);
+            float av2 = la(k0 + 4), bv2 = lb(k0 + 4);
+            acc = __builtin_amdgcn_mfma_f32_16x16x4f32(av, bv, acc, 0, 0, 0);
+            acc2 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                av2, bv2, acc2, 0, 0, 0);
+        }
+        for (; k0 < Kdim; k0 += 4) {
+            acc = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                la(k0), lb(k0), acc, 0, 0, 0);
+        }
+    }
+    for (int r = 0; r < 4; ++r) acc[r] += acc2[r];
+    return acc;
+}
+
 DEV_INLINE void mlp_update(const MlpArgs& a, int node, float* W,
                            float* act, float* grad, int& age)
 {
@@ -1119,40 +1170,19 @@ DEV_INLINE void mlp_update(const MlpArgs& a, int node, float* W,
                 int m16 = (m + 15) >> 4, o16 = (fout + 15) >> 4;
                 for (int tile = wave; tile < m16 * o16; tile += nwaves) {
                     int tr = (tile / o16) << 4, tc = (tile % o16) << 4;
-                    // two independent accumulator chains: back-to-back
-                    // dependent MFMAs stall on the 16x16x4 pipeline; the
-                    // f32 MFMA is exact, so only the summation order
-                    // changes (tests compare vs torch at tolerance)
-                    f32x4_t acc = {0.f, 0.f, 0.f, 0.f};
-                    f32x4_t acc2 = {0.f, 0.f, 0.f, 0.f};
-                    int k0 = 0;
-#pragma unroll 2
-                    for (; k0 + 8 <= fin; k0 += 8) {
-                        int q = k0 + fk, q2 = q + 4;
-                        float av = (tr + fr < m) ? cur[(tr + fr) * fin + q]
-                                                 : 0.f;
-                        float bv = (tc + fr < fout)
-                                       ? W[w_off + (tc + fr) * fin + q] : 0.f;
-                        float av2 = (tr + fr < m && q2 < fin)
-                                        ? cur[(tr + fr) * fin + q2] : 0.f;
-                        float bv2 = (tc + fr < fout && q2 < fin)
-                                        ? W[w_off + (tc + fr) * fin + q2]
-                                        : 0.f;
-                        acc = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                            av, bv, acc, 0, 0, 0);
-                        acc2 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                            av2, bv2, acc2, 0, 0, 0);
-                    }
-                    for (; k0 < fin; k0 += 4) {
-                        int q = k0 + fk;
-                        float av = (tr + fr < m && q < fin)
+                    f32x4_t acc = mfma_kloop(
+                        fin,
+                        [&](int k0) {
+                            int q = k0 + fk;
+                            return (tr + fr < m && q < fin)
                                        ? cur[(tr + fr) * fin + q] : 0.f;
-                        float bv = (tc + fr < fout && q < fin)
-                                       ? W[w_off + (tc + fr) * fin + q] : 0.f;
-                        acc = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                            av, bv, acc, 0, 0, 0);
-                    }
-                    for (int r = 0; r < 4; ++r) acc[r] += acc2[r];
+                        },
+                        [&](int k0) {
+                            int q = k0 + fk;
+                            return (tc + fr < fout && q < fin)
+                                       ? W[w_off + (tc + fr) * fin + q]
+                                       : 0.f;
+                        });
                     for (int r = 0; r < 4; ++r) {
                         int row = tr + (fk << 2) + r, col = tc + fr;
                         if (row < m && col < fout) {
@@ -1200,38 +1230,20 @@ DEV_INLINE void mlp_update(const MlpArgs& a, int node, float* W,
                     int m16 = (m + 15) >> 4, q16 = (fin + 15) >> 4;
                     for (int tile = wave; tile < m16 * q16; tile += nwaves) {
                         int tr = (tile / q16) << 4, tc = (tile % q16) << 4;
-                        f32x4_t acc = {0.f, 0.f, 0.f, 0.f};
-                        f32x4_t acc2 = {0.f, 0.f, 0.f, 0.f};
-                        int k0 = 0;
-#pragma unroll 2
-                        for (; k0 + 8 <= fout; k0 += 8) {
-                            int o = k0 + fk, o2 = o + 4;
-                            float av = (tr + fr < m)
-                                           ? gcur[(tr + fr) * fout + o] : 0.f;
-                            float bv = (tc + fr < fin)
+                        f32x4_t acc = mfma_kloop(
+                            fout,
+                            [&](int k0) {
+                                int o = k0 + fk;
+                                return (tr + fr < m && o < fout)
+                                           ? gcur[(tr + fr) * fout + o]
+                                           : 0.f;
+                            },
+                            [&](int k0) {
+                                int o = k0 + fk;
+                                return (o < fout && tc + fr < fin)
                                            ? W[w_off + o * fin + tc + fr]
                                            : 0.f;
-                            float av2 = (tr + fr < m)
-                                            ? gcur[(tr + fr) * fout + o2]
-                                            : 0.f;
-                            float bv2 = (tc + fr < fin)
-                                            ? W[w_off + o2 * fin + tc + fr]
-                                            : 0.f;
-                            acc = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                                av, bv, acc, 0, 0, 0);
-                            acc2 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                                av2, bv2, acc2, 0, 0, 0);
-                        }
-                        for (; k0 < fout; k0 += 4) {
-                            int o = k0 + fk;
-                            float av = (tr + fr < m && o < fout)
-                                           ? gcur[(tr + fr) * fout + o] : 0.f;
-                            float bv = (o < fout && tc + fr < fin)
-                                           ? W[w_off + o * fin + tc + fr] : 0.f;
-                            acc = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                                av, bv, acc, 0, 0, 0);
-                        }
-                        for (int r = 0; r < 4; ++r) acc[r] += acc2[r];
+                            });
                         for (int r = 0; r < 4; ++r) {
                             int row = tr + (fk << 2) + r, col = tc + fr;
                             if (row < m && col < fin) {
@@ -1249,38 +1261,20 @@ DEV_INLINE void mlp_update(const MlpArgs& a, int node, float* W,
                     int o16 = (fout + 15) >> 4, q16 = (fin + 15) >> 4;
                     for (int tile = wave; tile < o16 * q16; tile += nwaves) {
                         int tr = (tile / q16) << 4, tc = (tile % q16) << 4;
-                        f32x4_t acc = {0.f, 0.f, 0.f, 0.f};
-                        f32x4_t acc2 = {0.f, 0.f, 0.f, 0.f};
-                        int k0 = 0;
-#pragma unroll 2
-                        for (; k0 + 8 <= m; k0 += 8) {
-                            int sidx = k0 + fk, sidx2 = sidx + 4;
-                            float av = (tr + fr < fout)
+                        f32x4_t acc = mfma_kloop(
+                            m,
+                            [&](int k0) {
+                                int sidx = k0 + fk;
+                                return (tr + fr < fout && sidx < m)
                                            ? gcur[sidx * fout + tr + fr]
                                            : 0.f;
-                            float bv = (tc + fr < fin)
-                                           ? ain[sidx * fin + tc + fr] : 0.f;
-                            float av2 = (tr + fr < fout)
-                                            ? gcur[sidx2 * fout + tr + fr]
-                                            : 0.f;
-                            float bv2 = (tc + fr < fin)
-                                            ? ain[sidx2 * fin + tc + fr]
-                                            : 0.f;
-                            acc = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                                av, bv, acc, 0, 0, 0);
-                            acc2 = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                                av2, bv2, acc2, 0, 0, 0);
-                        }
-                        for (; k0 < m; k0 += 4) {
-                            int sidx = k0 + fk;
-                            float av = (tr + fr < fout && sidx < m)
-                                           ? gcur[sidx * fout + tr + fr] : 0.f;
-                            float bv = (sidx < m && tc + fr < fin)
-                                           ? ain[sidx * fin + tc + fr] : 0.f;
-                            acc = __builtin_amdgcn_mfma_f32_16x16x4f32(
-                                av, bv, acc, 0, 0, 0);
-                        }
-                        for (int r = 0; r < 4; ++r) acc[r] += acc2[r];
+                            },
+                            [&](int k0) {
+                                int sidx = k0 + fk;
+                                return (sidx < m && tc + fr < fin)
+                                           ? ain[sidx * fin + tc + fr]
+                                           : 0.f;
+                            });
                         for (int r = 0; r < 4; ++r) {
                             int row = tr + (fk << 2) + r, col = tc + fr;
                             if (row < fout && col < fin) {
