@@ -609,6 +609,35 @@ class TestKMeansGPU:
         assert _close(cs.params, gs.params, 1e-4)
         assert torch.equal(cs.ages.cpu(), gs.ages.cpu())
 
+    def test_cnn_graphs_match_eager(self, monkeypatch):
+        """hipGraph-captured SGD trajectories + eval graphs must produce
+        the same results as the eager path (GOSSIPY_NO_GRAPH=1)."""
+        from gossipy_amd.engine import TorchModuleSpec
+        from tests.test_engine import _cifar10net, _cnn_data
+
+        def run(no_graph):
+            if no_graph:
+                monkeypatch.setenv("GOSSIPY_NO_GRAPH", "1")
+            else:
+                monkeypatch.delenv("GOSSIPY_NO_GRAPH", raising=False)
+            spec = TorchModuleSpec(
+                _cifar10net, input_shape=(3, 32, 32), lr=0.1, batch_size=8
+            )
+            data = _cnn_data(device=CUDA)
+            cfg = EngineConfig(
+                n_nodes=8, delta=5, protocol=AntiEntropyProtocol.PUSH,
+                model_size=spec.D, sampling_eval=0.25, seed=41,
+            )
+            sim = BatchedGossipSimulator(cfg, spec, data, device=CUDA)
+            sim.init_nodes()
+            sim.start(n_rounds=3)
+            torch.cuda.synchronize()
+            return sim.local_params().cpu()
+
+        a = run(False)
+        b = run(True)
+        assert torch.allclose(a, b, atol=1e-4)
+
     def test_kmeans_hungarian_deliver_matches_oracle(self):
         """K12 engine path: wave-batched Hungarian
         merge on device == the bug-fixed torch-backend loop."""
