@@ -397,6 +397,16 @@ class TorchModuleSpec:
         self._numels = [int(np.prod(s)) for s in self._shapes]
         self._D = int(sum(self._numels))
 
+    def __getstate__(self):
+        # checkpoints dill the spec; the lazily-built template module
+        # (possibly holding device tensors) is derived state — drop it
+        st = dict(self.__dict__)
+        st.pop("_template", None)
+        return st
+
+    def __setstate__(self, st):
+        self.__dict__.update(st)
+
     def param_layout(self):
         """``(name, shape, offset, numel)`` per parameter tensor, in
         ``parameters()`` order — the arena row layout."""

@@ -916,6 +916,37 @@ class TestEngineCheckpoint:
         sim.init_nodes()
         return sim
 
+    def test_save_load_torchmod(self, tmp_path):
+        """CNN-family checkpoints resume bit-exactly (arena rows +
+        deterministic scheduler replay; the vmap/graph paths hold no
+        hidden state)."""
+        from gossipy_amd.engine import TorchModuleSpec
+
+        def build():
+            spec = TorchModuleSpec(
+                _cifar10net, input_shape=(3, 32, 32), lr=0.1, batch_size=8
+            )
+            data = _cnn_data()
+            cfg = EngineConfig(
+                n_nodes=8, delta=5, protocol=AntiEntropyProtocol.PUSH,
+                model_size=spec.D, sampling_eval=0.0, seed=41,
+            )
+            sim = BatchedGossipSimulator(cfg, spec, data)
+            sim.init_nodes()
+            return sim
+
+        ref = build()
+        ref.start(n_rounds=4)
+
+        sim = build()
+        sim.start(n_rounds=2)
+        f = str(tmp_path / "ckpt_cnn.dill")
+        sim.save(f)
+        restored = BatchedGossipSimulator.load(f, device=torch.device("cpu"))
+        restored.start(n_rounds=2)
+        assert torch.equal(ref.local_params(), restored.local_params())
+        assert torch.equal(ref.state.ages, restored.state.ages)
+
     @pytest.mark.parametrize("save_at", [2, 3, 5])
     def test_save_load_pens(self, tmp_path, save_at):
         """PENS resume is bit-exact whether the checkpoint lands before,

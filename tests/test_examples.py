@@ -32,3 +32,26 @@ def test_example_runs(script):
         timeout=280,
     )
     assert r.returncode == 0, f"{script} failed:\n{r.stdout}\n{r.stderr}"
+
+
+@pytest.mark.parametrize(
+    "extra",
+    [
+        ["--engine"],
+        ["--engine-cnn"],
+        ["--engine-cnn", "--pens"],
+    ],
+    ids=["engine", "engine-cnn", "engine-cnn-pens"],
+)
+@pytest.mark.timeout(600)
+def test_onoszko_engine_variants(extra):
+    """The Onoszko example's engine paths (incl. the paper's PENS+CNN
+    pairing) run end-to-end at a small scale."""
+    r = subprocess.run(
+        [sys.executable, "examples/main_onoszko_2021.py",
+         "--nodes", "8", "--rounds", "3", *extra],
+        capture_output=True,
+        text=True,
+        timeout=560,
+    )
+    assert r.returncode == 0, f"{extra} failed:\n{r.stdout}\n{r.stderr}"
