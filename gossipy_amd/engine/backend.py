@@ -639,6 +639,7 @@ class TorchBackend:
             cache[key] = (graph, srows, sx, sy)
         except Exception:
             cache[key] = None  # capture unsupported here — stay eager
+            torch.cuda.synchronize()  # clear any aborted-capture state
         return cache[key]
 
     def torchmod_scores(self, state, spec, nodes, X) -> torch.Tensor:
@@ -1417,6 +1418,11 @@ class HIPBackend(TorchBackend):
         # (keying on exact R re-captured EVERY round at 50k nodes).
         # Padded rows re-score a stale valid id; the caller slices [:R].
         Rp = self._pow2_bucket(R)
+        # skip capture for huge score matrices: padding waste is real
+        # memory, and an OOM INSIDE a capture invalidates the stream for
+        # the eager fallback too (hit at 1M nodes: Rp=16384 x 46k eval)
+        if Rp * int(gx.shape[0]) > 32_000_000:
+            return compute(ids_dev)
         key = ("evalg", spec.family, Rp, tuple(gx.shape))
         cache = getattr(self, "_eval_graphs", None)
         if cache is None:
@@ -1438,6 +1444,7 @@ class HIPBackend(TorchBackend):
                 entry = (graph, ids_static, out_static)
             except Exception:
                 entry = None  # capture unsupported — stay eager
+                torch.cuda.synchronize()  # clear any aborted-capture state
             cache[key] = entry
         if entry is None:
             return compute(ids_dev)
