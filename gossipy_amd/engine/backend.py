@@ -40,6 +40,16 @@ from typing import Optional
 # per-shape solver search runs once per bucket and then hits the cache —
 # FAST (immediate) mode was measured 2x slower steady-state on the
 # grouped convs (2.9 vs 5.9 ms per captured SGD trajectory, gfx950).
+#
+# A pre-tuned MIOpen user DB for the CNN configs' grouped-conv shapes
+# (MIOPEN_FIND_ENFORCE=SEARCH on an MI355X; +12% on the Onoszko bench,
+# 28.1 -> 31.5 r/s) ships in-tree and is used when present. The DB file
+# is keyed by arch + MIOpen version, so a mismatched stack just ignores
+# it; an explicit MIOPEN_USER_DB_PATH always wins.
+_udb = os.path.join(os.path.dirname(__file__), "..", "ops", "miopen_udb")
+if os.path.isdir(_udb):
+    os.environ.setdefault("MIOPEN_USER_DB_PATH", os.path.abspath(_udb))
+del _udb
 
 import numpy as np
 import torch
