@@ -1930,6 +1930,71 @@ def _cnn_worker(rank, world, port, q):
         dist.destroy_process_group()
 
 
+@pytest.mark.timeout(600)
+def test_cnn_four_rank_matches_single():
+    """The wave-batched torchmod deliver under the vectorized multi-rank
+    round path at 4 ranks (2 nodes/rank — uneven wave batches)."""
+    from gossipy_amd.engine import TorchModuleSpec
+
+    spec = TorchModuleSpec(
+        _cifar10net, input_shape=(3, 32, 32), lr=0.1, batch_size=0
+    )
+    data = _cnn_data()
+    cfg = EngineConfig(
+        n_nodes=8, delta=5, protocol=AntiEntropyProtocol.PUSH,
+        model_size=spec.D, sampling_eval=0.0, seed=41,
+    )
+    ref = BatchedGossipSimulator(cfg, spec, data, device=torch.device("cpu"))
+    ref.init_nodes()
+    ref.start(n_rounds=2)
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [
+        ctx.Process(target=_cnn_worker4, args=(r, 4, 29591, q))
+        for r in range(4)
+    ]
+    for p in procs:
+        p.start()
+    got = q.get(timeout=500)
+    for p in procs:
+        p.join(timeout=120)
+    assert np.allclose(ref.local_params().numpy(), got, atol=1e-5)
+
+
+def _cnn_worker4(rank, world, port, q):
+    import torch.distributed as dist
+
+    from gossipy_amd.engine import TorchModuleSpec
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        spec = TorchModuleSpec(
+            _cifar10net, input_shape=(3, 32, 32), lr=0.1, batch_size=0
+        )
+        full = _cnn_data()
+        per = 8 // world
+        lo = rank * per
+        data = DataArena(
+            full.x[lo : lo + per], full.y[lo : lo + per],
+            full.counts[lo : lo + per], gx=full.gx, gy=full.gy,
+        )
+        cfg = EngineConfig(
+            n_nodes=8, delta=5, protocol=AntiEntropyProtocol.PUSH,
+            model_size=spec.D, sampling_eval=0.0, seed=41,
+        )
+        sim = BatchedGossipSimulator(cfg, spec, data, device=torch.device("cpu"))
+        sim.init_nodes()
+        sim.start(n_rounds=2)
+        out = sim.gather_params()
+        if rank == 0:
+            q.put(out.numpy())
+    finally:
+        dist.destroy_process_group()
+
+
 @pytest.mark.timeout(300)
 def test_cnn_two_rank_matches_single():
     from gossipy_amd.engine import TorchModuleSpec
