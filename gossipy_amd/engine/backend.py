@@ -624,6 +624,10 @@ class TorchBackend:
         key = (id(spec), Bp, c, bs, epochs)
         if key in cache:
             return cache[key]
+        if len(cache) >= 64:
+            # very ragged shard lengths could otherwise accumulate one
+            # captured graph (with static buffers) per distinct shape
+            return None
         lr, wd = spec.lr, spec.weight_decay
         try:
             srows = params.new_zeros(Bp, params.shape[1])
