@@ -401,6 +401,11 @@ struct LogregPartArgs {
     int P, d, k, Smax, D;
     float lr, wd;
     int epochs, bs, mode, update_only;
+    //: xb holds the WHOLE shard (staged once per receiver row) instead of
+    //: one re-staged minibatch slice per step per delivery — decided
+    //: host-side from the LDS budget. The reference math is unchanged;
+    //: the staging was the tokenized row's repeated HBM round trip.
+    int stage_full;
 };
 
 // Minibatch SGD with whole-age-vector increment per batch and per-partition
@@ -417,8 +422,11 @@ DEV_INLINE void logreg_part_update(const LogregPartArgs& a, int node, float* W,
     for (int ep = 0; ep < a.epochs; ++ep) {
         for (int s0 = 0; s0 < c; s0 += bsz) {
             int m = min(bsz, c - s0);
-            for (int e = tid; e < m * a.d; e += blockDim.x)
-                xb[e] = Xn[(long)s0 * a.d + e];
+            float* xs = a.stage_full ? xb + (long)s0 * a.d : xb;
+            if (!a.stage_full) {
+                for (int e = tid; e < m * a.d; e += blockDim.x)
+                    xb[e] = Xn[(long)s0 * a.d + e];
+            }
             // self.n_updates += 1 happens before the step (handler.py:506)
             for (int p = tid; p < a.P; p += blockDim.x) agev[p] += 1;
             __syncthreads();
@@ -427,7 +435,7 @@ DEV_INLINE void logreg_part_update(const LogregPartArgs& a, int node, float* W,
                 for (int kk = 0; kk < a.k; ++kk) {
                     float acc = W[a.k * a.d + kk];
                     const float* wrow = W + kk * a.d;
-                    const float* xrow = xb + tid * a.d;
+                    const float* xrow = xs + tid * a.d;
                     for (int dd = 0; dd < a.d; ++dd) acc += wrow[dd] * xrow[dd];
                     z[kk] = acc;
                 }
@@ -453,7 +461,7 @@ DEV_INLINE void logreg_part_update(const LogregPartArgs& a, int node, float* W,
             for (int e = tid; e < a.k * a.d; e += blockDim.x) {
                 int kk = e / a.d, dd = e - kk * a.d;
                 float g = 0.f;
-                for (int s = 0; s < m; ++s) g += dz[s * a.k + kk] * xb[s * a.d + dd];
+                for (int s = 0; s < m; ++s) g += dz[s * a.k + kk] * xs[s * a.d + dd];
                 g /= (float)agev[a.apart[e]];  // _adjust_gradient
                 if (a.wd != 0.f) g += a.wd * W[e];
                 W[e] -= a.lr * g;
@@ -502,9 +510,10 @@ DEV_INLINE void logreg_part_process_node(const LogregPartArgs& a, int i)
     extern __shared__ float sm[];
     float* W = sm;                         // D
     float* W2 = W + a.D;                   // D (received-model scratch)
-    float* xb = W2 + a.D;                  // bsmax*d
+    float* xb = W2 + a.D;                  // (stage_full ? Smax : bsmax)*d
     int bsmax = (a.bs == 0) ? a.Smax : min(a.bs, a.Smax);
-    float* dz = xb + bsmax * a.d;          // bsmax*k
+    int xbn = a.stage_full ? a.Smax : bsmax;
+    float* dz = xb + xbn * a.d;            // bsmax*k
     int* agev = (int*)(dz + bsmax * a.k);  // P
     int* agev2 = agev + a.P;               // P
 
@@ -512,6 +521,14 @@ DEV_INLINE void logreg_part_process_node(const LogregPartArgs& a, int i)
         W[e] = a.params[(long)node * a.D + e];
     for (int p = tid; p < a.P; p += blockDim.x)
         agev[p] = a.ages[(long)node * a.P + p];
+    if (a.stage_full) {
+        // whole shard -> LDS once; every delivery's every minibatch step
+        // slices it (the re-staging was a repeated HBM round trip on the
+        // tokenized row's sequential chains)
+        int c = a.counts[node];
+        const float* Xn = a.X + (long)node * a.Smax * a.d;
+        for (int e = tid; e < c * a.d; e += blockDim.x) xb[e] = Xn[e];
+    }
     __syncthreads();
 
     if (a.update_only) {
@@ -1979,9 +1996,17 @@ void tick_logreg_part(
     a.lr = lr; a.wd = wd; a.epochs = epochs; a.bs = bs; a.mode = mode;
     a.update_only = update_only;
     int bsmax = (bs == 0) ? a.Smax : std::min<int>(bs, a.Smax);
-    size_t smem = sizeof(float) * (2 * a.D + (size_t)bsmax * a.d +
+    // whole-shard staging when it fits a 64 KB block budget (keeps >=2
+    // blocks/CU); otherwise per-step slices as before
+    size_t full = sizeof(float) * (2 * a.D + (size_t)a.Smax * a.d +
                                    (size_t)bsmax * a.k) +
                   sizeof(int) * 2 * a.P;
+    a.stage_full = (full <= 64 * 1024) ? 1 : 0;
+    size_t smem = a.stage_full
+        ? full
+        : sizeof(float) * (2 * a.D + (size_t)bsmax * a.d +
+                           (size_t)bsmax * a.k) +
+              sizeof(int) * 2 * a.P;
     TORCH_CHECK(smem <= 160 * 1024, "partitioned logreg LDS budget exceeded: ", smem);
     hipLaunchKernelGGL(tick_logreg_part_kernel, dim3(n), dim3(128), smem,
                        current_stream(), a);
@@ -2493,9 +2518,15 @@ void run_round_logreg_part(
     a.lr = lr; a.wd = wd; a.epochs = epochs; a.bs = bs; a.mode = mode;
     a.update_only = 0;
     int bsmax = (bs == 0) ? a.Smax : std::min<int>(bs, a.Smax);
-    size_t smem = sizeof(float) * (2 * a.D + (size_t)bsmax * a.d +
+    size_t full = sizeof(float) * (2 * a.D + (size_t)a.Smax * a.d +
                                    (size_t)bsmax * a.k) +
                   sizeof(int) * 2 * a.P;
+    a.stage_full = (full <= 64 * 1024) ? 1 : 0;
+    size_t smem = a.stage_full
+        ? full
+        : sizeof(float) * (2 * a.D + (size_t)bsmax * a.d +
+                           (size_t)bsmax * a.k) +
+              sizeof(int) * 2 * a.P;
     TORCH_CHECK(smem <= 160 * 1024, "partitioned logreg LDS budget exceeded");
     if (max_group(r) <= sb_threshold() && rep_rr == nullptr) {
         auto dev = params.device();
